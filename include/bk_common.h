@@ -1,0 +1,148 @@
+// bk_common.h — shared plain-C descriptors for the MI355X-native BaikalDB OLAP
+// hot-path engine ("bkgpu") and its CPU oracle.
+//
+// These structs cross the C-ABI boundary (include/bkgpu.h) and are also the
+// vocabulary of the CPU oracle (oracle/oracle.c). No C++/torch types here.
+//
+// Type tags mirror baidu/BaikalDB proto/common.proto:46-72 (pb::PrimitiveType)
+// so that a baikalStore host embedding this engine can pass its own enum
+// values through unchanged.
+#ifndef BK_COMMON_H
+#define BK_COMMON_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- pb::PrimitiveType (proto/common.proto:46) ---- */
+typedef enum BkType {
+    BK_INVALID_TYPE = 0,
+    BK_NULL_TYPE    = 1,
+    BK_BOOL         = 2,
+    BK_INT8         = 3,
+    BK_INT16        = 4,
+    BK_INT32        = 5,
+    BK_INT64        = 6,
+    BK_UINT8        = 7,
+    BK_UINT16       = 8,
+    BK_UINT32       = 9,
+    BK_UINT64       = 10,
+    BK_FLOAT        = 11,
+    BK_DOUBLE       = 12,
+    BK_STRING       = 13,   /* stored dict-encoded: int32 codes + host dict */
+} BkType;
+
+/* ---- pb::PlanNodeType subset (proto/plan.proto:10-23) ---- */
+typedef enum BkNodeType {
+    BK_SCAN_NODE         = 1,
+    BK_SORT_NODE         = 2,
+    BK_AGG_NODE          = 4,
+    BK_MERGE_AGG_NODE    = 5,
+    BK_TABLE_FILTER_NODE = 6,
+    BK_LIMIT_NODE        = 11,
+    BK_WHERE_FILTER_NODE = 12,
+} BkNodeType;
+
+/* ---- comparison ops of src/expr/operators.cpp:79-105 (eq/ne/gt/ge/lt/le) ---- */
+typedef enum BkCmpOp {
+    BK_OP_EQ = 0,
+    BK_OP_NE = 1,
+    BK_OP_GT = 2,
+    BK_OP_GE = 3,
+    BK_OP_LT = 4,
+    BK_OP_LE = 5,
+} BkCmpOp;
+
+/* ---- AggFnCall::AggType subset (include/expr/agg_fn_call.h:52-58) ---- */
+typedef enum BkAggType {
+    BK_AGG_COUNT_STAR = 0,
+    BK_AGG_COUNT      = 1,
+    BK_AGG_SUM        = 2,
+    BK_AGG_AVG        = 3,
+    BK_AGG_MIN        = 4,
+    BK_AGG_MAX        = 5,
+} BkAggType;
+
+/* ---- synthetic column distributions (bench configs) ---- */
+typedef enum BkDist {
+    BK_DIST_UNIFORM_I64 = 0,  /* uniform integer in [p0, p1) */
+    BK_DIST_CUBESKEW    = 1,  /* integer-only skewed ("Zipf-shaped") in [0, p0) */
+    BK_DIST_DICT        = 2,  /* dict code uniform in [0, p0) (VARCHAR via dict) */
+    BK_DIST_SUMU16      = 3,  /* approx N(0,1) double: sum of 4 u16 minus mean, scaled */
+} BkDist;
+
+/* One generated column. Physical storage by type:
+ *   BK_INT64  -> int64_t[nrows]
+ *   BK_DOUBLE -> double[nrows]
+ *   BK_STRING -> int32_t[nrows] dict codes (dict strings generated on host)
+ * Optional validity: uint8_t[nrows], 1 = present, 0 = SQL NULL. A column with
+ * null_frac_x1e6 == 0 has no validity array (all rows valid). */
+typedef struct BkColSpec {
+    int32_t col_type;        /* BkType */
+    int32_t dist;            /* BkDist */
+    int64_t p0, p1;          /* distribution params */
+    int32_t null_frac_x1e6;  /* NULL fraction in parts-per-million */
+    int32_t _pad;
+} BkColSpec;
+
+/* One WHERE conjunct: <col> <op> <literal>, evaluated with the reference's
+ * SQL ternary NULL logic (NULL operand => conjunct NULL => row rejected,
+ * src/exec/filter_node.cpp:726-734) and the reference's arg-cast rule
+ * (src/expr/scalar_fn_call.cpp:219-225: both args cast to the fn arg type).
+ * cmp_type selects the typed comparator of src/expr/operators.cpp:79-105:
+ *   BK_INT64  -> int64 compare, lit_i
+ *   BK_DOUBLE -> double compare (int64 col cast to double), lit_d
+ *   BK_STRING -> dict-code equality (EQ/NE only; dict codes are unique per
+ *                string so code equality == string equality)
+ */
+typedef struct BkConjunct {
+    int32_t col;
+    int32_t op;        /* BkCmpOp */
+    int32_t cmp_type;  /* BkType */
+    int32_t _pad;
+    int64_t lit_i;
+    double  lit_d;
+} BkConjunct;
+
+/* One aggregate call (reference: src/expr/agg_fn_call.cpp:496-555 update,
+ * 719-830 merge, 927-975 finalize). col == -1 for COUNT(*). */
+typedef struct BkAggSpec {
+    int32_t agg_type;  /* BkAggType */
+    int32_t col;       /* input column, -1 for COUNT_STAR */
+} BkAggSpec;
+
+/* ORDER BY key (reference: include/mem_row/mem_row_compare.h:23-45). */
+typedef struct BkOrderSpec {
+    int32_t col;
+    int32_t is_asc;        /* 1 asc, 0 desc */
+    int32_t is_null_first; /* NULLs first? */
+    int32_t _pad;
+} BkOrderSpec;
+
+/* Limits chosen for the hot path (north_star queries use <=2 group cols,
+ * <=3 conjuncts, <=4 aggregates; we allow a little headroom). */
+#define BK_MAX_COLS      16
+#define BK_MAX_CONJUNCTS 8
+#define BK_MAX_GROUP     2
+#define BK_MAX_AGGS      8
+
+/* A full query descriptor over one columnar table: the pb::Plan subset the
+ * store receives for the SELECT pipeline (SCAN -> FILTER -> AGG), flattened. */
+typedef struct BkQuerySpec {
+    int32_t    n_conjuncts;
+    int32_t    n_group;
+    int32_t    n_aggs;
+    int32_t    _pad;
+    BkConjunct conjuncts[BK_MAX_CONJUNCTS];
+    int32_t    group_cols[BK_MAX_GROUP];
+    int32_t    group_types[BK_MAX_GROUP];  /* BkType of each group col */
+    BkAggSpec  aggs[BK_MAX_AGGS];
+    int32_t    agg_in_types[BK_MAX_AGGS];  /* BkType of each agg input col */
+} BkQuerySpec;
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* BK_COMMON_H */

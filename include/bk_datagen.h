@@ -1,0 +1,93 @@
+// bk_datagen.h — deterministic synthetic table generator, shared verbatim by
+// the CPU oracle (gcc) and the GPU kernels (hipcc device code), so that both
+// sides see bit-identical inputs for any (seed, row, column).
+//
+// Replaces the data source below the drop-in boundary (the reference's
+// RocksDB scan, src/engine/table_iterator.cpp:446-660, is OUT of scope —
+// synthetic columnar generators stand in for the scan).
+//
+// Only integer ops and IEEE-754 double multiplies of exact constants are used
+// (no libm), so CPU and gfx950 results are bit-identical by construction.
+#ifndef BK_DATAGEN_H
+#define BK_DATAGEN_H
+
+#include <stdint.h>
+#include "bk_common.h"
+
+#if defined(__HIPCC__) || defined(__HIP_DEVICE_COMPILE__)
+#define BK_HD __host__ __device__ static inline
+#else
+#define BK_HD static inline
+#endif
+
+/* splitmix64 (public-domain PRNG finalizer): stateless per (seed,row,col). */
+BK_HD uint64_t bk_mix64(uint64_t x) {
+    x += 0x9E3779B97F4A7C15ull;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+    return x ^ (x >> 31);
+}
+
+BK_HD uint64_t bk_cell_bits(uint64_t seed, uint64_t row, uint32_t col) {
+    return bk_mix64(seed ^ bk_mix64(row ^ ((uint64_t)(col + 1) << 56)));
+}
+
+/* NULL decision: separate stream from the value stream. */
+BK_HD int bk_cell_valid(uint64_t seed, uint64_t row, uint32_t col,
+                        int32_t null_frac_x1e6) {
+    if (null_frac_x1e6 <= 0) return 1;
+    uint64_t u = bk_mix64(seed ^ 0xA5A5A5A5ull ^ bk_cell_bits(seed, row, col + 97));
+    return (int)((u % 1000000ull) >= (uint64_t)null_frac_x1e6);
+}
+
+/* uniform integer in [lo, hi) */
+BK_HD int64_t bk_gen_uniform_i64(uint64_t u, int64_t lo, int64_t hi) {
+    uint64_t span = (uint64_t)(hi - lo);
+    return lo + (int64_t)(u % span);
+}
+
+/* integer-only skewed distribution on [0, D): density concentrated near 0
+ * ("Zipf-shaped" group keys for the GROUP BY configs; exact distribution is
+ * a knob, identity CPU==GPU is the requirement). D must be <= 2^21. */
+BK_HD int64_t bk_gen_cubeskew(uint64_t u, int64_t D) {
+    uint64_t w = u >> 43;                 /* 21 bits */
+    uint64_t t = (w * w) >> 21;           /* w^2 / 2^21, <= 2^21 */
+    uint64_t c = (t * w) >> 21;           /* w^3 / 2^42, <= 2^21 */
+    return (int64_t)(c % (uint64_t)D);
+}
+
+/* dict code uniform in [0, nwords) */
+BK_HD int32_t bk_gen_dict(uint64_t u, int64_t nwords) {
+    return (int32_t)(u % (uint64_t)nwords);
+}
+
+/* approx N(0,1) double: (sum of four u16) centered and scaled. All steps are
+ * exact int ops plus one exact int->double convert and one double multiply by
+ * a constant, so CPU and GPU agree bitwise. Var(sum of 4 u16) = 4*(65536^2-1)/12;
+ * scale = 1/sqrt(that) precomputed as a decimal literal (closest double). */
+BK_HD double bk_gen_sumu16(uint64_t u) {
+    int64_t s = (int64_t)(u & 0xFFFF) + (int64_t)((u >> 16) & 0xFFFF)
+              + (int64_t)((u >> 32) & 0xFFFF) + (int64_t)((u >> 48) & 0xFFFF);
+    /* mean = 4*65535/2 = 131070; sd = sqrt(4*(65536*65536-1)/12) = 37837.16... */
+    return (double)(s - 131070) * 2.6429099261197387e-05;
+}
+
+/* Generate one cell. Returns value through the matching out-param; the
+ * caller dispatches storage by col_type. */
+BK_HD int64_t bk_gen_i64(const BkColSpec* cs, uint64_t seed, uint64_t row, uint32_t col) {
+    uint64_t u = bk_cell_bits(seed, row, col);
+    switch ((BkDist)cs->dist) {
+        case BK_DIST_UNIFORM_I64: return bk_gen_uniform_i64(u, cs->p0, cs->p1);
+        case BK_DIST_CUBESKEW:    return bk_gen_cubeskew(u, cs->p0);
+        case BK_DIST_DICT:        return (int64_t)bk_gen_dict(u, cs->p0);
+        default:                  return 0;
+    }
+}
+
+BK_HD double bk_gen_f64(const BkColSpec* cs, uint64_t seed, uint64_t row, uint32_t col) {
+    uint64_t u = bk_cell_bits(seed, row, col);
+    (void)cs;
+    return bk_gen_sumu16(u);
+}
+
+#endif /* BK_DATAGEN_H */

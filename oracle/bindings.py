@@ -1,0 +1,224 @@
+# oracle/bindings.py — *** TEST INFRASTRUCTURE ONLY *** (see __init__.py)
+# ctypes bindings over oracle/liboracle.so.
+import ctypes as C
+import os
+
+import numpy as np
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+
+# ---- enums mirroring include/bk_common.h ----
+TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
+DIST_UNIFORM, DIST_CUBESKEW, DIST_DICT, DIST_SUMU16 = 0, 1, 2, 3
+OP_EQ, OP_NE, OP_GT, OP_GE, OP_LT, OP_LE = 0, 1, 2, 3, 4, 5
+AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_AVG, AGG_MIN, AGG_MAX = 0, 1, 2, 3, 4, 5
+
+BK_MAX_GROUP = 2
+BK_MAX_CONJ = 8
+BK_MAX_AGGS = 8
+
+
+class BkColSpec(C.Structure):
+    _fields_ = [("col_type", C.c_int32), ("dist", C.c_int32),
+                ("p0", C.c_int64), ("p1", C.c_int64),
+                ("null_frac_x1e6", C.c_int32), ("_pad", C.c_int32)]
+
+
+class BkConjunct(C.Structure):
+    _fields_ = [("col", C.c_int32), ("op", C.c_int32),
+                ("cmp_type", C.c_int32), ("_pad", C.c_int32),
+                ("lit_i", C.c_int64), ("lit_d", C.c_double)]
+
+
+class BkAggSpec(C.Structure):
+    _fields_ = [("agg_type", C.c_int32), ("col", C.c_int32)]
+
+
+class BkOrderSpec(C.Structure):
+    _fields_ = [("col", C.c_int32), ("is_asc", C.c_int32),
+                ("is_null_first", C.c_int32), ("_pad", C.c_int32)]
+
+
+class BkQuerySpec(C.Structure):
+    _fields_ = [("n_conjuncts", C.c_int32), ("n_group", C.c_int32),
+                ("n_aggs", C.c_int32), ("_pad", C.c_int32),
+                ("conjuncts", BkConjunct * BK_MAX_CONJ),
+                ("group_cols", C.c_int32 * BK_MAX_GROUP),
+                ("group_types", C.c_int32 * BK_MAX_GROUP),
+                ("aggs", BkAggSpec * BK_MAX_AGGS),
+                ("agg_in_types", C.c_int32 * BK_MAX_AGGS)]
+
+
+class _OrcCol(C.Structure):
+    _fields_ = [("type", C.c_int32), ("data", C.c_void_p), ("valid", C.c_void_p)]
+
+
+class _OrcAggResult(C.Structure):
+    _fields_ = [("ngroups", C.c_int64), ("rows_passed", C.c_int64),
+                ("key_bytes", C.POINTER(C.c_uint8)), ("key_off", C.POINTER(C.c_int64)),
+                ("out_i", C.POINTER(C.c_int64)), ("out_d", C.POINTER(C.c_double)),
+                ("out_has", C.POINTER(C.c_uint8)),
+                ("g_flag", C.POINTER(C.c_uint8)), ("g_enc", C.POINTER(C.c_uint64))]
+
+
+def make_query(conjuncts=(), group=(), aggs=(), col_types=None):
+    """Build a BkQuerySpec.
+
+    conjuncts: list of (col, op, cmp_type, literal)
+    group:     list of col indices
+    aggs:      list of (agg_type, col)  (col=-1 for COUNT(*))
+    col_types: list of BkType per table column (needed for group/agg typing)
+    """
+    q = BkQuerySpec()
+    q.n_conjuncts = len(conjuncts)
+    for i, (col, op, cmp_type, lit) in enumerate(conjuncts):
+        cj = q.conjuncts[i]
+        cj.col, cj.op, cj.cmp_type = col, op, cmp_type
+        if cmp_type == TYPE_DOUBLE:
+            cj.lit_d = float(lit)
+            cj.lit_i = 0
+        else:
+            cj.lit_i = int(lit)
+            cj.lit_d = 0.0
+    q.n_group = len(group)
+    for i, col in enumerate(group):
+        q.group_cols[i] = col
+        q.group_types[i] = col_types[col]
+    q.n_aggs = len(aggs)
+    for i, (at, col) in enumerate(aggs):
+        q.aggs[i].agg_type = at
+        q.aggs[i].col = col
+        q.agg_in_types[i] = col_types[col] if col >= 0 else TYPE_INT64
+    return q
+
+
+class Oracle:
+    def __init__(self, path=None):
+        path = path or os.path.join(_HERE, "liboracle.so")
+        self.lib = C.CDLL(path)
+        lib = self.lib
+        lib.orc_generate_column.restype = C.c_int
+        lib.orc_generate_column.argtypes = [C.POINTER(BkColSpec), C.c_uint64, C.c_uint32,
+                                            C.c_int64, C.c_int64, C.c_void_p, C.c_void_p]
+        lib.orc_filter_agg.restype = C.POINTER(_OrcAggResult)
+        lib.orc_filter_agg.argtypes = [C.POINTER(_OrcCol), C.c_int,
+                                       C.POINTER(BkQuerySpec), C.c_int64, C.c_int64,
+                                       C.c_int, C.c_uint64, C.c_int]
+        lib.orc_agg_result_free.argtypes = [C.POINTER(_OrcAggResult)]
+        lib.orc_sort_topk.restype = C.c_int64
+        lib.orc_sort_topk.argtypes = [C.POINTER(_OrcCol), C.c_int,
+                                      C.POINTER(BkQuerySpec), C.POINTER(BkOrderSpec),
+                                      C.c_int, C.c_int64, C.c_int64, C.c_int64,
+                                      C.POINTER(C.c_int64)]
+        lib.orc_dict_word.restype = C.c_int
+        lib.orc_dict_word.argtypes = [C.c_uint64, C.c_int64, C.c_char_p, C.c_int]
+        for f in ("orc_encode_i64", "orc_decode_i64"):
+            getattr(lib, f).restype = C.c_uint64 if f.endswith("encode_i64") else C.c_int64
+        lib.orc_encode_i64.argtypes = [C.c_int64]
+        lib.orc_decode_i64.argtypes = [C.c_uint64]
+        lib.orc_encode_f64.restype = C.c_uint64
+        lib.orc_encode_f64.argtypes = [C.c_double]
+        lib.orc_decode_f64.restype = C.c_double
+        lib.orc_decode_f64.argtypes = [C.c_uint64]
+        lib.orc_mix64.restype = C.c_uint64
+        lib.orc_mix64.argtypes = [C.c_uint64]
+        lib.orc_cell_bits.restype = C.c_uint64
+        lib.orc_cell_bits.argtypes = [C.c_uint64, C.c_uint64, C.c_uint32]
+
+    # ---- data generation ----
+    def generate_table(self, specs, nrows, seed, row_begin=0):
+        """Generate host columns. Returns (columns, valids) lists of numpy arrays."""
+        cols, valids = [], []
+        for ci, spec in enumerate(specs):
+            if spec.col_type == TYPE_INT64:
+                arr = np.empty(nrows, dtype=np.int64)
+            elif spec.col_type == TYPE_DOUBLE:
+                arr = np.empty(nrows, dtype=np.float64)
+            elif spec.col_type == TYPE_STRING:
+                arr = np.empty(nrows, dtype=np.int32)
+            else:
+                raise ValueError(spec.col_type)
+            valid = None
+            vptr = None
+            if spec.null_frac_x1e6 > 0:
+                valid = np.empty(nrows, dtype=np.uint8)
+                vptr = valid.ctypes.data_as(C.c_void_p)
+            rc = self.lib.orc_generate_column(
+                C.byref(spec), seed, ci, row_begin, row_begin + nrows,
+                arr.ctypes.data_as(C.c_void_p), vptr)
+            assert rc == 0
+            cols.append(arr)
+            valids.append(valid)
+        return cols, valids
+
+    def _make_cols(self, cols, valids, col_types):
+        n = len(cols)
+        arr = (_OrcCol * n)()
+        for i in range(n):
+            arr[i].type = col_types[i]
+            arr[i].data = cols[i].ctypes.data_as(C.c_void_p)
+            arr[i].valid = (valids[i].ctypes.data_as(C.c_void_p)
+                            if valids and valids[i] is not None else None)
+        return arr
+
+    # ---- filter + aggregate ----
+    def filter_agg(self, cols, valids, col_types, q, nthreads=1, dict_seed=0,
+                   row_begin=0, row_end=None, sort_keys=True):
+        if row_end is None:
+            row_end = len(cols[0])
+        carr = self._make_cols(cols, valids, col_types)
+        res = self.lib.orc_filter_agg(carr, len(cols), C.byref(q),
+                                      row_begin, row_end, nthreads, dict_seed,
+                                      1 if sort_keys else 0)
+        try:
+            r = res.contents
+            ng = r.ngroups
+            na = q.n_aggs
+            out = {
+                "ngroups": ng,
+                "rows_passed": r.rows_passed,
+                "keys": [],
+                "flags": np.ctypeslib.as_array(r.g_flag, shape=(max(ng, 1),))[:ng].copy(),
+                "enc": np.ctypeslib.as_array(
+                    r.g_enc, shape=(max(ng, 1) * BK_MAX_GROUP,))[:ng * BK_MAX_GROUP]
+                    .copy().reshape(ng, BK_MAX_GROUP) if ng else np.zeros((0, 2), np.uint64),
+                "agg_i": np.ctypeslib.as_array(
+                    r.out_i, shape=(max(na * ng, 1),))[:na * ng].copy().reshape(na, ng)
+                    if ng else np.zeros((na, 0), np.int64),
+                "agg_d": np.ctypeslib.as_array(
+                    r.out_d, shape=(max(na * ng, 1),))[:na * ng].copy().reshape(na, ng)
+                    if ng else np.zeros((na, 0), np.float64),
+                "agg_has": np.ctypeslib.as_array(
+                    r.out_has, shape=(max(na * ng, 1),))[:na * ng].copy().reshape(na, ng)
+                    if ng else np.zeros((na, 0), np.uint8),
+            }
+            total = r.key_off[ng]
+            kb = bytes(bytearray(C.cast(r.key_bytes,
+                                        C.POINTER(C.c_uint8 * max(total, 1))).contents))[:total]
+            offs = [r.key_off[i] for i in range(ng + 1)]
+            out["keys"] = [kb[offs[i]:offs[i + 1]] for i in range(ng)]
+            return out
+        finally:
+            self.lib.orc_agg_result_free(res)
+
+    # ---- sort + top-N ----
+    def sort_topk(self, cols, valids, col_types, order, limit, q=None,
+                  row_begin=0, row_end=None):
+        if row_end is None:
+            row_end = len(cols[0])
+        carr = self._make_cols(cols, valids, col_types)
+        oarr = (BkOrderSpec * len(order))()
+        for i, (col, is_asc, null_first) in enumerate(order):
+            oarr[i].col, oarr[i].is_asc, oarr[i].is_null_first = col, is_asc, null_first
+        out = np.empty(max(limit, 1), dtype=np.int64)
+        if q is None:
+            q = BkQuerySpec()
+        n = self.lib.orc_sort_topk(carr, len(cols), C.byref(q), oarr, len(order),
+                                   row_begin, row_end, limit,
+                                   out.ctypes.data_as(C.POINTER(C.c_int64)))
+        return out[:n].copy()
+
+    def dict_word(self, dict_seed, code):
+        buf = C.create_string_buffer(64)
+        n = self.lib.orc_dict_word(dict_seed, code, buf, 64)
+        return buf.raw[:n].decode()

@@ -1,0 +1,676 @@
+/* oracle.c — CPU restatement of baidu/BaikalDB's OLAP row-engine hot path:
+ * the ScanNode -> FilterNode -> AggNode / SortNode pipeline plus the
+ * ExprValue/ScalarFnCall/AggFnCall evaluator semantics.
+ *
+ * *** TEST INFRASTRUCTURE ONLY ***
+ * This file is the parity ORACLE and the reported CPU baseline
+ * (bench.py cpu_baseline leg). It must never be linked into, imported by, or
+ * fallen back to from the product GPU path. Only tests/, __graft_entry__.smoke()
+ * and bench.py's cpu_baseline leg may call it.
+ *
+ * Reference sites restated (file:line against /root/reference):
+ *  - filter:      FilterNode::need_copy         src/exec/filter_node.cpp:726-734
+ *                 (row kept iff every conjunct is non-NULL and truthy)
+ *  - comparisons: operators.cpp:79-105 (typed eq/ne/gt/ge/lt/le, NULL=>NULL)
+ *                 with args cast to fn arg types, scalar_fn_call.cpp:219-225
+ *  - group key:   ExecNode::encode_exprs_key    src/exec/exec_node.cpp:555-571
+ *                 (null-flag byte + MutTableKey::append_value per value,
+ *                  mut_table_key.h:113-208, big-endian sign-flipped,
+ *                  key_encoder.h:104-173; strings: bytes + '\0')
+ *  - hash agg:    AggNode::process_row_batch    src/exec/agg_node.cpp:507-545
+ *  - agg fns:     AggFnCall::initialize/update/merge/finalize
+ *                 src/expr/agg_fn_call.cpp:370-455, 496-555, 719-830, 927-975
+ *                 (COUNT: int64 ++; SUM: starts NULL, first add adopts the
+ *                  value's own type then adds in that type (int64 wraps);
+ *                  AVG: {double sum, int64 count} pair, agg_fn_call.h:41-47;
+ *                  MIN/MAX: ExprValue::compare, expr_value.h:895-945)
+ *  - zero-row     AggNode::open                 src/exec/agg_node.cpp:490-505
+ *    aggregate:   (no GROUP BY + no rows => single all-initialized row, as the
+ *                  db-side merger produces for end-to-end SQL semantics)
+ *  - sort/top-N:  SortNode + Sorter/TopNSorter  src/exec/sort_node.cpp:278-440,
+ *                 src/runtime/sorter.cpp:18-110, include/runtime/topn_sorter.h:32-63
+ *                 (comparator: mem_row_compare.cpp:18-40 — per key: NULLs equal,
+ *                  is_null_first decides NULL order regardless of asc/desc;
+ *                  ties broken by arrival index, topn_sorter.h:46-54)
+ *
+ * Parity pinning: the JSON fixtures under tests/golden hold (a) vectors produced by the
+ * reference's own key_encoder.h compiled in-place (oracle/_ref, Makefile
+ * target 'ref'), and (b) expectation vectors re-expressed from the
+ * reference's unit tests test/test_key_encoder.cpp and test/test_expr_value.cpp.
+ *
+ * Internally the oracle groups by a fixed-width encoded key (null-flag byte +
+ * order-preserving u64 per group column — the same internal key the GPU path
+ * uses) and materializes the reference's variable-length MutTableKey byte
+ * string only when results are fetched. The mapping fixed-width -> byte key
+ * is injective for a fixed group-column type list, so group identity is
+ * exactly the reference's.
+ */
+
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <pthread.h>
+#include "../include/bk_common.h"
+#include "../include/bk_datagen.h"
+#include "../include/bk_keyenc.h"
+
+#define ORC_EXPORT __attribute__((visibility("default")))
+
+/* ------------------------------------------------------------------ */
+/* Column container                                                    */
+/* ------------------------------------------------------------------ */
+
+typedef struct OrcCol {
+    int32_t  type;      /* BkType: BK_INT64, BK_DOUBLE, BK_STRING(dict i32) */
+    void*    data;      /* int64_t* / double* / int32_t* */
+    uint8_t* valid;     /* NULL => all valid */
+} OrcCol;
+
+/* ------------------------------------------------------------------ */
+/* Synthetic generation (shared bit-exact generator, bk_datagen.h)     */
+/* ------------------------------------------------------------------ */
+
+ORC_EXPORT int orc_generate_column(const BkColSpec* cs, uint64_t seed, uint32_t col,
+                                   int64_t row_begin, int64_t row_end,
+                                   void* out_data, uint8_t* out_valid) {
+    if (cs->col_type == BK_INT64 || cs->col_type == BK_STRING) {
+        for (int64_t r = row_begin; r < row_end; r++) {
+            int64_t v = bk_gen_i64(cs, seed, (uint64_t)r, col);
+            if (cs->col_type == BK_INT64) ((int64_t*)out_data)[r - row_begin] = v;
+            else                          ((int32_t*)out_data)[r - row_begin] = (int32_t)v;
+        }
+    } else if (cs->col_type == BK_DOUBLE) {
+        for (int64_t r = row_begin; r < row_end; r++) {
+            ((double*)out_data)[r - row_begin] = bk_gen_f64(cs, seed, (uint64_t)r, col);
+        }
+    } else {
+        return -1;
+    }
+    if (out_valid) {
+        for (int64_t r = row_begin; r < row_end; r++) {
+            out_valid[r - row_begin] =
+                (uint8_t)bk_cell_valid(seed, (uint64_t)r, col, cs->null_frac_x1e6);
+        }
+    }
+    return 0;
+}
+
+/* Deterministic dict word for a code (host-side only; row data stores codes).
+ * Unique per code by construction (the code is embedded in the word). */
+ORC_EXPORT int orc_dict_word(uint64_t seed, int64_t code, char* out, int cap) {
+    uint32_t h = (uint32_t)bk_mix64(seed ^ 0xD1C7ull ^ (uint64_t)code);
+    return snprintf(out, (size_t)cap, "w%06lx_%08x", (unsigned long)code, h);
+}
+
+/* ------------------------------------------------------------------ */
+/* Filter (need_copy, filter_node.cpp:726-734)                          */
+/* ------------------------------------------------------------------ */
+
+static inline int cell_is_valid(const OrcCol* c, int64_t r) {
+    return c->valid == NULL || c->valid[r];
+}
+
+static inline int64_t cell_i64(const OrcCol* c, int64_t r) {
+    if (c->type == BK_STRING) return (int64_t)((int32_t*)c->data)[r];
+    return ((int64_t*)c->data)[r];
+}
+
+static inline double cell_f64_cast(const OrcCol* c, int64_t r) {
+    /* ExprValue::get_numberic<double> (expr_value.h:341-409): numeric types
+     * convert by value. */
+    if (c->type == BK_DOUBLE) return ((double*)c->data)[r];
+    return (double)cell_i64(c, r);
+}
+
+/* returns 1 iff row passes every conjunct (NULL or false rejects). */
+static int orc_row_passes(const OrcCol* cols, const BkQuerySpec* q, int64_t r) {
+    for (int32_t j = 0; j < q->n_conjuncts; j++) {
+        const BkConjunct* cj = &q->conjuncts[j];
+        const OrcCol* c = &cols[cj->col];
+        if (!cell_is_valid(c, r)) return 0;          /* NULL operand => NULL => reject */
+        int cmp; /* sign of (col - lit) */
+        if (cj->cmp_type == BK_DOUBLE) {
+            double v = cell_f64_cast(c, r);
+            cmp = (v > cj->lit_d) - (v < cj->lit_d);
+        } else { /* BK_INT64 or BK_STRING dict-code compare */
+            int64_t v = cell_i64(c, r);
+            cmp = (v > cj->lit_i) - (v < cj->lit_i);
+        }
+        int pass;
+        switch ((BkCmpOp)cj->op) {
+            case BK_OP_EQ: pass = (cmp == 0); break;
+            case BK_OP_NE: pass = (cmp != 0); break;
+            case BK_OP_GT: pass = (cmp > 0);  break;
+            case BK_OP_GE: pass = (cmp >= 0); break;
+            case BK_OP_LT: pass = (cmp < 0);  break;
+            case BK_OP_LE: pass = (cmp <= 0); break;
+            default: pass = 0;
+        }
+        if (!pass) return 0;
+    }
+    return 1;
+}
+
+/* ------------------------------------------------------------------ */
+/* Hash aggregate                                                      */
+/* ------------------------------------------------------------------ */
+
+/* Per-group aggregate state — one entry per agg call.
+ * ExprValue-typed per the reference: has==0 <=> intermediate is NULL. */
+typedef struct OrcAggState {
+    int64_t i;    /* COUNT / SUM(int64) / MIN/MAX(int64|dict) */
+    double  d;    /* SUM(double) / AVG sum / MIN/MAX(double) */
+    int64_t cnt;  /* AVG count */
+    uint8_t has;  /* 0 => NULL intermediate */
+} OrcAggState;
+
+typedef struct OrcGroup {
+    uint8_t  flag;      /* null-flag byte, exec_node.cpp:556-567 */
+    uint64_t e[BK_MAX_GROUP]; /* order-preserving encodes of group values */
+    OrcAggState st[BK_MAX_AGGS];
+} OrcGroup;
+
+typedef struct OrcMap {
+    uint64_t  cap;      /* power of 2 */
+    uint64_t  n;
+    uint64_t* hashes;   /* 0 = empty (hashes stored |1) */
+    OrcGroup* groups;
+} OrcMap;
+
+static uint64_t orc_key_hash(uint8_t flag, const uint64_t* e, int ng) {
+    uint64_t h = bk_mix64(0x9E1Eull ^ flag);
+    for (int i = 0; i < ng; i++) h = bk_mix64(h ^ e[i]);
+    return h | 1;
+}
+
+static void orc_map_init(OrcMap* m, uint64_t cap) {
+    uint64_t c = 64;
+    while (c < cap) c <<= 1;
+    m->cap = c; m->n = 0;
+    m->hashes = (uint64_t*)calloc(c, sizeof(uint64_t));
+    m->groups = (OrcGroup*)calloc(c, sizeof(OrcGroup));
+}
+
+static void orc_map_free(OrcMap* m) { free(m->hashes); free(m->groups); }
+
+static OrcGroup* orc_map_find_or_insert(OrcMap* m, uint8_t flag,
+                                        const uint64_t* e, int ng, int* created);
+
+static void orc_map_grow(OrcMap* m, int ng) {
+    OrcMap bigger;
+    orc_map_init(&bigger, m->cap * 2);
+    for (uint64_t i = 0; i < m->cap; i++) {
+        if (!m->hashes[i]) continue;
+        int created;
+        OrcGroup* g = orc_map_find_or_insert(&bigger, m->groups[i].flag,
+                                             m->groups[i].e, ng, &created);
+        memcpy(g->st, m->groups[i].st, sizeof(g->st));
+    }
+    orc_map_free(m);
+    *m = bigger;
+}
+
+static OrcGroup* orc_map_find_or_insert(OrcMap* m, uint8_t flag,
+                                        const uint64_t* e, int ng, int* created) {
+    uint64_t h = orc_key_hash(flag, e, ng);
+    uint64_t mask = m->cap - 1;
+    uint64_t i = h & mask;
+    for (;;) {
+        if (!m->hashes[i]) {
+            m->hashes[i] = h;
+            OrcGroup* g = &m->groups[i];
+            g->flag = flag;
+            for (int k = 0; k < ng; k++) g->e[k] = e[k];
+            m->n++;
+            *created = 1;
+            return g;
+        }
+        if (m->hashes[i] == h) {
+            OrcGroup* g = &m->groups[i];
+            int same = (g->flag == flag);
+            for (int k = 0; same && k < ng; k++) same = (g->e[k] == e[k]);
+            if (same) { *created = 0; return g; }
+        }
+        i = (i + 1) & mask;
+    }
+}
+
+/* AggFnCall::update (agg_fn_call.cpp:496-555). in_valid==0 means the input
+ * cell is SQL NULL. vtype is the input column's BkType. */
+static inline void orc_agg_update(OrcAggState* s, int agg_type, int vtype,
+                                  int in_valid, int64_t vi, double vd) {
+    switch (agg_type) {
+        case BK_AGG_COUNT_STAR:
+            s->i++; s->has = 1; return;
+        case BK_AGG_COUNT:
+            if (in_valid) { s->i++; } s->has = 1; return;
+        case BK_AGG_SUM:
+            if (!in_valid) return;
+            if (vtype == BK_DOUBLE) {
+                if (!s->has) { s->d = vd; s->has = 1; }
+                else s->d += vd;                      /* expr_value.h:869-871 */
+            } else {
+                if (!s->has) { s->i = vi; s->has = 1; }
+                else s->i = (int64_t)((uint64_t)s->i + (uint64_t)vi); /* int64 wrap */
+            }
+            return;
+        case BK_AGG_AVG:
+            if (!in_valid) return;
+            s->d += (vtype == BK_DOUBLE) ? vd : (double)vi; /* get_numberic<double> */
+            s->cnt++; s->has = 1;
+            return;
+        case BK_AGG_MIN:
+            if (!in_valid) return;
+            if (vtype == BK_DOUBLE) {
+                if (!s->has || s->d > vd) s->d = vd;
+            } else {
+                if (!s->has || s->i > vi) s->i = vi;
+            }
+            s->has = 1; return;
+        case BK_AGG_MAX:
+            if (!in_valid) return;
+            if (vtype == BK_DOUBLE) {
+                if (!s->has || s->d < vd) s->d = vd;
+            } else {
+                if (!s->has || s->i < vi) s->i = vi;
+            }
+            s->has = 1; return;
+        default: return;
+    }
+}
+
+/* AggFnCall::merge (agg_fn_call.cpp:781-830): partials combine. */
+static inline void orc_agg_merge(OrcAggState* dst, const OrcAggState* src,
+                                 int agg_type, int vtype) {
+    switch (agg_type) {
+        case BK_AGG_COUNT_STAR:
+        case BK_AGG_COUNT:
+            if (src->has) { dst->i += src->i; dst->has = 1; } return;
+        case BK_AGG_SUM:
+            if (!src->has) return;
+            if (vtype == BK_DOUBLE) {
+                if (!dst->has) { dst->d = src->d; dst->has = 1; }
+                else dst->d += src->d;
+            } else {
+                if (!dst->has) { dst->i = src->i; dst->has = 1; }
+                else dst->i = (int64_t)((uint64_t)dst->i + (uint64_t)src->i);
+            }
+            return;
+        case BK_AGG_AVG:
+            if (!src->has) return;
+            dst->d += src->d; dst->cnt += src->cnt; dst->has = 1; return;
+        case BK_AGG_MIN:
+            if (!src->has) return;
+            if (vtype == BK_DOUBLE) { if (!dst->has || dst->d > src->d) dst->d = src->d; }
+            else                    { if (!dst->has || dst->i > src->i) dst->i = src->i; }
+            dst->has = 1; return;
+        case BK_AGG_MAX:
+            if (!src->has) return;
+            if (vtype == BK_DOUBLE) { if (!dst->has || dst->d < src->d) dst->d = src->d; }
+            else                    { if (!dst->has || dst->i < src->i) dst->i = src->i; }
+            dst->has = 1; return;
+        default: return;
+    }
+}
+
+/* encode one group-column value to its order-preserving u64 (same internal
+ * key the GPU uses). */
+static inline uint64_t orc_enc_group(const OrcCol* c, int64_t r) {
+    switch (c->type) {
+        case BK_INT64:  return bk_enc_i64(((int64_t*)c->data)[r]);
+        case BK_DOUBLE: return bk_enc_f64(((double*)c->data)[r]);
+        case BK_STRING: return (uint64_t)(uint32_t)((int32_t*)c->data)[r];
+        default:        return 0;
+    }
+}
+
+/* ------------------------------------------------------------------ */
+/* filter+aggregate over a row range (one "region"'s work)             */
+/* ------------------------------------------------------------------ */
+
+typedef struct OrcAggTask {
+    const OrcCol*      cols;
+    const BkQuerySpec* q;
+    int64_t            row_begin, row_end;
+    OrcMap             map;
+    int64_t            rows_passed;
+} OrcAggTask;
+
+static void* orc_agg_worker(void* arg) {
+    OrcAggTask* t = (OrcAggTask*)arg;
+    const BkQuerySpec* q = t->q;
+    const OrcCol* cols = t->cols;
+    orc_map_init(&t->map, 1024);
+    int ng = q->n_group;
+    for (int64_t r = t->row_begin; r < t->row_end; r++) {
+        if (!orc_row_passes(cols, q, r)) continue;
+        t->rows_passed++;
+        uint8_t flag = 0;
+        uint64_t e[BK_MAX_GROUP] = {0, 0};
+        for (int k = 0; k < ng; k++) {
+            const OrcCol* c = &cols[q->group_cols[k]];
+            if (!cell_is_valid(c, r)) flag |= (uint8_t)(0x01u << (7 - k)); /* exec_node.cpp:561 */
+            else e[k] = orc_enc_group(c, r);
+        }
+        int created;
+        if (t->map.n * 10 >= t->map.cap * 6) orc_map_grow(&t->map, ng);
+        OrcGroup* g = orc_map_find_or_insert(&t->map, flag, e, ng, &created);
+        for (int32_t a = 0; a < q->n_aggs; a++) {
+            const BkAggSpec* as = &q->aggs[a];
+            int vtype = q->agg_in_types[a];
+            int in_valid = 1;
+            int64_t vi = 0; double vd = 0.0;
+            if (as->col >= 0) {
+                const OrcCol* c = &cols[as->col];
+                in_valid = cell_is_valid(c, r);
+                if (in_valid) {
+                    if (c->type == BK_DOUBLE) vd = ((double*)c->data)[r];
+                    else vi = cell_i64(c, r);
+                }
+            }
+            orc_agg_update(&g->st[a], as->agg_type, vtype, in_valid, vi, vd);
+        }
+    }
+    return NULL;
+}
+
+/* ------------------------------------------------------------------ */
+/* Result materialization                                              */
+/* ------------------------------------------------------------------ */
+
+typedef struct OrcAggResult {
+    int64_t  ngroups;
+    int64_t  rows_passed;  /* rows surviving the filter (RuntimeState counters) */
+    /* reference MutTableKey byte strings, concatenated */
+    uint8_t* key_bytes;
+    int64_t* key_off;      /* ngroups+1 offsets */
+    /* per agg a, per group g: value at [a*ngroups + g] */
+    int64_t* out_i;        /* int64-typed outputs (COUNT, SUM int64, MIN/MAX int64) */
+    double*  out_d;        /* double-typed outputs (SUM double, AVG, MIN/MAX double) */
+    uint8_t* out_has;      /* 0 => SQL NULL output */
+    /* raw group-key components for programmatic checks */
+    uint8_t*  g_flag;
+    uint64_t* g_enc;       /* [g*BK_MAX_GROUP + k] */
+} OrcAggResult;
+
+typedef struct KeySortRef { const OrcGroup* g; } KeySortRef;
+
+static int s_sort_ng;
+static int orc_group_cmp(const void* a, const void* b) {
+    const OrcGroup* ga = ((const KeySortRef*)a)->g;
+    const OrcGroup* gb = ((const KeySortRef*)b)->g;
+    /* order by the reference byte-key memcmp order == (flag, then per present
+     * column big-endian bytes). Comparing (flag, e[k] with nulls as omitted)
+     * — since both keys share the type list, compare flag first then encoded
+     * values; a null column (bit set in flag) contributes nothing. */
+    if (ga->flag != gb->flag) return ga->flag < gb->flag ? -1 : 1;
+    for (int k = 0; k < s_sort_ng; k++) {
+        int a_null = (ga->flag >> (7 - k)) & 1;
+        if (a_null) continue;
+        if (ga->e[k] != gb->e[k]) return ga->e[k] < gb->e[k] ? -1 : 1;
+    }
+    return 0;
+}
+
+/* Build the reference MutTableKey byte string for one group
+ * (exec_node.cpp:555-571 + mut_table_key.h append_value). */
+static int64_t orc_build_key_bytes(const OrcGroup* g, const BkQuerySpec* q,
+                                   uint64_t dict_seed, uint8_t* out /* may be NULL */) {
+    int64_t len = 0;
+    if (out) out[0] = g->flag;
+    len = 1;
+    for (int k = 0; k < q->n_group; k++) {
+        if ((g->flag >> (7 - k)) & 1) continue;  /* null values omitted */
+        int t = q->group_types[k];
+        if (t == BK_INT64 || t == BK_DOUBLE) {
+            if (out) {
+                uint64_t be = bk_bswap64(g->e[k]);  /* already sign-flip encoded */
+                memcpy(out + len, &be, 8);
+            }
+            len += 8;
+        } else if (t == BK_STRING) {
+            char word[64];
+            int wl = orc_dict_word(dict_seed, (int64_t)g->e[k], word, sizeof word);
+            if (out) { memcpy(out + len, word, (size_t)wl); out[len + wl] = 0; }
+            len += wl + 1;  /* append_string: bytes + '\0', mut_table_key.h:166-169 */
+        }
+    }
+    return len;
+}
+
+ORC_EXPORT void orc_agg_result_free(OrcAggResult* res) {
+    if (!res) return;
+    free(res->key_bytes); free(res->key_off);
+    free(res->out_i); free(res->out_d); free(res->out_has);
+    free(res->g_flag); free(res->g_enc);
+    free(res);
+}
+
+/* Finalize (agg_fn_call.cpp:927-975) one group's agg into output arrays. */
+static void orc_finalize_group(const OrcGroup* g, const BkQuerySpec* q,
+                               int64_t gi, int64_t ngroups, OrcAggResult* res) {
+    for (int32_t a = 0; a < q->n_aggs; a++) {
+        const OrcAggState* s = &g->st[a];
+        int64_t idx = (int64_t)a * ngroups + gi;
+        int at = q->aggs[a].agg_type;
+        int vtype = q->agg_in_types[a];
+        switch (at) {
+            case BK_AGG_COUNT_STAR:
+            case BK_AGG_COUNT:
+                res->out_i[idx] = s->i; res->out_has[idx] = 1; break;
+            case BK_AGG_SUM:
+                if (!s->has) { res->out_has[idx] = 0; break; }
+                if (vtype == BK_DOUBLE) res->out_d[idx] = s->d;
+                else res->out_i[idx] = s->i;
+                res->out_has[idx] = 1; break;
+            case BK_AGG_AVG:
+                if (!s->has || s->cnt == 0) { res->out_has[idx] = 0; break; }
+                res->out_d[idx] = s->d / (double)s->cnt;   /* agg_fn_call.cpp:958 */
+                res->out_has[idx] = 1; break;
+            case BK_AGG_MIN:
+            case BK_AGG_MAX:
+                if (!s->has) { res->out_has[idx] = 0; break; }
+                if (vtype == BK_DOUBLE) res->out_d[idx] = s->d;
+                else res->out_i[idx] = s->i;
+                res->out_has[idx] = 1; break;
+            default: res->out_has[idx] = 0;
+        }
+    }
+}
+
+/* Run filter+aggregate over [row_begin,row_end) with nthreads shards
+ * (each shard mirrors one region's AggNode, merged like MERGE_AGG —
+ * agg_node.cpp:29,539-543). Results sorted by reference key bytes iff
+ * sort_keys != 0. */
+ORC_EXPORT OrcAggResult* orc_filter_agg(const OrcCol* cols, int ncols,
+                                        const BkQuerySpec* q,
+                                        int64_t row_begin, int64_t row_end,
+                                        int nthreads, uint64_t dict_seed,
+                                        int sort_keys) {
+    (void)ncols;
+    if (nthreads < 1) nthreads = 1;
+    int64_t n = row_end - row_begin;
+    if (nthreads > 64) nthreads = 64;
+    if ((int64_t)nthreads > n && n > 0) nthreads = (int)n;
+    if (n <= 0) nthreads = 1;
+
+    OrcAggTask* tasks = (OrcAggTask*)calloc((size_t)nthreads, sizeof(OrcAggTask));
+    pthread_t* tids = (pthread_t*)calloc((size_t)nthreads, sizeof(pthread_t));
+    int64_t chunk = nthreads > 0 ? (n + nthreads - 1) / nthreads : 0;
+    for (int t = 0; t < nthreads; t++) {
+        tasks[t].cols = cols; tasks[t].q = q;
+        tasks[t].row_begin = row_begin + (int64_t)t * chunk;
+        tasks[t].row_end = tasks[t].row_begin + chunk;
+        if (tasks[t].row_end > row_end) tasks[t].row_end = row_end;
+        if (tasks[t].row_begin > row_end) tasks[t].row_begin = row_end;
+        if (nthreads == 1) orc_agg_worker(&tasks[t]);
+        else pthread_create(&tids[t], NULL, orc_agg_worker, &tasks[t]);
+    }
+    if (nthreads > 1)
+        for (int t = 0; t < nthreads; t++) pthread_join(tids[t], NULL);
+
+    /* merge partials into tasks[0].map (MERGE_AGG path, agg_node.cpp:539-543) */
+    OrcMap* m0 = &tasks[0].map;
+    int64_t rows_passed = tasks[0].rows_passed;
+    for (int t = 1; t < nthreads; t++) {
+        OrcMap* mt = &tasks[t].map;
+        rows_passed += tasks[t].rows_passed;
+        for (uint64_t i = 0; i < mt->cap; i++) {
+            if (!mt->hashes[i]) continue;
+            int created;
+            if (m0->n * 10 >= m0->cap * 6) orc_map_grow(m0, q->n_group);
+            OrcGroup* g = orc_map_find_or_insert(m0, mt->groups[i].flag,
+                                                 mt->groups[i].e, q->n_group, &created);
+            for (int32_t a = 0; a < q->n_aggs; a++)
+                orc_agg_merge(&g->st[a], &mt->groups[i].st[a],
+                              q->aggs[a].agg_type, q->agg_in_types[a]);
+        }
+        orc_map_free(mt);
+    }
+
+    /* zero-row, no-GROUP-BY query => one all-initialized row
+     * (agg_node.cpp:490-505; COUNT 0, SUM/AVG/MIN/MAX NULL) */
+    if (m0->n == 0 && q->n_group == 0) {
+        int created;
+        uint64_t e[BK_MAX_GROUP] = {0, 0};
+        OrcGroup* g = orc_map_find_or_insert(m0, 0, e, 0, &created);
+        for (int32_t a = 0; a < q->n_aggs; a++) {
+            int at = q->aggs[a].agg_type;
+            if (at == BK_AGG_COUNT_STAR || at == BK_AGG_COUNT) g->st[a].has = 1;
+        }
+    }
+
+    /* collect + optional canonical sort */
+    int64_t ngroups = (int64_t)m0->n;
+    KeySortRef* refs = (KeySortRef*)malloc((size_t)ngroups * sizeof(KeySortRef));
+    int64_t gi = 0;
+    for (uint64_t i = 0; i < m0->cap; i++)
+        if (m0->hashes[i]) refs[gi++].g = &m0->groups[i];
+    if (sort_keys) {
+        s_sort_ng = q->n_group;
+        qsort(refs, (size_t)ngroups, sizeof(KeySortRef), orc_group_cmp);
+    }
+
+    OrcAggResult* res = (OrcAggResult*)calloc(1, sizeof(OrcAggResult));
+    res->ngroups = ngroups;
+    res->rows_passed = rows_passed;
+    res->key_off = (int64_t*)malloc((size_t)(ngroups + 1) * sizeof(int64_t));
+    int64_t total = 0;
+    for (int64_t g2 = 0; g2 < ngroups; g2++) {
+        res->key_off[g2] = total;
+        total += orc_build_key_bytes(refs[g2].g, q, dict_seed, NULL);
+    }
+    res->key_off[ngroups] = total;
+    res->key_bytes = (uint8_t*)malloc((size_t)(total > 0 ? total : 1));
+    int64_t na = (int64_t)q->n_aggs * (ngroups > 0 ? ngroups : 1);
+    res->out_i = (int64_t*)calloc((size_t)na, sizeof(int64_t));
+    res->out_d = (double*)calloc((size_t)na, sizeof(double));
+    res->out_has = (uint8_t*)calloc((size_t)na, 1);
+    res->g_flag = (uint8_t*)calloc((size_t)(ngroups > 0 ? ngroups : 1), 1);
+    res->g_enc = (uint64_t*)calloc((size_t)(ngroups > 0 ? ngroups : 1) * BK_MAX_GROUP,
+                                   sizeof(uint64_t));
+    for (int64_t g2 = 0; g2 < ngroups; g2++) {
+        orc_build_key_bytes(refs[g2].g, q, dict_seed, res->key_bytes + res->key_off[g2]);
+        orc_finalize_group(refs[g2].g, q, g2, ngroups, res);
+        res->g_flag[g2] = refs[g2].g->flag;
+        for (int k = 0; k < BK_MAX_GROUP; k++)
+            res->g_enc[g2 * BK_MAX_GROUP + k] = refs[g2].g->e[k];
+    }
+    free(refs);
+    orc_map_free(m0);
+    free(tasks); free(tids);
+    return res;
+}
+
+/* ------------------------------------------------------------------ */
+/* ORDER BY ... LIMIT top-N (TopNSorter semantics)                     */
+/* ------------------------------------------------------------------ */
+
+/* comparator per mem_row_compare.cpp:18-40 + arrival-index tie-break
+ * (topn_sorter.h:46-54). Returns <0 if a orders before b. */
+static int orc_row_order_cmp(const OrcCol* cols, const BkOrderSpec* order, int norder,
+                             int64_t ra, int64_t rb) {
+    for (int i = 0; i < norder; i++) {
+        const OrcCol* c = &cols[order[i].col];
+        int va = cell_is_valid(c, ra), vb = cell_is_valid(c, rb);
+        if (!va && !vb) continue;
+        if (!va) return order[i].is_null_first ? -1 : 1;
+        if (!vb) return order[i].is_null_first ? 1 : -1;
+        int cmp;
+        if (c->type == BK_DOUBLE) {
+            double x = ((double*)c->data)[ra], y = ((double*)c->data)[rb];
+            cmp = (x > y) - (x < y);
+        } else {
+            int64_t x = cell_i64(c, ra), y = cell_i64(c, rb);
+            cmp = (x > y) - (x < y);
+        }
+        if (cmp != 0) return order[i].is_asc ? cmp : -cmp;
+    }
+    return (ra > rb) - (ra < rb);  /* stable by arrival order */
+}
+
+typedef struct OrcTopCtx {
+    const OrcCol* cols; const BkOrderSpec* order; int norder;
+} OrcTopCtx;
+static OrcTopCtx s_top_ctx;
+static int orc_top_qsort_cmp(const void* a, const void* b) {
+    return orc_row_order_cmp(s_top_ctx.cols, s_top_ctx.order, s_top_ctx.norder,
+                             *(const int64_t*)a, *(const int64_t*)b);
+}
+
+/* max-heap of the current top-N (root = worst kept row) */
+static void orc_heap_siftdown(int64_t* h, int64_t n, int64_t i, const OrcTopCtx* c) {
+    for (;;) {
+        int64_t l = 2 * i + 1, r = l + 1, m = i;
+        if (l < n && orc_row_order_cmp(c->cols, c->order, c->norder, h[l], h[m]) > 0) m = l;
+        if (r < n && orc_row_order_cmp(c->cols, c->order, c->norder, h[r], h[m]) > 0) m = r;
+        if (m == i) return;
+        int64_t t = h[i]; h[i] = h[m]; h[m] = t;
+        i = m;
+    }
+}
+
+/* Select + order the top `limit` rows of [row_begin,row_end) that pass the
+ * filter (SortNode drains FilterNode, sort_node.cpp:278-347). Returns number
+ * written to out_rows (global row indices, in final output order). */
+ORC_EXPORT int64_t orc_sort_topk(const OrcCol* cols, int ncols, const BkQuerySpec* q,
+                                 const BkOrderSpec* order, int norder,
+                                 int64_t row_begin, int64_t row_end,
+                                 int64_t limit, int64_t* out_rows) {
+    (void)ncols;
+    OrcTopCtx ctx = { cols, order, norder };
+    int64_t* heap = (int64_t*)malloc((size_t)limit * sizeof(int64_t));
+    int64_t hn = 0;
+    for (int64_t r = row_begin; r < row_end; r++) {
+        if (q && q->n_conjuncts > 0 && !orc_row_passes(cols, q, r)) continue;
+        if (hn < limit) {
+            heap[hn++] = r;
+            if (hn == limit)  /* heapify */
+                for (int64_t i = hn / 2 - 1; i >= 0; i--) orc_heap_siftdown(heap, hn, i, &ctx);
+        } else if (limit > 0 &&
+                   orc_row_order_cmp(cols, order, norder, r, heap[0]) < 0) {
+            heap[0] = r;
+            orc_heap_siftdown(heap, hn, 0, &ctx);
+        }
+    }
+    /* hn < limit => never heapified: contents are in arrival order, which
+     * qsort below handles the same way. */
+    s_top_ctx = ctx;
+    qsort(heap, (size_t)hn, sizeof(int64_t), orc_top_qsort_cmp);
+    memcpy(out_rows, heap, (size_t)hn * sizeof(int64_t));
+    free(heap);
+    return hn;
+}
+
+/* ------------------------------------------------------------------ */
+/* small exported helpers for tests                                    */
+/* ------------------------------------------------------------------ */
+
+ORC_EXPORT uint64_t orc_encode_i64(int64_t v)  { return bk_enc_i64(v); }
+ORC_EXPORT int64_t  orc_decode_i64(uint64_t u) { return bk_dec_i64(u); }
+ORC_EXPORT uint64_t orc_encode_f64(double v)   { return bk_enc_f64(v); }
+ORC_EXPORT double   orc_decode_f64(uint64_t u) { return bk_dec_f64(u); }
+ORC_EXPORT uint64_t orc_mix64(uint64_t x)      { return bk_mix64(x); }
+ORC_EXPORT uint64_t orc_cell_bits(uint64_t seed, uint64_t row, uint32_t col) {
+    return bk_cell_bits(seed, row, col);
+}
