@@ -1,0 +1,986 @@
+// bkgpu.hip — MI355X-native (gfx950/CDNA4) implementation of BaikalDB's OLAP
+// execution hot path: fused scan+filter+hash-aggregate, partial-aggregate
+// merge, and top-N selection, behind the C-ABI of include/bkgpu.h.
+//
+// This is NOT a port: the reference (baidu/BaikalDB src/exec, src/expr) is a
+// row-at-a-time Volcano interpreter over protobuf MemRows; here the same
+// *semantics* (cited per function) are computed columnar on HBM:
+//   - one grid-stride pass reads each referenced column byte exactly once
+//     (coalesced 8 B/lane loads; the path is HBM-bandwidth-bound, no MFMA —
+//     north_star: "no dense contraction here")
+//   - predicate evaluation is per-lane, wave-uniform control flow
+//   - GROUP BY uses a two-level hash aggregate: a per-workgroup LDS table
+//     (ds-atomics, absorbs hot/Zipf keys) flushed into a global open-
+//     addressing table in HBM (device-scope atomics, 8-byte-granule
+//     publish/consume per the CDNA4 visibility rules: sc1 stores + vmcnt
+//     drain + sc1 state word; see /opt/skills/guides G16/R1-R2)
+//   - aggregate states are order-independent: COUNT/SUM(int64) as wrapping
+//     u64 adds (bit-exact vs the reference's sequential int64 adds),
+//     MIN/MAX as atomicMax over the order-preserving encoding of
+//     include/bk_keyenc.h (bit-exact), SUM/AVG(double) as f64 atomic adds
+//     (reduction order differs from the reference's sequential
+//     ExprValue::add — tolerance stated in DESIGN.md and tests)
+//
+// Reference semantics implemented (file:line in /root/reference):
+//   filter:    FilterNode::need_copy            src/exec/filter_node.cpp:726-734
+//   compares:  operators.cpp:79-105 via ScalarFnCall arg casts
+//              (src/expr/scalar_fn_call.cpp:219-225)
+//   group key: ExecNode::encode_exprs_key       src/exec/exec_node.cpp:555-571
+//   hash agg:  AggNode::process_row_batch       src/exec/agg_node.cpp:507-545
+//   agg fns:   AggFnCall init/update/merge/finalize
+//              src/expr/agg_fn_call.cpp:370-455,496-555,719-830,927-975
+//   merge:     MERGE_AGG partial combine        src/exec/agg_node.cpp:29,539-543
+//   top-N:     TopNSorter (arrival-index ties)  include/runtime/topn_sorter.h:32-63
+
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+#include <vector>
+#include <algorithm>
+
+#include "../../include/bk_common.h"
+#include "../../include/bk_datagen.h"
+#include "../../include/bk_keyenc.h"
+#include "../../include/bkgpu.h"
+
+/* ------------------------------------------------------------------ */
+/* host error plumbing                                                 */
+/* ------------------------------------------------------------------ */
+
+static thread_local char g_err[512] = "";
+static void set_err(const char* msg) { snprintf(g_err, sizeof g_err, "%s", msg); }
+extern "C" const char* bkgpu_last_error(void) { return g_err; }
+
+#define HIP_CHECK(x) do { hipError_t _e = (x); if (_e != hipSuccess) { \
+    snprintf(g_err, sizeof g_err, "%s:%d %s: %s", __FILE__, __LINE__, #x, \
+             hipGetErrorString(_e)); return -1; } } while (0)
+#define HIP_CHECK_NULL(x) do { hipError_t _e = (x); if (_e != hipSuccess) { \
+    snprintf(g_err, sizeof g_err, "%s:%d %s: %s", __FILE__, __LINE__, #x, \
+             hipGetErrorString(_e)); return nullptr; } } while (0)
+
+/* ------------------------------------------------------------------ */
+/* device-side column view + query                                     */
+/* ------------------------------------------------------------------ */
+
+struct DevCol {
+    int32_t type;            /* BkType */
+    const void* data;
+    const uint8_t* valid;    /* null => all valid */
+};
+
+struct DevCols {
+    DevCol c[BK_MAX_COLS];
+};
+
+#define RLX __ATOMIC_RELAXED
+#define AGT __HIP_MEMORY_SCOPE_AGENT
+#define WGP __HIP_MEMORY_SCOPE_WORKGROUP
+
+/* slot layout, in u64 words, stride = 3 + 2*naggs:
+ *  w0: {u32 state (0 empty / 1 building / 2 ready), u32 null-flag}
+ *  w1: k0   w2: k1
+ *  w3+2a: agg value bits   w3+2a+1: agg non-null count                */
+#define SLOT_HDR 3
+#define MAX_PROBE 8192
+
+__device__ __forceinline__ uint64_t key_hash(uint32_t flag, uint64_t k0, uint64_t k1) {
+    uint64_t h = bk_mix64(k0 ^ 0x9E3779B97F4A7C15ull);
+    h = bk_mix64(h ^ k1);
+    h = bk_mix64(h ^ flag);
+    return h;
+}
+
+/* ---- cell access ---- */
+__device__ __forceinline__ bool cell_valid(const DevCol& c, int64_t r) {
+    return c.valid == nullptr || c.valid[r];
+}
+__device__ __forceinline__ int64_t cell_i64(const DevCol& c, int64_t r) {
+    if (c.type == BK_STRING) return (int64_t)((const int32_t*)c.data)[r];
+    return ((const int64_t*)c.data)[r];
+}
+__device__ __forceinline__ double cell_f64(const DevCol& c, int64_t r) {
+    if (c.type == BK_DOUBLE) return ((const double*)c.data)[r];
+    return (double)cell_i64(c, r);   /* get_numberic<double>, expr_value.h:341 */
+}
+
+/* need_copy (filter_node.cpp:726-734): all conjuncts non-NULL and true */
+__device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpec& q,
+                                           int64_t r) {
+    #pragma unroll 4
+    for (int32_t j = 0; j < q.n_conjuncts; j++) {
+        const BkConjunct& cj = q.conjuncts[j];
+        const DevCol& c = cols.c[cj.col];
+        if (!cell_valid(c, r)) return false;
+        int cmp;
+        if (cj.cmp_type == BK_DOUBLE) {
+            double v = cell_f64(c, r);
+            cmp = (v > cj.lit_d) - (v < cj.lit_d);
+        } else {
+            int64_t v = cell_i64(c, r);
+            cmp = (v > cj.lit_i) - (v < cj.lit_i);
+        }
+        bool pass;
+        switch (cj.op) {
+            case BK_OP_EQ: pass = (cmp == 0); break;
+            case BK_OP_NE: pass = (cmp != 0); break;
+            case BK_OP_GT: pass = (cmp > 0);  break;
+            case BK_OP_GE: pass = (cmp >= 0); break;
+            case BK_OP_LT: pass = (cmp < 0);  break;
+            default:       pass = (cmp <= 0); break;  /* LE */
+        }
+        if (!pass) return false;
+    }
+    return true;
+}
+
+/* order-preserving u64 encode of a group/minmax value (bk_keyenc.h) */
+__device__ __forceinline__ uint64_t enc_value(const DevCol& c, int64_t r) {
+    switch (c.type) {
+        case BK_INT64:  return bk_enc_i64(((const int64_t*)c.data)[r]);
+        case BK_DOUBLE: return bk_enc_f64(((const double*)c.data)[r]);
+        default:        return (uint64_t)(uint32_t)((const int32_t*)c.data)[r];
+    }
+}
+
+/* ---- double atomic adds (IEEE add; order unspecified) ---- */
+__device__ __forceinline__ void atomic_add_f64_global(uint64_t* addr, double v) {
+    unsafeAtomicAdd((double*)addr, v);   /* global_atomic_add_f64 on gfx950 */
+}
+__device__ __forceinline__ void atomic_add_f64_lds(uint64_t* addr, double v) {
+    unsafeAtomicAdd((double*)addr, v);   /* ds_add_f64 */
+}
+
+/* ---- per-agg atomic update into a slot (LDS or global templated) ---- */
+template <bool LDS>
+__device__ __forceinline__ void agg_update_slot(uint64_t* st, const BkQuerySpec& q,
+                                                const DevCols& cols, int64_t r) {
+    #pragma unroll 4
+    for (int32_t a = 0; a < q.n_aggs; a++) {
+        uint64_t* val = st + SLOT_HDR + 2 * a;
+        uint64_t* cnt = val + 1;
+        const BkAggSpec& as = q.aggs[a];
+        switch (as.agg_type) {
+            case BK_AGG_COUNT_STAR:
+                atomicAdd((unsigned long long*)val, 1ull);
+                break;
+            case BK_AGG_COUNT: {
+                if (cell_valid(cols.c[as.col], r))
+                    atomicAdd((unsigned long long*)val, 1ull);
+                break;
+            }
+            case BK_AGG_SUM: {
+                const DevCol& c = cols.c[as.col];
+                if (!cell_valid(c, r)) break;
+                if (q.agg_in_types[a] == BK_DOUBLE) {
+                    if (LDS) atomic_add_f64_lds(val, ((const double*)c.data)[r]);
+                    else     atomic_add_f64_global(val, ((const double*)c.data)[r]);
+                } else {
+                    atomicAdd((unsigned long long*)val,
+                              (unsigned long long)cell_i64(c, r)); /* int64 wrap */
+                }
+                atomicAdd((unsigned long long*)cnt, 1ull);
+                break;
+            }
+            case BK_AGG_AVG: {
+                const DevCol& c = cols.c[as.col];
+                if (!cell_valid(c, r)) break;
+                double v = cell_f64(c, r);
+                if (LDS) atomic_add_f64_lds(val, v);
+                else     atomic_add_f64_global(val, v);
+                atomicAdd((unsigned long long*)cnt, 1ull);
+                break;
+            }
+            case BK_AGG_MIN: {
+                const DevCol& c = cols.c[as.col];
+                if (!cell_valid(c, r)) break;
+                atomicMax((unsigned long long*)val,
+                          (unsigned long long)(~enc_value(c, r)));
+                atomicAdd((unsigned long long*)cnt, 1ull);
+                break;
+            }
+            case BK_AGG_MAX: {
+                const DevCol& c = cols.c[as.col];
+                if (!cell_valid(c, r)) break;
+                atomicMax((unsigned long long*)val,
+                          (unsigned long long)enc_value(c, r));
+                atomicAdd((unsigned long long*)cnt, 1ull);
+                break;
+            }
+            default: break;
+        }
+    }
+}
+
+/* merge one partial state array into a slot (MERGE_AGG, agg_node.cpp:539) */
+template <bool LDS>
+__device__ __forceinline__ void agg_merge_slot(uint64_t* dst, const uint64_t* src,
+                                               const BkQuerySpec& q) {
+    #pragma unroll 4
+    for (int32_t a = 0; a < q.n_aggs; a++) {
+        uint64_t* val = dst + SLOT_HDR + 2 * a;
+        uint64_t* cnt = val + 1;
+        uint64_t sv = src[2 * a];
+        uint64_t sc = src[2 * a + 1];
+        switch (q.aggs[a].agg_type) {
+            case BK_AGG_COUNT_STAR:
+            case BK_AGG_COUNT:
+                if (sv) atomicAdd((unsigned long long*)val, (unsigned long long)sv);
+                break;
+            case BK_AGG_SUM:
+            case BK_AGG_AVG:
+                if (sc) {
+                    if (q.agg_in_types[a] == BK_DOUBLE || q.aggs[a].agg_type == BK_AGG_AVG) {
+                        double d; memcpy(&d, &sv, 8);
+                        if (LDS) atomic_add_f64_lds(val, d);
+                        else     atomic_add_f64_global(val, d);
+                    } else {
+                        atomicAdd((unsigned long long*)val, (unsigned long long)sv);
+                    }
+                    atomicAdd((unsigned long long*)cnt, (unsigned long long)sc);
+                }
+                break;
+            case BK_AGG_MIN:
+            case BK_AGG_MAX:
+                if (sc) {
+                    atomicMax((unsigned long long*)val, (unsigned long long)sv);
+                    atomicAdd((unsigned long long*)cnt, (unsigned long long)sc);
+                }
+                break;
+            default: break;
+        }
+    }
+}
+
+/* ---- global-table claim: find or insert (k0,k1,flag), returns slot base.
+ * 8-byte-granule publish protocol (guide G16 R1): sc1 relaxed stores of
+ * flag/k0/k1, per-wave vmcnt drain, sc1 state=2; readers use sc1 relaxed
+ * loads only (no fences on the probe path). Deadlock-free within a wave:
+ * no lane ever waits in-place on a state it cannot observe progress on —
+ * `building` states are retried through the outer loop. */
+__device__ uint64_t* gtable_claim(uint64_t* table, uint64_t mask, int stride,
+                                  uint32_t flag, uint64_t k0, uint64_t k1,
+                                  uint64_t* fill, uint64_t fill_cap,
+                                  uint32_t* err) {
+    uint64_t slot = key_hash(flag, k0, k1) & mask;
+    for (int it = 0; it < MAX_PROBE; ++it) {
+        uint64_t* s = table + slot * (uint64_t)stride;
+        uint32_t* statep = (uint32_t*)s;
+        uint32_t st = __hip_atomic_load(statep, RLX, AGT);
+        if (st == 0) {
+            uint32_t old = atomicCAS(statep, 0u, 1u);
+            if (old == 0) {
+                uint64_t f = atomicAdd((unsigned long long*)fill, 1ull);
+                if (f + 1 > fill_cap)
+                    __hip_atomic_store(err, 2u, RLX, AGT);  /* overflow: rerun bigger */
+                __hip_atomic_store(statep + 1, flag, RLX, AGT);
+                __hip_atomic_store(&s[1], k0, RLX, AGT);
+                __hip_atomic_store(&s[2], k1, RLX, AGT);
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                __hip_atomic_store(statep, 2u, RLX, AGT);
+                return s;
+            }
+            st = old;
+        }
+        if (st == 2u) {
+            uint32_t f = __hip_atomic_load(statep + 1, RLX, AGT);
+            uint64_t a = __hip_atomic_load(&s[1], RLX, AGT);
+            uint64_t b = __hip_atomic_load(&s[2], RLX, AGT);
+            if (f == flag && a == k0 && b == k1) return s;
+            slot = (slot + 1) & mask;
+            continue;
+        }
+        /* st == 1: another wave is publishing this slot; yield briefly and
+         * retry (that wave is independently scheduled and will finish). */
+        __builtin_amdgcn_s_sleep(1);
+    }
+    __hip_atomic_store(err, 1u, RLX, AGT);  /* probe-limit livelock guard */
+    return nullptr;
+}
+
+/* ---- LDS-table claim (workgroup scope; CU-local, cheap) ---- */
+__device__ uint64_t* ltable_claim(uint64_t* ltab, uint32_t lmask, int stride,
+                                  uint32_t flag, uint64_t k0, uint64_t k1,
+                                  uint32_t* lfill, uint32_t lcap) {
+    uint64_t slot = (key_hash(flag, k0, k1) >> 32) & lmask;
+    for (int it = 0; it < 64; ++it) {
+        uint64_t* s = ltab + slot * (uint64_t)stride;
+        uint32_t* statep = (uint32_t*)s;
+        uint32_t st = __hip_atomic_load(statep, RLX, WGP);
+        if (st == 0) {
+            /* stop inserting NEW keys once the LDS table is crowded (long
+             * probes defeat the point) — overflow keys go to the global
+             * table; partials merge correctly since all ops are additive. */
+            if (__hip_atomic_load(lfill, RLX, WGP) >= lcap) return nullptr;
+            uint32_t old = atomicCAS(statep, 0u, 1u);
+            if (old == 0) {
+                atomicAdd(lfill, 1u);
+                statep[1] = flag;
+                s[1] = k0;
+                s[2] = k1;
+                __hip_atomic_store(statep, 2u, __ATOMIC_RELEASE, WGP);
+                return s;
+            }
+            st = old;
+        }
+        if (st == 2u) {
+            (void)__hip_atomic_load(statep, __ATOMIC_ACQUIRE, WGP);
+            if (statep[1] == flag && s[1] == k0 && s[2] == k1) return s;
+            slot = (slot + 1) & lmask;
+            continue;
+        }
+        /* st == 1: publisher is another wave (same CU); retry. */
+    }
+    return nullptr;  /* fall through to global table */
+}
+
+/* ------------------------------------------------------------------ */
+/* kernels                                                             */
+/* ------------------------------------------------------------------ */
+
+/* data generation: one pass, all columns (bk_datagen.h shared generator) */
+struct DevSpecs { BkColSpec s[BK_MAX_COLS]; };
+
+__global__ void k_generate(DevCols cols, int ncols, DevSpecs specs,
+                           int64_t nrows, uint64_t seed, int64_t row_begin) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nrows;
+         i += stride) {
+        uint64_t grow = (uint64_t)(row_begin + i);
+        for (int c = 0; c < ncols; c++) {
+            const BkColSpec& cs = specs.s[c];
+            if (cs.col_type == BK_DOUBLE) {
+                ((double*)cols.c[c].data)[i] = bk_gen_f64(&cs, seed, grow, (uint32_t)c);
+            } else if (cs.col_type == BK_STRING) {
+                ((int32_t*)cols.c[c].data)[i] =
+                    (int32_t)bk_gen_i64(&cs, seed, grow, (uint32_t)c);
+            } else {
+                ((int64_t*)cols.c[c].data)[i] = bk_gen_i64(&cs, seed, grow, (uint32_t)c);
+            }
+            if (cols.c[c].valid)
+                ((uint8_t*)cols.c[c].valid)[i] =
+                    (uint8_t)bk_cell_valid(seed, grow, (uint32_t)c, cs.null_frac_x1e6);
+        }
+    }
+}
+
+/* fused filter + grouped hash aggregate.
+ * dynamic LDS: LDS_SLOTS * stride u64 words (per-workgroup pre-agg table). */
+__global__ void __launch_bounds__(256)
+k_filter_agg_group(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
+                   uint64_t* gtable, uint64_t gmask, uint64_t fill_cap,
+                   uint64_t* fill, uint64_t* rows_passed, uint32_t* err,
+                   uint32_t lds_slots) {
+    /* ALL LDS in one dynamic carve (guide G17: a static __shared__ ahead of
+     * the dynamic region misaligns the u64 table -> 64-cycle replays):
+     *   [ lds_slots*stride table ][ laux[0]=rows_passed ][ laux[1].lo=fill ] */
+    extern __shared__ __attribute__((aligned(16))) uint64_t ltab[];
+    const int stride = SLOT_HDR + 2 * q.n_aggs;
+    uint64_t* laux = ltab + (size_t)lds_slots * stride;
+    uint32_t* lfill = (uint32_t*)&laux[1];
+    /* zero LDS table + aux */
+    for (uint32_t w = threadIdx.x; w < lds_slots * (uint32_t)stride + 2;
+         w += blockDim.x)
+        ltab[w] = 0;
+    __syncthreads();
+
+    const uint32_t lmask = lds_slots - 1;
+    const uint32_t lcap = (lds_slots * 3u) / 4u;
+    int64_t my_passed = 0;
+    int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < row_end; r += gstride) {
+        if (!row_passes(cols, q, r)) continue;
+        my_passed++;
+        /* group key (exec_node.cpp:555-571: null-flag byte + encoded values) */
+        uint32_t flag = 0;
+        uint64_t k0 = 0, k1 = 0;
+        if (q.n_group >= 1) {
+            const DevCol& c = cols.c[q.group_cols[0]];
+            if (!cell_valid(c, r)) flag |= 0x80u;      /* 0x01 << (7-0) */
+            else k0 = enc_value(c, r);
+        }
+        if (q.n_group >= 2) {
+            const DevCol& c = cols.c[q.group_cols[1]];
+            if (!cell_valid(c, r)) flag |= 0x40u;      /* 0x01 << (7-1) */
+            else k1 = enc_value(c, r);
+        }
+        uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1, lfill, lcap);
+        if (slot) {
+            agg_update_slot<true>(slot, q, cols, r);
+        } else {
+            slot = gtable_claim(gtable, gmask, stride, flag, k0, k1, fill, fill_cap, err);
+            if (slot) agg_update_slot<false>(slot, q, cols, r);
+            else break;  /* table exploded; err set, host reruns */
+        }
+    }
+    /* block-level rows_passed reduction into laux[0] */
+    long long w = my_passed;
+    for (int off = 32; off > 0; off >>= 1) w += __shfl_down(w, off, 64);
+    if ((threadIdx.x & 63) == 0)
+        atomicAdd((unsigned long long*)&laux[0], (unsigned long long)w);
+    __syncthreads();
+    if (threadIdx.x == 0)
+        atomicAdd((unsigned long long*)rows_passed, (unsigned long long)laux[0]);
+    /* flush LDS partials into the global table */
+    for (uint32_t sl = threadIdx.x; sl < lds_slots; sl += blockDim.x) {
+        uint64_t* s = ltab + (uint64_t)sl * stride;
+        uint32_t st = ((uint32_t*)s)[0];
+        if (st != 2u) continue;
+        uint32_t flag = ((uint32_t*)s)[1];
+        uint64_t* g = gtable_claim(gtable, gmask, stride, flag, s[1], s[2],
+                                   fill, fill_cap, err);
+        if (!g) break;
+        agg_merge_slot<false>(g, s + SLOT_HDR, q);
+    }
+}
+
+/* no-GROUP-BY aggregate: register accumulation, one wave-reduce, one merge
+ * per wave into the single pre-initialized slot 0 (config-1 COUNT(*) path). */
+__global__ void __launch_bounds__(256)
+k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
+                    uint64_t* gtable, uint64_t* rows_passed) {
+    int64_t my_passed = 0;
+    uint64_t acc_v[BK_MAX_AGGS];   /* wrapping-int or cnt accum; doubles separate */
+    uint64_t acc_c[BK_MAX_AGGS];
+    double   acc_d[BK_MAX_AGGS];
+    for (int a = 0; a < q.n_aggs; a++) { acc_v[a] = 0; acc_c[a] = 0; acc_d[a] = 0.0; }
+
+    int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < row_end; r += gstride) {
+        if (!row_passes(cols, q, r)) continue;
+        my_passed++;
+        for (int32_t a = 0; a < q.n_aggs; a++) {
+            const BkAggSpec& as = q.aggs[a];
+            switch (as.agg_type) {
+                case BK_AGG_COUNT_STAR: acc_v[a]++; break;
+                case BK_AGG_COUNT:
+                    if (cell_valid(cols.c[as.col], r)) acc_v[a]++;
+                    break;
+                case BK_AGG_SUM: {
+                    const DevCol& c = cols.c[as.col];
+                    if (!cell_valid(c, r)) break;
+                    if (q.agg_in_types[a] == BK_DOUBLE) acc_d[a] += ((const double*)c.data)[r];
+                    else acc_v[a] += (uint64_t)cell_i64(c, r);
+                    acc_c[a]++;
+                    break;
+                }
+                case BK_AGG_AVG: {
+                    const DevCol& c = cols.c[as.col];
+                    if (!cell_valid(c, r)) break;
+                    acc_d[a] += cell_f64(c, r);
+                    acc_c[a]++;
+                    break;
+                }
+                case BK_AGG_MIN: {
+                    const DevCol& c = cols.c[as.col];
+                    if (!cell_valid(c, r)) break;
+                    uint64_t e = ~enc_value(c, r);
+                    if (e > acc_v[a]) acc_v[a] = e;
+                    acc_c[a]++;
+                    break;
+                }
+                case BK_AGG_MAX: {
+                    const DevCol& c = cols.c[as.col];
+                    if (!cell_valid(c, r)) break;
+                    uint64_t e = enc_value(c, r);
+                    if (e > acc_v[a]) acc_v[a] = e;
+                    acc_c[a]++;
+                    break;
+                }
+                default: break;
+            }
+        }
+    }
+    /* wave reduce + one atomic merge per wave into slot 0 */
+    for (int off = 32; off > 0; off >>= 1) my_passed += __shfl_down(my_passed, off, 64);
+    for (int32_t a = 0; a < q.n_aggs; a++) {
+        int at = q.aggs[a].agg_type;
+        for (int off = 32; off > 0; off >>= 1) {
+            uint64_t ov = __shfl_down(acc_v[a], off, 64);
+            acc_c[a] += __shfl_down(acc_c[a], off, 64);
+            acc_d[a] += __shfl_down(acc_d[a], off, 64);
+            if (at == BK_AGG_MIN || at == BK_AGG_MAX) acc_v[a] = acc_v[a] > ov ? acc_v[a] : ov;
+            else acc_v[a] += ov;
+        }
+    }
+    if ((threadIdx.x & 63) == 0) {
+        atomicAdd((unsigned long long*)rows_passed, (unsigned long long)my_passed);
+        uint64_t* s = gtable;  /* slot 0, pre-initialized state=2 flag=0 */
+        for (int32_t a = 0; a < q.n_aggs; a++) {
+            uint64_t* val = s + SLOT_HDR + 2 * a;
+            uint64_t* cnt = val + 1;
+            int at = q.aggs[a].agg_type;
+            if (at == BK_AGG_COUNT_STAR || at == BK_AGG_COUNT) {
+                if (acc_v[a]) atomicAdd((unsigned long long*)val, (unsigned long long)acc_v[a]);
+            } else if (at == BK_AGG_MIN || at == BK_AGG_MAX) {
+                if (acc_c[a]) {
+                    atomicMax((unsigned long long*)val, (unsigned long long)acc_v[a]);
+                    atomicAdd((unsigned long long*)cnt, (unsigned long long)acc_c[a]);
+                }
+            } else {  /* SUM / AVG */
+                if (acc_c[a]) {
+                    if (q.agg_in_types[a] == BK_DOUBLE || at == BK_AGG_AVG)
+                        atomic_add_f64_global(val, acc_d[a]);
+                    else
+                        atomicAdd((unsigned long long*)val, (unsigned long long)acc_v[a]);
+                    atomicAdd((unsigned long long*)cnt, (unsigned long long)acc_c[a]);
+                }
+            }
+        }
+    }
+}
+
+/* compact the global table into the wire blob:
+ * [flags u32*n][k0 u64*n][k1 u64*n][states u64*n*2*naggs] */
+__global__ void k_compact(const uint64_t* gtable, uint64_t nslots, int naggs,
+                          uint64_t* counter, uint32_t* out_flags, uint64_t* out_k0,
+                          uint64_t* out_k1, uint64_t* out_states, uint64_t cap) {
+    const int stride = SLOT_HDR + 2 * naggs;
+    uint64_t gs = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t sl = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; sl < nslots;
+         sl += gs) {
+        const uint64_t* s = gtable + sl * stride;
+        if (((const uint32_t*)s)[0] != 2u) continue;
+        uint64_t i = atomicAdd((unsigned long long*)counter, 1ull);
+        if (i >= cap) continue;
+        out_flags[i] = ((const uint32_t*)s)[1];
+        out_k0[i] = s[1];
+        out_k1[i] = s[2];
+        for (int w = 0; w < 2 * naggs; w++)
+            out_states[i * (uint64_t)(2 * naggs) + w] = s[SLOT_HDR + w];
+    }
+}
+
+/* merge a peer's compact blob into the global table (MERGE_AGG path) */
+__global__ void k_merge_blob(BkQuerySpec q, const uint32_t* flags, const uint64_t* k0,
+                             const uint64_t* k1, const uint64_t* states, int64_t n,
+                             uint64_t* gtable, uint64_t gmask, uint64_t fill_cap,
+                             uint64_t* fill, uint32_t* err) {
+    const int stride = SLOT_HDR + 2 * q.n_aggs;
+    int64_t gs = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += gs) {
+        uint64_t* s = gtable_claim(gtable, gmask, stride, flags[i], k0[i], k1[i],
+                                   fill, fill_cap, err);
+        if (!s) return;
+        agg_merge_slot<false>(s, states + i * (uint64_t)(2 * q.n_aggs), q);
+    }
+}
+
+/* ------------------------------------------------------------------ */
+/* host: tables                                                        */
+/* ------------------------------------------------------------------ */
+
+struct BkgTable {
+    int ncols = 0;
+    int64_t nrows = 0;
+    BkColSpec specs[BK_MAX_COLS];
+    void* data[BK_MAX_COLS] = {};
+    uint8_t* valid[BK_MAX_COLS] = {};
+};
+
+static int g_device_set = 0;
+static int ensure_device() {
+    if (!g_device_set) {
+        HIP_CHECK(hipSetDevice(0));
+        g_device_set = 1;
+    }
+    return 0;
+}
+
+extern "C" int bkgpu_device_count(void) {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+    return n;
+}
+
+extern "C" int bkgpu_set_device(int dev) {
+    HIP_CHECK(hipSetDevice(dev));
+    g_device_set = 1;
+    return 0;
+}
+
+extern "C" int bkgpu_sync(void) { HIP_CHECK(hipDeviceSynchronize()); return 0; }
+
+static size_t elem_size(int32_t t) {
+    switch (t) {
+        case BK_INT64: case BK_DOUBLE: return 8;
+        case BK_STRING: return 4;
+        default: return 0;
+    }
+}
+
+extern "C" BkgTable* bkgpu_table_create(int ncols, const BkColSpec* specs,
+                                        int64_t nrows) {
+    if (ensure_device() != 0) return nullptr;
+    if (ncols <= 0 || ncols > BK_MAX_COLS) { set_err("bad ncols"); return nullptr; }
+    BkgTable* t = new BkgTable();
+    t->ncols = ncols;
+    t->nrows = nrows;
+    for (int c = 0; c < ncols; c++) {
+        t->specs[c] = specs[c];
+        size_t es = elem_size(specs[c].col_type);
+        if (es == 0) { set_err("unsupported col type"); delete t; return nullptr; }
+        if (hipMalloc(&t->data[c], (size_t)nrows * es) != hipSuccess) {
+            set_err("hipMalloc column failed");
+            bkgpu_table_free(t);
+            return nullptr;
+        }
+        if (specs[c].null_frac_x1e6 > 0) {
+            if (hipMalloc((void**)&t->valid[c], (size_t)nrows) != hipSuccess) {
+                set_err("hipMalloc validity failed");
+                bkgpu_table_free(t);
+                return nullptr;
+            }
+        }
+    }
+    return t;
+}
+
+extern "C" int64_t bkgpu_table_nrows(const BkgTable* t) { return t ? t->nrows : 0; }
+
+extern "C" void bkgpu_table_free(BkgTable* t) {
+    if (!t) return;
+    for (int c = 0; c < t->ncols; c++) {
+        if (t->data[c]) hipFree(t->data[c]);
+        if (t->valid[c]) hipFree(t->valid[c]);
+    }
+    delete t;
+}
+
+static DevCols table_cols(const BkgTable* t) {
+    DevCols dc{};
+    for (int c = 0; c < t->ncols; c++) {
+        dc.c[c].type = t->specs[c].col_type;
+        dc.c[c].data = t->data[c];
+        dc.c[c].valid = t->valid[c];
+    }
+    return dc;
+}
+
+extern "C" int bkgpu_table_generate(BkgTable* t, uint64_t seed, int64_t row_begin) {
+    DevCols dc = table_cols(t);
+    DevSpecs ds{};
+    for (int c = 0; c < t->ncols; c++) ds.s[c] = t->specs[c];
+    /* memory-bound: ~2048 blocks grid-stride (guide G11) */
+    int blocks = 2048, threads = 256;
+    hipLaunchKernelGGL(k_generate, dim3(blocks), dim3(threads), 0, 0,
+                       dc, t->ncols, ds, t->nrows, seed, row_begin);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipDeviceSynchronize());
+    return 0;
+}
+
+extern "C" int bkgpu_table_upload(BkgTable* t, int col, const void* data,
+                                  const uint8_t* valid) {
+    size_t es = elem_size(t->specs[col].col_type);
+    HIP_CHECK(hipMemcpy(t->data[col], data, (size_t)t->nrows * es,
+                        hipMemcpyHostToDevice));
+    if (valid) {
+        if (!t->valid[col]) HIP_CHECK(hipMalloc((void**)&t->valid[col], (size_t)t->nrows));
+        HIP_CHECK(hipMemcpy(t->valid[col], valid, (size_t)t->nrows,
+                            hipMemcpyHostToDevice));
+    }
+    return 0;
+}
+
+/* ------------------------------------------------------------------ */
+/* host: aggregation                                                   */
+/* ------------------------------------------------------------------ */
+
+struct BkgAggOut {
+    BkQuerySpec q;
+    uint64_t* table = nullptr;     /* device */
+    uint64_t nslots = 0;
+    uint64_t* ctrs = nullptr;      /* device: [fill, rows_passed, compact_n] */
+    uint32_t* err = nullptr;       /* device */
+    /* compact blob (device) */
+    uint8_t* blob = nullptr;
+    int64_t blob_groups = 0;       /* capacity in groups */
+    int64_t ngroups = -1;          /* valid after compact */
+    int64_t rows_passed = 0;
+    float kernel_ms = 0.f;
+    bool dirty = true;             /* table modified since last compact */
+};
+
+static int64_t next_pow2(int64_t x) {
+    int64_t p = 1;
+    while (p < x) p <<= 1;
+    return p;
+}
+
+static size_t blob_bytes_for(int naggs, int64_t n) {
+    return (size_t)n * (4 + 8 + 8 + 8 * 2 * (size_t)naggs);
+}
+
+static int agg_alloc(BkgAggOut* o, int64_t nslots) {
+    const int stride = SLOT_HDR + 2 * o->q.n_aggs;
+    o->nslots = (uint64_t)nslots;
+    HIP_CHECK(hipMalloc(&o->table, (size_t)nslots * stride * 8));
+    HIP_CHECK(hipMemset(o->table, 0, (size_t)nslots * stride * 8));
+    if (!o->ctrs) {
+        HIP_CHECK(hipMalloc(&o->ctrs, 3 * 8));
+        HIP_CHECK(hipMalloc((void**)&o->err, 4));
+    }
+    HIP_CHECK(hipMemset(o->ctrs, 0, 3 * 8));
+    HIP_CHECK(hipMemset(o->err, 0, 4));
+    return 0;
+}
+
+static void agg_release_table(BkgAggOut* o) {
+    if (o->table) { hipFree(o->table); o->table = nullptr; }
+}
+
+extern "C" void bkgpu_agg_free(BkgAggOut* o) {
+    if (!o) return;
+    agg_release_table(o);
+    if (o->ctrs) hipFree(o->ctrs);
+    if (o->err) hipFree(o->err);
+    if (o->blob) hipFree(o->blob);
+    delete o;
+}
+
+/* run compact if table changed; updates o->ngroups and o->blob */
+static int agg_compact(BkgAggOut* o) {
+    if (!o->dirty && o->ngroups >= 0) return 0;
+    const int naggs = o->q.n_aggs;
+    int64_t cap = (int64_t)o->nslots;
+    /* read fill count to size the blob */
+    uint64_t ctr_host[3];
+    HIP_CHECK(hipMemcpy(ctr_host, o->ctrs, 24, hipMemcpyDeviceToHost));
+    int64_t fill = (int64_t)ctr_host[0];
+    if (o->q.n_group == 0) fill = 1;
+    cap = fill > 0 ? fill : 1;
+    if (o->blob) { hipFree(o->blob); o->blob = nullptr; }
+    HIP_CHECK(hipMalloc(&o->blob, blob_bytes_for(naggs, cap)));
+    o->blob_groups = cap;
+    HIP_CHECK(hipMemset(o->ctrs + 2, 0, 8));
+    uint32_t* flags = (uint32_t*)o->blob;
+    uint64_t* k0 = (uint64_t*)(flags + cap);
+    uint64_t* k1 = k0 + cap;
+    uint64_t* states = k1 + cap;
+    hipLaunchKernelGGL(k_compact, dim3(1024), dim3(256), 0, 0,
+                       o->table, o->nslots, naggs, o->ctrs + 2,
+                       flags, k0, k1, states, (uint64_t)cap);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipMemcpy(ctr_host, o->ctrs, 24, hipMemcpyDeviceToHost));
+    o->ngroups = (int64_t)ctr_host[2];
+    o->rows_passed = (int64_t)ctr_host[1];
+    o->dirty = false;
+    return 0;
+}
+
+extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
+                                       int64_t row_begin, int64_t row_end,
+                                       int64_t expected_groups) {
+    if (ensure_device() != 0) return nullptr;
+    BkgAggOut* o = new BkgAggOut();
+    o->q = *q;
+    const int stride = SLOT_HDR + 2 * q->n_aggs;
+    int64_t nslots = next_pow2(std::max<int64_t>(1024, expected_groups * 2));
+    DevCols dc = table_cols(t);
+
+    for (int attempt = 0; attempt < 4; attempt++) {
+        if (agg_alloc(o, nslots) != 0) { bkgpu_agg_free(o); return nullptr; }
+        hipEvent_t ev0, ev1;
+        HIP_CHECK_NULL(hipEventCreate(&ev0));
+        HIP_CHECK_NULL(hipEventCreate(&ev1));
+        uint32_t lds_slots = 512;
+        /* keep >=2 blocks/CU: LDS per block <= 64 KiB */
+        while ((size_t)lds_slots * stride * 8 > 63 * 1024) lds_slots >>= 1;
+        size_t lds_bytes = ((size_t)lds_slots * stride + 2) * 8;
+        int blocks = 2048, threads = 256;
+        HIP_CHECK_NULL(hipEventRecord(ev0));
+        if (q->n_group == 0) {
+            /* pre-initialize slot 0 as the single group (state=2, flag 0):
+             * mirrors agg_node.cpp:490-505's always-present row */
+            uint64_t hdr[3] = {2ull /* state=2,flag=0 */, 0, 0};
+            HIP_CHECK_NULL(hipMemcpy(o->table, hdr, 24, hipMemcpyHostToDevice));
+            uint64_t one = 1;
+            HIP_CHECK_NULL(hipMemcpy(o->ctrs, &one, 8, hipMemcpyHostToDevice));
+            hipLaunchKernelGGL(k_filter_agg_scalar, dim3(blocks), dim3(threads), 0, 0,
+                               dc, *q, row_begin, row_end, o->table, o->ctrs + 1);
+        } else {
+            hipLaunchKernelGGL(k_filter_agg_group, dim3(blocks), dim3(threads),
+                               lds_bytes, 0,
+                               dc, *q, row_begin, row_end, o->table,
+                               o->nslots - 1, (o->nslots * 7) / 8,
+                               o->ctrs, o->ctrs + 1, o->err, lds_slots);
+        }
+        HIP_CHECK_NULL(hipEventRecord(ev1));
+        hipError_t lerr = hipGetLastError();
+        if (lerr != hipSuccess) {
+            snprintf(g_err, sizeof g_err, "filter_agg launch: %s", hipGetErrorString(lerr));
+            hipEventDestroy(ev0); hipEventDestroy(ev1);
+            bkgpu_agg_free(o);
+            return nullptr;
+        }
+        HIP_CHECK_NULL(hipEventSynchronize(ev1));
+        HIP_CHECK_NULL(hipEventElapsedTime(&o->kernel_ms, ev0, ev1));
+        hipEventDestroy(ev0); hipEventDestroy(ev1);
+        uint32_t err_host = 0;
+        HIP_CHECK_NULL(hipMemcpy(&err_host, o->err, 4, hipMemcpyDeviceToHost));
+        if (err_host == 0) break;
+        /* overflow / probe livelock: grow 4x and rerun */
+        agg_release_table(o);
+        nslots *= 4;
+        if (attempt == 3) {
+            set_err("hash table overflow after retries");
+            bkgpu_agg_free(o);
+            return nullptr;
+        }
+    }
+    o->dirty = true;
+    if (agg_compact(o) != 0) { bkgpu_agg_free(o); return nullptr; }
+    return o;
+}
+
+extern "C" int64_t bkgpu_agg_ngroups(const BkgAggOut* o) { return o->ngroups; }
+extern "C" int64_t bkgpu_agg_rows_passed(const BkgAggOut* o) { return o->rows_passed; }
+extern "C" double bkgpu_agg_kernel_ms(const BkgAggOut* o) { return o->kernel_ms; }
+
+extern "C" int64_t bkgpu_agg_export_bytes(const BkgAggOut* o) {
+    return (int64_t)blob_bytes_for(o->q.n_aggs, o->ngroups);
+}
+
+extern "C" int bkgpu_agg_export(const BkgAggOut* o, void* dst, int64_t cap) {
+    int64_t need = bkgpu_agg_export_bytes(o);
+    if (cap < need) { set_err("export buffer too small"); return -1; }
+    /* blob already compacted with capacity == ngroups layout? The blob's
+     * internal layout used blob_groups as the stride; re-pack if they differ. */
+    if (o->blob_groups == o->ngroups) {
+        HIP_CHECK(hipMemcpy(dst, o->blob, (size_t)need, hipMemcpyDeviceToDevice));
+        return 0;
+    }
+    /* repack section by section */
+    int64_t n = o->ngroups, c = o->blob_groups;
+    const uint8_t* b = o->blob;
+    uint8_t* d = (uint8_t*)dst;
+    HIP_CHECK(hipMemcpy(d, b, (size_t)n * 4, hipMemcpyDeviceToDevice));
+    HIP_CHECK(hipMemcpy(d + n * 4, b + c * 4, (size_t)n * 8, hipMemcpyDeviceToDevice));
+    HIP_CHECK(hipMemcpy(d + n * 12, b + c * 12, (size_t)n * 8, hipMemcpyDeviceToDevice));
+    HIP_CHECK(hipMemcpy(d + n * 20, b + c * 20,
+                        (size_t)n * 16 * o->q.n_aggs, hipMemcpyDeviceToDevice));
+    return 0;
+}
+
+extern "C" int bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups) {
+    const uint8_t* b = (const uint8_t*)blob;
+    const uint32_t* flags = (const uint32_t*)b;
+    const uint64_t* k0 = (const uint64_t*)(b + n_groups * 4);
+    const uint64_t* k1 = (const uint64_t*)(b + n_groups * 12);
+    const uint64_t* states = (const uint64_t*)(b + n_groups * 20);
+    hipLaunchKernelGGL(k_merge_blob, dim3(512), dim3(256), 0, 0,
+                       o->q, flags, k0, k1, states, n_groups,
+                       o->table, o->nslots - 1, (o->nslots * 7) / 8,
+                       o->ctrs, o->err);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipDeviceSynchronize());
+    uint32_t err_host = 0;
+    HIP_CHECK(hipMemcpy(&err_host, o->err, 4, hipMemcpyDeviceToHost));
+    if (err_host) { set_err("merge overflowed dst table"); return -1; }
+    o->dirty = true;
+    return 0;
+}
+
+/* ---- fetch: download + finalize (agg_fn_call.cpp:927-975) ---- */
+
+struct HostGroups {
+    std::vector<uint32_t> flags;
+    std::vector<uint64_t> k0, k1, states;
+};
+
+static int download_groups(BkgAggOut* o, HostGroups& hg) {
+    if (agg_compact(o) != 0) return -1;
+    int64_t n = o->ngroups, c = o->blob_groups;
+    int naggs = o->q.n_aggs;
+    hg.flags.resize(n);
+    hg.k0.resize(n);
+    hg.k1.resize(n);
+    hg.states.resize((size_t)n * 2 * naggs);
+    const uint8_t* b = o->blob;
+    HIP_CHECK(hipMemcpy(hg.flags.data(), b, (size_t)n * 4, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(hg.k0.data(), b + c * 4, (size_t)n * 8, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(hg.k1.data(), b + c * 12, (size_t)n * 8, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(hg.states.data(), b + c * 20, (size_t)n * 16 * naggs,
+                        hipMemcpyDeviceToHost));
+    return 0;
+}
+
+extern "C" int64_t bkgpu_agg_fetch(BkgAggOut* o, int sorted, int64_t max_groups,
+                                   uint8_t* flags, uint64_t* enc,
+                                   int64_t* out_i, double* out_d, uint8_t* out_has) {
+    HostGroups hg;
+    if (download_groups(o, hg) != 0) return -1;
+    int64_t n = o->ngroups;
+    const int naggs = o->q.n_aggs;
+    std::vector<int64_t> order(n);
+    for (int64_t i = 0; i < n; i++) order[i] = i;
+    if (sorted) {
+        const BkQuerySpec& q = o->q;
+        std::sort(order.begin(), order.end(), [&](int64_t a, int64_t b) {
+            /* canonical MutTableKey byte order (see oracle orc_group_cmp) */
+            if (hg.flags[a] != hg.flags[b]) return hg.flags[a] < hg.flags[b];
+            for (int k = 0; k < q.n_group; k++) {
+                if ((hg.flags[a] >> (7 - k)) & 1) continue;
+                uint64_t ea = k == 0 ? hg.k0[a] : hg.k1[a];
+                uint64_t eb = k == 0 ? hg.k0[b] : hg.k1[b];
+                if (ea != eb) return ea < eb;
+            }
+            return false;
+        });
+    }
+    int64_t out_n = std::min(n, max_groups);
+    for (int64_t i = 0; i < out_n; i++) {
+        int64_t g = order[i];
+        flags[i] = (uint8_t)hg.flags[g];
+        enc[i * BK_MAX_GROUP + 0] = hg.k0[g];
+        enc[i * BK_MAX_GROUP + 1] = hg.k1[g];
+        for (int a = 0; a < naggs; a++) {
+            uint64_t val = hg.states[(size_t)g * 2 * naggs + 2 * a];
+            uint64_t cnt = hg.states[(size_t)g * 2 * naggs + 2 * a + 1];
+            int64_t idx = (int64_t)a * out_n + i;
+            int at = o->q.aggs[a].agg_type;
+            int vt = o->q.agg_in_types[a];
+            out_i[idx] = 0; out_d[idx] = 0.0; out_has[idx] = 0;
+            switch (at) {
+                case BK_AGG_COUNT_STAR:
+                case BK_AGG_COUNT:
+                    out_i[idx] = (int64_t)val; out_has[idx] = 1; break;
+                case BK_AGG_SUM:
+                    if (!cnt) break;
+                    if (vt == BK_DOUBLE) memcpy(&out_d[idx], &val, 8);
+                    else out_i[idx] = (int64_t)val;
+                    out_has[idx] = 1; break;
+                case BK_AGG_AVG:
+                    if (!cnt) break;
+                    { double s; memcpy(&s, &val, 8); out_d[idx] = s / (double)cnt; }
+                    out_has[idx] = 1; break;
+                case BK_AGG_MIN:
+                case BK_AGG_MAX: {
+                    if (!cnt) break;
+                    uint64_t e = (at == BK_AGG_MIN) ? ~val : val;
+                    if (vt == BK_DOUBLE) out_d[idx] = bk_dec_f64(e);
+                    else if (vt == BK_STRING) out_i[idx] = (int64_t)(uint32_t)e;
+                    else out_i[idx] = bk_dec_i64(e);
+                    out_has[idx] = 1; break;
+                }
+                default: break;
+            }
+        }
+    }
+    return out_n;
+}
+
+/* ------------------------------------------------------------------ */
+/* top-N (placeholder — implemented in bkgpu_sort.hip part 2)          */
+/* ------------------------------------------------------------------ */
+
+extern "C" int64_t bkgpu_sort_topk(BkgTable*, const BkQuerySpec*, const BkOrderSpec*,
+                                   int, int64_t, int64_t, int64_t, int64_t*) {
+    set_err("bkgpu_sort_topk: not implemented yet");
+    return -1;
+}
+extern "C" double bkgpu_topk_kernel_ms(void) { return 0.0; }

@@ -1,0 +1,213 @@
+# engine.py — ctypes host driver over the C-ABI of include/bkgpu.h.
+# This is plumbing for tests/bench; the production embedding is the C++ host
+# layer mirroring ExecNode. No CPU fallback exists: if
+# the HIP extension is missing on a GPU machine, constructing GpuEngine fails.
+import ctypes as C
+import os
+
+import numpy as np
+
+from .plan import BkQuerySpec, BkOrderSpec, QueryPlan, BK_MAX_GROUP
+from .plan import TYPE_INT64, TYPE_DOUBLE, TYPE_STRING
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_LIB = os.path.join(_HERE, "libbkgpu.so")
+
+
+class NativeEngineMissing(RuntimeError):
+    pass
+
+
+class _BkColSpec(C.Structure):
+    _fields_ = [("col_type", C.c_int32), ("dist", C.c_int32),
+                ("p0", C.c_int64), ("p1", C.c_int64),
+                ("null_frac_x1e6", C.c_int32), ("_pad", C.c_int32)]
+
+
+def _load():
+    if not os.path.exists(_LIB):
+        raise NativeEngineMissing(
+            f"native HIP engine not built: {_LIB} missing — run "
+            f"__graft_entry__.build() (there is no CPU fallback)")
+    lib = C.CDLL(_LIB)
+    lib.bkgpu_last_error.restype = C.c_char_p
+    lib.bkgpu_device_count.restype = C.c_int
+    lib.bkgpu_set_device.argtypes = [C.c_int]
+    lib.bkgpu_table_create.restype = C.c_void_p
+    lib.bkgpu_table_create.argtypes = [C.c_int, C.POINTER(_BkColSpec), C.c_int64]
+    lib.bkgpu_table_generate.argtypes = [C.c_void_p, C.c_uint64, C.c_int64]
+    lib.bkgpu_table_upload.argtypes = [C.c_void_p, C.c_int, C.c_void_p, C.c_void_p]
+    lib.bkgpu_table_nrows.restype = C.c_int64
+    lib.bkgpu_table_nrows.argtypes = [C.c_void_p]
+    lib.bkgpu_table_free.argtypes = [C.c_void_p]
+    lib.bkgpu_filter_agg.restype = C.c_void_p
+    lib.bkgpu_filter_agg.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
+                                     C.c_int64, C.c_int64, C.c_int64]
+    lib.bkgpu_agg_ngroups.restype = C.c_int64
+    lib.bkgpu_agg_ngroups.argtypes = [C.c_void_p]
+    lib.bkgpu_agg_rows_passed.restype = C.c_int64
+    lib.bkgpu_agg_rows_passed.argtypes = [C.c_void_p]
+    lib.bkgpu_agg_kernel_ms.restype = C.c_double
+    lib.bkgpu_agg_kernel_ms.argtypes = [C.c_void_p]
+    lib.bkgpu_agg_export_bytes.restype = C.c_int64
+    lib.bkgpu_agg_export_bytes.argtypes = [C.c_void_p]
+    lib.bkgpu_agg_export.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
+    lib.bkgpu_agg_merge.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
+    lib.bkgpu_agg_fetch.restype = C.c_int64
+    lib.bkgpu_agg_fetch.argtypes = [C.c_void_p, C.c_int, C.c_int64,
+                                    C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+                                    C.POINTER(C.c_int64), C.POINTER(C.c_double),
+                                    C.POINTER(C.c_uint8)]
+    lib.bkgpu_agg_free.argtypes = [C.c_void_p]
+    lib.bkgpu_sort_topk.restype = C.c_int64
+    lib.bkgpu_sort_topk.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
+                                    C.POINTER(BkOrderSpec), C.c_int,
+                                    C.c_int64, C.c_int64, C.c_int64,
+                                    C.POINTER(C.c_int64)]
+    lib.bkgpu_topk_kernel_ms.restype = C.c_double
+    lib.bkgpu_sync.restype = C.c_int
+    return lib
+
+
+class GpuTable:
+    def __init__(self, engine, handle, col_types, nrows):
+        self.engine = engine
+        self.handle = handle
+        self.col_types = col_types
+        self.nrows = nrows
+
+    def free(self):
+        if self.handle:
+            self.engine.lib.bkgpu_table_free(self.handle)
+            self.handle = None
+
+
+class AggResult:
+    def __init__(self, engine, handle, plan):
+        self.engine = engine
+        self.handle = handle
+        self.plan = plan
+
+    @property
+    def ngroups(self):
+        return self.engine.lib.bkgpu_agg_ngroups(self.handle)
+
+    @property
+    def rows_passed(self):
+        return self.engine.lib.bkgpu_agg_rows_passed(self.handle)
+
+    @property
+    def kernel_ms(self):
+        return self.engine.lib.bkgpu_agg_kernel_ms(self.handle)
+
+    def export_bytes(self):
+        return self.engine.lib.bkgpu_agg_export_bytes(self.handle)
+
+    def export_to(self, dev_ptr, cap):
+        rc = self.engine.lib.bkgpu_agg_export(self.handle, C.c_void_p(dev_ptr), cap)
+        self.engine._check(rc, "agg_export")
+
+    def merge_blob(self, dev_ptr, n_groups):
+        rc = self.engine.lib.bkgpu_agg_merge(self.handle, C.c_void_p(dev_ptr), n_groups)
+        self.engine._check(rc, "agg_merge")
+
+    def fetch(self, sorted=True, max_groups=None):
+        n = self.ngroups if max_groups is None else min(max_groups, self.ngroups)
+        n = max(n, 0)
+        na = self.plan and len(self.plan.aggs) or 0
+        flags = np.zeros(max(n, 1), dtype=np.uint8)
+        enc = np.zeros(max(n, 1) * BK_MAX_GROUP, dtype=np.uint64)
+        out_i = np.zeros(max(na * n, 1), dtype=np.int64)
+        out_d = np.zeros(max(na * n, 1), dtype=np.float64)
+        out_has = np.zeros(max(na * n, 1), dtype=np.uint8)
+        got = self.engine.lib.bkgpu_agg_fetch(
+            self.handle, 1 if sorted else 0, n,
+            flags.ctypes.data_as(C.POINTER(C.c_uint8)),
+            enc.ctypes.data_as(C.POINTER(C.c_uint64)),
+            out_i.ctypes.data_as(C.POINTER(C.c_int64)),
+            out_d.ctypes.data_as(C.POINTER(C.c_double)),
+            out_has.ctypes.data_as(C.POINTER(C.c_uint8)))
+        self.engine._check(int(got), "agg_fetch")
+        return {
+            "ngroups": int(got),
+            "rows_passed": self.rows_passed,
+            "flags": flags[:got],
+            "enc": enc[:got * BK_MAX_GROUP].reshape(got, BK_MAX_GROUP),
+            "agg_i": out_i[:na * got].reshape(na, got) if got else
+                     np.zeros((na, 0), np.int64),
+            "agg_d": out_d[:na * got].reshape(na, got) if got else
+                     np.zeros((na, 0), np.float64),
+            "agg_has": out_has[:na * got].reshape(na, got) if got else
+                       np.zeros((na, 0), np.uint8),
+        }
+
+    def free(self):
+        if self.handle:
+            self.engine.lib.bkgpu_agg_free(self.handle)
+            self.handle = None
+
+
+class GpuEngine:
+    def __init__(self, device=0):
+        self.lib = _load()
+        if self.lib.bkgpu_device_count() < 1:
+            raise NativeEngineMissing("no HIP device visible")
+        self._check(self.lib.bkgpu_set_device(device), "set_device")
+
+    def _check(self, rc, what):
+        if rc < 0:
+            raise RuntimeError(f"bkgpu {what}: {self.lib.bkgpu_last_error().decode()}")
+        return rc
+
+    def create_table(self, specs, nrows):
+        """specs: list of (col_type, dist, p0, p1, null_frac_x1e6)."""
+        arr = (_BkColSpec * len(specs))()
+        types = []
+        for i, s in enumerate(specs):
+            arr[i].col_type, arr[i].dist, arr[i].p0, arr[i].p1, arr[i].null_frac_x1e6 = s
+            types.append(s[0])
+        h = self.lib.bkgpu_table_create(len(specs), arr, nrows)
+        if not h:
+            raise RuntimeError(f"table_create: {self.lib.bkgpu_last_error().decode()}")
+        return GpuTable(self, h, types, nrows)
+
+    def generate(self, table, seed, row_begin=0):
+        self._check(self.lib.bkgpu_table_generate(table.handle, seed, row_begin),
+                    "table_generate")
+
+    def upload(self, table, col, data, valid=None):
+        vptr = valid.ctypes.data_as(C.c_void_p) if valid is not None else None
+        self._check(self.lib.bkgpu_table_upload(
+            table.handle, col, data.ctypes.data_as(C.c_void_p), vptr), "table_upload")
+
+    def filter_agg(self, table, plan: QueryPlan, row_begin=0, row_end=None,
+                   expected_groups=1 << 16):
+        if row_end is None:
+            row_end = table.nrows
+        q = plan.to_spec()
+        h = self.lib.bkgpu_filter_agg(table.handle, C.byref(q), row_begin, row_end,
+                                      expected_groups)
+        if not h:
+            raise RuntimeError(f"filter_agg: {self.lib.bkgpu_last_error().decode()}")
+        return AggResult(self, h, plan)
+
+    def sort_topk(self, table, order, limit, plan: QueryPlan = None,
+                  row_begin=0, row_end=None):
+        if row_end is None:
+            row_end = table.nrows
+        q = (plan or QueryPlan(table.col_types)).to_spec()
+        oarr = (BkOrderSpec * len(order))()
+        for i, (col, is_asc, null_first) in enumerate(order):
+            oarr[i].col, oarr[i].is_asc, oarr[i].is_null_first = col, is_asc, null_first
+        out = np.empty(max(limit, 1), dtype=np.int64)
+        n = self.lib.bkgpu_sort_topk(table.handle, C.byref(q), oarr, len(order),
+                                     row_begin, row_end, limit,
+                                     out.ctypes.data_as(C.POINTER(C.c_int64)))
+        self._check(int(n), "sort_topk")
+        return out[:n].copy()
+
+    def topk_kernel_ms(self):
+        return self.lib.bkgpu_topk_kernel_ms()
+
+    def sync(self):
+        self._check(self.lib.bkgpu_sync(), "sync")

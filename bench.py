@@ -1,0 +1,248 @@
+#!/usr/bin/env python3
+"""bench.py — measures BASELINE.json's metric: scanned rows/sec (whole node)
+plus achieved HBM GB/s on the 1e9-row filter+GROUP BY workload (config 3 of
+BASELINE.json, the configuration the metric is quoted on; it fits one GPU).
+
+One "step" = one pass of the fused scan+filter+hash-aggregate pipeline over
+the full synthetic table resident in HBM (data generated on device before the
+timed region; `data: synthetic`). With N>1 ranks each rank owns its region
+set (weak scaling: per-GPU rows fixed) and the step includes the RCCL
+merge-aggregate of partial group tables (the MERGE_AGG exchange the reference
+does over brpc, exchange_sender_node.h:228-235 — here all-gather over xGMI).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--config NAME]
+The driver launches N>1 via torch.distributed.run; ranks read RANK/WORLD_SIZE.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
+D_UNI, D_SKEW, D_DICT, D_SUM16 = 0, 1, 2, 3
+SEED = 20260915
+
+# ---------------- workload definitions (BASELINE.json configs) -------------
+# config3: 1e9 rows mixed INT64/DOUBLE/VARCHAR(32)-dict; 3 predicates;
+# GROUP BY (c2 skew-1e5, c7 dict-65536); aggs COUNT(*), SUM(c3:I64),
+# SUM(c4:DBL), AVG(c5:DBL).
+CONFIGS = {
+    "config2_1e8_8int64": dict(
+        nrows=100_000_000,
+        specs=[(TYPE_INT64, D_UNI, 0, 1 << 31, 0)] * 2
+            + [(TYPE_INT64, D_SKEW, 100_000, 0, 0)]
+            + [(TYPE_INT64, D_UNI, 0, 1000, 0)]
+            + [(TYPE_INT64, D_UNI, 0, 1 << 31, 0)] * 4,
+        conjuncts=[(0, "<", 1 << 30), (1, "<", int((1 << 31) * 0.9))],
+        group=[2], aggs=[("sum", 3)],
+        expected_groups=1 << 18,
+        # algorithmic bytes/row: c0,c1 always (16); group c2 + agg c3 for the
+        # 0.5*0.9 surviving fraction (16 * 0.45)
+        bytes_per_row=16 + 16 * 0.45),
+    "config3_1e9_mixed": dict(
+        nrows=1_000_000_000,
+        specs=[(TYPE_INT64, D_UNI, 0, 1 << 31, 0),      # c0 predicate
+               (TYPE_INT64, D_UNI, 0, 1 << 31, 0),      # c1 predicate
+               (TYPE_INT64, D_SKEW, 100_000, 0, 0),     # c2 group key 1
+               (TYPE_INT64, D_UNI, 0, 1000, 0),         # c3 SUM int64
+               (TYPE_DOUBLE, D_SUM16, 0, 0, 0),         # c4 SUM double
+               (TYPE_DOUBLE, D_SUM16, 0, 0, 0),         # c5 AVG double
+               (TYPE_INT64, D_UNI, 0, 1 << 31, 0),      # c6 untouched
+               (TYPE_STRING, D_DICT, 65536, 0, 0)],     # c7 group key 2 (dict)
+        conjuncts=[(0, "<", 1 << 30),                    # sel 0.5
+                   (1, "<", int((1 << 31) * 0.9)),       # sel 0.9
+                   (7, "!=", 123)],                      # sel ~1
+        group=[2, 7],
+        aggs=[("count_star", -1), ("sum", 3), ("sum", 4), ("avg", 5)],
+        expected_groups=1 << 22,
+        # c0,c1 for all rows (16 B); dict c7 for the 0.45 surviving c0&c1
+        # fraction (4 B); group c2 + aggs c3,c4,c5 for the 0.45*(65535/65536)
+        # survivors (32 B)
+        bytes_per_row=16 + 4 * 0.45 + 32 * 0.45 * (65535 / 65536)),
+}
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def cpu_baseline_leg(cfg, sample_rows):
+    """Time the oracle (the reference row engine restated; kind='port') on the
+    host cores, on a bounded sample of the same workload."""
+    from oracle import Oracle, BkColSpec
+    from oracle.bindings import make_query
+    orc = Oracle()
+    specs = (BkColSpec * len(cfg["specs"]))()
+    for i, s in enumerate(cfg["specs"]):
+        (specs[i].col_type, specs[i].dist, specs[i].p0, specs[i].p1,
+         specs[i].null_frac_x1e6) = s
+    col_types = [s[0] for s in cfg["specs"]]
+    cols, valids = orc.generate_table(list(specs), sample_rows, SEED)
+    ops = {"=": 0, "!=": 1, ">": 2, ">=": 3, "<": 4, "<=": 5}
+    aggmap = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5}
+    conj = []
+    for col, op, lit in cfg["conjuncts"]:
+        ct = TYPE_DOUBLE if (col_types[col] == TYPE_DOUBLE or isinstance(lit, float)) \
+            else TYPE_INT64
+        conj.append((col, ops[op], ct, lit))
+    q = make_query(conj, cfg["group"], [(aggmap[a], c) for a, c in cfg["aggs"]],
+                   col_types)
+    cores = os.cpu_count() or 1
+    t0 = time.perf_counter()
+    res = orc.filter_agg(cols, valids, col_types, q, nthreads=cores,
+                         dict_seed=SEED, sort_keys=False)
+    dt = time.perf_counter() - t0
+    return {"value": sample_rows / dt, "unit": "rows/s", "cores": cores,
+            "kind": "port",
+            "sample": f"{sample_rows} rows of the same workload, {dt:.1f}s"}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--config", default="config3_1e9_mixed")
+    ap.add_argument("--rows", type=int, default=0, help="override rows per GPU")
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(world, 1)
+
+    import torch
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        tdist.init_process_group(backend="nccl")
+        dist = tdist
+        torch.cuda.set_device(local_rank)
+
+    from baikaldb_amd import GpuEngine, QueryPlan
+    eng = GpuEngine(device=local_rank)
+
+    cfg = CONFIGS[args.config]
+    nrows = args.rows or cfg["nrows"]
+    # weak scaling: each rank owns its own region set of `nrows` rows
+    row_begin = rank * nrows
+
+    log(f"[bench] generating {nrows} rows on rank {rank} (device {local_rank})")
+    t = eng.create_table(cfg["specs"], nrows)
+    eng.generate(t, SEED, row_begin=row_begin)
+    eng.sync()
+
+    plan = QueryPlan(t.col_types, conjuncts=cfg["conjuncts"], group=cfg["group"],
+                     aggs=cfg["aggs"])
+
+    def one_step(timed_kernels):
+        res = eng.filter_agg(t, plan, expected_groups=cfg["expected_groups"])
+        if timed_kernels is not None:
+            timed_kernels.append(res.kernel_ms)
+        if dist is not None:
+            eng.sync()
+            ng = torch.tensor([res.ngroups], dtype=torch.int64, device="cuda")
+            sizes = [torch.zeros(1, dtype=torch.int64, device="cuda")
+                     for _ in range(world)]
+            dist.all_gather(sizes, ng)
+            sizes = [int(s.item()) for s in sizes]
+            nbytes = res.export_bytes()
+            per_group = nbytes // max(res.ngroups, 1)
+            maxg = max(sizes)
+            buf = torch.zeros(maxg * per_group, dtype=torch.uint8, device="cuda")
+            if res.ngroups > 0:
+                res.export_to(buf.data_ptr(), maxg * per_group)
+            gath = [torch.zeros_like(buf) for _ in range(world)]
+            dist.all_gather(gath, buf)
+            torch.cuda.synchronize()
+            if rank == 0:
+                for peer in range(1, world):
+                    if sizes[peer] > 0:
+                        # repack: peer blob was laid out for sizes[peer] groups
+                        res.merge_blob(gath[peer].data_ptr(), sizes[peer])
+                eng.sync()
+        rp = res.rows_passed
+        ng = res.ngroups
+        res.free()
+        return rp, ng
+
+    log(f"[bench] warmup x{args.warmup}")
+    for _ in range(args.warmup):
+        one_step(None)
+    eng.sync()
+    if dist is not None:
+        dist.barrier()
+    torch.cuda.synchronize() if torch.cuda.is_available() else None
+
+    kms = []
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        rp, ng = one_step(kms)
+    eng.sync()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    if dist is not None:
+        dist.barrier()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+    if dist is not None:  # max over ranks
+        e = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    total_rows = nrows * n_gpus  # whole-job rows scanned per step
+    rows_per_sec = total_rows / (elapsed / args.steps)
+
+    if rank != 0:
+        return
+
+    # roofline of the dominant kernel (fused filter+agg), HIP-event timed on
+    # its launch stream inside libbkgpu (bkgpu_agg_kernel_ms)
+    avg_kernel_ms = sum(kms) / len(kms) if kms else 0.0
+    algo_bytes = cfg["bytes_per_row"] * nrows  # per launch (this rank)
+    achieved_gbs = (algo_bytes / (avg_kernel_ms / 1000.0)) / 1e9 if avg_kernel_ms else 0.0
+    peak_gbs = 8000.0  # HBM3E spec peak (MI355X_MICROARCH.md)
+    traffic = os.environ.get("BK_TRAFFIC_BYTES")  # from separate rocprof --pmc run
+    roofline = {"bound": "hbm", "achieved": round(achieved_gbs, 1),
+                "peak": peak_gbs, "unit": "GB/s",
+                "frac": round(achieved_gbs / peak_gbs, 4),
+                "traffic": float(traffic) if traffic else None}
+
+    cpu_baseline = None
+    if n_gpus == 1 and not args.no_cpu_baseline:
+        sample = min(nrows, 20_000_000)
+        log(f"[bench] cpu baseline leg on {sample} rows")
+        cpu_baseline = cpu_baseline_leg(cfg, sample)
+
+    line = {
+        "metric": "scanned rows/sec (whole node), 1e9-row filter+GROUP BY",
+        "value": round(rows_per_sec, 1),
+        "unit": "rows/s",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,  # reference publishes no number
+        "dtype": "int64/f64",
+        "data": "synthetic",
+        "config": {"workload": args.config, "rows_per_gpu": nrows,
+                   "groups": ng, "rows_passed_rank0": rp,
+                   "predicates": 3 if args.config.startswith("config3") else 2,
+                   "group_keys": len(cfg["group"]), "aggs": len(cfg["aggs"])},
+        "roofline": roofline,
+        "cpu_baseline": cpu_baseline,
+    }
+    print(json.dumps(line), flush=True)
+
+
+if __name__ == "__main__":
+    main()

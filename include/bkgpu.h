@@ -1,0 +1,112 @@
+/* bkgpu.h — C-ABI of the MI355X-native OLAP execution engine ("bkgpu").
+ *
+ * This is the drop-in boundary for baikalStore: the host-side
+ * exec nodes (C++ mirror of ExecNode, see include/bk_exec.h) call ONLY this
+ * flat C API, exactly as the reference's row nodes hand off to Arrow Acero
+ * today (src/runtime/arrow_io_excutor.cpp:265, src/store/region.cpp:2793-2923).
+ * No C++ or torch types cross this boundary: plain pointers and sizes.
+ *
+ * Entry points and the reference interface each replaces:
+ *  - bkgpu_table_*        : the columnar batch source standing in for
+ *                           RocksdbScanNode's decoded output
+ *                           (src/exec/rocksdb_scan_node.cpp:748-800; the
+ *                           KV-decode itself is below the boundary)
+ *  - bkgpu_filter_agg     : FilterNode::get_next + AggNode::open/process
+ *                           (src/exec/filter_node.cpp:736-795,
+ *                            src/exec/agg_node.cpp:405-545) fused into one
+ *                           GPU pipeline pass
+ *  - bkgpu_agg_merge      : the db-side MERGE_AGG combine
+ *                           (src/exec/agg_node.cpp:29,539-543) — used by the
+ *                           multi-GPU region-set merge (RCCL over xGMI)
+ *  - bkgpu_agg_fetch      : AggNode::get_next finalize+emit
+ *                           (src/exec/agg_node.cpp:548-573,
+ *                            src/expr/agg_fn_call.cpp:927-975)
+ *  - bkgpu_sort_topk      : SortNode + TopNSorter (src/exec/sort_node.cpp:278-440,
+ *                           include/runtime/topn_sorter.h:32-63)
+ *
+ * All functions return 0 on success, negative on error (mirrors ExecNode's
+ * int-return error convention, include/exec/exec_node.h:140-153);
+ * bkgpu_last_error() returns a description.
+ */
+#ifndef BKGPU_H
+#define BKGPU_H
+
+#include <stdint.h>
+#include "bk_common.h"
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct BkgTable BkgTable;     /* columnar table resident in HBM */
+typedef struct BkgAggOut BkgAggOut;   /* aggregation result resident in HBM */
+
+/* ---- device / error ---- */
+int         bkgpu_device_count(void);
+int         bkgpu_set_device(int dev);
+const char* bkgpu_last_error(void);
+int         bkgpu_sync(void);
+
+/* ---- table lifecycle ---- */
+/* Create a table of nrows rows whose columns follow specs[0..ncols).
+ * Storage: BK_INT64 -> int64, BK_DOUBLE -> double, BK_STRING -> int32 dict
+ * codes. Columns with null_frac_x1e6 > 0 also get a validity byte array. */
+BkgTable* bkgpu_table_create(int ncols, const BkColSpec* specs, int64_t nrows);
+/* Fill all columns on device with the deterministic shared generator
+ * (bk_datagen.h) for global row ids [row_begin, row_begin+nrows). */
+int  bkgpu_table_generate(BkgTable* t, uint64_t seed, int64_t row_begin);
+/* Upload host column data (and optional validity bytes) instead. */
+int  bkgpu_table_upload(BkgTable* t, int col, const void* data, const uint8_t* valid);
+int64_t bkgpu_table_nrows(const BkgTable* t);
+void bkgpu_table_free(BkgTable* t);
+
+/* ---- fused scan+filter+aggregate ---- */
+/* Run the SELECT pipeline `WHERE conjuncts GROUP BY group aggs` over rows
+ * [row_begin, row_end) of t. expected_groups sizes the hash table (engine
+ * retries with a larger table on overflow). Returns NULL on error. */
+BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
+                            int64_t row_begin, int64_t row_end,
+                            int64_t expected_groups);
+
+int64_t bkgpu_agg_ngroups(const BkgAggOut* o);
+int64_t bkgpu_agg_rows_passed(const BkgAggOut* o);
+/* wall time of the fused kernel (HIP events), for bench roofline only */
+double  bkgpu_agg_kernel_ms(const BkgAggOut* o);
+
+/* ---- partial-aggregate exchange (multi-GPU merge over RCCL) ----
+ * Compact wire format of one partial result, an SoA byte blob:
+ *   [ flags: u32 * n ][ k0: u64 * n ][ k1: u64 * n ][ states: u64 * n * 2*naggs ]
+ * (AVG state = {sum double bits, count}; SUM = {val bits, nonnull count};
+ *  MIN/MAX = {order-encoded u64, nonnull count}; COUNT = {count, _}.) */
+int64_t bkgpu_agg_export_bytes(const BkgAggOut* o);
+/* Write the blob to dst (DEVICE pointer, capacity cap bytes). */
+int  bkgpu_agg_export(const BkgAggOut* o, void* dst, int64_t cap);
+/* Merge a peer blob (DEVICE pointer, n groups) into o with the reference's
+ * AggFnCall::merge semantics (src/expr/agg_fn_call.cpp:781-830). */
+int  bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups);
+
+/* ---- result fetch (finalize + emit) ----
+ * Downloads up to max_groups finalized groups to host arrays (each sized by
+ * caller: flags[n], enc[n*BK_MAX_GROUP], out_i/out_d/out_has[naggs*n]).
+ * Groups arrive in canonical order (sorted by the reference MutTableKey byte
+ * order) iff sorted != 0. Finalization follows agg_fn_call.cpp:927-975. */
+int64_t bkgpu_agg_fetch(BkgAggOut* o, int sorted, int64_t max_groups,
+                        uint8_t* flags, uint64_t* enc,
+                        int64_t* out_i, double* out_d, uint8_t* out_has);
+void bkgpu_agg_free(BkgAggOut* o);
+
+/* ---- ORDER BY ... LIMIT top-N ----
+ * Select the `limit` smallest rows of [row_begin,row_end) passing q's filter
+ * under `order` (non-null INT64/DOUBLE keys; ties broken by arrival index,
+ * topn_sorter.h:46-54), write their global row ids in final order to
+ * out_rows (host array). Returns count written, negative on error. */
+int64_t bkgpu_sort_topk(BkgTable* t, const BkQuerySpec* q,
+                        const BkOrderSpec* order, int norder,
+                        int64_t row_begin, int64_t row_end,
+                        int64_t limit, int64_t* out_rows);
+double bkgpu_topk_kernel_ms(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* BKGPU_H */

@@ -1,0 +1,288 @@
+# GPU parity tests: the HIP fused filter+aggregate vs the CPU oracle on
+# identical seeded inputs (bk_datagen.h generates bit-identical data on both
+# sides). Bit-exact for COUNT / integer SUM / integer keys / MIN/MAX;
+# stated tolerance for SUM/AVG DOUBLE (GPU reduction order differs from the
+# reference's sequential ExprValue::add — north_star fp tolerance clause).
+import ctypes as C
+
+import numpy as np
+import pytest
+
+from oracle import BkColSpec
+from oracle.bindings import make_query
+
+pytestmark = pytest.mark.gpu
+
+SEED = 0xBADC0DE
+TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
+D_UNI, D_SKEW, D_DICT, D_SUM16 = 0, 1, 2, 3
+
+# tolerance for double sums: |err| <= DTOL_REL * (|sum| + n_rows * 1.0) — the
+# inputs are ~N(0,1) so per-element |x| ~ 1; fp error of a pairwise-vs-
+# sequential reorder is ~1e-16 per element accumulated.
+DTOL_REL = 1e-10
+
+
+@pytest.fixture(scope="module")
+def eng():
+    from baikaldb_amd import GpuEngine
+    return GpuEngine()
+
+
+@pytest.fixture(scope="module")
+def orc():
+    from oracle import Oracle
+    return Oracle()
+
+
+def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
+             expected_groups=1 << 14, seed=SEED):
+    """conjuncts: (col, op_sym, lit); aggs: (name, col)."""
+    t = eng.create_table(spec_rows, n)
+    try:
+        eng.generate(t, seed)
+        from baikaldb_amd import QueryPlan
+        plan = QueryPlan(t.col_types, conjuncts=conjuncts, group=group, aggs=aggs)
+        res = eng.filter_agg(t, plan, expected_groups=expected_groups)
+        try:
+            got = res.fetch(sorted=True)
+        finally:
+            res.free()
+    finally:
+        t.free()
+
+    specs = (BkColSpec * len(spec_rows))()
+    for i, s in enumerate(spec_rows):
+        (specs[i].col_type, specs[i].dist, specs[i].p0, specs[i].p1,
+         specs[i].null_frac_x1e6) = s
+    cols, valids = orc.generate_table(list(specs), n, seed)
+    col_types = [s[0] for s in spec_rows]
+    ops = {"=": 0, "!=": 1, ">": 2, ">=": 3, "<": 4, "<=": 5}
+    aggmap = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5}
+    oconj = []
+    for col, op, lit in conjuncts:
+        ct = TYPE_DOUBLE if (col_types[col] == TYPE_DOUBLE or
+                             isinstance(lit, float)) else TYPE_INT64
+        oconj.append((col, ops[op], ct, lit))
+    q = make_query(oconj, group, [(aggmap[a], c) for a, c in aggs], col_types)
+    exp = orc.filter_agg(cols, valids, col_types, q, nthreads=nthreads,
+                         dict_seed=seed)
+    return got, exp
+
+
+def assert_parity(got, exp, aggs, col_types):
+    assert got["rows_passed"] == exp["rows_passed"]
+    assert got["ngroups"] == exp["ngroups"]
+    assert np.array_equal(got["flags"], exp["flags"])
+    assert np.array_equal(got["enc"], exp["enc"])
+    assert np.array_equal(got["agg_has"], exp["agg_has"])
+    for a, (name, col) in enumerate(aggs):
+        is_double = col >= 0 and col_types[col] == TYPE_DOUBLE
+        if name in ("count_star", "count") or not is_double:
+            assert np.array_equal(got["agg_i"][a], exp["agg_i"][a]), f"agg {a} {name}"
+        elif name in ("min", "max"):
+            # min/max double: same element selected => bit-exact
+            assert np.array_equal(got["agg_d"][a], exp["agg_d"][a]), f"agg {a} {name}"
+        else:  # sum/avg double: reduction-order tolerance
+            denom = np.abs(exp["agg_d"][a]) + np.maximum(exp["agg_i"][0], 1)
+            err = np.abs(got["agg_d"][a] - exp["agg_d"][a])
+            assert np.all(err <= DTOL_REL * denom), \
+                f"agg {a} {name} max err {err.max()}"
+
+
+BASE5 = [(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+         (TYPE_INT64, D_SKEW, 100000, 0, 0),
+         (TYPE_INT64, D_UNI, 0, 1000, 0),
+         (TYPE_DOUBLE, D_SUM16, 0, 0, 0),
+         (TYPE_STRING, D_DICT, 4096, 0, 0)]
+
+
+def test_count_star_selectivities(eng, orc):
+    for frac in (0.0, 0.1, 0.5, 0.9, 1.0):
+        k = int((1 << 31) * frac)
+        got, exp = run_both(eng, orc, BASE5, 300_000,
+                            [(0, "<", k)], [], [("count_star", -1)])
+        assert_parity(got, exp, [("count_star", -1)], [s[0] for s in BASE5])
+
+
+def test_group_by_single_int_key(eng, orc):
+    aggs = [("count_star", -1), ("sum", 2), ("sum", 3), ("avg", 3)]
+    got, exp = run_both(eng, orc, BASE5, 500_000,
+                        [(0, "<", int((1 << 31) * 0.7))], [1], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in BASE5])
+
+
+def test_group_by_two_keys_with_dict(eng, orc):
+    aggs = [("count_star", -1), ("sum", 2), ("avg", 3), ("min", 0), ("max", 3)]
+    got, exp = run_both(eng, orc, BASE5, 400_000,
+                        [(0, "<", int((1 << 31) * 0.8)), (2, "!=", 17)],
+                        [1, 4], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in BASE5])
+
+
+def test_double_predicate(eng, orc):
+    aggs = [("count_star", -1), ("sum", 3)]
+    got, exp = run_both(eng, orc, BASE5, 200_000,
+                        [(3, ">", 0.5)], [2], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in BASE5])
+
+
+def test_nulls(eng, orc):
+    specs = [(TYPE_INT64, D_UNI, 0, 1 << 31, 150_000),
+             (TYPE_INT64, D_SKEW, 500, 0, 300_000),
+             (TYPE_INT64, D_UNI, 0, 100, 100_000),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 200_000)]
+    aggs = [("count_star", -1), ("count", 2), ("sum", 2), ("avg", 3),
+            ("min", 3), ("max", 0)]
+    got, exp = run_both(eng, orc, specs, 250_000,
+                        [(0, "<", int((1 << 31) * 0.9))], [1], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
+
+
+def test_empty_selection_no_group(eng, orc):
+    got, exp = run_both(eng, orc, BASE5, 100_000,
+                        [(0, "<", -1)], [], [("count_star", -1), ("sum", 2)])
+    assert got["ngroups"] == 1 == exp["ngroups"]
+    assert got["agg_i"][0][0] == 0
+    assert got["agg_has"][1][0] == 0 == exp["agg_has"][1][0]
+
+
+def test_empty_selection_with_group(eng, orc):
+    got, exp = run_both(eng, orc, BASE5, 100_000,
+                        [(0, "<", -1)], [1], [("count_star", -1)])
+    assert got["ngroups"] == 0 == exp["ngroups"]
+
+
+def test_empty_table(eng, orc):
+    got, exp = run_both(eng, orc, BASE5, 0, [], [1], [("count_star", -1)])
+    assert got["ngroups"] == 0 == exp["ngroups"]
+    got, exp = run_both(eng, orc, BASE5, 0, [], [], [("count_star", -1)])
+    assert got["ngroups"] == 1 == exp["ngroups"]
+    assert got["agg_i"][0][0] == 0
+
+
+def test_many_groups_overflow_regrow(eng, orc):
+    """More groups than the initial table sizing => engine regrows and reruns."""
+    specs = [(TYPE_INT64, D_UNI, 0, 200_000, 0),
+             (TYPE_INT64, D_UNI, 0, 10, 0)]
+    aggs = [("count_star", -1), ("sum", 1)]
+    got, exp = run_both(eng, orc, specs, 400_000, [], [0], aggs,
+                        expected_groups=64)  # deliberately undersized
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
+    assert got["ngroups"] > 100_000
+
+
+def test_group_key_edge_values(eng, orc):
+    """Keys spanning the full int64 range incl. INT64_MIN (encodes to 0) and
+    INT64_MAX (encodes to ~0) — exercises sentinel-free slot claims."""
+    specs = [(TYPE_INT64, D_UNI, -(1 << 62), (1 << 62), 0),
+             (TYPE_INT64, D_UNI, 0, 5, 0)]
+    aggs = [("count_star", -1), ("min", 0), ("max", 0)]
+    got, exp = run_both(eng, orc, specs, 100_000, [], [1], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
+
+
+def test_merge_partials_equals_whole(eng, orc):
+    """Region-sharded execution + MERGE_AGG combine == single-pass execution
+    (the multi-GPU merge path, agg_node.cpp:539-543), exercised on one GPU by
+    splitting rows into 4 'region sets' and merging their partials."""
+    import torch
+    from baikaldb_amd import QueryPlan
+    n = 400_000
+    spec_rows = BASE5
+    aggs = [("count_star", -1), ("sum", 2), ("sum", 3), ("avg", 3),
+            ("min", 0), ("max", 3)]
+    conj = [(0, "<", int((1 << 31) * 0.75))]
+    t = eng.create_table(spec_rows, n)
+    try:
+        eng.generate(t, SEED)
+        plan = QueryPlan(t.col_types, conjuncts=conj, group=[1, 4], aggs=aggs)
+        # whole-range run
+        whole = eng.filter_agg(t, plan, expected_groups=1 << 14)
+        expect = whole.fetch(sorted=True)
+        whole.free()
+        # 4 shard runs + merge into the first
+        shards = []
+        bounds = [0, n // 4, n // 2, 3 * n // 4, n]
+        for i in range(4):
+            shards.append(eng.filter_agg(t, plan, row_begin=bounds[i],
+                                         row_end=bounds[i + 1],
+                                         expected_groups=1 << 14))
+        dst = shards[0]
+        for s in shards[1:]:
+            nbytes = s.export_bytes()
+            buf = torch.empty(nbytes, dtype=torch.uint8, device="cuda")
+            s.export_to(buf.data_ptr(), nbytes)
+            dst.merge_blob(buf.data_ptr(), s.ngroups)
+            s.free()
+        merged = dst.fetch(sorted=True)
+        dst.free()
+    finally:
+        t.free()
+    assert merged["ngroups"] == expect["ngroups"]
+    assert np.array_equal(merged["flags"], expect["flags"])
+    assert np.array_equal(merged["enc"], expect["enc"])
+    assert np.array_equal(merged["agg_i"], expect["agg_i"])
+    assert np.array_equal(merged["agg_has"], expect["agg_has"])
+    np.testing.assert_allclose(merged["agg_d"], expect["agg_d"],
+                               rtol=1e-9, atol=1e-9)
+    # COUNT(*) sums to the whole-range rows_passed (rows_passed itself is a
+    # per-partial counter; the multi-GPU bench sums it across ranks)
+    assert merged["agg_i"][0].sum() == expect["rows_passed"]
+
+
+def test_full_size_properties(eng, orc):
+    """Property checks at a size the oracle cannot cover row-by-row quickly:
+    1e8 rows, config-2 shape (8 x INT64, WHERE c1<K AND c2=K2 GROUP BY c3
+    SUM(c4)). Properties: sum of per-group COUNTs == rows_passed; COUNT over
+    a partition of the group domain is conserved; result is identical across
+    two runs (determinism of everything except double sums)."""
+    from baikaldb_amd import QueryPlan
+    n = 100_000_000
+    specs = [(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+             (TYPE_INT64, D_UNI, 0, 100, 0),
+             (TYPE_INT64, D_SKEW, 100_000, 0, 0),
+             (TYPE_INT64, D_UNI, 0, 1000, 0)]
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED)
+        plan = QueryPlan(t.col_types,
+                         conjuncts=[(0, "<", int((1 << 31) * 0.5)), (1, "=", 42)],
+                         group=[2], aggs=[("count_star", -1), ("sum", 3)])
+        r1 = eng.filter_agg(t, plan, expected_groups=1 << 18)
+        g1 = r1.fetch(sorted=True)
+        r1.free()
+        r2 = eng.filter_agg(t, plan, expected_groups=1 << 18)
+        g2 = r2.fetch(sorted=True)
+        r2.free()
+    finally:
+        t.free()
+    assert g1["agg_i"][0].sum() == g1["rows_passed"]
+    assert g1["ngroups"] == g2["ngroups"]
+    assert np.array_equal(g1["enc"], g2["enc"])
+    assert np.array_equal(g1["agg_i"], g2["agg_i"])  # int sums deterministic
+    # oracle checks a 2M-row sample of the same table slice bit-exactly
+    sspecs = (BkColSpec * len(specs))()
+    for i, s in enumerate(specs):
+        (sspecs[i].col_type, sspecs[i].dist, sspecs[i].p0, sspecs[i].p1,
+         sspecs[i].null_frac_x1e6) = s
+    m = 2_000_000
+    cols, valids = orc.generate_table(list(sspecs), m, SEED)
+    col_types = [s[0] for s in specs]
+    q = make_query([(0, 4, TYPE_INT64, int((1 << 31) * 0.5)), (1, 0, TYPE_INT64, 42)],
+                   [2], [(0, -1), (2, 3)], col_types)
+    exp = orc.filter_agg(cols, valids, col_types, q, nthreads=8, dict_seed=SEED)
+    from baikaldb_amd import QueryPlan as QP
+    t2 = eng.create_table(specs, m)
+    try:
+        eng.generate(t2, SEED)
+        plan2 = QP(col_types, conjuncts=[(0, "<", int((1 << 31) * 0.5)), (1, "=", 42)],
+                   group=[2], aggs=[("count_star", -1), ("sum", 3)])
+        rs = eng.filter_agg(t2, plan2, expected_groups=1 << 18)
+        gs = rs.fetch(sorted=True)
+        rs.free()
+    finally:
+        t2.free()
+    assert gs["rows_passed"] == exp["rows_passed"]
+    assert np.array_equal(gs["enc"], exp["enc"])
+    assert np.array_equal(gs["agg_i"], exp["agg_i"])
