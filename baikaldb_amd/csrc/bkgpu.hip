@@ -782,7 +782,7 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
     int64_t nslots = next_pow2(std::max<int64_t>(1024, expected_groups * 2));
     DevCols dc = table_cols(t);
 
-    for (int attempt = 0; attempt < 4; attempt++) {
+    for (int attempt = 0; attempt < 8; attempt++) {
         if (agg_alloc(o, nslots) != 0) { bkgpu_agg_free(o); return nullptr; }
         hipEvent_t ev0, ev1;
         HIP_CHECK_NULL(hipEventCreate(&ev0));
@@ -826,7 +826,7 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
         /* overflow / probe livelock: grow 4x and rerun */
         agg_release_table(o);
         nslots *= 4;
-        if (attempt == 3) {
+        if (attempt == 7) {
             set_err("hash table overflow after retries");
             bkgpu_agg_free(o);
             return nullptr;

@@ -92,7 +92,7 @@ def cpu_baseline_leg(cfg, sample_rows):
         conj.append((col, ops[op], ct, lit))
     q = make_query(conj, cfg["group"], [(aggmap[a], c) for a, c in cfg["aggs"]],
                    col_types)
-    cores = os.cpu_count() or 1
+    cores = min(os.cpu_count() or 1, 64)  # oracle caps at 64 worker threads
     t0 = time.perf_counter()
     res = orc.filter_agg(cols, valids, col_types, q, nthreads=cores,
                          dict_seed=SEED, sort_keys=False)
