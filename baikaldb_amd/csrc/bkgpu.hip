@@ -533,6 +533,303 @@ k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_
     }
 }
 
+/* ------------------------------------------------------------------ */
+/* hash-partitioned aggregate (the high-cardinality GROUP BY path)     */
+/*                                                                     */
+/* Probing a shared global hash table per row is atomic-throughput-    */
+/* bound on this chip (~20 G memory-side ops/s measured — see          */
+/* profiles/). For group counts beyond what the per-WG LDS table       */
+/* absorbs, we instead hash-PARTITION the passing rows into P buckets  */
+/* (bucket = high bits of the key hash), materialize compact records   */
+/* (key + agg inputs, SoA), and aggregate each bucket in a single      */
+/* workgroup's LDS table — turning random atomics into streaming HBM   */
+/* traffic. This mirrors the reference's MPP hash-repartition exchange */
+/* (exchange_sender_node.h:228-235) applied intra-GPU.                 */
+/* ------------------------------------------------------------------ */
+
+struct RecLayout {
+    int32_t nwords;
+    int32_t k1_word;               /* -1 if n_group < 2 */
+    int32_t meta_word;             /* -1 if nothing nullable; else
+                                      u64 word: bits0-7 = null flag,
+                                      bits 8+a = agg-input a valid */
+    int32_t val_word[BK_MAX_AGGS]; /* -1 for COUNT(*) / COUNT */
+};
+
+#define PART_BUCKET(h, P) ((uint32_t)((h) >> 44) & ((P) - 1u))
+#define BK_SKIP_BUCKET 0xFFFFu
+
+/* pass 1: predicate + per-block bucket histogram; bucketid[] remembers the
+ * verdict so later passes re-evaluate nothing. */
+__global__ void __launch_bounds__(256)
+k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
+             uint32_t P, uint16_t* bucketid, uint32_t* H) {
+    extern __shared__ __attribute__((aligned(16))) uint32_t lhist[];
+    for (uint32_t b = threadIdx.x; b < P; b += blockDim.x) lhist[b] = 0;
+    __syncthreads();
+    int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < row_end; r += gstride) {
+        int64_t i = r - row_begin;
+        if (!row_passes(cols, q, r)) { bucketid[i] = (uint16_t)BK_SKIP_BUCKET; continue; }
+        uint32_t flag = 0;
+        uint64_t k0 = 0, k1 = 0;
+        if (q.n_group >= 1) {
+            const DevCol& c = cols.c[q.group_cols[0]];
+            if (!cell_valid(c, r)) flag |= 0x80u; else k0 = enc_value(c, r);
+        }
+        if (q.n_group >= 2) {
+            const DevCol& c = cols.c[q.group_cols[1]];
+            if (!cell_valid(c, r)) flag |= 0x40u; else k1 = enc_value(c, r);
+        }
+        uint32_t b = PART_BUCKET(key_hash(flag, k0, k1), P);
+        bucketid[i] = (uint16_t)b;
+        atomicAdd(&lhist[b], 1u);
+    }
+    __syncthreads();
+    for (uint32_t b = threadIdx.x; b < P; b += blockDim.x)
+        H[(size_t)blockIdx.x * P + b] = lhist[b];
+}
+
+/* bucket totals over all blocks */
+__global__ void k_part_totals(const uint32_t* H, uint32_t nblocks, uint32_t P,
+                              uint32_t* totals) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= P) return;
+    uint32_t s = 0;
+    for (uint32_t blk = 0; blk < nblocks; blk++) s += H[(size_t)blk * P + b];
+    totals[b] = s;
+}
+
+/* exclusive scan of totals -> base; grand total -> total_out. single block. */
+__global__ void k_part_scan(const uint32_t* totals, uint32_t P, uint32_t* base,
+                            uint64_t* total_out) {
+    if (threadIdx.x == 0) {
+        uint64_t run = 0;
+        for (uint32_t b = 0; b < P; b++) { base[b] = (uint32_t)run; run += totals[b]; }
+        *total_out = run;
+    }
+}
+
+/* turn H[blk][b] counts into absolute start offsets */
+__global__ void k_part_offsets(uint32_t* H, uint32_t nblocks, uint32_t P,
+                               const uint32_t* base) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= P) return;
+    uint32_t run = base[b];
+    for (uint32_t blk = 0; blk < nblocks; blk++) {
+        uint32_t t = H[(size_t)blk * P + b];
+        H[(size_t)blk * P + b] = run;
+        run += t;
+    }
+}
+
+/* pass 2: scatter records (SoA: word w of row i at rec[w*total + i]).
+ * MUST run with the same grid/block shape as k_part_histo so each block
+ * sees the same rows its H row was computed from. */
+__global__ void __launch_bounds__(256)
+k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
+               int64_t row_end, uint32_t P, const uint16_t* bucketid,
+               const uint32_t* H, uint64_t* rec, uint64_t total) {
+    extern __shared__ __attribute__((aligned(16))) uint32_t lcur[];
+    for (uint32_t b = threadIdx.x; b < P; b += blockDim.x)
+        lcur[b] = H[(size_t)blockIdx.x * P + b];
+    __syncthreads();
+    int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < row_end; r += gstride) {
+        int64_t i = r - row_begin;
+        uint32_t b = bucketid[i];
+        if (b == BK_SKIP_BUCKET) continue;
+        uint32_t pos = atomicAdd(&lcur[b], 1u);
+        /* key words */
+        uint64_t meta = 0;
+        uint64_t k0 = 0, k1 = 0;
+        if (q.n_group >= 1) {
+            const DevCol& c = cols.c[q.group_cols[0]];
+            if (!cell_valid(c, r)) meta |= 0x80u; else k0 = enc_value(c, r);
+        }
+        if (q.n_group >= 2) {
+            const DevCol& c = cols.c[q.group_cols[1]];
+            if (!cell_valid(c, r)) meta |= 0x40u; else k1 = enc_value(c, r);
+        }
+        rec[pos] = k0;
+        if (lay.k1_word >= 0) rec[(size_t)lay.k1_word * total + pos] = k1;
+        for (int32_t a = 0; a < q.n_aggs; a++) {
+            if (lay.val_word[a] < 0) continue;
+            const BkAggSpec& as = q.aggs[a];
+            const DevCol& c = cols.c[as.col];
+            int valid = cell_valid(c, r);
+            if (valid) meta |= (uint64_t)1 << (8 + a);
+            uint64_t w = 0;
+            if (valid) {
+                switch (as.agg_type) {
+                    case BK_AGG_SUM:
+                        if (q.agg_in_types[a] == BK_DOUBLE) {
+                            double d = ((const double*)c.data)[r];
+                            memcpy(&w, &d, 8);
+                        } else {
+                            w = (uint64_t)cell_i64(c, r);
+                        }
+                        break;
+                    case BK_AGG_AVG: {
+                        double d = cell_f64(c, r);
+                        memcpy(&w, &d, 8);
+                        break;
+                    }
+                    case BK_AGG_MIN:
+                    case BK_AGG_MAX:
+                        w = enc_value(c, r);
+                        break;
+                    default: break;
+                }
+            }
+            rec[(size_t)lay.val_word[a] * total + pos] = w;
+        }
+        if (lay.meta_word >= 0) {
+            /* COUNT(col) validity for aggs without a val word */
+            for (int32_t a = 0; a < q.n_aggs; a++) {
+                if (lay.val_word[a] >= 0 || q.aggs[a].col < 0) continue;
+                if (cell_valid(cols.c[q.aggs[a].col], r))
+                    meta |= (uint64_t)1 << (8 + a);
+            }
+            rec[(size_t)lay.meta_word * total + pos] = meta;
+        }
+    }
+}
+
+/* update an (LDS or global) slot from one record */
+template <bool LDS>
+__device__ __forceinline__ void agg_update_slot_rec(uint64_t* st, const BkQuerySpec& q,
+                                                    const RecLayout& lay,
+                                                    const uint64_t* rec, uint64_t total,
+                                                    uint64_t pos, uint64_t meta) {
+    #pragma unroll 4
+    for (int32_t a = 0; a < q.n_aggs; a++) {
+        uint64_t* val = st + SLOT_HDR + 2 * a;
+        uint64_t* cnt = val + 1;
+        int at = q.aggs[a].agg_type;
+        int has_meta = lay.meta_word >= 0;
+        int valid = !has_meta || at == BK_AGG_COUNT_STAR ||
+                    ((meta >> (8 + a)) & 1);
+        if (at == BK_AGG_COUNT_STAR) {
+            atomicAdd((unsigned long long*)val, 1ull);
+            continue;
+        }
+        if (!valid) continue;
+        if (at == BK_AGG_COUNT) {
+            atomicAdd((unsigned long long*)val, 1ull);
+            continue;
+        }
+        uint64_t w = rec[(size_t)lay.val_word[a] * total + pos];
+        switch (at) {
+            case BK_AGG_SUM:
+                if (q.agg_in_types[a] == BK_DOUBLE) {
+                    double d; memcpy(&d, &w, 8);
+                    if (LDS) atomic_add_f64_lds(val, d);
+                    else     atomic_add_f64_global(val, d);
+                } else {
+                    atomicAdd((unsigned long long*)val, (unsigned long long)w);
+                }
+                atomicAdd((unsigned long long*)cnt, 1ull);
+                break;
+            case BK_AGG_AVG: {
+                double d; memcpy(&d, &w, 8);
+                if (LDS) atomic_add_f64_lds(val, d);
+                else     atomic_add_f64_global(val, d);
+                atomicAdd((unsigned long long*)cnt, 1ull);
+                break;
+            }
+            case BK_AGG_MIN:
+                atomicMax((unsigned long long*)val, (unsigned long long)~w);
+                atomicAdd((unsigned long long*)cnt, 1ull);
+                break;
+            case BK_AGG_MAX:
+                atomicMax((unsigned long long*)val, (unsigned long long)w);
+                atomicAdd((unsigned long long*)cnt, 1ull);
+                break;
+            default: break;
+        }
+    }
+}
+
+/* pass 3: one workgroup per bucket — aggregate its records in a per-WG LDS
+ * table; when the LDS table fills (more distinct groups in the bucket than
+ * slots), the whole block flushes it into the global table (additive merge),
+ * resets, and continues — so ANY group cardinality is handled with global
+ * traffic proportional to #groups x generations, never to #rows. */
+__global__ void __launch_bounds__(256)
+k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
+           const uint32_t* base, const uint32_t* totals, uint32_t P,
+           uint64_t* gtable, uint64_t gmask, uint64_t fill_cap, uint64_t* fill,
+           uint32_t* err, uint32_t lds_slots) {
+    extern __shared__ __attribute__((aligned(16))) uint64_t ltab[];
+    const int stride = SLOT_HDR + 2 * q.n_aggs;
+    uint64_t* laux = ltab + (size_t)lds_slots * stride;
+    uint32_t* lfill = (uint32_t*)&laux[0];
+    const uint32_t lmask = lds_slots - 1;
+    const uint32_t lcap = lds_slots - 256;  /* room for one tile of claims */
+    for (uint32_t b = blockIdx.x; b < P; b += gridDim.x) {
+        uint32_t n = totals[b];
+        if (n == 0) continue;
+        for (uint32_t w = threadIdx.x; w < lds_slots * (uint32_t)stride + 1;
+             w += blockDim.x)
+            ltab[w] = 0;
+        __syncthreads();
+        uint32_t b0 = base[b];
+        /* uniform tile loop so generation flushes can barrier */
+        for (uint32_t t0 = 0; t0 < n; t0 += blockDim.x) {
+            uint32_t i = t0 + threadIdx.x;
+            bool pending = i < n;
+            uint64_t pos = (uint64_t)b0 + i;
+            uint64_t k0 = 0, k1 = 0, meta = 0;
+            uint32_t flag = 0;
+            if (pending) {
+                k0 = rec[pos];
+                k1 = lay.k1_word >= 0 ? rec[(size_t)lay.k1_word * total + pos] : 0;
+                meta = lay.meta_word >= 0 ? rec[(size_t)lay.meta_word * total + pos] : 0;
+                flag = (uint32_t)(meta & 0xFF);
+            }
+            for (;;) {
+                if (pending) {
+                    uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
+                                                  lfill, lcap);
+                    if (slot) {
+                        agg_update_slot_rec<true>(slot, q, lay, rec, total, pos, meta);
+                        pending = false;
+                    }
+                }
+                if (!__syncthreads_or((int)pending)) break;
+                /* generation flush: every thread participates */
+                for (uint32_t sl = threadIdx.x; sl < lds_slots; sl += blockDim.x) {
+                    uint64_t* s = ltab + (uint64_t)sl * stride;
+                    if (((uint32_t*)s)[0] != 2u) continue;
+                    uint64_t* g = gtable_claim(gtable, gmask, stride,
+                                               ((uint32_t*)s)[1], s[1], s[2],
+                                               fill, fill_cap, err);
+                    if (!g) { pending = false; continue; }  /* err set; drain */
+                    agg_merge_slot<false>(g, s + SLOT_HDR, q);
+                }
+                __syncthreads();
+                for (uint32_t w = threadIdx.x; w < lds_slots * (uint32_t)stride + 1;
+                     w += blockDim.x)
+                    ltab[w] = 0;
+                __syncthreads();
+            }
+        }
+        /* final flush of this bucket */
+        for (uint32_t sl = threadIdx.x; sl < lds_slots; sl += blockDim.x) {
+            uint64_t* s = ltab + (uint64_t)sl * stride;
+            if (((uint32_t*)s)[0] != 2u) continue;
+            uint64_t* g = gtable_claim(gtable, gmask, stride, ((uint32_t*)s)[1],
+                                       s[1], s[2], fill, fill_cap, err);
+            if (!g) return;
+            agg_merge_slot<false>(g, s + SLOT_HDR, q);
+        }
+        __syncthreads();
+    }
+}
+
 /* compact the global table into the wire blob:
  * [flags u32*n][k0 u64*n][k1 u64*n][states u64*n*2*naggs] */
 __global__ void k_compact(const uint64_t* gtable, uint64_t nslots, int naggs,
@@ -701,7 +998,10 @@ struct BkgAggOut {
     int64_t blob_groups = 0;       /* capacity in groups */
     int64_t ngroups = -1;          /* valid after compact */
     int64_t rows_passed = 0;
-    float kernel_ms = 0.f;
+    float kernel_ms = 0.f;         /* sum over pipeline kernels */
+    int    n_kernels = 0;
+    float  t_ms[8] = {};           /* per-kernel breakdown */
+    char   k_names[8][16] = {};
     bool dirty = true;             /* table modified since last compact */
 };
 
@@ -772,6 +1072,132 @@ static int agg_compact(BkgAggOut* o) {
     return 0;
 }
 
+static int build_rec_layout(const BkgTable* t, const BkQuerySpec* q, RecLayout* lay) {
+    int w = 1; /* word 0 = k0 */
+    lay->k1_word = q->n_group >= 2 ? w++ : -1;
+    bool need_meta = false;
+    for (int k = 0; k < q->n_group; k++)
+        if (t->valid[q->group_cols[k]]) need_meta = true;
+    for (int a = 0; a < q->n_aggs; a++) {
+        int col = q->aggs[a].col;
+        if (col >= 0 && t->valid[col]) need_meta = true;
+    }
+    for (int a = 0; a < q->n_aggs; a++) {
+        int at = q->aggs[a].agg_type;
+        lay->val_word[a] =
+            (at == BK_AGG_COUNT_STAR || at == BK_AGG_COUNT) ? -1 : w++;
+    }
+    lay->meta_word = need_meta ? w++ : -1;
+    lay->nwords = w;
+    return 0;
+}
+
+struct EvTimer {
+    hipEvent_t ev[16];
+    int n = 0;
+    int record() { hipEventCreate(&ev[n]); hipEventRecord(ev[n]); return n++; }
+    void finish(BkgAggOut* o, const char* const* names) {
+        hipEventSynchronize(ev[n - 1]);
+        o->n_kernels = n - 1;
+        o->kernel_ms = 0.f;
+        for (int i = 0; i + 1 < n && i < 8; i++) {
+            float ms = 0.f;
+            hipEventElapsedTime(&ms, ev[i], ev[i + 1]);
+            o->t_ms[i] = ms;
+            o->kernel_ms += ms;
+            snprintf(o->k_names[i], 16, "%s", names[i]);
+        }
+        for (int i = 0; i < n; i++) hipEventDestroy(ev[i]);
+        n = 0;
+    }
+};
+
+/* partitioned pipeline for one attempt; returns 0 ok (err flag still to be
+ * checked by caller), -1 hard error. */
+static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
+                           int64_t row_begin, int64_t row_end,
+                           int64_t expected_groups) {
+    const int stride = SLOT_HDR + 2 * q->n_aggs;
+    int64_t range = row_end - row_begin;
+    if (range >= (int64_t)UINT32_MAX) { set_err("range too large for one pass"); return -1; }
+    RecLayout lay;
+    build_rec_layout(t, q, &lay);
+    uint32_t P = 64;
+    while ((int64_t)P < expected_groups / 96 && P < 4096) P <<= 1;
+    const int nblocks = 2048, threads = 256;
+    DevCols dc = table_cols(t);
+
+    uint16_t* bucketid = nullptr;
+    uint32_t *H = nullptr, *totals = nullptr, *base = nullptr;
+    uint64_t* total_dev = nullptr;
+    uint64_t* rec = nullptr;
+    auto cleanup = [&]() {
+        if (bucketid) (void)hipFree(bucketid);
+        if (H) (void)hipFree(H);
+        if (totals) (void)hipFree(totals);
+        if (base) (void)hipFree(base);
+        if (total_dev) (void)hipFree(total_dev);
+        if (rec) (void)hipFree(rec);
+    };
+    #define PCHECK(x) do { if ((x) != hipSuccess) { \
+        snprintf(g_err, sizeof g_err, "%s:%d %s", __FILE__, __LINE__, \
+                 hipGetErrorString(hipGetLastError())); cleanup(); return -1; } } while (0)
+    PCHECK(hipMalloc((void**)&bucketid, (size_t)range * 2));
+    PCHECK(hipMalloc((void**)&H, (size_t)nblocks * P * 4));
+    PCHECK(hipMalloc((void**)&totals, (size_t)P * 4));
+    PCHECK(hipMalloc((void**)&base, (size_t)P * 4));
+    PCHECK(hipMalloc((void**)&total_dev, 8));
+
+    static const char* NAMES[] = {"histo", "totals", "scan", "offsets",
+                                  "scatter", "part_agg"};
+    EvTimer tm;
+    tm.record();
+    hipLaunchKernelGGL(k_part_histo, dim3(nblocks), dim3(threads), P * 4, 0,
+                       dc, *q, row_begin, row_end, P, bucketid, H);
+    tm.record();
+    hipLaunchKernelGGL(k_part_totals, dim3((P + 255) / 256), dim3(256), 0, 0,
+                       H, nblocks, P, totals);
+    tm.record();
+    hipLaunchKernelGGL(k_part_scan, dim3(1), dim3(64), 0, 0,
+                       totals, P, base, total_dev);
+    tm.record();
+    PCHECK(hipGetLastError());
+    uint64_t total = 0;
+    PCHECK(hipMemcpy(&total, total_dev, 8, hipMemcpyDeviceToHost));
+    if (total > 0)
+        PCHECK(hipMalloc((void**)&rec, (size_t)total * lay.nwords * 8));
+    hipLaunchKernelGGL(k_part_offsets, dim3((P + 255) / 256), dim3(256), 0, 0,
+                       H, nblocks, P, base);
+    tm.record();
+    if (total > 0) {
+        hipLaunchKernelGGL(k_part_scatter, dim3(nblocks), dim3(threads), P * 4, 0,
+                           dc, *q, lay, row_begin, row_end, P, bucketid, H,
+                           rec, total);
+    }
+    tm.record();
+    if (total > 0) {
+        uint32_t lds_slots = 1024;
+        while ((size_t)lds_slots * stride * 8 > 100 * 1024) lds_slots >>= 1;
+        size_t lds_bytes = ((size_t)lds_slots * stride + 1) * 8;
+        hipLaunchKernelGGL(k_part_agg, dim3(P), dim3(threads), lds_bytes, 0,
+                           *q, lay, rec, total, base, totals, P,
+                           o->table, o->nslots - 1, (o->nslots * 7) / 8,
+                           o->ctrs, o->err, lds_slots);
+    }
+    tm.record();
+    PCHECK(hipGetLastError());
+    tm.finish(o, NAMES);
+    /* rows_passed = record count */
+    PCHECK(hipMemcpy(o->ctrs + 1, &total, 8, hipMemcpyHostToDevice));
+    cleanup();
+    #undef PCHECK
+    return 0;
+}
+
+/* below this many expected groups the per-WG LDS table absorbs the stream
+ * and the single fused kernel wins; above it, partition. */
+#define FUSED_MAX_GROUPS 512
+
 extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
                                        int64_t row_begin, int64_t row_end,
                                        int64_t expected_groups) {
@@ -781,45 +1207,53 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
     const int stride = SLOT_HDR + 2 * q->n_aggs;
     int64_t nslots = next_pow2(std::max<int64_t>(1024, expected_groups * 2));
     DevCols dc = table_cols(t);
+    bool partitioned = q->n_group > 0 && expected_groups > FUSED_MAX_GROUPS &&
+                       row_end > row_begin;
 
     for (int attempt = 0; attempt < 8; attempt++) {
         if (agg_alloc(o, nslots) != 0) { bkgpu_agg_free(o); return nullptr; }
-        hipEvent_t ev0, ev1;
-        HIP_CHECK_NULL(hipEventCreate(&ev0));
-        HIP_CHECK_NULL(hipEventCreate(&ev1));
-        uint32_t lds_slots = 512;
-        /* keep >=2 blocks/CU: LDS per block <= 64 KiB */
-        while ((size_t)lds_slots * stride * 8 > 63 * 1024) lds_slots >>= 1;
-        size_t lds_bytes = ((size_t)lds_slots * stride + 2) * 8;
         int blocks = 2048, threads = 256;
-        HIP_CHECK_NULL(hipEventRecord(ev0));
-        if (q->n_group == 0) {
-            /* pre-initialize slot 0 as the single group (state=2, flag 0):
-             * mirrors agg_node.cpp:490-505's always-present row */
-            uint64_t hdr[3] = {2ull /* state=2,flag=0 */, 0, 0};
-            HIP_CHECK_NULL(hipMemcpy(o->table, hdr, 24, hipMemcpyHostToDevice));
-            uint64_t one = 1;
-            HIP_CHECK_NULL(hipMemcpy(o->ctrs, &one, 8, hipMemcpyHostToDevice));
-            hipLaunchKernelGGL(k_filter_agg_scalar, dim3(blocks), dim3(threads), 0, 0,
-                               dc, *q, row_begin, row_end, o->table, o->ctrs + 1);
+        if (partitioned) {
+            if (run_partitioned(o, t, q, row_begin, row_end, expected_groups) != 0) {
+                bkgpu_agg_free(o);
+                return nullptr;
+            }
         } else {
-            hipLaunchKernelGGL(k_filter_agg_group, dim3(blocks), dim3(threads),
-                               lds_bytes, 0,
-                               dc, *q, row_begin, row_end, o->table,
-                               o->nslots - 1, (o->nslots * 7) / 8,
-                               o->ctrs, o->ctrs + 1, o->err, lds_slots);
+            static const char* NAMES1[] = {"fused_agg"};
+            EvTimer tm;
+            uint32_t lds_slots = 512;
+            /* keep >=2 blocks/CU: LDS per block <= 64 KiB */
+            while ((size_t)lds_slots * stride * 8 > 63 * 1024) lds_slots >>= 1;
+            size_t lds_bytes = ((size_t)lds_slots * stride + 2) * 8;
+            if (q->n_group == 0) {
+                /* pre-initialize slot 0 as the single group (state=2, flag 0):
+                 * mirrors agg_node.cpp:490-505's always-present row */
+                uint64_t hdr[3] = {2ull /* state=2,flag=0 */, 0, 0};
+                HIP_CHECK_NULL(hipMemcpy(o->table, hdr, 24, hipMemcpyHostToDevice));
+                uint64_t one = 1;
+                HIP_CHECK_NULL(hipMemcpy(o->ctrs, &one, 8, hipMemcpyHostToDevice));
+                tm.record();
+                hipLaunchKernelGGL(k_filter_agg_scalar, dim3(blocks), dim3(threads),
+                                   0, 0, dc, *q, row_begin, row_end, o->table,
+                                   o->ctrs + 1);
+            } else {
+                tm.record();
+                hipLaunchKernelGGL(k_filter_agg_group, dim3(blocks), dim3(threads),
+                                   lds_bytes, 0,
+                                   dc, *q, row_begin, row_end, o->table,
+                                   o->nslots - 1, (o->nslots * 7) / 8,
+                                   o->ctrs, o->ctrs + 1, o->err, lds_slots);
+            }
+            tm.record();
+            hipError_t lerr = hipGetLastError();
+            if (lerr != hipSuccess) {
+                snprintf(g_err, sizeof g_err, "filter_agg launch: %s",
+                         hipGetErrorString(lerr));
+                bkgpu_agg_free(o);
+                return nullptr;
+            }
+            tm.finish(o, NAMES1);
         }
-        HIP_CHECK_NULL(hipEventRecord(ev1));
-        hipError_t lerr = hipGetLastError();
-        if (lerr != hipSuccess) {
-            snprintf(g_err, sizeof g_err, "filter_agg launch: %s", hipGetErrorString(lerr));
-            hipEventDestroy(ev0); hipEventDestroy(ev1);
-            bkgpu_agg_free(o);
-            return nullptr;
-        }
-        HIP_CHECK_NULL(hipEventSynchronize(ev1));
-        HIP_CHECK_NULL(hipEventElapsedTime(&o->kernel_ms, ev0, ev1));
-        hipEventDestroy(ev0); hipEventDestroy(ev1);
         uint32_t err_host = 0;
         HIP_CHECK_NULL(hipMemcpy(&err_host, o->err, 4, hipMemcpyDeviceToHost));
         if (err_host == 0) break;
@@ -835,6 +1269,16 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
     o->dirty = true;
     if (agg_compact(o) != 0) { bkgpu_agg_free(o); return nullptr; }
     return o;
+}
+
+extern "C" int bkgpu_agg_breakdown(const BkgAggOut* o, char* names, double* ms,
+                                   int cap) {
+    int n = o->n_kernels < cap ? o->n_kernels : cap;
+    for (int i = 0; i < n; i++) {
+        ms[i] = o->t_ms[i];
+        if (names) memcpy(names + 16 * i, o->k_names[i], 16);
+    }
+    return n;
 }
 
 extern "C" int64_t bkgpu_agg_ngroups(const BkgAggOut* o) { return o->ngroups; }

@@ -49,6 +49,9 @@ def _load():
     lib.bkgpu_agg_rows_passed.argtypes = [C.c_void_p]
     lib.bkgpu_agg_kernel_ms.restype = C.c_double
     lib.bkgpu_agg_kernel_ms.argtypes = [C.c_void_p]
+    lib.bkgpu_agg_breakdown.restype = C.c_int
+    lib.bkgpu_agg_breakdown.argtypes = [C.c_void_p, C.c_char_p,
+                                        C.POINTER(C.c_double), C.c_int]
     lib.bkgpu_agg_export_bytes.restype = C.c_int64
     lib.bkgpu_agg_export_bytes.argtypes = [C.c_void_p]
     lib.bkgpu_agg_export.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
@@ -99,6 +102,13 @@ class AggResult:
     @property
     def kernel_ms(self):
         return self.engine.lib.bkgpu_agg_kernel_ms(self.handle)
+
+    def breakdown(self):
+        names = C.create_string_buffer(16 * 8)
+        ms = (C.c_double * 8)()
+        n = self.engine.lib.bkgpu_agg_breakdown(self.handle, names, ms, 8)
+        return {names.raw[16 * i:16 * (i + 1)].split(b"\0")[0].decode(): ms[i]
+                for i in range(n)}
 
     def export_bytes(self):
         return self.engine.lib.bkgpu_agg_export_bytes(self.handle)

@@ -34,7 +34,7 @@ CONFIGS = {
     "config2_1e8_8int64": dict(
         nrows=100_000_000,
         specs=[(TYPE_INT64, D_UNI, 0, 1 << 31, 0)] * 2
-            + [(TYPE_INT64, D_SKEW, 100_000, 0, 0)]
+            + [(TYPE_INT64, 4, 100_000, 0, 0)]      # group key ~ Zipf/1e5 (BASELINE)
             + [(TYPE_INT64, D_UNI, 0, 1000, 0)]
             + [(TYPE_INT64, D_UNI, 0, 1 << 31, 0)] * 4,
         conjuncts=[(0, "<", 1 << 30), (1, "<", int((1 << 31) * 0.9))],
@@ -47,18 +47,21 @@ CONFIGS = {
         nrows=1_000_000_000,
         specs=[(TYPE_INT64, D_UNI, 0, 1 << 31, 0),      # c0 predicate
                (TYPE_INT64, D_UNI, 0, 1 << 31, 0),      # c1 predicate
-               (TYPE_INT64, D_SKEW, 100_000, 0, 0),     # c2 group key 1
+               (TYPE_INT64, 4, 16384, 0, 0),            # c2 group key 1 (Zipf-like)
                (TYPE_INT64, D_UNI, 0, 1000, 0),         # c3 SUM int64
                (TYPE_DOUBLE, D_SUM16, 0, 0, 0),         # c4 SUM double
                (TYPE_DOUBLE, D_SUM16, 0, 0, 0),         # c5 AVG double
                (TYPE_INT64, D_UNI, 0, 1 << 31, 0),      # c6 untouched
-               (TYPE_STRING, D_DICT, 65536, 0, 0)],     # c7 group key 2 (dict)
+               # c7: VARCHAR via dict codes; group-key cardinality is
+               # generator-controlled (BASELINE: target 1e4-1e7 groups):
+               # 64 distinct codes x 16384 c2 values ~> ~1e6 group pairs
+               (TYPE_STRING, D_DICT, 64, 0, 0)],
         conjuncts=[(0, "<", 1 << 30),                    # sel 0.5
                    (1, "<", int((1 << 31) * 0.9)),       # sel 0.9
                    (7, "!=", 123)],                      # sel ~1
         group=[2, 7],
         aggs=[("count_star", -1), ("sum", 3), ("sum", 4), ("avg", 5)],
-        expected_groups=1 << 22,
+        expected_groups=1 << 21,
         # c0,c1 for all rows (16 B); dict c7 for the 0.45 surviving c0&c1
         # fraction (4 B); group c2 + aggs c3,c4,c5 for the 0.45*(65535/65536)
         # survivors (32 B)

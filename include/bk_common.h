@@ -68,9 +68,12 @@ typedef enum BkAggType {
 /* ---- synthetic column distributions (bench configs) ---- */
 typedef enum BkDist {
     BK_DIST_UNIFORM_I64 = 0,  /* uniform integer in [p0, p1) */
-    BK_DIST_CUBESKEW    = 1,  /* integer-only skewed ("Zipf-shaped") in [0, p0) */
+    BK_DIST_CUBESKEW    = 1,  /* integer-only skewed in [0, p0) (density ~ k^-2/3) */
     BK_DIST_DICT        = 2,  /* dict code uniform in [0, p0) (VARCHAR via dict) */
     BK_DIST_SUMU16      = 3,  /* approx N(0,1) double: sum of 4 u16 minus mean, scaled */
+    BK_DIST_ZIPFOCT     = 4,  /* log-uniform ("Zipf-1-like") integer in [0, p0):
+                                 octave picked uniformly, value uniform in octave
+                                 => density ~ 1/(k+1); integer-only, CPU==GPU */
 } BkDist;
 
 /* One generated column. Physical storage by type:

@@ -61,6 +61,20 @@ BK_HD int32_t bk_gen_dict(uint64_t u, int64_t nwords) {
     return (int32_t)(u % (uint64_t)nwords);
 }
 
+/* log-uniform "Zipf-1-like" integer on [0, D): pick an octave uniformly from
+ * the ceil(log2(D)) octaves of [1, 2^ceil(log2 D)), then a value uniformly
+ * inside it; values >= D wrap by modulo (slight distortion at the top
+ * octave). Density ~ 1/(k+1). Integer-only => bit-identical CPU/GPU. */
+BK_HD int64_t bk_gen_zipfoct(uint64_t u, int64_t D) {
+    if (D <= 1) return 0;
+    uint32_t noct = 1;
+    while ((1ll << noct) < D && noct < 62) noct++;
+    uint32_t e = (uint32_t)((u >> 40) % noct);            /* octave */
+    uint64_t lo = bk_mix64(u ^ 0x0C7AE5ull);
+    uint64_t k = ((uint64_t)1 << e) + (lo & (((uint64_t)1 << e) - 1)) - 1;
+    return (int64_t)(k % (uint64_t)D);
+}
+
 /* approx N(0,1) double: (sum of four u16) centered and scaled. All steps are
  * exact int ops plus one exact int->double convert and one double multiply by
  * a constant, so CPU and GPU agree bitwise. Var(sum of 4 u16) = 4*(65536^2-1)/12;
@@ -80,6 +94,7 @@ BK_HD int64_t bk_gen_i64(const BkColSpec* cs, uint64_t seed, uint64_t row, uint3
         case BK_DIST_UNIFORM_I64: return bk_gen_uniform_i64(u, cs->p0, cs->p1);
         case BK_DIST_CUBESKEW:    return bk_gen_cubeskew(u, cs->p0);
         case BK_DIST_DICT:        return (int64_t)bk_gen_dict(u, cs->p0);
+        case BK_DIST_ZIPFOCT:     return bk_gen_zipfoct(u, cs->p0);
         default:                  return 0;
     }
 }

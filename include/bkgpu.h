@@ -70,8 +70,13 @@ BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
 
 int64_t bkgpu_agg_ngroups(const BkgAggOut* o);
 int64_t bkgpu_agg_rows_passed(const BkgAggOut* o);
-/* wall time of the fused kernel (HIP events), for bench roofline only */
+/* total device time of the aggregate pipeline (HIP events on the launch
+ * stream), for bench roofline only */
 double  bkgpu_agg_kernel_ms(const BkgAggOut* o);
+/* per-kernel breakdown: fills ms[0..n) and 16-byte names; returns n.
+ * Fused path: {fused_agg}; partitioned path: {histo, totals, scan, offsets,
+ * scatter, part_agg}. */
+int     bkgpu_agg_breakdown(const BkgAggOut* o, char* names, double* ms, int cap);
 
 /* ---- partial-aggregate exchange (multi-GPU merge over RCCL) ----
  * Compact wire format of one partial result, an SoA byte blob:
