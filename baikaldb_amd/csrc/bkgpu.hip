@@ -753,6 +753,78 @@ __device__ __forceinline__ void agg_update_slot_rec(uint64_t* st, const BkQueryS
     }
 }
 
+/* wave-level combine: when ALL 64 lanes of a wave hold records of the SAME
+ * group (the common case inside a hot key's bucket — hash partitioning sends
+ * a hot group's rows to one bucket, so its waves are single-key), butterfly-
+ * reduce the agg inputs across the wave and issue ONE slot update instead of
+ * 64 serialized same-address LDS atomics. Returns true if handled. */
+__device__ __forceinline__ bool wave_combine_update(
+        uint64_t* slot, const BkQuerySpec& q, const RecLayout& lay,
+        const uint64_t* rec, uint64_t total, uint64_t pos, uint64_t meta,
+        bool lds) {
+    int lane = threadIdx.x & 63;
+    #pragma unroll 4
+    for (int32_t a = 0; a < q.n_aggs; a++) {
+        int at = q.aggs[a].agg_type;
+        int has_meta = lay.meta_word >= 0;
+        int valid = !has_meta || at == BK_AGG_COUNT_STAR || ((meta >> (8 + a)) & 1);
+        uint64_t addv = 0;   /* combined value */
+        double addd = 0.0;
+        uint64_t addc = 0;   /* combined count */
+        uint64_t w = (valid && lay.val_word[a] >= 0)
+                         ? rec[(size_t)lay.val_word[a] * total + pos] : 0;
+        switch (at) {
+            case BK_AGG_COUNT_STAR: addv = 1; break;
+            case BK_AGG_COUNT:      addv = valid ? 1 : 0; break;
+            case BK_AGG_SUM:
+                if (q.agg_in_types[a] == BK_DOUBLE) { if (valid) memcpy(&addd, &w, 8); }
+                else addv = valid ? w : 0;
+                addc = valid ? 1 : 0;
+                break;
+            case BK_AGG_AVG:
+                if (valid) memcpy(&addd, &w, 8);
+                addc = valid ? 1 : 0;
+                break;
+            case BK_AGG_MIN: addv = valid ? ~w : 0; addc = valid ? 1 : 0; break;
+            case BK_AGG_MAX: addv = valid ? w : 0;  addc = valid ? 1 : 0; break;
+            default: break;
+        }
+        /* butterfly over the full wave */
+        bool is_minmax = (at == BK_AGG_MIN || at == BK_AGG_MAX);
+        #pragma unroll
+        for (int off = 32; off > 0; off >>= 1) {
+            uint64_t ov = __shfl_xor((unsigned long long)addv, off, 64);
+            addc += __shfl_xor((unsigned long long)addc, off, 64);
+            addd += __shfl_xor(addd, off, 64);
+            addv = is_minmax ? (addv > ov ? addv : ov) : addv + ov;
+        }
+        if (lane == 0) {
+            uint64_t* val = slot + SLOT_HDR + 2 * a;
+            uint64_t* cnt = val + 1;
+            if (at == BK_AGG_COUNT_STAR || at == BK_AGG_COUNT) {
+                if (addv) atomicAdd((unsigned long long*)val, (unsigned long long)addv);
+            } else if (is_minmax) {
+                if (addc) {
+                    atomicMax((unsigned long long*)val, (unsigned long long)addv);
+                    atomicAdd((unsigned long long*)cnt, (unsigned long long)addc);
+                }
+            } else if (q.agg_in_types[a] == BK_DOUBLE || at == BK_AGG_AVG) {
+                if (addc) {
+                    if (lds) atomic_add_f64_lds(val, addd);
+                    else     atomic_add_f64_global(val, addd);
+                    atomicAdd((unsigned long long*)cnt, (unsigned long long)addc);
+                }
+            } else {
+                if (addc) {
+                    atomicAdd((unsigned long long*)val, (unsigned long long)addv);
+                    atomicAdd((unsigned long long*)cnt, (unsigned long long)addc);
+                }
+            }
+        }
+    }
+    return true;
+}
+
 /* pass 3: one workgroup per bucket — aggregate its records in a per-WG LDS
  * table; when the LDS table fills (more distinct groups in the bucket than
  * slots), the whole block flushes it into the global table (additive merge),
@@ -791,6 +863,26 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
                 flag = (uint32_t)(meta & 0xFF);
             }
             for (;;) {
+                /* hot-bucket fast path: whole wave carries one group */
+                uint64_t wm = __ballot(pending);
+                if (wm == 0xFFFFFFFFFFFFFFFFull) {
+                    uint64_t f0 = __shfl((unsigned long long)k0, 0, 64);
+                    uint64_t f1 = __shfl((unsigned long long)k1, 0, 64);
+                    uint32_t ff = (uint32_t)__shfl((int)flag, 0, 64);
+                    if (__all(k0 == f0 && k1 == f1 && flag == ff)) {
+                        uint64_t* slot = nullptr;
+                        if ((threadIdx.x & 63) == 0)
+                            slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
+                                                lfill, lcap);
+                        int ok = __shfl((int)(slot != nullptr), 0, 64);
+                        if (ok) {
+                            wave_combine_update(slot, q, lay, rec, total, pos,
+                                                meta, true);
+                            pending = false;
+                        }
+                        goto vote;  /* skip per-lane path this round */
+                    }
+                }
                 if (pending) {
                     uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
                                                   lfill, lcap);
@@ -799,6 +891,7 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
                         pending = false;
                     }
                 }
+vote:
                 if (!__syncthreads_or((int)pending)) break;
                 /* generation flush: every thread participates */
                 for (uint32_t sl = threadIdx.x; sl < lds_slots; sl += blockDim.x) {
@@ -1310,7 +1403,7 @@ extern "C" int bkgpu_agg_export(const BkgAggOut* o, void* dst, int64_t cap) {
     return 0;
 }
 
-extern "C" int bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups) {
+static int merge_blob_into(BkgAggOut* o, const void* blob, int64_t n_groups) {
     const uint8_t* b = (const uint8_t*)blob;
     const uint32_t* flags = (const uint32_t*)b;
     const uint64_t* k0 = (const uint64_t*)(b + n_groups * 4);
@@ -1325,7 +1418,37 @@ extern "C" int bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups)
     uint32_t err_host = 0;
     HIP_CHECK(hipMemcpy(&err_host, o->err, 4, hipMemcpyDeviceToHost));
     if (err_host) { set_err("merge overflowed dst table"); return -1; }
+    return 0;
+}
+
+extern "C" int bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups) {
+    /* k_merge_blob mutates the table as it goes, so an overflow mid-merge is
+     * unrecoverable — ensure capacity FIRST: if fill+n could cross the cap,
+     * rebuild the table at a larger size from our own compact blob (additive
+     * re-insert from zero = exact), then merge the peer. */
+    uint64_t fill = 0;
+    HIP_CHECK(hipMemcpy(&fill, o->ctrs, 8, hipMemcpyDeviceToHost));
+    if ((fill + (uint64_t)n_groups) * 8 > o->nslots * 7) {
+        if (agg_compact(o) != 0) return -1;  /* own groups -> o->blob */
+        uint8_t* own = nullptr;
+        int64_t own_n = o->ngroups;
+        int64_t own_bytes = bkgpu_agg_export_bytes(o);
+        HIP_CHECK(hipMalloc((void**)&own, (size_t)(own_bytes > 0 ? own_bytes : 1)));
+        if (bkgpu_agg_export(o, own, own_bytes) != 0) { (void)hipFree(own); return -1; }
+        uint64_t rows_passed = (uint64_t)o->rows_passed;
+        agg_release_table(o);
+        int64_t nslots = next_pow2((int64_t)((fill + n_groups) * 2 + 1024));
+        if (agg_alloc(o, nslots) != 0) { (void)hipFree(own); return -1; }
+        HIP_CHECK(hipMemcpy(o->ctrs + 1, &rows_passed, 8, hipMemcpyHostToDevice));
+        if (own_n > 0 && merge_blob_into(o, own, own_n) != 0) {
+            (void)hipFree(own);
+            return -1;
+        }
+        (void)hipFree(own);
+    }
+    if (merge_blob_into(o, blob, n_groups) != 0) return -1;
     o->dirty = true;
+    o->ngroups = -1;
     return 0;
 }
 
