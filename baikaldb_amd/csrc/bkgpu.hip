@@ -832,7 +832,7 @@ __device__ __forceinline__ bool wave_combine_update(
  * traffic proportional to #groups x generations, never to #rows. */
 __global__ void __launch_bounds__(256)
 k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
-           const uint32_t* base, const uint32_t* totals, uint32_t P,
+           uint64_t chunk,
            uint64_t* gtable, uint64_t gmask, uint64_t fill_cap, uint64_t* fill,
            uint32_t* err, uint32_t lds_slots) {
     extern __shared__ __attribute__((aligned(16))) uint64_t ltab[];
@@ -841,14 +841,17 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
     uint32_t* lfill = (uint32_t*)&laux[0];
     const uint32_t lmask = lds_slots - 1;
     const uint32_t lcap = lds_slots - 256;  /* room for one tile of claims */
-    for (uint32_t b = blockIdx.x; b < P; b += gridDim.x) {
-        uint32_t n = totals[b];
-        if (n == 0) continue;
+    /* fixed-size CHUNKS of the (bucket-sorted) record array, not buckets:
+     * a hot bucket gets many workgroups, and a chunk still spans only 1-2
+     * buckets' worth of distinct groups for the LDS table. */
+    uint64_t nchunks = (total + chunk - 1) / chunk;
+    for (uint64_t c = blockIdx.x; c < nchunks; c += gridDim.x) {
+        uint64_t b0 = c * chunk;
+        uint32_t n = (uint32_t)((total - b0 < chunk) ? (total - b0) : chunk);
         for (uint32_t w = threadIdx.x; w < lds_slots * (uint32_t)stride + 1;
              w += blockDim.x)
             ltab[w] = 0;
         __syncthreads();
-        uint32_t b0 = base[b];
         /* uniform tile loop so generation flushes can barrier */
         for (uint32_t t0 = 0; t0 < n; t0 += blockDim.x) {
             uint32_t i = t0 + threadIdx.x;
@@ -1272,8 +1275,11 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
         uint32_t lds_slots = 1024;
         while ((size_t)lds_slots * stride * 8 > 100 * 1024) lds_slots >>= 1;
         size_t lds_bytes = ((size_t)lds_slots * stride + 1) * 8;
-        hipLaunchKernelGGL(k_part_agg, dim3(P), dim3(threads), lds_bytes, 0,
-                           *q, lay, rec, total, base, totals, P,
+        uint64_t chunk = 32768;
+        uint64_t nchunks = (total + chunk - 1) / chunk;
+        uint32_t grid = (uint32_t)std::min<uint64_t>(nchunks, 32768);
+        hipLaunchKernelGGL(k_part_agg, dim3(grid), dim3(threads), lds_bytes, 0,
+                           *q, lay, rec, total, chunk,
                            o->table, o->nslots - 1, (o->nslots * 7) / 8,
                            o->ctrs, o->err, lds_slots);
     }
@@ -1374,7 +1380,13 @@ extern "C" int bkgpu_agg_breakdown(const BkgAggOut* o, char* names, double* ms,
     return n;
 }
 
-extern "C" int64_t bkgpu_agg_ngroups(const BkgAggOut* o) { return o->ngroups; }
+extern "C" int64_t bkgpu_agg_ngroups(const BkgAggOut* o) {
+    BkgAggOut* m = const_cast<BkgAggOut*>(o);
+    if (m->dirty || m->ngroups < 0) {
+        if (agg_compact(m) != 0) return -1;
+    }
+    return o->ngroups;
+}
 extern "C" int64_t bkgpu_agg_rows_passed(const BkgAggOut* o) { return o->rows_passed; }
 extern "C" double bkgpu_agg_kernel_ms(const BkgAggOut* o) { return o->kernel_ms; }
 
