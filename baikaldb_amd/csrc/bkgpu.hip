@@ -591,33 +591,65 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
         H[(size_t)blockIdx.x * P + b] = lhist[b];
 }
 
-/* bucket totals over all blocks */
+/* bucket totals over all blocks, tiled (b, chunk-of-blocks) for parallelism;
+ * per-tile partials S[chunk][b] are reused by k_part_offsets. */
+#define OFFS_CHUNK 128u
 __global__ void k_part_totals(const uint32_t* H, uint32_t nblocks, uint32_t P,
-                              uint32_t* totals) {
-    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
-    if (b >= P) return;
+                              uint32_t* S, uint32_t* totals) {
+    uint32_t nchunks = (nblocks + OFFS_CHUNK - 1) / OFFS_CHUNK;
+    uint32_t idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= P * nchunks) return;
+    uint32_t b = idx % P, chunk = idx / P;
+    uint32_t b1 = (chunk + 1) * OFFS_CHUNK;
+    if (b1 > nblocks) b1 = nblocks;
     uint32_t s = 0;
-    for (uint32_t blk = 0; blk < nblocks; blk++) s += H[(size_t)blk * P + b];
-    totals[b] = s;
+    for (uint32_t blk = chunk * OFFS_CHUNK; blk < b1; blk++)
+        s += H[(size_t)blk * P + b];
+    S[(size_t)chunk * P + b] = s;
+    atomicAdd(&totals[b], s);
 }
 
-/* exclusive scan of totals -> base; grand total -> total_out. single block. */
-__global__ void k_part_scan(const uint32_t* totals, uint32_t P, uint32_t* base,
-                            uint64_t* total_out) {
-    if (threadIdx.x == 0) {
-        uint64_t run = 0;
-        for (uint32_t b = 0; b < P; b++) { base[b] = (uint32_t)run; run += totals[b]; }
-        *total_out = run;
+/* exclusive scan of totals -> base; grand total -> total_out. One workgroup,
+ * per-thread serial pre-scan of P/T elements + Hillis-Steele over partials. */
+__global__ void __launch_bounds__(1024)
+k_part_scan(const uint32_t* totals, uint32_t P, uint32_t* base,
+            uint64_t* total_out) {
+    __shared__ uint32_t part[1024];
+    uint32_t T = blockDim.x;
+    uint32_t per = (P + T - 1) / T;
+    uint32_t lo = threadIdx.x * per, hi = lo + per;
+    if (hi > P) hi = P;
+    uint32_t s = 0;
+    for (uint32_t b = lo; b < hi; b++) s += totals[b];
+    part[threadIdx.x] = s;
+    __syncthreads();
+    /* Hillis-Steele inclusive scan over partials */
+    for (uint32_t off = 1; off < T; off <<= 1) {
+        uint32_t v = threadIdx.x >= off ? part[threadIdx.x - off] : 0;
+        __syncthreads();
+        part[threadIdx.x] += v;
+        __syncthreads();
     }
+    uint32_t run = threadIdx.x ? part[threadIdx.x - 1] : 0;  /* exclusive */
+    for (uint32_t b = lo; b < hi; b++) {
+        base[b] = run;
+        run += totals[b];
+    }
+    if (threadIdx.x == T - 1 && total_out) *total_out = part[T - 1];
 }
 
-/* turn H[blk][b] counts into absolute start offsets */
+/* turn H[blk][b] counts into absolute start offsets, tiled like totals */
 __global__ void k_part_offsets(uint32_t* H, uint32_t nblocks, uint32_t P,
-                               const uint32_t* base) {
-    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
-    if (b >= P) return;
+                               const uint32_t* base, const uint32_t* S) {
+    uint32_t nchunks = (nblocks + OFFS_CHUNK - 1) / OFFS_CHUNK;
+    uint32_t idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= P * nchunks) return;
+    uint32_t b = idx % P, chunk = idx / P;
     uint32_t run = base[b];
-    for (uint32_t blk = 0; blk < nblocks; blk++) {
+    for (uint32_t c = 0; c < chunk; c++) run += S[(size_t)c * P + b];
+    uint32_t b1 = (chunk + 1) * OFFS_CHUNK;
+    if (b1 > nblocks) b1 = nblocks;
+    for (uint32_t blk = chunk * OFFS_CHUNK; blk < b1; blk++) {
         uint32_t t = H[(size_t)blk * P + b];
         H[(size_t)blk * P + b] = run;
         run += t;
@@ -1224,7 +1256,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     DevCols dc = table_cols(t);
 
     uint16_t* bucketid = nullptr;
-    uint32_t *H = nullptr, *totals = nullptr, *base = nullptr;
+    uint32_t *H = nullptr, *totals = nullptr, *base = nullptr, *S = nullptr;
     uint64_t* total_dev = nullptr;
     uint64_t* rec = nullptr;
     auto cleanup = [&]() {
@@ -1232,16 +1264,20 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
         if (H) (void)hipFree(H);
         if (totals) (void)hipFree(totals);
         if (base) (void)hipFree(base);
+        if (S) (void)hipFree(S);
         if (total_dev) (void)hipFree(total_dev);
         if (rec) (void)hipFree(rec);
     };
     #define PCHECK(x) do { if ((x) != hipSuccess) { \
         snprintf(g_err, sizeof g_err, "%s:%d %s", __FILE__, __LINE__, \
                  hipGetErrorString(hipGetLastError())); cleanup(); return -1; } } while (0)
+    const uint32_t nchunks = (nblocks + OFFS_CHUNK - 1) / OFFS_CHUNK;
     PCHECK(hipMalloc((void**)&bucketid, (size_t)range * 2));
     PCHECK(hipMalloc((void**)&H, (size_t)nblocks * P * 4));
     PCHECK(hipMalloc((void**)&totals, (size_t)P * 4));
+    PCHECK(hipMemset(totals, 0, (size_t)P * 4));
     PCHECK(hipMalloc((void**)&base, (size_t)P * 4));
+    PCHECK(hipMalloc((void**)&S, (size_t)nchunks * P * 4));
     PCHECK(hipMalloc((void**)&total_dev, 8));
 
     static const char* NAMES[] = {"histo", "totals", "scan", "offsets",
@@ -1251,10 +1287,10 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     hipLaunchKernelGGL(k_part_histo, dim3(nblocks), dim3(threads), P * 4, 0,
                        dc, *q, row_begin, row_end, P, bucketid, H);
     tm.record();
-    hipLaunchKernelGGL(k_part_totals, dim3((P + 255) / 256), dim3(256), 0, 0,
-                       H, nblocks, P, totals);
+    hipLaunchKernelGGL(k_part_totals, dim3((P * nchunks + 255) / 256), dim3(256),
+                       0, 0, H, nblocks, P, S, totals);
     tm.record();
-    hipLaunchKernelGGL(k_part_scan, dim3(1), dim3(64), 0, 0,
+    hipLaunchKernelGGL(k_part_scan, dim3(1), dim3(1024), 0, 0,
                        totals, P, base, total_dev);
     tm.record();
     PCHECK(hipGetLastError());
@@ -1262,8 +1298,8 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     PCHECK(hipMemcpy(&total, total_dev, 8, hipMemcpyDeviceToHost));
     if (total > 0)
         PCHECK(hipMalloc((void**)&rec, (size_t)total * lay.nwords * 8));
-    hipLaunchKernelGGL(k_part_offsets, dim3((P + 255) / 256), dim3(256), 0, 0,
-                       H, nblocks, P, base);
+    hipLaunchKernelGGL(k_part_offsets, dim3((P * nchunks + 255) / 256), dim3(256),
+                       0, 0, H, nblocks, P, base, S);
     tm.record();
     if (total > 0) {
         hipLaunchKernelGGL(k_part_scatter, dim3(nblocks), dim3(threads), P * 4, 0,
