@@ -1006,6 +1006,43 @@ struct BkgTable {
     uint8_t* valid[BK_MAX_COLS] = {};
 };
 
+/* ---- device memory pool: query-scoped buffers (hash tables, partition
+ * records, blobs) are reallocated every query at identical sizes; hipMalloc
+ * of an 18 GB record buffer costs ~100s of ms, so cache freed buffers by
+ * exact size. Host driving is single-threaded (one bthread drives the exec
+ * tree in the reference too — SURVEY §8b threading contract). ---- */
+#include <map>
+static std::multimap<size_t, void*> g_pool_free;
+static std::map<void*, size_t> g_pool_sizes;
+
+static hipError_t pool_alloc(void** p, size_t bytes) {
+    auto it = g_pool_free.lower_bound(bytes);
+    /* reuse only if within 2x of the request (avoid hoarding) */
+    if (it != g_pool_free.end() && it->first <= bytes * 2) {
+        *p = it->second;
+        g_pool_free.erase(it);
+        return hipSuccess;
+    }
+    hipError_t e = hipMalloc(p, bytes);
+    if (e == hipSuccess) g_pool_sizes[*p] = bytes;
+    return e;
+}
+
+static void pool_free(void* p) {
+    if (!p) return;
+    auto it = g_pool_sizes.find(p);
+    if (it == g_pool_sizes.end()) { (void)hipFree(p); return; }
+    g_pool_free.insert({it->second, p});
+}
+
+extern "C" void bkgpu_pool_trim(void) {
+    for (auto& kv : g_pool_free) {
+        (void)hipFree(kv.second);
+        g_pool_sizes.erase(kv.second);
+    }
+    g_pool_free.clear();
+}
+
 static int g_device_set = 0;
 static int ensure_device() {
     if (!g_device_set) {
@@ -1146,7 +1183,7 @@ static size_t blob_bytes_for(int naggs, int64_t n) {
 static int agg_alloc(BkgAggOut* o, int64_t nslots) {
     const int stride = SLOT_HDR + 2 * o->q.n_aggs;
     o->nslots = (uint64_t)nslots;
-    HIP_CHECK(hipMalloc(&o->table, (size_t)nslots * stride * 8));
+    HIP_CHECK(pool_alloc((void**)&o->table, (size_t)nslots * stride * 8));
     HIP_CHECK(hipMemset(o->table, 0, (size_t)nslots * stride * 8));
     if (!o->ctrs) {
         HIP_CHECK(hipMalloc(&o->ctrs, 3 * 8));
@@ -1158,7 +1195,7 @@ static int agg_alloc(BkgAggOut* o, int64_t nslots) {
 }
 
 static void agg_release_table(BkgAggOut* o) {
-    if (o->table) { hipFree(o->table); o->table = nullptr; }
+    if (o->table) { pool_free(o->table); o->table = nullptr; }
 }
 
 extern "C" void bkgpu_agg_free(BkgAggOut* o) {
@@ -1166,7 +1203,7 @@ extern "C" void bkgpu_agg_free(BkgAggOut* o) {
     agg_release_table(o);
     if (o->ctrs) hipFree(o->ctrs);
     if (o->err) hipFree(o->err);
-    if (o->blob) hipFree(o->blob);
+    if (o->blob) pool_free(o->blob);
     delete o;
 }
 
@@ -1181,8 +1218,8 @@ static int agg_compact(BkgAggOut* o) {
     int64_t fill = (int64_t)ctr_host[0];
     if (o->q.n_group == 0) fill = 1;
     cap = fill > 0 ? fill : 1;
-    if (o->blob) { hipFree(o->blob); o->blob = nullptr; }
-    HIP_CHECK(hipMalloc(&o->blob, blob_bytes_for(naggs, cap)));
+    if (o->blob) { pool_free(o->blob); o->blob = nullptr; }
+    HIP_CHECK(pool_alloc((void**)&o->blob, blob_bytes_for(naggs, cap)));
     o->blob_groups = cap;
     HIP_CHECK(hipMemset(o->ctrs + 2, 0, 8));
     uint32_t* flags = (uint32_t*)o->blob;
@@ -1260,25 +1297,20 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     uint64_t* total_dev = nullptr;
     uint64_t* rec = nullptr;
     auto cleanup = [&]() {
-        if (bucketid) (void)hipFree(bucketid);
-        if (H) (void)hipFree(H);
-        if (totals) (void)hipFree(totals);
-        if (base) (void)hipFree(base);
-        if (S) (void)hipFree(S);
-        if (total_dev) (void)hipFree(total_dev);
-        if (rec) (void)hipFree(rec);
+        pool_free(bucketid); pool_free(H); pool_free(totals); pool_free(base);
+        pool_free(S); pool_free(total_dev); pool_free(rec);
     };
     #define PCHECK(x) do { if ((x) != hipSuccess) { \
         snprintf(g_err, sizeof g_err, "%s:%d %s", __FILE__, __LINE__, \
                  hipGetErrorString(hipGetLastError())); cleanup(); return -1; } } while (0)
     const uint32_t nchunks = (nblocks + OFFS_CHUNK - 1) / OFFS_CHUNK;
-    PCHECK(hipMalloc((void**)&bucketid, (size_t)range * 2));
-    PCHECK(hipMalloc((void**)&H, (size_t)nblocks * P * 4));
-    PCHECK(hipMalloc((void**)&totals, (size_t)P * 4));
+    PCHECK(pool_alloc((void**)&bucketid, (size_t)range * 2));
+    PCHECK(pool_alloc((void**)&H, (size_t)nblocks * P * 4));
+    PCHECK(pool_alloc((void**)&totals, (size_t)P * 4));
     PCHECK(hipMemset(totals, 0, (size_t)P * 4));
-    PCHECK(hipMalloc((void**)&base, (size_t)P * 4));
-    PCHECK(hipMalloc((void**)&S, (size_t)nchunks * P * 4));
-    PCHECK(hipMalloc((void**)&total_dev, 8));
+    PCHECK(pool_alloc((void**)&base, (size_t)P * 4));
+    PCHECK(pool_alloc((void**)&S, (size_t)nchunks * P * 4));
+    PCHECK(pool_alloc((void**)&total_dev, 8));
 
     static const char* NAMES[] = {"histo", "totals", "scan", "offsets",
                                   "scatter", "part_agg"};
@@ -1297,7 +1329,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     uint64_t total = 0;
     PCHECK(hipMemcpy(&total, total_dev, 8, hipMemcpyDeviceToHost));
     if (total > 0)
-        PCHECK(hipMalloc((void**)&rec, (size_t)total * lay.nwords * 8));
+        PCHECK(pool_alloc((void**)&rec, (size_t)total * lay.nwords * 8));
     hipLaunchKernelGGL(k_part_offsets, dim3((P * nchunks + 255) / 256), dim3(256),
                        0, 0, H, nblocks, P, base, S);
     tm.record();
@@ -1481,18 +1513,18 @@ extern "C" int bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups)
         uint8_t* own = nullptr;
         int64_t own_n = o->ngroups;
         int64_t own_bytes = bkgpu_agg_export_bytes(o);
-        HIP_CHECK(hipMalloc((void**)&own, (size_t)(own_bytes > 0 ? own_bytes : 1)));
-        if (bkgpu_agg_export(o, own, own_bytes) != 0) { (void)hipFree(own); return -1; }
+        HIP_CHECK(pool_alloc((void**)&own, (size_t)(own_bytes > 0 ? own_bytes : 1)));
+        if (bkgpu_agg_export(o, own, own_bytes) != 0) { pool_free(own); return -1; }
         uint64_t rows_passed = (uint64_t)o->rows_passed;
         agg_release_table(o);
         int64_t nslots = next_pow2((int64_t)((fill + n_groups) * 2 + 1024));
-        if (agg_alloc(o, nslots) != 0) { (void)hipFree(own); return -1; }
+        if (agg_alloc(o, nslots) != 0) { pool_free(own); return -1; }
         HIP_CHECK(hipMemcpy(o->ctrs + 1, &rows_passed, 8, hipMemcpyHostToDevice));
         if (own_n > 0 && merge_blob_into(o, own, own_n) != 0) {
-            (void)hipFree(own);
+            pool_free(own);
             return -1;
         }
-        (void)hipFree(own);
+        pool_free(own);
     }
     if (merge_blob_into(o, blob, n_groups) != 0) return -1;
     o->dirty = true;
