@@ -1289,7 +1289,12 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     build_rec_layout(t, q, &lay);
     uint32_t P = 64;
     while ((int64_t)P < expected_groups / 96 && P < 4096) P <<= 1;
-    const int nblocks = 2048, threads = 256;
+    const char* envP = getenv("BK_PART_P");
+    if (envP) P = (uint32_t)atoi(envP);
+    int nblocks = 2048;
+    const char* envB = getenv("BK_PART_BLOCKS");
+    if (envB) nblocks = atoi(envB);
+    const int threads = 256;
     DevCols dc = table_cols(t);
 
     uint16_t* bucketid = nullptr;
