@@ -558,20 +558,37 @@ struct RecLayout {
 
 #define PART_BUCKET(h, P) ((uint32_t)((h) >> 44) & ((P) - 1u))
 #define BK_SKIP_BUCKET 0xFFFFu
+#define BK_HOT_BUCKET  0xFFFEu
 
-/* pass 1: predicate + per-block bucket histogram; bucketid[] remembers the
- * verdict so later passes re-evaluate nothing. */
+/* pass 1 (hot/cold hybrid): predicate, then try the per-WG LDS aggregate
+ * table first — under Zipf-shaped keys the first-come LDS set absorbs the
+ * hot head of the distribution right here (no record, no later passes).
+ * Cold rows get a bucket id + per-block histogram for the partition passes.
+ * LDS carve: [ hot table: lds_slots*stride u64 | laux 2 u64 | lhist P u32 ]. */
 __global__ void __launch_bounds__(256)
 k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
-             uint32_t P, uint16_t* bucketid, uint32_t* H) {
-    extern __shared__ __attribute__((aligned(16))) uint32_t lhist[];
+             uint32_t P, uint16_t* bucketid, uint32_t* H,
+             uint64_t* gtable, uint64_t gmask, uint64_t fill_cap, uint64_t* fill,
+             uint64_t* rows_passed, uint32_t* err, uint32_t lds_slots) {
+    extern __shared__ __attribute__((aligned(16))) uint64_t ltab[];
+    const int stride = SLOT_HDR + 2 * q.n_aggs;
+    uint64_t* laux = ltab + (size_t)lds_slots * stride;
+    uint32_t* lfill = (uint32_t*)&laux[0];
+    uint32_t* lhist = (uint32_t*)&laux[2];
+    for (uint32_t w = threadIdx.x; w < lds_slots * (uint32_t)stride + 2;
+         w += blockDim.x)
+        ltab[w] = 0;
     for (uint32_t b = threadIdx.x; b < P; b += blockDim.x) lhist[b] = 0;
     __syncthreads();
+    const uint32_t lmask = lds_slots - 1;
+    const uint32_t lcap = (lds_slots * 3u) / 4u;
+    int64_t my_passed = 0;
     int64_t gstride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          r < row_end; r += gstride) {
         int64_t i = r - row_begin;
         if (!row_passes(cols, q, r)) { bucketid[i] = (uint16_t)BK_SKIP_BUCKET; continue; }
+        my_passed++;
         uint32_t flag = 0;
         uint64_t k0 = 0, k1 = 0;
         if (q.n_group >= 1) {
@@ -582,13 +599,37 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
             const DevCol& c = cols.c[q.group_cols[1]];
             if (!cell_valid(c, r)) flag |= 0x40u; else k1 = enc_value(c, r);
         }
+        uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
+                                      lfill, lcap);
+        if (slot) {
+            agg_update_slot<true>(slot, q, cols, r);
+            bucketid[i] = (uint16_t)BK_HOT_BUCKET;
+            continue;
+        }
         uint32_t b = PART_BUCKET(key_hash(flag, k0, k1), P);
         bucketid[i] = (uint16_t)b;
         atomicAdd(&lhist[b], 1u);
     }
+    /* rows_passed: reduce per wave, then one atomic per block via laux[1] */
+    long long w = my_passed;
+    for (int off = 32; off > 0; off >>= 1) w += __shfl_down(w, off, 64);
+    if ((threadIdx.x & 63) == 0)
+        atomicAdd((unsigned long long*)&laux[1], (unsigned long long)w);
     __syncthreads();
+    if (threadIdx.x == 0)
+        atomicAdd((unsigned long long*)rows_passed, (unsigned long long)laux[1]);
     for (uint32_t b = threadIdx.x; b < P; b += blockDim.x)
         H[(size_t)blockIdx.x * P + b] = lhist[b];
+    /* flush this block's hot groups into the global table */
+    __syncthreads();
+    for (uint32_t sl = threadIdx.x; sl < lds_slots; sl += blockDim.x) {
+        uint64_t* s = ltab + (uint64_t)sl * stride;
+        if (((uint32_t*)s)[0] != 2u) continue;
+        uint64_t* g = gtable_claim(gtable, gmask, stride, ((uint32_t*)s)[1],
+                                   s[1], s[2], fill, fill_cap, err);
+        if (!g) break;
+        agg_merge_slot<false>(g, s + SLOT_HDR, q);
+    }
 }
 
 /* bucket totals over all blocks, tiled (b, chunk-of-blocks) for parallelism;
@@ -672,7 +713,7 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
          r < row_end; r += gstride) {
         int64_t i = r - row_begin;
         uint32_t b = bucketid[i];
-        if (b == BK_SKIP_BUCKET) continue;
+        if (b >= BK_HOT_BUCKET) continue;  /* filtered out or absorbed hot */
         uint32_t pos = atomicAdd(&lcur[b], 1u);
         /* key words */
         uint64_t meta = 0;
@@ -1319,10 +1360,15 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
 
     static const char* NAMES[] = {"histo", "totals", "scan", "offsets",
                                   "scatter", "part_agg"};
+    uint32_t hot_slots = 512;
+    while ((size_t)hot_slots * stride * 8 > 50 * 1024) hot_slots >>= 1;
+    size_t histo_lds = ((size_t)hot_slots * stride + 2) * 8 + (size_t)P * 4;
     EvTimer tm;
     tm.record();
-    hipLaunchKernelGGL(k_part_histo, dim3(nblocks), dim3(threads), P * 4, 0,
-                       dc, *q, row_begin, row_end, P, bucketid, H);
+    hipLaunchKernelGGL(k_part_histo, dim3(nblocks), dim3(threads), histo_lds, 0,
+                       dc, *q, row_begin, row_end, P, bucketid, H,
+                       o->table, o->nslots - 1, (o->nslots * 7) / 8,
+                       o->ctrs, o->ctrs + 1, o->err, hot_slots);
     tm.record();
     hipLaunchKernelGGL(k_part_totals, dim3((P * nchunks + 255) / 256), dim3(256),
                        0, 0, H, nblocks, P, S, totals);
@@ -1359,8 +1405,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     tm.record();
     PCHECK(hipGetLastError());
     tm.finish(o, NAMES);
-    /* rows_passed = record count */
-    PCHECK(hipMemcpy(o->ctrs + 1, &total, 8, hipMemcpyHostToDevice));
+    /* rows_passed is accumulated by k_part_histo (hot + cold alike) */
     cleanup();
     #undef PCHECK
     return 0;

@@ -31,18 +31,22 @@ SEED = 20260915
 # GROUP BY (c2 skew-1e5, c7 dict-65536); aggs COUNT(*), SUM(c3:I64),
 # SUM(c4:DBL), AVG(c5:DBL).
 CONFIGS = {
+    # BASELINE config 2: "1e8-row 8xINT64, WHERE c1<K AND c2=K2 GROUP BY c3
+    # SUM(c4)" — c2 is a low-cardinality column (sysbench-style), K2 one of
+    # its values; c3 ~ Zipf over 1e5 distinct.
     "config2_1e8_8int64": dict(
         nrows=100_000_000,
-        specs=[(TYPE_INT64, D_UNI, 0, 1 << 31, 0)] * 2
-            + [(TYPE_INT64, 4, 100_000, 0, 0)]      # group key ~ Zipf/1e5 (BASELINE)
+        specs=[(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+               (TYPE_INT64, D_UNI, 0, 20, 0)]
+            + [(TYPE_INT64, 4, 100_000, 0, 0)]      # group key ~ Zipf/1e5
             + [(TYPE_INT64, D_UNI, 0, 1000, 0)]
             + [(TYPE_INT64, D_UNI, 0, 1 << 31, 0)] * 4,
-        conjuncts=[(0, "<", 1 << 30), (1, "<", int((1 << 31) * 0.9))],
+        conjuncts=[(0, "<", 1 << 30), (1, "=", 7)],
         group=[2], aggs=[("sum", 3)],
         expected_groups=1 << 18,
-        # algorithmic bytes/row: c0,c1 always (16); group c2 + agg c3 for the
-        # 0.5*0.9 surviving fraction (16 * 0.45)
-        bytes_per_row=16 + 16 * 0.45),
+        # algorithmic bytes/row: c0,c1 for every row (16 B); group c2 + agg
+        # c3 only for the 0.5 * 0.05 surviving fraction
+        bytes_per_row=16 + 16 * 0.5 * 0.05),
     "config3_1e9_mixed": dict(
         nrows=1_000_000_000,
         specs=[(TYPE_INT64, D_UNI, 0, 1 << 31, 0),      # c0 predicate

@@ -25,6 +25,12 @@ DTOL_REL = 1e-10
 
 @pytest.fixture(scope="module")
 def eng():
+    # initialize torch's HIP context before the engine's (torch lazy-init
+    # after libbkgpu has initialized HIP was observed to fail sporadically
+    # with "No HIP GPUs are available")
+    import torch
+    if torch.cuda.is_available():
+        torch.cuda.init()
     from baikaldb_amd import GpuEngine
     return GpuEngine()
 
