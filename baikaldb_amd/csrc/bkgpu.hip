@@ -302,9 +302,10 @@ __device__ uint64_t* gtable_claim(uint64_t* table, uint64_t mask, int stride,
 /* ---- LDS-table claim (workgroup scope; CU-local, cheap) ---- */
 __device__ uint64_t* ltable_claim(uint64_t* ltab, uint32_t lmask, int stride,
                                   uint32_t flag, uint64_t k0, uint64_t k1,
-                                  uint32_t* lfill, uint32_t lcap) {
+                                  uint32_t* lfill, uint32_t lcap,
+                                  int max_probe = 64) {
     uint64_t slot = (key_hash(flag, k0, k1) >> 32) & lmask;
-    for (int it = 0; it < 64; ++it) {
+    for (int it = 0; it < max_probe; ++it) {
         uint64_t* s = ltab + slot * (uint64_t)stride;
         uint32_t* statep = (uint32_t*)s;
         uint32_t st = __hip_atomic_load(statep, RLX, WGP);
@@ -581,7 +582,10 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
     for (uint32_t b = threadIdx.x; b < P; b += blockDim.x) lhist[b] = 0;
     __syncthreads();
     const uint32_t lmask = lds_slots - 1;
-    const uint32_t lcap = (lds_slots * 3u) / 4u;
+    /* low load cap + short probes: a MISS must exit fast (it pays on every
+     * cold row); hot keys the short probe skips just stay cold — correct,
+     * only less absorbed. */
+    const uint32_t lcap = lds_slots / 2u;
     int64_t my_passed = 0;
     int64_t gstride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -600,7 +604,7 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
             if (!cell_valid(c, r)) flag |= 0x40u; else k1 = enc_value(c, r);
         }
         uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
-                                      lfill, lcap);
+                                      lfill, lcap, /*max_probe=*/4);
         if (slot) {
             agg_update_slot<true>(slot, q, cols, r);
             bucketid[i] = (uint16_t)BK_HOT_BUCKET;
