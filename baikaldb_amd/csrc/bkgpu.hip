@@ -575,8 +575,12 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
     const int stride = SLOT_HDR + 2 * q.n_aggs;
     uint64_t* laux = ltab + (size_t)lds_slots * stride;
     uint32_t* lfill = (uint32_t*)&laux[0];
-    uint32_t* lhist = (uint32_t*)&laux[2];
-    for (uint32_t w = threadIdx.x; w < lds_slots * (uint32_t)stride + 2;
+    /* adaptive hot-path state: laux[2] = {attempts, hits}, laux[3].lo = mode
+     * (0 warmup / 1 disabled / 2 locked-on) */
+    uint32_t* lctr = (uint32_t*)&laux[2];
+    volatile uint32_t* lmode = (volatile uint32_t*)&laux[3];
+    uint32_t* lhist = (uint32_t*)&laux[4];
+    for (uint32_t w = threadIdx.x; w < lds_slots * (uint32_t)stride + 4;
          w += blockDim.x)
         ltab[w] = 0;
     for (uint32_t b = threadIdx.x; b < P; b += blockDim.x) lhist[b] = 0;
@@ -603,12 +607,24 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
             const DevCol& c = cols.c[q.group_cols[1]];
             if (!cell_valid(c, r)) flag |= 0x40u; else k1 = enc_value(c, r);
         }
-        uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
-                                      lfill, lcap, /*max_probe=*/4);
-        if (slot) {
-            agg_update_slot<true>(slot, q, cols, r);
-            bucketid[i] = (uint16_t)BK_HOT_BUCKET;
-            continue;
+        /* adaptive: pay the LDS-claim probe only while it absorbs >= 1/4
+         * of the stream (per-block warmup decides; Zipf-headed keys keep it
+         * on, flat/high-cardinality keys turn it off). */
+        uint32_t mode = lmode[0];
+        if (mode != 1u) {
+            uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
+                                          lfill, lcap, /*max_probe=*/4);
+            if (mode == 0u) {
+                uint32_t att = atomicAdd(&lctr[0], 1u);
+                if (slot) atomicAdd(&lctr[1], 1u);
+                if (att == 4095u)
+                    lmode[0] = (lctr[1] * 4u >= 4096u) ? 2u : 1u;
+            }
+            if (slot) {
+                agg_update_slot<true>(slot, q, cols, r);
+                bucketid[i] = (uint16_t)BK_HOT_BUCKET;
+                continue;
+            }
         }
         uint32_t b = PART_BUCKET(key_hash(flag, k0, k1), P);
         bucketid[i] = (uint16_t)b;
@@ -1333,10 +1349,10 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     RecLayout lay;
     build_rec_layout(t, q, &lay);
     uint32_t P = 64;
-    while ((int64_t)P < expected_groups / 96 && P < 4096) P <<= 1;
+    while ((int64_t)P < expected_groups / 512 && P < 4096) P <<= 1;
     const char* envP = getenv("BK_PART_P");
     if (envP) P = (uint32_t)atoi(envP);
-    int nblocks = 2048;
+    int nblocks = 1024;
     const char* envB = getenv("BK_PART_BLOCKS");
     if (envB) nblocks = atoi(envB);
     const int threads = 256;
@@ -1366,7 +1382,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
                                   "scatter", "part_agg"};
     uint32_t hot_slots = 512;
     while ((size_t)hot_slots * stride * 8 > 50 * 1024) hot_slots >>= 1;
-    size_t histo_lds = ((size_t)hot_slots * stride + 2) * 8 + (size_t)P * 4;
+    size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     EvTimer tm;
     tm.record();
     hipLaunchKernelGGL(k_part_histo, dim3(nblocks), dim3(threads), histo_lds, 0,
