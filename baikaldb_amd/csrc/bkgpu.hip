@@ -1164,6 +1164,11 @@ extern "C" BkgTable* bkgpu_table_create(int ncols, const BkColSpec* specs,
 
 extern "C" int64_t bkgpu_table_nrows(const BkgTable* t) { return t ? t->nrows : 0; }
 
+extern "C" int32_t bkgpu_table_col_type(const BkgTable* t, int col) {
+    if (!t || col < 0 || col >= t->ncols) return BK_INVALID_TYPE;
+    return t->specs[col].col_type;
+}
+
 extern "C" void bkgpu_table_free(BkgTable* t) {
     if (!t) return;
     for (int c = 0; c < t->ncols; c++) {
@@ -1692,12 +1697,86 @@ extern "C" int64_t bkgpu_agg_fetch(BkgAggOut* o, int sorted, int64_t max_groups,
 }
 
 /* ------------------------------------------------------------------ */
-/* top-N (placeholder — implemented in bkgpu_sort.hip part 2)          */
+/* filtered row-id collection + column gather (SELECT without GROUP BY:
+ * FilterNode::get_next row emission, filter_node.cpp:736-795)         */
 /* ------------------------------------------------------------------ */
 
-extern "C" int64_t bkgpu_sort_topk(BkgTable*, const BkQuerySpec*, const BkOrderSpec*,
-                                   int, int64_t, int64_t, int64_t, int64_t*) {
-    set_err("bkgpu_sort_topk: not implemented yet");
-    return -1;
+__global__ void __launch_bounds__(256)
+k_filter_collect(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
+                 int64_t limit, int64_t* out_rowids, uint64_t* counter) {
+    int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < row_end; r += gstride) {
+        if (!row_passes(cols, q, r)) continue;
+        uint64_t i = atomicAdd((unsigned long long*)counter, 1ull);
+        if ((int64_t)i < limit) out_rowids[i] = r;
+    }
 }
-extern "C" double bkgpu_topk_kernel_ms(void) { return 0.0; }
+
+/* Collect up to `limit` passing row ids (ascending order NOT guaranteed —
+ * the reference's row order within a scan is the iterator's; callers that
+ * need arrival order sort the ids). Returns count written (<0 error). */
+extern "C" int64_t bkgpu_filter_collect(BkgTable* t, const BkQuerySpec* q,
+                                        int64_t row_begin, int64_t row_end,
+                                        int64_t limit, int64_t* out_rowids_host) {
+    if (ensure_device() != 0) return -1;
+    if (limit <= 0) return 0;
+    int64_t* d_ids = nullptr;
+    uint64_t* d_ctr = nullptr;
+    HIP_CHECK(pool_alloc((void**)&d_ids, (size_t)limit * 8));
+    HIP_CHECK(pool_alloc((void**)&d_ctr, 8));
+    HIP_CHECK(hipMemset(d_ctr, 0, 8));
+    DevCols dc = table_cols(t);
+    hipLaunchKernelGGL(k_filter_collect, dim3(1024), dim3(256), 0, 0,
+                       dc, *q, row_begin, row_end, limit, d_ids, d_ctr);
+    HIP_CHECK(hipGetLastError());
+    uint64_t n = 0;
+    HIP_CHECK(hipMemcpy(&n, d_ctr, 8, hipMemcpyDeviceToHost));
+    int64_t got = std::min<int64_t>((int64_t)n, limit);
+    HIP_CHECK(hipMemcpy(out_rowids_host, d_ids, (size_t)got * 8,
+                        hipMemcpyDeviceToHost));
+    pool_free(d_ids);
+    pool_free(d_ctr);
+    return got;
+}
+
+__global__ void k_gather(DevCol c, const int64_t* rowids, int64_t n,
+                         int64_t* out_i, double* out_d, uint8_t* out_null) {
+    int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gstride) {
+        int64_t r = rowids[i];
+        uint8_t valid = cell_valid(c, r) ? 1 : 0;
+        out_null[i] = !valid;
+        if (!valid) { out_i[i] = 0; out_d[i] = 0.0; continue; }
+        if (c.type == BK_DOUBLE) { out_d[i] = ((const double*)c.data)[r]; out_i[i] = 0; }
+        else { out_i[i] = cell_i64(c, r); out_d[i] = 0.0; }
+    }
+}
+
+/* Materialize column `col` for the given global row ids (host arrays). */
+extern "C" int bkgpu_gather(BkgTable* t, int col, const int64_t* rowids_host,
+                            int64_t n, int64_t* out_i, double* out_d,
+                            uint8_t* out_null) {
+    if (n <= 0) return 0;
+    int64_t* d_ids = nullptr;
+    int64_t* d_i = nullptr;
+    double* d_d = nullptr;
+    uint8_t* d_n = nullptr;
+    HIP_CHECK(pool_alloc((void**)&d_ids, (size_t)n * 8));
+    HIP_CHECK(pool_alloc((void**)&d_i, (size_t)n * 8));
+    HIP_CHECK(pool_alloc((void**)&d_d, (size_t)n * 8));
+    HIP_CHECK(pool_alloc((void**)&d_n, (size_t)n));
+    HIP_CHECK(hipMemcpy(d_ids, rowids_host, (size_t)n * 8, hipMemcpyHostToDevice));
+    DevCols dc = table_cols(t);
+    hipLaunchKernelGGL(k_gather, dim3(512), dim3(256), 0, 0,
+                       dc.c[col], d_ids, n, d_i, d_d, d_n);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipMemcpy(out_i, d_i, (size_t)n * 8, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(out_d, d_d, (size_t)n * 8, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(out_null, d_n, (size_t)n, hipMemcpyDeviceToHost));
+    pool_free(d_ids); pool_free(d_i); pool_free(d_d); pool_free(d_n);
+    return 0;
+}
+
+#include "bksort.inc"

@@ -70,6 +70,20 @@ CONFIGS = {
         # fraction (4 B); group c2 + aggs c3,c4,c5 for the 0.45*(65535/65536)
         # survivors (32 B)
         bytes_per_row=16 + 4 * 0.45 + 32 * 0.45 * (65535 / 65536)),
+    # BASELINE config 5: 1e9-row ORDER BY c0,c1 LIMIT 1e6 (GPU radix top-N
+    # selection; with N>1 each rank owns a region set and rank 0 merges the
+    # gathered per-rank top-Ks — SelectManagerNode's heap-merge role,
+    # select_manager_node.cpp:304-344).
+    "config5_1e9_sort": dict(
+        nrows=1_000_000_000,
+        specs=[(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+               (TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+               (TYPE_INT64, D_UNI, 0, 1 << 31, 0)],
+        order=[(0, 1, 1), (1, 1, 1)],
+        limit=1_000_000,
+        conjuncts=[], group=[], aggs=[],
+        expected_groups=0,
+        bytes_per_row=16.0),  # the two 8-B key columns, read once
 }
 
 
@@ -101,8 +115,15 @@ def cpu_baseline_leg(cfg, sample_rows):
                    col_types)
     cores = min(os.cpu_count() or 1, 64)  # oracle caps at 64 worker threads
     t0 = time.perf_counter()
-    res = orc.filter_agg(cols, valids, col_types, q, nthreads=cores,
-                         dict_seed=SEED, sort_keys=False)
+    if "order" in cfg:
+        # top-N selection (single-threaded heap walk — the reference's
+        # TopNSorter is per-region single-threaded too, topn_sorter.h:32)
+        cores = 1
+        lim = min(cfg["limit"], sample_rows)
+        orc.sort_topk(cols, valids, col_types, cfg["order"], lim, q=q)
+    else:
+        orc.filter_agg(cols, valids, col_types, q, nthreads=cores,
+                       dict_seed=SEED, sort_keys=False)
     dt = time.perf_counter() - t0
     return {"value": sample_rows / dt, "unit": "rows/s", "cores": cores,
             "kind": "port",
@@ -148,7 +169,45 @@ def main():
     plan = QueryPlan(t.col_types, conjuncts=cfg["conjuncts"], group=cfg["group"],
                      aggs=cfg["aggs"])
 
+    is_sort = "order" in cfg
+
+    def one_sort_step(timed_kernels):
+        rowids = eng.sort_topk(t, cfg["order"], cfg["limit"], plan=plan)
+        if timed_kernels is not None:
+            timed_kernels.append(eng.topk_kernel_ms())
+        if dist is not None:
+            # gather per-rank top-Ks; rank 0 merges (stable LSD argsorts)
+            import numpy as np
+            k0 = np.zeros(len(rowids), dtype=np.int64)
+            k1 = np.zeros(len(rowids), dtype=np.int64)
+            dscr = np.zeros(len(rowids), dtype=np.float64)
+            nul = np.zeros(len(rowids), dtype=np.uint8)
+            import ctypes as Ct
+            for col, dst in ((cfg["order"][0][0], k0), (cfg["order"][1][0], k1)):
+                eng.lib.bkgpu_gather(t.handle, col,
+                                     rowids.ctypes.data_as(Ct.POINTER(Ct.c_int64)),
+                                     len(rowids),
+                                     dst.ctypes.data_as(Ct.POINTER(Ct.c_int64)),
+                                     dscr.ctypes.data_as(Ct.POINTER(Ct.c_double)),
+                                     nul.ctypes.data_as(Ct.POINTER(Ct.c_uint8)))
+            pack = torch.stack([torch.from_numpy(k0), torch.from_numpy(k1),
+                                torch.from_numpy(rowids)]).cuda()
+            buf = torch.zeros(3, cfg["limit"], dtype=torch.int64, device="cuda")
+            buf[:, :pack.shape[1]] = pack
+            gath = [torch.zeros_like(buf) for _ in range(world)]
+            dist.all_gather(gath, buf)
+            if rank == 0:
+                allk = torch.cat(gath, dim=1)
+                idx = torch.argsort(allk[2], stable=True)
+                for row in (1, 0):
+                    idx = idx[torch.argsort(allk[row][idx], stable=True)]
+                _final = allk[:, idx[:cfg["limit"]]]
+            torch.cuda.synchronize()
+        return len(rowids), 0
+
     def one_step(timed_kernels):
+        if is_sort:
+            return one_sort_step(timed_kernels)
         res = eng.filter_agg(t, plan, expected_groups=cfg["expected_groups"])
         if timed_kernels is not None:
             timed_kernels.append(res.kernel_ms)

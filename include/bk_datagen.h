@@ -105,4 +105,13 @@ BK_HD double bk_gen_f64(const BkColSpec* cs, uint64_t seed, uint64_t row, uint32
     return bk_gen_sumu16(u);
 }
 
+/* Deterministic dict word for a code (host-side; rows store codes, words are
+ * materialized only when emitting VARCHAR values). Unique per code by
+ * construction (the code is embedded in the word). */
+#include <stdio.h>
+static inline int bk_dict_word(uint64_t seed, int64_t code, char* out, int cap) {
+    uint32_t h = (uint32_t)bk_mix64(seed ^ 0xD1C7ull ^ (uint64_t)code);
+    return snprintf(out, (size_t)cap, "w%06lx_%08x", (unsigned long)code, h);
+}
+
 #endif /* BK_DATAGEN_H */

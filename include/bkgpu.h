@@ -58,6 +58,7 @@ int  bkgpu_table_generate(BkgTable* t, uint64_t seed, int64_t row_begin);
 /* Upload host column data (and optional validity bytes) instead. */
 int  bkgpu_table_upload(BkgTable* t, int col, const void* data, const uint8_t* valid);
 int64_t bkgpu_table_nrows(const BkgTable* t);
+int32_t bkgpu_table_col_type(const BkgTable* t, int col);
 void bkgpu_table_free(BkgTable* t);
 
 /* ---- fused scan+filter+aggregate ---- */
@@ -110,6 +111,17 @@ int64_t bkgpu_sort_topk(BkgTable* t, const BkQuerySpec* q,
                         int64_t row_begin, int64_t row_end,
                         int64_t limit, int64_t* out_rows);
 double bkgpu_topk_kernel_ms(void);
+
+/* ---- SELECT without GROUP BY (FilterNode row emission, filter_node.cpp:
+ * 736-795): collect up to limit passing global row ids (order unspecified);
+ * materialize columns for row ids. Host arrays. ---- */
+int64_t bkgpu_filter_collect(BkgTable* t, const BkQuerySpec* q,
+                             int64_t row_begin, int64_t row_end,
+                             int64_t limit, int64_t* out_rowids_host);
+int bkgpu_gather(BkgTable* t, int col, const int64_t* rowids_host, int64_t n,
+                 int64_t* out_i, double* out_d, uint8_t* out_null);
+/* release pooled device buffers */
+void bkgpu_pool_trim(void);
 
 #ifdef __cplusplus
 }

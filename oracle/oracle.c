@@ -95,11 +95,9 @@ ORC_EXPORT int orc_generate_column(const BkColSpec* cs, uint64_t seed, uint32_t 
     return 0;
 }
 
-/* Deterministic dict word for a code (host-side only; row data stores codes).
- * Unique per code by construction (the code is embedded in the word). */
+/* Deterministic dict word for a code (shared generator, bk_datagen.h). */
 ORC_EXPORT int orc_dict_word(uint64_t seed, int64_t code, char* out, int cap) {
-    uint32_t h = (uint32_t)bk_mix64(seed ^ 0xD1C7ull ^ (uint64_t)code);
-    return snprintf(out, (size_t)cap, "w%06lx_%08x", (unsigned long)code, h);
+    return bk_dict_word(seed, code, out, cap);
 }
 
 /* ------------------------------------------------------------------ */

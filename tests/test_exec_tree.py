@@ -1,0 +1,131 @@
+# GPU tests of the C++ ExecNode plugin-surface mirror (bk_exec.h): the
+# SELECT pipeline driven through create_tree + open/get_next/close must
+# produce the same rows as the CPU oracle on the same seeded inputs.
+import numpy as np
+import pytest
+
+from oracle import BkColSpec
+from oracle.bindings import make_query
+
+pytestmark = pytest.mark.gpu
+
+SEED = 0xE4EC
+TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
+
+
+@pytest.fixture(scope="module")
+def eng():
+    import torch
+    if torch.cuda.is_available():
+        torch.cuda.init()
+    from baikaldb_amd import GpuEngine
+    return GpuEngine()
+
+
+@pytest.fixture(scope="module")
+def orc():
+    from oracle import Oracle
+    return Oracle()
+
+
+def make_table(eng, orc, n=120_000):
+    spec_rows = [(TYPE_INT64, 0, 0, 1 << 31, 0),
+                 (TYPE_INT64, 1, 2000, 0, 0),
+                 (TYPE_INT64, 0, 0, 500, 0),
+                 (TYPE_DOUBLE, 3, 0, 0, 0)]
+    t = eng.create_table(spec_rows, n)
+    eng.generate(t, SEED)
+    specs = (BkColSpec * len(spec_rows))()
+    for i, s in enumerate(spec_rows):
+        (specs[i].col_type, specs[i].dist, specs[i].p0, specs[i].p1,
+         specs[i].null_frac_x1e6) = s
+    cols, valids = orc.generate_table(list(specs), n, SEED)
+    types = [s[0] for s in spec_rows]
+    return t, cols, valids, types
+
+
+def test_agg_pipeline_through_exec_surface(eng, orc):
+    from baikaldb_amd import exec as bx
+    t, cols, valids, types = make_table(eng, orc)
+    try:
+        # AGG -> FILTER -> SCAN  (the store-side SELECT pipeline shape)
+        nodes = [bx.agg_node(group=[1], aggs=[("count_star", -1), ("sum", 2),
+                                              ("avg", 3)]),
+                 bx.filter_node(types, [(0, "<", int((1 << 31) * 0.6))]),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        nscan, nfilt = tree.num_scan_rows, tree.num_filter_rows
+        tree.close()
+    finally:
+        t.free()
+
+    q = make_query([(0, 4, TYPE_INT64, int((1 << 31) * 0.6))], [1],
+                   [(0, -1), (2, 2), (3, 3)], types)
+    exp = orc.filter_agg(cols, valids, types, q, nthreads=4, dict_seed=SEED)
+
+    assert nscan == len(cols[0])
+    assert nscan - nfilt == exp["rows_passed"]
+    assert tags.shape[0] == exp["ngroups"]
+    # slot 0 = group key (sorted canonical order matches oracle's)
+    keys = np.array([orc.lib.orc_decode_i64(int(e)) for e in exp["enc"][:, 0]])
+    assert np.array_equal(vi[:, 0], keys)
+    assert np.array_equal(vi[:, 1], exp["agg_i"][0])   # COUNT(*)
+    assert np.array_equal(vi[:, 2], exp["agg_i"][1])   # SUM int64
+    np.testing.assert_allclose(vd[:, 3], exp["agg_d"][2], rtol=1e-12)  # AVG
+    assert tags[0, 1] == TYPE_INT64 and tags[0, 3] == TYPE_DOUBLE
+
+
+def test_limit_over_agg(eng, orc):
+    from baikaldb_amd import exec as bx
+    t, cols, valids, types = make_table(eng, orc, n=50_000)
+    try:
+        nodes = [bx.limit_node(7),
+                 bx.agg_node(group=[1], aggs=[("count_star", -1)]),
+                 bx.filter_node(types, []),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all(batch=3)
+        tree.close()
+    finally:
+        t.free()
+    assert tags.shape[0] == 7
+
+
+def test_sort_topn_through_exec_surface(eng, orc):
+    from baikaldb_amd import exec as bx
+    t, cols, valids, types = make_table(eng, orc, n=80_000)
+    try:
+        nodes = [bx.sort_node(order=[(2, 1, 1), (0, 1, 1)], out_cols=[2, 0, 3],
+                              limit=1000),
+                 bx.filter_node(types, [(0, ">", 1000)]),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
+
+    q = make_query([(0, 2, TYPE_INT64, 1000)], [], [], types)
+    expect_rows = orc.sort_topk(cols, valids, types,
+                                [(2, 1, 1), (0, 1, 1)], limit=1000, q=q)
+    assert tags.shape[0] == 1000
+    assert np.array_equal(vi[:, 0], cols[2][expect_rows])
+    assert np.array_equal(vi[:, 1], cols[0][expect_rows])
+    assert np.array_equal(vd[:, 2].view(np.float64), cols[3][expect_rows])
+
+
+def test_scan_root_refuses_row_mode(eng, orc):
+    from baikaldb_amd import exec as bx
+    t, *_ = make_table(eng, orc, n=1000)
+    try:
+        tree = bx.ExecTree([bx.scan_node(t)])
+        tree.open()
+        with pytest.raises(RuntimeError):
+            tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
