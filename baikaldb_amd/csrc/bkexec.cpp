@@ -66,6 +66,7 @@ public:
     void reset() { _idx = 0; }
     bool is_traverse_over() const { return _idx >= _rows.size(); }
     void clear() { _rows.clear(); _idx = 0; }
+    void truncate(size_t n) { if (n < _rows.size()) _rows.resize(n); }
 private:
     std::vector<std::unique_ptr<MemRow>> _rows;
     size_t _idx = 0;
@@ -429,6 +430,11 @@ public:
         int ret = _children[0]->get_next(state, batch, eos);
         if (ret < 0) return ret;
         _num_rows_returned += (int64_t)(batch->size() - before);
+        if (_limit > 0 && _num_rows_returned > _limit) {
+            /* truncate the overshoot so exactly _limit rows are emitted */
+            batch->truncate(batch->size() - (size_t)(_num_rows_returned - _limit));
+            _num_rows_returned = _limit;
+        }
         if (reached_limit()) *eos = true;
         return 0;
     }

@@ -69,6 +69,16 @@ def _load():
                                     C.POINTER(C.c_int64)]
     lib.bkgpu_topk_kernel_ms.restype = C.c_double
     lib.bkgpu_sync.restype = C.c_int
+    lib.bkgpu_filter_collect.restype = C.c_int64
+    lib.bkgpu_filter_collect.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
+                                         C.c_int64, C.c_int64, C.c_int64,
+                                         C.POINTER(C.c_int64)]
+    lib.bkgpu_gather.restype = C.c_int
+    lib.bkgpu_gather.argtypes = [C.c_void_p, C.c_int, C.POINTER(C.c_int64),
+                                 C.c_int64, C.POINTER(C.c_int64),
+                                 C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
+    lib.bkgpu_table_col_type.restype = C.c_int32
+    lib.bkgpu_table_col_type.argtypes = [C.c_void_p, C.c_int]
     return lib
 
 

@@ -302,8 +302,10 @@ def main():
         "data": "synthetic",
         "config": {"workload": args.config, "rows_per_gpu": nrows,
                    "groups": ng, "rows_passed_rank0": rp,
-                   "predicates": 3 if args.config.startswith("config3") else 2,
-                   "group_keys": len(cfg["group"]), "aggs": len(cfg["aggs"])},
+                   "predicates": len(cfg["conjuncts"]),
+                   "group_keys": len(cfg["group"]), "aggs": len(cfg["aggs"]),
+                   **({"order_keys": len(cfg["order"]), "limit": cfg["limit"]}
+                      if "order" in cfg else {})},
         "roofline": roofline,
         "cpu_baseline": cpu_baseline,
     }
