@@ -1416,10 +1416,16 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     }
     tm.record();
     if (total > 0) {
-        uint32_t lds_slots = 1024;
-        while ((size_t)lds_slots * stride * 8 > 100 * 1024) lds_slots >>= 1;
+        uint32_t lds_slots = 2048;
+        while ((size_t)lds_slots * stride * 8 > 135 * 1024) lds_slots >>= 1;
         size_t lds_bytes = ((size_t)lds_slots * stride + 1) * 8;
-        uint64_t chunk = 32768;
+        /* chunk ~ bucket size: fewer generation flushes (flush traffic ~
+         * chunks x distinct-per-chunk), while >=2048 chunks keep the chip
+         * balanced */
+        uint64_t chunk = std::max<uint64_t>(
+            32768, std::min<uint64_t>(1u << 20, total / 2048));
+        const char* envC = getenv("BK_AGG_CHUNK");
+        if (envC) chunk = (uint64_t)atoll(envC);
         uint64_t nchunks = (total + chunk - 1) / chunk;
         uint32_t grid = (uint32_t)std::min<uint64_t>(nchunks, 32768);
         hipLaunchKernelGGL(k_part_agg, dim3(grid), dim3(threads), lds_bytes, 0,

@@ -28,7 +28,7 @@ plan = QueryPlan(t.col_types,
                  group=[2, 6],
                  aggs=[("count_star", -1), ("sum", 3), ("sum", 4), ("avg", 5)])
 
-for P, blocks in itertools.product([512, 1024, 2048, 4096], [512, 1024, 2048, 4096]):
+for P, blocks in itertools.product([128, 256, 512, 1024], [1024, 2048]):
     os.environ["BK_PART_P"] = str(P)
     os.environ["BK_PART_BLOCKS"] = str(blocks)
     best = None
