@@ -1354,7 +1354,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     RecLayout lay;
     build_rec_layout(t, q, &lay);
     uint32_t P = 64;
-    while ((int64_t)P < expected_groups / 512 && P < 4096) P <<= 1;
+    while ((int64_t)P < expected_groups / 1024 && P < 4096) P <<= 1;
     const char* envP = getenv("BK_PART_P");
     if (envP) P = (uint32_t)atoi(envP);
     int nblocks = 1024;
