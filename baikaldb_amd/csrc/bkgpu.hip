@@ -746,8 +746,9 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
             const DevCol& c = cols.c[q.group_cols[1]];
             if (!cell_valid(c, r)) meta |= 0x40u; else k1 = enc_value(c, r);
         }
-        rec[pos] = k0;
-        if (lay.k1_word >= 0) rec[(size_t)lay.k1_word * total + pos] = k1;
+        uint64_t* my = rec + (size_t)pos * lay.nwords;   /* AoS record */
+        my[0] = k0;
+        if (lay.k1_word >= 0) my[lay.k1_word] = k1;
         for (int32_t a = 0; a < q.n_aggs; a++) {
             if (lay.val_word[a] < 0) continue;
             const BkAggSpec& as = q.aggs[a];
@@ -777,7 +778,7 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
                     default: break;
                 }
             }
-            rec[(size_t)lay.val_word[a] * total + pos] = w;
+            my[lay.val_word[a]] = w;
         }
         if (lay.meta_word >= 0) {
             /* COUNT(col) validity for aggs without a val word */
@@ -786,7 +787,7 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
                 if (cell_valid(cols.c[q.aggs[a].col], r))
                     meta |= (uint64_t)1 << (8 + a);
             }
-            rec[(size_t)lay.meta_word * total + pos] = meta;
+            my[lay.meta_word] = meta;
         }
     }
 }
@@ -795,8 +796,8 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
 template <bool LDS>
 __device__ __forceinline__ void agg_update_slot_rec(uint64_t* st, const BkQuerySpec& q,
                                                     const RecLayout& lay,
-                                                    const uint64_t* rec, uint64_t total,
-                                                    uint64_t pos, uint64_t meta) {
+                                                    const uint64_t* my,
+                                                    uint64_t meta) {
     #pragma unroll 4
     for (int32_t a = 0; a < q.n_aggs; a++) {
         uint64_t* val = st + SLOT_HDR + 2 * a;
@@ -814,7 +815,7 @@ __device__ __forceinline__ void agg_update_slot_rec(uint64_t* st, const BkQueryS
             atomicAdd((unsigned long long*)val, 1ull);
             continue;
         }
-        uint64_t w = rec[(size_t)lay.val_word[a] * total + pos];
+        uint64_t w = my[lay.val_word[a]];
         switch (at) {
             case BK_AGG_SUM:
                 if (q.agg_in_types[a] == BK_DOUBLE) {
@@ -853,8 +854,7 @@ __device__ __forceinline__ void agg_update_slot_rec(uint64_t* st, const BkQueryS
  * 64 serialized same-address LDS atomics. Returns true if handled. */
 __device__ __forceinline__ bool wave_combine_update(
         uint64_t* slot, const BkQuerySpec& q, const RecLayout& lay,
-        const uint64_t* rec, uint64_t total, uint64_t pos, uint64_t meta,
-        bool lds) {
+        const uint64_t* my, uint64_t meta, bool lds) {
     int lane = threadIdx.x & 63;
     #pragma unroll 4
     for (int32_t a = 0; a < q.n_aggs; a++) {
@@ -864,8 +864,7 @@ __device__ __forceinline__ bool wave_combine_update(
         uint64_t addv = 0;   /* combined value */
         double addd = 0.0;
         uint64_t addc = 0;   /* combined count */
-        uint64_t w = (valid && lay.val_word[a] >= 0)
-                         ? rec[(size_t)lay.val_word[a] * total + pos] : 0;
+        uint64_t w = (valid && lay.val_word[a] >= 0) ? my[lay.val_word[a]] : 0;
         switch (at) {
             case BK_AGG_COUNT_STAR: addv = 1; break;
             case BK_AGG_COUNT:      addv = valid ? 1 : 0; break;
@@ -950,12 +949,13 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
             uint32_t i = t0 + threadIdx.x;
             bool pending = i < n;
             uint64_t pos = (uint64_t)b0 + i;
+            const uint64_t* my = rec + (size_t)pos * lay.nwords;
             uint64_t k0 = 0, k1 = 0, meta = 0;
             uint32_t flag = 0;
             if (pending) {
-                k0 = rec[pos];
-                k1 = lay.k1_word >= 0 ? rec[(size_t)lay.k1_word * total + pos] : 0;
-                meta = lay.meta_word >= 0 ? rec[(size_t)lay.meta_word * total + pos] : 0;
+                k0 = my[0];
+                k1 = lay.k1_word >= 0 ? my[lay.k1_word] : 0;
+                meta = lay.meta_word >= 0 ? my[lay.meta_word] : 0;
                 flag = (uint32_t)(meta & 0xFF);
             }
             for (;;) {
@@ -972,8 +972,7 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
                                                 lfill, lcap);
                         int ok = __shfl((int)(slot != nullptr), 0, 64);
                         if (ok) {
-                            wave_combine_update(slot, q, lay, rec, total, pos,
-                                                meta, true);
+                            wave_combine_update(slot, q, lay, my, meta, true);
                             pending = false;
                         }
                         goto vote;  /* skip per-lane path this round */
@@ -983,7 +982,7 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
                     uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
                                                   lfill, lcap);
                     if (slot) {
-                        agg_update_slot_rec<true>(slot, q, lay, rec, total, pos, meta);
+                        agg_update_slot_rec<true>(slot, q, lay, my, meta);
                         pending = false;
                     }
                 }
