@@ -1445,10 +1445,24 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
  * and the single fused kernel wins; above it, partition. */
 #define FUSED_MAX_GROUPS 512
 
+#include <sys/time.h>
+static double now_ms() {
+    struct timeval tv;
+    gettimeofday(&tv, nullptr);
+    return tv.tv_sec * 1e3 + tv.tv_usec * 1e-3;
+}
+static int g_debug_timing = -1;
+static bool debug_timing() {
+    if (g_debug_timing < 0) g_debug_timing = getenv("BK_DEBUG") ? 1 : 0;
+    return g_debug_timing == 1;
+}
+
 extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
                                        int64_t row_begin, int64_t row_end,
                                        int64_t expected_groups) {
     if (ensure_device() != 0) return nullptr;
+    double t_start = debug_timing() ? now_ms() : 0;
+    double t_alloc = 0, t_pipe = 0;
     BkgAggOut* o = new BkgAggOut();
     o->q = *q;
     const int stride = SLOT_HDR + 2 * q->n_aggs;
@@ -1459,6 +1473,7 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
 
     for (int attempt = 0; attempt < 8; attempt++) {
         if (agg_alloc(o, nslots) != 0) { bkgpu_agg_free(o); return nullptr; }
+        if (debug_timing()) { (void)hipDeviceSynchronize(); t_alloc = now_ms(); }
         int blocks = 2048, threads = 256;
         if (partitioned) {
             if (run_partitioned(o, t, q, row_begin, row_end, expected_groups) != 0) {
@@ -1514,7 +1529,13 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
         }
     }
     o->dirty = true;
+    if (debug_timing()) { (void)hipDeviceSynchronize(); t_pipe = now_ms(); }
     if (agg_compact(o) != 0) { bkgpu_agg_free(o); return nullptr; }
+    if (debug_timing()) {
+        fprintf(stderr, "[bkgpu] alloc %.1f ms, pipeline %.1f ms (events %.1f), "
+                "compact %.1f ms\n", t_alloc - t_start, t_pipe - t_alloc,
+                o->kernel_ms, now_ms() - t_pipe);
+    }
     return o;
 }
 

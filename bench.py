@@ -246,10 +246,15 @@ def main():
         dist.barrier()
     torch.cuda.synchronize() if torch.cuda.is_available() else None
 
+    dbg = os.environ.get("BK_BENCH_DEBUG")
     kms = []
     t0 = time.perf_counter()
     for _ in range(args.steps):
+        s0 = time.perf_counter()
         rp, ng = one_step(kms)
+        if dbg:
+            log(f"[dbg] step wall {(time.perf_counter()-s0)*1e3:.1f} ms, "
+                f"kernels {kms[-1] if kms else 0:.1f} ms")
     eng.sync()
     if torch.cuda.is_available():
         torch.cuda.synchronize()
