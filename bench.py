@@ -153,6 +153,14 @@ def main():
         dist = tdist
         torch.cuda.set_device(local_rank)
 
+    # pin clocks: the box's DVFS governor drops to ~100 MHz between steps and
+    # ramps slowly, making otherwise-identical steps bimodal (measured in
+    # profiles/; junction temps ~46C, nowhere near thermal limits)
+    if local_rank == 0:
+        import subprocess
+        subprocess.run(["rocm-smi", "--setperflevel", "high"],
+                       capture_output=True, check=False)
+
     from baikaldb_amd import GpuEngine, QueryPlan
     eng = GpuEngine(device=local_rank)
 
