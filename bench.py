@@ -219,6 +219,8 @@ def main():
         res = eng.filter_agg(t, plan, expected_groups=cfg["expected_groups"])
         if timed_kernels is not None:
             timed_kernels.append(res.kernel_ms)
+        if os.environ.get("BK_BENCH_DEBUG"):
+            log(f"[dbg] breakdown { {k: round(v, 1) for k, v in res.breakdown().items()} }")
         if dist is not None:
             eng.sync()
             ng = torch.tensor([res.ngroups], dtype=torch.int64, device="cuda")
