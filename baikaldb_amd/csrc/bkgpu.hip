@@ -1076,6 +1076,11 @@ static std::multimap<size_t, void*> g_pool_free;
 static std::map<void*, size_t> g_pool_sizes;
 
 static hipError_t pool_alloc(void** p, size_t bytes) {
+    /* round to a size class so runs whose buffer sizes wobble slightly
+     * (e.g. the adaptive hot path makes cold-record counts timing-dependent)
+     * still hit the cache: 4 KiB classes below 64 MiB, 64 MiB classes above */
+    size_t cls = bytes >= (64u << 20) ? (64u << 20) : 4096;
+    bytes = (bytes + cls - 1) & ~(cls - 1);
     auto it = g_pool_free.lower_bound(bytes);
     /* reuse only if within 2x of the request (avoid hoarding) */
     if (it != g_pool_free.end() && it->first <= bytes * 2) {
@@ -1083,6 +1088,9 @@ static hipError_t pool_alloc(void** p, size_t bytes) {
         g_pool_free.erase(it);
         return hipSuccess;
     }
+    if (getenv("BK_DEBUG"))
+        fprintf(stderr, "[bkgpu] pool MISS %zu bytes (free chunks: %zu)\n",
+                bytes, g_pool_free.size());
     hipError_t e = hipMalloc(p, bytes);
     if (e == hipSuccess) g_pool_sizes[*p] = bytes;
     return e;
