@@ -95,6 +95,8 @@ def log(msg):
 def cpu_baseline_leg(cfg, sample_rows):
     """Time the oracle (the reference row engine restated; kind='port') on the
     host cores, on a bounded sample of the same workload."""
+    from concurrent.futures import ThreadPoolExecutor
+    import numpy as np
     from oracle import Oracle, BkColSpec
     from oracle.bindings import make_query
     orc = Oracle()
@@ -103,7 +105,19 @@ def cpu_baseline_leg(cfg, sample_rows):
         (specs[i].col_type, specs[i].dist, specs[i].p0, specs[i].p1,
          specs[i].null_frac_x1e6) = s
     col_types = [s[0] for s in cfg["specs"]]
-    cols, valids = orc.generate_table(list(specs), sample_rows, SEED)
+    # generate columns in parallel (untimed setup; ctypes releases the GIL)
+    import ctypes as Ct
+    cols = []
+    for ci, sp in enumerate(specs):
+        dt = {6: np.int64, 12: np.float64, 13: np.int32}[sp.col_type]
+        cols.append(np.empty(sample_rows, dtype=dt))
+    def gen(ci):
+        orc.lib.orc_generate_column(Ct.byref(specs[ci]), SEED, ci, 0,
+                                    sample_rows,
+                                    cols[ci].ctypes.data_as(Ct.c_void_p), None)
+    with ThreadPoolExecutor(max_workers=len(cols)) as ex:
+        list(ex.map(gen, range(len(cols))))
+    valids = [None] * len(cols)
     ops = {"=": 0, "!=": 1, ">": 2, ">=": 3, "<": 4, "<=": 5}
     aggmap = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5}
     conj = []
@@ -113,7 +127,7 @@ def cpu_baseline_leg(cfg, sample_rows):
         conj.append((col, ops[op], ct, lit))
     q = make_query(conj, cfg["group"], [(aggmap[a], c) for a, c in cfg["aggs"]],
                    col_types)
-    cores = min(os.cpu_count() or 1, 64)  # oracle caps at 64 worker threads
+    cores = min(os.cpu_count() or 1, 128)  # oracle caps at 128 worker threads
     t0 = time.perf_counter()
     if "order" in cfg:
         # top-N selection (single-threaded heap walk — the reference's
@@ -147,7 +161,9 @@ def main():
 
     import torch
     dist = None
-    if world > 1:
+    if world > 1 or (world == 1 and os.environ.get("BK_FORCE_DIST")):
+        # BK_FORCE_DIST exercises the full RCCL exchange path on a single
+        # rank (validation on 1-GPU boxes; the driver launches real N>1)
         import torch.distributed as tdist
         tdist.init_process_group(backend="nccl")
         dist = tdist
@@ -298,7 +314,7 @@ def main():
 
     cpu_baseline = None
     if n_gpus == 1 and not args.no_cpu_baseline:
-        sample = min(nrows, 20_000_000)
+        sample = min(nrows, 50_000_000)
         log(f"[bench] cpu baseline leg on {sample} rows")
         cpu_baseline = cpu_baseline_leg(cfg, sample)
 

@@ -488,7 +488,7 @@ ORC_EXPORT OrcAggResult* orc_filter_agg(const OrcCol* cols, int ncols,
     (void)ncols;
     if (nthreads < 1) nthreads = 1;
     int64_t n = row_end - row_begin;
-    if (nthreads > 64) nthreads = 64;
+    if (nthreads > 128) nthreads = 128;
     if ((int64_t)nthreads > n && n > 0) nthreads = (int)n;
     if (n <= 0) nthreads = 1;
 
