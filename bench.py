@@ -314,7 +314,7 @@ def main():
 
     cpu_baseline = None
     if n_gpus == 1 and not args.no_cpu_baseline:
-        sample = min(nrows, 50_000_000)
+        sample = min(nrows, 100_000_000)
         log(f"[bench] cpu baseline leg on {sample} rows")
         cpu_baseline = cpu_baseline_leg(cfg, sample)
 

@@ -476,6 +476,36 @@ static void orc_finalize_group(const OrcGroup* g, const BkQuerySpec* q,
     }
 }
 
+typedef struct OrcMergeTask {
+    OrcAggTask* tasks;
+    int ntasks;
+    const BkQuerySpec* q;
+    int part, nparts;
+    OrcMap out;
+} OrcMergeTask;
+
+static void* orc_merge_worker(void* arg) {
+    OrcMergeTask* m = (OrcMergeTask*)arg;
+    const BkQuerySpec* q = m->q;
+    orc_map_init(&m->out, 1024);
+    for (int t = 0; t < m->ntasks; t++) {
+        OrcMap* mt = &m->tasks[t].map;
+        for (uint64_t i = 0; i < mt->cap; i++) {
+            if (!mt->hashes[i]) continue;
+            if ((int)(mt->hashes[i] % (uint64_t)m->nparts) != m->part) continue;
+            int created;
+            if (m->out.n * 10 >= m->out.cap * 6) orc_map_grow(&m->out, q->n_group);
+            OrcGroup* g = orc_map_find_or_insert(&m->out, mt->groups[i].flag,
+                                                 mt->groups[i].e, q->n_group,
+                                                 &created);
+            for (int32_t a = 0; a < q->n_aggs; a++)
+                orc_agg_merge(&g->st[a], &mt->groups[i].st[a],
+                              q->aggs[a].agg_type, q->agg_in_types[a]);
+        }
+    }
+    return NULL;
+}
+
 /* Run filter+aggregate over [row_begin,row_end) with nthreads shards
  * (each shard mirrors one region's AggNode, merged like MERGE_AGG —
  * agg_node.cpp:29,539-543). Results sorted by reference key bytes iff
@@ -507,28 +537,37 @@ ORC_EXPORT OrcAggResult* orc_filter_agg(const OrcCol* cols, int ncols,
     if (nthreads > 1)
         for (int t = 0; t < nthreads; t++) pthread_join(tids[t], NULL);
 
-    /* merge partials into tasks[0].map (MERGE_AGG path, agg_node.cpp:539-543) */
-    OrcMap* m0 = &tasks[0].map;
-    int64_t rows_passed = tasks[0].rows_passed;
-    for (int t = 1; t < nthreads; t++) {
-        OrcMap* mt = &tasks[t].map;
-        rows_passed += tasks[t].rows_passed;
-        for (uint64_t i = 0; i < mt->cap; i++) {
-            if (!mt->hashes[i]) continue;
-            int created;
-            if (m0->n * 10 >= m0->cap * 6) orc_map_grow(m0, q->n_group);
-            OrcGroup* g = orc_map_find_or_insert(m0, mt->groups[i].flag,
-                                                 mt->groups[i].e, q->n_group, &created);
-            for (int32_t a = 0; a < q->n_aggs; a++)
-                orc_agg_merge(&g->st[a], &mt->groups[i].st[a],
-                              q->aggs[a].agg_type, q->agg_in_types[a]);
-        }
-        orc_map_free(mt);
+    /* merge partials (MERGE_AGG path, agg_node.cpp:539-543) — parallel by
+     * key-hash partition: merge thread t owns groups with hash%T == t, so
+     * the combine scales with cores instead of serializing. */
+    int64_t rows_passed = 0;
+    for (int t = 0; t < nthreads; t++) rows_passed += tasks[t].rows_passed;
+    int nmerge = nthreads;
+    OrcMergeTask* mtasks = (OrcMergeTask*)calloc((size_t)nmerge, sizeof(OrcMergeTask));
+    pthread_t* mtids = (pthread_t*)calloc((size_t)nmerge, sizeof(pthread_t));
+    for (int t = 0; t < nmerge; t++) {
+        mtasks[t].tasks = tasks;
+        mtasks[t].ntasks = nthreads;
+        mtasks[t].q = q;
+        mtasks[t].part = t;
+        mtasks[t].nparts = nmerge;
+        if (nmerge == 1) orc_merge_worker(&mtasks[t]);
+        else pthread_create(&mtids[t], NULL, orc_merge_worker, &mtasks[t]);
     }
+    if (nmerge > 1)
+        for (int t = 0; t < nmerge; t++) pthread_join(mtids[t], NULL);
+    for (int t = 0; t < nthreads; t++) orc_map_free(&tasks[t].map);
+    /* steal the merged maps into one logical view */
+    OrcMap* parts = (OrcMap*)calloc((size_t)nmerge, sizeof(OrcMap));
+    for (int t = 0; t < nmerge; t++) parts[t] = mtasks[t].out;
+    free(mtasks); free(mtids);
+    OrcMap* m0 = &parts[0];  /* used only for the zero-row special case */
 
     /* zero-row, no-GROUP-BY query => one all-initialized row
      * (agg_node.cpp:490-505; COUNT 0, SUM/AVG/MIN/MAX NULL) */
-    if (m0->n == 0 && q->n_group == 0) {
+    uint64_t total_groups = 0;
+    for (int t = 0; t < nmerge; t++) total_groups += parts[t].n;
+    if (total_groups == 0 && q->n_group == 0) {
         int created;
         uint64_t e[BK_MAX_GROUP] = {0, 0};
         OrcGroup* g = orc_map_find_or_insert(m0, 0, e, 0, &created);
@@ -539,11 +578,15 @@ ORC_EXPORT OrcAggResult* orc_filter_agg(const OrcCol* cols, int ncols,
     }
 
     /* collect + optional canonical sort */
-    int64_t ngroups = (int64_t)m0->n;
-    KeySortRef* refs = (KeySortRef*)malloc((size_t)ngroups * sizeof(KeySortRef));
+    total_groups = 0;
+    for (int t = 0; t < nmerge; t++) total_groups += parts[t].n;
+    int64_t ngroups = (int64_t)total_groups;
+    KeySortRef* refs = (KeySortRef*)malloc((size_t)(ngroups > 0 ? ngroups : 1)
+                                           * sizeof(KeySortRef));
     int64_t gi = 0;
-    for (uint64_t i = 0; i < m0->cap; i++)
-        if (m0->hashes[i]) refs[gi++].g = &m0->groups[i];
+    for (int t = 0; t < nmerge; t++)
+        for (uint64_t i = 0; i < parts[t].cap; i++)
+            if (parts[t].hashes[i]) refs[gi++].g = &parts[t].groups[i];
     if (sort_keys) {
         s_sort_ng = q->n_group;
         qsort(refs, (size_t)ngroups, sizeof(KeySortRef), orc_group_cmp);
@@ -575,7 +618,8 @@ ORC_EXPORT OrcAggResult* orc_filter_agg(const OrcCol* cols, int ncols,
             res->g_enc[g2 * BK_MAX_GROUP + k] = refs[g2].g->e[k];
     }
     free(refs);
-    orc_map_free(m0);
+    for (int t = 0; t < nmerge; t++) orc_map_free(&parts[t]);
+    free(parts);
     free(tasks); free(tids);
     return res;
 }
