@@ -105,11 +105,54 @@ __device__ __forceinline__ double cell_f64(const DevCol& c, int64_t r) {
     return (double)cell_i64(c, r);   /* get_numberic<double>, expr_value.h:341 */
 }
 
-/* need_copy (filter_node.cpp:726-734): all conjuncts non-NULL and true */
+/* need_copy (filter_node.cpp:726-734): all conjuncts non-NULL and true.
+ * All conjunct columns are loaded EAGERLY (no short-circuit between loads):
+ * dependent loads serialize ~900-cycle HBM latencies per conjunct, while at
+ * the bench selectivities nearly every cacheline is touched anyway, so eager
+ * issue trades no traffic for full memory-level parallelism. */
 __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpec& q,
                                            int64_t r) {
-    #pragma unroll 4
-    for (int32_t j = 0; j < q.n_conjuncts; j++) {
+    /* manually scalarized so the staged values live in registers (indexed
+     * locals spill to scratch — guide §5.4 rule 20); the first 4 conjuncts
+     * (all bench queries) issue eagerly, the rare rest evaluates lazily */
+    #define BK_EVAL1(J, VI, VD, OK)                                         \
+        int64_t VI = 0; double VD = 0.0; bool OK = true;                    \
+        if (q.n_conjuncts > (J)) {                                          \
+            const BkConjunct& cj = q.conjuncts[J];                          \
+            const DevCol& c = cols.c[cj.col];                               \
+            OK = cell_valid(c, r);                                          \
+            if (cj.cmp_type == BK_DOUBLE) VD = cell_f64(c, r);              \
+            else VI = cell_i64(c, r);                                       \
+        }
+    BK_EVAL1(0, vi0, vd0, ok0)
+    BK_EVAL1(1, vi1, vd1, ok1)
+    BK_EVAL1(2, vi2, vd2, ok2)
+    BK_EVAL1(3, vi3, vd3, ok3)
+    #undef BK_EVAL1
+    #define BK_TEST1(J, VI, VD, OK)                                          \
+        if (q.n_conjuncts > (J)) {                                          \
+            const BkConjunct& cj = q.conjuncts[J];                          \
+            int cmp = (cj.cmp_type == BK_DOUBLE)                            \
+                ? ((VD > cj.lit_d) - (VD < cj.lit_d))                       \
+                : ((VI > cj.lit_i) - (VI < cj.lit_i));                      \
+            bool pass;                                                      \
+            switch (cj.op) {                                                \
+                case BK_OP_EQ: pass = (cmp == 0); break;                    \
+                case BK_OP_NE: pass = (cmp != 0); break;                    \
+                case BK_OP_GT: pass = (cmp > 0);  break;                    \
+                case BK_OP_GE: pass = (cmp >= 0); break;                    \
+                case BK_OP_LT: pass = (cmp < 0);  break;                    \
+                default:       pass = (cmp <= 0); break;                    \
+            }                                                               \
+            pass_all = pass_all && (OK) && pass;                            \
+        }
+    bool pass_all = true;
+    BK_TEST1(0, vi0, vd0, ok0)
+    BK_TEST1(1, vi1, vd1, ok1)
+    BK_TEST1(2, vi2, vd2, ok2)
+    BK_TEST1(3, vi3, vd3, ok3)
+    #undef BK_TEST1
+    for (int32_t j = 4; j < q.n_conjuncts && pass_all; j++) {
         const BkConjunct& cj = q.conjuncts[j];
         const DevCol& c = cols.c[cj.col];
         if (!cell_valid(c, r)) return false;
@@ -128,11 +171,11 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
             case BK_OP_GT: pass = (cmp > 0);  break;
             case BK_OP_GE: pass = (cmp >= 0); break;
             case BK_OP_LT: pass = (cmp < 0);  break;
-            default:       pass = (cmp <= 0); break;  /* LE */
+            default:       pass = (cmp <= 0); break;
         }
-        if (!pass) return false;
+        pass_all = pass_all && pass;
     }
-    return true;
+    return pass_all;
 }
 
 /* order-preserving u64 encode of a group/minmax value (bk_keyenc.h) */
