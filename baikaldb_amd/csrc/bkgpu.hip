@@ -1784,8 +1784,17 @@ k_filter_collect(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end
     int64_t gstride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          r < row_end; r += gstride) {
-        if (!row_passes(cols, q, r)) continue;
-        uint64_t i = atomicAdd((unsigned long long*)counter, 1ull);
+        bool want = row_passes(cols, q, r);
+        uint64_t m = __ballot(want);
+        if (m == 0) continue;
+        int lead = __ffsll((unsigned long long)m) - 1;
+        unsigned long long base_i = 0;
+        if ((int)(threadIdx.x & 63) == lead)
+            base_i = atomicAdd((unsigned long long*)counter,
+                               (unsigned long long)__popcll(m));
+        base_i = __shfl(base_i, lead, 64);
+        if (!want) continue;
+        uint64_t i = base_i + __popcll(m & ((1ull << (threadIdx.x & 63)) - 1));
         if ((int64_t)i < limit) out_rowids[i] = r;
     }
 }
