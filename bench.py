@@ -306,7 +306,15 @@ def main():
     algo_bytes = cfg["bytes_per_row"] * nrows  # per launch (this rank)
     achieved_gbs = (algo_bytes / (avg_kernel_ms / 1000.0)) / 1e9 if avg_kernel_ms else 0.0
     peak_gbs = 8000.0  # HBM3E spec peak (MI355X_MICROARCH.md)
-    traffic = os.environ.get("BK_TRAFFIC_BYTES")  # from separate rocprof --pmc run
+    # traffic: HBM bytes per launch from separate rocprofv3 --pmc passes
+    # (FETCH_SIZE x2 wide-read calibration + WRITE_SIZE; see profiles/).
+    # Regenerated by tools/pmc_traffic.py; env BK_TRAFFIC_BYTES overrides.
+    traffic = os.environ.get("BK_TRAFFIC_BYTES")
+    if traffic is None:
+        tf = os.path.join(REPO, "profiles", f"traffic_{args.config}.json")
+        if os.path.exists(tf) and nrows == cfg["nrows"]:
+            with open(tf) as f:
+                traffic = json.load(f).get("bytes_per_launch")
     roofline = {"bound": "hbm", "achieved": round(achieved_gbs, 1),
                 "peak": peak_gbs, "unit": "GB/s",
                 "frac": round(achieved_gbs / peak_gbs, 4),
