@@ -792,6 +792,7 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
         uint64_t* my = rec + (size_t)pos * lay.nwords;   /* AoS record */
         my[0] = k0;
         if (lay.k1_word >= 0) my[lay.k1_word] = k1;
+        else if (lay.k1_word == -2) meta |= k1 << 32;
         for (int32_t a = 0; a < q.n_aggs; a++) {
             if (lay.val_word[a] < 0) continue;
             const BkAggSpec& as = q.aggs[a];
@@ -997,8 +998,9 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
             uint32_t flag = 0;
             if (pending) {
                 k0 = my[0];
-                k1 = lay.k1_word >= 0 ? my[lay.k1_word] : 0;
                 meta = lay.meta_word >= 0 ? my[lay.meta_word] : 0;
+                k1 = lay.k1_word >= 0 ? my[lay.k1_word]
+                     : (lay.k1_word == -2 ? (meta >> 32) : 0);
                 flag = (uint32_t)(meta & 0xFF);
             }
             for (;;) {
@@ -1355,8 +1357,13 @@ static int agg_compact(BkgAggOut* o) {
 
 static int build_rec_layout(const BkgTable* t, const BkQuerySpec* q, RecLayout* lay) {
     int w = 1; /* word 0 = k0 */
-    lay->k1_word = q->n_group >= 2 ? w++ : -1;
-    bool need_meta = false;
+    /* a dict-encoded (BK_STRING) second group key is a 32-bit code: pack it
+     * into the meta word's high half (k1_word == -2) instead of spending a
+     * whole record word — 20% narrower records on the config-3 shape */
+    bool k1_in_meta = q->n_group >= 2 &&
+                      t->specs[q->group_cols[1]].col_type == BK_STRING;
+    lay->k1_word = q->n_group >= 2 ? (k1_in_meta ? -2 : w++) : -1;
+    bool need_meta = k1_in_meta;
     for (int k = 0; k < q->n_group; k++)
         if (t->valid[q->group_cols[k]]) need_meta = true;
     for (int a = 0; a < q->n_aggs; a++) {
