@@ -244,10 +244,13 @@ def main():
                      for _ in range(world)]
             dist.all_gather(sizes, ng)
             sizes = [int(s.item()) for s in sizes]
-            nbytes = res.export_bytes()
-            per_group = nbytes // max(res.ngroups, 1)
+            # wire-blob bytes per group (flags4 + k0/k1 16 + states 16*naggs)
+            # — computed analytically so every rank allocates identically
+            # even when its own result is empty
+            per_group = 20 + 16 * len(cfg["aggs"])
             maxg = max(sizes)
-            buf = torch.zeros(maxg * per_group, dtype=torch.uint8, device="cuda")
+            buf = torch.zeros(max(maxg * per_group, 1), dtype=torch.uint8,
+                              device="cuda")
             if res.ngroups > 0:
                 res.export_to(buf.data_ptr(), maxg * per_group)
             gath = [torch.zeros_like(buf) for _ in range(world)]

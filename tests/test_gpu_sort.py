@@ -147,3 +147,25 @@ def test_large_property(eng, orc):
     order = np.lexsort((got, k1, k0))
     assert np.array_equal(order, np.arange(limit)), "output not sorted"
     _ = comp
+
+
+def test_nullable_order_columns(eng, orc):
+    """NULLs in ORDER BY columns: is_null_first decides NULL placement
+    regardless of asc/desc (mem_row_compare.cpp:22-31)."""
+    specs = [(TYPE_INT64, 0, 0, 1000, 400_000),      # 40% NULLs, heavy ties
+             (TYPE_INT64, 0, 0, 1 << 31, 150_000),
+             (TYPE_DOUBLE, 3, 0, 0, 250_000)]
+    for order in ([(0, 1, 1), (1, 1, 1)],     # nulls first, asc
+                  [(0, 1, 0), (1, 1, 0)],     # nulls last, asc
+                  [(0, 0, 1), (2, 1, 0)],     # desc + nulls-first, mixed
+                  [(2, 0, 0)]):               # double desc nulls-last
+        got, exp = run_both(eng, orc, specs, 60_000, order, 2000)
+        assert np.array_equal(got, exp), order
+
+
+def test_nullable_order_with_filter(eng, orc):
+    specs = [(TYPE_INT64, 0, 0, 500, 300_000),
+             (TYPE_INT64, 0, 0, 1 << 31, 0)]
+    got, exp = run_both(eng, orc, specs, 50_000, [(0, 1, 1), (1, 1, 1)], 1500,
+                        conjuncts=[(1, "<", int((1 << 31) * 0.6))])
+    assert np.array_equal(got, exp)
