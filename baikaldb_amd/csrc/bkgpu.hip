@@ -839,7 +839,8 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
             }
             regs[lay.meta_word] = meta;
         }
-        /* flush the record with nt stores, widest first */
+        /* flush the record with the widest stores available (nt measured
+         * worse: partial-line nt stores are unmerged fabric writes) */
         typedef long long ll2 __attribute__((ext_vector_type(2)));
         #pragma unroll
         for (int w = 0; w < 8; w += 2) {
@@ -847,9 +848,9 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
                 ll2 v2;
                 v2.x = (long long)regs[w];
                 v2.y = (long long)regs[w + 1];
-                __builtin_nontemporal_store(v2, (ll2*)&my[w]);
+                *(ll2*)&my[w] = v2;
             } else if (w < lay.nwords) {
-                __builtin_nontemporal_store(regs[w], &my[w]);
+                my[w] = regs[w];
             }
         }
     }
