@@ -171,12 +171,112 @@ public:
     }
     int n_conjuncts() const { return _n_conjuncts; }
     const BkConjunct* conjuncts() const { return _conjuncts; }
+    /* As the EFFECTIVE ROOT (SELECT without GROUP BY/ORDER BY) FilterNode
+     * emits the passing rows itself (filter_node.cpp:736-795): one GPU
+     * filter_collect for the row ids, then column gathers. All table
+     * columns become the row's slots (the child scan's tuple). */
+    int get_next(RuntimeState* state, RowBatch* batch, bool* eos) override;
+    int n_slots();  /* lazily resolved from the scan's table */
+    void close(RuntimeState* state) override {
+        ExecNode::close(state);
+        _materialized = false;
+        _rowids.clear();
+        _cols_i.clear(); _cols_d.clear(); _cols_n.clear();
+        _iter = 0;
+    }
 private:
+    int materialize(RuntimeState* state);
     int32_t _n_conjuncts = 0;
     BkConjunct _conjuncts[BK_MAX_CONJUNCTS] = {};
+    bool _materialized = false;
+    int _ncols = 0;
+    std::vector<int64_t> _rowids;
+    std::vector<std::vector<int64_t>> _cols_i;
+    std::vector<std::vector<double>> _cols_d;
+    std::vector<std::vector<uint8_t>> _cols_n;
+    std::vector<int32_t> _col_types;
+    int64_t _iter = 0;
 };
 
 /* helpers to locate the pipeline pieces below a blocking node */
+static ScanNode* find_scan(ExecNode* n);
+
+int FilterNode::n_slots() {
+    if (_ncols == 0) {
+        ScanNode* scan = find_scan(this);
+        if (scan) _ncols = bkgpu_table_ncols(scan->table());
+    }
+    return _ncols;
+}
+
+int FilterNode::materialize(RuntimeState* state) {
+    ScanNode* scan = find_scan(this);
+    if (!scan) { state->error_msg = "FilterNode: no scan below"; return -1; }
+    BkgTable* t = scan->table();
+    BkQuerySpec q{};
+    q.n_conjuncts = _n_conjuncts;
+    memcpy(q.conjuncts, _conjuncts, sizeof(q.conjuncts));
+    int64_t nrows = bkgpu_table_nrows(t);
+    int64_t limit = _limit > 0 ? _limit : nrows;
+    _rowids.resize(limit > 0 ? limit : 1);
+    int64_t got = bkgpu_filter_collect(t, &q, 0, nrows, limit, _rowids.data());
+    if (got < 0) { state->error_msg = bkgpu_last_error(); return -1; }
+    _rowids.resize(got);
+    /* arrival order: the reference scan emits rows in iterator order */
+    std::sort(_rowids.begin(), _rowids.end());
+    state->inc_num_scan_rows(nrows);
+    state->inc_num_filter_rows(nrows - got);
+    _ncols = bkgpu_table_ncols(t);
+    _cols_i.assign(_ncols, {});
+    _cols_d.assign(_ncols, {});
+    _cols_n.assign(_ncols, {});
+    _col_types.resize(_ncols);
+    for (int c = 0; c < _ncols; c++) {
+        _col_types[c] = bkgpu_table_col_type(t, c);
+        _cols_i[c].resize(got ? got : 1);
+        _cols_d[c].resize(got ? got : 1);
+        _cols_n[c].resize(got ? got : 1);
+        if (got > 0 && bkgpu_gather(t, c, _rowids.data(), got,
+                                    _cols_i[c].data(), _cols_d[c].data(),
+                                    _cols_n[c].data()) != 0) {
+            state->error_msg = bkgpu_last_error();
+            return -1;
+        }
+    }
+    _iter = 0;
+    _materialized = true;
+    return 0;
+}
+
+int FilterNode::get_next(RuntimeState* state, RowBatch* batch, bool* eos) {
+    if (!_materialized) {
+        int ret = materialize(state);
+        if (ret < 0) return ret;
+    }
+    while (true) {
+        if (state->is_cancelled()) { *eos = true; return 0; }
+        if (reached_limit() || _iter >= (int64_t)_rowids.size()) {
+            *eos = true;
+            return 0;
+        }
+        if (batch->is_full()) return 0;
+        auto row = std::make_unique<MemRow>(_ncols);
+        for (int c = 0; c < _ncols; c++) {
+            ExprValue v;
+            v.type = _col_types[c];
+            v.is_null_ = _cols_n[c][_iter] != 0;
+            if (!v.is_null_) {
+                if (v.type == BK_DOUBLE) v.d = _cols_d[c][_iter];
+                else v.i = _cols_i[c][_iter];
+            }
+            row->set_value(c, v);
+        }
+        batch->move_row(std::move(row));
+        _num_rows_returned++;
+        _iter++;
+    }
+}
+
 static ScanNode* find_scan(ExecNode* n) {
     if (!n) return nullptr;
     if (n->node_type() == BK_SCAN_NODE) return static_cast<ScanNode*>(n);
@@ -501,6 +601,9 @@ static int tree_n_slots(const BkExecTree* t) {
         return static_cast<AggNode*>(n)->n_slots();
     if (n->node_type() == BK_SORT_NODE)
         return static_cast<SortNode*>(n)->n_slots();
+    if (n->node_type() == BK_TABLE_FILTER_NODE ||
+        n->node_type() == BK_WHERE_FILTER_NODE)
+        return static_cast<FilterNode*>(n)->n_slots();
     return 0;
 }
 

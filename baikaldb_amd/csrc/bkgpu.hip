@@ -1236,6 +1236,10 @@ extern "C" BkgTable* bkgpu_table_create(int ncols, const BkColSpec* specs,
 
 extern "C" int64_t bkgpu_table_nrows(const BkgTable* t) { return t ? t->nrows : 0; }
 
+extern "C" int32_t bkgpu_table_ncols(const BkgTable* t) {
+    return t ? t->ncols : 0;
+}
+
 extern "C" int32_t bkgpu_table_col_type(const BkgTable* t, int col) {
     if (!t || col < 0 || col >= t->ncols) return BK_INVALID_TYPE;
     return t->specs[col].col_type;

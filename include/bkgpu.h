@@ -59,6 +59,7 @@ int  bkgpu_table_generate(BkgTable* t, uint64_t seed, int64_t row_begin);
 int  bkgpu_table_upload(BkgTable* t, int col, const void* data, const uint8_t* valid);
 int64_t bkgpu_table_nrows(const BkgTable* t);
 int32_t bkgpu_table_col_type(const BkgTable* t, int col);
+int32_t bkgpu_table_ncols(const BkgTable* t);
 void bkgpu_table_free(BkgTable* t);
 
 /* ---- fused scan+filter+aggregate ---- */
@@ -103,9 +104,10 @@ void bkgpu_agg_free(BkgAggOut* o);
 
 /* ---- ORDER BY ... LIMIT top-N ----
  * Select the `limit` smallest rows of [row_begin,row_end) passing q's filter
- * under `order` (non-null INT64/DOUBLE keys; ties broken by arrival index,
- * topn_sorter.h:46-54), write their global row ids in final order to
- * out_rows (host array). Returns count written, negative on error. */
+ * under `order` (INT64/DOUBLE/dict keys, nullable columns ordered per
+ * is_null_first; ties broken by arrival index, topn_sorter.h:46-54), write
+ * their global row ids in final order to out_rows (host array). Returns
+ * count written, negative on error. */
 int64_t bkgpu_sort_topk(BkgTable* t, const BkQuerySpec* q,
                         const BkOrderSpec* order, int norder,
                         int64_t row_begin, int64_t row_end,

@@ -129,3 +129,62 @@ def test_scan_root_refuses_row_mode(eng, orc):
         tree.close()
     finally:
         t.free()
+
+
+def test_filter_root_emits_rows(eng, orc):
+    """FilterNode as effective root (SELECT * WHERE ...): rows materialize in
+    scan/arrival order with all table columns as slots."""
+    from baikaldb_amd import exec as bx
+    t, cols, valids, types = make_table(eng, orc, n=40_000)
+    try:
+        nodes = [bx.filter_node(types, [(0, "<", 1 << 26)]),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
+    mask = cols[0] < (1 << 26)
+    idx = np.nonzero(mask)[0]
+    assert tags.shape == (len(idx), 4)
+    assert np.array_equal(vi[:, 0], cols[0][idx])
+    assert np.array_equal(vi[:, 2], cols[2][idx])
+    np.testing.assert_array_equal(vd[:, 3], cols[3][idx])
+
+
+def test_limit_over_filter_root(eng, orc):
+    from baikaldb_amd import exec as bx
+    t, cols, valids, types = make_table(eng, orc, n=30_000)
+    try:
+        nodes = [bx.limit_node(25),
+                 bx.filter_node(types, [(2, ">=", 100)]),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all(batch=10)
+        tree.close()
+    finally:
+        t.free()
+    idx = np.nonzero(cols[2] >= 100)[0][:25]
+    assert np.array_equal(vi[:, 2], cols[2][idx])
+
+
+def test_merge_agg_node_type(eng, orc):
+    """MERGE_AGG node type shares the aggregate implementation."""
+    from baikaldb_amd import exec as bx
+    t, cols, valids, types = make_table(eng, orc, n=30_000)
+    try:
+        nodes = [bx.agg_node(group=[1], aggs=[("count_star", -1)], merge=True),
+                 bx.filter_node(types, []),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
+    import numpy as _np
+    vals, counts = _np.unique(cols[1], return_counts=True)
+    assert tags.shape[0] == len(vals)
+    assert _np.array_equal(vi[:, 1], counts)
