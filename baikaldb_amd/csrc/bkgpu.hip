@@ -760,16 +760,46 @@ __global__ void k_part_offsets(uint32_t* H, uint32_t nblocks, uint32_t P,
     }
 }
 
-/* pass 2: scatter records (SoA: word w of row i at rec[w*total + i]).
+/* widest-stores record flush (nt measured worse: partial-line nt stores are
+ * unmerged fabric writes) */
+__device__ __forceinline__ void scatter_store_rec(uint64_t* dst,
+                                                  const uint64_t* regs,
+                                                  int nwords) {
+    typedef long long ll2 __attribute__((ext_vector_type(2)));
+    #pragma unroll
+    for (int w = 0; w < 8; w += 2) {
+        if (w + 1 < nwords) {
+            ll2 v2;
+            v2.x = (long long)regs[w];
+            v2.y = (long long)regs[w + 1];
+            *(ll2*)&dst[w] = v2;
+        } else if (w < nwords) {
+            dst[w] = regs[w];
+        }
+    }
+}
+
+/* pass 2: scatter AoS records into bucket-contiguous regions.
  * MUST run with the same grid/block shape as k_part_histo so each block
  * sees the same rows its H row was computed from. */
 __global__ void __launch_bounds__(256)
 k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
                int64_t row_end, uint32_t P, const uint16_t* bucketid,
-               const uint32_t* H, uint64_t* rec, uint64_t total) {
-    extern __shared__ __attribute__((aligned(16))) uint32_t lcur[];
-    for (uint32_t b = threadIdx.x; b < P; b += blockDim.x)
+               const uint32_t* H, uint64_t* rec, uint64_t total, int paired) {
+    /* LDS carve: [ stash: P*nwords u64 (pairing only) ][ lcur: P u32 ]
+     *            [ bstate: P u32 ]
+     * Pair staging: a lone 32-40 B record store touches a 64-128 B line that
+     * gets written back before its neighbour arrives (~2.5x write
+     * amplification by PMC). Stash ONE record per bucket in LDS; the second
+     * arrival writes BOTH contiguously. */
+    extern __shared__ __attribute__((aligned(16))) uint64_t lmem[];
+    uint64_t* stash = lmem;
+    uint32_t* lcur = (uint32_t*)(lmem + (paired ? (size_t)P * lay.nwords : 0));
+    uint32_t* bstate = lcur + P;
+    for (uint32_t b = threadIdx.x; b < P; b += blockDim.x) {
         lcur[b] = H[(size_t)blockIdx.x * P + b];
+        if (paired) bstate[b] = 0;
+    }
     __syncthreads();
     int64_t gstride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -777,7 +807,6 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
         int64_t i = r - row_begin;
         uint32_t b = bucketid[i];
         if (b >= BK_HOT_BUCKET) continue;  /* filtered out or absorbed hot */
-        uint32_t pos = atomicAdd(&lcur[b], 1u);
         /* key words */
         uint64_t meta = 0;
         uint64_t k0 = 0, k1 = 0;
@@ -789,10 +818,6 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
             const DevCol& c = cols.c[q.group_cols[1]];
             if (!cell_valid(c, r)) meta |= 0x40u; else k1 = enc_value(c, r);
         }
-        /* AoS record, non-temporal stores: scattered 32-40 B records would
-         * otherwise read-for-ownership whole lines through L2 (2.5x write
-         * amplification measured by PMC) */
-        uint64_t* my = rec + (size_t)pos * lay.nwords;
         uint64_t regs[8];
         #pragma unroll
         for (int w = 0; w < 8; w++) regs[w] = 0;
@@ -839,19 +864,58 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
             }
             regs[lay.meta_word] = meta;
         }
-        /* flush the record with the widest stores available (nt measured
-         * worse: partial-line nt stores are unmerged fabric writes) */
-        typedef long long ll2 __attribute__((ext_vector_type(2)));
-        #pragma unroll
-        for (int w = 0; w < 8; w += 2) {
-            if (w + 1 < lay.nwords) {
-                ll2 v2;
-                v2.x = (long long)regs[w];
-                v2.y = (long long)regs[w + 1];
-                *(ll2*)&my[w] = v2;
-            } else if (w < lay.nwords) {
-                my[w] = regs[w];
+        if (!paired) {
+            uint32_t pos = atomicAdd(&lcur[b], 1u);
+            scatter_store_rec(rec + (size_t)pos * lay.nwords, regs, lay.nwords);
+            continue;
+        }
+        /* pairing: claim the bucket stash, or take it and write a pair */
+        bool done = false;
+        for (int tries = 0; tries < 64 && !done; tries++) {
+            uint32_t st = __hip_atomic_load(&bstate[b], __ATOMIC_RELAXED, WGP);
+            if (st == 0u) {
+                uint32_t exp = 0u;
+                if (__hip_atomic_compare_exchange_strong(
+                        &bstate[b], &exp, 1u, __ATOMIC_ACQUIRE,
+                        __ATOMIC_RELAXED, WGP)) {
+                    uint64_t* sl = stash + (size_t)b * lay.nwords;
+                    for (int w = 0; w < lay.nwords; w++) sl[w] = regs[w];
+                    __hip_atomic_store(&bstate[b], 2u, __ATOMIC_RELEASE, WGP);
+                    done = true;
+                }
+            } else if (st == 2u) {
+                uint32_t exp = 2u;
+                if (__hip_atomic_compare_exchange_strong(
+                        &bstate[b], &exp, 3u, __ATOMIC_ACQUIRE,
+                        __ATOMIC_RELAXED, WGP)) {
+                    uint64_t prev[8];
+                    uint64_t* sl = stash + (size_t)b * lay.nwords;
+                    #pragma unroll
+                    for (int w = 0; w < 8; w++)
+                        prev[w] = w < lay.nwords ? sl[w] : 0;
+                    __hip_atomic_store(&bstate[b], 0u, __ATOMIC_RELEASE, WGP);
+                    uint32_t pos = atomicAdd(&lcur[b], 2u);
+                    uint64_t* dst = rec + (size_t)pos * lay.nwords;
+                    scatter_store_rec(dst, prev, lay.nwords);
+                    scatter_store_rec(dst + lay.nwords, regs, lay.nwords);
+                    done = true;
+                }
             }
+            /* st 1/3: another wave mid-transfer — retry (no in-place spin) */
+        }
+        if (!done) {  /* contention fallback: single-record write */
+            uint32_t pos = atomicAdd(&lcur[b], 1u);
+            scatter_store_rec(rec + (size_t)pos * lay.nwords, regs, lay.nwords);
+        }
+    }
+    /* drain leftover stashes (one half-pair per bucket at most) */
+    if (paired) {
+        __syncthreads();
+        for (uint32_t b = threadIdx.x; b < P; b += blockDim.x) {
+            if (bstate[b] != 2u) continue;
+            uint64_t* sl = stash + (size_t)b * lay.nwords;
+            uint32_t pos = lcur[b]++;
+            scatter_store_rec(rec + (size_t)pos * lay.nwords, sl, lay.nwords);
         }
     }
 }
@@ -1491,9 +1555,13 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
                        0, 0, H, nblocks, P, base, S);
     tm.record();
     if (total > 0) {
-        hipLaunchKernelGGL(k_part_scatter, dim3(nblocks), dim3(threads), P * 4, 0,
+        size_t pair_lds = (size_t)P * lay.nwords * 8 + (size_t)P * 8;
+        int paired = pair_lds <= 130 * 1024 && !getenv("BK_NO_PAIR");
+        size_t sc_lds = paired ? pair_lds : (size_t)P * 8;
+        hipLaunchKernelGGL(k_part_scatter, dim3(nblocks), dim3(threads),
+                           sc_lds, 0,
                            dc, *q, lay, row_begin, row_end, P, bucketid, H,
-                           rec, total);
+                           rec, total, paired);
     }
     tm.record();
     if (total > 0) {
