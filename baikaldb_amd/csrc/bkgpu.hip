@@ -1555,8 +1555,12 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
                        0, 0, H, nblocks, P, base, S);
     tm.record();
     if (total > 0) {
+        /* pair staging halves write traffic (45->24 GB measured) but the
+         * 72 KB stash drops occupancy to 2 blocks/CU and the scatter's
+         * gather reads become latency-bound (33->77 ms): net loss. Kept
+         * opt-in (BK_PAIR=1) until an occupancy-neutral variant exists. */
         size_t pair_lds = (size_t)P * lay.nwords * 8 + (size_t)P * 8;
-        int paired = pair_lds <= 130 * 1024 && !getenv("BK_NO_PAIR");
+        int paired = pair_lds <= 130 * 1024 && getenv("BK_PAIR") != nullptr;
         size_t sc_lds = paired ? pair_lds : (size_t)P * 8;
         hipLaunchKernelGGL(k_part_scatter, dim3(nblocks), dim3(threads),
                            sc_lds, 0,
