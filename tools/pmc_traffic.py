@@ -42,9 +42,9 @@ def per_kernel(dbpat):
 
 def main(config, dominant):
     fetch, ff = per_kernel(
-        f"{REPO}/gpurun_out/pmcF_{config}_FETCH_SIZE/runc/*_results.db")
+        f"{REPO}/gpurun_out/pmcZ_{config}_FETCH_SIZE/runc/*_results.db")
     write, wf = per_kernel(
-        f"{REPO}/gpurun_out/pmcF_{config}_WRITE_SIZE/runc/*_results.db")
+        f"{REPO}/gpurun_out/pmcZ_{config}_WRITE_SIZE/runc/*_results.db")
     lines = [f"# PMC HBM traffic — {config} (separate FETCH_SIZE / WRITE_SIZE passes)",
              f"# sources: {os.path.basename(ff)}, {os.path.basename(wf)}",
              f"# FETCH calibrated x{FETCH_CAL} (wide coalesced reads report 1/2; "
