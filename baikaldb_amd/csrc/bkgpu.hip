@@ -1570,7 +1570,10 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     tm.record();
     if (total > 0) {
         uint32_t lds_slots = 2048;
-        while ((size_t)lds_slots * stride * 8 > 135 * 1024) lds_slots >>= 1;
+        size_t lds_cap = 135 * 1024;
+        const char* envL = getenv("BK_AGG_LDS_KB");
+        if (envL) lds_cap = (size_t)atoi(envL) * 1024;
+        while ((size_t)lds_slots * stride * 8 > lds_cap) lds_slots >>= 1;
         size_t lds_bytes = ((size_t)lds_slots * stride + 1) * 8;
         /* chunk ~ bucket size: fewer generation flushes (flush traffic ~
          * chunks x distinct-per-chunk), while >=2048 chunks keep the chip
