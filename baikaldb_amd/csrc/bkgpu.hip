@@ -1502,7 +1502,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     while ((int64_t)P < expected_groups / 1024 && P < 4096) P <<= 1;
     const char* envP = getenv("BK_PART_P");
     if (envP) P = (uint32_t)atoi(envP);
-    int nblocks = 1024;
+    int nblocks = 2048;
     const char* envB = getenv("BK_PART_BLOCKS");
     if (envB) nblocks = atoi(envB);
     const int threads = 256;
