@@ -132,17 +132,24 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
     #define BK_TEST1(J, VI, VD, OK)                                          \
         if (q.n_conjuncts > (J)) {                                          \
             const BkConjunct& cj = q.conjuncts[J];                          \
-            int cmp = (cj.cmp_type == BK_DOUBLE)                            \
-                ? ((VD > cj.lit_d) - (VD < cj.lit_d))                       \
-                : ((VI > cj.lit_i) - (VI < cj.lit_i));                      \
             bool pass;                                                      \
-            switch (cj.op) {                                                \
-                case BK_OP_EQ: pass = (cmp == 0); break;                    \
-                case BK_OP_NE: pass = (cmp != 0); break;                    \
-                case BK_OP_GT: pass = (cmp > 0);  break;                    \
-                case BK_OP_GE: pass = (cmp >= 0); break;                    \
-                case BK_OP_LT: pass = (cmp < 0);  break;                    \
-                default:       pass = (cmp <= 0); break;                    \
+            if (cj.op >= BK_OP_IN) {                                        \
+                bool found = false;                                         \
+                for (int32_t m = 0; m < cj.n_in; m++)                       \
+                    found = found || (cj.in_list[m] == (VI));               \
+                pass = cj.op == BK_OP_IN ? found : !found;                  \
+            } else {                                                        \
+                int cmp = (cj.cmp_type == BK_DOUBLE)                        \
+                    ? ((VD > cj.lit_d) - (VD < cj.lit_d))                   \
+                    : ((VI > cj.lit_i) - (VI < cj.lit_i));                  \
+                switch (cj.op) {                                            \
+                    case BK_OP_EQ: pass = (cmp == 0); break;                \
+                    case BK_OP_NE: pass = (cmp != 0); break;                \
+                    case BK_OP_GT: pass = (cmp > 0);  break;                \
+                    case BK_OP_GE: pass = (cmp >= 0); break;                \
+                    case BK_OP_LT: pass = (cmp < 0);  break;                \
+                    default:       pass = (cmp <= 0); break;                \
+                }                                                           \
             }                                                               \
             pass_all = pass_all && (OK) && pass;                            \
         }
@@ -156,22 +163,30 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
         const BkConjunct& cj = q.conjuncts[j];
         const DevCol& c = cols.c[cj.col];
         if (!cell_valid(c, r)) return false;
-        int cmp;
-        if (cj.cmp_type == BK_DOUBLE) {
-            double v = cell_f64(c, r);
-            cmp = (v > cj.lit_d) - (v < cj.lit_d);
-        } else {
-            int64_t v = cell_i64(c, r);
-            cmp = (v > cj.lit_i) - (v < cj.lit_i);
-        }
         bool pass;
-        switch (cj.op) {
-            case BK_OP_EQ: pass = (cmp == 0); break;
-            case BK_OP_NE: pass = (cmp != 0); break;
-            case BK_OP_GT: pass = (cmp > 0);  break;
-            case BK_OP_GE: pass = (cmp >= 0); break;
-            case BK_OP_LT: pass = (cmp < 0);  break;
-            default:       pass = (cmp <= 0); break;
+        if (cj.op >= BK_OP_IN) {
+            int64_t v = cell_i64(c, r);
+            bool found = false;
+            for (int32_t m = 0; m < cj.n_in; m++)
+                found = found || (cj.in_list[m] == v);
+            pass = cj.op == BK_OP_IN ? found : !found;
+        } else {
+            int cmp;
+            if (cj.cmp_type == BK_DOUBLE) {
+                double v = cell_f64(c, r);
+                cmp = (v > cj.lit_d) - (v < cj.lit_d);
+            } else {
+                int64_t v = cell_i64(c, r);
+                cmp = (v > cj.lit_i) - (v < cj.lit_i);
+            }
+            switch (cj.op) {
+                case BK_OP_EQ: pass = (cmp == 0); break;
+                case BK_OP_NE: pass = (cmp != 0); break;
+                case BK_OP_GT: pass = (cmp > 0);  break;
+                case BK_OP_GE: pass = (cmp >= 0); break;
+                case BK_OP_LT: pass = (cmp < 0);  break;
+                default:       pass = (cmp <= 0); break;
+            }
         }
         pass_all = pass_all && pass;
     }

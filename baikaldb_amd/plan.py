@@ -13,15 +13,17 @@ OP_EQ, OP_NE, OP_GT, OP_GE, OP_LT, OP_LE = 0, 1, 2, 3, 4, 5
 AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_AVG, AGG_MIN, AGG_MAX = 0, 1, 2, 3, 4, 5
 BK_MAX_GROUP, BK_MAX_CONJ, BK_MAX_AGGS = 2, 8, 8
 
-_OPS = {"=": OP_EQ, "!=": OP_NE, ">": OP_GT, ">=": OP_GE, "<": OP_LT, "<=": OP_LE}
+_OPS = {"=": OP_EQ, "!=": OP_NE, ">": OP_GT, ">=": OP_GE, "<": OP_LT,
+        "<=": OP_LE, "in": 6, "not_in": 7}
 _AGGS = {"count_star": AGG_COUNT_STAR, "count": AGG_COUNT, "sum": AGG_SUM,
          "avg": AGG_AVG, "min": AGG_MIN, "max": AGG_MAX}
 
 
 class BkConjunct(C.Structure):
     _fields_ = [("col", C.c_int32), ("op", C.c_int32),
-                ("cmp_type", C.c_int32), ("_pad", C.c_int32),
-                ("lit_i", C.c_int64), ("lit_d", C.c_double)]
+                ("cmp_type", C.c_int32), ("n_in", C.c_int32),
+                ("lit_i", C.c_int64), ("lit_d", C.c_double),
+                ("in_list", C.c_int64 * 16)]
 
 
 class BkAggSpec(C.Structure):
@@ -66,7 +68,12 @@ class QueryPlan:
             cj.col = col
             cj.op = _OPS[op] if isinstance(op, str) else op
             ct = self.col_types[col]
-            if ct == TYPE_DOUBLE or isinstance(lit, float):
+            if cj.op >= 6:  # IN / NOT IN: lit is a list of int/dict literals
+                cj.cmp_type = TYPE_INT64
+                cj.n_in = len(lit)
+                for m, v in enumerate(lit):
+                    cj.in_list[m] = int(v)
+            elif ct == TYPE_DOUBLE or isinstance(lit, float):
                 cj.cmp_type = TYPE_DOUBLE
                 cj.lit_d = float(lit)
             else:

@@ -53,7 +53,14 @@ typedef enum BkCmpOp {
     BK_OP_GE = 3,
     BK_OP_LT = 4,
     BK_OP_LE = 5,
+    /* IN-list predicates (src/expr/predicate.h InPredicate; NULL operand =>
+     * NULL => row rejected; literal lists carry no NULLs — the planner only
+     * pushes literal IN lists) */
+    BK_OP_IN     = 6,
+    BK_OP_NOT_IN = 7,
 } BkCmpOp;
+
+#define BK_MAX_INLIST 16
 
 /* ---- AggFnCall::AggType subset (include/expr/agg_fn_call.h:52-58) ---- */
 typedef enum BkAggType {
@@ -104,9 +111,10 @@ typedef struct BkConjunct {
     int32_t col;
     int32_t op;        /* BkCmpOp */
     int32_t cmp_type;  /* BkType */
-    int32_t _pad;
+    int32_t n_in;      /* IN-list length (BK_OP_IN / BK_OP_NOT_IN) */
     int64_t lit_i;
     double  lit_d;
+    int64_t in_list[BK_MAX_INLIST];  /* int64 / dict-code IN literals */
 } BkConjunct;
 
 /* One aggregate call (reference: src/expr/agg_fn_call.cpp:496-555 update,

@@ -26,8 +26,9 @@ class BkColSpec(C.Structure):
 
 class BkConjunct(C.Structure):
     _fields_ = [("col", C.c_int32), ("op", C.c_int32),
-                ("cmp_type", C.c_int32), ("_pad", C.c_int32),
-                ("lit_i", C.c_int64), ("lit_d", C.c_double)]
+                ("cmp_type", C.c_int32), ("n_in", C.c_int32),
+                ("lit_i", C.c_int64), ("lit_d", C.c_double),
+                ("in_list", C.c_int64 * 16)]
 
 
 class BkAggSpec(C.Structure):
@@ -74,7 +75,11 @@ def make_query(conjuncts=(), group=(), aggs=(), col_types=None):
     for i, (col, op, cmp_type, lit) in enumerate(conjuncts):
         cj = q.conjuncts[i]
         cj.col, cj.op, cj.cmp_type = col, op, cmp_type
-        if cmp_type == TYPE_DOUBLE:
+        if op >= 6:  # IN / NOT IN: lit is a list
+            cj.n_in = len(lit)
+            for m, v in enumerate(lit):
+                cj.in_list[m] = int(v)
+        elif cmp_type == TYPE_DOUBLE:
             cj.lit_d = float(lit)
             cj.lit_i = 0
         else:

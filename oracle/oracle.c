@@ -135,14 +135,23 @@ static int orc_row_passes(const OrcCol* cols, const BkQuerySpec* q, int64_t r) {
             cmp = (v > cj->lit_i) - (v < cj->lit_i);
         }
         int pass;
-        switch ((BkCmpOp)cj->op) {
-            case BK_OP_EQ: pass = (cmp == 0); break;
-            case BK_OP_NE: pass = (cmp != 0); break;
-            case BK_OP_GT: pass = (cmp > 0);  break;
-            case BK_OP_GE: pass = (cmp >= 0); break;
-            case BK_OP_LT: pass = (cmp < 0);  break;
-            case BK_OP_LE: pass = (cmp <= 0); break;
-            default: pass = 0;
+        if (cj->op == BK_OP_IN || cj->op == BK_OP_NOT_IN) {
+            /* predicate.h InPredicate semantics over literal lists */
+            int64_t v = cell_i64(c, r);
+            int found = 0;
+            for (int32_t m = 0; m < cj->n_in; m++)
+                if (cj->in_list[m] == v) { found = 1; break; }
+            pass = (cj->op == BK_OP_IN) ? found : !found;
+        } else {
+            switch ((BkCmpOp)cj->op) {
+                case BK_OP_EQ: pass = (cmp == 0); break;
+                case BK_OP_NE: pass = (cmp != 0); break;
+                case BK_OP_GT: pass = (cmp > 0);  break;
+                case BK_OP_GE: pass = (cmp >= 0); break;
+                case BK_OP_LT: pass = (cmp < 0);  break;
+                case BK_OP_LE: pass = (cmp <= 0); break;
+                default: pass = 0;
+            }
         }
         if (!pass) return 0;
     }

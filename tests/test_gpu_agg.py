@@ -63,12 +63,13 @@ def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
          specs[i].null_frac_x1e6) = s
     cols, valids = orc.generate_table(list(specs), n, seed)
     col_types = [s[0] for s in spec_rows]
-    ops = {"=": 0, "!=": 1, ">": 2, ">=": 3, "<": 4, "<=": 5}
+    ops = {"=": 0, "!=": 1, ">": 2, ">=": 3, "<": 4, "<=": 5,
+           "in": 6, "not_in": 7}
     aggmap = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5}
     oconj = []
     for col, op, lit in conjuncts:
-        ct = TYPE_DOUBLE if (col_types[col] == TYPE_DOUBLE or
-                             isinstance(lit, float)) else TYPE_INT64
+        ct = TYPE_DOUBLE if (col_types[col] == TYPE_DOUBLE and
+                             not isinstance(lit, (list, tuple))) or             isinstance(lit, float) else TYPE_INT64
         oconj.append((col, ops[op], ct, lit))
     q = make_query(oconj, group, [(aggmap[a], c) for a, c in aggs], col_types)
     exp = orc.filter_agg(cols, valids, col_types, q, nthreads=nthreads,
@@ -292,3 +293,12 @@ def test_full_size_properties(eng, orc):
     assert gs["rows_passed"] == exp["rows_passed"]
     assert np.array_equal(gs["enc"], exp["enc"])
     assert np.array_equal(gs["agg_i"], exp["agg_i"])
+
+
+def test_in_predicates(eng, orc):
+    """IN / NOT IN list predicates (src/expr/predicate.h InPredicate)."""
+    aggs = [("count_star", -1), ("sum", 2), ("avg", 3)]
+    got, exp = run_both(eng, orc, BASE5, 300_000,
+                        [(2, "in", [1, 2, 5, 77, 999]), (4, "not_in", [7, 8])],
+                        [1], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in BASE5])

@@ -10,6 +10,8 @@ from oracle import (BkColSpec, TYPE_INT64, TYPE_DOUBLE, TYPE_STRING,
                     DIST_UNIFORM, DIST_CUBESKEW, DIST_DICT, DIST_SUMU16,
                     OP_LT, OP_EQ, OP_NE, OP_GT,
                     AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_AVG, AGG_MIN, AGG_MAX)
+
+OP_IN, OP_NOT_IN = 6, 7
 from oracle.bindings import make_query
 
 SEED = 0x5EED
@@ -43,9 +45,13 @@ def numpy_reference(cols, valids, types, conjuncts, group, aggs):
             litv = float(lit)
         else:
             x = v.astype(np.int64)
-            litv = int(lit)
-        res = {OP_LT: x < litv, OP_EQ: x == litv, OP_NE: x != litv,
-               OP_GT: x > litv}[op]
+            litv = lit if isinstance(lit, (list, tuple)) else int(lit)
+        if op in (OP_IN, OP_NOT_IN):
+            member = np.isin(x, np.array(list(lit), dtype=np.int64))
+            res = member if op == OP_IN else ~member
+        else:
+            res = {OP_LT: x < litv, OP_EQ: x == litv, OP_NE: x != litv,
+                   OP_GT: x > litv}[op]
         mask &= ok & res
     idx = np.nonzero(mask)[0]
     # group keys: tuple of (is_null, value)
@@ -259,3 +265,11 @@ def test_sort_topk_desc_and_ties(oracle):
 def test_dict_words_unique(oracle):
     words = {oracle.dict_word(SEED, c) for c in range(5000)}
     assert len(words) == 5000
+
+
+def test_in_predicates(oracle):
+    cols, valids, types = small_table(oracle, n=30000, null_frac=100000)
+    conj = [(1, OP_IN, TYPE_INT64, [3, 7, 11, 42]),
+            (4, OP_NOT_IN, TYPE_STRING, [0, 1, 2])]
+    aggs = [(AGG_COUNT_STAR, -1), (AGG_SUM, 1)]
+    check_against_numpy(oracle, cols, valids, types, conj, [2], aggs, nthreads=4)
