@@ -1327,8 +1327,8 @@ extern "C" int32_t bkgpu_table_col_type(const BkgTable* t, int col) {
 extern "C" void bkgpu_table_free(BkgTable* t) {
     if (!t) return;
     for (int c = 0; c < t->ncols; c++) {
-        if (t->data[c]) hipFree(t->data[c]);
-        if (t->valid[c]) hipFree(t->valid[c]);
+        if (t->data[c]) (void)hipFree(t->data[c]);
+        if (t->valid[c]) (void)hipFree(t->valid[c]);
     }
     delete t;
 }
@@ -1422,8 +1422,8 @@ static void agg_release_table(BkgAggOut* o) {
 extern "C" void bkgpu_agg_free(BkgAggOut* o) {
     if (!o) return;
     agg_release_table(o);
-    if (o->ctrs) hipFree(o->ctrs);
-    if (o->err) hipFree(o->err);
+    if (o->ctrs) (void)hipFree(o->ctrs);
+    if (o->err) (void)hipFree(o->err);
     if (o->blob) pool_free(o->blob);
     delete o;
 }
@@ -1486,19 +1486,19 @@ static int build_rec_layout(const BkgTable* t, const BkQuerySpec* q, RecLayout* 
 struct EvTimer {
     hipEvent_t ev[16];
     int n = 0;
-    int record() { hipEventCreate(&ev[n]); hipEventRecord(ev[n]); return n++; }
+    int record() { (void)hipEventCreate(&ev[n]); (void)hipEventRecord(ev[n]); return n++; }
     void finish(BkgAggOut* o, const char* const* names) {
-        hipEventSynchronize(ev[n - 1]);
+        (void)hipEventSynchronize(ev[n - 1]);
         o->n_kernels = n - 1;
         o->kernel_ms = 0.f;
         for (int i = 0; i + 1 < n && i < 8; i++) {
             float ms = 0.f;
-            hipEventElapsedTime(&ms, ev[i], ev[i + 1]);
+            (void)hipEventElapsedTime(&ms, ev[i], ev[i + 1]);
             o->t_ms[i] = ms;
             o->kernel_ms += ms;
             snprintf(o->k_names[i], 16, "%s", names[i]);
         }
-        for (int i = 0; i < n; i++) hipEventDestroy(ev[i]);
+        for (int i = 0; i < n; i++) (void)hipEventDestroy(ev[i]);
         n = 0;
     }
 };
