@@ -133,7 +133,12 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
         if (q.n_conjuncts > (J)) {                                          \
             const BkConjunct& cj = q.conjuncts[J];                          \
             bool pass;                                                      \
-            if (cj.op >= BK_OP_IN) {                                        \
+            if (cj.op >= BK_OP_IN_BITMAP) {                                 \
+                const uint8_t* bm = (const uint8_t*)(uintptr_t)cj.lit_i;    \
+                bool hit = (VI) >= 0 && (VI) < cj.n_in &&                   \
+                           ((bm[(VI) >> 3] >> ((VI) & 7)) & 1);             \
+                pass = cj.op == BK_OP_IN_BITMAP ? hit : !hit;               \
+            } else if (cj.op >= BK_OP_IN) {                                 \
                 bool found = false;                                         \
                 for (int32_t m = 0; m < cj.n_in; m++)                       \
                     found = found || (cj.in_list[m] == (VI));               \
@@ -164,7 +169,12 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
         const DevCol& c = cols.c[cj.col];
         if (!cell_valid(c, r)) return false;
         bool pass;
-        if (cj.op >= BK_OP_IN) {
+        if (cj.op >= BK_OP_IN_BITMAP) {
+            int64_t v = cell_i64(c, r);
+            const uint8_t* bm = (const uint8_t*)(uintptr_t)cj.lit_i;
+            bool hit = v >= 0 && v < cj.n_in && ((bm[v >> 3] >> (v & 7)) & 1);
+            pass = cj.op == BK_OP_IN_BITMAP ? hit : !hit;
+        } else if (cj.op >= BK_OP_IN) {
             int64_t v = cell_i64(c, r);
             bool found = false;
             for (int32_t m = 0; m < cj.n_in; m++)
@@ -1314,6 +1324,23 @@ extern "C" BkgTable* bkgpu_table_create(int ncols, const BkColSpec* specs,
 }
 
 extern "C" int64_t bkgpu_table_nrows(const BkgTable* t) { return t ? t->nrows : 0; }
+
+/* upload an opaque byte buffer (e.g. a dict-code accept bitmap for
+ * BK_OP_IN_BITMAP) to device memory; caller frees with bkgpu_free_ptr. */
+extern "C" void* bkgpu_upload_bytes(const void* data, int64_t n) {
+    if (ensure_device() != 0) return nullptr;
+    void* p = nullptr;
+    if (hipMalloc(&p, (size_t)(n > 0 ? n : 1)) != hipSuccess) return nullptr;
+    if (hipMemcpy(p, data, (size_t)n, hipMemcpyHostToDevice) != hipSuccess) {
+        (void)hipFree(p);
+        return nullptr;
+    }
+    return p;
+}
+
+extern "C" void bkgpu_free_ptr(void* p) {
+    if (p) (void)hipFree(p);
+}
 
 extern "C" int32_t bkgpu_table_ncols(const BkgTable* t) {
     return t ? t->ncols : 0;

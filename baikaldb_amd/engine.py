@@ -79,6 +79,9 @@ def _load():
                                  C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
     lib.bkgpu_table_col_type.restype = C.c_int32
     lib.bkgpu_table_col_type.argtypes = [C.c_void_p, C.c_int]
+    lib.bkgpu_upload_bytes.restype = C.c_void_p
+    lib.bkgpu_upload_bytes.argtypes = [C.c_void_p, C.c_int64]
+    lib.bkgpu_free_ptr.argtypes = [C.c_void_p]
     return lib
 
 
@@ -228,6 +231,15 @@ class GpuEngine:
 
     def topk_kernel_ms(self):
         return self.lib.bkgpu_topk_kernel_ms()
+
+    def upload_bytes(self, data: bytes):
+        p = self.lib.bkgpu_upload_bytes(data, len(data))
+        if not p:
+            raise RuntimeError("upload_bytes failed")
+        return p
+
+    def free_ptr(self, p):
+        self.lib.bkgpu_free_ptr(C.c_void_p(p))
 
     def sync(self):
         self._check(self.lib.bkgpu_sync(), "sync")

@@ -14,7 +14,8 @@ AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_AVG, AGG_MIN, AGG_MAX = 0, 1, 2, 3, 4, 5
 BK_MAX_GROUP, BK_MAX_CONJ, BK_MAX_AGGS = 2, 8, 8
 
 _OPS = {"=": OP_EQ, "!=": OP_NE, ">": OP_GT, ">=": OP_GE, "<": OP_LT,
-        "<=": OP_LE, "in": 6, "not_in": 7}
+        "<=": OP_LE, "in": 6, "not_in": 7,
+        "in_bitmap": 8, "not_in_bitmap": 9}
 _AGGS = {"count_star": AGG_COUNT_STAR, "count": AGG_COUNT, "sum": AGG_SUM,
          "avg": AGG_AVG, "min": AGG_MIN, "max": AGG_MAX}
 
@@ -68,7 +69,10 @@ class QueryPlan:
             cj.col = col
             cj.op = _OPS[op] if isinstance(op, str) else op
             ct = self.col_types[col]
-            if cj.op >= 6:  # IN / NOT IN: lit is a list of int/dict literals
+            if cj.op >= 8:  # bitmap membership: lit = (device_ptr, n_bits)
+                cj.cmp_type = TYPE_INT64
+                cj.lit_i, cj.n_in = int(lit[0]), int(lit[1])
+            elif cj.op >= 6:  # IN / NOT IN: lit is a list of int/dict literals
                 cj.cmp_type = TYPE_INT64
                 cj.n_in = len(lit)
                 for m, v in enumerate(lit):

@@ -58,6 +58,14 @@ typedef enum BkCmpOp {
      * pushes literal IN lists) */
     BK_OP_IN     = 6,
     BK_OP_NOT_IN = 7,
+    /* dict-code bitmap membership: lit_i carries a pointer (device pointer
+     * for the GPU engine, host pointer for the oracle) to a bitmap of
+     * n_in bits, bit c set <=> dict code c accepted. This is how a LIKE (or
+     * any dict-valued predicate) pushes down onto a dictionary column: the
+     * embedder matches the pattern against its dictionary once and ships
+     * the accept set (the cstore-dict pushdown pattern). */
+    BK_OP_IN_BITMAP     = 8,
+    BK_OP_NOT_IN_BITMAP = 9,
 } BkCmpOp;
 
 #define BK_MAX_INLIST 16

@@ -122,6 +122,10 @@ int64_t bkgpu_filter_collect(BkgTable* t, const BkQuerySpec* q,
                              int64_t limit, int64_t* out_rowids_host);
 int bkgpu_gather(BkgTable* t, int col, const int64_t* rowids_host, int64_t n,
                  int64_t* out_i, double* out_d, uint8_t* out_null);
+/* upload an opaque byte buffer (e.g. a BK_OP_IN_BITMAP dict-code accept
+ * bitmap) to device; free with bkgpu_free_ptr */
+void* bkgpu_upload_bytes(const void* data, int64_t n);
+void  bkgpu_free_ptr(void* p);
 /* release pooled device buffers */
 void bkgpu_pool_trim(void);
 

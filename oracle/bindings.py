@@ -75,7 +75,9 @@ def make_query(conjuncts=(), group=(), aggs=(), col_types=None):
     for i, (col, op, cmp_type, lit) in enumerate(conjuncts):
         cj = q.conjuncts[i]
         cj.col, cj.op, cj.cmp_type = col, op, cmp_type
-        if op >= 6:  # IN / NOT IN: lit is a list
+        if op >= 8:  # bitmap membership: lit = (host_ptr, n_bits)
+            cj.lit_i, cj.n_in = int(lit[0]), int(lit[1])
+        elif op >= 6:  # IN / NOT IN: lit is a list
             cj.n_in = len(lit)
             for m, v in enumerate(lit):
                 cj.in_list[m] = int(v)

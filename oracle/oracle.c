@@ -135,7 +135,12 @@ static int orc_row_passes(const OrcCol* cols, const BkQuerySpec* q, int64_t r) {
             cmp = (v > cj->lit_i) - (v < cj->lit_i);
         }
         int pass;
-        if (cj->op == BK_OP_IN || cj->op == BK_OP_NOT_IN) {
+        if (cj->op == BK_OP_IN_BITMAP || cj->op == BK_OP_NOT_IN_BITMAP) {
+            int64_t v = cell_i64(c, r);
+            const uint8_t* bm = (const uint8_t*)(uintptr_t)cj->lit_i;
+            int hit = v >= 0 && v < cj->n_in && ((bm[v >> 3] >> (v & 7)) & 1);
+            pass = (cj->op == BK_OP_IN_BITMAP) ? hit : !hit;
+        } else if (cj->op == BK_OP_IN || cj->op == BK_OP_NOT_IN) {
             /* predicate.h InPredicate semantics over literal lists */
             int64_t v = cell_i64(c, r);
             int found = 0;

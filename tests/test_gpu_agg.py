@@ -302,3 +302,64 @@ def test_in_predicates(eng, orc):
                         [(2, "in", [1, 2, 5, 77, 999]), (4, "not_in", [7, 8])],
                         [1], aggs)
     assert_parity(got, exp, aggs, [s[0] for s in BASE5])
+
+
+def test_like_pushdown_via_dict_bitmap(eng, orc):
+    """LIKE on a dictionary VARCHAR pushes down as a code-accept bitmap
+    (BK_OP_IN_BITMAP): the host matches the pattern against the dictionary
+    once, the engine filters by code membership (cstore-dict pushdown)."""
+    import fnmatch
+    import numpy as np
+    from baikaldb_amd import QueryPlan
+    from oracle.bindings import make_query as mq
+
+    n = 200_000
+    nwords = 512
+    specs = [(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+             (TYPE_STRING, D_DICT, nwords, 0, 0),
+             (TYPE_INT64, D_UNI, 0, 1000, 0)]
+    # dict words are deterministic from the seed: LIKE 'w%3_%' etc.
+    pattern = "w*1*"  # SQL: LIKE 'w%1%'
+    accept = bytearray((nwords + 7) // 8)
+    naccept = 0
+    for c in range(nwords):
+        if fnmatch.fnmatch(orc.dict_word(SEED, c), pattern):
+            accept[c >> 3] |= 1 << (c & 7)
+            naccept += 1
+    assert 0 < naccept < nwords
+    accept = bytes(accept)
+
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED)
+        dev_bm = eng.upload_bytes(accept)
+        try:
+            plan = QueryPlan(t.col_types,
+                             conjuncts=[(1, "in_bitmap", (dev_bm, nwords)),
+                                        (0, "<", int((1 << 31) * 0.7))],
+                             group=[1], aggs=[("count_star", -1), ("sum", 2)])
+            res = eng.filter_agg(t, plan, expected_groups=1 << 12)
+            got = res.fetch(sorted=True)
+            res.free()
+        finally:
+            eng.free_ptr(dev_bm)
+    finally:
+        t.free()
+
+    specs_c = (BkColSpec * len(specs))()
+    for i, s in enumerate(specs):
+        (specs_c[i].col_type, specs_c[i].dist, specs_c[i].p0, specs_c[i].p1,
+         specs_c[i].null_frac_x1e6) = s
+    cols, valids = orc.generate_table(list(specs_c), n, SEED)
+    types = [s[0] for s in specs]
+    import ctypes as Ct
+    host_bm = Ct.create_string_buffer(accept, len(accept))
+    q = mq([(1, 8, TYPE_INT64, (Ct.addressof(host_bm), nwords)),
+            (0, 4, TYPE_INT64, int((1 << 31) * 0.7))],
+           [1], [(0, -1), (2, 2)], types)
+    exp = orc.filter_agg(cols, valids, types, q, nthreads=4, dict_seed=SEED)
+    assert got["rows_passed"] == exp["rows_passed"]
+    assert got["ngroups"] == exp["ngroups"] == naccept or \
+        got["ngroups"] == exp["ngroups"]  # some codes may not occur in n rows
+    assert np.array_equal(got["enc"], exp["enc"])
+    assert np.array_equal(got["agg_i"], exp["agg_i"])
