@@ -638,7 +638,8 @@ __global__ void __launch_bounds__(256)
 k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
              uint32_t P, uint16_t* bucketid, uint32_t* H,
              uint64_t* gtable, uint64_t gmask, uint64_t fill_cap, uint64_t* fill,
-             uint64_t* rows_passed, uint32_t* err, uint32_t lds_slots) {
+             uint64_t* rows_passed, uint32_t* err, uint32_t lds_slots,
+             uint32_t lcap, uint32_t hot_probe, uint32_t hot_min) {
     extern __shared__ __attribute__((aligned(16))) uint64_t ltab[];
     const int stride = SLOT_HDR + 2 * q.n_aggs;
     uint64_t* laux = ltab + (size_t)lds_slots * stride;
@@ -656,8 +657,7 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
     const uint32_t lmask = lds_slots - 1;
     /* low load cap + short probes: a MISS must exit fast (it pays on every
      * cold row); hot keys the short probe skips just stay cold — correct,
-     * only less absorbed. */
-    const uint32_t lcap = lds_slots / 2u;
+     * only less absorbed. lcap/hot_probe/hot_min are host-tuned. */
     int64_t my_passed = 0;
     int64_t gstride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -681,12 +681,12 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
         uint32_t mode = lmode[0];
         if (mode != 1u) {
             uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
-                                          lfill, lcap, /*max_probe=*/4);
+                                          lfill, lcap, hot_probe);
             if (mode == 0u) {
                 uint32_t att = atomicAdd(&lctr[0], 1u);
                 if (slot) atomicAdd(&lctr[1], 1u);
                 if (att == 4095u)
-                    lmode[0] = (lctr[1] * 4u >= 4096u) ? 2u : 1u;
+                    lmode[0] = (lctr[1] >= hot_min) ? 2u : 1u;
             }
             if (slot) {
                 agg_update_slot<true>(slot, q, cols, r);
@@ -1572,15 +1572,31 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
 
     static const char* NAMES[] = {"histo", "totals", "scan", "offsets",
                                   "scatter", "part_agg"};
+    /* hot-table knobs (defaults from the config3/config2 sweeps; see
+     * profiles/ and DESIGN.md §6). lcap = keys the first-come table can
+     * hold; under octave-shaped key mass every doubling of lcap absorbs
+     * ~1/noct more of the stream, so bigger is better until the LDS carve
+     * costs histo occupancy. */
+    size_t hot_lds_cap = 50 * 1024;
+    if (const char* e = getenv("BK_HOT_LDS_KB")) hot_lds_cap = (size_t)atoi(e) * 1024;
     uint32_t hot_slots = 512;
-    while ((size_t)hot_slots * stride * 8 > 50 * 1024) hot_slots >>= 1;
+    if (const char* e = getenv("BK_HOT_SLOTS")) hot_slots = (uint32_t)atoi(e);
+    while ((size_t)hot_slots * stride * 8 > hot_lds_cap) hot_slots >>= 1;
+    uint32_t hot_cap = hot_slots / 2u;
+    if (const char* e = getenv("BK_HOT_CAP")) hot_cap = (uint32_t)atoi(e);
+    if (hot_cap > hot_slots - hot_slots / 8u) hot_cap = hot_slots - hot_slots / 8u;
+    uint32_t hot_probe = 4;
+    if (const char* e = getenv("BK_HOT_PROBE")) hot_probe = (uint32_t)atoi(e);
+    uint32_t hot_min = 1024;  /* lock on when >= hot_min hits per 4096 rows */
+    if (const char* e = getenv("BK_HOT_MIN")) hot_min = (uint32_t)atoi(e);
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     EvTimer tm;
     tm.record();
     hipLaunchKernelGGL(k_part_histo, dim3(nblocks), dim3(threads), histo_lds, 0,
                        dc, *q, row_begin, row_end, P, bucketid, H,
                        o->table, o->nslots - 1, (o->nslots * 7) / 8,
-                       o->ctrs, o->ctrs + 1, o->err, hot_slots);
+                       o->ctrs, o->ctrs + 1, o->err, hot_slots,
+                       hot_cap, hot_probe, hot_min);
     tm.record();
     hipLaunchKernelGGL(k_part_totals, dim3((P * nchunks + 255) / 256), dim3(256),
                        0, 0, H, nblocks, P, S, totals);
@@ -1591,6 +1607,13 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     PCHECK(hipGetLastError());
     uint64_t total = 0;
     PCHECK(hipMemcpy(&total, total_dev, 8, hipMemcpyDeviceToHost));
+    if (getenv("BK_DEBUG")) {
+        uint64_t passed = 0;
+        hipMemcpy(&passed, o->ctrs + 1, 8, hipMemcpyDeviceToHost);
+        fprintf(stderr, "[bkgpu] passed=%llu cold_records=%llu hot_absorbed=%.1f%%\n",
+                (unsigned long long)passed, (unsigned long long)total,
+                passed ? 100.0 * (double)(passed - total) / (double)passed : 0.0);
+    }
     if (total > 0)
         PCHECK(pool_alloc((void**)&rec, (size_t)total * lay.nwords * 8));
     hipLaunchKernelGGL(k_part_offsets, dim3((P * nchunks + 255) / 256), dim3(256),
