@@ -1075,7 +1075,8 @@ __device__ __forceinline__ bool wave_combine_update(
  * slots), the whole block flushes it into the global table (additive merge),
  * resets, and continues — so ANY group cardinality is handled with global
  * traffic proportional to #groups x generations, never to #rows. */
-__global__ void __launch_bounds__(256)
+template <int BS>
+__global__ void __launch_bounds__(BS)
 k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
            uint64_t chunk,
            uint64_t* gtable, uint64_t gmask, uint64_t fill_cap, uint64_t* fill,
@@ -1085,7 +1086,8 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
     uint64_t* laux = ltab + (size_t)lds_slots * stride;
     uint32_t* lfill = (uint32_t*)&laux[0];
     const uint32_t lmask = lds_slots - 1;
-    const uint32_t lcap = lds_slots - 256;  /* room for one tile of claims */
+    /* room for one tile of claims; keep at least half the table usable */
+    const uint32_t lcap = lds_slots > 2u * BS ? lds_slots - BS : lds_slots / 2u;
     /* fixed-size CHUNKS of the (bucket-sorted) record array, not buckets:
      * a hot bucket gets many workgroups, and a chunk still spans only 1-2
      * buckets' worth of distinct groups for the LDS table. */
@@ -1609,7 +1611,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     PCHECK(hipMemcpy(&total, total_dev, 8, hipMemcpyDeviceToHost));
     if (getenv("BK_DEBUG")) {
         uint64_t passed = 0;
-        hipMemcpy(&passed, o->ctrs + 1, 8, hipMemcpyDeviceToHost);
+        (void)hipMemcpy(&passed, o->ctrs + 1, 8, hipMemcpyDeviceToHost);
         fprintf(stderr, "[bkgpu] passed=%llu cold_records=%llu hot_absorbed=%.1f%%\n",
                 (unsigned long long)passed, (unsigned long long)total,
                 passed ? 100.0 * (double)(passed - total) / (double)passed : 0.0);
@@ -1649,7 +1651,15 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
         if (envC) chunk = (uint64_t)atoll(envC);
         uint64_t nchunks = (total + chunk - 1) / chunk;
         uint32_t grid = (uint32_t)std::min<uint64_t>(nchunks, 32768);
-        hipLaunchKernelGGL(k_part_agg, dim3(grid), dim3(threads), lds_bytes, 0,
+        /* block size: more waves per block hide LDS-atomic latency (the
+         * 135 KB table allows only 1 block/CU) */
+        int at = 256;
+        if (const char* e = getenv("BK_AGG_THREADS")) at = atoi(e);
+        auto kfn = k_part_agg<256>;
+        if (at == 512) kfn = k_part_agg<512>;
+        else if (at == 1024) kfn = k_part_agg<1024>;
+        else at = 256;
+        hipLaunchKernelGGL(kfn, dim3(grid), dim3(at), lds_bytes, 0,
                            *q, lay, rec, total, chunk,
                            o->table, o->nslots - 1, (o->nslots * 7) / 8,
                            o->ctrs, o->err, lds_slots);
