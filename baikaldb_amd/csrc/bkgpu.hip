@@ -634,7 +634,8 @@ struct RecLayout {
  * hot head of the distribution right here (no record, no later passes).
  * Cold rows get a bucket id + per-block histogram for the partition passes.
  * LDS carve: [ hot table: lds_slots*stride u64 | laux 2 u64 | lhist P u32 ]. */
-__global__ void __launch_bounds__(256)
+template <int BS>
+__global__ void __launch_bounds__(BS)
 k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
              uint32_t P, uint16_t* bucketid, uint32_t* H,
              uint64_t* gtable, uint64_t gmask, uint64_t fill_cap, uint64_t* fill,
@@ -807,7 +808,8 @@ __device__ __forceinline__ void scatter_store_rec(uint64_t* dst,
 /* pass 2: scatter AoS records into bucket-contiguous regions.
  * MUST run with the same grid/block shape as k_part_histo so each block
  * sees the same rows its H row was computed from. */
-__global__ void __launch_bounds__(256)
+template <int BS>
+__global__ void __launch_bounds__(BS)
 k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
                int64_t row_end, uint32_t P, const uint16_t* bucketid,
                const uint32_t* H, uint64_t* rec, uint64_t total, int paired) {
@@ -1549,7 +1551,14 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     int nblocks = 2048;
     const char* envB = getenv("BK_PART_BLOCKS");
     if (envB) nblocks = atoi(envB);
-    const int threads = 256;
+    /* histo+scatter MUST share grid AND block shape (H rows are per-block) */
+    int threads = 256;
+    if (const char* e = getenv("BK_PART_THREADS")) threads = atoi(e);
+    if (threads != 512 && threads != 1024) threads = 256;
+    auto histo_fn = k_part_histo<256>;
+    auto scat_fn = k_part_scatter<256>;
+    if (threads == 512) { histo_fn = k_part_histo<512>; scat_fn = k_part_scatter<512>; }
+    else if (threads == 1024) { histo_fn = k_part_histo<1024>; scat_fn = k_part_scatter<1024>; }
     DevCols dc = table_cols(t);
 
     uint16_t* bucketid = nullptr;
@@ -1594,7 +1603,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     EvTimer tm;
     tm.record();
-    hipLaunchKernelGGL(k_part_histo, dim3(nblocks), dim3(threads), histo_lds, 0,
+    hipLaunchKernelGGL(histo_fn, dim3(nblocks), dim3(threads), histo_lds, 0,
                        dc, *q, row_begin, row_end, P, bucketid, H,
                        o->table, o->nslots - 1, (o->nslots * 7) / 8,
                        o->ctrs, o->ctrs + 1, o->err, hot_slots,
@@ -1629,7 +1638,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
         size_t pair_lds = (size_t)P * lay.nwords * 8 + (size_t)P * 8;
         int paired = pair_lds <= 130 * 1024 && getenv("BK_PAIR") != nullptr;
         size_t sc_lds = paired ? pair_lds : (size_t)P * 8;
-        hipLaunchKernelGGL(k_part_scatter, dim3(nblocks), dim3(threads),
+        hipLaunchKernelGGL(scat_fn, dim3(nblocks), dim3(threads),
                            sc_lds, 0,
                            dc, *q, lay, row_begin, row_end, P, bucketid, H,
                            rec, total, paired);
@@ -1651,9 +1660,11 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
         if (envC) chunk = (uint64_t)atoll(envC);
         uint64_t nchunks = (total + chunk - 1) / chunk;
         uint32_t grid = (uint32_t)std::min<uint64_t>(nchunks, 32768);
-        /* block size: more waves per block hide LDS-atomic latency (the
-         * 135 KB table allows only 1 block/CU) */
-        int at = 256;
+        /* block size: the 135 KB table allows only 1 block/CU, so waves per
+         * BLOCK are the only latency-hiding lever. 1024 (16 waves/CU)
+         * measured 2.4x faster than 256 on both ~1M-group and 1000-group
+         * shapes (see profiles/; LDS-atomic chains dominate this kernel). */
+        int at = 1024;
         if (const char* e = getenv("BK_AGG_THREADS")) at = atoi(e);
         auto kfn = k_part_agg<256>;
         if (at == 512) kfn = k_part_agg<512>;
