@@ -78,6 +78,14 @@ typedef enum BkAggType {
     BK_AGG_AVG        = 3,
     BK_AGG_MIN        = 4,
     BK_AGG_MAX        = 5,
+    /* DISTINCT aggregates (reference "count_distinct"/"sum_distinct",
+     * agg_fn_call.cpp:35-78). Executed as the reference's planner rewrite
+     * (agg_node.cpp:247-258): level 1 groups by (group_keys + distinct col),
+     * level 2 rolls the dedup key up via bkgpu_agg_rollup. These types only
+     * appear in the LEVEL-2 (rollup) spec; their states are COUNT-/SUM-
+     * shaped, so export/merge/fetch treat them like COUNT/SUM. */
+    BK_AGG_COUNT_DISTINCT = 6,
+    BK_AGG_SUM_DISTINCT   = 7,
 } BkAggType;
 
 /* ---- synthetic column distributions (bench configs) ---- */

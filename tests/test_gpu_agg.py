@@ -363,3 +363,20 @@ def test_like_pushdown_via_dict_bitmap(eng, orc):
         got["ngroups"] == exp["ngroups"]  # some codes may not occur in n rows
     assert np.array_equal(got["enc"], exp["enc"])
     assert np.array_equal(got["agg_i"], exp["agg_i"])
+
+
+def test_string_minmax_dict_order(eng, orc):
+    """MIN/MAX over a dict-encoded VARCHAR column: integer min/max on dict
+    codes == string min/max because the dictionary encoding is order-
+    preserving (tests/test_dict_order.py pins the invariant; reference
+    semantics ExprValue::compare STRING = byte compare, expr_value.h:895-945).
+    """
+    specs = [(TYPE_INT64, 0, 0, 1 << 20, 0),     # group key
+             (TYPE_STRING, 2, 900, 0, 0),        # dict column, no NULLs
+             (TYPE_STRING, 2, 50, 0, 250_000)]   # dict column, 25% NULLs
+    aggs = [("count_star", -1), ("min", 1), ("max", 1),
+            ("min", 2), ("max", 2)]
+    got, exp = run_both(eng, orc, specs, 150_000,
+                        [(0, "<", 1 << 19)], [0], aggs,
+                        expected_groups=1 << 16)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
