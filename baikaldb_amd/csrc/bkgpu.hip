@@ -294,10 +294,12 @@ __device__ __forceinline__ void agg_merge_slot(uint64_t* dst, const uint64_t* sr
         switch (q.aggs[a].agg_type) {
             case BK_AGG_COUNT_STAR:
             case BK_AGG_COUNT:
+            case BK_AGG_COUNT_DISTINCT:  /* additive; ranks exchange at L1 */
                 if (sv) atomicAdd((unsigned long long*)val, (unsigned long long)sv);
                 break;
             case BK_AGG_SUM:
             case BK_AGG_AVG:
+            case BK_AGG_SUM_DISTINCT:
                 if (sc) {
                     if (q.agg_in_types[a] == BK_DOUBLE || q.aggs[a].agg_type == BK_AGG_AVG) {
                         double d; memcpy(&d, &sv, 8);
@@ -1213,6 +1215,103 @@ __global__ void k_merge_blob(BkQuerySpec q, const uint32_t* flags, const uint64_
 }
 
 /* ------------------------------------------------------------------ */
+/* DISTINCT rollup (the reference's multi-distinct planner rewrite,
+ * agg_node.cpp:247-258): level 1 grouped by (user group keys + distinct
+ * col), so each live level-1 slot is one DEDUPED (group, d) pair. This
+ * kernel folds level-1 slots into a level-2 table keyed by the user group
+ * keys alone: plain aggs merge additively (their level-1 states already
+ * merged per (g,d), and (g,d) partitions g), COUNT(DISTINCT d) adds 1 per
+ * non-null d, SUM(DISTINCT d) adds the decoded dedup key.              */
+/* ------------------------------------------------------------------ */
+
+struct SrcIdx { int32_t v[BK_MAX_AGGS]; };  /* level-1 agg index per level-2
+                                               agg; -1 = synthesize from d */
+
+__global__ void k_rollup(BkQuerySpec q2, int in_naggs, const uint64_t* in_table,
+                         uint64_t in_nslots, uint64_t* out, uint64_t omask,
+                         uint64_t fill_cap, uint64_t* fill, uint32_t* err,
+                         SrcIdx src) {
+    const int in_stride = SLOT_HDR + 2 * in_naggs;
+    const int stride = SLOT_HDR + 2 * q2.n_aggs;
+    uint64_t gs = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t sl = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         sl < in_nslots; sl += gs) {
+        const uint64_t* s = in_table + sl * in_stride;
+        if (((const uint32_t*)s)[0] != 2u) continue;
+        uint32_t f = ((const uint32_t*)s)[1];
+        uint64_t e_d;
+        int d_null;
+        uint64_t* g;
+        if (q2.n_group == 0) {
+            /* no user group keys: level-1 key IS d; out slot 0 pre-claimed */
+            e_d = s[1]; d_null = (f & 0x80u) != 0;
+            g = out;
+        } else {
+            e_d = s[2]; d_null = (f & 0x40u) != 0;
+            g = gtable_claim(out, omask, stride, f & 0x80u, s[1], 0,
+                             fill, fill_cap, err);
+            if (!g) return;
+        }
+        const uint64_t* st_in = s + SLOT_HDR;
+        for (int32_t a = 0; a < q2.n_aggs; a++) {
+            uint64_t* val = g + SLOT_HDR + 2 * a;
+            uint64_t* cnt = val + 1;
+            int at = q2.aggs[a].agg_type;
+            if (at == BK_AGG_COUNT_DISTINCT) {
+                if (!d_null) atomicAdd((unsigned long long*)val, 1ull);
+                continue;
+            }
+            if (at == BK_AGG_SUM_DISTINCT) {
+                if (!d_null) {
+                    if (q2.agg_in_types[a] == BK_DOUBLE)
+                        atomic_add_f64_global(val, bk_dec_f64(e_d));
+                    else
+                        atomicAdd((unsigned long long*)val,
+                                  (unsigned long long)(uint64_t)bk_dec_i64(e_d));
+                    atomicAdd((unsigned long long*)cnt, 1ull);
+                }
+                continue;
+            }
+            /* plain agg: additive merge of the matching level-1 state */
+            int sa = src.v[a];
+            uint64_t sv = st_in[2 * sa], sc = st_in[2 * sa + 1];
+            switch (at) {
+                case BK_AGG_COUNT_STAR:
+                case BK_AGG_COUNT:
+                    if (sv) atomicAdd((unsigned long long*)val,
+                                      (unsigned long long)sv);
+                    break;
+                case BK_AGG_SUM:
+                case BK_AGG_AVG:
+                    if (sc) {
+                        if (q2.agg_in_types[a] == BK_DOUBLE ||
+                            at == BK_AGG_AVG) {
+                            double d; memcpy(&d, &sv, 8);
+                            atomic_add_f64_global(val, d);
+                        } else {
+                            atomicAdd((unsigned long long*)val,
+                                      (unsigned long long)sv);
+                        }
+                        atomicAdd((unsigned long long*)cnt,
+                                  (unsigned long long)sc);
+                    }
+                    break;
+                case BK_AGG_MIN:
+                case BK_AGG_MAX:
+                    if (sc) {
+                        atomicMax((unsigned long long*)val,
+                                  (unsigned long long)sv);
+                        atomicAdd((unsigned long long*)cnt,
+                                  (unsigned long long)sc);
+                    }
+                    break;
+                default: break;
+            }
+        }
+    }
+}
+
+/* ------------------------------------------------------------------ */
 /* host: tables                                                        */
 /* ------------------------------------------------------------------ */
 
@@ -1785,6 +1884,86 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
     return o;
 }
 
+/* Fold a level-1 aggregate (grouped by user keys + distinct col) into the
+ * level-2 result (see k_rollup). q2 describes level 2: n_group = level-1
+ * n_group - 1, same leading group cols, aggs may use BK_AGG_COUNT_DISTINCT /
+ * BK_AGG_SUM_DISTINCT (src_idx[a] = -1) or reference a level-1 agg state by
+ * index. Cross-rank exchange for distinct queries must happen on the LEVEL-1
+ * blob (dedup collapses duplicate (g,d) pairs); level-2 results only merge
+ * with disjoint group sets. */
+extern "C" BkgAggOut* bkgpu_agg_rollup(const BkgAggOut* in, const BkQuerySpec* q2,
+                                       const int32_t* src_idx,
+                                       int64_t expected_groups) {
+    if (!in || !q2 || !src_idx) { set_err("rollup: null arg"); return nullptr; }
+    if (in->q.n_group != q2->n_group + 1 || q2->n_group > 1) {
+        set_err("rollup: level-1 must group by (user keys + distinct col), "
+                "user keys <= 1");
+        return nullptr;
+    }
+    if (q2->n_aggs < 1 || q2->n_aggs > BK_MAX_AGGS) {
+        set_err("rollup: bad n_aggs"); return nullptr;
+    }
+    SrcIdx src;
+    for (int a = 0; a < q2->n_aggs; a++) {
+        int at = q2->aggs[a].agg_type;
+        bool synth = at == BK_AGG_COUNT_DISTINCT || at == BK_AGG_SUM_DISTINCT;
+        if (at == BK_AGG_SUM_DISTINCT && q2->agg_in_types[a] == BK_STRING) {
+            set_err("rollup: SUM(DISTINCT string) unsupported"); return nullptr;
+        }
+        if (synth != (src_idx[a] < 0) ||
+            (!synth && src_idx[a] >= in->q.n_aggs)) {
+            set_err("rollup: bad src_idx"); return nullptr;
+        }
+        src.v[a] = src_idx[a];
+    }
+    /* ensure in->rows_passed / fill counters are settled */
+    if (agg_compact(const_cast<BkgAggOut*>(in)) != 0) return nullptr;
+    BkgAggOut* o = new BkgAggOut();
+    o->q = *q2;
+    int64_t nslots = next_pow2(std::max<int64_t>(expected_groups * 2, 1024));
+    for (int attempt = 0; attempt < 8; attempt++) {
+        if (agg_alloc(o, nslots) != 0) { bkgpu_agg_free(o); return nullptr; }
+        if (q2->n_group == 0) {
+            uint64_t hdr[3] = {2ull, 0, 0};
+            HIP_CHECK_NULL(hipMemcpy(o->table, hdr, 24, hipMemcpyHostToDevice));
+            uint64_t one = 1;
+            HIP_CHECK_NULL(hipMemcpy(o->ctrs, &one, 8, hipMemcpyHostToDevice));
+        }
+        static const char* NAMESR[] = {"rollup"};
+        EvTimer tm;
+        tm.record();
+        hipLaunchKernelGGL(k_rollup, dim3(1024), dim3(256), 0, 0,
+                           *q2, in->q.n_aggs, in->table, in->nslots,
+                           o->table, o->nslots - 1, (o->nslots * 7) / 8,
+                           o->ctrs, o->err, src);
+        tm.record();
+        hipError_t lerr = hipGetLastError();
+        if (lerr != hipSuccess) {
+            snprintf(g_err, sizeof g_err, "rollup launch: %s",
+                     hipGetErrorString(lerr));
+            bkgpu_agg_free(o);
+            return nullptr;
+        }
+        tm.finish(o, NAMESR);
+        uint32_t err_host = 0;
+        HIP_CHECK_NULL(hipMemcpy(&err_host, o->err, 4, hipMemcpyDeviceToHost));
+        if (err_host == 0) break;
+        agg_release_table(o);
+        nslots *= 4;
+        if (attempt == 7) {
+            set_err("rollup table overflow after retries");
+            bkgpu_agg_free(o);
+            return nullptr;
+        }
+    }
+    /* rows scanned are the level-1 rows (the rollup reads groups, not rows) */
+    uint64_t rp = (uint64_t)in->rows_passed;
+    HIP_CHECK_NULL(hipMemcpy(o->ctrs + 1, &rp, 8, hipMemcpyHostToDevice));
+    o->dirty = true;
+    if (agg_compact(o) != 0) { bkgpu_agg_free(o); return nullptr; }
+    return o;
+}
+
 extern "C" int bkgpu_agg_breakdown(const BkgAggOut* o, char* names, double* ms,
                                    int cap) {
     int n = o->n_kernels < cap ? o->n_kernels : cap;
@@ -1942,7 +2121,9 @@ extern "C" int64_t bkgpu_agg_fetch(BkgAggOut* o, int sorted, int64_t max_groups,
             switch (at) {
                 case BK_AGG_COUNT_STAR:
                 case BK_AGG_COUNT:
+                case BK_AGG_COUNT_DISTINCT:  /* COUNT-shaped state */
                     out_i[idx] = (int64_t)val; out_has[idx] = 1; break;
+                case BK_AGG_SUM_DISTINCT:    /* SUM-shaped state */
                 case BK_AGG_SUM:
                     if (!cnt) break;
                     if (vt == BK_DOUBLE) memcpy(&out_d[idx], &val, 8);

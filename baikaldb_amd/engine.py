@@ -56,6 +56,9 @@ def _load():
     lib.bkgpu_agg_export_bytes.argtypes = [C.c_void_p]
     lib.bkgpu_agg_export.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
     lib.bkgpu_agg_merge.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
+    lib.bkgpu_agg_rollup.restype = C.c_void_p
+    lib.bkgpu_agg_rollup.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
+                                     C.POINTER(C.c_int32), C.c_int64]
     lib.bkgpu_agg_fetch.restype = C.c_int64
     lib.bkgpu_agg_fetch.argtypes = [C.c_void_p, C.c_int, C.c_int64,
                                     C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
@@ -212,6 +215,25 @@ class GpuEngine:
                                       expected_groups)
         if not h:
             raise RuntimeError(f"filter_agg: {self.lib.bkgpu_last_error().decode()}")
+        return AggResult(self, h, plan)
+
+    def filter_agg_distinct(self, table, plan: QueryPlan, row_begin=0,
+                            row_end=None, expected_l1_groups=1 << 18,
+                            expected_groups=1 << 14):
+        """COUNT/SUM(DISTINCT d): level-1 filter_agg grouped by
+        (user keys + d), then bkgpu_agg_rollup (the reference's multi-
+        distinct planner rewrite, agg_node.cpp:247-258)."""
+        l1_plan, q2, src_idx = plan.split_distinct()
+        l1 = self.filter_agg(table, l1_plan, row_begin, row_end,
+                             expected_groups=expected_l1_groups)
+        try:
+            h = self.lib.bkgpu_agg_rollup(l1.handle, C.byref(q2), src_idx,
+                                          expected_groups)
+            if not h:
+                raise RuntimeError(
+                    f"agg_rollup: {self.lib.bkgpu_last_error().decode()}")
+        finally:
+            l1.free()
         return AggResult(self, h, plan)
 
     def sort_topk(self, table, order, limit, plan: QueryPlan = None,

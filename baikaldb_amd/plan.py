@@ -11,13 +11,16 @@ import ctypes as C
 TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
 OP_EQ, OP_NE, OP_GT, OP_GE, OP_LT, OP_LE = 0, 1, 2, 3, 4, 5
 AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_AVG, AGG_MIN, AGG_MAX = 0, 1, 2, 3, 4, 5
+AGG_COUNT_DISTINCT, AGG_SUM_DISTINCT = 6, 7
 BK_MAX_GROUP, BK_MAX_CONJ, BK_MAX_AGGS = 2, 8, 8
 
 _OPS = {"=": OP_EQ, "!=": OP_NE, ">": OP_GT, ">=": OP_GE, "<": OP_LT,
         "<=": OP_LE, "in": 6, "not_in": 7,
         "in_bitmap": 8, "not_in_bitmap": 9}
 _AGGS = {"count_star": AGG_COUNT_STAR, "count": AGG_COUNT, "sum": AGG_SUM,
-         "avg": AGG_AVG, "min": AGG_MIN, "max": AGG_MAX}
+         "avg": AGG_AVG, "min": AGG_MIN, "max": AGG_MAX,
+         "count_distinct": AGG_COUNT_DISTINCT,
+         "sum_distinct": AGG_SUM_DISTINCT}
 
 
 class BkConjunct(C.Structure):
@@ -93,3 +96,48 @@ class QueryPlan:
             q.aggs[i].col = col
             q.agg_in_types[i] = self.col_types[col] if col >= 0 else TYPE_INT64
         return q
+
+    def has_distinct(self):
+        return any((_AGGS[n] if isinstance(n, str) else n) >= AGG_COUNT_DISTINCT
+                   for n, _ in self.aggs)
+
+    def split_distinct(self):
+        """The reference planner's multi-distinct rewrite (agg_node.cpp:
+        247-258): return (level1_plan, level2_spec, src_idx) where level 1
+        groups by (user keys + distinct col) and carries the plain aggs, and
+        level 2 (bkgpu_agg_rollup / orc_filter_agg_distinct) folds the dedup
+        key back out. v1 envelope: one distinct column, <= 1 user group key.
+        """
+        dist = [(i, n, c) for i, (n, c) in enumerate(self.aggs)
+                if (_AGGS[n] if isinstance(n, str) else n) >= AGG_COUNT_DISTINCT]
+        dcols = {c for _, _, c in dist}
+        if len(dcols) != 1:
+            raise ValueError("exactly one DISTINCT column supported")
+        if len(self.group) > 1:
+            raise ValueError("DISTINCT aggs support <= 1 group key")
+        dcol = dcols.pop()
+        plain = [(n, c) for n, c in self.aggs
+                 if (_AGGS[n] if isinstance(n, str) else n) < AGG_COUNT_DISTINCT]
+        l1 = QueryPlan(self.col_types, conjuncts=self.conjuncts,
+                       group=self.group + [dcol],
+                       aggs=plain or [("count_star", -1)])
+        q2 = BkQuerySpec()
+        q2.n_conjuncts = 0
+        q2.n_group = len(self.group)
+        for i, col in enumerate(self.group):
+            q2.group_cols[i] = col
+            q2.group_types[i] = self.col_types[col]
+        q2.n_aggs = len(self.aggs)
+        src_idx = (C.c_int32 * len(self.aggs))()
+        next_plain = 0
+        for i, (name, col) in enumerate(self.aggs):
+            t = _AGGS[name] if isinstance(name, str) else name
+            q2.aggs[i].agg_type = t
+            q2.aggs[i].col = col
+            q2.agg_in_types[i] = self.col_types[col] if col >= 0 else TYPE_INT64
+            if t >= AGG_COUNT_DISTINCT:
+                src_idx[i] = -1
+            else:
+                src_idx[i] = next_plain
+                next_plain += 1
+        return l1, q2, src_idx

@@ -92,6 +92,16 @@ int  bkgpu_agg_export(const BkgAggOut* o, void* dst, int64_t cap);
  * AggFnCall::merge semantics (src/expr/agg_fn_call.cpp:781-830). */
 int  bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups);
 
+/* ---- DISTINCT rollup (reference planner rewrite, agg_node.cpp:247-258) ----
+ * Fold a level-1 aggregate grouped by (user group keys + distinct col) into
+ * the level-2 result grouped by the user keys alone. q2: n_group = level-1
+ * n_group - 1 (0 or 1), aggs may be BK_AGG_COUNT_DISTINCT/BK_AGG_SUM_DISTINCT
+ * (synthesized from the dedup key; src_idx[a] = -1) or plain aggs merged
+ * additively from the level-1 state at src_idx[a]. For multi-GPU, exchange
+ * LEVEL-1 blobs (bkgpu_agg_merge dedups (g,d) pairs) before rolling up. */
+BkgAggOut* bkgpu_agg_rollup(const BkgAggOut* in, const BkQuerySpec* q2,
+                            const int32_t* src_idx, int64_t expected_groups);
+
 /* ---- result fetch (finalize + emit) ----
  * Downloads up to max_groups finalized groups to host arrays (each sized by
  * caller: flags[n], enc[n*BK_MAX_GROUP], out_i/out_d/out_has[naggs*n]).

@@ -111,6 +111,11 @@ class Oracle:
         lib.orc_filter_agg.argtypes = [C.POINTER(_OrcCol), C.c_int,
                                        C.POINTER(BkQuerySpec), C.c_int64, C.c_int64,
                                        C.c_int, C.c_uint64, C.c_int]
+        lib.orc_filter_agg_distinct.restype = C.POINTER(_OrcAggResult)
+        lib.orc_filter_agg_distinct.argtypes = [
+            C.POINTER(_OrcCol), C.c_int, C.POINTER(BkQuerySpec),
+            C.POINTER(BkQuerySpec), C.POINTER(C.c_int32),
+            C.c_int64, C.c_int64, C.c_int, C.c_uint64, C.c_int]
         lib.orc_agg_result_free.argtypes = [C.POINTER(_OrcAggResult)]
         lib.orc_sort_topk.restype = C.c_int64
         lib.orc_sort_topk.argtypes = [C.POINTER(_OrcCol), C.c_int,
@@ -177,10 +182,27 @@ class Oracle:
         res = self.lib.orc_filter_agg(carr, len(cols), C.byref(q),
                                       row_begin, row_end, nthreads, dict_seed,
                                       1 if sort_keys else 0)
+        return self._unpack_agg(res, q.n_aggs)
+
+    def filter_agg_distinct(self, cols, valids, col_types, q1, q2, src_idx,
+                            nthreads=1, dict_seed=0, row_begin=0, row_end=None,
+                            sort_keys=True):
+        """COUNT/SUM(DISTINCT) via the planner rewrite; mirrors
+        bkgpu_agg_rollup (see oracle.c orc_filter_agg_distinct)."""
+        if row_end is None:
+            row_end = len(cols[0])
+        carr = self._make_cols(cols, valids, col_types)
+        res = self.lib.orc_filter_agg_distinct(
+            carr, len(cols), C.byref(q1), C.byref(q2), src_idx,
+            row_begin, row_end, nthreads, dict_seed, 1 if sort_keys else 0)
+        if not res:
+            raise RuntimeError("orc_filter_agg_distinct failed")
+        return self._unpack_agg(res, q2.n_aggs)
+
+    def _unpack_agg(self, res, na):
         try:
             r = res.contents
             ng = r.ngroups
-            na = q.n_aggs
             out = {
                 "ngroups": ng,
                 "rows_passed": r.rows_passed,

@@ -297,7 +297,9 @@ static inline void orc_agg_merge(OrcAggState* dst, const OrcAggState* src,
     switch (agg_type) {
         case BK_AGG_COUNT_STAR:
         case BK_AGG_COUNT:
+        case BK_AGG_COUNT_DISTINCT:
             if (src->has) { dst->i += src->i; dst->has = 1; } return;
+        case BK_AGG_SUM_DISTINCT:
         case BK_AGG_SUM:
             if (!src->has) return;
             if (vtype == BK_DOUBLE) {
@@ -450,6 +452,119 @@ static int64_t orc_build_key_bytes(const OrcGroup* g, const BkQuerySpec* q,
     return len;
 }
 
+static OrcAggResult* orc_materialize_parts(OrcMap* parts, int nparts,
+                                           const BkQuerySpec* q,
+                                           int64_t rows_passed,
+                                           uint64_t dict_seed, int sort_keys);
+
+/* COUNT/SUM(DISTINCT d) via the reference's planner rewrite
+ * (agg_node.cpp:247-258): level 1 = filter+aggregate grouped by (user group
+ * keys + d) — each surviving group is one deduped (g, d) pair — then fold
+ * into the level-2 result keyed by the user keys alone. Mirrors
+ * bkgpu_agg_rollup: src_idx[a] >= 0 merges level-1 agg state a additively,
+ * src_idx[a] < 0 synthesizes COUNT_DISTINCT/SUM_DISTINCT from the dedup key. */
+ORC_EXPORT OrcAggResult* orc_filter_agg_distinct(
+        const OrcCol* cols, int ncols, const BkQuerySpec* q1,
+        const BkQuerySpec* q2, const int32_t* src_idx,
+        int64_t row_begin, int64_t row_end, int nthreads,
+        uint64_t dict_seed, int sort_keys) {
+    (void)ncols;
+    if (q1->n_group != q2->n_group + 1 || q2->n_group > 1) return NULL;
+    int ng2 = q2->n_group;
+    if (nthreads < 1) nthreads = 1;
+    if (nthreads > 128) nthreads = 128;
+    int64_t n = row_end - row_begin;
+    if ((int64_t)nthreads > n && n > 0) nthreads = (int)n;
+    if (n <= 0) nthreads = 1;
+    OrcAggTask* tasks = (OrcAggTask*)calloc((size_t)nthreads, sizeof(OrcAggTask));
+    pthread_t* tids = (pthread_t*)calloc((size_t)nthreads, sizeof(pthread_t));
+    int64_t chunk = (n + nthreads - 1) / nthreads;
+    for (int t = 0; t < nthreads; t++) {
+        tasks[t].cols = cols; tasks[t].q = q1;
+        tasks[t].row_begin = row_begin + (int64_t)t * chunk;
+        tasks[t].row_end = tasks[t].row_begin + chunk;
+        if (tasks[t].row_end > row_end) tasks[t].row_end = row_end;
+        if (tasks[t].row_begin > row_end) tasks[t].row_begin = row_end;
+        if (nthreads == 1) orc_agg_worker(&tasks[t]);
+        else pthread_create(&tids[t], NULL, orc_agg_worker, &tasks[t]);
+    }
+    if (nthreads > 1)
+        for (int t = 0; t < nthreads; t++) pthread_join(tids[t], NULL);
+    int64_t rows_passed = 0;
+    for (int t = 0; t < nthreads; t++) rows_passed += tasks[t].rows_passed;
+
+    /* serial level-1 merge (MERGE_AGG over (g,d)) */
+    OrcMap l1;
+    orc_map_init(&l1, 4096);
+    for (int t = 0; t < nthreads; t++) {
+        OrcMap* mt = &tasks[t].map;
+        for (uint64_t i = 0; i < mt->cap; i++) {
+            if (!mt->hashes[i]) continue;
+            int created;
+            if (l1.n * 10 >= l1.cap * 6) orc_map_grow(&l1, q1->n_group);
+            OrcGroup* g = orc_map_find_or_insert(&l1, mt->groups[i].flag,
+                                                 mt->groups[i].e, q1->n_group,
+                                                 &created);
+            for (int32_t a = 0; a < q1->n_aggs; a++)
+                orc_agg_merge(&g->st[a], &mt->groups[i].st[a],
+                              q1->aggs[a].agg_type, q1->agg_in_types[a]);
+        }
+        orc_map_free(mt);
+    }
+    free(tasks); free(tids);
+
+    /* rollup */
+    OrcMap out;
+    orc_map_init(&out, 1024);
+    for (uint64_t i = 0; i < l1.cap; i++) {
+        if (!l1.hashes[i]) continue;
+        OrcGroup* gi = &l1.groups[i];
+        uint8_t f2 = ng2 ? (uint8_t)(gi->flag & 0x80u) : 0;
+        int d_null = (gi->flag >> (7 - ng2)) & 1;
+        uint64_t e_d = gi->e[ng2];
+        uint64_t e2[BK_MAX_GROUP] = {0, 0};
+        for (int k = 0; k < ng2; k++) e2[k] = gi->e[k];
+        int created;
+        if (out.n * 10 >= out.cap * 6) orc_map_grow(&out, ng2);
+        OrcGroup* g2 = orc_map_find_or_insert(&out, f2, e2, ng2, &created);
+        for (int32_t a = 0; a < q2->n_aggs; a++) {
+            int at = q2->aggs[a].agg_type;
+            int vt = q2->agg_in_types[a];
+            if (at == BK_AGG_COUNT_DISTINCT) {
+                orc_agg_update(&g2->st[a], BK_AGG_COUNT, vt, !d_null, 0, 0);
+            } else if (at == BK_AGG_SUM_DISTINCT) {
+                if (!d_null) {
+                    int64_t vi = 0; double vd = 0.0;
+                    if (vt == BK_DOUBLE) vd = bk_dec_f64(e_d);
+                    else vi = bk_dec_i64(e_d);
+                    orc_agg_update(&g2->st[a], BK_AGG_SUM, vt, 1, vi, vd);
+                }
+            } else {
+                orc_agg_merge(&g2->st[a], &gi->st[src_idx[a]], at, vt);
+            }
+        }
+    }
+    orc_map_free(&l1);
+
+    /* zero-row, no-GROUP-BY => one all-initialized row (agg_node.cpp:490-505) */
+    if (out.n == 0 && ng2 == 0) {
+        int created;
+        uint64_t e[BK_MAX_GROUP] = {0, 0};
+        OrcGroup* g = orc_map_find_or_insert(&out, 0, e, 0, &created);
+        for (int32_t a = 0; a < q2->n_aggs; a++) {
+            int at = q2->aggs[a].agg_type;
+            if (at == BK_AGG_COUNT_STAR || at == BK_AGG_COUNT ||
+                at == BK_AGG_COUNT_DISTINCT)
+                g->st[a].has = 1;
+        }
+    }
+
+    OrcAggResult* res = orc_materialize_parts(&out, 1, q2, rows_passed,
+                                              dict_seed, sort_keys);
+    orc_map_free(&out);
+    return res;
+}
+
 ORC_EXPORT void orc_agg_result_free(OrcAggResult* res) {
     if (!res) return;
     free(res->key_bytes); free(res->key_off);
@@ -469,7 +584,9 @@ static void orc_finalize_group(const OrcGroup* g, const BkQuerySpec* q,
         switch (at) {
             case BK_AGG_COUNT_STAR:
             case BK_AGG_COUNT:
+            case BK_AGG_COUNT_DISTINCT:  /* COUNT-shaped state */
                 res->out_i[idx] = s->i; res->out_has[idx] = 1; break;
+            case BK_AGG_SUM_DISTINCT:    /* SUM-shaped state */
             case BK_AGG_SUM:
                 if (!s->has) { res->out_has[idx] = 0; break; }
                 if (vtype == BK_DOUBLE) res->out_d[idx] = s->d;
@@ -518,6 +635,55 @@ static void* orc_merge_worker(void* arg) {
         }
     }
     return NULL;
+}
+
+/* collect groups from `parts`, canonical-sort, build the result arrays.
+ * Does NOT free parts. */
+static OrcAggResult* orc_materialize_parts(OrcMap* parts, int nparts,
+                                           const BkQuerySpec* q,
+                                           int64_t rows_passed,
+                                           uint64_t dict_seed, int sort_keys) {
+    uint64_t total_groups = 0;
+    for (int t = 0; t < nparts; t++) total_groups += parts[t].n;
+    int64_t ngroups = (int64_t)total_groups;
+    KeySortRef* refs = (KeySortRef*)malloc((size_t)(ngroups > 0 ? ngroups : 1)
+                                           * sizeof(KeySortRef));
+    int64_t gi = 0;
+    for (int t = 0; t < nparts; t++)
+        for (uint64_t i = 0; i < parts[t].cap; i++)
+            if (parts[t].hashes[i]) refs[gi++].g = &parts[t].groups[i];
+    if (sort_keys) {
+        s_sort_ng = q->n_group;
+        qsort(refs, (size_t)ngroups, sizeof(KeySortRef), orc_group_cmp);
+    }
+
+    OrcAggResult* res = (OrcAggResult*)calloc(1, sizeof(OrcAggResult));
+    res->ngroups = ngroups;
+    res->rows_passed = rows_passed;
+    res->key_off = (int64_t*)malloc((size_t)(ngroups + 1) * sizeof(int64_t));
+    int64_t total = 0;
+    for (int64_t g2 = 0; g2 < ngroups; g2++) {
+        res->key_off[g2] = total;
+        total += orc_build_key_bytes(refs[g2].g, q, dict_seed, NULL);
+    }
+    res->key_off[ngroups] = total;
+    res->key_bytes = (uint8_t*)malloc((size_t)(total > 0 ? total : 1));
+    int64_t na = (int64_t)q->n_aggs * (ngroups > 0 ? ngroups : 1);
+    res->out_i = (int64_t*)calloc((size_t)na, sizeof(int64_t));
+    res->out_d = (double*)calloc((size_t)na, sizeof(double));
+    res->out_has = (uint8_t*)calloc((size_t)na, 1);
+    res->g_flag = (uint8_t*)calloc((size_t)(ngroups > 0 ? ngroups : 1), 1);
+    res->g_enc = (uint64_t*)calloc((size_t)(ngroups > 0 ? ngroups : 1) * BK_MAX_GROUP,
+                                   sizeof(uint64_t));
+    for (int64_t g2 = 0; g2 < ngroups; g2++) {
+        orc_build_key_bytes(refs[g2].g, q, dict_seed, res->key_bytes + res->key_off[g2]);
+        orc_finalize_group(refs[g2].g, q, g2, ngroups, res);
+        res->g_flag[g2] = refs[g2].g->flag;
+        for (int k = 0; k < BK_MAX_GROUP; k++)
+            res->g_enc[g2 * BK_MAX_GROUP + k] = refs[g2].g->e[k];
+    }
+    free(refs);
+    return res;
 }
 
 /* Run filter+aggregate over [row_begin,row_end) with nthreads shards
@@ -591,47 +757,8 @@ ORC_EXPORT OrcAggResult* orc_filter_agg(const OrcCol* cols, int ncols,
         }
     }
 
-    /* collect + optional canonical sort */
-    total_groups = 0;
-    for (int t = 0; t < nmerge; t++) total_groups += parts[t].n;
-    int64_t ngroups = (int64_t)total_groups;
-    KeySortRef* refs = (KeySortRef*)malloc((size_t)(ngroups > 0 ? ngroups : 1)
-                                           * sizeof(KeySortRef));
-    int64_t gi = 0;
-    for (int t = 0; t < nmerge; t++)
-        for (uint64_t i = 0; i < parts[t].cap; i++)
-            if (parts[t].hashes[i]) refs[gi++].g = &parts[t].groups[i];
-    if (sort_keys) {
-        s_sort_ng = q->n_group;
-        qsort(refs, (size_t)ngroups, sizeof(KeySortRef), orc_group_cmp);
-    }
-
-    OrcAggResult* res = (OrcAggResult*)calloc(1, sizeof(OrcAggResult));
-    res->ngroups = ngroups;
-    res->rows_passed = rows_passed;
-    res->key_off = (int64_t*)malloc((size_t)(ngroups + 1) * sizeof(int64_t));
-    int64_t total = 0;
-    for (int64_t g2 = 0; g2 < ngroups; g2++) {
-        res->key_off[g2] = total;
-        total += orc_build_key_bytes(refs[g2].g, q, dict_seed, NULL);
-    }
-    res->key_off[ngroups] = total;
-    res->key_bytes = (uint8_t*)malloc((size_t)(total > 0 ? total : 1));
-    int64_t na = (int64_t)q->n_aggs * (ngroups > 0 ? ngroups : 1);
-    res->out_i = (int64_t*)calloc((size_t)na, sizeof(int64_t));
-    res->out_d = (double*)calloc((size_t)na, sizeof(double));
-    res->out_has = (uint8_t*)calloc((size_t)na, 1);
-    res->g_flag = (uint8_t*)calloc((size_t)(ngroups > 0 ? ngroups : 1), 1);
-    res->g_enc = (uint64_t*)calloc((size_t)(ngroups > 0 ? ngroups : 1) * BK_MAX_GROUP,
-                                   sizeof(uint64_t));
-    for (int64_t g2 = 0; g2 < ngroups; g2++) {
-        orc_build_key_bytes(refs[g2].g, q, dict_seed, res->key_bytes + res->key_off[g2]);
-        orc_finalize_group(refs[g2].g, q, g2, ngroups, res);
-        res->g_flag[g2] = refs[g2].g->flag;
-        for (int k = 0; k < BK_MAX_GROUP; k++)
-            res->g_enc[g2 * BK_MAX_GROUP + k] = refs[g2].g->e[k];
-    }
-    free(refs);
+    OrcAggResult* res = orc_materialize_parts(parts, nmerge, q, rows_passed,
+                                              dict_seed, sort_keys);
     for (int t = 0; t < nmerge; t++) orc_map_free(&parts[t]);
     free(parts);
     free(tasks); free(tids);

@@ -1,0 +1,250 @@
+# COUNT/SUM(DISTINCT) — the reference's multi-distinct planner rewrite
+# (agg_node.cpp:247-258) executed as level-1 GROUP BY (keys + d) plus the
+# bkgpu_agg_rollup / orc_filter_agg_distinct fold.
+#
+# The oracle itself is validated against a brute-force numpy recompute
+# (CPU test); the GPU path is validated against the oracle (gpu tests).
+import ctypes as C
+
+import numpy as np
+import pytest
+
+from tests.test_gpu_agg import assert_parity, SEED
+
+TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
+D_UNI, D_SKEW, D_DICT, D_SUM16, D_ZIPF = 0, 1, 2, 3, 4
+AGGMAP = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5,
+          "count_distinct": 6, "sum_distinct": 7}
+OPS = {"=": 0, "!=": 1, ">": 2, ">=": 3, "<": 4, "<=": 5}
+
+
+def oracle_distinct(orc, specs, n, conjuncts, group, aggs, seed=SEED,
+                    nthreads=4):
+    """Run the oracle's two-level distinct path; returns the fetch dict."""
+    from oracle.bindings import make_query, BkColSpec
+    from baikaldb_amd.plan import QueryPlan
+
+    arr = (BkColSpec * len(specs))()
+    for i, s in enumerate(specs):
+        (arr[i].col_type, arr[i].dist, arr[i].p0, arr[i].p1,
+         arr[i].null_frac_x1e6) = s
+    cols, valids = orc.generate_table(list(arr), n, seed)
+    col_types = [s[0] for s in specs]
+
+    plan = QueryPlan(col_types, conjuncts=conjuncts, group=group, aggs=aggs)
+    l1_plan, _, src_idx = plan.split_distinct()
+    oconj = []
+    for col, op, lit in conjuncts:
+        ct = TYPE_DOUBLE if isinstance(lit, float) else TYPE_INT64
+        oconj.append((col, OPS[op], ct, lit))
+    q1 = make_query(oconj, l1_plan.group,
+                    [(AGGMAP[a], c) for a, c in l1_plan.aggs], col_types)
+    q2 = make_query((), group, [(AGGMAP[a], c) for a, c in aggs], col_types)
+    exp = orc.filter_agg_distinct(cols, valids, col_types, q1, q2, src_idx,
+                                  nthreads=nthreads, dict_seed=seed)
+    return exp, (cols, valids, col_types)
+
+
+def brute_distinct(cols, valids, col_types, conjuncts, group, aggs):
+    """Numpy recompute (independent of both engine and oracle code paths)."""
+    n = len(cols[0])
+    mask = np.ones(n, dtype=bool)
+    for col, op, lit in conjuncts:
+        v = cols[col]
+        ok = {"=": v == lit, "!=": v != lit, "<": v < lit, "<=": v <= lit,
+              ">": v > lit, ">=": v >= lit}[op]
+        if valids[col] is not None:
+            ok = ok & (valids[col] != 0)
+        mask &= ok
+    idx = np.nonzero(mask)[0]
+    gcol = group[0] if group else None
+    if gcol is not None:
+        gvals = cols[gcol][idx]
+        gnull = (valids[gcol][idx] == 0) if valids[gcol] is not None \
+            else np.zeros(len(idx), bool)
+        keys = [(bool(gn), None if gn else int(gv))
+                for gn, gv in zip(gnull, gvals)]
+    else:
+        keys = [(False, None)] * len(idx)
+    out = {}
+    for i, k in zip(idx, keys):
+        out.setdefault(k, []).append(i)
+    result = {}
+    for k, rows in out.items():
+        rows = np.array(rows)
+        vals = {}
+        for a, (name, col) in enumerate(aggs):
+            if name == "count_star":
+                vals[a] = len(rows)
+                continue
+            v = cols[col][rows]
+            nn = (valids[col][rows] != 0) if valids[col] is not None \
+                else np.ones(len(rows), bool)
+            v = v[nn]
+            if name == "count_distinct":
+                vals[a] = len(np.unique(v))
+            elif name == "sum_distinct":
+                vals[a] = np.unique(v).sum()
+            elif name == "sum":
+                vals[a] = v.sum() if len(v) else None
+            elif name == "count":
+                vals[a] = len(v)
+        result[k] = vals
+    return result
+
+
+@pytest.fixture(scope="module")
+def orc():
+    from oracle import Oracle
+    return Oracle()
+
+
+def test_oracle_distinct_vs_brute(orc):
+    specs = [(TYPE_INT64, D_UNI, 0, 40, 0),          # group key
+             (TYPE_INT64, D_UNI, 0, 25, 200_000),    # distinct col, 20% NULL
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 0),
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 0)]
+    conj = [(3, "<", 1 << 30)]
+    aggs = [("count_star", -1), ("count_distinct", 1), ("sum_distinct", 1),
+            ("sum", 2), ("count", 1)]
+    exp, (cols, valids, col_types) = oracle_distinct(
+        orc, specs, 30_000, conj, [0], aggs)
+    brute = brute_distinct(cols, valids, col_types, conj, [0], aggs)
+    assert exp["ngroups"] == len(brute)
+    # oracle sorts canonically: nulls-first=flag order then encoded key
+    enc = exp["enc"][:, 0]
+    flags = exp["flags"]
+    for g in range(exp["ngroups"]):
+        if flags[g] & 0x80:
+            k = (True, None)
+        else:
+            k = (False, int(orc.lib.orc_decode_i64(C.c_uint64(int(enc[g])))))
+        b = brute[k]
+        assert exp["agg_i"][0][g] == b[0]                      # count_star
+        assert exp["agg_i"][1][g] == b[1], f"count_distinct group {k}"
+        assert exp["agg_i"][2][g] == b[2], f"sum_distinct group {k}"
+        if b[3] is None:
+            assert exp["agg_has"][3][g] == 0
+        else:
+            assert abs(exp["agg_d"][3][g] - b[3]) < 1e-9 * (abs(b[3]) + 1)
+        assert exp["agg_i"][4][g] == b[4]                      # count
+
+
+def test_oracle_distinct_no_group(orc):
+    specs = [(TYPE_INT64, D_UNI, 0, 1000, 100_000),
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 0)]
+    aggs = [("count_star", -1), ("count_distinct", 0)]
+    exp, (cols, valids, col_types) = oracle_distinct(
+        orc, specs, 20_000, [(1, "<", 1 << 29)], [], aggs)
+    brute = brute_distinct(cols, valids, col_types,
+                           [(1, "<", 1 << 29)], [], aggs)
+    assert exp["ngroups"] == 1
+    b = brute[(False, None)]
+    assert exp["agg_i"][0][0] == b[0]
+    assert exp["agg_i"][1][0] == b[1]
+
+
+def test_oracle_distinct_zero_rows(orc):
+    specs = [(TYPE_INT64, D_UNI, 0, 1000, 0)]
+    aggs = [("count_star", -1), ("count_distinct", 0)]
+    exp, _ = oracle_distinct(orc, specs, 1000, [(0, "<", -5)], [], aggs)
+    assert exp["ngroups"] == 1          # agg_node.cpp:490-505 single row
+    assert exp["agg_i"][0][0] == 0
+    assert exp["agg_i"][1][0] == 0
+
+
+# ---------------- GPU parity ----------------
+
+@pytest.fixture(scope="module")
+def eng():
+    import torch
+    if torch.cuda.is_available():
+        torch.cuda.init()
+    from baikaldb_amd import GpuEngine
+    return GpuEngine()
+
+
+def run_both_distinct(eng, orc, specs, n, conjuncts, group, aggs,
+                      seed=SEED, expected_groups=1 << 12):
+    from baikaldb_amd import QueryPlan
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, seed)
+        plan = QueryPlan(t.col_types, conjuncts=conjuncts, group=group,
+                         aggs=aggs)
+        res = eng.filter_agg_distinct(t, plan,
+                                      expected_l1_groups=1 << 16,
+                                      expected_groups=expected_groups)
+        try:
+            got = res.fetch(sorted=True)
+        finally:
+            res.free()
+    finally:
+        t.free()
+    exp, _ = oracle_distinct(orc, specs, n, conjuncts, group, aggs, seed=seed)
+    return got, exp
+
+
+def _parity_names(aggs):
+    """map distinct names onto their state-shaped plain kin for assert_parity"""
+    m = {"count_distinct": "count", "sum_distinct": "sum"}
+    return [(m.get(n, n), c) for n, c in aggs]
+
+
+@pytest.mark.gpu
+def test_gpu_count_distinct_grouped(eng, orc):
+    specs = [(TYPE_INT64, D_UNI, 0, 64, 0),
+             (TYPE_INT64, D_UNI, 0, 300, 150_000),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 0),
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 0)]
+    aggs = [("count_star", -1), ("count_distinct", 1), ("sum", 3),
+            ("avg", 2), ("sum_distinct", 1)]
+    got, exp = run_both_distinct(eng, orc, specs, 200_000,
+                                 [(3, "<", 1 << 30)], [0], aggs)
+    assert_parity(got, exp, _parity_names(aggs), [s[0] for s in specs])
+
+
+@pytest.mark.gpu
+def test_gpu_count_distinct_dict(eng, orc):
+    # distinct over a dict-encoded VARCHAR; group key nullable
+    specs = [(TYPE_INT64, D_UNI, 0, 40, 120_000),
+             (TYPE_STRING, D_DICT, 700, 0, 80_000)]
+    aggs = [("count_star", -1), ("count_distinct", 1)]
+    got, exp = run_both_distinct(eng, orc, specs, 150_000, [], [0], aggs)
+    assert_parity(got, exp, _parity_names(aggs), [s[0] for s in specs])
+
+
+@pytest.mark.gpu
+def test_gpu_distinct_double_no_group(eng, orc):
+    specs = [(TYPE_DOUBLE, D_SUM16, 0, 0, 100_000),
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 0)]
+    aggs = [("count_star", -1), ("count_distinct", 0), ("sum_distinct", 0)]
+    got, exp = run_both_distinct(eng, orc, specs, 120_000,
+                                 [(1, ">", 1 << 28)], [], aggs)
+    assert got["ngroups"] == exp["ngroups"] == 1
+    assert got["agg_i"][0][0] == exp["agg_i"][0][0]
+    assert got["agg_i"][1][0] == exp["agg_i"][1][0]
+    # sum over identical distinct sets: tolerance for reduction order
+    d = abs(got["agg_d"][2][0] - exp["agg_d"][2][0])
+    assert d <= 1e-9 * (abs(exp["agg_d"][2][0]) + 1)
+
+
+@pytest.mark.gpu
+def test_gpu_distinct_zero_rows(eng, orc):
+    specs = [(TYPE_INT64, D_UNI, 0, 1000, 0)]
+    aggs = [("count_star", -1), ("count_distinct", 0)]
+    got, exp = run_both_distinct(eng, orc, specs, 50_000,
+                                 [(0, "<", -1)], [], aggs)
+    assert got["ngroups"] == exp["ngroups"] == 1
+    assert got["agg_i"][0][0] == 0 and got["agg_i"][1][0] == 0
+
+
+@pytest.mark.gpu
+def test_gpu_distinct_high_cardinality(eng, orc):
+    # distinct cardinality near row count: level-1 table must regrow
+    specs = [(TYPE_INT64, D_UNI, 0, 16, 0),
+             (TYPE_INT64, D_UNI, 0, 1 << 40, 0)]
+    aggs = [("count_star", -1), ("count_distinct", 1)]
+    got, exp = run_both_distinct(eng, orc, specs, 400_000, [], [0], aggs,
+                                 expected_groups=64)
+    assert_parity(got, exp, _parity_names(aggs), [s[0] for s in specs])
