@@ -311,7 +311,8 @@ struct FetchedGroups {
 static int32_t agg_out_type(int agg_type, int32_t in_type) {
     switch (agg_type) {
         case BK_AGG_COUNT_STAR:
-        case BK_AGG_COUNT: return BK_INT64;
+        case BK_AGG_COUNT:
+        case BK_AGG_COUNT_DISTINCT: return BK_INT64;  /* agg_fn_call.cpp:96-99 */
         case BK_AGG_AVG:   return BK_DOUBLE;
         default:           return in_type;  /* SUM/MIN/MAX keep input type */
     }
@@ -354,7 +355,60 @@ public:
         }
         _q = q;
         int64_t expected = _desc.expected_groups > 0 ? _desc.expected_groups : 65536;
-        BkgAggOut* out = bkgpu_filter_agg(t, &q, 0, bkgpu_table_nrows(t), expected);
+        /* DISTINCT aggs: the reference's planner rewrite (agg_node.cpp:
+         * 247-258) — level 1 groups by (user keys + distinct col), then
+         * bkgpu_agg_rollup folds the dedup key out. */
+        bool has_distinct = false;
+        int dcol = -1;
+        for (int a = 0; a < q.n_aggs; a++) {
+            int at = q.aggs[a].agg_type;
+            if (at == BK_AGG_COUNT_DISTINCT || at == BK_AGG_SUM_DISTINCT) {
+                has_distinct = true;
+                if (dcol >= 0 && dcol != q.aggs[a].col) {
+                    state->error_msg = "one DISTINCT column supported";
+                    return -1;
+                }
+                dcol = q.aggs[a].col;
+            }
+        }
+        BkgAggOut* out = nullptr;
+        if (has_distinct) {
+            if (q.n_group > 1 || dcol < 0) {
+                state->error_msg = "DISTINCT aggs support <= 1 group key";
+                return -1;
+            }
+            BkQuerySpec q1 = q;
+            q1.n_group = q.n_group + 1;
+            q1.group_cols[q.n_group] = dcol;
+            q1.group_types[q.n_group] = bkgpu_table_col_type(t, dcol);
+            int32_t src_idx[BK_MAX_AGGS];
+            int np = 0;
+            q1.n_aggs = 0;
+            for (int a = 0; a < q.n_aggs; a++) {
+                int at = q.aggs[a].agg_type;
+                if (at == BK_AGG_COUNT_DISTINCT || at == BK_AGG_SUM_DISTINCT) {
+                    src_idx[a] = -1;
+                } else {
+                    q1.aggs[np] = q.aggs[a];
+                    q1.agg_in_types[np] = q.agg_in_types[a];
+                    src_idx[a] = np++;
+                }
+            }
+            if (np == 0) {  /* level 1 still needs one state column */
+                q1.aggs[0].agg_type = BK_AGG_COUNT_STAR;
+                q1.aggs[0].col = -1;
+                q1.agg_in_types[0] = BK_INT64;
+                np = 1;
+            }
+            q1.n_aggs = np;
+            BkgAggOut* l1 = bkgpu_filter_agg(t, &q1, 0, bkgpu_table_nrows(t),
+                                             expected * 16);
+            if (!l1) { state->error_msg = bkgpu_last_error(); return -1; }
+            out = bkgpu_agg_rollup(l1, &q, src_idx, expected);
+            bkgpu_agg_free(l1);
+        } else {
+            out = bkgpu_filter_agg(t, &q, 0, bkgpu_table_nrows(t), expected);
+        }
         if (!out) { state->error_msg = bkgpu_last_error(); return -1; }
         state->inc_num_scan_rows(bkgpu_table_nrows(t));
         state->inc_num_filter_rows(bkgpu_table_nrows(t) - bkgpu_agg_rows_passed(out));

@@ -188,3 +188,34 @@ def test_merge_agg_node_type(eng, orc):
     vals, counts = _np.unique(cols[1], return_counts=True)
     assert tags.shape[0] == len(vals)
     assert _np.array_equal(vi[:, 1], counts)
+
+
+def test_count_distinct_through_exec_surface(eng, orc):
+    """COUNT(DISTINCT c2) GROUP BY c1 through the ExecNode mirror: AggNode
+    applies the reference's multi-distinct rewrite internally
+    (agg_node.cpp:247-258 -> bkgpu_agg_rollup)."""
+    from baikaldb_amd import exec as bx
+    t, cols, valids, types = make_table(eng, orc, n=60_000)
+    try:
+        nodes = [bx.agg_node(group=[1], aggs=[("count_star", -1),
+                                              ("count_distinct", 2),
+                                              ("sum", 2)]),
+                 bx.filter_node(types, [(0, "<", int((1 << 31) * 0.5))]),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
+    import numpy as _np
+    idx = _np.nonzero(cols[0] < int((1 << 31) * 0.5))[0]
+    g = cols[1][idx]
+    d = cols[2][idx]
+    vals = _np.unique(g)
+    assert tags.shape[0] == len(vals)
+    for r, gv in enumerate(vals):
+        sel = d[g == gv]
+        assert vi[r, 1] == (g == gv).sum()          # count_star
+        assert vi[r, 2] == len(_np.unique(sel))     # count_distinct
+        assert vi[r, 3] == sel.sum()                # sum
