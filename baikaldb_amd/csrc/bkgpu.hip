@@ -122,7 +122,8 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
             const DevCol& c = cols.c[cj.col];                               \
             OK = cell_valid(c, r);                                          \
             if (cj.cmp_type == BK_DOUBLE) VD = cell_f64(c, r);              \
-            else VI = cell_i64(c, r);                                       \
+            else { VI = cell_i64(c, r);                                     \
+                   if (cj.fn) VI = bk_scalar_fn(cj.fn, VI); }               \
         }
     BK_EVAL1(0, vi0, vd0, ok0)
     BK_EVAL1(1, vi1, vd1, ok1)
@@ -171,11 +172,13 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
         bool pass;
         if (cj.op >= BK_OP_IN_BITMAP) {
             int64_t v = cell_i64(c, r);
+            if (cj.fn) v = bk_scalar_fn(cj.fn, v);
             const uint8_t* bm = (const uint8_t*)(uintptr_t)cj.lit_i;
             bool hit = v >= 0 && v < cj.n_in && ((bm[v >> 3] >> (v & 7)) & 1);
             pass = cj.op == BK_OP_IN_BITMAP ? hit : !hit;
         } else if (cj.op >= BK_OP_IN) {
             int64_t v = cell_i64(c, r);
+            if (cj.fn) v = bk_scalar_fn(cj.fn, v);
             bool found = false;
             for (int32_t m = 0; m < cj.n_in; m++)
                 found = found || (cj.in_list[m] == v);
@@ -187,6 +190,7 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
                 cmp = (v > cj.lit_d) - (v < cj.lit_d);
             } else {
                 int64_t v = cell_i64(c, r);
+                if (cj.fn) v = bk_scalar_fn(cj.fn, v);
                 cmp = (v > cj.lit_i) - (v < cj.lit_i);
             }
             switch (cj.op) {
@@ -1394,6 +1398,7 @@ extern "C" int bkgpu_sync(void) { HIP_CHECK(hipDeviceSynchronize()); return 0; }
 static size_t elem_size(int32_t t) {
     switch (t) {
         case BK_INT64: case BK_DOUBLE: return 8;
+        case BK_DATETIME: return 8;   /* packed u64 in an int64 column */
         case BK_STRING: return 4;
         default: return 0;
     }

@@ -70,6 +70,10 @@ def filter_node(col_types, conjuncts, num_children=1):
     d.n_conjuncts = len(conjuncts)
     for i, (col, op, lit) in enumerate(conjuncts):
         cj = d.conjuncts[i]
+        if isinstance(col, tuple):   # ("hour", col): pushed-down scalar fn
+            from .plan import _FNS
+            cj.fn = _FNS[col[0]]
+            col = col[1]
         cj.col = col
         cj.op = _OPS[op] if isinstance(op, str) else op
         if col_types[col] == TYPE_DOUBLE or isinstance(lit, float):

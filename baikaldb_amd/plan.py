@@ -8,7 +8,7 @@
 import ctypes as C
 
 # mirror include/bk_common.h (shared with oracle/bindings.py)
-TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
+TYPE_INT64, TYPE_DOUBLE, TYPE_STRING, TYPE_DATETIME = 6, 12, 13, 14
 OP_EQ, OP_NE, OP_GT, OP_GE, OP_LT, OP_LE = 0, 1, 2, 3, 4, 5
 AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_AVG, AGG_MIN, AGG_MAX = 0, 1, 2, 3, 4, 5
 AGG_COUNT_DISTINCT, AGG_SUM_DISTINCT = 6, 7
@@ -17,6 +17,8 @@ BK_MAX_GROUP, BK_MAX_CONJ, BK_MAX_AGGS = 2, 8, 8
 _OPS = {"=": OP_EQ, "!=": OP_NE, ">": OP_GT, ">=": OP_GE, "<": OP_LT,
         "<=": OP_LE, "in": 6, "not_in": 7,
         "in_bitmap": 8, "not_in_bitmap": 9}
+_FNS = {"year": 1, "month": 2, "day": 3, "dayofmonth": 3, "hour": 4,
+        "minute": 5, "second": 6}
 _AGGS = {"count_star": AGG_COUNT_STAR, "count": AGG_COUNT, "sum": AGG_SUM,
          "avg": AGG_AVG, "min": AGG_MIN, "max": AGG_MAX,
          "count_distinct": AGG_COUNT_DISTINCT,
@@ -27,7 +29,8 @@ class BkConjunct(C.Structure):
     _fields_ = [("col", C.c_int32), ("op", C.c_int32),
                 ("cmp_type", C.c_int32), ("n_in", C.c_int32),
                 ("lit_i", C.c_int64), ("lit_d", C.c_double),
-                ("in_list", C.c_int64 * 16)]
+                ("in_list", C.c_int64 * 16),
+                ("fn", C.c_int32), ("_pad2", C.c_int32)]
 
 
 class BkAggSpec(C.Structure):
@@ -69,6 +72,9 @@ class QueryPlan:
         q.n_conjuncts = len(self.conjuncts)
         for i, (col, op, lit) in enumerate(self.conjuncts):
             cj = q.conjuncts[i]
+            if isinstance(col, tuple):   # ("hour", col): pushed-down scalar fn
+                cj.fn = _FNS[col[0]]
+                col = col[1]
             cj.col = col
             cj.op = _OPS[op] if isinstance(op, str) else op
             ct = self.col_types[col]

@@ -32,6 +32,10 @@ typedef enum BkType {
     BK_FLOAT        = 11,
     BK_DOUBLE       = 12,
     BK_STRING       = 13,   /* stored dict-encoded: int32 codes + host dict */
+    BK_DATETIME     = 14,   /* MySQL-packed u64 in an int64 column
+                               (yearmonth<<46 | day<<41 | hour<<36 |
+                                minute<<30 | second<<24; reference
+                                include/common/datetime.h:35-68) */
 } BkType;
 
 /* ---- pb::PlanNodeType subset (proto/plan.proto:10-23) ---- */
@@ -97,6 +101,7 @@ typedef enum BkDist {
     BK_DIST_ZIPFOCT     = 4,  /* log-uniform ("Zipf-1-like") integer in [0, p0):
                                  octave picked uniformly, value uniform in octave
                                  => density ~ 1/(k+1); integer-only, CPU==GPU */
+    BK_DIST_DATETIME    = 5,  /* valid packed DATETIME in years [2019, 2026) */
 } BkDist;
 
 /* One generated column. Physical storage by type:
@@ -123,6 +128,20 @@ typedef struct BkColSpec {
  *   BK_STRING -> dict-code equality (EQ/NE only; dict codes are unique per
  *                string so code equality == string equality)
  */
+/* unary scalar fn applied to the column value before the compare — the
+ * pushed-down datetime extraction calls (reference internal_functions.cpp
+ * hour/minute/second/month/year/dayofmonth, fn_manager.cpp:219-230;
+ * bit layout datetime.h:35-45 + datetime.cpp:410-419). */
+typedef enum BkScalarFn {
+    BK_FN_NONE   = 0,
+    BK_FN_YEAR   = 1,
+    BK_FN_MONTH  = 2,
+    BK_FN_DAY    = 3,   /* dayofmonth */
+    BK_FN_HOUR   = 4,
+    BK_FN_MINUTE = 5,
+    BK_FN_SECOND = 6,
+} BkScalarFn;
+
 typedef struct BkConjunct {
     int32_t col;
     int32_t op;        /* BkCmpOp */
@@ -131,6 +150,8 @@ typedef struct BkConjunct {
     int64_t lit_i;
     double  lit_d;
     int64_t in_list[BK_MAX_INLIST];  /* int64 / dict-code IN literals */
+    int32_t fn;        /* BkScalarFn on the column value (int64 paths only) */
+    int32_t _pad2;
 } BkConjunct;
 
 /* One aggregate call (reference: src/expr/agg_fn_call.cpp:496-555 update,

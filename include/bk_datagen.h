@@ -86,6 +86,35 @@ BK_HD double bk_gen_sumu16(uint64_t u) {
     return (double)(s - 131070) * 2.6429099261197387e-05;
 }
 
+/* apply a BkScalarFn to a packed-DATETIME int64 (identity for BK_FN_NONE).
+ * Restates datetime.h:35-45 (year/month/day) and datetime.cpp:410-419 +
+ * internal_functions.cpp hour/minute/second (time bit fields). */
+BK_HD int64_t bk_scalar_fn(int32_t fn, int64_t v) {
+    uint64_t dt = (uint64_t)v;
+    switch (fn) {
+        case 1 /*YEAR*/:   return (int64_t)(((dt >> 46) & 0x1FFFFull) / 13u);
+        case 2 /*MONTH*/:  return (int64_t)(((dt >> 46) & 0x1FFFFull) % 13u);
+        case 3 /*DAY*/:    return (int64_t)((dt >> 41) & 0x1Full);
+        case 4 /*HOUR*/:   return (int64_t)((dt >> 36) & 0x1Full);
+        case 5 /*MINUTE*/: return (int64_t)((dt >> 30) & 0x3Full);
+        case 6 /*SECOND*/: return (int64_t)((dt >> 24) & 0x3Full);
+        default:           return v;
+    }
+}
+
+/* a valid packed DATETIME: year in [2019,2026), month 1-12, day 1-28,
+ * h/m/s uniform — deterministic, integer-only (CPU == GPU) */
+BK_HD int64_t bk_gen_datetime(uint64_t u) {
+    uint64_t year  = 2019u + (u & 0xFF) % 7u;
+    uint64_t month = 1u + ((u >> 8) & 0xFF) % 12u;
+    uint64_t day   = 1u + ((u >> 16) & 0xFF) % 28u;
+    uint64_t hour  = ((u >> 24) & 0xFF) % 24u;
+    uint64_t minu  = ((u >> 33) & 0xFF) % 60u;
+    uint64_t sec   = ((u >> 42) & 0xFF) % 60u;
+    return (int64_t)(((year * 13u + month) << 46) | (day << 41) |
+                     (hour << 36) | (minu << 30) | (sec << 24));
+}
+
 /* Generate one cell. Returns value through the matching out-param; the
  * caller dispatches storage by col_type. */
 BK_HD int64_t bk_gen_i64(const BkColSpec* cs, uint64_t seed, uint64_t row, uint32_t col) {
@@ -95,6 +124,7 @@ BK_HD int64_t bk_gen_i64(const BkColSpec* cs, uint64_t seed, uint64_t row, uint3
         case BK_DIST_CUBESKEW:    return bk_gen_cubeskew(u, cs->p0);
         case BK_DIST_DICT:        return (int64_t)bk_gen_dict(u, cs->p0);
         case BK_DIST_ZIPFOCT:     return bk_gen_zipfoct(u, cs->p0);
+        case BK_DIST_DATETIME:    return bk_gen_datetime(u);
         default:                  return 0;
     }
 }

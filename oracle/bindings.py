@@ -8,7 +8,7 @@ import numpy as np
 _HERE = os.path.dirname(os.path.abspath(__file__))
 
 # ---- enums mirroring include/bk_common.h ----
-TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
+TYPE_INT64, TYPE_DOUBLE, TYPE_STRING, TYPE_DATETIME = 6, 12, 13, 14
 DIST_UNIFORM, DIST_CUBESKEW, DIST_DICT, DIST_SUMU16 = 0, 1, 2, 3
 OP_EQ, OP_NE, OP_GT, OP_GE, OP_LT, OP_LE = 0, 1, 2, 3, 4, 5
 AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_AVG, AGG_MIN, AGG_MAX = 0, 1, 2, 3, 4, 5
@@ -28,7 +28,8 @@ class BkConjunct(C.Structure):
     _fields_ = [("col", C.c_int32), ("op", C.c_int32),
                 ("cmp_type", C.c_int32), ("n_in", C.c_int32),
                 ("lit_i", C.c_int64), ("lit_d", C.c_double),
-                ("in_list", C.c_int64 * 16)]
+                ("in_list", C.c_int64 * 16),
+                ("fn", C.c_int32), ("_pad2", C.c_int32)]
 
 
 class BkAggSpec(C.Structure):
@@ -65,16 +66,18 @@ class _OrcAggResult(C.Structure):
 def make_query(conjuncts=(), group=(), aggs=(), col_types=None):
     """Build a BkQuerySpec.
 
-    conjuncts: list of (col, op, cmp_type, literal)
+    conjuncts: list of (col, op, cmp_type, literal[, fn])
     group:     list of col indices
     aggs:      list of (agg_type, col)  (col=-1 for COUNT(*))
     col_types: list of BkType per table column (needed for group/agg typing)
     """
     q = BkQuerySpec()
     q.n_conjuncts = len(conjuncts)
-    for i, (col, op, cmp_type, lit) in enumerate(conjuncts):
+    for i, cjt in enumerate(conjuncts):
+        col, op, cmp_type, lit = cjt[:4]
         cj = q.conjuncts[i]
         cj.col, cj.op, cj.cmp_type = col, op, cmp_type
+        cj.fn = cjt[4] if len(cjt) > 4 else 0
         if op >= 8:  # bitmap membership: lit = (host_ptr, n_bits)
             cj.lit_i, cj.n_in = int(lit[0]), int(lit[1])
         elif op >= 6:  # IN / NOT IN: lit is a list
@@ -142,7 +145,7 @@ class Oracle:
         """Generate host columns. Returns (columns, valids) lists of numpy arrays."""
         cols, valids = [], []
         for ci, spec in enumerate(specs):
-            if spec.col_type == TYPE_INT64:
+            if spec.col_type in (TYPE_INT64, TYPE_DATETIME):
                 arr = np.empty(nrows, dtype=np.int64)
             elif spec.col_type == TYPE_DOUBLE:
                 arr = np.empty(nrows, dtype=np.float64)

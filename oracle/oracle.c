@@ -73,11 +73,12 @@ typedef struct OrcCol {
 ORC_EXPORT int orc_generate_column(const BkColSpec* cs, uint64_t seed, uint32_t col,
                                    int64_t row_begin, int64_t row_end,
                                    void* out_data, uint8_t* out_valid) {
-    if (cs->col_type == BK_INT64 || cs->col_type == BK_STRING) {
+    if (cs->col_type == BK_INT64 || cs->col_type == BK_STRING ||
+        cs->col_type == BK_DATETIME) {
         for (int64_t r = row_begin; r < row_end; r++) {
             int64_t v = bk_gen_i64(cs, seed, (uint64_t)r, col);
-            if (cs->col_type == BK_INT64) ((int64_t*)out_data)[r - row_begin] = v;
-            else                          ((int32_t*)out_data)[r - row_begin] = (int32_t)v;
+            if (cs->col_type == BK_STRING) ((int32_t*)out_data)[r - row_begin] = (int32_t)v;
+            else                           ((int64_t*)out_data)[r - row_begin] = v;
         }
     } else if (cs->col_type == BK_DOUBLE) {
         for (int64_t r = row_begin; r < row_end; r++) {
@@ -132,17 +133,20 @@ static int orc_row_passes(const OrcCol* cols, const BkQuerySpec* q, int64_t r) {
             cmp = (v > cj->lit_d) - (v < cj->lit_d);
         } else { /* BK_INT64 or BK_STRING dict-code compare */
             int64_t v = cell_i64(c, r);
+            if (cj->fn) v = bk_scalar_fn(cj->fn, v);
             cmp = (v > cj->lit_i) - (v < cj->lit_i);
         }
         int pass;
         if (cj->op == BK_OP_IN_BITMAP || cj->op == BK_OP_NOT_IN_BITMAP) {
             int64_t v = cell_i64(c, r);
+            if (cj->fn) v = bk_scalar_fn(cj->fn, v);
             const uint8_t* bm = (const uint8_t*)(uintptr_t)cj->lit_i;
             int hit = v >= 0 && v < cj->n_in && ((bm[v >> 3] >> (v & 7)) & 1);
             pass = (cj->op == BK_OP_IN_BITMAP) ? hit : !hit;
         } else if (cj->op == BK_OP_IN || cj->op == BK_OP_NOT_IN) {
             /* predicate.h InPredicate semantics over literal lists */
             int64_t v = cell_i64(c, r);
+            if (cj->fn) v = bk_scalar_fn(cj->fn, v);
             int found = 0;
             for (int32_t m = 0; m < cj->n_in; m++)
                 if (cj->in_list[m] == v) { found = 1; break; }
