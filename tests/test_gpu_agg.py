@@ -68,9 +68,15 @@ def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
     aggmap = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5}
     oconj = []
     for col, op, lit in conjuncts:
+        fn = 0
+        if isinstance(col, tuple):   # ("hour", col) scalar-fn pushdown
+            from baikaldb_amd.plan import _FNS
+            fn = _FNS[col[0]]
+            col = col[1]
         ct = TYPE_DOUBLE if (col_types[col] == TYPE_DOUBLE and
-                             not isinstance(lit, (list, tuple))) or             isinstance(lit, float) else TYPE_INT64
-        oconj.append((col, ops[op], ct, lit))
+                             not isinstance(lit, (list, tuple))) or \
+            isinstance(lit, float) else TYPE_INT64
+        oconj.append((col, ops[op], ct, lit, fn))
     q = make_query(oconj, group, [(aggmap[a], c) for a, c in aggs], col_types)
     exp = orc.filter_agg(cols, valids, col_types, q, nthreads=nthreads,
                          dict_seed=seed)
