@@ -1083,7 +1083,43 @@ __device__ __forceinline__ bool wave_combine_update(
  * slots), the whole block flushes it into the global table (additive merge),
  * resets, and continues — so ANY group cardinality is handled with global
  * traffic proportional to #groups x generations, never to #rows. */
-template <int BS>
+/* one progress attempt for one record: wave same-key fast path, else a
+ * per-lane LDS claim. Clears `pending` on success. */
+__device__ __forceinline__ void agg_try_one(
+        bool& pending, const uint64_t* my, uint64_t k0, uint64_t k1,
+        uint64_t meta, uint32_t flag, uint64_t* ltab, uint32_t lmask,
+        int stride, uint32_t* lfill, uint32_t lcap,
+        const BkQuerySpec& q, const RecLayout& lay) {
+    /* hot-bucket fast path: whole wave carries one group */
+    uint64_t wm = __ballot(pending);
+    if (wm == 0xFFFFFFFFFFFFFFFFull) {
+        uint64_t f0 = __shfl((unsigned long long)k0, 0, 64);
+        uint64_t f1 = __shfl((unsigned long long)k1, 0, 64);
+        uint32_t ff = (uint32_t)__shfl((int)flag, 0, 64);
+        if (__all(k0 == f0 && k1 == f1 && flag == ff)) {
+            uint64_t* slot = nullptr;
+            if ((threadIdx.x & 63) == 0)
+                slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
+                                    lfill, lcap);
+            int ok = __shfl((int)(slot != nullptr), 0, 64);
+            if (ok) {
+                wave_combine_update(slot, q, lay, my, meta, true);
+                pending = false;
+            }
+            return;  /* skip per-lane path this round */
+        }
+    }
+    if (pending) {
+        uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
+                                      lfill, lcap);
+        if (slot) {
+            agg_update_slot_rec<true>(slot, q, lay, my, meta);
+            pending = false;
+        }
+    }
+}
+
+template <int BS, int ILP>
 __global__ void __launch_bounds__(BS)
 k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
            uint64_t chunk,
@@ -1095,7 +1131,8 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
     uint32_t* lfill = (uint32_t*)&laux[0];
     const uint32_t lmask = lds_slots - 1;
     /* room for one tile of claims; keep at least half the table usable */
-    const uint32_t lcap = lds_slots > 2u * BS ? lds_slots - BS : lds_slots / 2u;
+    const uint32_t lcap = lds_slots > 2u * ILP * BS ? lds_slots - ILP * BS
+                                                    : lds_slots / 2u;
     /* fixed-size CHUNKS of the (bucket-sorted) record array, not buckets:
      * a hot bucket gets many workgroups, and a chunk still spans only 1-2
      * buckets' worth of distinct groups for the LDS table. */
@@ -1107,50 +1144,38 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
              w += blockDim.x)
             ltab[w] = 0;
         __syncthreads();
-        /* uniform tile loop so generation flushes can barrier */
-        for (uint32_t t0 = 0; t0 < n; t0 += blockDim.x) {
-            uint32_t i = t0 + threadIdx.x;
-            bool pending = i < n;
-            uint64_t pos = (uint64_t)b0 + i;
-            const uint64_t* my = rec + (size_t)pos * lay.nwords;
-            uint64_t k0 = 0, k1 = 0, meta = 0;
-            uint32_t flag = 0;
-            if (pending) {
-                k0 = my[0];
-                meta = lay.meta_word >= 0 ? my[lay.meta_word] : 0;
-                k1 = lay.k1_word >= 0 ? my[lay.k1_word]
-                     : (lay.k1_word == -2 ? (meta >> 32) : 0);
-                flag = (uint32_t)(meta & 0xFF);
+        /* uniform tile loop so generation flushes can barrier. ILP==2 keeps
+         * two independent records in flight per lane — two LDS-claim chains
+         * the scheduler can overlap (this kernel is claim-latency-bound). */
+        for (uint32_t t0 = 0; t0 < n; t0 += (uint32_t)ILP * blockDim.x) {
+            uint32_t iA = t0 + threadIdx.x;
+            uint32_t iB = ILP == 2 ? t0 + blockDim.x + threadIdx.x : n;
+            bool pA = iA < n, pB = iB < n;
+            const uint64_t* myA = rec + (size_t)(b0 + iA) * lay.nwords;
+            const uint64_t* myB = rec + (size_t)(b0 + iB) * lay.nwords;
+            uint64_t k0A = 0, k1A = 0, metaA = 0, k0B = 0, k1B = 0, metaB = 0;
+            uint32_t fA = 0, fB = 0;
+            if (pA) {
+                k0A = myA[0];
+                metaA = lay.meta_word >= 0 ? myA[lay.meta_word] : 0;
+                k1A = lay.k1_word >= 0 ? myA[lay.k1_word]
+                      : (lay.k1_word == -2 ? (metaA >> 32) : 0);
+                fA = (uint32_t)(metaA & 0xFF);
+            }
+            if (pB) {
+                k0B = myB[0];
+                metaB = lay.meta_word >= 0 ? myB[lay.meta_word] : 0;
+                k1B = lay.k1_word >= 0 ? myB[lay.k1_word]
+                      : (lay.k1_word == -2 ? (metaB >> 32) : 0);
+                fB = (uint32_t)(metaB & 0xFF);
             }
             for (;;) {
-                /* hot-bucket fast path: whole wave carries one group */
-                uint64_t wm = __ballot(pending);
-                if (wm == 0xFFFFFFFFFFFFFFFFull) {
-                    uint64_t f0 = __shfl((unsigned long long)k0, 0, 64);
-                    uint64_t f1 = __shfl((unsigned long long)k1, 0, 64);
-                    uint32_t ff = (uint32_t)__shfl((int)flag, 0, 64);
-                    if (__all(k0 == f0 && k1 == f1 && flag == ff)) {
-                        uint64_t* slot = nullptr;
-                        if ((threadIdx.x & 63) == 0)
-                            slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
-                                                lfill, lcap);
-                        int ok = __shfl((int)(slot != nullptr), 0, 64);
-                        if (ok) {
-                            wave_combine_update(slot, q, lay, my, meta, true);
-                            pending = false;
-                        }
-                        goto vote;  /* skip per-lane path this round */
-                    }
-                }
-                if (pending) {
-                    uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
-                                                  lfill, lcap);
-                    if (slot) {
-                        agg_update_slot_rec<true>(slot, q, lay, my, meta);
-                        pending = false;
-                    }
-                }
-vote:
+                agg_try_one(pA, myA, k0A, k1A, metaA, fA, ltab, lmask, stride,
+                            lfill, lcap, q, lay);
+                if (ILP == 2)
+                    agg_try_one(pB, myB, k0B, k1B, metaB, fB, ltab, lmask,
+                                stride, lfill, lcap, q, lay);
+                bool pending = pA || pB;
                 if (!__syncthreads_or((int)pending)) break;
                 /* generation flush: every thread participates */
                 for (uint32_t sl = threadIdx.x; sl < lds_slots; sl += blockDim.x) {
@@ -1159,7 +1184,7 @@ vote:
                     uint64_t* g = gtable_claim(gtable, gmask, stride,
                                                ((uint32_t*)s)[1], s[1], s[2],
                                                fill, fill_cap, err);
-                    if (!g) { pending = false; continue; }  /* err set; drain */
+                    if (!g) { pA = pB = false; continue; }  /* err set; drain */
                     agg_merge_slot<false>(g, s + SLOT_HDR, q);
                 }
                 __syncthreads();
@@ -1773,10 +1798,13 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
          * shapes (see profiles/; LDS-atomic chains dominate this kernel). */
         int at = 1024;
         if (const char* e = getenv("BK_AGG_THREADS")) at = atoi(e);
-        auto kfn = k_part_agg<256>;
-        if (at == 512) kfn = k_part_agg<512>;
-        else if (at == 1024) kfn = k_part_agg<1024>;
-        else at = 256;
+        int ilp = 1;
+        if (const char* e = getenv("BK_AGG_ILP")) ilp = atoi(e);
+        auto kfn = k_part_agg<256, 1>;
+        if (at == 512) kfn = ilp == 2 ? k_part_agg<512, 2> : k_part_agg<512, 1>;
+        else if (at == 1024) kfn = ilp == 2 ? k_part_agg<1024, 2>
+                                            : k_part_agg<1024, 1>;
+        else { at = 256; if (ilp == 2) kfn = k_part_agg<256, 2>; }
         hipLaunchKernelGGL(kfn, dim3(grid), dim3(at), lds_bytes, 0,
                            *q, lay, rec, total, chunk,
                            o->table, o->nslots - 1, (o->nslots * 7) / 8,
