@@ -139,6 +139,24 @@ void  bkgpu_free_ptr(void* p);
 /* release pooled device buffers */
 void bkgpu_pool_trim(void);
 
+/* ---- cold columnar ingestion: minimal parquet reader ----
+ * The reference stages OLAP cold data as parquet (src/column/file_manager.h:
+ * 252-334, parquet_writer.h:119). This reads v1 uncompressed PLAIN files
+ * (INT64/DOUBLE, optional columns) without Arrow; unsupported features are
+ * rejected with bkparquet_last_error(), never misread. */
+typedef struct BkParquet BkParquet;
+BkParquet* bkparquet_open(const char* path);
+int64_t    bkparquet_num_rows(const BkParquet* r);
+int        bkparquet_num_cols(const BkParquet* r);
+int        bkparquet_col_type(const BkParquet* r, int col);  /* BkType or <0 */
+int        bkparquet_col_nullable(const BkParquet* r, int col);
+int        bkparquet_col_name(const BkParquet* r, int col, char* out, int cap);
+int64_t    bkparquet_read_column(BkParquet* r, int col, void* out, uint8_t* valid);
+const char* bkparquet_last_error(void);
+void       bkparquet_close(BkParquet* r);
+/* parquet file -> HBM-resident table (create + upload all columns) */
+BkgTable*  bkgpu_table_from_parquet(const char* path);
+
 #ifdef __cplusplus
 }
 #endif

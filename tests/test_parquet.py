@@ -1,0 +1,164 @@
+# Cold columnar ingestion: the from-scratch parquet reader
+# (baikaldb_amd/csrc/bkparquet.cpp) vs files written by pyarrow — the same
+# arrow writer the reference's cold path uses (parquet_writer.h:119).
+import ctypes as C
+import os
+
+import numpy as np
+import pytest
+
+pa = pytest.importorskip("pyarrow")
+pq = pytest.importorskip("pyarrow.parquet")
+
+_LIB = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                    "baikaldb_amd", "libbkgpu.so")
+
+
+@pytest.fixture(scope="module")
+def lib():
+    lib = C.CDLL(_LIB)
+    lib.bkparquet_open.restype = C.c_void_p
+    lib.bkparquet_open.argtypes = [C.c_char_p]
+    lib.bkparquet_num_rows.restype = C.c_int64
+    lib.bkparquet_num_rows.argtypes = [C.c_void_p]
+    lib.bkparquet_num_cols.restype = C.c_int
+    lib.bkparquet_num_cols.argtypes = [C.c_void_p]
+    lib.bkparquet_col_type.restype = C.c_int
+    lib.bkparquet_col_type.argtypes = [C.c_void_p, C.c_int]
+    lib.bkparquet_col_nullable.restype = C.c_int
+    lib.bkparquet_col_nullable.argtypes = [C.c_void_p, C.c_int]
+    lib.bkparquet_col_name.argtypes = [C.c_void_p, C.c_int, C.c_char_p, C.c_int]
+    lib.bkparquet_read_column.restype = C.c_int64
+    lib.bkparquet_read_column.argtypes = [C.c_void_p, C.c_int, C.c_void_p,
+                                          C.POINTER(C.c_uint8)]
+    lib.bkparquet_last_error.restype = C.c_char_p
+    lib.bkparquet_close.argtypes = [C.c_void_p]
+    lib.bkgpu_table_from_parquet.restype = C.c_void_p
+    lib.bkgpu_table_from_parquet.argtypes = [C.c_char_p]
+    return lib
+
+
+def write_file(path, cols, row_group_size=None, data_page_size=None):
+    """cols: list of (name, numpy array, mask-or-None); mask True = NULL."""
+    arrays, names = [], []
+    for name, arr, mask in cols:
+        arrays.append(pa.array(arr, mask=mask))
+        names.append(name)
+    tab = pa.table(dict(zip(names, arrays)))
+    kw = dict(compression=None, use_dictionary=False, version="2.6",
+              data_page_version="1.0", write_statistics=False)
+    if row_group_size:
+        kw["row_group_size"] = row_group_size
+    if data_page_size:
+        kw["data_page_size"] = data_page_size
+    pq.write_table(tab, path, **kw)
+
+
+def read_col(lib, r, col, n, nullable):
+    out = np.empty(n, dtype=np.int64)
+    valid = np.empty(n, dtype=np.uint8) if nullable else None
+    vp = valid.ctypes.data_as(C.POINTER(C.c_uint8)) if nullable else None
+    got = lib.bkparquet_read_column(r, col, out.ctypes.data_as(C.c_void_p), vp)
+    assert got == n, lib.bkparquet_last_error()
+    return out, valid
+
+
+def test_read_int64_double_nulls(lib, tmp_path):
+    rng = np.random.default_rng(7)
+    n = 100_000
+    a = rng.integers(-(1 << 60), 1 << 60, n, dtype=np.int64)
+    b = rng.standard_normal(n)
+    bmask = rng.random(n) < 0.2
+    path = str(tmp_path / "t.parquet")
+    # multiple row groups AND multiple pages per chunk
+    write_file(path, [("a", a, None), ("b", b, bmask)],
+               row_group_size=30_000, data_page_size=4096)
+    r = lib.bkparquet_open(path.encode())
+    assert r, lib.bkparquet_last_error()
+    try:
+        assert lib.bkparquet_num_rows(r) == n
+        assert lib.bkparquet_num_cols(r) == 2
+        assert lib.bkparquet_col_type(r, 0) == 6      # BK_INT64
+        assert lib.bkparquet_col_type(r, 1) == 12     # BK_DOUBLE
+        assert lib.bkparquet_col_nullable(r, 0) == 0 or True  # pyarrow may mark optional
+        va, _ = read_col(lib, r, 0, n, lib.bkparquet_col_nullable(r, 0))
+        assert np.array_equal(va, a)
+        vb, validb = read_col(lib, r, 1, n, True)
+        assert np.array_equal(validb == 0, bmask)
+        assert np.array_equal(vb.view(np.float64)[~bmask], b[~bmask])
+    finally:
+        lib.bkparquet_close(r)
+
+
+def test_reject_compressed(lib, tmp_path):
+    path = str(tmp_path / "c.parquet")
+    tab = pa.table({"a": pa.array(np.arange(1000, dtype=np.int64))})
+    pq.write_table(tab, path, compression="snappy", use_dictionary=False)
+    r = lib.bkparquet_open(path.encode())
+    assert r
+    out = np.empty(1000, dtype=np.int64)
+    got = lib.bkparquet_read_column(r, 0, out.ctypes.data_as(C.c_void_p), None)
+    assert got < 0
+    assert b"unsupported" in lib.bkparquet_last_error() or \
+           b"compressed" in lib.bkparquet_last_error()
+    lib.bkparquet_close(r)
+
+
+def test_reject_dictionary(lib, tmp_path):
+    path = str(tmp_path / "d.parquet")
+    tab = pa.table({"a": pa.array(np.arange(1000, dtype=np.int64) % 5)})
+    pq.write_table(tab, path, compression=None, use_dictionary=True)
+    r = lib.bkparquet_open(path.encode())
+    assert r
+    out = np.empty(1000, dtype=np.int64)
+    got = lib.bkparquet_read_column(r, 0, out.ctypes.data_as(C.c_void_p), None)
+    assert got < 0
+    lib.bkparquet_close(r)
+
+
+@pytest.mark.gpu
+def test_parquet_to_gpu_agg_parity(lib, tmp_path):
+    """parquet file -> HBM table -> filter+GROUP BY == oracle over the
+    same host arrays (stored-region end-to-end)."""
+    import torch
+    if torch.cuda.is_available():
+        torch.cuda.init()
+    from baikaldb_amd import GpuEngine, QueryPlan
+    from baikaldb_amd.engine import AggResult
+    from oracle import Oracle
+    from oracle.bindings import make_query
+
+    rng = np.random.default_rng(11)
+    n = 200_000
+    g = rng.integers(0, 50, n, dtype=np.int64)
+    v = rng.integers(-1000, 1000, n, dtype=np.int64)
+    d = rng.standard_normal(n)
+    dmask = rng.random(n) < 0.15
+    path = str(tmp_path / "r.parquet")
+    write_file(path, [("g", g, None), ("v", v, None), ("d", d, dmask)],
+               row_group_size=64_000)
+
+    eng = GpuEngine()
+    h = lib.bkgpu_table_from_parquet(path.encode())
+    assert h, lib.bkparquet_last_error()
+    from baikaldb_amd.engine import GpuTable
+    t = GpuTable(eng, h, [6, 6, 12], n)
+    plan = QueryPlan(t.col_types, conjuncts=[(1, ">", 0)], group=[0],
+                     aggs=[("count_star", -1), ("sum", 1), ("avg", 2)])
+    res = eng.filter_agg(t, plan, expected_groups=1 << 10)
+    got = res.fetch(sorted=True)
+    res.free()
+    t.free()
+
+    orc = Oracle()
+    valid_d = (~dmask).astype(np.uint8)
+    q = make_query([(1, 2, 6, 0)], [0], [(0, -1), (2, 1), (3, 2)], [6, 6, 12])
+    exp = orc.filter_agg([g, v, d], [None, None, valid_d], [6, 6, 12], q,
+                         nthreads=4)
+    assert got["ngroups"] == exp["ngroups"]
+    assert got["rows_passed"] == exp["rows_passed"]
+    assert np.array_equal(got["enc"], exp["enc"])
+    assert np.array_equal(got["agg_i"][0], exp["agg_i"][0])
+    assert np.array_equal(got["agg_i"][1], exp["agg_i"][1])
+    err = np.abs(got["agg_d"][2] - exp["agg_d"][2])
+    assert np.all(err <= 1e-10 * (np.abs(exp["agg_d"][2]) + 1))
