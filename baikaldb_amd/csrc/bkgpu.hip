@@ -1350,7 +1350,33 @@ struct BkgTable {
     BkColSpec specs[BK_MAX_COLS];
     void* data[BK_MAX_COLS] = {};
     uint8_t* valid[BK_MAX_COLS] = {};
+    /* host-side dictionary for ingested BK_STRING columns (codes are
+     * order-preserving; see bkparquet.cpp). Null for generated tables,
+     * whose words come from bk_dict_word. */
+    std::vector<std::string>* dict[BK_MAX_COLS] = {};
 };
+
+/* attach an ingested column dictionary (concatenated words + offsets) */
+extern "C" int bkgpu_table_set_dict(BkgTable* t, int col, const char* concat,
+                                    const int64_t* offs, int64_t n) {
+    if (!t || col < 0 || col >= t->ncols) { set_err("set_dict: bad col"); return -1; }
+    delete t->dict[col];
+    auto* d = new std::vector<std::string>();
+    d->reserve((size_t)n);
+    for (int64_t i = 0; i < n; i++)
+        d->emplace_back(concat + offs[i], (size_t)(offs[i + 1] - offs[i]));
+    t->dict[col] = d;
+    return 0;
+}
+
+/* materialize a dict word of an ingested column (fetch-side, host) */
+extern "C" int bkgpu_table_dict_word(const BkgTable* t, int col, int64_t code,
+                                     char* out, int cap) {
+    if (!t || col < 0 || col >= t->ncols || !t->dict[col]) return -1;
+    const auto& d = *t->dict[col];
+    if (code < 0 || (size_t)code >= d.size()) return -1;
+    return snprintf(out, (size_t)cap, "%s", d[(size_t)code].c_str());
+}
 
 /* ---- device memory pool: query-scoped buffers (hash tables, partition
  * records, blobs) are reallocated every query at identical sizes; hipMalloc
@@ -1489,6 +1515,7 @@ extern "C" void bkgpu_table_free(BkgTable* t) {
     for (int c = 0; c < t->ncols; c++) {
         if (t->data[c]) (void)hipFree(t->data[c]);
         if (t->valid[c]) (void)hipFree(t->valid[c]);
+        delete t->dict[c];
     }
     delete t;
 }

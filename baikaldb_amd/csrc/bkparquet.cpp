@@ -17,7 +17,7 @@
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
-#include <string>
+#include <algorithm>\n#include <string>
 #include <vector>
 
 namespace bkparquet {
@@ -314,10 +314,54 @@ static bool parse_page_header(Cursor& c, PageHeader& ph) {
     return c.ok;
 }
 
-/* RLE/bit-packed hybrid definition levels, bit width 1 (max_def_level 1).
- * Layout (v1 page): i32 LE byte length, then runs:
+/* RLE/bit-packed hybrid decoder (parquet Encodings.md):
  *   header = varint; header & 1 ? bit-packed group of (header>>1)*8 values
- *                                : RLE run of (header>>1) copies of 1 value. */
+ *                                : RLE run of (header>>1) copies of 1 value
+ * (RLE value stored in ceil(width/8) bytes LE). Decodes into u32. */
+static bool rle_hybrid(const uint8_t*& p, const uint8_t* rend, int width,
+                       int64_t nvals, uint32_t* out) {
+    int vbytes = (width + 7) / 8;
+    int64_t i = 0;
+    Cursor c{p, rend};
+    while (i < nvals && c.p < rend && c.ok) {
+        uint64_t h = c.varint();
+        if (!c.ok) return false;
+        if (h & 1) {              /* bit-packed: (h>>1) groups of 8 values */
+            uint64_t groups = h >> 1;
+            uint64_t acc = 0;
+            int nbits = 0;
+            for (uint64_t g = 0; g < groups && i < nvals; g++) {
+                for (int k = 0; k < 8 && i < nvals; k++) {
+                    while (nbits < width) {
+                        if (c.p >= rend) return false;
+                        acc |= (uint64_t)(*c.p++) << nbits;
+                        nbits += 8;
+                    }
+                    out[i++] = (uint32_t)(acc & ((width == 32)
+                                   ? 0xFFFFFFFFull
+                                   : ((1ull << width) - 1)));
+                    acc >>= width;
+                    nbits -= width;
+                }
+                /* a full group consumes exactly width bytes; partial final
+                 * group already consumed what it needed via the loop */
+            }
+        } else {                  /* RLE run */
+            uint64_t run = h >> 1;
+            uint32_t v = 0;
+            for (int b = 0; b < vbytes; b++) {
+                if (c.p >= rend) return false;
+                v |= (uint32_t)(*c.p++) << (8 * b);
+            }
+            for (uint64_t r = 0; r < run && i < nvals; r++) out[i++] = v;
+        }
+    }
+    p = c.p;
+    return i == nvals;
+}
+
+/* definition levels, width 1 (flat optional column), v1 page layout:
+ * i32 LE byte length then RLE/bit-packed hybrid */
 static bool read_def_levels(Cursor& c, int32_t nvals, uint8_t* def) {
     if ((size_t)(c.end - c.p) < 4) return false;
     uint32_t len;
@@ -325,27 +369,12 @@ static bool read_def_levels(Cursor& c, int32_t nvals, uint8_t* def) {
     c.p += 4;
     const uint8_t* rend = c.p + len;
     if (rend > c.end) return false;
-    int32_t i = 0;
-    while (i < nvals && c.p < rend) {
-        uint64_t h = c.varint();
-        if (!c.ok) return false;
-        if (h & 1) {              /* bit-packed: (h>>1) groups of 8, width 1 */
-            uint64_t groups = h >> 1;
-            for (uint64_t g = 0; g < groups && i < nvals; g++) {
-                if (c.p >= rend) return false;
-                uint8_t byte = *c.p++;
-                for (int b = 0; b < 8 && i < nvals; b++)
-                    def[i++] = (byte >> b) & 1;
-            }
-        } else {                  /* RLE: h>>1 copies of one width-1 value */
-            uint64_t run = h >> 1;
-            if (c.p >= rend) return false;
-            uint8_t v = *c.p++ & 1;
-            for (uint64_t r = 0; r < run && i < nvals; r++) def[i++] = v;
-        }
-    }
+    std::vector<uint32_t> tmp((size_t)nvals);
+    const uint8_t* p = c.p;
+    if (!rle_hybrid(p, rend, 1, nvals, tmp.data())) return false;
+    for (int32_t i = 0; i < nvals; i++) def[i] = (uint8_t)(tmp[i] & 1);
     c.p = rend;
-    return i == nvals;
+    return true;
 }
 
 struct Reader {
@@ -388,25 +417,135 @@ static Reader* open_reader(const char* path) {
     return r;
 }
 
+/* decode the dictionary page of a chunk (PLAIN-encoded entries).
+ * For INT64/DOUBLE: 8-byte values. For BYTE_ARRAY: [u32 len][bytes]. */
+static bool read_dict_page(Reader* r, const ColChunk& cc, int ptype,
+                           std::vector<uint64_t>* vals,
+                           std::vector<std::string>* strs) {
+    Cursor c{r->buf.data() + cc.dict_page_offset,
+             r->buf.data() + r->buf.size()};
+    PageHeader ph;
+    if (!parse_page_header(c, ph)) { seterr("dict page header parse failed"); return false; }
+    if (ph.type != 2) { seterr("expected dictionary page"); return false; }
+    Cursor pc{c.p, c.p + ph.uncompressed_size};
+    /* DictionaryPageHeader num_values lives in field 7; our PageHeader parser
+     * keeps only data_page_header, so recover the count from entries. */
+    if (ptype == 6) {
+        while (pc.p + 4 <= pc.end) {
+            uint32_t len;
+            memcpy(&len, pc.p, 4);
+            pc.p += 4;
+            if ((size_t)(pc.end - pc.p) < len) { seterr("dict entry underrun"); return false; }
+            strs->emplace_back((const char*)pc.p, (size_t)len);
+            pc.p += len;
+        }
+    } else {
+        while (pc.p + 8 <= pc.end) {
+            uint64_t v;
+            memcpy(&v, pc.p, 8);
+            pc.p += 8;
+            vals->push_back(v);
+        }
+    }
+    return true;
+}
+
 /* read one leaf column across all row groups into host buffers.
- * out: int64/double array of meta.num_rows; valid: per-row 1/0 (may be null
- * for required columns). Returns rows read, < 0 on error. */
-static int64_t read_column(Reader* r, int col, void* out, uint8_t* valid) {
+ * INT64/DOUBLE -> out is int64/double[num_rows];
+ * BYTE_ARRAY   -> out is int32[num_rows] ORDER-PRESERVING dict codes and
+ *                 *dict_out receives the sorted word list (code c = c-th
+ *                 smallest string, so code order == byte order — the
+ *                 invariant the engine's string MIN/MAX/ORDER BY relies on).
+ * valid: per-row 1/0 (may be null for required columns).
+ * Returns rows read, < 0 on error. */
+static int64_t read_column(Reader* r, int col, void* out, uint8_t* valid,
+                           std::vector<std::string>* dict_out) {
     const FileMeta& fm = r->meta;
     if (col < 0 || (size_t)col >= fm.schema.size()) { seterr("bad column"); return -1; }
     const SchemaCol& sc = fm.schema[col];
-    if (sc.physical_type != 2 && sc.physical_type != 5) {
-        seterr("unsupported physical type (need INT64 or DOUBLE): col " +
+    int ptype = sc.physical_type;
+    if (ptype != 2 && ptype != 5 && ptype != 6) {
+        seterr("unsupported physical type (need INT64/DOUBLE/BYTE_ARRAY): col " +
                sc.name);
         return -1;
     }
+    bool is_str = ptype == 6;
+    if (is_str && !dict_out) { seterr("string column needs dict_out"); return -1; }
+
+    /* pass 1 for strings: collect the global word set to build the
+     * order-preserving code assignment across all row groups */
+    std::vector<std::string> words;   /* sorted unique words */
+    if (is_str) {
+        std::vector<std::string> all;
+        for (const RowGroup& rg : fm.groups) {
+            const ColChunk& cc = rg.cols[col];
+            if (cc.dict_page_offset >= 0) {
+                std::vector<uint64_t> dv;
+                std::vector<std::string> ds;
+                if (!read_dict_page(r, cc, ptype, &dv, &ds)) return -1;
+                for (auto& w : ds) all.push_back(std::move(w));
+            }
+        }
+        /* PLAIN (non-dict) string pages contribute words during pass 2; to
+         * keep one code space we scan them here too */
+        for (const RowGroup& rg : fm.groups) {
+            const ColChunk& cc = rg.cols[col];
+            if (cc.dict_page_offset >= 0) continue;
+            int64_t remaining = cc.num_values;
+            int64_t off = cc.data_page_offset;
+            while (remaining > 0) {
+                Cursor c{r->buf.data() + off, r->buf.data() + r->buf.size()};
+                PageHeader ph;
+                if (!parse_page_header(c, ph)) { seterr("page header parse failed"); return -1; }
+                const uint8_t* data = c.p;
+                off = (int64_t)(data - r->buf.data()) + ph.compressed_size;
+                if (ph.type != 0 || ph.encoding != 0) { seterr("unsupported string page"); return -1; }
+                Cursor pc{data, data + ph.uncompressed_size};
+                std::vector<uint8_t> def((size_t)ph.num_values, 1);
+                if (sc.optional && !read_def_levels(pc, ph.num_values, def.data()))
+                    { seterr("def level decode failed"); return -1; }
+                for (int32_t i = 0; i < ph.num_values; i++) {
+                    if (!def[(size_t)i]) continue;
+                    if ((size_t)(pc.end - pc.p) < 4) { seterr("string underrun"); return -1; }
+                    uint32_t len;
+                    memcpy(&len, pc.p, 4);
+                    pc.p += 4;
+                    if ((size_t)(pc.end - pc.p) < len) { seterr("string underrun"); return -1; }
+                    all.emplace_back((const char*)pc.p, (size_t)len);
+                    pc.p += len;
+                }
+                remaining -= ph.num_values;
+            }
+        }
+        std::sort(all.begin(), all.end());
+        all.erase(std::unique(all.begin(), all.end()), all.end());
+        words = std::move(all);
+        if (words.size() > (1u << 30)) { seterr("dictionary too large"); return -1; }
+    }
+    auto code_of = [&](const std::string& w) -> int32_t {
+        return (int32_t)(std::lower_bound(words.begin(), words.end(), w) -
+                         words.begin());
+    };
+
     int64_t row = 0;
     std::vector<uint8_t> def;
+    std::vector<uint32_t> idx;
     for (const RowGroup& rg : fm.groups) {
         if ((size_t)col >= rg.cols.size()) { seterr("row group missing column"); return -1; }
         const ColChunk& cc = rg.cols[col];
         if (cc.codec != 0) { seterr("compressed parquet unsupported (codec != UNCOMPRESSED)"); return -1; }
-        if (cc.dict_page_offset >= 0) { seterr("dictionary-encoded parquet unsupported (write with use_dictionary=False)"); return -1; }
+        /* chunk dictionary (original parquet code order, NOT our code space) */
+        std::vector<uint64_t> dictv;
+        std::vector<std::string> dicts;
+        std::vector<int32_t> remap;   /* parquet dict idx -> engine code */
+        if (cc.dict_page_offset >= 0) {
+            if (!read_dict_page(r, cc, ptype, &dictv, &dicts)) return -1;
+            if (is_str) {
+                remap.resize(dicts.size());
+                for (size_t i = 0; i < dicts.size(); i++)
+                    remap[i] = code_of(dicts[i]);
+            }
+        }
         int64_t remaining = cc.num_values;
         int64_t off = cc.data_page_offset;
         while (remaining > 0) {
@@ -416,8 +555,7 @@ static int64_t read_column(Reader* r, int col, void* out, uint8_t* valid) {
             if (!parse_page_header(c, ph)) { seterr("page header parse failed"); return -1; }
             const uint8_t* data = c.p;
             off = (int64_t)(data - r->buf.data()) + ph.compressed_size;
-            if (ph.type != 0) { seterr("unsupported page type (v2 pages / dict page)"); return -1; }
-            if (ph.encoding != 0) { seterr("unsupported page encoding (need PLAIN)"); return -1; }
+            if (ph.type != 0) { seterr("unsupported page type (need v1 data pages)"); return -1; }
             Cursor pc{data, data + ph.uncompressed_size};
             int32_t nv = ph.num_values;
             def.assign((size_t)nv, 1);
@@ -425,23 +563,73 @@ static int64_t read_column(Reader* r, int col, void* out, uint8_t* valid) {
                 if (ph.def_encoding != 3) { seterr("def levels must be RLE"); return -1; }
                 if (!read_def_levels(pc, nv, def.data())) { seterr("def level decode failed"); return -1; }
             }
-            size_t esz = 8;
-            for (int32_t i = 0; i < nv; i++) {
-                if (row >= fm.num_rows) { seterr("row overflow"); return -1; }
-                if (def[i]) {
-                    if ((size_t)(pc.end - pc.p) < esz) { seterr("page data underrun"); return -1; }
-                    memcpy((uint8_t*)out + (size_t)row * esz, pc.p, esz);
-                    pc.p += esz;
-                    if (valid) valid[row] = 1;
-                } else {
-                    memset((uint8_t*)out + (size_t)row * esz, 0, esz);
-                    if (valid) valid[row] = 0;
+            int64_t npresent = 0;
+            for (int32_t i = 0; i < nv; i++) npresent += def[(size_t)i];
+            if (ph.encoding == 2 || ph.encoding == 8) {
+                /* PLAIN_DICTIONARY / RLE_DICTIONARY indices */
+                if (cc.dict_page_offset < 0) { seterr("dict-encoded page without dict"); return -1; }
+                if (pc.p >= pc.end) { seterr("missing index bit width"); return -1; }
+                int width = *pc.p++;
+                idx.assign((size_t)npresent, 0);
+                const uint8_t* p = pc.p;
+                if (width > 0) {
+                    if (!rle_hybrid(p, pc.end, width, npresent, idx.data()))
+                        { seterr("index decode failed"); return -1; }
                 }
-                row++;
+                int64_t k = 0;
+                for (int32_t i = 0; i < nv; i++) {
+                    if (row >= fm.num_rows) { seterr("row overflow"); return -1; }
+                    if (def[(size_t)i]) {
+                        uint32_t ix = idx[(size_t)k++];
+                        if (is_str) {
+                            if (ix >= remap.size()) { seterr("dict index out of range"); return -1; }
+                            ((int32_t*)out)[row] = remap[ix];
+                        } else {
+                            if (ix >= dictv.size()) { seterr("dict index out of range"); return -1; }
+                            memcpy((uint8_t*)out + (size_t)row * 8, &dictv[ix], 8);
+                        }
+                        if (valid) valid[row] = 1;
+                    } else {
+                        if (is_str) ((int32_t*)out)[row] = 0;
+                        else memset((uint8_t*)out + (size_t)row * 8, 0, 8);
+                        if (valid) valid[row] = 0;
+                    }
+                    row++;
+                }
+            } else if (ph.encoding == 0) {
+                for (int32_t i = 0; i < nv; i++) {
+                    if (row >= fm.num_rows) { seterr("row overflow"); return -1; }
+                    if (def[(size_t)i]) {
+                        if (is_str) {
+                            uint32_t len;
+                            if ((size_t)(pc.end - pc.p) < 4) { seterr("string underrun"); return -1; }
+                            memcpy(&len, pc.p, 4);
+                            pc.p += 4;
+                            if ((size_t)(pc.end - pc.p) < len) { seterr("string underrun"); return -1; }
+                            ((int32_t*)out)[row] =
+                                code_of(std::string((const char*)pc.p, (size_t)len));
+                            pc.p += len;
+                        } else {
+                            if ((size_t)(pc.end - pc.p) < 8) { seterr("page data underrun"); return -1; }
+                            memcpy((uint8_t*)out + (size_t)row * 8, pc.p, 8);
+                            pc.p += 8;
+                        }
+                        if (valid) valid[row] = 1;
+                    } else {
+                        if (is_str) ((int32_t*)out)[row] = 0;
+                        else memset((uint8_t*)out + (size_t)row * 8, 0, 8);
+                        if (valid) valid[row] = 0;
+                    }
+                    row++;
+                }
+            } else {
+                seterr("unsupported page encoding");
+                return -1;
             }
             remaining -= nv;
         }
     }
+    if (is_str) *dict_out = std::move(words);
     return row;
 }
 
@@ -464,12 +652,13 @@ int bkparquet_num_cols(const BkParquet* r) {
     return (int)((const bkparquet::Reader*)r)->meta.schema.size();
 }
 
-/* BkType of a column: 6 (BK_INT64) or 12 (BK_DOUBLE); <0 unsupported */
+/* BkType of a column: 6 INT64, 12 DOUBLE, 13 STRING(dict); <0 unsupported */
 int bkparquet_col_type(const BkParquet* r, int col) {
     const auto& s = ((const bkparquet::Reader*)r)->meta.schema;
     if (col < 0 || (size_t)col >= s.size()) return -1;
     if (s[col].physical_type == 2) return 6;
     if (s[col].physical_type == 5) return 12;
+    if (s[col].physical_type == 6) return 13;
     return -2;
 }
 
@@ -488,7 +677,33 @@ int bkparquet_col_name(const BkParquet* r, int col, char* out, int cap) {
 /* read a whole column to host buffers (out sized num_rows * 8 bytes; valid
  * sized num_rows or NULL for required columns). Returns rows read, <0 err. */
 int64_t bkparquet_read_column(BkParquet* r, int col, void* out, uint8_t* valid) {
-    return bkparquet::read_column((bkparquet::Reader*)r, col, out, valid);
+    return bkparquet::read_column((bkparquet::Reader*)r, col, out, valid,
+                                  nullptr);
+}
+
+/* BYTE_ARRAY column -> int32 ORDER-PRESERVING dict codes + the dictionary.
+ * *dict_handle receives an opaque handle (free with bkparquet_dict_free);
+ * *dict_n its entry count. */
+int64_t bkparquet_read_string_column(BkParquet* r, int col, int32_t* codes,
+                                     uint8_t* valid, void** dict_handle,
+                                     int64_t* dict_n) {
+    auto* d = new std::vector<std::string>();
+    int64_t got = bkparquet::read_column((bkparquet::Reader*)r, col, codes,
+                                         valid, d);
+    if (got < 0) { delete d; return got; }
+    if (dict_n) *dict_n = (int64_t)d->size();
+    if (dict_handle) *dict_handle = d; else delete d;
+    return got;
+}
+
+int bkparquet_dict_word(void* dict_handle, int64_t code, char* out, int cap) {
+    auto* d = (std::vector<std::string>*)dict_handle;
+    if (code < 0 || (size_t)code >= d->size()) return -1;
+    return snprintf(out, (size_t)cap, "%s", (*d)[(size_t)code].c_str());
+}
+
+void bkparquet_dict_free(void* dict_handle) {
+    delete (std::vector<std::string>*)dict_handle;
 }
 
 const char* bkparquet_last_error(void) { return bkparquet::g_err.c_str(); }
@@ -505,6 +720,9 @@ extern "C" void* bkgpu_table_create(int ncols, const void* specs, int64_t nrows)
 extern "C" int bkgpu_table_upload(void* t, int col, const void* data,
                                   const uint8_t* valid);
 extern "C" void bkgpu_table_free(void* t);
+
+extern "C" int bkgpu_table_set_dict(void* t, int col, const char* concat,
+                                    const int64_t* offs, int64_t n);
 
 extern "C" void* bkgpu_table_from_parquet(const char* path) {
     BkParquet* r = bkparquet_open(path);
@@ -532,8 +750,30 @@ extern "C" void* bkgpu_table_from_parquet(const char* path) {
             valid.assign((size_t)nrows, 1);
             vp = valid.data();
         }
-        if (bkparquet_read_column(r, c, data.data(), vp) != nrows ||
-            bkgpu_table_upload(tab, c, data.data(), vp) != 0) {
+        bool ok;
+        if (bkparquet_col_type(r, c) == 13) {
+            void* dh = nullptr;
+            int64_t dn = 0;
+            ok = bkparquet_read_string_column(r, c, (int32_t*)data.data(), vp,
+                                              &dh, &dn) == nrows;
+            if (ok) ok = bkgpu_table_upload(tab, c, data.data(), vp) == 0;
+            if (ok && dh) {
+                auto* d = (std::vector<std::string>*)dh;
+                std::string concat;
+                std::vector<int64_t> offs(d->size() + 1, 0);
+                for (size_t i = 0; i < d->size(); i++) {
+                    concat += (*d)[i];
+                    offs[i + 1] = (int64_t)concat.size();
+                }
+                ok = bkgpu_table_set_dict(tab, c, concat.data(), offs.data(),
+                                          (int64_t)d->size()) == 0;
+            }
+            if (dh) bkparquet_dict_free(dh);
+        } else {
+            ok = bkparquet_read_column(r, c, data.data(), vp) == nrows &&
+                 bkgpu_table_upload(tab, c, data.data(), vp) == 0;
+        }
+        if (!ok) {
             bkgpu_table_free(tab);
             bkparquet_close(r);
             return nullptr;

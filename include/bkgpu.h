@@ -152,10 +152,22 @@ int        bkparquet_col_type(const BkParquet* r, int col);  /* BkType or <0 */
 int        bkparquet_col_nullable(const BkParquet* r, int col);
 int        bkparquet_col_name(const BkParquet* r, int col, char* out, int cap);
 int64_t    bkparquet_read_column(BkParquet* r, int col, void* out, uint8_t* valid);
+/* BYTE_ARRAY column -> int32 ORDER-PRESERVING dict codes (code order ==
+ * byte order of the words: the invariant string MIN/MAX/ORDER BY rely on) */
+int64_t    bkparquet_read_string_column(BkParquet* r, int col, int32_t* codes,
+                                        uint8_t* valid, void** dict_handle,
+                                        int64_t* dict_n);
+int        bkparquet_dict_word(void* dict_handle, int64_t code, char* out, int cap);
+void       bkparquet_dict_free(void* dict_handle);
 const char* bkparquet_last_error(void);
 void       bkparquet_close(BkParquet* r);
-/* parquet file -> HBM-resident table (create + upload all columns) */
+/* parquet file -> HBM-resident table (create + upload all columns; string
+ * columns keep their dictionary on the table, bkgpu_table_dict_word) */
 BkgTable*  bkgpu_table_from_parquet(const char* path);
+int        bkgpu_table_set_dict(BkgTable* t, int col, const char* concat,
+                                const int64_t* offs, int64_t n);
+int        bkgpu_table_dict_word(const BkgTable* t, int col, int64_t code,
+                                 char* out, int cap);
 
 #ifdef __cplusplus
 }

@@ -35,6 +35,15 @@ def lib():
     lib.bkparquet_close.argtypes = [C.c_void_p]
     lib.bkgpu_table_from_parquet.restype = C.c_void_p
     lib.bkgpu_table_from_parquet.argtypes = [C.c_char_p]
+    lib.bkparquet_read_string_column.restype = C.c_int64
+    lib.bkparquet_read_string_column.argtypes = [
+        C.c_void_p, C.c_int, C.POINTER(C.c_int32), C.POINTER(C.c_uint8),
+        C.POINTER(C.c_void_p), C.POINTER(C.c_int64)]
+    lib.bkparquet_dict_word.argtypes = [C.c_void_p, C.c_int64, C.c_char_p,
+                                        C.c_int]
+    lib.bkparquet_dict_free.argtypes = [C.c_void_p]
+    lib.bkgpu_table_dict_word.argtypes = [C.c_void_p, C.c_int, C.c_int64,
+                                          C.c_char_p, C.c_int]
     return lib
 
 
@@ -104,15 +113,57 @@ def test_reject_compressed(lib, tmp_path):
     lib.bkparquet_close(r)
 
 
-def test_reject_dictionary(lib, tmp_path):
+def test_dictionary_encoded_int64(lib, tmp_path):
+    """pyarrow's default dict encoding: dict page + RLE_DICTIONARY indices."""
     path = str(tmp_path / "d.parquet")
-    tab = pa.table({"a": pa.array(np.arange(1000, dtype=np.int64) % 5)})
-    pq.write_table(tab, path, compression=None, use_dictionary=True)
+    rng = np.random.default_rng(3)
+    a = rng.integers(0, 37, 50_000, dtype=np.int64) * 1000 - 5000
+    tab = pa.table({"a": pa.array(a)})
+    pq.write_table(tab, path, compression=None, use_dictionary=True,
+                   data_page_version="1.0", write_statistics=False)
     r = lib.bkparquet_open(path.encode())
     assert r
-    out = np.empty(1000, dtype=np.int64)
+    out = np.empty(len(a), dtype=np.int64)
     got = lib.bkparquet_read_column(r, 0, out.ctypes.data_as(C.c_void_p), None)
-    assert got < 0
+    assert got == len(a), lib.bkparquet_last_error()
+    assert np.array_equal(out, a)
+    lib.bkparquet_close(r)
+
+
+def test_string_column_order_preserving_codes(lib, tmp_path):
+    """BYTE_ARRAY -> dict codes whose order == byte order of the words."""
+    path = str(tmp_path / "s.parquet")
+    rng = np.random.default_rng(5)
+    words = [f"city_{i:04d}" for i in rng.integers(0, 300, 40_000)]
+    mask = rng.random(40_000) < 0.1
+    tab = pa.table({"w": pa.array(words, mask=mask)})
+    pq.write_table(tab, path, compression=None, use_dictionary=True,
+                   data_page_version="1.0", write_statistics=False)
+    r = lib.bkparquet_open(path.encode())
+    assert r
+    assert lib.bkparquet_col_type(r, 0) == 13    # BK_STRING
+    codes = np.empty(40_000, dtype=np.int32)
+    valid = np.empty(40_000, dtype=np.uint8)
+    dh = C.c_void_p()
+    dn = C.c_int64()
+    got = lib.bkparquet_read_string_column(
+        r, 0, codes.ctypes.data_as(C.POINTER(C.c_int32)),
+        valid.ctypes.data_as(C.POINTER(C.c_uint8)), C.byref(dh), C.byref(dn))
+    assert got == 40_000, lib.bkparquet_last_error()
+    uniq = sorted(set(w for w, m in zip(words, mask) if not m))
+    assert dn.value == len(uniq)
+    buf = C.create_string_buffer(64)
+    for c in (0, 1, dn.value // 2, dn.value - 1):
+        lib.bkparquet_dict_word(dh, c, buf, 64)
+        assert buf.value.decode() == uniq[c]
+    # codes match the sorted-unique rank of each word
+    rank = {w: i for i, w in enumerate(uniq)}
+    exp = np.array([0 if m else rank[w] for w, m in zip(words, mask)],
+                   dtype=np.int32)
+    sel = valid != 0
+    assert np.array_equal(codes[sel], exp[sel])
+    assert np.array_equal(sel, ~mask)
+    lib.bkparquet_dict_free(dh)
     lib.bkparquet_close(r)
 
 
@@ -162,3 +213,56 @@ def test_parquet_to_gpu_agg_parity(lib, tmp_path):
     assert np.array_equal(got["agg_i"][1], exp["agg_i"][1])
     err = np.abs(got["agg_d"][2] - exp["agg_d"][2])
     assert np.all(err <= 1e-10 * (np.abs(exp["agg_d"][2]) + 1))
+
+
+@pytest.mark.gpu
+def test_parquet_string_group_minmax(lib, tmp_path):
+    """Ingested VARCHAR column: GROUP BY string, MIN/MAX over another string
+    column, dict words materialized via bkgpu_table_dict_word."""
+    import torch
+    if torch.cuda.is_available():
+        torch.cuda.init()
+    from baikaldb_amd import GpuEngine, QueryPlan
+    from baikaldb_amd.engine import GpuTable
+
+    rng = np.random.default_rng(21)
+    n = 120_000
+    g = [f"grp_{i:02d}" for i in rng.integers(0, 12, n)]
+    w = [f"val_{i:05d}" for i in rng.integers(0, 5000, n)]
+    v = rng.integers(0, 100, n, dtype=np.int64)
+    path = str(tmp_path / "sv.parquet")
+    tab = pa.table({"g": pa.array(g), "w": pa.array(w),
+                    "v": pa.array(v)})
+    pq.write_table(tab, path, compression=None, use_dictionary=True,
+                   data_page_version="1.0", write_statistics=False)
+
+    eng = GpuEngine()
+    h = lib.bkgpu_table_from_parquet(path.encode())
+    assert h, lib.bkparquet_last_error()
+    t = GpuTable(eng, h, [13, 13, 6], n)
+    plan = QueryPlan(t.col_types, conjuncts=[(2, "<", 80)], group=[0],
+                     aggs=[("count_star", -1), ("min", 1), ("max", 1)])
+    res = eng.filter_agg(t, plan, expected_groups=1 << 8)
+    got = res.fetch(sorted=True)
+    res.free()
+
+    # brute force over the host data
+    sel = v < 80
+    groups = sorted(set(np.array(g)[sel]))
+    assert got["ngroups"] == len(groups)
+    buf = C.create_string_buffer(64)
+    garr, warr = np.array(g), np.array(w)
+    for r_i, gname in enumerate(groups):
+        # group key code -> word
+        code = int(got["enc"][r_i][0])
+        lib.bkgpu_table_dict_word(C.c_void_p(t.handle), 0, code, buf, 64)
+        assert buf.value.decode() == gname
+        rows = sel & (garr == gname)
+        assert got["agg_i"][0][r_i] == rows.sum()
+        lib.bkgpu_table_dict_word(C.c_void_p(t.handle), 1,
+                                  int(got["agg_i"][1][r_i]), buf, 64)
+        assert buf.value.decode() == min(warr[rows])
+        lib.bkgpu_table_dict_word(C.c_void_p(t.handle), 1,
+                                  int(got["agg_i"][2][r_i]), buf, 64)
+        assert buf.value.decode() == max(warr[rows])
+    t.free()
