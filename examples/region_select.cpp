@@ -1,0 +1,146 @@
+// region_select.cpp — a standalone C++ embedding of the engine, re-creating
+// exactly the driver loop a baikalStore region runs for one SELECT
+// (Region::select_normal, src/store/region.cpp:3166-3216):
+//
+//   build plan -> ExecNode::create_tree -> open -> while(!eos) get_next -> close
+//
+// Usage: region_select <file.parquet>
+//   runs  SELECT g, COUNT(*), SUM(v), MIN(w) FROM t WHERE v < 80 GROUP BY g
+//   and   SELECT g, v FROM t ORDER BY v, g LIMIT 5
+// over the ingested file and prints the result rows. No Python anywhere:
+// this is the product path an embedder links (libbkgpu.so + headers).
+#include <cstdio>
+#include <cstring>
+#include <vector>
+
+#include "../include/bk_exec.h"
+
+static int run_agg(BkgTable* table) {
+    BkPlanNodeDesc plan[3];
+    memset(plan, 0, sizeof plan);
+    /* pre-order: AGG -> FILTER -> SCAN (plan.proto:495-510 flattening) */
+    plan[0].node_type = BK_AGG_NODE;
+    plan[0].num_children = 1;
+    plan[0].limit = -1;
+    plan[0].n_group = 1;
+    plan[0].group_cols[0] = 0;
+    plan[0].n_aggs = 3;
+    plan[0].aggs[0] = {BK_AGG_COUNT_STAR, -1};
+    plan[0].aggs[1] = {BK_AGG_SUM, 2};
+    plan[0].aggs[2] = {BK_AGG_MIN, 1};
+    plan[0].expected_groups = 1 << 8;
+    plan[1].node_type = BK_WHERE_FILTER_NODE;
+    plan[1].num_children = 1;
+    plan[1].limit = -1;
+    plan[1].n_conjuncts = 1;
+    plan[1].conjuncts[0].col = 2;
+    plan[1].conjuncts[0].op = BK_OP_LT;
+    plan[1].conjuncts[0].cmp_type = BK_INT64;
+    plan[1].conjuncts[0].lit_i = 80;
+    plan[2].node_type = BK_SCAN_NODE;
+    plan[2].limit = -1;
+    plan[2].table = table;
+
+    BkExecTree* t = bkexec_create_tree(plan, 3);
+    if (!t) { fprintf(stderr, "create_tree: %s\n", bkgpu_last_error()); return 1; }
+    if (bkexec_open(t) < 0) { fprintf(stderr, "open failed\n"); return 1; }
+    int ns = bkexec_n_slots(t);
+    std::vector<int32_t> tag(16 * ns);
+    std::vector<int64_t> vi(16 * ns);
+    std::vector<double> vd(16 * ns);
+    std::vector<uint8_t> nul(16 * ns);
+    int eos = 0;
+    int64_t total = 0;
+    char word[64];
+    while (!eos) {
+        int64_t n = bkexec_get_next(t, 16, tag.data(), vi.data(), vd.data(),
+                                    nul.data(), &eos);
+        if (n < 0) { fprintf(stderr, "get_next failed\n"); return 1; }
+        for (int64_t r = 0; r < n; r++, total++) {
+            printf("row %lld:", (long long)total);
+            for (int s = 0; s < ns; s++) {
+                size_t i = (size_t)r * ns + s;
+                if (nul[i]) { printf(" NULL"); continue; }
+                if (tag[i] == BK_DOUBLE) printf(" %.6g", vd[i]);
+                else if (tag[i] == BK_STRING) {
+                    if (bkgpu_table_dict_word(table, s == 0 ? 0 : 1,
+                                              vi[i], word, sizeof word) >= 0)
+                        printf(" %s", word);
+                    else
+                        printf(" code:%lld", (long long)vi[i]);
+                } else printf(" %lld", (long long)vi[i]);
+            }
+            printf("\n");
+        }
+    }
+    printf("scan_rows=%lld filtered=%lld returned=%lld\n",
+           (long long)bkexec_num_scan_rows(t),
+           (long long)bkexec_num_filter_rows(t),
+           (long long)bkexec_num_rows_returned(t));
+    bkexec_close(t);
+    return 0;
+}
+
+static int run_sort(BkgTable* table) {
+    BkPlanNodeDesc plan[2];
+    memset(plan, 0, sizeof plan);
+    plan[0].node_type = BK_SORT_NODE;
+    plan[0].num_children = 1;
+    plan[0].limit = 5;
+    plan[0].n_order = 2;
+    plan[0].order[0] = {2, 1, 1, 0};   /* v asc */
+    plan[0].order[1] = {0, 1, 1, 0};   /* g asc */
+    plan[0].n_out_cols = 2;
+    plan[0].out_cols[0] = 0;
+    plan[0].out_cols[1] = 2;
+    plan[1].node_type = BK_SCAN_NODE;
+    plan[1].limit = -1;
+    plan[1].table = table;
+
+    BkExecTree* t = bkexec_create_tree(plan, 2);
+    if (!t) { fprintf(stderr, "create_tree: %s\n", bkgpu_last_error()); return 1; }
+    if (bkexec_open(t) < 0) { fprintf(stderr, "sort open failed\n"); return 1; }
+    int ns = bkexec_n_slots(t);
+    std::vector<int32_t> tag(8 * ns);
+    std::vector<int64_t> vi(8 * ns);
+    std::vector<double> vd(8 * ns);
+    std::vector<uint8_t> nul(8 * ns);
+    int eos = 0;
+    char word[64];
+    printf("top-5 by (v, g):\n");
+    while (!eos) {
+        int64_t n = bkexec_get_next(t, 8, tag.data(), vi.data(), vd.data(),
+                                    nul.data(), &eos);
+        if (n < 0) { fprintf(stderr, "sort get_next failed\n"); return 1; }
+        for (int64_t r = 0; r < n; r++) {
+            size_t i0 = (size_t)r * ns;
+            if (bkgpu_table_dict_word(table, 0, vi[i0], word, sizeof word) < 0)
+                snprintf(word, sizeof word, "code:%lld", (long long)vi[i0]);
+            printf("  g=%s v=%lld\n", word, (long long)vi[i0 + 1]);
+        }
+    }
+    bkexec_close(t);
+    return 0;
+}
+
+int main(int argc, char** argv) {
+    if (argc != 2) {
+        fprintf(stderr, "usage: %s <file.parquet>\n", argv[0]);
+        return 2;
+    }
+    if (bkgpu_device_count() < 1) {
+        fprintf(stderr, "no HIP device (this example needs a GPU)\n");
+        return 3;
+    }
+    BkgTable* table = (BkgTable*)bkgpu_table_from_parquet(argv[1]);
+    if (!table) {
+        fprintf(stderr, "ingest failed: %s\n", bkparquet_last_error());
+        return 1;
+    }
+    printf("ingested %lld rows x %d cols\n",
+           (long long)bkgpu_table_nrows(table), bkgpu_table_ncols(table));
+    int rc = run_agg(table);
+    if (rc == 0) rc = run_sort(table);
+    bkgpu_table_free(table);
+    return rc;
+}

@@ -266,3 +266,36 @@ def test_parquet_string_group_minmax(lib, tmp_path):
                                   int(got["agg_i"][2][r_i]), buf, 64)
         assert buf.value.decode() == max(warr[rows])
     t.free()
+
+
+@pytest.mark.gpu
+def test_cpp_embedding_example(tmp_path):
+    """examples/region_select: the pure-C++ driver loop (no Python in the
+    query path) over an ingested parquet file — Region::select_normal's
+    loop re-created by an embedder."""
+    import subprocess
+    exe = os.path.join(os.path.dirname(_LIB), "..", "examples",
+                       "region_select")
+    if not os.path.exists(exe):
+        pytest.skip("example binary not built")
+    rng = np.random.default_rng(9)
+    n = 50_000
+    g = [f"grp_{i:02d}" for i in rng.integers(0, 6, n)]
+    w = [f"val_{i:04d}" for i in rng.integers(0, 500, n)]
+    v = rng.integers(0, 100, n, dtype=np.int64)
+    path = str(tmp_path / "e.parquet")
+    tab = pa.table({"g": pa.array(g), "w": pa.array(w), "v": pa.array(v)})
+    pq.write_table(tab, path, compression=None, use_dictionary=True,
+                   data_page_version="1.0", write_statistics=False)
+    out = subprocess.run([exe, path], capture_output=True, text=True,
+                         timeout=120)
+    assert out.returncode == 0, out.stderr
+    assert f"ingested {n} rows x 3 cols" in out.stdout
+    # verify one group row against a brute-force recompute
+    garr, warr = np.array(g), np.array(w)
+    sel = v < 80
+    g0 = sorted(set(garr[sel]))[0]
+    rows = sel & (garr == g0)
+    expect = f"row 0: {g0} {rows.sum()} {v[rows].sum()} {min(warr[rows])}"
+    assert expect in out.stdout, f"wanted {expect!r} in:\n{out.stdout[:2000]}"
+    assert "top-5 by (v, g):" in out.stdout
