@@ -7,7 +7,8 @@ import os
 
 import numpy as np
 
-from .plan import BkQuerySpec, BkOrderSpec, QueryPlan, BK_MAX_GROUP
+from .plan import (BkQuerySpec, BkOrderSpec, BkWindowFn, QueryPlan,
+                   BK_MAX_GROUP, _WINFNS)
 from .plan import TYPE_INT64, TYPE_DOUBLE, TYPE_STRING
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
@@ -56,6 +57,13 @@ def _load():
     lib.bkgpu_agg_export_bytes.argtypes = [C.c_void_p]
     lib.bkgpu_agg_export.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
     lib.bkgpu_agg_merge.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
+    lib.bkgpu_window.restype = C.c_int64
+    lib.bkgpu_window.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec), C.c_int32,
+                                 C.POINTER(BkOrderSpec), C.c_int,
+                                 C.POINTER(BkWindowFn), C.c_int,
+                                 C.c_int64, C.c_int64,
+                                 C.POINTER(C.c_int64), C.POINTER(C.c_int64),
+                                 C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
     lib.bkgpu_agg_rollup.restype = C.c_void_p
     lib.bkgpu_agg_rollup.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
                                      C.POINTER(C.c_int32), C.c_int64]
@@ -253,6 +261,46 @@ class GpuEngine:
 
     def topk_kernel_ms(self):
         return self.lib.bkgpu_topk_kernel_ms()
+
+    def window(self, table, fns, part_col=-1, order=(), plan: QueryPlan = None,
+               row_begin=0, row_end=None):
+        """WindowNode (non-frame): fns = (name, col[, param]). Returns dict
+        with sorted rowids and fn-major out_i/out_d/out_null."""
+        if row_end is None:
+            row_end = table.nrows
+        q = (plan or QueryPlan(table.col_types)).to_spec()
+        oarr = (BkOrderSpec * max(len(order), 1))()
+        for i, (col, is_asc, null_first) in enumerate(order):
+            oarr[i].col, oarr[i].is_asc, oarr[i].is_null_first = \
+                col, is_asc, null_first
+        farr = (BkWindowFn * len(fns))()
+        for i, f in enumerate(fns):
+            name, col = f[0], f[1]
+            farr[i].fn_type = _WINFNS[name] if isinstance(name, str) else name
+            farr[i].col = col
+            farr[i].param = f[2] if len(f) > 2 else 0
+        cap = row_end - row_begin
+        rowids = np.empty(cap, dtype=np.int64)
+        out_i = np.zeros(len(fns) * cap, dtype=np.int64)
+        out_d = np.zeros(len(fns) * cap, dtype=np.float64)
+        out_null = np.zeros(len(fns) * cap, dtype=np.uint8)
+        n = self.lib.bkgpu_window(
+            table.handle, C.byref(q), part_col, oarr, len(order),
+            farr, len(fns), row_begin, row_end,
+            rowids.ctypes.data_as(C.POINTER(C.c_int64)),
+            out_i.ctypes.data_as(C.POINTER(C.c_int64)),
+            out_d.ctypes.data_as(C.POINTER(C.c_double)),
+            out_null.ctypes.data_as(C.POINTER(C.c_uint8)))
+        self._check(int(n), "window")
+        n = int(n)
+        nf = len(fns)
+        return {
+            "n": n,
+            "rowids": rowids[:n].copy(),
+            "out_i": out_i[:nf * n].reshape(nf, n).copy(),
+            "out_d": out_d[:nf * n].reshape(nf, n).copy(),
+            "out_null": out_null[:nf * n].reshape(nf, n).copy(),
+        }
 
     def upload_bytes(self, data: bytes):
         p = self.lib.bkgpu_upload_bytes(data, len(data))

@@ -37,6 +37,17 @@ class BkAggSpec(C.Structure):
     _fields_ = [("agg_type", C.c_int32), ("col", C.c_int32)]
 
 
+_WINFNS = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4,
+           "max": 5, "row_number": 10, "rank": 11, "dense_rank": 12,
+           "percent_rank": 13, "first_value": 14, "last_value": 15,
+           "nth_value": 16, "lead": 17, "lag": 18}
+
+
+class BkWindowFn(C.Structure):
+    _fields_ = [("fn_type", C.c_int32), ("col", C.c_int32),
+                ("param", C.c_int64)]
+
+
 class BkOrderSpec(C.Structure):
     _fields_ = [("col", C.c_int32), ("is_asc", C.c_int32),
                 ("is_null_first", C.c_int32), ("_pad", C.c_int32)]

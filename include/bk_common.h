@@ -161,6 +161,35 @@ typedef struct BkAggSpec {
     int32_t col;       /* input column, -1 for COUNT_STAR */
 } BkAggSpec;
 
+/* ---- window functions (reference src/expr/window_fn_call.cpp:20-38 name
+ * map; executed in the reference's NON-FRAME mode, window_node.cpp:39-41 —
+ * every fn sees the whole partition). ---- */
+typedef enum BkWinType {
+    BK_WIN_COUNT_STAR   = 0,
+    BK_WIN_COUNT        = 1,
+    BK_WIN_SUM          = 2,
+    BK_WIN_AVG          = 3,
+    BK_WIN_MIN          = 4,
+    BK_WIN_MAX          = 5,
+    BK_WIN_ROW_NUMBER   = 10,
+    BK_WIN_RANK         = 11,
+    BK_WIN_DENSE_RANK   = 12,
+    BK_WIN_PERCENT_RANK = 13,
+    BK_WIN_FIRST_VALUE  = 14,
+    BK_WIN_LAST_VALUE   = 15,
+    BK_WIN_NTH_VALUE    = 16,   /* param = n (1-based) */
+    BK_WIN_LEAD         = 17,   /* param = offset; out-of-partition => NULL */
+    BK_WIN_LAG          = 18,
+} BkWinType;
+
+typedef struct BkWindowFn {
+    int32_t fn_type;   /* BkWinType */
+    int32_t col;       /* input column, -1 for COUNT_STAR / pure rank fns */
+    int64_t param;     /* NTH_VALUE n / LEAD/LAG offset */
+} BkWindowFn;
+
+#define BK_MAX_WINFNS 8
+
 /* ORDER BY key (reference: include/mem_row/mem_row_compare.h:23-45). */
 typedef struct BkOrderSpec {
     int32_t col;

@@ -124,6 +124,17 @@ int64_t bkgpu_sort_topk(BkgTable* t, const BkQuerySpec* q,
                         int64_t limit, int64_t* out_rows);
 double bkgpu_topk_kernel_ms(void);
 
+/* ---- window functions (WindowNode NON-FRAME mode, window_node.cpp:39-41,
+ * window_fn_call.cpp:364-700): rows sorted by (partition, order, arrival);
+ * outputs fn-major host arrays out_i/out_d/out_null[f*n + i]; out_rowids
+ * receives the sorted global row ids. Returns rows produced, <0 error. ---- */
+int64_t bkgpu_window(BkgTable* t, const BkQuerySpec* q, int32_t part_col,
+                     const BkOrderSpec* order, int norder,
+                     const BkWindowFn* fns, int nfns,
+                     int64_t row_begin, int64_t row_end,
+                     int64_t* out_rowids, int64_t* out_i, double* out_d,
+                     uint8_t* out_null);
+
 /* ---- SELECT without GROUP BY (FilterNode row emission, filter_node.cpp:
  * 736-795): collect up to limit passing global row ids (order unspecified);
  * materialize columns for row ids. Host arrays. ---- */

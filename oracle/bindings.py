@@ -36,6 +36,11 @@ class BkAggSpec(C.Structure):
     _fields_ = [("agg_type", C.c_int32), ("col", C.c_int32)]
 
 
+class BkWindowFn(C.Structure):
+    _fields_ = [("fn_type", C.c_int32), ("col", C.c_int32),
+                ("param", C.c_int64)]
+
+
 class BkOrderSpec(C.Structure):
     _fields_ = [("col", C.c_int32), ("is_asc", C.c_int32),
                 ("is_null_first", C.c_int32), ("_pad", C.c_int32)]
@@ -125,6 +130,12 @@ class Oracle:
                                       C.POINTER(BkQuerySpec), C.POINTER(BkOrderSpec),
                                       C.c_int, C.c_int64, C.c_int64, C.c_int64,
                                       C.POINTER(C.c_int64)]
+        lib.orc_window.restype = C.c_int64
+        lib.orc_window.argtypes = [
+            C.POINTER(_OrcCol), C.c_int, C.POINTER(BkQuerySpec), C.c_int32,
+            C.POINTER(BkOrderSpec), C.c_int, C.POINTER(BkWindowFn), C.c_int,
+            C.c_int64, C.c_int64, C.POINTER(C.c_int64), C.POINTER(C.c_int64),
+            C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
         lib.orc_dict_word.restype = C.c_int
         lib.orc_dict_word.argtypes = [C.c_uint64, C.c_int64, C.c_char_p, C.c_int]
         for f in ("orc_encode_i64", "orc_decode_i64"):
@@ -232,6 +243,46 @@ class Oracle:
             return out
         finally:
             self.lib.orc_agg_result_free(res)
+
+    def window(self, cols, valids, col_types, fns, part_col=-1, order=(),
+               q=None, row_begin=0, row_end=None):
+        """fns: (fn_type:int, col, param). Mirrors bkgpu_window."""
+        if row_end is None:
+            row_end = len(cols[0])
+        if q is None:
+            q = make_query((), (), ((0, -1),), col_types)
+            q.n_aggs = 0
+        carr = self._make_cols(cols, valids, col_types)
+        oarr = (BkOrderSpec * max(len(order), 1))()
+        for i, (col, is_asc, null_first) in enumerate(order):
+            oarr[i].col, oarr[i].is_asc, oarr[i].is_null_first = \
+                col, is_asc, null_first
+        farr = (BkWindowFn * len(fns))()
+        for i, f in enumerate(fns):
+            farr[i].fn_type, farr[i].col = f[0], f[1]
+            farr[i].param = f[2] if len(f) > 2 else 0
+        cap = row_end - row_begin
+        rowids = np.empty(cap, dtype=np.int64)
+        out_i = np.zeros(len(fns) * cap, dtype=np.int64)
+        out_d = np.zeros(len(fns) * cap, dtype=np.float64)
+        out_null = np.zeros(len(fns) * cap, dtype=np.uint8)
+        n = self.lib.orc_window(
+            carr, len(cols), C.byref(q), part_col, oarr, len(order),
+            farr, len(fns), row_begin, row_end,
+            rowids.ctypes.data_as(C.POINTER(C.c_int64)),
+            out_i.ctypes.data_as(C.POINTER(C.c_int64)),
+            out_d.ctypes.data_as(C.POINTER(C.c_double)),
+            out_null.ctypes.data_as(C.POINTER(C.c_uint8)))
+        assert n >= 0
+        n = int(n)
+        nf = len(fns)
+        return {
+            "n": n,
+            "rowids": rowids[:n].copy(),
+            "out_i": out_i[:nf * n].reshape(nf, n).copy(),
+            "out_d": out_d[:nf * n].reshape(nf, n).copy(),
+            "out_null": out_null[:nf * n].reshape(nf, n).copy(),
+        }
 
     # ---- sort + top-N ----
     def sort_topk(self, cols, valids, col_types, order, limit, q=None,

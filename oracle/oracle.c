@@ -850,6 +850,183 @@ ORC_EXPORT int64_t orc_sort_topk(const OrcCol* cols, int ncols, const BkQuerySpe
 }
 
 /* ------------------------------------------------------------------ */
+/* WindowNode, NON-FRAME mode (window_node.cpp:39-41: every fn sees its
+ * whole partition; fn semantics window_fn_call.cpp:364-700).
+ * Rows sorted by (partition asc nulls-first, order keys, arrival), then
+ * partitions walked start-to-end. Mirrors bkgpu_window.                */
+/* ------------------------------------------------------------------ */
+
+static int orc_cell_eq(const OrcCol* c, int64_t ra, int64_t rb) {
+    int va = cell_is_valid(c, ra), vb = cell_is_valid(c, rb);
+    if (va != vb) return 0;
+    if (!va) return 1;   /* NULLs equal (mem_row_compare.cpp:18-40) */
+    if (c->type == BK_DOUBLE)
+        return ((double*)c->data)[ra] == ((double*)c->data)[rb];
+    return cell_i64(c, ra) == cell_i64(c, rb);
+}
+
+static void orc_win_value(const OrcCol* c, int64_t r, int64_t idx,
+                          int64_t* out_i, double* out_d, uint8_t* out_null) {
+    if (!cell_is_valid(c, r)) { out_null[idx] = 1; return; }
+    out_null[idx] = 0;
+    if (c->type == BK_DOUBLE) out_d[idx] = ((double*)c->data)[r];
+    else out_i[idx] = cell_i64(c, r);
+}
+
+ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
+                              const BkQuerySpec* q, int32_t part_col,
+                              const BkOrderSpec* order, int norder,
+                              const BkWindowFn* fns, int nfns,
+                              int64_t row_begin, int64_t row_end,
+                              int64_t* out_rows, int64_t* out_i,
+                              double* out_d, uint8_t* out_null) {
+    BkOrderSpec full[4];
+    int nf = 0;
+    if (part_col >= 0) {
+        full[nf].col = part_col; full[nf].is_asc = 1;
+        full[nf].is_null_first = 1; nf++;
+    }
+    for (int k = 0; k < norder && nf < 4; k++) full[nf++] = order[k];
+    int64_t n = orc_sort_topk(cols, ncols, q, full, nf, row_begin, row_end,
+                              row_end - row_begin, out_rows);
+    if (n <= 0) return n;
+
+    int64_t ps = 0;
+    while (ps < n) {
+        int64_t pe = ps + 1;
+        while (pe < n && (part_col < 0 ||
+               orc_cell_eq(&cols[part_col], out_rows[pe], out_rows[pe - 1])))
+            pe++;
+        int64_t pn = pe - ps;
+        /* per-partition aggregate states */
+        for (int f = 0; f < nfns; f++) {
+            int ft = fns[f].fn_type;
+            if (ft > BK_WIN_MAX) continue;
+            const OrcCol* c = fns[f].col >= 0 ? &cols[fns[f].col] : NULL;
+            int64_t cnt = 0, vi = 0;
+            double vd = 0.0;
+            int has = 0;
+            for (int64_t j = ps; j < pe; j++) {
+                int64_t r = out_rows[j];
+                if (ft == BK_WIN_COUNT_STAR) { cnt++; continue; }
+                if (!cell_is_valid(c, r)) continue;
+                switch (ft) {
+                    case BK_WIN_COUNT: cnt++; break;
+                    case BK_WIN_SUM:
+                        if (c->type == BK_DOUBLE) vd += ((double*)c->data)[r];
+                        else vi = (int64_t)((uint64_t)vi +
+                                            (uint64_t)cell_i64(c, r));
+                        cnt++;
+                        break;
+                    case BK_WIN_AVG: vd += cell_f64_cast(c, r); cnt++; break;
+                    case BK_WIN_MIN:
+                        if (c->type == BK_DOUBLE) {
+                            double v = ((double*)c->data)[r];
+                            if (!has || v < vd) vd = v;
+                        } else {
+                            int64_t v = cell_i64(c, r);
+                            if (!has || v < vi) vi = v;
+                        }
+                        has = 1; cnt++;
+                        break;
+                    case BK_WIN_MAX:
+                        if (c->type == BK_DOUBLE) {
+                            double v = ((double*)c->data)[r];
+                            if (!has || v > vd) vd = v;
+                        } else {
+                            int64_t v = cell_i64(c, r);
+                            if (!has || v > vi) vi = v;
+                        }
+                        has = 1; cnt++;
+                        break;
+                    default: break;
+                }
+            }
+            for (int64_t j = ps; j < pe; j++) {
+                int64_t idx = (int64_t)f * n + j;
+                out_i[idx] = 0; out_d[idx] = 0.0; out_null[idx] = 0;
+                switch (ft) {
+                    case BK_WIN_COUNT_STAR:
+                    case BK_WIN_COUNT: out_i[idx] = cnt; break;
+                    case BK_WIN_SUM:
+                        if (!cnt) { out_null[idx] = 1; break; }
+                        if (c->type == BK_DOUBLE) out_d[idx] = vd;
+                        else out_i[idx] = vi;
+                        break;
+                    case BK_WIN_AVG:
+                        if (!cnt) { out_null[idx] = 1; break; }
+                        out_d[idx] = vd / (double)cnt;
+                        break;
+                    case BK_WIN_MIN:
+                    case BK_WIN_MAX:
+                        if (!cnt) { out_null[idx] = 1; break; }
+                        if (c->type == BK_DOUBLE) out_d[idx] = vd;
+                        else out_i[idx] = vi;
+                        break;
+                    default: break;
+                }
+            }
+        }
+        /* rank / positional fns */
+        int64_t peer_head = ps, peers_seen = 1;
+        for (int64_t j = ps; j < pe; j++) {
+            if (j > ps) {
+                int changed = 0;
+                for (int k = 0; k < norder; k++)
+                    if (!orc_cell_eq(&cols[order[k].col], out_rows[j],
+                                     out_rows[j - 1])) { changed = 1; break; }
+                if (changed) { peer_head = j; peers_seen++; }
+            }
+            for (int f = 0; f < nfns; f++) {
+                int ft = fns[f].fn_type;
+                if (ft <= BK_WIN_MAX) continue;
+                int64_t idx = (int64_t)f * n + j;
+                out_i[idx] = 0; out_d[idx] = 0.0; out_null[idx] = 0;
+                switch (ft) {
+                    case BK_WIN_ROW_NUMBER: out_i[idx] = j - ps + 1; break;
+                    case BK_WIN_RANK: out_i[idx] = peer_head - ps + 1; break;
+                    case BK_WIN_DENSE_RANK: out_i[idx] = peers_seen; break;
+                    case BK_WIN_PERCENT_RANK:
+                        out_d[idx] = pn > 1
+                            ? (double)(peer_head - ps) / (double)(pn - 1)
+                            : 0.0;
+                        break;
+                    case BK_WIN_FIRST_VALUE:
+                        orc_win_value(&cols[fns[f].col], out_rows[ps], idx,
+                                      out_i, out_d, out_null);
+                        break;
+                    case BK_WIN_LAST_VALUE:
+                        orc_win_value(&cols[fns[f].col], out_rows[pe - 1], idx,
+                                      out_i, out_d, out_null);
+                        break;
+                    case BK_WIN_NTH_VALUE: {
+                        int64_t jj = ps + fns[f].param - 1;
+                        if (jj >= ps && jj < pe)
+                            orc_win_value(&cols[fns[f].col], out_rows[jj], idx,
+                                          out_i, out_d, out_null);
+                        else out_null[idx] = 1;
+                        break;
+                    }
+                    case BK_WIN_LEAD:
+                    case BK_WIN_LAG: {
+                        int64_t off = fns[f].param > 0 ? fns[f].param : 1;
+                        int64_t jj = ft == BK_WIN_LEAD ? j + off : j - off;
+                        if (jj >= ps && jj < pe)
+                            orc_win_value(&cols[fns[f].col], out_rows[jj], idx,
+                                          out_i, out_d, out_null);
+                        else out_null[idx] = 1;
+                        break;
+                    }
+                    default: out_null[idx] = 1; break;
+                }
+            }
+        }
+        ps = pe;
+    }
+    return n;
+}
+
+/* ------------------------------------------------------------------ */
 /* small exported helpers for tests                                    */
 /* ------------------------------------------------------------------ */
 

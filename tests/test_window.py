@@ -1,0 +1,264 @@
+# WindowNode (NON-FRAME mode, window_node.cpp:39-41 + window_fn_call.cpp):
+# oracle vs numpy brute force (CPU) and GPU vs oracle (parity).
+import numpy as np
+import pytest
+
+TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
+D_UNI, D_SKEW, D_DICT, D_SUM16 = 0, 1, 2, 3
+W = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5,
+     "row_number": 10, "rank": 11, "dense_rank": 12, "percent_rank": 13,
+     "first_value": 14, "last_value": 15, "nth_value": 16, "lead": 17,
+     "lag": 18}
+SEED = 77
+
+
+def gen(orc, specs, n, seed=SEED):
+    import ctypes as C
+    from oracle.bindings import BkColSpec
+    arr = (BkColSpec * len(specs))()
+    for i, s in enumerate(specs):
+        (arr[i].col_type, arr[i].dist, arr[i].p0, arr[i].p1,
+         arr[i].null_frac_x1e6) = s
+    cols, valids = orc.generate_table(list(arr), n, seed)
+    return cols, valids, [s[0] for s in specs]
+
+
+@pytest.fixture(scope="module")
+def orc():
+    from oracle import Oracle
+    return Oracle()
+
+
+@pytest.fixture(scope="module")
+def eng():
+    import torch
+    if torch.cuda.is_available():
+        torch.cuda.init()
+    from baikaldb_amd import GpuEngine
+    return GpuEngine()
+
+
+def brute_window(cols, valids, part_col, order, fns):
+    """numpy/python reference independent of oracle code paths."""
+    n = len(cols[0])
+
+    def null(c, r):
+        return valids[c] is not None and valids[c][r] == 0
+
+    def keyf(r):
+        ks = []
+        if part_col >= 0:
+            ks.append((0 if null(part_col, r) else 1,
+                       None if null(part_col, r) else cols[part_col][r]))
+        for col, asc, nf in order:
+            isn = null(col, r)
+            v = None if isn else cols[col][r]
+            # null_first under asc: nulls smallest; invert value for desc
+            nk = (0 if nf else 2) if isn else 1
+            ks.append((nk, (v if asc else (-v if v is not None else None))
+                       if not isn else 0))
+        ks.append(r)
+        return tuple(ks)
+
+    idx = sorted(range(n), key=keyf)
+    out = {f: [None] * n for f in range(len(fns))}
+    # partitions
+    def peq(a, b):
+        if part_col < 0:
+            return True
+        na, nb = null(part_col, a), null(part_col, b)
+        if na != nb:
+            return False
+        return na or cols[part_col][a] == cols[part_col][b]
+
+    def oeq(a, b):
+        for col, _, _ in order:
+            na, nb = null(col, a), null(col, b)
+            if na != nb:
+                return False
+            if not na and cols[col][a] != cols[col][b]:
+                return False
+        return True
+
+    ps = 0
+    while ps < n:
+        pe = ps + 1
+        while pe < n and peq(idx[pe], idx[pe - 1]):
+            pe += 1
+        rows = [idx[j] for j in range(ps, pe)]
+        pn = len(rows)
+        for f, (name, col, *rest) in enumerate(fns):
+            param = rest[0] if rest else 0
+            if name == "count_star":
+                vals = [pn] * pn
+            elif name == "count":
+                c = sum(0 if null(col, r) else 1 for r in rows)
+                vals = [c] * pn
+            elif name in ("sum", "avg", "min", "max"):
+                vv = [cols[col][r] for r in rows if not null(col, r)]
+                if not vv:
+                    vals = [None] * pn
+                elif name == "sum":
+                    vals = [np.sum(np.array(vv))] * pn
+                elif name == "avg":
+                    vals = [float(np.mean(np.array(vv, dtype=np.float64)))] * pn
+                elif name == "min":
+                    vals = [min(vv)] * pn
+                else:
+                    vals = [max(vv)] * pn
+            elif name == "row_number":
+                vals = list(range(1, pn + 1))
+            elif name in ("rank", "dense_rank", "percent_rank"):
+                vals = []
+                rank, dense = 1, 1
+                for j in range(pn):
+                    if j > 0 and not oeq(rows[j], rows[j - 1]):
+                        rank = j + 1
+                        dense += 1
+                    if name == "rank":
+                        vals.append(rank)
+                    elif name == "dense_rank":
+                        vals.append(dense)
+                    else:
+                        vals.append((rank - 1) / (pn - 1) if pn > 1 else 0.0)
+            elif name == "first_value":
+                r0 = rows[0]
+                vals = [None if null(col, r0) else cols[col][r0]] * pn
+            elif name == "last_value":
+                r0 = rows[-1]
+                vals = [None if null(col, r0) else cols[col][r0]] * pn
+            elif name == "nth_value":
+                j = param - 1
+                if 0 <= j < pn:
+                    r0 = rows[j]
+                    vals = [None if null(col, r0) else cols[col][r0]] * pn
+                else:
+                    vals = [None] * pn
+            elif name in ("lead", "lag"):
+                off = param if param > 0 else 1
+                vals = []
+                for j in range(pn):
+                    jj = j + off if name == "lead" else j - off
+                    if 0 <= jj < pn and not null(col, rows[jj]):
+                        vals.append(cols[col][rows[jj]])
+                    else:
+                        vals.append(None)
+            else:
+                raise ValueError(name)
+            for j in range(pn):
+                out[f][ps + j] = vals[j]
+        ps = pe
+    return idx, out
+
+
+FNS = [("count_star", -1), ("count", 2), ("sum", 2), ("avg", 2),
+       ("min", 2), ("max", 2), ("row_number", -1), ("rank", -1)]
+FNS2 = [("dense_rank", -1), ("percent_rank", -1), ("first_value", 2),
+        ("last_value", 2), ("nth_value", 2, 3), ("lead", 2, 1), ("lag", 2, 2)]
+
+
+def check_against_brute(res, col_types, fns, idx, brute):
+    assert res["n"] == len(idx)
+    assert np.array_equal(res["rowids"], np.array(idx))
+    for f, fdesc in enumerate(fns):
+        name, col = fdesc[0], fdesc[1]
+        is_double = (name in ("avg", "percent_rank") or
+                     (col >= 0 and col_types[col] == TYPE_DOUBLE))
+        for i in range(res["n"]):
+            b = brute[f][i]
+            if b is None:
+                assert res["out_null"][f][i] == 1, (name, i)
+            else:
+                assert res["out_null"][f][i] == 0, (name, i, b)
+                if is_double:
+                    got = res["out_d"][f][i]
+                    assert abs(got - b) <= 1e-9 * (abs(b) + 1), (name, i, got, b)
+                else:
+                    assert res["out_i"][f][i] == b, (name, i)
+
+
+def oracle_window(orc, cols, valids, col_types, fns, part_col, order):
+    ofns = [(W[f[0]], f[1], f[2] if len(f) > 2 else 0) for f in fns]
+    return orc.window(cols, valids, col_types, ofns, part_col=part_col,
+                      order=order)
+
+
+def test_oracle_window_vs_brute(orc):
+    specs = [(TYPE_INT64, D_UNI, 0, 12, 0),          # partition
+             (TYPE_INT64, D_UNI, 0, 40, 100_000),    # order key, 10% null
+             (TYPE_INT64, D_UNI, -500, 500, 200_000)]  # value, 20% null
+    cols, valids, types = gen(orc, specs, 4000)
+    order = [(1, 1, 1)]
+    for fns in (FNS, FNS2):
+        res = oracle_window(orc, cols, valids, types, fns, 0, order)
+        idx, brute = brute_window(cols, valids, 0, order, fns)
+        check_against_brute(res, types, fns, idx, brute)
+
+
+def test_oracle_window_no_partition_no_order(orc):
+    specs = [(TYPE_INT64, D_UNI, 0, 100, 0)]
+    cols, valids, types = gen(orc, specs, 500)
+    fns = [("row_number", -1), ("rank", -1), ("count_star", -1)]
+    res = oracle_window(orc, cols, valids, types, fns, -1, [])
+    assert res["n"] == 500
+    assert np.array_equal(res["out_i"][0], np.arange(1, 501))  # row_number
+    assert np.all(res["out_i"][1] == 1)                        # rank: no order
+    assert np.all(res["out_i"][2] == 500)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("fns", [FNS, FNS2])
+def test_gpu_window_parity(eng, orc, fns):
+    specs = [(TYPE_INT64, D_UNI, 0, 300, 50_000),      # partition, nullable
+             (TYPE_INT64, D_UNI, 0, 25, 100_000),      # order
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 150_000),    # double value
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 0)]
+    # swap value col to the double col for a double-typed run
+    fns = [tuple([f[0], 2 if f[1] == 2 else f[1]] + list(f[2:])) for f in fns]
+    n = 200_000
+    from baikaldb_amd import QueryPlan
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED)
+        plan = QueryPlan(t.col_types, conjuncts=[(3, "<", 1 << 30)])
+        got = eng.window(t, fns, part_col=0, order=[(1, 1, 1)], plan=plan)
+    finally:
+        t.free()
+    cols, valids, types = gen(orc, specs, n)
+    from oracle.bindings import make_query
+    q = make_query([(3, 4, TYPE_INT64, 1 << 30)], (), ((0, -1),), types)
+    q.n_aggs = 0
+    exp = orc.window(cols, valids, types,
+                     [(W[f[0]], f[1], f[2] if len(f) > 2 else 0) for f in fns],
+                     part_col=0, order=[(1, 1, 1)], q=q)
+    assert got["n"] == exp["n"]
+    assert np.array_equal(got["rowids"], exp["rowids"])
+    assert np.array_equal(got["out_null"], exp["out_null"])
+    assert np.array_equal(got["out_i"], exp["out_i"])
+    mask = exp["out_null"] == 0
+    d = np.abs(got["out_d"] - exp["out_d"])
+    tol = 1e-10 * (np.abs(exp["out_d"]) + 100)
+    assert np.all(d[mask] <= tol[mask])
+
+
+@pytest.mark.gpu
+def test_gpu_window_desc_order_and_big_partitions(eng, orc):
+    specs = [(TYPE_INT64, D_UNI, 0, 3, 0),             # 3 huge partitions
+             (TYPE_INT64, D_UNI, 0, 1 << 40, 0),
+             (TYPE_INT64, D_UNI, 0, 1000, 0)]
+    n = 300_000
+    fns = [("row_number", -1), ("rank", -1), ("sum", 2), ("lag", 2, 3)]
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED + 1)
+        got = eng.window(t, fns, part_col=0, order=[(1, 0, 0)])  # desc
+    finally:
+        t.free()
+    cols, valids, types = gen(orc, specs, n, SEED + 1)
+    exp = orc.window(cols, valids, types,
+                     [(W[f[0]], f[1], f[2] if len(f) > 2 else 0) for f in fns],
+                     part_col=0, order=[(1, 0, 0)])
+    assert got["n"] == exp["n"] == n
+    assert np.array_equal(got["rowids"], exp["rowids"])
+    assert np.array_equal(got["out_i"], exp["out_i"])
+    assert np.array_equal(got["out_null"], exp["out_null"])
