@@ -40,7 +40,8 @@ class BkAggSpec(C.Structure):
 _WINFNS = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4,
            "max": 5, "row_number": 10, "rank": 11, "dense_rank": 12,
            "percent_rank": 13, "first_value": 14, "last_value": 15,
-           "nth_value": 16, "lead": 17, "lag": 18}
+           "nth_value": 16, "lead": 17, "lag": 18, "cume_dist": 19,
+           "ntile": 20}
 
 
 class BkWindowFn(C.Structure):

@@ -180,6 +180,8 @@ typedef enum BkWinType {
     BK_WIN_NTH_VALUE    = 16,   /* param = n (1-based) */
     BK_WIN_LEAD         = 17,   /* param = offset; out-of-partition => NULL */
     BK_WIN_LAG          = 18,
+    BK_WIN_CUME_DIST    = 19,
+    BK_WIN_NTILE        = 20,   /* param = n buckets */
 } BkWinType;
 
 typedef struct BkWindowFn {

@@ -968,7 +968,7 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
             }
         }
         /* rank / positional fns */
-        int64_t peer_head = ps, peers_seen = 1;
+        int64_t peer_head = ps, peers_seen = 1, peer_end = ps;
         for (int64_t j = ps; j < pe; j++) {
             if (j > ps) {
                 int changed = 0;
@@ -976,6 +976,18 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                     if (!orc_cell_eq(&cols[order[k].col], out_rows[j],
                                      out_rows[j - 1])) { changed = 1; break; }
                 if (changed) { peer_head = j; peers_seen++; }
+            }
+            if (j >= peer_end) {   /* advance to current peer group's end */
+                peer_end = j + 1;
+                while (peer_end < pe) {
+                    int same = 1;
+                    for (int k = 0; k < norder; k++)
+                        if (!orc_cell_eq(&cols[order[k].col],
+                                         out_rows[peer_end], out_rows[j]))
+                            { same = 0; break; }
+                    if (!same) break;
+                    peer_end++;
+                }
             }
             for (int f = 0; f < nfns; f++) {
                 int ft = fns[f].fn_type;
@@ -1015,6 +1027,18 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                             orc_win_value(&cols[fns[f].col], out_rows[jj], idx,
                                           out_i, out_d, out_null);
                         else out_null[idx] = 1;
+                        break;
+                    }
+                    case BK_WIN_CUME_DIST:
+                        out_d[idx] = (double)(peer_end - ps) / (double)pn;
+                        break;
+                    case BK_WIN_NTILE: {
+                        int64_t k2 = fns[f].param > 0 ? fns[f].param : 1;
+                        int64_t quot = pn / k2, rem = pn % k2;
+                        int64_t jl = j - ps, fat = rem * (quot + 1);
+                        out_i[idx] = jl < fat
+                            ? jl / (quot + 1) + 1
+                            : rem + (quot > 0 ? (jl - fat) / quot : 0) + 1;
                         break;
                     }
                     default: out_null[idx] = 1; break;

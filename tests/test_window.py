@@ -8,7 +8,7 @@ D_UNI, D_SKEW, D_DICT, D_SUM16 = 0, 1, 2, 3
 W = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5,
      "row_number": 10, "rank": 11, "dense_rank": 12, "percent_rank": 13,
      "first_value": 14, "last_value": 15, "nth_value": 16, "lead": 17,
-     "lag": 18}
+     "lag": 18, "cume_dist": 19, "ntile": 20}
 SEED = 77
 
 
@@ -134,6 +134,20 @@ def brute_window(cols, valids, part_col, order, fns):
                     vals = [None if null(col, r0) else cols[col][r0]] * pn
                 else:
                     vals = [None] * pn
+            elif name == "cume_dist":
+                vals = []
+                for j in range(pn):
+                    le = j + 1
+                    while le < pn and oeq(rows[le], rows[j]):
+                        le += 1
+                    vals.append(le / pn)
+            elif name == "ntile":
+                k = param if param > 0 else 1
+                quot, rem = pn // k, pn % k
+                fat = rem * (quot + 1)
+                vals = [(j // (quot + 1) + 1) if j < fat
+                        else rem + ((j - fat) // quot if quot else 0) + 1
+                        for j in range(pn)]
             elif name in ("lead", "lag"):
                 off = param if param > 0 else 1
                 vals = []
@@ -154,7 +168,8 @@ def brute_window(cols, valids, part_col, order, fns):
 FNS = [("count_star", -1), ("count", 2), ("sum", 2), ("avg", 2),
        ("min", 2), ("max", 2), ("row_number", -1), ("rank", -1)]
 FNS2 = [("dense_rank", -1), ("percent_rank", -1), ("first_value", 2),
-        ("last_value", 2), ("nth_value", 2, 3), ("lead", 2, 1), ("lag", 2, 2)]
+        ("last_value", 2), ("nth_value", 2, 3), ("lead", 2, 1), ("lag", 2, 2),
+        ("cume_dist", -1, 0), ("ntile", -1, 4)]
 
 
 def check_against_brute(res, col_types, fns, idx, brute):
@@ -162,7 +177,7 @@ def check_against_brute(res, col_types, fns, idx, brute):
     assert np.array_equal(res["rowids"], np.array(idx))
     for f, fdesc in enumerate(fns):
         name, col = fdesc[0], fdesc[1]
-        is_double = (name in ("avg", "percent_rank") or
+        is_double = (name in ("avg", "percent_rank", "cume_dist") or
                      (col >= 0 and col_types[col] == TYPE_DOUBLE))
         for i in range(res["n"]):
             b = brute[f][i]
