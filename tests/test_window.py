@@ -168,8 +168,9 @@ def brute_window(cols, valids, part_col, order, fns):
 FNS = [("count_star", -1), ("count", 2), ("sum", 2), ("avg", 2),
        ("min", 2), ("max", 2), ("row_number", -1), ("rank", -1)]
 FNS2 = [("dense_rank", -1), ("percent_rank", -1), ("first_value", 2),
-        ("last_value", 2), ("nth_value", 2, 3), ("lead", 2, 1), ("lag", 2, 2),
-        ("cume_dist", -1, 0), ("ntile", -1, 4)]
+        ("last_value", 2), ("nth_value", 2, 3), ("lead", 2, 1), ("lag", 2, 2)]
+FNS3 = [("cume_dist", -1, 0), ("ntile", -1, 4), ("ntile", -1, 7),
+        ("rank", -1), ("row_number", -1)]
 
 
 def check_against_brute(res, col_types, fns, idx, brute):
@@ -204,7 +205,7 @@ def test_oracle_window_vs_brute(orc):
              (TYPE_INT64, D_UNI, -500, 500, 200_000)]  # value, 20% null
     cols, valids, types = gen(orc, specs, 4000)
     order = [(1, 1, 1)]
-    for fns in (FNS, FNS2):
+    for fns in (FNS, FNS2, FNS3):
         res = oracle_window(orc, cols, valids, types, fns, 0, order)
         idx, brute = brute_window(cols, valids, 0, order, fns)
         check_against_brute(res, types, fns, idx, brute)
@@ -222,7 +223,7 @@ def test_oracle_window_no_partition_no_order(orc):
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("fns", [FNS, FNS2])
+@pytest.mark.parametrize("fns", [FNS, FNS2, FNS3])
 def test_gpu_window_parity(eng, orc, fns):
     specs = [(TYPE_INT64, D_UNI, 0, 300, 50_000),      # partition, nullable
              (TYPE_INT64, D_UNI, 0, 25, 100_000),      # order
