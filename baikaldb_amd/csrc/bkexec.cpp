@@ -575,6 +575,131 @@ private:
 };
 
 /* ---- LimitNode ---- */
+/* ---- WindowNode (window_node.cpp, NON-FRAME mode; fns evaluated by
+ * bkgpu_window over rows sorted by (partition, order, arrival)). Slots:
+ * [out_cols...][window fn outputs...]. ---- */
+class WindowNode : public ExecNode {
+public:
+    int init(const BkPlanNodeDesc& node) override {
+        ExecNode::init(node);
+        _desc = node;
+        if (_desc.n_winfns <= 0) return -1;
+        return 0;
+    }
+    static int32_t fn_out_type(const BkWindowFn& f, BkgTable* t) {
+        switch (f.fn_type) {
+            case BK_WIN_COUNT_STAR: case BK_WIN_COUNT: case BK_WIN_ROW_NUMBER:
+            case BK_WIN_RANK: case BK_WIN_DENSE_RANK: case BK_WIN_NTILE:
+                return BK_INT64;                 /* window_fn_call.cpp:213-258 */
+            case BK_WIN_AVG: case BK_WIN_PERCENT_RANK: case BK_WIN_CUME_DIST:
+                return BK_DOUBLE;
+            default:
+                return bkgpu_table_col_type(t, f.col);  /* input type */
+        }
+    }
+    int open(RuntimeState* state) override {
+        int ret = ExecNode::open(state);
+        if (ret < 0) return ret;
+        ScanNode* scan = find_scan(this);
+        if (!scan) { state->error_msg = "WindowNode: no scan below"; return -1; }
+        FilterNode* filter = find_filter(this);
+        BkQuerySpec q{};
+        if (filter) {
+            q.n_conjuncts = filter->n_conjuncts();
+            memcpy(q.conjuncts, filter->conjuncts(), sizeof(q.conjuncts));
+        }
+        BkgTable* t = scan->table();
+        int64_t nrows = bkgpu_table_nrows(t);
+        _rowids.resize(nrows ? nrows : 1);
+        _win_i.resize((size_t)_desc.n_winfns * (nrows ? nrows : 1));
+        _win_d.resize((size_t)_desc.n_winfns * (nrows ? nrows : 1));
+        _win_n.resize((size_t)_desc.n_winfns * (nrows ? nrows : 1));
+        int64_t got = bkgpu_window(t, &q, _desc.part_col, _desc.order,
+                                   _desc.n_order, _desc.winfns, _desc.n_winfns,
+                                   0, nrows, _rowids.data(), _win_i.data(),
+                                   _win_d.data(), _win_n.data());
+        if (got < 0) { state->error_msg = bkgpu_last_error(); return -1; }
+        _n = got;
+        state->inc_num_scan_rows(nrows);
+        state->inc_num_filter_rows(nrows - got);
+        _fn_types.resize(_desc.n_winfns);
+        for (int f = 0; f < _desc.n_winfns; f++)
+            _fn_types[f] = fn_out_type(_desc.winfns[f], t);
+        /* materialize out_cols of the sorted rows */
+        _cols_i.assign(_desc.n_out_cols, {});
+        _cols_d.assign(_desc.n_out_cols, {});
+        _cols_n.assign(_desc.n_out_cols, {});
+        _col_types.resize(_desc.n_out_cols);
+        for (int c = 0; c < _desc.n_out_cols; c++) {
+            int col = _desc.out_cols[c];
+            _col_types[c] = bkgpu_table_col_type(t, col);
+            _cols_i[c].resize(_n ? _n : 1);
+            _cols_d[c].resize(_n ? _n : 1);
+            _cols_n[c].resize(_n ? _n : 1);
+            if (_n > 0 &&
+                bkgpu_gather(t, col, _rowids.data(), _n, _cols_i[c].data(),
+                             _cols_d[c].data(), _cols_n[c].data()) != 0) {
+                state->error_msg = bkgpu_last_error();
+                return -1;
+            }
+        }
+        _iter = 0;
+        return 0;
+    }
+    int get_next(RuntimeState* state, RowBatch* batch, bool* eos) override {
+        while (true) {
+            if (state->is_cancelled()) { *eos = true; return 0; }
+            if (reached_limit() || _iter >= _n) { *eos = true; return 0; }
+            if (batch->is_full()) return 0;
+            auto row = std::make_unique<MemRow>(n_slots());
+            int s = 0;
+            for (int c = 0; c < _desc.n_out_cols; c++, s++) {
+                ExprValue v;
+                v.type = _col_types[c];
+                v.is_null_ = _cols_n[c][_iter] != 0;
+                if (!v.is_null_) {
+                    if (v.type == BK_DOUBLE) v.d = _cols_d[c][_iter];
+                    else v.i = _cols_i[c][_iter];
+                }
+                row->set_value(s, v);
+            }
+            for (int f = 0; f < _desc.n_winfns; f++, s++) {
+                ExprValue v;
+                v.type = _fn_types[f];
+                size_t idx = (size_t)f * _n + _iter;
+                v.is_null_ = _win_n[idx] != 0;
+                if (!v.is_null_) {
+                    if (v.type == BK_DOUBLE) v.d = _win_d[idx];
+                    else v.i = _win_i[idx];
+                }
+                row->set_value(s, v);
+            }
+            batch->move_row(std::move(row));
+            _num_rows_returned++;
+            _iter++;
+        }
+    }
+    void close(RuntimeState* state) override {
+        ExecNode::close(state);
+        _rowids.clear(); _win_i.clear(); _win_d.clear(); _win_n.clear();
+        _iter = 0; _n = 0;
+    }
+    int n_slots() const { return _desc.n_out_cols + _desc.n_winfns; }
+
+private:
+    BkPlanNodeDesc _desc{};
+    std::vector<int64_t> _rowids;
+    std::vector<int64_t> _win_i;
+    std::vector<double> _win_d;
+    std::vector<uint8_t> _win_n;
+    std::vector<int32_t> _fn_types;
+    std::vector<std::vector<int64_t>> _cols_i;
+    std::vector<std::vector<double>> _cols_d;
+    std::vector<std::vector<uint8_t>> _cols_n;
+    std::vector<int32_t> _col_types;
+    int64_t _iter = 0, _n = 0;
+};
+
 class LimitNode : public ExecNode {
 public:
     int get_next(RuntimeState* state, RowBatch* batch, bool* eos) override {
@@ -599,6 +724,7 @@ ExecNode* ExecNode::create_exec_node(const BkPlanNodeDesc& node) {
     switch (node.node_type) {
         case BK_SCAN_NODE:         return new ScanNode();
         case BK_SORT_NODE:         return new SortNode();
+        case BK_WINDOW_NODE:       return new WindowNode();
         case BK_AGG_NODE:
         case BK_MERGE_AGG_NODE:    return new AggNode();
         case BK_TABLE_FILTER_NODE:
@@ -655,6 +781,8 @@ static int tree_n_slots(const BkExecTree* t) {
         return static_cast<AggNode*>(n)->n_slots();
     if (n->node_type() == BK_SORT_NODE)
         return static_cast<SortNode*>(n)->n_slots();
+    if (n->node_type() == BK_WINDOW_NODE)
+        return static_cast<WindowNode*>(n)->n_slots();
     if (n->node_type() == BK_TABLE_FILTER_NODE ||
         n->node_type() == BK_WHERE_FILTER_NODE)
         return static_cast<FilterNode*>(n)->n_slots();

@@ -9,11 +9,13 @@ import numpy as np
 
 from .engine import _load
 from .plan import (BkQuerySpec, BkConjunct, BkAggSpec, BkOrderSpec,  # noqa
+                   BkWindowFn, _WINFNS,
                    BK_MAX_GROUP, BK_MAX_CONJ, BK_MAX_AGGS, _OPS, _AGGS,
                    TYPE_INT64, TYPE_DOUBLE, TYPE_STRING)
 
 BK_MAX_COLS = 16
 SCAN, SORT, AGG, MERGE_AGG, TABLE_FILTER, LIMIT, WHERE_FILTER = 1, 2, 4, 5, 6, 11, 12
+WINDOW = 42
 
 
 class BkPlanNodeDesc(C.Structure):
@@ -30,7 +32,10 @@ class BkPlanNodeDesc(C.Structure):
                 ("n_order", C.c_int32),
                 ("order", BkOrderSpec * 4),
                 ("n_out_cols", C.c_int32),
-                ("out_cols", C.c_int32 * BK_MAX_COLS)]
+                ("out_cols", C.c_int32 * BK_MAX_COLS),
+                ("part_col", C.c_int32),
+                ("n_winfns", C.c_int32),
+                ("winfns", BkWindowFn * 8)]
 
 
 def _bind(lib):
@@ -80,6 +85,29 @@ def filter_node(col_types, conjuncts, num_children=1):
             cj.cmp_type, cj.lit_d = TYPE_DOUBLE, float(lit)
         else:
             cj.cmp_type, cj.lit_i = TYPE_INT64, int(lit)
+    return d
+
+
+def window_node(part_col, order, fns, out_cols, num_children=1, limit=-1):
+    """WINDOW_NODE (window_node.cpp, non-frame): fns = (name, col[, param]);
+    slots = [out_cols...][fn outputs...]."""
+    d = BkPlanNodeDesc()
+    d.node_type, d.num_children = WINDOW, num_children
+    d.limit = limit
+    d.part_col = part_col
+    d.n_order = len(order)
+    for i, (col, is_asc, null_first) in enumerate(order):
+        d.order[i].col, d.order[i].is_asc, d.order[i].is_null_first = \
+            col, is_asc, null_first
+    d.n_winfns = len(fns)
+    for i, f in enumerate(fns):
+        name = f[0]
+        d.winfns[i].fn_type = _WINFNS[name] if isinstance(name, str) else name
+        d.winfns[i].col = f[1]
+        d.winfns[i].param = f[2] if len(f) > 2 else 0
+    d.n_out_cols = len(out_cols)
+    for i, c in enumerate(out_cols):
+        d.out_cols[i] = c
     return d
 
 

@@ -47,6 +47,7 @@ typedef enum BkNodeType {
     BK_TABLE_FILTER_NODE = 6,
     BK_LIMIT_NODE        = 11,
     BK_WHERE_FILTER_NODE = 12,
+    BK_WINDOW_NODE       = 42,
 } BkNodeType;
 
 /* ---- comparison ops of src/expr/operators.cpp:79-105 (eq/ne/gt/ge/lt/le) ---- */

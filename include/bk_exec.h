@@ -52,6 +52,12 @@ typedef struct BkPlanNodeDesc {
     BkOrderSpec order[4];
     int32_t     n_out_cols;          /* columns materialized per output row */
     int32_t     out_cols[BK_MAX_COLS];
+    /* WINDOW_NODE payload (NON-FRAME mode, window_node.cpp:39-41); reuses
+     * n_order/order for the in-partition sort and n_out_cols/out_cols for
+     * the materialized input columns */
+    int32_t     part_col;            /* -1 = single whole-set partition */
+    int32_t     n_winfns;
+    BkWindowFn  winfns[BK_MAX_WINFNS];
 } BkPlanNodeDesc;
 
 typedef struct BkExecTree BkExecTree;      /* root ExecNode + RuntimeState */

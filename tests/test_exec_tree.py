@@ -219,3 +219,54 @@ def test_count_distinct_through_exec_surface(eng, orc):
         assert vi[r, 1] == (g == gv).sum()          # count_star
         assert vi[r, 2] == len(_np.unique(sel))     # count_distinct
         assert vi[r, 3] == sel.sum()                # sum
+
+
+def test_window_through_exec_surface(eng, orc):
+    """WINDOW node (non-frame, window_node.cpp): row_number/rank/sum OVER
+    (PARTITION BY c2 ORDER BY c0) through the ExecNode mirror."""
+    from baikaldb_amd import exec as bx
+    import numpy as _np
+    t, cols, valids, types = make_table(eng, orc, n=40_000)
+    try:
+        nodes = [bx.window_node(part_col=2, order=[(0, 1, 1)],
+                                fns=[("row_number", -1), ("rank", -1),
+                                     ("sum", 2), ("lag", 0, 1)],
+                                out_cols=[2, 0]),
+                 bx.filter_node(types, [(0, "<", int((1 << 31) * 0.4))]),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
+    sel = cols[0] < int((1 << 31) * 0.4)
+    idx = _np.nonzero(sel)[0]
+    # sort by (partition c2, order c0, arrival)
+    orderk = _np.lexsort((idx, cols[0][idx], cols[2][idx]))
+    sidx = idx[orderk]
+    assert tags.shape[0] == len(sidx)
+    # slots: [c2, c0, row_number, rank, sum, lag]
+    assert _np.array_equal(vi[:, 0], cols[2][sidx])
+    assert _np.array_equal(vi[:, 1], cols[0][sidx])
+    g = cols[2][sidx]
+    # brute per partition
+    rn = _np.zeros(len(sidx), dtype=_np.int64)
+    sm = _np.zeros(len(sidx), dtype=_np.int64)
+    lag = _np.full(len(sidx), -1, dtype=_np.int64)
+    lagnull = _np.zeros(len(sidx), dtype=bool)
+    start = 0
+    while start < len(sidx):
+        end = start + 1
+        while end < len(sidx) and g[end] == g[start]:
+            end += 1
+        rn[start:end] = _np.arange(1, end - start + 1)
+        sm[start:end] = cols[2][sidx[start:end]].sum()
+        lagnull[start] = True
+        lag[start + 1:end] = cols[0][sidx[start:end - 1]]
+        start = end
+    assert _np.array_equal(vi[:, 2], rn)
+    assert _np.array_equal(vi[:, 4], sm)
+    assert _np.array_equal(nulls[:, 5] != 0, lagnull)
+    m = ~lagnull
+    assert _np.array_equal(vi[:, 5][m], lag[m])
