@@ -278,3 +278,59 @@ def test_gpu_window_desc_order_and_big_partitions(eng, orc):
     assert np.array_equal(got["rowids"], exp["rowids"])
     assert np.array_equal(got["out_i"], exp["out_i"])
     assert np.array_equal(got["out_null"], exp["out_null"])
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("case_seed", range(3))
+def test_gpu_window_fuzz(eng, orc, case_seed):
+    """Randomized schemas x window specs: GPU vs oracle (oracle itself is
+    pinned against the numpy brute force above)."""
+    import random
+    rng = random.Random(4200 + case_seed)
+    for sub in range(4):
+        ncols = rng.randint(2, 5)
+        specs = []
+        for _ in range(ncols):
+            tcol = rng.choice([TYPE_INT64, TYPE_INT64, TYPE_DOUBLE])
+            nf = rng.choice([0, 0, 200_000])
+            if tcol == TYPE_INT64:
+                specs.append((tcol, D_UNI, 0,
+                              rng.choice([5, 50, 1 << 20, 1 << 40]), nf))
+            else:
+                specs.append((tcol, D_SUM16, 0, 0, nf))
+        part = rng.choice([-1] + list(range(ncols)))
+        n_ord = rng.randint(0, min(2, ncols))
+        order = [(c, rng.randint(0, 1), rng.randint(0, 1))
+                 for c in rng.sample(range(ncols), n_ord)]
+        pool = list(W.keys())
+        fns = []
+        for _ in range(rng.randint(1, 6)):
+            name = rng.choice(pool)
+            needs_col = name not in ("count_star", "row_number", "rank",
+                                     "dense_rank", "percent_rank",
+                                     "cume_dist", "ntile")
+            col = rng.randrange(ncols) if needs_col else -1
+            param = rng.randint(1, 5) if name in ("nth_value", "lead",
+                                                  "lag", "ntile") else 0
+            fns.append((name, col, param))
+        n = rng.choice([3000, 20_000])
+        seed = rng.randrange(1 << 40)
+        t = eng.create_table(specs, n)
+        try:
+            eng.generate(t, seed)
+            got = eng.window(t, fns, part_col=part, order=order)
+        finally:
+            t.free()
+        cols, valids, types = gen(orc, specs, n, seed)
+        exp = orc.window(cols, valids, types,
+                         [(W[f[0]], f[1], f[2]) for f in fns],
+                         part_col=part, order=order)
+        ctx = f"fuzz {case_seed}/{sub} part={part} order={order} fns={fns}"
+        assert got["n"] == exp["n"], ctx
+        assert np.array_equal(got["rowids"], exp["rowids"]), ctx
+        assert np.array_equal(got["out_null"], exp["out_null"]), ctx
+        assert np.array_equal(got["out_i"], exp["out_i"]), ctx
+        mask = exp["out_null"] == 0
+        d = np.abs(got["out_d"] - exp["out_d"])
+        tol = 1e-10 * (np.abs(exp["out_d"]) + 100)
+        assert np.all(d[mask] <= tol[mask]), ctx
