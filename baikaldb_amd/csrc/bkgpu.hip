@@ -624,11 +624,15 @@ k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_
 
 struct RecLayout {
     int32_t nwords;
-    int32_t k1_word;               /* -1 if n_group < 2 */
+    int32_t k1_word;               /* -1 if n_group < 2; -2 dict code in the
+                                      meta word's high half; -3 FUSED: word 0
+                                      = [k0-k0_base:32 | k1:24 | flags:8]
+                                      (narrow-key mode — no meta word) */
     int32_t meta_word;             /* -1 if nothing nullable; else
                                       u64 word: bits0-7 = null flag,
                                       bits 8+a = agg-input a valid */
     int32_t val_word[BK_MAX_AGGS]; /* -1 for COUNT(*) / COUNT */
+    uint64_t k0_base;              /* fused mode: min enc of group col 0 */
 };
 
 #define PART_BUCKET(h, P) ((uint32_t)((h) >> 44) & ((P) - 1u))
@@ -854,9 +858,14 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
         uint64_t regs[8];
         #pragma unroll
         for (int w = 0; w < 8; w++) regs[w] = 0;
-        regs[0] = k0;
-        if (lay.k1_word >= 0) regs[lay.k1_word] = k1;
-        else if (lay.k1_word == -2) meta |= k1 << 32;
+        if (lay.k1_word == -3) {
+            uint64_t d0 = (meta & 0x80u) ? 0 : (k0 - lay.k0_base);
+            regs[0] = (d0 << 32) | (k1 << 8) | (meta & 0xFFu);
+        } else {
+            regs[0] = k0;
+            if (lay.k1_word >= 0) regs[lay.k1_word] = k1;
+            else if (lay.k1_word == -2) meta |= k1 << 32;
+        }
         for (int32_t a = 0; a < q.n_aggs; a++) {
             if (lay.val_word[a] < 0) continue;
             const BkAggSpec& as = q.aggs[a];
@@ -1156,18 +1165,32 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
             uint64_t k0A = 0, k1A = 0, metaA = 0, k0B = 0, k1B = 0, metaB = 0;
             uint32_t fA = 0, fB = 0;
             if (pA) {
-                k0A = myA[0];
-                metaA = lay.meta_word >= 0 ? myA[lay.meta_word] : 0;
-                k1A = lay.k1_word >= 0 ? myA[lay.k1_word]
-                      : (lay.k1_word == -2 ? (metaA >> 32) : 0);
-                fA = (uint32_t)(metaA & 0xFF);
+                uint64_t w0 = myA[0];
+                if (lay.k1_word == -3) {
+                    fA = (uint32_t)(w0 & 0xFFu);
+                    k1A = (w0 >> 8) & 0xFFFFFFull;
+                    k0A = (fA & 0x80u) ? 0 : lay.k0_base + (w0 >> 32);
+                } else {
+                    k0A = w0;
+                    metaA = lay.meta_word >= 0 ? myA[lay.meta_word] : 0;
+                    k1A = lay.k1_word >= 0 ? myA[lay.k1_word]
+                          : (lay.k1_word == -2 ? (metaA >> 32) : 0);
+                    fA = (uint32_t)(metaA & 0xFF);
+                }
             }
             if (pB) {
-                k0B = myB[0];
-                metaB = lay.meta_word >= 0 ? myB[lay.meta_word] : 0;
-                k1B = lay.k1_word >= 0 ? myB[lay.k1_word]
-                      : (lay.k1_word == -2 ? (metaB >> 32) : 0);
-                fB = (uint32_t)(metaB & 0xFF);
+                uint64_t w0 = myB[0];
+                if (lay.k1_word == -3) {
+                    fB = (uint32_t)(w0 & 0xFFu);
+                    k1B = (w0 >> 8) & 0xFFFFFFull;
+                    k0B = (fB & 0x80u) ? 0 : lay.k0_base + (w0 >> 32);
+                } else {
+                    k0B = w0;
+                    metaB = lay.meta_word >= 0 ? myB[lay.meta_word] : 0;
+                    k1B = lay.k1_word >= 0 ? myB[lay.k1_word]
+                          : (lay.k1_word == -2 ? (metaB >> 32) : 0);
+                    fB = (uint32_t)(metaB & 0xFF);
+                }
             }
             for (;;) {
                 agg_try_one(pA, myA, k0A, k1A, metaA, fA, ltab, lmask, stride,
@@ -1354,7 +1377,63 @@ struct BkgTable {
      * order-preserving; see bkparquet.cpp). Null for generated tables,
      * whose words come from bk_dict_word. */
     std::vector<std::string>* dict[BK_MAX_COLS] = {};
+    /* cached per-column order-encoding range over VALID cells (for the
+     * fused narrow-key record layout; invalidated on generate/upload) */
+    uint64_t stat_min[BK_MAX_COLS] = {};
+    uint64_t stat_max[BK_MAX_COLS] = {};
+    uint8_t  stat_ok[BK_MAX_COLS] = {};
 };
+
+/* per-column enc_value range over valid cells (block-reduce + one atomic) */
+__global__ void k_enc_range(DevCol c, int64_t n, uint64_t* mn, uint64_t* mx) {
+    uint64_t lmn = ~0ull, lmx = 0;
+    int64_t gs = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += gs) {
+        if (!cell_valid(c, r)) continue;
+        uint64_t e = enc_value(c, r);
+        lmn = e < lmn ? e : lmn;
+        lmx = e > lmx ? e : lmx;
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+        uint64_t a = __shfl_down((unsigned long long)lmn, off, 64);
+        uint64_t b = __shfl_down((unsigned long long)lmx, off, 64);
+        lmn = a < lmn ? a : lmn;
+        lmx = b > lmx ? b : lmx;
+    }
+    if ((threadIdx.x & 63) == 0) {
+        atomicMin((unsigned long long*)mn, (unsigned long long)lmn);
+        atomicMax((unsigned long long*)mx, (unsigned long long)lmx);
+    }
+}
+
+static DevCols table_cols(const BkgTable* t);
+static hipError_t pool_alloc(void** p, size_t bytes);
+static void pool_free(void* p);
+
+static int ensure_stats(BkgTable* t, int col) {
+    if (t->stat_ok[col]) return 0;
+    uint64_t* d = nullptr;
+    if (pool_alloc((void**)&d, 16) != hipSuccess) return -1;
+    uint64_t init[2] = {~0ull, 0};
+    if (hipMemcpy(d, init, 16, hipMemcpyHostToDevice) != hipSuccess) {
+        pool_free(d);
+        return -1;
+    }
+    DevCols dc = table_cols(t);
+    hipLaunchKernelGGL(k_enc_range, dim3(1024), dim3(256), 0, 0,
+                       dc.c[col], t->nrows, d, d + 1);
+    uint64_t out[2];
+    if (hipMemcpy(out, d, 16, hipMemcpyDeviceToHost) != hipSuccess) {
+        pool_free(d);
+        return -1;
+    }
+    pool_free(d);
+    t->stat_min[col] = out[0];
+    t->stat_max[col] = out[1];
+    t->stat_ok[col] = 1;
+    return 0;
+}
 
 /* attach an ingested column dictionary (concatenated words + offsets) */
 extern "C" int bkgpu_table_set_dict(BkgTable* t, int col, const char* concat,
@@ -1531,6 +1610,7 @@ static DevCols table_cols(const BkgTable* t) {
 }
 
 extern "C" int bkgpu_table_generate(BkgTable* t, uint64_t seed, int64_t row_begin) {
+    memset(t->stat_ok, 0, sizeof t->stat_ok);
     DevCols dc = table_cols(t);
     DevSpecs ds{};
     for (int c = 0; c < t->ncols; c++) ds.s[c] = t->specs[c];
@@ -1545,6 +1625,7 @@ extern "C" int bkgpu_table_generate(BkgTable* t, uint64_t seed, int64_t row_begi
 
 extern "C" int bkgpu_table_upload(BkgTable* t, int col, const void* data,
                                   const uint8_t* valid) {
+    t->stat_ok[col] = 0;
     size_t es = elem_size(t->specs[col].col_type);
     HIP_CHECK(hipMemcpy(t->data[col], data, (size_t)t->nrows * es,
                         hipMemcpyHostToDevice));
@@ -1645,21 +1726,52 @@ static int agg_compact(BkgAggOut* o) {
     return 0;
 }
 
-static int build_rec_layout(const BkgTable* t, const BkQuerySpec* q, RecLayout* lay) {
-    int w = 1; /* word 0 = k0 */
+static int build_rec_layout(BkgTable* t, const BkQuerySpec* q, RecLayout* lay) {
+    int w = 1; /* word 0 = k0 (or the fused key word) */
     /* a dict-encoded (BK_STRING) second group key is a 32-bit code: pack it
      * into the meta word's high half (k1_word == -2) instead of spending a
      * whole record word — 20% narrower records on the config-3 shape */
     bool k1_in_meta = q->n_group >= 2 &&
                       t->specs[q->group_cols[1]].col_type == BK_STRING;
-    lay->k1_word = q->n_group >= 2 ? (k1_in_meta ? -2 : w++) : -1;
-    bool need_meta = k1_in_meta;
-    for (int k = 0; k < q->n_group; k++)
-        if (t->valid[q->group_cols[k]]) need_meta = true;
+    bool agg_meta = false;   /* meta bits beyond the key null flags */
     for (int a = 0; a < q->n_aggs; a++) {
         int col = q->aggs[a].col;
-        if (col >= 0 && t->valid[col]) need_meta = true;
+        if (col >= 0 && t->valid[col]) agg_meta = true;
     }
+    /* FUSED narrow-key mode (adaptive, from cached column stats): when the
+     * first key's valid-cell encoding span fits 32 bits, the second key (if
+     * any) is a small dict code, and no agg-input validity bits are needed,
+     * key + flags collapse into ONE word — 20-25% narrower records, which
+     * is scatter write traffic and part_agg read traffic. */
+    lay->k0_base = 0;
+    bool fused = q->n_group >= 1 && !agg_meta && getenv("BK_NO_FUSE") == nullptr;
+    if (fused) {
+        int c0 = q->group_cols[0];
+        fused = ensure_stats(t, c0) == 0 && t->stat_ok[c0] &&
+                t->stat_max[c0] >= t->stat_min[c0] &&
+                t->stat_max[c0] - t->stat_min[c0] < (1ull << 32);
+        if (fused && q->n_group >= 2) {
+            int c1 = q->group_cols[1];
+            fused = k1_in_meta && ensure_stats(t, c1) == 0 && t->stat_ok[c1] &&
+                    t->stat_max[c1] < (1ull << 24);
+        }
+        if (fused) lay->k0_base = t->stat_min[q->group_cols[0]];
+    }
+    if (fused) {
+        lay->k1_word = -3;
+        for (int a = 0; a < q->n_aggs; a++) {
+            int at = q->aggs[a].agg_type;
+            lay->val_word[a] =
+                (at == BK_AGG_COUNT_STAR || at == BK_AGG_COUNT) ? -1 : w++;
+        }
+        lay->meta_word = -1;
+        lay->nwords = w;
+        return 0;
+    }
+    lay->k1_word = q->n_group >= 2 ? (k1_in_meta ? -2 : w++) : -1;
+    bool need_meta = k1_in_meta || agg_meta;
+    for (int k = 0; k < q->n_group; k++)
+        if (t->valid[q->group_cols[k]]) need_meta = true;
     for (int a = 0; a < q->n_aggs; a++) {
         int at = q->aggs[a].agg_type;
         lay->val_word[a] =
