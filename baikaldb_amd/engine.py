@@ -61,6 +61,7 @@ def _load():
     lib.bkgpu_window.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec), C.c_int32,
                                  C.POINTER(BkOrderSpec), C.c_int,
                                  C.POINTER(BkWindowFn), C.c_int,
+                                 C.c_int32, C.c_int64, C.c_int64,
                                  C.c_int64, C.c_int64,
                                  C.POINTER(C.c_int64), C.POINTER(C.c_int64),
                                  C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
@@ -263,7 +264,7 @@ class GpuEngine:
         return self.lib.bkgpu_topk_kernel_ms()
 
     def window(self, table, fns, part_col=-1, order=(), plan: QueryPlan = None,
-               row_begin=0, row_end=None):
+               frame=None, row_begin=0, row_end=None):
         """WindowNode (non-frame): fns = (name, col[, param]). Returns dict
         with sorted rowids and fn-major out_i/out_d/out_null."""
         if row_end is None:
@@ -284,9 +285,10 @@ class GpuEngine:
         out_i = np.zeros(len(fns) * cap, dtype=np.int64)
         out_d = np.zeros(len(fns) * cap, dtype=np.float64)
         out_null = np.zeros(len(fns) * cap, dtype=np.uint8)
+        fr, fpre, ffol = (1, frame[0], frame[1]) if frame else (0, -1, -1)
         n = self.lib.bkgpu_window(
             table.handle, C.byref(q), part_col, oarr, len(order),
-            farr, len(fns), row_begin, row_end,
+            farr, len(fns), fr, fpre, ffol, row_begin, row_end,
             rowids.ctypes.data_as(C.POINTER(C.c_int64)),
             out_i.ctypes.data_as(C.POINTER(C.c_int64)),
             out_d.ctypes.data_as(C.POINTER(C.c_double)),

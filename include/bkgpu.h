@@ -124,13 +124,16 @@ int64_t bkgpu_sort_topk(BkgTable* t, const BkQuerySpec* q,
                         int64_t limit, int64_t* out_rows);
 double bkgpu_topk_kernel_ms(void);
 
-/* ---- window functions (WindowNode NON-FRAME mode, window_node.cpp:39-41,
- * window_fn_call.cpp:364-700): rows sorted by (partition, order, arrival);
+/* ---- window functions (WindowNode, window_node.cpp; non-frame mode and
+ * ROWS frames — frame_rows=1, bounds f_pre/f_fol rows, negative =
+ * UNBOUNDED; RowFrameWindowProcessor semantics for SUM/COUNT/AVG and
+ * FIRST/LAST/NTH_VALUE): rows sorted by (partition, order, arrival);
  * outputs fn-major host arrays out_i/out_d/out_null[f*n + i]; out_rowids
  * receives the sorted global row ids. Returns rows produced, <0 error. ---- */
 int64_t bkgpu_window(BkgTable* t, const BkQuerySpec* q, int32_t part_col,
                      const BkOrderSpec* order, int norder,
                      const BkWindowFn* fns, int nfns,
+                     int32_t frame_rows, int64_t f_pre, int64_t f_fol,
                      int64_t row_begin, int64_t row_end,
                      int64_t* out_rowids, int64_t* out_i, double* out_d,
                      uint8_t* out_null);
