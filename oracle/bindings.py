@@ -134,6 +134,7 @@ class Oracle:
         lib.orc_window.argtypes = [
             C.POINTER(_OrcCol), C.c_int, C.POINTER(BkQuerySpec), C.c_int32,
             C.POINTER(BkOrderSpec), C.c_int, C.POINTER(BkWindowFn), C.c_int,
+            C.c_int32, C.c_int64, C.c_int64,
             C.c_int64, C.c_int64, C.POINTER(C.c_int64), C.POINTER(C.c_int64),
             C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
         lib.orc_dict_word.restype = C.c_int
@@ -245,7 +246,7 @@ class Oracle:
             self.lib.orc_agg_result_free(res)
 
     def window(self, cols, valids, col_types, fns, part_col=-1, order=(),
-               q=None, row_begin=0, row_end=None):
+               q=None, frame=None, row_begin=0, row_end=None):
         """fns: (fn_type:int, col, param). Mirrors bkgpu_window."""
         if row_end is None:
             row_end = len(cols[0])
@@ -266,9 +267,10 @@ class Oracle:
         out_i = np.zeros(len(fns) * cap, dtype=np.int64)
         out_d = np.zeros(len(fns) * cap, dtype=np.float64)
         out_null = np.zeros(len(fns) * cap, dtype=np.uint8)
+        fr, fpre, ffol = (1, frame[0], frame[1]) if frame else (0, -1, -1)
         n = self.lib.orc_window(
             carr, len(cols), C.byref(q), part_col, oarr, len(order),
-            farr, len(fns), row_begin, row_end,
+            farr, len(fns), fr, fpre, ffol, row_begin, row_end,
             rowids.ctypes.data_as(C.POINTER(C.c_int64)),
             out_i.ctypes.data_as(C.POINTER(C.c_int64)),
             out_d.ctypes.data_as(C.POINTER(C.c_double)),

@@ -877,6 +877,7 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                               const BkQuerySpec* q, int32_t part_col,
                               const BkOrderSpec* order, int norder,
                               const BkWindowFn* fns, int nfns,
+                              int32_t frame_rows, int64_t f_pre, int64_t f_fol,
                               int64_t row_begin, int64_t row_end,
                               int64_t* out_rows, int64_t* out_i,
                               double* out_d, uint8_t* out_null) {
@@ -902,6 +903,46 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
         for (int f = 0; f < nfns; f++) {
             int ft = fns[f].fn_type;
             if (ft > BK_WIN_MAX) continue;
+            if (frame_rows && ft <= BK_WIN_AVG) {
+                /* ROWS frame (RowFrameWindowProcessor): per-row recompute
+                 * over [fl, fr] clamped inside the partition */
+                const OrcCol* fc = fns[f].col >= 0 ? &cols[fns[f].col] : NULL;
+                for (int64_t j = ps; j < pe; j++) {
+                    int64_t fl = f_pre >= 0 && j - f_pre > ps ? j - f_pre : ps;
+                    int64_t fr = f_fol >= 0 && j + f_fol < pe - 1 ? j + f_fol
+                                                                  : pe - 1;
+                    int64_t idx = (int64_t)f * n + j;
+                    out_i[idx] = 0; out_d[idx] = 0.0; out_null[idx] = 0;
+                    int64_t fcnt = 0, fvi = 0;
+                    double fvd = 0.0;
+                    for (int64_t jj = fl; jj <= fr; jj++) {
+                        int64_t r = out_rows[jj];
+                        if (ft == BK_WIN_COUNT_STAR) { fcnt++; continue; }
+                        if (!cell_is_valid(fc, r)) continue;
+                        fcnt++;
+                        if (ft == BK_WIN_SUM) {
+                            if (fc->type == BK_DOUBLE)
+                                fvd += ((double*)fc->data)[r];
+                            else fvi = (int64_t)((uint64_t)fvi +
+                                                 (uint64_t)cell_i64(fc, r));
+                        } else if (ft == BK_WIN_AVG) {
+                            fvd += cell_f64_cast(fc, r);
+                        }
+                    }
+                    if (ft == BK_WIN_COUNT_STAR || ft == BK_WIN_COUNT) {
+                        out_i[idx] = fcnt;
+                    } else if (!fcnt) {
+                        out_null[idx] = 1;
+                    } else if (ft == BK_WIN_AVG) {
+                        out_d[idx] = fvd / (double)fcnt;
+                    } else if (fc->type == BK_DOUBLE) {
+                        out_d[idx] = fvd;
+                    } else {
+                        out_i[idx] = fvi;
+                    }
+                }
+                continue;
+            }
             const OrcCol* c = fns[f].col >= 0 ? &cols[fns[f].col] : NULL;
             int64_t cnt = 0, vi = 0;
             double vd = 0.0;
@@ -994,6 +1035,11 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                 if (ft <= BK_WIN_MAX) continue;
                 int64_t idx = (int64_t)f * n + j;
                 out_i[idx] = 0; out_d[idx] = 0.0; out_null[idx] = 0;
+                int64_t fl = ps, fr = pe - 1;
+                if (frame_rows) {
+                    if (f_pre >= 0 && j - f_pre > ps) fl = j - f_pre;
+                    if (f_fol >= 0 && j + f_fol < pe - 1) fr = j + f_fol;
+                }
                 switch (ft) {
                     case BK_WIN_ROW_NUMBER: out_i[idx] = j - ps + 1; break;
                     case BK_WIN_RANK: out_i[idx] = peer_head - ps + 1; break;
@@ -1004,16 +1050,18 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                             : 0.0;
                         break;
                     case BK_WIN_FIRST_VALUE:
-                        orc_win_value(&cols[fns[f].col], out_rows[ps], idx,
+                        if (fr < fl) { out_null[idx] = 1; break; }
+                        orc_win_value(&cols[fns[f].col], out_rows[fl], idx,
                                       out_i, out_d, out_null);
                         break;
                     case BK_WIN_LAST_VALUE:
-                        orc_win_value(&cols[fns[f].col], out_rows[pe - 1], idx,
+                        if (fr < fl) { out_null[idx] = 1; break; }
+                        orc_win_value(&cols[fns[f].col], out_rows[fr], idx,
                                       out_i, out_d, out_null);
                         break;
                     case BK_WIN_NTH_VALUE: {
-                        int64_t jj = ps + fns[f].param - 1;
-                        if (jj >= ps && jj < pe)
+                        int64_t jj = fl + fns[f].param - 1;
+                        if (jj >= fl && jj <= fr)
                             orc_win_value(&cols[fns[f].col], out_rows[jj], idx,
                                           out_i, out_d, out_null);
                         else out_null[idx] = 1;

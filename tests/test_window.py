@@ -334,3 +334,111 @@ def test_gpu_window_fuzz(eng, orc, case_seed):
         d = np.abs(got["out_d"] - exp["out_d"])
         tol = 1e-10 * (np.abs(exp["out_d"]) + 100)
         assert np.all(d[mask] <= tol[mask]), ctx
+
+
+def brute_frame(cols, valids, part_col, order, fns, pre, fol):
+    """ROWS-frame brute force for SUM/COUNT/AVG/first/last/nth."""
+    idx, _ = brute_window(cols, valids, part_col, order,
+                          [("row_number", -1)])
+    n = len(idx)
+
+    def null(c, r):
+        return valids[c] is not None and valids[c][r] == 0
+
+    def peq(a, b):
+        if part_col < 0:
+            return True
+        na, nb = null(part_col, a), null(part_col, b)
+        return na == nb and (na or cols[part_col][a] == cols[part_col][b])
+
+    out = {f: [None] * n for f in range(len(fns))}
+    ps = 0
+    while ps < n:
+        pe = ps + 1
+        while pe < n and peq(idx[pe], idx[pe - 1]):
+            pe += 1
+        for j in range(ps, pe):
+            fl = ps if pre < 0 else max(ps, j - pre)
+            fr = pe - 1 if fol < 0 else min(pe - 1, j + fol)
+            frame_rows = [idx[k] for k in range(fl, fr + 1)]
+            for f, (name, col, *rest) in enumerate(fns):
+                param = rest[0] if rest else 0
+                if name == "count_star":
+                    out[f][j] = len(frame_rows)
+                elif name == "count":
+                    out[f][j] = sum(0 if null(col, r) else 1
+                                    for r in frame_rows)
+                elif name in ("sum", "avg"):
+                    vv = [cols[col][r] for r in frame_rows if not null(col, r)]
+                    if not vv:
+                        out[f][j] = None
+                    elif name == "sum":
+                        out[f][j] = np.sum(np.array(vv))
+                    else:
+                        out[f][j] = float(np.mean(np.array(vv,
+                                                           dtype=np.float64)))
+                elif name == "first_value":
+                    r = frame_rows[0]
+                    out[f][j] = None if null(col, r) else cols[col][r]
+                elif name == "last_value":
+                    r = frame_rows[-1]
+                    out[f][j] = None if null(col, r) else cols[col][r]
+                elif name == "nth_value":
+                    k = param - 1
+                    if 0 <= k < len(frame_rows):
+                        r = frame_rows[k]
+                        out[f][j] = None if null(col, r) else cols[col][r]
+                    else:
+                        out[f][j] = None
+        ps = pe
+    return idx, out
+
+
+FRAME_FNS = [("count_star", -1), ("count", 2), ("sum", 2), ("avg", 2),
+             ("first_value", 2), ("last_value", 2), ("nth_value", 2, 2)]
+
+
+@pytest.mark.parametrize("frame", [(-1, 0), (3, 0), (0, 3), (2, 2)])
+def test_oracle_rows_frame_vs_brute(orc, frame):
+    specs = [(TYPE_INT64, D_UNI, 0, 10, 0),
+             (TYPE_INT64, D_UNI, 0, 30, 100_000),
+             (TYPE_INT64, D_UNI, -100, 100, 200_000)]
+    cols, valids, types = gen(orc, specs, 2500)
+    order = [(1, 1, 1)]
+    res = orc.window(cols, valids, types,
+                     [(W[f[0]], f[1], f[2] if len(f) > 2 else 0)
+                      for f in FRAME_FNS],
+                     part_col=0, order=order, frame=frame)
+    idx, brute = brute_frame(cols, valids, 0, order, FRAME_FNS, *frame)
+    check_against_brute(res, types, FRAME_FNS, idx, brute)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("frame", [(-1, 0), (5, 0), (1, 4)])
+def test_gpu_rows_frame_parity(eng, orc, frame):
+    specs = [(TYPE_INT64, D_UNI, 0, 200, 50_000),
+             (TYPE_INT64, D_UNI, 0, 40, 0),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 150_000),
+             (TYPE_INT64, D_UNI, -500, 500, 100_000)]
+    fns = [("count_star", -1), ("sum", 3), ("sum", 2), ("avg", 2),
+           ("count", 3), ("first_value", 3), ("last_value", 2),
+           ("nth_value", 3, 3)]
+    n = 150_000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED + 5)
+        got = eng.window(t, fns, part_col=0, order=[(1, 1, 1)], frame=frame)
+    finally:
+        t.free()
+    cols, valids, types = gen(orc, specs, n, SEED + 5)
+    exp = orc.window(cols, valids, types,
+                     [(W[f[0]], f[1], f[2] if len(f) > 2 else 0) for f in fns],
+                     part_col=0, order=[(1, 1, 1)], frame=frame)
+    assert got["n"] == exp["n"]
+    assert np.array_equal(got["rowids"], exp["rowids"])
+    assert np.array_equal(got["out_null"], exp["out_null"])
+    assert np.array_equal(got["out_i"], exp["out_i"])
+    mask = exp["out_null"] == 0
+    d = np.abs(got["out_d"] - exp["out_d"])
+    tol = 1e-9 * (np.abs(exp["out_d"]) + 100)
+    assert np.all(d[mask] <= tol[mask])
