@@ -285,7 +285,14 @@ class GpuEngine:
         out_i = np.zeros(len(fns) * cap, dtype=np.int64)
         out_d = np.zeros(len(fns) * cap, dtype=np.float64)
         out_null = np.zeros(len(fns) * cap, dtype=np.uint8)
-        fr, fpre, ffol = (1, frame[0], frame[1]) if frame else (0, -1, -1)
+        if frame is None:
+            fr, fpre, ffol = 0, -1, -1
+        elif frame == "range_upc":    # RANGE UNBOUNDED PRECEDING..CURRENT ROW
+            fr, fpre, ffol = 2, -1, -1
+        elif frame == "range_crf":    # RANGE CURRENT ROW..UNBOUNDED FOLLOWING
+            fr, fpre, ffol = 3, -1, -1
+        else:
+            fr, fpre, ffol = 1, frame[0], frame[1]
         n = self.lib.bkgpu_window(
             table.handle, C.byref(q), part_col, oarr, len(order),
             farr, len(fns), fr, fpre, ffol, row_begin, row_end,

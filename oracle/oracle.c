@@ -907,10 +907,29 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                 /* ROWS frame (RowFrameWindowProcessor): per-row recompute
                  * over [fl, fr] clamped inside the partition */
                 const OrcCol* fc = fns[f].col >= 0 ? &cols[fns[f].col] : NULL;
+                int64_t rgl = ps, rgr = ps - 1;  /* current peer range */
                 for (int64_t j = ps; j < pe; j++) {
-                    int64_t fl = f_pre >= 0 && j - f_pre > ps ? j - f_pre : ps;
-                    int64_t fr = f_fol >= 0 && j + f_fol < pe - 1 ? j + f_fol
-                                                                  : pe - 1;
+                    if (j > rgr) {   /* advance the peer group for RANGE */
+                        rgl = j; rgr = j;
+                        while (rgr + 1 < pe) {
+                            int same = 1;
+                            for (int k = 0; k < norder; k++)
+                                if (!orc_cell_eq(&cols[order[k].col],
+                                                 out_rows[rgr + 1],
+                                                 out_rows[j]))
+                                    { same = 0; break; }
+                            if (!same) break;
+                            rgr++;
+                        }
+                    }
+                    int64_t fl, fr;
+                    if (frame_rows == 2) { fl = ps; fr = rgr; }
+                    else if (frame_rows == 3) { fl = rgl; fr = pe - 1; }
+                    else {
+                        fl = f_pre >= 0 && j - f_pre > ps ? j - f_pre : ps;
+                        fr = f_fol >= 0 && j + f_fol < pe - 1 ? j + f_fol
+                                                              : pe - 1;
+                    }
                     int64_t idx = (int64_t)f * n + j;
                     out_i[idx] = 0; out_d[idx] = 0.0; out_null[idx] = 0;
                     int64_t fcnt = 0, fvi = 0;
@@ -1036,9 +1055,13 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                 int64_t idx = (int64_t)f * n + j;
                 out_i[idx] = 0; out_d[idx] = 0.0; out_null[idx] = 0;
                 int64_t fl = ps, fr = pe - 1;
-                if (frame_rows) {
+                if (frame_rows == 1) {
                     if (f_pre >= 0 && j - f_pre > ps) fl = j - f_pre;
                     if (f_fol >= 0 && j + f_fol < pe - 1) fr = j + f_fol;
+                } else if (frame_rows == 2) {
+                    fr = peer_end - 1;
+                } else if (frame_rows == 3) {
+                    fl = peer_head;
                 }
                 switch (ft) {
                     case BK_WIN_ROW_NUMBER: out_i[idx] = j - ps + 1; break;

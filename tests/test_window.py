@@ -442,3 +442,93 @@ def test_gpu_rows_frame_parity(eng, orc, frame):
     d = np.abs(got["out_d"] - exp["out_d"])
     tol = 1e-9 * (np.abs(exp["out_d"]) + 100)
     assert np.all(d[mask] <= tol[mask])
+
+
+def brute_range_frame(cols, valids, part_col, order, fns, mode):
+    """RANGE frames: upc = [ps, peer_end), crf = [peer_head, pe)."""
+    idx, _ = brute_window(cols, valids, part_col, order, [("row_number", -1)])
+    n = len(idx)
+
+    def null(c, r):
+        return valids[c] is not None and valids[c][r] == 0
+
+    def peq(a, b):
+        if part_col < 0:
+            return True
+        na, nb = null(part_col, a), null(part_col, b)
+        return na == nb and (na or cols[part_col][a] == cols[part_col][b])
+
+    def oeq(a, b):
+        for col, _, _ in order:
+            na, nb = null(col, a), null(col, b)
+            if na != nb:
+                return False
+            if not na and cols[col][a] != cols[col][b]:
+                return False
+        return True
+
+    out = {f: [None] * n for f in range(len(fns))}
+    ps = 0
+    while ps < n:
+        pe = ps + 1
+        while pe < n and peq(idx[pe], idx[pe - 1]):
+            pe += 1
+        for j in range(ps, pe):
+            lo, hi = j, j
+            while lo > ps and oeq(idx[lo - 1], idx[j]):
+                lo -= 1
+            while hi + 1 < pe and oeq(idx[hi + 1], idx[j]):
+                hi += 1
+            fl, fr = (ps, hi) if mode == "range_upc" else (lo, pe - 1)
+            frame_rows = [idx[k] for k in range(fl, fr + 1)]
+            for f, (name, col, *rest) in enumerate(fns):
+                if name == "count_star":
+                    out[f][j] = len(frame_rows)
+                elif name == "sum":
+                    vv = [cols[col][r] for r in frame_rows if not null(col, r)]
+                    out[f][j] = np.sum(np.array(vv)) if vv else None
+        ps = pe
+    return idx, out
+
+
+@pytest.mark.parametrize("mode", ["range_upc", "range_crf"])
+def test_oracle_range_frame_vs_brute(orc, mode):
+    specs = [(TYPE_INT64, D_UNI, 0, 8, 0),
+             (TYPE_INT64, D_UNI, 0, 12, 100_000),   # coarse: many ties
+             (TYPE_INT64, D_UNI, -50, 50, 0)]
+    cols, valids, types = gen(orc, specs, 2000)
+    order = [(1, 1, 1)]
+    fns = [("count_star", -1), ("sum", 2)]
+    res = orc.window(cols, valids, types,
+                     [(W[f[0]], f[1], 0) for f in fns],
+                     part_col=0, order=order, frame=mode)
+    idx, brute = brute_range_frame(cols, valids, 0, order, fns, mode)
+    check_against_brute(res, types, fns, idx, brute)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("mode", ["range_upc", "range_crf"])
+def test_gpu_range_frame_parity(eng, orc, mode):
+    specs = [(TYPE_INT64, D_UNI, 0, 100, 0),
+             (TYPE_INT64, D_UNI, 0, 15, 50_000),
+             (TYPE_INT64, D_UNI, -500, 500, 100_000)]
+    fns = [("count_star", -1), ("sum", 2), ("avg", 2), ("last_value", 2)]
+    n = 120_000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED + 9)
+        got = eng.window(t, fns, part_col=0, order=[(1, 1, 1)], frame=mode)
+    finally:
+        t.free()
+    cols, valids, types = gen(orc, specs, n, SEED + 9)
+    exp = orc.window(cols, valids, types,
+                     [(W[f[0]], f[1], 0) for f in fns],
+                     part_col=0, order=[(1, 1, 1)], frame=mode)
+    assert got["n"] == exp["n"]
+    assert np.array_equal(got["rowids"], exp["rowids"])
+    assert np.array_equal(got["out_null"], exp["out_null"])
+    assert np.array_equal(got["out_i"], exp["out_i"])
+    mask = exp["out_null"] == 0
+    d = np.abs(got["out_d"] - exp["out_d"])
+    tol = 1e-9 * (np.abs(exp["out_d"]) + 100)
+    assert np.all(d[mask] <= tol[mask])
