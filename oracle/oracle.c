@@ -1125,6 +1125,7 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
 /* small exported helpers for tests                                    */
 /* ------------------------------------------------------------------ */
 
+ORC_EXPORT int64_t orc_scalar_fn(int32_t fn, int64_t v) { return bk_scalar_fn(fn, v); }
 ORC_EXPORT uint64_t orc_encode_i64(int64_t v)  { return bk_enc_i64(v); }
 ORC_EXPORT int64_t  orc_decode_i64(uint64_t u) { return bk_dec_i64(u); }
 ORC_EXPORT uint64_t orc_encode_f64(double v)   { return bk_enc_f64(v); }
