@@ -33,10 +33,14 @@
  *                  is_null_first decides NULL order regardless of asc/desc;
  *                  ties broken by arrival index, topn_sorter.h:46-54)
  *
- * Parity pinning: the JSON fixtures under tests/golden hold (a) vectors produced by the
- * reference's own key_encoder.h compiled in-place (oracle/_ref, Makefile
- * target 'ref'), and (b) expectation vectors re-expressed from the
- * reference's unit tests test/test_key_encoder.cpp and test/test_expr_value.cpp.
+ * Parity pinning: the JSON fixtures under tests/golden hold vectors produced
+ * by the reference's OWN code compiled in place (oracle/_ref, Makefile target
+ * 'ref'): key_encoder.h encode/decode (keyenc_golden.json, covering the
+ * expectation classes of test/test_key_encoder.cpp) and datetime.h's packed
+ * bit-layout extractions (datetime_golden.json). The reference's
+ * test_expr_value.cpp cases are proto-round-trip self-compares and add no
+ * cross-implementation signal; ExprValue compare/arith semantics are instead
+ * pinned by restatement citations + the numpy cross-checks in tests/.
  *
  * Internally the oracle groups by a fixed-width encoded key (null-flag byte +
  * order-preserving u64 per group column — the same internal key the GPU path
