@@ -110,6 +110,26 @@ __device__ __forceinline__ double cell_f64(const DevCol& c, int64_t r) {
  * dependent loads serialize ~900-cycle HBM latencies per conjunct, while at
  * the bench selectivities nearly every cacheline is touched anyway, so eager
  * issue trades no traffic for full memory-level parallelism. */
+/* IN-list membership: inline list or (big lists) sorted device array via
+ * binary search (bk_common.h BK_OP_IN) */
+__device__ __forceinline__ bool in_list_hit(const BkConjunct& cj, int64_t v) {
+    if (cj.n_in <= BK_MAX_INLIST) {
+        bool found = false;
+        for (int32_t m = 0; m < cj.n_in; m++)
+            found = found || (cj.in_list[m] == v);
+        return found;
+    }
+    const int64_t* a = (const int64_t*)(uintptr_t)cj.lit_i;
+    int32_t lo = 0, hi = cj.n_in - 1;
+    while (lo <= hi) {
+        int32_t mid = (lo + hi) >> 1;
+        int64_t x = a[mid];
+        if (x == v) return true;
+        if (x < v) lo = mid + 1; else hi = mid - 1;
+    }
+    return false;
+}
+
 __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpec& q,
                                            int64_t r) {
     /* manually scalarized so the staged values live in registers (indexed
@@ -140,9 +160,7 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
                            ((bm[(VI) >> 3] >> ((VI) & 7)) & 1);             \
                 pass = cj.op == BK_OP_IN_BITMAP ? hit : !hit;               \
             } else if (cj.op >= BK_OP_IN) {                                 \
-                bool found = false;                                         \
-                for (int32_t m = 0; m < cj.n_in; m++)                       \
-                    found = found || (cj.in_list[m] == (VI));               \
+                bool found = in_list_hit(cj, (VI));                         \
                 pass = cj.op == BK_OP_IN ? found : !found;                  \
             } else {                                                        \
                 int cmp = (cj.cmp_type == BK_DOUBLE)                        \
@@ -179,9 +197,7 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
         } else if (cj.op >= BK_OP_IN) {
             int64_t v = cell_i64(c, r);
             if (cj.fn) v = bk_scalar_fn(cj.fn, v);
-            bool found = false;
-            for (int32_t m = 0; m < cj.n_in; m++)
-                found = found || (cj.in_list[m] == v);
+            bool found = in_list_hit(cj, v);
             pass = cj.op == BK_OP_IN ? found : !found;
         } else {
             int cmp;

@@ -93,11 +93,15 @@ class QueryPlan:
             if cj.op >= 8:  # bitmap membership: lit = (device_ptr, n_bits)
                 cj.cmp_type = TYPE_INT64
                 cj.lit_i, cj.n_in = int(lit[0]), int(lit[1])
-            elif cj.op >= 6:  # IN / NOT IN: lit is a list of int/dict literals
-                cj.cmp_type = TYPE_INT64
-                cj.n_in = len(lit)
-                for m, v in enumerate(lit):
-                    cj.in_list[m] = int(v)
+            elif cj.op >= 6:  # IN / NOT IN: small lists inline; big lists
+                cj.cmp_type = TYPE_INT64  # as (device_ptr, n) sorted arrays
+                if isinstance(lit, tuple) and len(lit) == 2 and \
+                        isinstance(lit[0], int) and lit[1] > len(q.conjuncts[i].in_list):
+                    cj.lit_i, cj.n_in = int(lit[0]), int(lit[1])
+                else:
+                    cj.n_in = len(lit)
+                    for m, v in enumerate(lit):
+                        cj.in_list[m] = int(v)
             elif ct == TYPE_DOUBLE or isinstance(lit, float):
                 cj.cmp_type = TYPE_DOUBLE
                 cj.lit_d = float(lit)

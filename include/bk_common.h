@@ -60,7 +60,10 @@ typedef enum BkCmpOp {
     BK_OP_LE = 5,
     /* IN-list predicates (src/expr/predicate.h InPredicate; NULL operand =>
      * NULL => row rejected; literal lists carry no NULLs — the planner only
-     * pushes literal IN lists) */
+     * pushes literal IN lists). n_in <= BK_MAX_INLIST: literals inline in
+     * in_list. n_in > BK_MAX_INLIST: lit_i carries a pointer (device pointer
+     * for the GPU engine, host pointer for the oracle) to a SORTED int64
+     * array of n_in literals, probed by binary search. */
     BK_OP_IN     = 6,
     BK_OP_NOT_IN = 7,
     /* dict-code bitmap membership: lit_i carries a pointer (device pointer

@@ -85,10 +85,14 @@ def make_query(conjuncts=(), group=(), aggs=(), col_types=None):
         cj.fn = cjt[4] if len(cjt) > 4 else 0
         if op >= 8:  # bitmap membership: lit = (host_ptr, n_bits)
             cj.lit_i, cj.n_in = int(lit[0]), int(lit[1])
-        elif op >= 6:  # IN / NOT IN: lit is a list
-            cj.n_in = len(lit)
-            for m, v in enumerate(lit):
-                cj.in_list[m] = int(v)
+        elif op >= 6:  # IN / NOT IN: small lists inline; big lists as
+            if isinstance(lit, tuple) and len(lit) == 2 and \
+                    isinstance(lit[0], int) and lit[1] > 16:
+                cj.lit_i, cj.n_in = int(lit[0]), int(lit[1])  # (host_ptr, n)
+            else:
+                cj.n_in = len(lit)
+                for m, v in enumerate(lit):
+                    cj.in_list[m] = int(v)
         elif cmp_type == TYPE_DOUBLE:
             cj.lit_d = float(lit)
             cj.lit_i = 0

@@ -148,12 +148,23 @@ static int orc_row_passes(const OrcCol* cols, const BkQuerySpec* q, int64_t r) {
             int hit = v >= 0 && v < cj->n_in && ((bm[v >> 3] >> (v & 7)) & 1);
             pass = (cj->op == BK_OP_IN_BITMAP) ? hit : !hit;
         } else if (cj->op == BK_OP_IN || cj->op == BK_OP_NOT_IN) {
-            /* predicate.h InPredicate semantics over literal lists */
+            /* predicate.h InPredicate semantics over literal lists; big
+             * lists (> BK_MAX_INLIST) are a sorted host array in lit_i */
             int64_t v = cell_i64(c, r);
             if (cj->fn) v = bk_scalar_fn(cj->fn, v);
             int found = 0;
-            for (int32_t m = 0; m < cj->n_in; m++)
-                if (cj->in_list[m] == v) { found = 1; break; }
+            if (cj->n_in <= BK_MAX_INLIST) {
+                for (int32_t m = 0; m < cj->n_in; m++)
+                    if (cj->in_list[m] == v) { found = 1; break; }
+            } else {
+                const int64_t* a = (const int64_t*)(uintptr_t)cj->lit_i;
+                int32_t lo = 0, hi = cj->n_in - 1;
+                while (lo <= hi) {
+                    int32_t mid = (lo + hi) >> 1;
+                    if (a[mid] == v) { found = 1; break; }
+                    if (a[mid] < v) lo = mid + 1; else hi = mid - 1;
+                }
+            }
             pass = (cj->op == BK_OP_IN) ? found : !found;
         } else {
             switch ((BkCmpOp)cj->op) {

@@ -386,3 +386,48 @@ def test_string_minmax_dict_order(eng, orc):
                         [(0, "<", 1 << 19)], [0], aggs,
                         expected_groups=1 << 16)
     assert_parity(got, exp, aggs, [s[0] for s in specs])
+
+
+def test_large_in_list(eng, orc):
+    """IN list beyond BK_MAX_INLIST: sorted device/host array, binary
+    search (bk_common.h BK_OP_IN big-list convention)."""
+    from baikaldb_amd import QueryPlan
+    rng = np.random.default_rng(13)
+    vals = np.unique(rng.integers(0, 5000, 700)).astype(np.int64)
+    specs = [(TYPE_INT64, D_UNI, 0, 5000, 0),
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 120_000),
+             (TYPE_INT64, D_UNI, 0, 1000, 0)]
+    n = 200_000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED)
+        dev = eng.upload_bytes(vals.tobytes())
+        try:
+            plan = QueryPlan(t.col_types,
+                             conjuncts=[(0, "in", (dev, len(vals))),
+                                        (1, ">", 1 << 28)],
+                             group=[2], aggs=[("count_star", -1), ("sum", 0)])
+            res = eng.filter_agg(t, plan, expected_groups=1 << 11)
+            got = res.fetch(sorted=True)
+            res.free()
+        finally:
+            eng.free_ptr(dev)
+    finally:
+        t.free()
+    specs_c = (BkColSpec * len(specs))()
+    for i, s in enumerate(specs):
+        (specs_c[i].col_type, specs_c[i].dist, specs_c[i].p0, specs_c[i].p1,
+         specs_c[i].null_frac_x1e6) = s
+    cols, valids = orc.generate_table(list(specs_c), n, SEED)
+    types = [s[0] for s in specs]
+    host = vals.ctypes.data
+    from oracle.bindings import make_query as mq
+    q = mq([(0, 6, TYPE_INT64, (host, len(vals))),
+            (1, 2, TYPE_INT64, 1 << 28)], [2],
+           [(0, -1), (2, 0)], types)
+    exp = orc.filter_agg(cols, valids, types, q, nthreads=4, dict_seed=SEED)
+    # numpy cross-check of the oracle too
+    sel = np.isin(cols[0], vals) & (cols[1] > (1 << 28)) & \
+        (valids[1] != 0 if valids[1] is not None else True)
+    assert exp["rows_passed"] == int(sel.sum())
+    assert_parity(got, exp, [("count_star", -1), ("sum", 0)], types)
