@@ -356,75 +356,145 @@ public:
         _q = q;
         int64_t expected = _desc.expected_groups > 0 ? _desc.expected_groups : 65536;
         /* DISTINCT aggs: the reference's planner rewrite (agg_node.cpp:
-         * 247-258) — level 1 groups by (user keys + distinct col), then
-         * bkgpu_agg_rollup folds the dedup key out. */
-        bool has_distinct = false;
-        int dcol = -1;
+         * 247-258) — per distinct column, level 1 groups by (user keys +
+         * that column) and bkgpu_agg_rollup folds the dedup key out. The
+         * reference's MULTI_COUNT_DISTINCT (several distinct columns) runs
+         * one such pass per column; every pass shares the same filter and
+         * group key so the canonical-sorted group sets align row for row
+         * and the output columns stitch together. */
+        std::vector<int> dcols;            /* distinct columns, deduped */
+        std::vector<int> agg_dcol(q.n_aggs, -1);
         for (int a = 0; a < q.n_aggs; a++) {
             int at = q.aggs[a].agg_type;
-            if (at == BK_AGG_COUNT_DISTINCT || at == BK_AGG_SUM_DISTINCT) {
-                has_distinct = true;
-                if (dcol >= 0 && dcol != q.aggs[a].col) {
-                    state->error_msg = "one DISTINCT column supported";
-                    return -1;
-                }
-                dcol = q.aggs[a].col;
-            }
+            if (at != BK_AGG_COUNT_DISTINCT && at != BK_AGG_SUM_DISTINCT)
+                continue;
+            int c = q.aggs[a].col;
+            size_t j = 0;
+            while (j < dcols.size() && dcols[j] != c) j++;
+            if (j == dcols.size()) dcols.push_back(c);
+            agg_dcol[a] = (int)j;
         }
-        BkgAggOut* out = nullptr;
-        if (has_distinct) {
-            if (q.n_group > 1 || dcol < 0) {
+        bool has_distinct = !dcols.empty();
+        /* passes: pass 0 = plain aggs (or everything when no distinct);
+         * pass 1+j = distinct col j. src[a] = (pass, idx-in-pass). */
+        struct Src { int pass; int idx; };
+        std::vector<Src> src(q.n_aggs);
+        std::vector<BkgAggOut*> outs;
+        int64_t nrows_t = bkgpu_table_nrows(t);
+        if (!has_distinct) {
+            BkgAggOut* out = bkgpu_filter_agg(t, &q, 0, nrows_t, expected);
+            if (!out) { state->error_msg = bkgpu_last_error(); return -1; }
+            outs.push_back(out);
+            for (int a = 0; a < q.n_aggs; a++) src[a] = {0, a};
+        } else {
+            if (q.n_group > 1) {
                 state->error_msg = "DISTINCT aggs support <= 1 group key";
                 return -1;
             }
-            BkQuerySpec q1 = q;
-            q1.n_group = q.n_group + 1;
-            q1.group_cols[q.n_group] = dcol;
-            q1.group_types[q.n_group] = bkgpu_table_col_type(t, dcol);
-            int32_t src_idx[BK_MAX_AGGS];
-            int np = 0;
-            q1.n_aggs = 0;
+            /* pass 0: the plain aggs (always run — it also anchors the
+             * group set when a group has rows but only NULL d values) */
+            BkQuerySpec q0 = q;
+            q0.n_aggs = 0;
             for (int a = 0; a < q.n_aggs; a++) {
-                int at = q.aggs[a].agg_type;
-                if (at == BK_AGG_COUNT_DISTINCT || at == BK_AGG_SUM_DISTINCT) {
-                    src_idx[a] = -1;
-                } else {
-                    q1.aggs[np] = q.aggs[a];
-                    q1.agg_in_types[np] = q.agg_in_types[a];
-                    src_idx[a] = np++;
-                }
+                if (agg_dcol[a] >= 0) continue;
+                src[a] = {0, q0.n_aggs};
+                q0.aggs[q0.n_aggs] = q.aggs[a];
+                q0.agg_in_types[q0.n_aggs] = q.agg_in_types[a];
+                q0.n_aggs++;
             }
-            if (np == 0) {  /* level 1 still needs one state column */
+            if (q0.n_aggs == 0) {
+                q0.aggs[0].agg_type = BK_AGG_COUNT_STAR;
+                q0.aggs[0].col = -1;
+                q0.agg_in_types[0] = BK_INT64;
+                q0.n_aggs = 1;
+            }
+            BkgAggOut* p0 = bkgpu_filter_agg(t, &q0, 0, nrows_t, expected);
+            if (!p0) { state->error_msg = bkgpu_last_error(); return -1; }
+            outs.push_back(p0);
+            for (size_t j = 0; j < dcols.size(); j++) {
+                BkQuerySpec q1 = q;
+                q1.n_group = q.n_group + 1;
+                q1.group_cols[q.n_group] = dcols[j];
+                q1.group_types[q.n_group] = bkgpu_table_col_type(t, dcols[j]);
                 q1.aggs[0].agg_type = BK_AGG_COUNT_STAR;
                 q1.aggs[0].col = -1;
                 q1.agg_in_types[0] = BK_INT64;
-                np = 1;
+                q1.n_aggs = 1;
+                /* level-2 spec: only this column's distinct aggs */
+                BkQuerySpec q2 = q;
+                q2.n_aggs = 0;
+                int32_t src_idx[BK_MAX_AGGS];
+                for (int a = 0; a < q.n_aggs; a++) {
+                    if (agg_dcol[a] != (int)j) continue;
+                    src[a] = {(int)j + 1, q2.n_aggs};
+                    src_idx[q2.n_aggs] = -1;
+                    q2.aggs[q2.n_aggs] = q.aggs[a];
+                    q2.agg_in_types[q2.n_aggs] = q.agg_in_types[a];
+                    q2.n_aggs++;
+                }
+                BkgAggOut* l1 = bkgpu_filter_agg(t, &q1, 0, nrows_t,
+                                                 expected * 16);
+                BkgAggOut* r = l1 ? bkgpu_agg_rollup(l1, &q2, src_idx,
+                                                     expected) : nullptr;
+                if (l1) bkgpu_agg_free(l1);
+                if (!r) {
+                    state->error_msg = bkgpu_last_error();
+                    for (auto* o : outs) bkgpu_agg_free(o);
+                    return -1;
+                }
+                outs.push_back(r);
             }
-            q1.n_aggs = np;
-            BkgAggOut* l1 = bkgpu_filter_agg(t, &q1, 0, bkgpu_table_nrows(t),
-                                             expected * 16);
-            if (!l1) { state->error_msg = bkgpu_last_error(); return -1; }
-            out = bkgpu_agg_rollup(l1, &q, src_idx, expected);
-            bkgpu_agg_free(l1);
-        } else {
-            out = bkgpu_filter_agg(t, &q, 0, bkgpu_table_nrows(t), expected);
         }
-        if (!out) { state->error_msg = bkgpu_last_error(); return -1; }
-        state->inc_num_scan_rows(bkgpu_table_nrows(t));
-        state->inc_num_filter_rows(bkgpu_table_nrows(t) - bkgpu_agg_rows_passed(out));
-        /* fetch finalized groups to host (canonical key order) */
-        int64_t n = bkgpu_agg_ngroups(out);
+        state->inc_num_scan_rows(nrows_t);
+        state->inc_num_filter_rows(nrows_t - bkgpu_agg_rows_passed(outs[0]));
+        /* fetch every pass in canonical key order; group sets align (same
+         * filter + same user keys), assemble the output columns */
+        int64_t n = bkgpu_agg_ngroups(outs[0]);
         _g.n = n;
         _g.flags.resize(n ? n : 1);
         _g.enc.resize((n ? n : 1) * BK_MAX_GROUP);
         _g.out_i.resize((size_t)(n ? n : 1) * q.n_aggs);
         _g.out_d.resize((size_t)(n ? n : 1) * q.n_aggs);
         _g.out_has.resize((size_t)(n ? n : 1) * q.n_aggs);
-        int64_t got = bkgpu_agg_fetch(out, /*sorted=*/1, n, _g.flags.data(),
-                                      _g.enc.data(), _g.out_i.data(),
-                                      _g.out_d.data(), _g.out_has.data());
-        bkgpu_agg_free(out);
-        if (got < 0) { state->error_msg = bkgpu_last_error(); return -1; }
+        int64_t got = -1;
+        for (size_t p = 0; p < outs.size(); p++) {
+            int pn_aggs = 0;
+            for (int a = 0; a < q.n_aggs; a++)
+                if (src[a].pass == (int)p && src[a].idx + 1 > pn_aggs)
+                    pn_aggs = src[a].idx + 1;
+            if (p == 0 && pn_aggs == 0) pn_aggs = 1;  /* synthetic count(*) */
+            std::vector<uint8_t> flags(n ? n : 1);
+            std::vector<uint64_t> enc((n ? n : 1) * BK_MAX_GROUP);
+            std::vector<int64_t> oi((size_t)(n ? n : 1) * pn_aggs);
+            std::vector<double> od((size_t)(n ? n : 1) * pn_aggs);
+            std::vector<uint8_t> oh((size_t)(n ? n : 1) * pn_aggs);
+            int64_t gp = bkgpu_agg_fetch(outs[p], /*sorted=*/1, n,
+                                         flags.data(), enc.data(), oi.data(),
+                                         od.data(), oh.data());
+            if (gp < 0 || (p > 0 && gp != got)) {
+                state->error_msg = gp < 0 ? bkgpu_last_error()
+                                          : "distinct pass group mismatch";
+                for (auto* o : outs) bkgpu_agg_free(o);
+                return -1;
+            }
+            if (p == 0) {
+                got = gp;
+                memcpy(_g.flags.data(), flags.data(), (size_t)(gp ? gp : 1));
+                memcpy(_g.enc.data(), enc.data(),
+                       (size_t)(gp ? gp : 1) * BK_MAX_GROUP * 8);
+            }
+            for (int a = 0; a < q.n_aggs; a++) {
+                if (src[a].pass != (int)p) continue;
+                for (int64_t g = 0; g < gp; g++) {
+                    size_t di = (size_t)a * gp + g;
+                    size_t si = (size_t)src[a].idx * gp + g;
+                    _g.out_i[di] = oi[si];
+                    _g.out_d[di] = od[si];
+                    _g.out_has[di] = oh[si];
+                }
+            }
+        }
+        for (auto* o : outs) bkgpu_agg_free(o);
         _g.n = got;
         _iter = 0;
         return 0;

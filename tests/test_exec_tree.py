@@ -270,3 +270,38 @@ def test_window_through_exec_surface(eng, orc):
     assert _np.array_equal(nulls[:, 5] != 0, lagnull)
     m = ~lagnull
     assert _np.array_equal(vi[:, 5][m], lag[m])
+
+
+def test_multi_column_distinct_through_exec_surface(eng, orc):
+    """COUNT(DISTINCT c2), COUNT(DISTINCT c0), SUM(c2) GROUP BY c1 — the
+    reference's MULTI_COUNT_DISTINCT shape (one rollup pass per distinct
+    column, stitched on the aligned canonical group order)."""
+    from baikaldb_amd import exec as bx
+    import numpy as _np
+    t, cols, valids, types = make_table(eng, orc, n=50_000)
+    try:
+        nodes = [bx.agg_node(group=[1],
+                             aggs=[("count_star", -1),
+                                   ("count_distinct", 2),
+                                   ("sum", 2),
+                                   ("count_distinct", 0),
+                                   ("sum_distinct", 2)]),
+                 bx.filter_node(types, [(0, "<", int((1 << 31) * 0.7))]),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
+    sel = cols[0] < int((1 << 31) * 0.7)
+    g = cols[1][sel]
+    vals = _np.unique(g)
+    assert tags.shape[0] == len(vals)
+    for r, gv in enumerate(vals):
+        m = sel & (cols[1] == gv)
+        assert vi[r, 1] == m.sum()                              # count_star
+        assert vi[r, 2] == len(_np.unique(cols[2][m]))          # cd(c2)
+        assert vi[r, 3] == cols[2][m].sum()                     # sum(c2)
+        assert vi[r, 4] == len(_np.unique(cols[0][m]))          # cd(c0)
+        assert vi[r, 5] == _np.unique(cols[2][m]).sum()         # sd(c2)
