@@ -918,7 +918,7 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
         for (int f = 0; f < nfns; f++) {
             int ft = fns[f].fn_type;
             if (ft > BK_WIN_MAX) continue;
-            if (frame_rows && ft <= BK_WIN_AVG) {
+            if (frame_rows && ft <= BK_WIN_MAX) {
                 /* ROWS frame (RowFrameWindowProcessor): per-row recompute
                  * over [fl, fr] clamped inside the partition */
                 const OrcCol* fc = fns[f].col >= 0 ? &cols[fns[f].col] : NULL;
@@ -953,7 +953,6 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                         int64_t r = out_rows[jj];
                         if (ft == BK_WIN_COUNT_STAR) { fcnt++; continue; }
                         if (!cell_is_valid(fc, r)) continue;
-                        fcnt++;
                         if (ft == BK_WIN_SUM) {
                             if (fc->type == BK_DOUBLE)
                                 fvd += ((double*)fc->data)[r];
@@ -961,7 +960,20 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                                                  (uint64_t)cell_i64(fc, r));
                         } else if (ft == BK_WIN_AVG) {
                             fvd += cell_f64_cast(fc, r);
+                        } else if (ft == BK_WIN_MIN || ft == BK_WIN_MAX) {
+                            if (fc->type == BK_DOUBLE) {
+                                double v = ((double*)fc->data)[r];
+                                if (!fcnt || (ft == BK_WIN_MIN ? v < fvd
+                                                               : v > fvd))
+                                    fvd = v;
+                            } else {
+                                int64_t v = cell_i64(fc, r);
+                                if (!fcnt || (ft == BK_WIN_MIN ? v < fvi
+                                                               : v > fvi))
+                                    fvi = v;
+                            }
                         }
+                        fcnt++;
                     }
                     if (ft == BK_WIN_COUNT_STAR || ft == BK_WIN_COUNT) {
                         out_i[idx] = fcnt;

@@ -377,6 +377,10 @@ def brute_frame(cols, valids, part_col, order, fns, pre, fol):
                     else:
                         out[f][j] = float(np.mean(np.array(vv,
                                                            dtype=np.float64)))
+                elif name in ("min", "max"):
+                    vv = [cols[col][r] for r in frame_rows if not null(col, r)]
+                    out[f][j] = (min(vv) if name == "min" else max(vv)) \
+                        if vv else None
                 elif name == "first_value":
                     r = frame_rows[0]
                     out[f][j] = None if null(col, r) else cols[col][r]
@@ -395,7 +399,8 @@ def brute_frame(cols, valids, part_col, order, fns, pre, fol):
 
 
 FRAME_FNS = [("count_star", -1), ("count", 2), ("sum", 2), ("avg", 2),
-             ("first_value", 2), ("last_value", 2), ("nth_value", 2, 2)]
+             ("first_value", 2), ("last_value", 2), ("nth_value", 2, 2),
+             ("min", 2)]
 
 
 @pytest.mark.parametrize("frame", [(-1, 0), (3, 0), (0, 3), (2, 2)])
@@ -420,7 +425,7 @@ def test_gpu_rows_frame_parity(eng, orc, frame):
              (TYPE_INT64, D_UNI, 0, 40, 0),
              (TYPE_DOUBLE, D_SUM16, 0, 0, 150_000),
              (TYPE_INT64, D_UNI, -500, 500, 100_000)]
-    fns = [("count_star", -1), ("sum", 3), ("sum", 2), ("avg", 2),
+    fns = [("count_star", -1), ("sum", 3), ("min", 3), ("max", 2),
            ("count", 3), ("first_value", 3), ("last_value", 2),
            ("nth_value", 3, 3)]
     n = 150_000
