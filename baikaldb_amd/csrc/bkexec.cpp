@@ -313,7 +313,8 @@ static int32_t agg_out_type(int agg_type, int32_t in_type) {
         case BK_AGG_COUNT_STAR:
         case BK_AGG_COUNT:
         case BK_AGG_COUNT_DISTINCT: return BK_INT64;  /* agg_fn_call.cpp:96-99 */
-        case BK_AGG_AVG:   return BK_DOUBLE;
+        case BK_AGG_AVG:
+        case BK_AGG_AVG_DISTINCT: return BK_DOUBLE;
         default:           return in_type;  /* SUM/MIN/MAX keep input type */
     }
 }
@@ -366,8 +367,7 @@ public:
         std::vector<int> agg_dcol(q.n_aggs, -1);
         for (int a = 0; a < q.n_aggs; a++) {
             int at = q.aggs[a].agg_type;
-            if (at != BK_AGG_COUNT_DISTINCT && at != BK_AGG_SUM_DISTINCT)
-                continue;
+            if (at < BK_AGG_COUNT_DISTINCT) continue;
             int c = q.aggs[a].col;
             size_t j = 0;
             while (j < dcols.size() && dcols[j] != c) j++;

@@ -320,8 +320,11 @@ __device__ __forceinline__ void agg_merge_slot(uint64_t* dst, const uint64_t* sr
             case BK_AGG_SUM:
             case BK_AGG_AVG:
             case BK_AGG_SUM_DISTINCT:
+            case BK_AGG_AVG_DISTINCT:
                 if (sc) {
-                    if (q.agg_in_types[a] == BK_DOUBLE || q.aggs[a].agg_type == BK_AGG_AVG) {
+                    if (q.agg_in_types[a] == BK_DOUBLE ||
+                        q.aggs[a].agg_type == BK_AGG_AVG ||
+                        q.aggs[a].agg_type == BK_AGG_AVG_DISTINCT) {
                         double d; memcpy(&d, &sv, 8);
                         if (LDS) atomic_add_f64_lds(val, d);
                         else     atomic_add_f64_global(val, d);
@@ -1329,10 +1332,14 @@ __global__ void k_rollup(BkQuerySpec q2, int in_naggs, const uint64_t* in_table,
                 if (!d_null) atomicAdd((unsigned long long*)val, 1ull);
                 continue;
             }
-            if (at == BK_AGG_SUM_DISTINCT) {
+            if (at == BK_AGG_SUM_DISTINCT || at == BK_AGG_AVG_DISTINCT) {
                 if (!d_null) {
-                    if (q2.agg_in_types[a] == BK_DOUBLE)
-                        atomic_add_f64_global(val, bk_dec_f64(e_d));
+                    if (q2.agg_in_types[a] == BK_DOUBLE ||
+                        at == BK_AGG_AVG_DISTINCT)
+                        atomic_add_f64_global(val,
+                            q2.agg_in_types[a] == BK_DOUBLE
+                                ? bk_dec_f64(e_d)
+                                : (double)bk_dec_i64(e_d));
                     else
                         atomicAdd((unsigned long long*)val,
                                   (unsigned long long)(uint64_t)bk_dec_i64(e_d));
@@ -2094,8 +2101,10 @@ extern "C" BkgAggOut* bkgpu_agg_rollup(const BkgAggOut* in, const BkQuerySpec* q
     SrcIdx src;
     for (int a = 0; a < q2->n_aggs; a++) {
         int at = q2->aggs[a].agg_type;
-        bool synth = at == BK_AGG_COUNT_DISTINCT || at == BK_AGG_SUM_DISTINCT;
-        if (at == BK_AGG_SUM_DISTINCT && q2->agg_in_types[a] == BK_STRING) {
+        bool synth = at == BK_AGG_COUNT_DISTINCT || at == BK_AGG_SUM_DISTINCT ||
+                     at == BK_AGG_AVG_DISTINCT;
+        if ((at == BK_AGG_SUM_DISTINCT || at == BK_AGG_AVG_DISTINCT) &&
+            q2->agg_in_types[a] == BK_STRING) {
             set_err("rollup: SUM(DISTINCT string) unsupported"); return nullptr;
         }
         if (synth != (src_idx[a] < 0) ||
@@ -2318,6 +2327,7 @@ extern "C" int64_t bkgpu_agg_fetch(BkgAggOut* o, int sorted, int64_t max_groups,
                     else out_i[idx] = (int64_t)val;
                     out_has[idx] = 1; break;
                 case BK_AGG_AVG:
+                case BK_AGG_AVG_DISTINCT:
                     if (!cnt) break;
                     { double s; memcpy(&s, &val, 8); out_d[idx] = s / (double)cnt; }
                     out_has[idx] = 1; break;

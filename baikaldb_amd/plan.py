@@ -22,7 +22,7 @@ _FNS = {"year": 1, "month": 2, "day": 3, "dayofmonth": 3, "hour": 4,
 _AGGS = {"count_star": AGG_COUNT_STAR, "count": AGG_COUNT, "sum": AGG_SUM,
          "avg": AGG_AVG, "min": AGG_MIN, "max": AGG_MAX,
          "count_distinct": AGG_COUNT_DISTINCT,
-         "sum_distinct": AGG_SUM_DISTINCT}
+         "sum_distinct": AGG_SUM_DISTINCT, "avg_distinct": 8}
 
 
 class BkConjunct(C.Structure):

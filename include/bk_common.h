@@ -94,6 +94,9 @@ typedef enum BkAggType {
      * shaped, so export/merge/fetch treat them like COUNT/SUM. */
     BK_AGG_COUNT_DISTINCT = 6,
     BK_AGG_SUM_DISTINCT   = 7,
+    BK_AGG_AVG_DISTINCT   = 8,   /* {sum, count} over the dedup keys;
+                                    finalizes as sum/count (reference
+                                    "avg_distinct", agg_fn_call.cpp:39) */
 } BkAggType;
 
 /* ---- synthetic column distributions (bench configs) ---- */

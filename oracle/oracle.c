@@ -330,6 +330,7 @@ static inline void orc_agg_merge(OrcAggState* dst, const OrcAggState* src,
             }
             return;
         case BK_AGG_AVG:
+        case BK_AGG_AVG_DISTINCT:
             if (!src->has) return;
             dst->d += src->d; dst->cnt += src->cnt; dst->has = 1; return;
         case BK_AGG_MIN:
@@ -551,12 +552,15 @@ ORC_EXPORT OrcAggResult* orc_filter_agg_distinct(
             int vt = q2->agg_in_types[a];
             if (at == BK_AGG_COUNT_DISTINCT) {
                 orc_agg_update(&g2->st[a], BK_AGG_COUNT, vt, !d_null, 0, 0);
-            } else if (at == BK_AGG_SUM_DISTINCT) {
+            } else if (at == BK_AGG_SUM_DISTINCT || at == BK_AGG_AVG_DISTINCT) {
                 if (!d_null) {
                     int64_t vi = 0; double vd = 0.0;
                     if (vt == BK_DOUBLE) vd = bk_dec_f64(e_d);
                     else vi = bk_dec_i64(e_d);
-                    orc_agg_update(&g2->st[a], BK_AGG_SUM, vt, 1, vi, vd);
+                    orc_agg_update(&g2->st[a],
+                                   at == BK_AGG_AVG_DISTINCT ? BK_AGG_AVG
+                                                             : BK_AGG_SUM,
+                                   vt, 1, vi, vd);
                 }
             } else {
                 orc_agg_merge(&g2->st[a], &gi->st[src_idx[a]], at, vt);
@@ -612,6 +616,7 @@ static void orc_finalize_group(const OrcGroup* g, const BkQuerySpec* q,
                 else res->out_i[idx] = s->i;
                 res->out_has[idx] = 1; break;
             case BK_AGG_AVG:
+            case BK_AGG_AVG_DISTINCT:
                 if (!s->has || s->cnt == 0) { res->out_has[idx] = 0; break; }
                 res->out_d[idx] = s->d / (double)s->cnt;   /* agg_fn_call.cpp:958 */
                 res->out_has[idx] = 1; break;

@@ -14,7 +14,7 @@ from tests.test_gpu_agg import assert_parity, SEED
 TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
 D_UNI, D_SKEW, D_DICT, D_SUM16, D_ZIPF = 0, 1, 2, 3, 4
 AGGMAP = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5,
-          "count_distinct": 6, "sum_distinct": 7}
+          "count_distinct": 6, "sum_distinct": 7, "avg_distinct": 8}
 OPS = {"=": 0, "!=": 1, ">": 2, ">=": 3, "<": 4, "<=": 5}
 
 
@@ -85,6 +85,9 @@ def brute_distinct(cols, valids, col_types, conjuncts, group, aggs):
                 vals[a] = len(np.unique(v))
             elif name == "sum_distinct":
                 vals[a] = np.unique(v).sum()
+            elif name == "avg_distinct":
+                u = np.unique(v)
+                vals[a] = float(u.mean()) if len(u) else None
             elif name == "sum":
                 vals[a] = v.sum() if len(v) else None
             elif name == "count":
@@ -106,7 +109,7 @@ def test_oracle_distinct_vs_brute(orc):
              (TYPE_INT64, D_UNI, 0, 1 << 31, 0)]
     conj = [(3, "<", 1 << 30)]
     aggs = [("count_star", -1), ("count_distinct", 1), ("sum_distinct", 1),
-            ("sum", 2), ("count", 1)]
+            ("sum", 2), ("count", 1), ("avg_distinct", 1)]
     exp, (cols, valids, col_types) = oracle_distinct(
         orc, specs, 30_000, conj, [0], aggs)
     brute = brute_distinct(cols, valids, col_types, conj, [0], aggs)
@@ -128,6 +131,10 @@ def test_oracle_distinct_vs_brute(orc):
         else:
             assert abs(exp["agg_d"][3][g] - b[3]) < 1e-9 * (abs(b[3]) + 1)
         assert exp["agg_i"][4][g] == b[4]                      # count
+        if b[5] is None:
+            assert exp["agg_has"][5][g] == 0
+        else:
+            assert abs(exp["agg_d"][5][g] - b[5]) < 1e-9 * (abs(b[5]) + 1)
 
 
 def test_oracle_distinct_no_group(orc):
@@ -187,7 +194,8 @@ def run_both_distinct(eng, orc, specs, n, conjuncts, group, aggs,
 
 def _parity_names(aggs):
     """map distinct names onto their state-shaped plain kin for assert_parity"""
-    m = {"count_distinct": "count", "sum_distinct": "sum"}
+    m = {"count_distinct": "count", "sum_distinct": "sum",
+         "avg_distinct": "avg"}
     return [(m.get(n, n), c) for n, c in aggs]
 
 
@@ -198,7 +206,7 @@ def test_gpu_count_distinct_grouped(eng, orc):
              (TYPE_DOUBLE, D_SUM16, 0, 0, 0),
              (TYPE_INT64, D_UNI, 0, 1 << 31, 0)]
     aggs = [("count_star", -1), ("count_distinct", 1), ("sum", 3),
-            ("avg", 2), ("sum_distinct", 1)]
+            ("avg", 2), ("sum_distinct", 1), ("avg_distinct", 1)]
     got, exp = run_both_distinct(eng, orc, specs, 200_000,
                                  [(3, "<", 1 << 30)], [0], aggs)
     assert_parity(got, exp, _parity_names(aggs), [s[0] for s in specs])
