@@ -10,15 +10,17 @@ from tests.test_gpu_agg import run_both, assert_parity  # reuse harness
 
 pytestmark = pytest.mark.gpu
 
-TYPE_INT64, TYPE_DOUBLE, TYPE_STRING = 6, 12, 13
-D_UNI, D_SKEW, D_DICT, D_SUM16, D_ZIPF = 0, 1, 2, 3, 4
+TYPE_INT64, TYPE_DOUBLE, TYPE_STRING, TYPE_DATETIME = 6, 12, 13, 14
+D_UNI, D_SKEW, D_DICT, D_SUM16, D_ZIPF, D_DT = 0, 1, 2, 3, 4, 5
+_DT_FNS = ["year", "month", "day", "hour", "minute", "second"]
 
 
 def random_case(rng):
     ncols = rng.randint(2, 6)
     specs = []
     for _ in range(ncols):
-        t = rng.choice([TYPE_INT64, TYPE_INT64, TYPE_DOUBLE, TYPE_STRING])
+        t = rng.choice([TYPE_INT64, TYPE_INT64, TYPE_DOUBLE, TYPE_STRING,
+                        TYPE_DATETIME])
         nf = rng.choice([0, 0, 0, 120_000, 400_000])
         if t == TYPE_INT64:
             dist = rng.choice([D_UNI, D_SKEW, D_ZIPF])
@@ -32,6 +34,8 @@ def random_case(rng):
                 specs.append((t, dist, rng.choice([5, 300, 20_000]), 0, nf))
         elif t == TYPE_DOUBLE:
             specs.append((t, D_SUM16, 0, 0, nf))
+        elif t == TYPE_DATETIME:
+            specs.append((t, D_DT, 0, 0, nf))
         else:
             specs.append((t, D_DICT, rng.choice([3, 64, 3000]), 0, nf))
     # conjuncts
@@ -42,6 +46,12 @@ def random_case(rng):
         if t == TYPE_DOUBLE:
             conjuncts.append((c, rng.choice(["<", ">", ">=", "<="]),
                               rng.uniform(-1.5, 1.5)))
+        elif t == TYPE_DATETIME:
+            fn = rng.choice(_DT_FNS)
+            lim = {"year": (2019, 2026), "month": (1, 13), "day": (1, 29),
+                   "hour": (0, 24), "minute": (0, 60), "second": (0, 60)}[fn]
+            conjuncts.append(((fn, c), rng.choice(["<", ">", "=", "!="]),
+                              rng.randint(*lim)))
         elif rng.random() < 0.3:
             vals = [rng.randint(0, 3000) for _ in range(rng.randint(1, 8))]
             conjuncts.append((c, rng.choice(["in", "not_in"]), vals))

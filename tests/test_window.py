@@ -315,16 +315,27 @@ def test_gpu_window_fuzz(eng, orc, case_seed):
             fns.append((name, col, param))
         n = rng.choice([3000, 20_000])
         seed = rng.randrange(1 << 40)
+        # random frame mode; frames restrict to the fns they define
+        frame = rng.choice([None, None, (rng.randint(0, 4), rng.randint(0, 4)),
+                            (-1, 0), "range_upc", "range_crf"])
+        if frame is not None:
+            fns = [f for f in fns
+                   if f[0] in ("count_star", "count", "sum", "avg", "min",
+                               "max", "first_value", "last_value",
+                               "nth_value", "row_number", "rank",
+                               "dense_rank", "lead", "lag")]
+            if not fns:
+                fns = [("count_star", -1, 0)]
         t = eng.create_table(specs, n)
         try:
             eng.generate(t, seed)
-            got = eng.window(t, fns, part_col=part, order=order)
+            got = eng.window(t, fns, part_col=part, order=order, frame=frame)
         finally:
             t.free()
         cols, valids, types = gen(orc, specs, n, seed)
         exp = orc.window(cols, valids, types,
                          [(W[f[0]], f[1], f[2]) for f in fns],
-                         part_col=part, order=order)
+                         part_col=part, order=order, frame=frame)
         ctx = f"fuzz {case_seed}/{sub} part={part} order={order} fns={fns}"
         assert got["n"] == exp["n"], ctx
         assert np.array_equal(got["rowids"], exp["rowids"]), ctx
