@@ -226,7 +226,8 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
 /* order-preserving u64 encode of a group/minmax value (bk_keyenc.h) */
 __device__ __forceinline__ uint64_t enc_value(const DevCol& c, int64_t r) {
     switch (c.type) {
-        case BK_INT64:  return bk_enc_i64(((const int64_t*)c.data)[r]);
+        case BK_INT64:
+        case BK_DATETIME: return bk_enc_i64(((const int64_t*)c.data)[r]);
         case BK_DOUBLE: return bk_enc_f64(((const double*)c.data)[r]);
         default:        return (uint64_t)(uint32_t)((const int32_t*)c.data)[r];
     }

@@ -351,7 +351,8 @@ static inline void orc_agg_merge(OrcAggState* dst, const OrcAggState* src,
  * key the GPU uses). */
 static inline uint64_t orc_enc_group(const OrcCol* c, int64_t r) {
     switch (c->type) {
-        case BK_INT64:  return bk_enc_i64(((int64_t*)c->data)[r]);
+        case BK_INT64:
+        case BK_DATETIME: return bk_enc_i64(((int64_t*)c->data)[r]);
         case BK_DOUBLE: return bk_enc_f64(((double*)c->data)[r]);
         case BK_STRING: return (uint64_t)(uint32_t)((int32_t*)c->data)[r];
         default:        return 0;
@@ -456,7 +457,7 @@ static int64_t orc_build_key_bytes(const OrcGroup* g, const BkQuerySpec* q,
     for (int k = 0; k < q->n_group; k++) {
         if ((g->flag >> (7 - k)) & 1) continue;  /* null values omitted */
         int t = q->group_types[k];
-        if (t == BK_INT64 || t == BK_DOUBLE) {
+        if (t == BK_INT64 || t == BK_DOUBLE || t == BK_DATETIME) {
             if (out) {
                 uint64_t be = bk_bswap64(g->e[k]);  /* already sign-flip encoded */
                 memcpy(out + len, &be, 8);
