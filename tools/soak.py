@@ -1,0 +1,79 @@
+# Long randomized soak: many more fuzz iterations than the CI-sized suite.
+import random
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+sys.path.insert(0, "/root/repo")
+import numpy as np
+import torch
+if torch.cuda.is_available():
+    torch.cuda.init()
+from baikaldb_amd import GpuEngine
+from oracle import Oracle
+
+import tests.test_gpu_fuzz as fz
+import tests.test_window as tw
+from tests.test_gpu_agg import run_both, assert_parity
+from tests.test_gpu_sort import run_both as sort_both
+
+eng = GpuEngine()
+orc = Oracle()
+
+AGG_CASES = int(sys.argv[1]) if len(sys.argv) > 1 else 40
+fails = 0
+for cs in range(AGG_CASES):
+    rng = random.Random(90_000 + cs)
+    specs, conjuncts, group, aggs = fz.random_case(rng)
+    n = rng.choice([1000, 20_000, 120_000])
+    try:
+        got, exp = run_both(eng, orc, specs, n, conjuncts, group, aggs,
+                            seed=rng.randrange(1 << 40), expected_groups=1 << 12)
+        assert_parity(got, exp, aggs, [s[0] for s in specs])
+    except Exception as e:
+        fails += 1
+        print(f"AGG-FAIL {cs}: specs={specs} conj={conjuncts} group={group} "
+              f"aggs={aggs}: {e}", flush=True)
+print(f"agg soak: {AGG_CASES - fails}/{AGG_CASES} ok", flush=True)
+
+WIN_CASES = AGG_CASES // 2
+wfails = 0
+for cs in range(WIN_CASES):
+    try:
+        tw.test_gpu_window_fuzz.__wrapped__(eng, orc, 100_000 + cs) \
+            if hasattr(tw.test_gpu_window_fuzz, "__wrapped__") else \
+            tw.test_gpu_window_fuzz(eng, orc, 100_000 + cs)
+    except Exception as e:
+        wfails += 1
+        print(f"WIN-FAIL {cs}: {e}", flush=True)
+print(f"window soak: {WIN_CASES - wfails}/{WIN_CASES} ok", flush=True)
+
+SORT_CASES = AGG_CASES // 2
+sfails = 0
+for cs in range(SORT_CASES):
+    rng = random.Random(110_000 + cs)
+    ncols = rng.randint(2, 4)
+    specs = []
+    for _ in range(ncols):
+        t = rng.choice([fz.TYPE_INT64, fz.TYPE_INT64, fz.TYPE_DOUBLE])
+        nf = rng.choice([0, 0, 300_000])
+        if t == fz.TYPE_INT64:
+            specs.append((t, 0, rng.choice([0, -(1 << 50)]),
+                          rng.choice([50, 1 << 20, 1 << 51]), nf))
+        else:
+            specs.append((t, 3, 0, 0, nf))
+    norder = rng.randint(1, min(3, ncols))
+    order = [(c, rng.randint(0, 1), rng.randint(0, 1))
+             for c in rng.sample(range(ncols), norder)]
+    limit = rng.choice([1, 100, 3000, 50_000])
+    n = rng.choice([5000, 40_000, 150_000])
+    try:
+        got, exp = sort_both(eng, orc, specs, n, order, min(limit, n))
+        assert np.array_equal(got, exp)
+    except Exception as e:
+        sfails += 1
+        print(f"SORT-FAIL {cs}: specs={specs} order={order} "
+              f"limit={limit}: {e}", flush=True)
+print(f"sort soak: {SORT_CASES - sfails}/{SORT_CASES} ok", flush=True)
+total_fails = fails + wfails + sfails
+print(f"SOAK {'PASS' if total_fails == 0 else 'FAIL'} "
+      f"({total_fails} failures)", flush=True)
+sys.exit(1 if total_fails else 0)
