@@ -1178,7 +1178,7 @@ k_part_agg(BkQuerySpec q, RecLayout lay, const uint64_t* rec, uint64_t total,
          * the scheduler can overlap (this kernel is claim-latency-bound). */
         for (uint32_t t0 = 0; t0 < n; t0 += (uint32_t)ILP * blockDim.x) {
             uint32_t iA = t0 + threadIdx.x;
-            uint32_t iB = ILP == 2 ? t0 + blockDim.x + threadIdx.x : n;
+            uint32_t iB = ILP >= 2 ? t0 + blockDim.x + threadIdx.x : n;
             bool pA = iA < n, pB = iB < n;
             const uint64_t* myA = rec + (size_t)(b0 + iA) * lay.nwords;
             const uint64_t* myB = rec + (size_t)(b0 + iB) * lay.nwords;
