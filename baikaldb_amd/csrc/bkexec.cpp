@@ -686,7 +686,9 @@ public:
         _win_n.resize((size_t)_desc.n_winfns * (nrows ? nrows : 1));
         int64_t got = bkgpu_window(t, &q, _desc.part_col, _desc.order,
                                    _desc.n_order, _desc.winfns, _desc.n_winfns,
-                                   /*frame_rows=*/0, -1, -1,
+                                   _desc.frame_mode,
+                                   _desc.frame_mode ? _desc.frame_pre : -1,
+                                   _desc.frame_mode ? _desc.frame_fol : -1,
                                    0, nrows, _rowids.data(), _win_i.data(),
                                    _win_d.data(), _win_n.data());
         if (got < 0) { state->error_msg = bkgpu_last_error(); return -1; }

@@ -35,7 +35,10 @@ class BkPlanNodeDesc(C.Structure):
                 ("out_cols", C.c_int32 * BK_MAX_COLS),
                 ("part_col", C.c_int32),
                 ("n_winfns", C.c_int32),
-                ("winfns", BkWindowFn * 8)]
+                ("winfns", BkWindowFn * 8),
+                ("frame_mode", C.c_int32),
+                ("_pad_w", C.c_int32),
+                ("frame_pre", C.c_int64), ("frame_fol", C.c_int64)]
 
 
 def _bind(lib):
@@ -88,7 +91,8 @@ def filter_node(col_types, conjuncts, num_children=1):
     return d
 
 
-def window_node(part_col, order, fns, out_cols, num_children=1, limit=-1):
+def window_node(part_col, order, fns, out_cols, num_children=1, limit=-1,
+                frame=None):
     """WINDOW_NODE (window_node.cpp, non-frame): fns = (name, col[, param]);
     slots = [out_cols...][fn outputs...]."""
     d = BkPlanNodeDesc()
@@ -108,6 +112,12 @@ def window_node(part_col, order, fns, out_cols, num_children=1, limit=-1):
     d.n_out_cols = len(out_cols)
     for i, c in enumerate(out_cols):
         d.out_cols[i] = c
+    if frame == "range_upc":
+        d.frame_mode = 2
+    elif frame == "range_crf":
+        d.frame_mode = 3
+    elif frame is not None:
+        d.frame_mode, d.frame_pre, d.frame_fol = 1, frame[0], frame[1]
     return d
 
 

@@ -52,12 +52,17 @@ typedef struct BkPlanNodeDesc {
     BkOrderSpec order[4];
     int32_t     n_out_cols;          /* columns materialized per output row */
     int32_t     out_cols[BK_MAX_COLS];
-    /* WINDOW_NODE payload (NON-FRAME mode, window_node.cpp:39-41); reuses
-     * n_order/order for the in-partition sort and n_out_cols/out_cols for
-     * the materialized input columns */
+    /* WINDOW_NODE payload (window_node.cpp); reuses n_order/order for the
+     * in-partition sort and n_out_cols/out_cols for the materialized input
+     * columns. frame_mode: 0 non-frame, 1 ROWS [f_pre PRECEDING, f_fol
+     * FOLLOWING] (negative = UNBOUNDED), 2 RANGE UNBOUNDED..CURRENT,
+     * 3 RANGE CURRENT..UNBOUNDED (see bkgpu_window). */
     int32_t     part_col;            /* -1 = single whole-set partition */
     int32_t     n_winfns;
     BkWindowFn  winfns[BK_MAX_WINFNS];
+    int32_t     frame_mode;
+    int32_t     _pad_w;
+    int64_t     frame_pre, frame_fol;
 } BkPlanNodeDesc;
 
 typedef struct BkExecTree BkExecTree;      /* root ExecNode + RuntimeState */

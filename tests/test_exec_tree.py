@@ -305,3 +305,33 @@ def test_multi_column_distinct_through_exec_surface(eng, orc):
         assert vi[r, 3] == cols[2][m].sum()                     # sum(c2)
         assert vi[r, 4] == len(_np.unique(cols[0][m]))          # cd(c0)
         assert vi[r, 5] == _np.unique(cols[2][m]).sum()         # sd(c2)
+
+
+def test_window_frame_through_exec_surface(eng, orc):
+    """ROWS frame through the WINDOW node: running SUM (UNBOUNDED
+    PRECEDING..CURRENT ROW)."""
+    from baikaldb_amd import exec as bx
+    import numpy as _np
+    t, cols, valids, types = make_table(eng, orc, n=20_000)
+    try:
+        nodes = [bx.window_node(part_col=2, order=[(0, 1, 1)],
+                                fns=[("sum", 2), ("row_number", -1)],
+                                out_cols=[2, 0], frame=(-1, 0)),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
+    order = _np.lexsort((_np.arange(len(cols[0])), cols[0], cols[2]))
+    g = cols[2][order]
+    run = _np.zeros(len(order), dtype=_np.int64)
+    start = 0
+    while start < len(order):
+        end = start + 1
+        while end < len(order) and g[end] == g[start]:
+            end += 1
+        run[start:end] = _np.cumsum(cols[2][order[start:end]])
+        start = end
+    assert _np.array_equal(vi[:, 2], run)
