@@ -280,6 +280,12 @@ class GpuEngine:
             farr[i].fn_type = _WINFNS[name] if isinstance(name, str) else name
             farr[i].col = col
             farr[i].param = f[2] if len(f) > 2 else 0
+            if len(f) > 3 and f[3] is not None:
+                farr[i].has_def = 1
+                if isinstance(f[3], float):
+                    farr[i].def_d = f[3]
+                else:
+                    farr[i].def_i = int(f[3])
         cap = row_end - row_begin
         rowids = np.empty(cap, dtype=np.int64)
         out_i = np.zeros(len(fns) * cap, dtype=np.int64)

@@ -46,7 +46,9 @@ _WINFNS = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4,
 
 class BkWindowFn(C.Structure):
     _fields_ = [("fn_type", C.c_int32), ("col", C.c_int32),
-                ("param", C.c_int64)]
+                ("param", C.c_int64),
+                ("has_def", C.c_int32), ("_pad", C.c_int32),
+                ("def_i", C.c_int64), ("def_d", C.c_double)]
 
 
 class BkOrderSpec(C.Structure):

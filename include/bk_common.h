@@ -195,6 +195,12 @@ typedef struct BkWindowFn {
     int32_t fn_type;   /* BkWinType */
     int32_t col;       /* input column, -1 for COUNT_STAR / pure rank fns */
     int64_t param;     /* NTH_VALUE n / LEAD/LAG offset */
+    /* LEAD/LAG optional literal default (window_fn_call.cpp:144-150):
+     * returned when the offset row leaves the partition; NULL otherwise */
+    int32_t has_def;
+    int32_t _pad;
+    int64_t def_i;
+    double  def_d;
 } BkWindowFn;
 
 #define BK_MAX_WINFNS 8

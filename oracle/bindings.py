@@ -38,7 +38,9 @@ class BkAggSpec(C.Structure):
 
 class BkWindowFn(C.Structure):
     _fields_ = [("fn_type", C.c_int32), ("col", C.c_int32),
-                ("param", C.c_int64)]
+                ("param", C.c_int64),
+                ("has_def", C.c_int32), ("_pad", C.c_int32),
+                ("def_i", C.c_int64), ("def_d", C.c_double)]
 
 
 class BkOrderSpec(C.Structure):
@@ -266,6 +268,12 @@ class Oracle:
         for i, f in enumerate(fns):
             farr[i].fn_type, farr[i].col = f[0], f[1]
             farr[i].param = f[2] if len(f) > 2 else 0
+            if len(f) > 3 and f[3] is not None:
+                farr[i].has_def = 1
+                if isinstance(f[3], float):
+                    farr[i].def_d = f[3]
+                else:
+                    farr[i].def_i = int(f[3])
         cap = row_end - row_begin
         rowids = np.empty(cap, dtype=np.int64)
         out_i = np.zeros(len(fns) * cap, dtype=np.int64)

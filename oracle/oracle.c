@@ -1127,10 +1127,16 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                     case BK_WIN_LAG: {
                         int64_t off = fns[f].param > 0 ? fns[f].param : 1;
                         int64_t jj = ft == BK_WIN_LEAD ? j + off : j - off;
-                        if (jj >= ps && jj < pe)
+                        if (jj >= ps && jj < pe) {
                             orc_win_value(&cols[fns[f].col], out_rows[jj], idx,
                                           out_i, out_d, out_null);
-                        else out_null[idx] = 1;
+                        } else if (fns[f].has_def) {
+                            if (cols[fns[f].col].type == BK_DOUBLE)
+                                out_d[idx] = fns[f].def_d;
+                            else out_i[idx] = fns[f].def_i;
+                        } else {
+                            out_null[idx] = 1;
+                        }
                         break;
                     }
                     case BK_WIN_CUME_DIST:

@@ -548,3 +548,43 @@ def test_gpu_range_frame_parity(eng, orc, mode):
     d = np.abs(got["out_d"] - exp["out_d"])
     tol = 1e-9 * (np.abs(exp["out_d"]) + 100)
     assert np.all(d[mask] <= tol[mask])
+
+
+def test_oracle_lead_lag_default(orc):
+    """LEAD/LAG with a literal default (window_fn_call.cpp:144-150)."""
+    specs = [(TYPE_INT64, D_UNI, 0, 5, 0),
+             (TYPE_INT64, D_UNI, 0, 100, 0)]
+    cols, valids, types = gen(orc, specs, 400)
+    fns = [(W["lag"], 1, 2, -777), (W["lead"], 1, 1, 999)]
+    res = orc.window(cols, valids, types, fns, part_col=0, order=[(1, 1, 1)])
+    idx, brute = brute_window(cols, valids, 0, [(1, 1, 1)],
+                              [("lag", 1, 2), ("lead", 1, 1)])
+    for f, dflt in ((0, -777), (1, 999)):
+        for i in range(res["n"]):
+            b = brute[f][i]
+            assert res["out_null"][f][i] == 0
+            assert res["out_i"][f][i] == (dflt if b is None else b), (f, i)
+
+
+@pytest.mark.gpu
+def test_gpu_lead_lag_default(eng, orc):
+    specs = [(TYPE_INT64, D_UNI, 0, 50, 0),
+             (TYPE_INT64, D_UNI, 0, 1 << 20, 0),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 100_000)]
+    fns = [("lag", 1, 1, -5), ("lead", 2, 3, 2.5), ("lead", 1, 2)]
+    n = 80_000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED + 3)
+        got = eng.window(t, fns, part_col=0, order=[(1, 1, 1)])
+    finally:
+        t.free()
+    cols, valids, types = gen(orc, specs, n, SEED + 3)
+    exp = orc.window(cols, valids, types,
+                     [(W[f[0]], f[1], f[2], f[3] if len(f) > 3 else None)
+                      for f in fns],
+                     part_col=0, order=[(1, 1, 1)])
+    assert np.array_equal(got["out_null"], exp["out_null"])
+    assert np.array_equal(got["out_i"], exp["out_i"])
+    mask = exp["out_null"] == 0
+    assert np.allclose(got["out_d"][mask], exp["out_d"][mask], rtol=1e-12)
