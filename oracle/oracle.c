@@ -946,6 +946,25 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                     int64_t fl, fr;
                     if (frame_rows == 2) { fl = ps; fr = rgr; }
                     else if (frame_rows == 3) { fl = rgl; fr = pe - 1; }
+                    else if (frame_rows == 4) {
+                        /* RANGE by VALUE over the single ASC int order key */
+                        const OrcCol* okc = &cols[order[0].col];
+                        if (!cell_is_valid(okc, out_rows[j])) {
+                            fl = rgl; fr = rgr;   /* null peers */
+                        } else {
+                            int64_t v = cell_i64(okc, out_rows[j]);
+                            fl = j; fr = j;
+                            while (fl > ps && cell_is_valid(okc, out_rows[fl - 1]) &&
+                                   (f_pre < 0 ||
+                                    cell_i64(okc, out_rows[fl - 1]) >= v - f_pre))
+                                fl--;
+                            while (fr + 1 < pe &&
+                                   cell_is_valid(okc, out_rows[fr + 1]) &&
+                                   (f_fol < 0 ||
+                                    cell_i64(okc, out_rows[fr + 1]) <= v + f_fol))
+                                fr++;
+                        }
+                    }
                     else {
                         fl = f_pre >= 0 && j - f_pre > ps ? j - f_pre : ps;
                         fr = f_fol >= 0 && j + f_fol < pe - 1 ? j + f_fol

@@ -588,3 +588,100 @@ def test_gpu_lead_lag_default(eng, orc):
     assert np.array_equal(got["out_i"], exp["out_i"])
     mask = exp["out_null"] == 0
     assert np.allclose(got["out_d"][mask], exp["out_d"][mask], rtol=1e-12)
+
+
+def brute_range_val(cols, valids, part_col, order, fns, pre, fol):
+    """RANGE value-offset frames, single ASC int key."""
+    idx, _ = brute_window(cols, valids, part_col, order, [("row_number", -1)])
+    n = len(idx)
+    oc = order[0][0]
+
+    def null(c, r):
+        return valids[c] is not None and valids[c][r] == 0
+
+    def peq(a, b):
+        if part_col < 0:
+            return True
+        na, nb = null(part_col, a), null(part_col, b)
+        return na == nb and (na or cols[part_col][a] == cols[part_col][b])
+
+    out = {f: [None] * n for f in range(len(fns))}
+    ps = 0
+    while ps < n:
+        pe = ps + 1
+        while pe < n and peq(idx[pe], idx[pe - 1]):
+            pe += 1
+        for j in range(ps, pe):
+            if null(oc, idx[j]):
+                fl = j
+                while fl > ps and null(oc, idx[fl - 1]):
+                    fl -= 1
+                fr = j
+                while fr + 1 < pe and null(oc, idx[fr + 1]):
+                    fr += 1
+            else:
+                v = cols[oc][idx[j]]
+                fl = fr = j
+                while fl > ps and not null(oc, idx[fl - 1]) and \
+                        (pre < 0 or cols[oc][idx[fl - 1]] >= v - pre):
+                    fl -= 1
+                while fr + 1 < pe and not null(oc, idx[fr + 1]) and \
+                        (fol < 0 or cols[oc][idx[fr + 1]] <= v + fol):
+                    fr += 1
+            frame_rows = [idx[k] for k in range(fl, fr + 1)]
+            for f, (name, col, *rest) in enumerate(fns):
+                if name == "count_star":
+                    out[f][j] = len(frame_rows)
+                elif name == "sum":
+                    vv = [cols[col][r] for r in frame_rows if not null(col, r)]
+                    out[f][j] = np.sum(np.array(vv)) if vv else None
+                elif name == "min":
+                    vv = [cols[col][r] for r in frame_rows if not null(col, r)]
+                    out[f][j] = min(vv) if vv else None
+        ps = pe
+    return idx, out
+
+
+@pytest.mark.parametrize("vframe", [(10, 10), (0, 25), (-1, 5)])
+def test_oracle_range_val_vs_brute(orc, vframe):
+    specs = [(TYPE_INT64, D_UNI, 0, 6, 0),
+             (TYPE_INT64, D_UNI, 0, 200, 150_000),
+             (TYPE_INT64, D_UNI, -40, 40, 0)]
+    cols, valids, types = gen(orc, specs, 2000)
+    order = [(1, 1, 1)]
+    fns = [("count_star", -1), ("sum", 2), ("min", 2)]
+    res = orc.window(cols, valids, types, [(W[f[0]], f[1], 0) for f in fns],
+                     part_col=0, order=order,
+                     frame=("range_val", vframe[0], vframe[1]))
+    idx, brute = brute_range_val(cols, valids, 0, order, fns, *vframe)
+    check_against_brute(res, types, fns, idx, brute)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("vframe", [(100, 100), (-1, 50)])
+def test_gpu_range_val_parity(eng, orc, vframe):
+    specs = [(TYPE_INT64, D_UNI, 0, 60, 0),
+             (TYPE_INT64, D_UNI, 0, 5000, 120_000),
+             (TYPE_INT64, D_UNI, -900, 900, 80_000)]
+    fns = [("count_star", -1), ("sum", 2), ("avg", 2), ("min", 2),
+           ("max", 2), ("last_value", 2)]
+    n = 120_000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED + 11)
+        got = eng.window(t, fns, part_col=0, order=[(1, 1, 1)],
+                         frame=("range_val", vframe[0], vframe[1]))
+    finally:
+        t.free()
+    cols, valids, types = gen(orc, specs, n, SEED + 11)
+    exp = orc.window(cols, valids, types, [(W[f[0]], f[1], 0) for f in fns],
+                     part_col=0, order=[(1, 1, 1)],
+                     frame=("range_val", vframe[0], vframe[1]))
+    assert got["n"] == exp["n"]
+    assert np.array_equal(got["rowids"], exp["rowids"])
+    assert np.array_equal(got["out_null"], exp["out_null"])
+    assert np.array_equal(got["out_i"], exp["out_i"])
+    mask = exp["out_null"] == 0
+    d = np.abs(got["out_d"] - exp["out_d"])
+    tol = 1e-9 * (np.abs(exp["out_d"]) + 100)
+    assert np.all(d[mask] <= tol[mask])
