@@ -297,6 +297,8 @@ class GpuEngine:
             fr, fpre, ffol = 2, -1, -1
         elif frame == "range_crf":    # RANGE CURRENT ROW..UNBOUNDED FOLLOWING
             fr, fpre, ffol = 3, -1, -1
+        elif isinstance(frame, tuple) and frame[0] == "range_val":
+            fr, fpre, ffol = 4, frame[1], frame[2]
         else:
             fr, fpre, ffol = 1, frame[0], frame[1]
         n = self.lib.bkgpu_window(
