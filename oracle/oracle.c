@@ -1114,6 +1114,24 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                     fr = peer_end - 1;
                 } else if (frame_rows == 3) {
                     fl = peer_head;
+                } else if (frame_rows == 4) {
+                    const OrcCol* okc = &cols[order[0].col];
+                    if (!cell_is_valid(okc, out_rows[j])) {
+                        fl = peer_head;
+                        fr = peer_end - 1;
+                    } else {
+                        int64_t v = cell_i64(okc, out_rows[j]);
+                        fl = j; fr = j;
+                        while (fl > ps && cell_is_valid(okc, out_rows[fl - 1]) &&
+                               (f_pre < 0 ||
+                                cell_i64(okc, out_rows[fl - 1]) >= v - f_pre))
+                            fl--;
+                        while (fr + 1 < pe &&
+                               cell_is_valid(okc, out_rows[fr + 1]) &&
+                               (f_fol < 0 ||
+                                cell_i64(okc, out_rows[fr + 1]) <= v + f_fol))
+                            fr++;
+                    }
                 }
                 switch (ft) {
                     case BK_WIN_ROW_NUMBER: out_i[idx] = j - ps + 1; break;

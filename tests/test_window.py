@@ -638,6 +638,12 @@ def brute_range_val(cols, valids, part_col, order, fns, pre, fol):
                 elif name == "min":
                     vv = [cols[col][r] for r in frame_rows if not null(col, r)]
                     out[f][j] = min(vv) if vv else None
+                elif name == "first_value":
+                    r0 = frame_rows[0]
+                    out[f][j] = None if null(col, r0) else cols[col][r0]
+                elif name == "last_value":
+                    r0 = frame_rows[-1]
+                    out[f][j] = None if null(col, r0) else cols[col][r0]
         ps = pe
     return idx, out
 
@@ -649,7 +655,8 @@ def test_oracle_range_val_vs_brute(orc, vframe):
              (TYPE_INT64, D_UNI, -40, 40, 0)]
     cols, valids, types = gen(orc, specs, 2000)
     order = [(1, 1, 1)]
-    fns = [("count_star", -1), ("sum", 2), ("min", 2)]
+    fns = [("count_star", -1), ("sum", 2), ("min", 2), ("last_value", 2),
+           ("first_value", 2)]
     res = orc.window(cols, valids, types, [(W[f[0]], f[1], 0) for f in fns],
                      part_col=0, order=order,
                      frame=("range_val", vframe[0], vframe[1]))
