@@ -116,6 +116,8 @@ def window_node(part_col, order, fns, out_cols, num_children=1, limit=-1,
         d.frame_mode = 2
     elif frame == "range_crf":
         d.frame_mode = 3
+    elif isinstance(frame, tuple) and frame and frame[0] == "range_val":
+        d.frame_mode, d.frame_pre, d.frame_fol = 4, frame[1], frame[2]
     elif frame is not None:
         d.frame_mode, d.frame_pre, d.frame_fol = 1, frame[0], frame[1]
     return d
