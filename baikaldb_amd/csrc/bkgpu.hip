@@ -306,7 +306,6 @@ __device__ __forceinline__ void agg_update_slot(uint64_t* st, const BkQuerySpec&
 template <bool LDS>
 __device__ __forceinline__ void agg_merge_slot(uint64_t* dst, const uint64_t* src,
                                                const BkQuerySpec& q) {
-    #pragma unroll 4
     for (int32_t a = 0; a < q.n_aggs; a++) {
         uint64_t* val = dst + SLOT_HDR + 2 * a;
         uint64_t* cnt = val + 1;

@@ -17,7 +17,8 @@
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
-#include <algorithm>\n#include <string>
+#include <algorithm>
+#include <string>
 #include <vector>
 
 namespace bkparquet {
