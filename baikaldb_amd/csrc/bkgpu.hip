@@ -987,7 +987,6 @@ __device__ __forceinline__ void agg_update_slot_rec(uint64_t* st, const BkQueryS
                                                     const RecLayout& lay,
                                                     const uint64_t* my,
                                                     uint64_t meta) {
-    #pragma unroll 4
     for (int32_t a = 0; a < q.n_aggs; a++) {
         uint64_t* val = st + SLOT_HDR + 2 * a;
         uint64_t* cnt = val + 1;
@@ -1045,7 +1044,6 @@ __device__ __forceinline__ bool wave_combine_update(
         uint64_t* slot, const BkQuerySpec& q, const RecLayout& lay,
         const uint64_t* my, uint64_t meta, bool lds) {
     int lane = threadIdx.x & 63;
-    #pragma unroll 4
     for (int32_t a = 0; a < q.n_aggs; a++) {
         int at = q.aggs[a].agg_type;
         int has_meta = lay.meta_word >= 0;
