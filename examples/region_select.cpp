@@ -123,6 +123,56 @@ static int run_sort(BkgTable* table) {
     return 0;
 }
 
+static int run_window(BkgTable* table) {
+    /* SELECT g, v, ROW_NUMBER() OVER w, SUM(v) OVER w, RANK() OVER w
+     * FROM t WINDOW w AS (PARTITION BY g ORDER BY v)  -- non-frame mode */
+    BkPlanNodeDesc plan[2];
+    memset(plan, 0, sizeof plan);
+    plan[0].node_type = BK_WINDOW_NODE;
+    plan[0].num_children = 1;
+    plan[0].limit = 6;                  /* print the first rows only */
+    plan[0].part_col = 0;
+    plan[0].n_order = 1;
+    plan[0].order[0] = {2, 1, 1, 0};
+    plan[0].n_winfns = 3;
+    plan[0].winfns[0] = {BK_WIN_ROW_NUMBER, -1, 0, 0, 0, 0, 0.0};
+    plan[0].winfns[1] = {BK_WIN_SUM, 2, 0, 0, 0, 0, 0.0};
+    plan[0].winfns[2] = {BK_WIN_RANK, -1, 0, 0, 0, 0, 0.0};
+    plan[0].n_out_cols = 2;
+    plan[0].out_cols[0] = 0;
+    plan[0].out_cols[1] = 2;
+    plan[1].node_type = BK_SCAN_NODE;
+    plan[1].limit = -1;
+    plan[1].table = table;
+
+    BkExecTree* t = bkexec_create_tree(plan, 2);
+    if (!t) { fprintf(stderr, "create_tree: %s\n", bkgpu_last_error()); return 1; }
+    if (bkexec_open(t) < 0) { fprintf(stderr, "window open failed\n"); return 1; }
+    int ns = bkexec_n_slots(t);
+    std::vector<int32_t> tag(8 * ns);
+    std::vector<int64_t> vi(8 * ns);
+    std::vector<double> vd(8 * ns);
+    std::vector<uint8_t> nul(8 * ns);
+    int eos = 0;
+    char word[64];
+    printf("window (PARTITION BY g ORDER BY v): first rows\n");
+    while (!eos) {
+        int64_t n = bkexec_get_next(t, 8, tag.data(), vi.data(), vd.data(),
+                                    nul.data(), &eos);
+        if (n < 0) { fprintf(stderr, "window get_next failed\n"); return 1; }
+        for (int64_t r = 0; r < n; r++) {
+            size_t i0 = (size_t)r * ns;
+            if (bkgpu_table_dict_word(table, 0, vi[i0], word, sizeof word) < 0)
+                snprintf(word, sizeof word, "code:%lld", (long long)vi[i0]);
+            printf("  g=%s v=%lld row_number=%lld sum_over_g=%lld rank=%lld\n",
+                   word, (long long)vi[i0 + 1], (long long)vi[i0 + 2],
+                   (long long)vi[i0 + 3], (long long)vi[i0 + 4]);
+        }
+    }
+    bkexec_close(t);
+    return 0;
+}
+
 int main(int argc, char** argv) {
     if (argc != 2) {
         fprintf(stderr, "usage: %s <file.parquet>\n", argv[0]);
@@ -141,6 +191,7 @@ int main(int argc, char** argv) {
            (long long)bkgpu_table_nrows(table), bkgpu_table_ncols(table));
     int rc = run_agg(table);
     if (rc == 0) rc = run_sort(table);
+    if (rc == 0) rc = run_window(table);
     bkgpu_table_free(table);
     return rc;
 }

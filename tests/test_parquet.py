@@ -299,3 +299,4 @@ def test_cpp_embedding_example(tmp_path):
     expect = f"row 0: {g0} {rows.sum()} {v[rows].sum()} {min(warr[rows])}"
     assert expect in out.stdout, f"wanted {expect!r} in:\n{out.stdout[:2000]}"
     assert "top-5 by (v, g):" in out.stdout
+    assert "row_number=1" in out.stdout
