@@ -347,6 +347,8 @@ public:
         for (int k = 0; k < q.n_group; k++) {
             q.group_cols[k] = _desc.group_cols[k];
             q.group_types[k] = bkgpu_table_col_type(t, _desc.group_cols[k]);
+            q.group_bits[k] = _desc.group_bits[k];
+            q.group_base[k] = _desc.group_base[k];
         }
         q.n_aggs = _desc.n_aggs;
         for (int a = 0; a < q.n_aggs; a++) {

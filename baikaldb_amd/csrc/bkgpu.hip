@@ -233,6 +233,41 @@ __device__ __forceinline__ uint64_t enc_value(const DevCol& c, int64_t r) {
     }
 }
 
+/* spec-driven group-key packing (bk_common.h group_bits/group_base):
+ * per key, (enc - enc(base)) packs into `bits` bits; bits == 0 = a full
+ * 64-bit word. With <= 2 keys and all-zero bits this degenerates to the
+ * plain [k0][k1] layout (shift stays 0, each key takes its own word).
+ * Widths come from the QUERY, never from data statistics, so the packed
+ * keys are identical on every rank and partial blobs merge across GPUs. */
+struct KeyPack { uint64_t k0, k1; uint32_t flag; };
+
+__device__ __forceinline__ KeyPack pack_group_keys(const DevCols& cols,
+                                                   const BkQuerySpec& q,
+                                                   int64_t r) {
+    KeyPack kp{0, 0, 0};
+    int shift = 0, word = 0;
+    for (int32_t k = 0; k < q.n_group; k++) {
+        int bits = q.group_bits[k] ? q.group_bits[k] : 64;
+        if (shift + bits > 64) { word++; shift = 0; }
+        const DevCol& c = cols.c[q.group_cols[k]];
+        uint64_t e = 0;
+        if (!cell_valid(c, r)) {
+            kp.flag |= 0x80u >> k;   /* null-flag bit, exec_node.cpp:561 */
+        } else {
+            e = enc_value(c, r);
+            if (bits < 64) {
+                uint64_t eb = c.type == BK_STRING
+                                  ? (uint64_t)q.group_base[k]
+                                  : bk_enc_i64(q.group_base[k]);
+                e = (e - eb) & ((1ull << bits) - 1);
+            }
+        }
+        if (word == 0) kp.k0 |= e << shift; else kp.k1 |= e << shift;
+        shift += bits;
+    }
+    return kp;
+}
+
 /* ---- double atomic adds (IEEE add; order unspecified) ---- */
 __device__ __forceinline__ void atomic_add_f64_global(uint64_t* addr, double v) {
     unsafeAtomicAdd((double*)addr, v);   /* global_atomic_add_f64 on gfx950 */
@@ -487,19 +522,11 @@ k_filter_agg_group(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_e
          r < row_end; r += gstride) {
         if (!row_passes(cols, q, r)) continue;
         my_passed++;
-        /* group key (exec_node.cpp:555-571: null-flag byte + encoded values) */
-        uint32_t flag = 0;
-        uint64_t k0 = 0, k1 = 0;
-        if (q.n_group >= 1) {
-            const DevCol& c = cols.c[q.group_cols[0]];
-            if (!cell_valid(c, r)) flag |= 0x80u;      /* 0x01 << (7-0) */
-            else k0 = enc_value(c, r);
-        }
-        if (q.n_group >= 2) {
-            const DevCol& c = cols.c[q.group_cols[1]];
-            if (!cell_valid(c, r)) flag |= 0x40u;      /* 0x01 << (7-1) */
-            else k1 = enc_value(c, r);
-        }
+        /* group key (exec_node.cpp:555-571: null-flag byte + encoded values,
+         * spec-packed into two words for up to 4 keys) */
+        KeyPack kp = pack_group_keys(cols, q, r);
+        uint32_t flag = kp.flag;
+        uint64_t k0 = kp.k0, k1 = kp.k1;
         uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1, lfill, lcap);
         if (slot) {
             agg_update_slot<true>(slot, q, cols, r);
@@ -695,16 +722,9 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
         int64_t i = r - row_begin;
         if (!row_passes(cols, q, r)) { bucketid[i] = (uint16_t)BK_SKIP_BUCKET; continue; }
         my_passed++;
-        uint32_t flag = 0;
-        uint64_t k0 = 0, k1 = 0;
-        if (q.n_group >= 1) {
-            const DevCol& c = cols.c[q.group_cols[0]];
-            if (!cell_valid(c, r)) flag |= 0x80u; else k0 = enc_value(c, r);
-        }
-        if (q.n_group >= 2) {
-            const DevCol& c = cols.c[q.group_cols[1]];
-            if (!cell_valid(c, r)) flag |= 0x40u; else k1 = enc_value(c, r);
-        }
+        KeyPack kp = pack_group_keys(cols, q, r);
+        uint32_t flag = kp.flag;
+        uint64_t k0 = kp.k0, k1 = kp.k1;
         /* adaptive: pay the LDS-claim probe only while it absorbs >= 1/4
          * of the stream (per-block warmup decides; Zipf-headed keys keep it
          * on, flat/high-cardinality keys turn it off). */
@@ -864,16 +884,9 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
         uint32_t b = bucketid[i];
         if (b >= BK_HOT_BUCKET) continue;  /* filtered out or absorbed hot */
         /* key words */
-        uint64_t meta = 0;
-        uint64_t k0 = 0, k1 = 0;
-        if (q.n_group >= 1) {
-            const DevCol& c = cols.c[q.group_cols[0]];
-            if (!cell_valid(c, r)) meta |= 0x80u; else k0 = enc_value(c, r);
-        }
-        if (q.n_group >= 2) {
-            const DevCol& c = cols.c[q.group_cols[1]];
-            if (!cell_valid(c, r)) meta |= 0x40u; else k1 = enc_value(c, r);
-        }
+        KeyPack kp = pack_group_keys(cols, q, r);
+        uint64_t meta = kp.flag;
+        uint64_t k0 = kp.k0, k1 = kp.k1;
         uint64_t regs[8];
         #pragma unroll
         for (int w = 0; w < 8; w++) regs[w] = 0;
@@ -1747,12 +1760,33 @@ static int agg_compact(BkgAggOut* o) {
     return 0;
 }
 
+/* number of 64-bit key words the spec-packed group keys occupy */
+static int pack_key_words(const BkQuerySpec* q) {
+    if (q->n_group == 0) return 0;
+    int shift = 0, word = 0;
+    for (int32_t k = 0; k < q->n_group; k++) {
+        int bits = q->group_bits[k] ? q->group_bits[k] : 64;
+        if (shift + bits > 64) { word++; shift = 0; }
+        shift += bits;
+    }
+    return word + 1;
+}
+
+static bool any_group_bits(const BkQuerySpec* q) {
+    for (int32_t k = 0; k < q->n_group; k++)
+        if (q->group_bits[k]) return true;
+    return false;
+}
+
 static int build_rec_layout(BkgTable* t, const BkQuerySpec* q, RecLayout* lay) {
     int w = 1; /* word 0 = k0 (or the fused key word) */
+    bool plain2 = q->n_group <= 2 && !any_group_bits(q);
     /* a dict-encoded (BK_STRING) second group key is a 32-bit code: pack it
      * into the meta word's high half (k1_word == -2) instead of spending a
-     * whole record word — 20% narrower records on the config-3 shape */
-    bool k1_in_meta = q->n_group >= 2 &&
+     * whole record word — 20% narrower records on the config-3 shape.
+     * Only in the plain two-word layout; spec-packed keys manage their own
+     * word budget. */
+    bool k1_in_meta = plain2 && q->n_group == 2 &&
                       t->specs[q->group_cols[1]].col_type == BK_STRING;
     bool agg_meta = false;   /* meta bits beyond the key null flags */
     for (int a = 0; a < q->n_aggs; a++) {
@@ -1765,7 +1799,8 @@ static int build_rec_layout(BkgTable* t, const BkQuerySpec* q, RecLayout* lay) {
      * key + flags collapse into ONE word — 20-25% narrower records, which
      * is scatter write traffic and part_agg read traffic. */
     lay->k0_base = 0;
-    bool fused = q->n_group >= 1 && !agg_meta && getenv("BK_NO_FUSE") == nullptr;
+    bool fused = plain2 && q->n_group >= 1 && !agg_meta &&
+                 getenv("BK_NO_FUSE") == nullptr;
     if (fused) {
         int c0 = q->group_cols[0];
         fused = ensure_stats(t, c0) == 0 && t->stat_ok[c0] &&
@@ -1789,7 +1824,7 @@ static int build_rec_layout(BkgTable* t, const BkQuerySpec* q, RecLayout* lay) {
         lay->nwords = w;
         return 0;
     }
-    lay->k1_word = q->n_group >= 2 ? (k1_in_meta ? -2 : w++) : -1;
+    lay->k1_word = pack_key_words(q) >= 2 ? (k1_in_meta ? -2 : w++) : -1;
     bool need_meta = k1_in_meta || agg_meta;
     for (int k = 0; k < q->n_group; k++)
         if (t->valid[q->group_cols[k]]) need_meta = true;
@@ -1998,6 +2033,19 @@ static bool debug_timing() {
 extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
                                        int64_t row_begin, int64_t row_end,
                                        int64_t expected_groups) {
+    if (q && pack_key_words(q) > 2) {
+        set_err("group keys exceed two packed 64-bit words "
+                "(set group_bits per key, bk_common.h)");
+        return nullptr;
+    }
+    if (q)
+        for (int32_t k = 0; k < q->n_group; k++)
+            if (q->group_bits[k] &&
+                (q->group_bits[k] < 0 || q->group_bits[k] > 63 ||
+                 q->group_types[k] == BK_DOUBLE)) {
+                set_err("bad group_bits (1..63; DOUBLE keys need bits==0)");
+                return nullptr;
+            }
     if (ensure_device() != 0) return nullptr;
     double t_start = debug_timing() ? now_ms() : 0;
     double t_alloc = 0, t_pipe = 0;
@@ -2277,6 +2325,32 @@ static int download_groups(BkgAggOut* o, HostGroups& hg) {
     return 0;
 }
 
+/* reverse of pack_group_keys: recover each key's full order-preserving
+ * encoding from the two packed words (spec widths + bases) */
+static void unpack_group_keys(const BkQuerySpec& q, uint64_t k0, uint64_t k1,
+                              uint32_t flag, uint64_t* enc_out) {
+    int shift = 0, word = 0;
+    for (int32_t k = 0; k < q.n_group; k++) {
+        int bits = q.group_bits[k] ? q.group_bits[k] : 64;
+        if (shift + bits > 64) { word++; shift = 0; }
+        uint64_t w = word == 0 ? k0 : k1;
+        uint64_t e = 0;
+        if (!((flag >> (7 - k)) & 1)) {
+            if (bits == 64) {
+                e = w >> shift;   /* shift is 0 for a full word */
+            } else {
+                uint64_t eb = q.group_types[k] == BK_STRING
+                                  ? (uint64_t)q.group_base[k]
+                                  : bk_enc_i64(q.group_base[k]);
+                e = ((w >> shift) & ((1ull << bits) - 1)) + eb;
+            }
+        }
+        enc_out[k] = e;
+        shift += bits;
+    }
+    for (int32_t k = q.n_group; k < BK_MAX_GROUP; k++) enc_out[k] = 0;
+}
+
 extern "C" int64_t bkgpu_agg_fetch(BkgAggOut* o, int sorted, int64_t max_groups,
                                    uint8_t* flags, uint64_t* enc,
                                    int64_t* out_i, double* out_d, uint8_t* out_has) {
@@ -2286,6 +2360,11 @@ extern "C" int64_t bkgpu_agg_fetch(BkgAggOut* o, int sorted, int64_t max_groups,
     const int naggs = o->q.n_aggs;
     std::vector<int64_t> order(n);
     for (int64_t i = 0; i < n; i++) order[i] = i;
+    /* recover per-key encodings (spec-packed keys, pack_group_keys) */
+    std::vector<uint64_t> unp((size_t)(n > 0 ? n : 1) * BK_MAX_GROUP);
+    for (int64_t i = 0; i < n; i++)
+        unpack_group_keys(o->q, hg.k0[i], hg.k1[i], hg.flags[i],
+                          unp.data() + (size_t)i * BK_MAX_GROUP);
     if (sorted) {
         const BkQuerySpec& q = o->q;
         std::sort(order.begin(), order.end(), [&](int64_t a, int64_t b) {
@@ -2293,8 +2372,8 @@ extern "C" int64_t bkgpu_agg_fetch(BkgAggOut* o, int sorted, int64_t max_groups,
             if (hg.flags[a] != hg.flags[b]) return hg.flags[a] < hg.flags[b];
             for (int k = 0; k < q.n_group; k++) {
                 if ((hg.flags[a] >> (7 - k)) & 1) continue;
-                uint64_t ea = k == 0 ? hg.k0[a] : hg.k1[a];
-                uint64_t eb = k == 0 ? hg.k0[b] : hg.k1[b];
+                uint64_t ea = unp[(size_t)a * BK_MAX_GROUP + k];
+                uint64_t eb = unp[(size_t)b * BK_MAX_GROUP + k];
                 if (ea != eb) return ea < eb;
             }
             return false;
@@ -2304,8 +2383,8 @@ extern "C" int64_t bkgpu_agg_fetch(BkgAggOut* o, int sorted, int64_t max_groups,
     for (int64_t i = 0; i < out_n; i++) {
         int64_t g = order[i];
         flags[i] = (uint8_t)hg.flags[g];
-        enc[i * BK_MAX_GROUP + 0] = hg.k0[g];
-        enc[i * BK_MAX_GROUP + 1] = hg.k1[g];
+        for (int k = 0; k < BK_MAX_GROUP; k++)
+            enc[i * BK_MAX_GROUP + k] = unp[(size_t)g * BK_MAX_GROUP + k];
         for (int a = 0; a < naggs; a++) {
             uint64_t val = hg.states[(size_t)g * 2 * naggs + 2 * a];
             uint64_t cnt = hg.states[(size_t)g * 2 * naggs + 2 * a + 1];

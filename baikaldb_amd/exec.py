@@ -26,6 +26,8 @@ class BkPlanNodeDesc(C.Structure):
                 ("conjuncts", BkConjunct * BK_MAX_CONJ),
                 ("n_group", C.c_int32),
                 ("group_cols", C.c_int32 * BK_MAX_GROUP),
+                ("group_bits", C.c_int32 * BK_MAX_GROUP),
+                ("group_base", C.c_int64 * BK_MAX_GROUP),
                 ("n_aggs", C.c_int32),
                 ("aggs", BkAggSpec * BK_MAX_AGGS),
                 ("expected_groups", C.c_int64),
@@ -124,13 +126,17 @@ def window_node(part_col, order, fns, out_cols, num_children=1, limit=-1,
 
 
 def agg_node(group, aggs, expected_groups=1 << 16, merge=False, num_children=1,
-             limit=-1):
+             limit=-1, group_bits=(), group_base=()):
     d = BkPlanNodeDesc()
     d.node_type, d.num_children = (MERGE_AGG if merge else AGG), num_children
     d.limit = limit
     d.n_group = len(group)
     for i, c in enumerate(group):
         d.group_cols[i] = c
+        if i < len(group_bits):
+            d.group_bits[i] = group_bits[i]
+        if i < len(group_base):
+            d.group_base[i] = group_base[i]
     d.n_aggs = len(aggs)
     for i, (name, col) in enumerate(aggs):
         d.aggs[i].agg_type = _AGGS[name] if isinstance(name, str) else name

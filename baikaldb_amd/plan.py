@@ -12,7 +12,7 @@ TYPE_INT64, TYPE_DOUBLE, TYPE_STRING, TYPE_DATETIME = 6, 12, 13, 14
 OP_EQ, OP_NE, OP_GT, OP_GE, OP_LT, OP_LE = 0, 1, 2, 3, 4, 5
 AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_AVG, AGG_MIN, AGG_MAX = 0, 1, 2, 3, 4, 5
 AGG_COUNT_DISTINCT, AGG_SUM_DISTINCT = 6, 7
-BK_MAX_GROUP, BK_MAX_CONJ, BK_MAX_AGGS = 2, 8, 8
+BK_MAX_GROUP, BK_MAX_CONJ, BK_MAX_AGGS = 4, 8, 8
 
 _OPS = {"=": OP_EQ, "!=": OP_NE, ">": OP_GT, ">=": OP_GE, "<": OP_LT,
         "<=": OP_LE, "in": 6, "not_in": 7,
@@ -62,6 +62,8 @@ class BkQuerySpec(C.Structure):
                 ("conjuncts", BkConjunct * BK_MAX_CONJ),
                 ("group_cols", C.c_int32 * BK_MAX_GROUP),
                 ("group_types", C.c_int32 * BK_MAX_GROUP),
+                ("group_bits", C.c_int32 * BK_MAX_GROUP),
+                ("group_base", C.c_int64 * BK_MAX_GROUP),
                 ("aggs", BkAggSpec * BK_MAX_AGGS),
                 ("agg_in_types", C.c_int32 * BK_MAX_AGGS)]
 
@@ -75,11 +77,16 @@ class QueryPlan:
     aggs:      (name, col) with name in count_star/count/sum/avg/min/max.
     """
 
-    def __init__(self, col_types, conjuncts=(), group=(), aggs=()):
+    def __init__(self, col_types, conjuncts=(), group=(), aggs=(),
+                 group_bits=(), group_base=()):
         self.col_types = list(col_types)
         self.conjuncts = list(conjuncts)
         self.group = list(group)
         self.aggs = list(aggs)
+        # >2 group keys pack into two 64-bit words: per-key bit width and
+        # base value, declared by the caller (bk_common.h BkQuerySpec)
+        self.group_bits = list(group_bits)
+        self.group_base = list(group_base)
 
     def to_spec(self):
         q = BkQuerySpec()
@@ -114,6 +121,10 @@ class QueryPlan:
         for i, col in enumerate(self.group):
             q.group_cols[i] = col
             q.group_types[i] = self.col_types[col]
+            if i < len(self.group_bits):
+                q.group_bits[i] = self.group_bits[i]
+            if i < len(self.group_base):
+                q.group_base[i] = self.group_base[i]
         q.n_aggs = len(self.aggs)
         for i, (name, col) in enumerate(self.aggs):
             q.aggs[i].agg_type = _AGGS[name] if isinstance(name, str) else name

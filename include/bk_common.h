@@ -217,7 +217,7 @@ typedef struct BkOrderSpec {
  * <=3 conjuncts, <=4 aggregates; we allow a little headroom). */
 #define BK_MAX_COLS      16
 #define BK_MAX_CONJUNCTS 8
-#define BK_MAX_GROUP     2
+#define BK_MAX_GROUP     4
 #define BK_MAX_AGGS      8
 
 /* A full query descriptor over one columnar table: the pb::Plan subset the
@@ -230,6 +230,18 @@ typedef struct BkQuerySpec {
     BkConjunct conjuncts[BK_MAX_CONJUNCTS];
     int32_t    group_cols[BK_MAX_GROUP];
     int32_t    group_types[BK_MAX_GROUP];  /* BkType of each group col */
+    /* >2 group keys pack into the engine's two 64-bit key words. The CALLER
+     * declares, per key, a bit width and a base VALUE (inclusive minimum);
+     * the engine packs (enc(v) - enc(base)) into `bits` bits. bits == 0
+     * (the default) means a full 64-bit word — with <= 2 keys and all-zero
+     * bits this reproduces the plain two-word layout. Declaring the widths
+     * in the QUERY (not from data statistics) keeps the packed keys
+     * identical on every rank, so partial-aggregate blobs merge across
+     * GPUs. Values outside [base, base + 2^bits) are caller error.
+     * Widths must sum to <= 64 bits per word (keys never straddle words);
+     * DOUBLE keys require bits == 0. */
+    int32_t    group_bits[BK_MAX_GROUP];
+    int64_t    group_base[BK_MAX_GROUP];
     BkAggSpec  aggs[BK_MAX_AGGS];
     int32_t    agg_in_types[BK_MAX_AGGS];  /* BkType of each agg input col */
 } BkQuerySpec;

@@ -44,6 +44,8 @@ typedef struct BkPlanNodeDesc {
     /* AGG/MERGE_AGG payload */
     int32_t    n_group;
     int32_t    group_cols[BK_MAX_GROUP];
+    int32_t    group_bits[BK_MAX_GROUP];   /* key packing, bk_common.h */
+    int64_t    group_base[BK_MAX_GROUP];
     int32_t    n_aggs;
     BkAggSpec  aggs[BK_MAX_AGGS];
     int64_t    expected_groups;

@@ -13,7 +13,7 @@ DIST_UNIFORM, DIST_CUBESKEW, DIST_DICT, DIST_SUMU16 = 0, 1, 2, 3
 OP_EQ, OP_NE, OP_GT, OP_GE, OP_LT, OP_LE = 0, 1, 2, 3, 4, 5
 AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_AVG, AGG_MIN, AGG_MAX = 0, 1, 2, 3, 4, 5
 
-BK_MAX_GROUP = 2
+BK_MAX_GROUP = 4
 BK_MAX_CONJ = 8
 BK_MAX_AGGS = 8
 
@@ -54,6 +54,8 @@ class BkQuerySpec(C.Structure):
                 ("conjuncts", BkConjunct * BK_MAX_CONJ),
                 ("group_cols", C.c_int32 * BK_MAX_GROUP),
                 ("group_types", C.c_int32 * BK_MAX_GROUP),
+                ("group_bits", C.c_int32 * BK_MAX_GROUP),
+                ("group_base", C.c_int64 * BK_MAX_GROUP),
                 ("aggs", BkAggSpec * BK_MAX_AGGS),
                 ("agg_in_types", C.c_int32 * BK_MAX_AGGS)]
 
@@ -70,7 +72,8 @@ class _OrcAggResult(C.Structure):
                 ("g_flag", C.POINTER(C.c_uint8)), ("g_enc", C.POINTER(C.c_uint64))]
 
 
-def make_query(conjuncts=(), group=(), aggs=(), col_types=None):
+def make_query(conjuncts=(), group=(), aggs=(), col_types=None,
+               group_bits=(), group_base=()):
     """Build a BkQuerySpec.
 
     conjuncts: list of (col, op, cmp_type, literal[, fn])
@@ -105,6 +108,10 @@ def make_query(conjuncts=(), group=(), aggs=(), col_types=None):
     for i, col in enumerate(group):
         q.group_cols[i] = col
         q.group_types[i] = col_types[col]
+        if i < len(group_bits):
+            q.group_bits[i] = group_bits[i]
+        if i < len(group_base):
+            q.group_base[i] = group_base[i]
     q.n_aggs = len(aggs)
     for i, (at, col) in enumerate(aggs):
         q.aggs[i].agg_type = at

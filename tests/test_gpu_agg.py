@@ -42,13 +42,15 @@ def orc():
 
 
 def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
-             expected_groups=1 << 14, seed=SEED):
+             expected_groups=1 << 14, seed=SEED, group_bits=(), group_base=()):
     """conjuncts: (col, op_sym, lit); aggs: (name, col)."""
     t = eng.create_table(spec_rows, n)
     try:
         eng.generate(t, seed)
         from baikaldb_amd import QueryPlan
-        plan = QueryPlan(t.col_types, conjuncts=conjuncts, group=group, aggs=aggs)
+        plan = QueryPlan(t.col_types, conjuncts=conjuncts, group=group,
+                         aggs=aggs, group_bits=group_bits,
+                         group_base=group_base)
         res = eng.filter_agg(t, plan, expected_groups=expected_groups)
         try:
             got = res.fetch(sorted=True)
@@ -77,7 +79,8 @@ def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
                              not isinstance(lit, (list, tuple))) or \
             isinstance(lit, float) else TYPE_INT64
         oconj.append((col, ops[op], ct, lit, fn))
-    q = make_query(oconj, group, [(aggmap[a], c) for a, c in aggs], col_types)
+    q = make_query(oconj, group, [(aggmap[a], c) for a, c in aggs], col_types,
+                   group_bits=group_bits, group_base=group_base)
     exp = orc.filter_agg(cols, valids, col_types, q, nthreads=nthreads,
                          dict_seed=seed)
     return got, exp
@@ -431,3 +434,71 @@ def test_large_in_list(eng, orc):
         (valids[1] != 0 if valids[1] is not None else True)
     assert exp["rows_passed"] == int(sel.sum())
     assert_parity(got, exp, [("count_star", -1), ("sum", 0)], types)
+
+
+def test_three_and_four_group_keys(eng, orc):
+    """3-4 group keys spec-packed into the two 64-bit key words
+    (bk_common.h group_bits/group_base): parity incl NULL keys and the
+    canonical per-key output order."""
+    specs = [(TYPE_INT64, D_UNI, 0, 1000, 0),          # 10 bits
+             (TYPE_STRING, D_DICT, 50, 0, 150_000),    # 6 bits, nullable
+             (TYPE_INT64, D_UNI, -500, 500, 120_000),  # 10 bits, base -500
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 0),       # full word
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 0)]
+    aggs = [("count_star", -1), ("sum", 3), ("min", 2), ("avg", 4)]
+    # 3 keys: 10 + 6 + 10 = 26 bits in word 0
+    got, exp = run_both(eng, orc, specs, 120_000,
+                        [(3, "<", int((1 << 31) * 0.8))], [0, 1, 2], aggs,
+                        expected_groups=1 << 18,
+                        group_bits=[10, 6, 10], group_base=[0, 0, -500])
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
+    assert got["ngroups"] > 10_000          # genuinely multi-key
+    # 4 keys: 26 bits + full word -> both words
+    aggs4 = [("count_star", -1), ("min", 4)]
+    got, exp = run_both(eng, orc, specs, 60_000, [], [0, 1, 2, 3], aggs4,
+                        expected_groups=1 << 17,
+                        group_bits=[10, 6, 10, 0], group_base=[0, 0, -500, 0])
+    assert_parity(got, exp, aggs4, [s[0] for s in specs])
+
+
+def test_packed_keys_shard_merge_equals_whole(eng, orc):
+    """Cross-shard merge with spec-packed 3-key groups: packing is derived
+    from the QUERY, so shards pack identically and blobs merge correctly
+    (the property multi-GPU depends on)."""
+    import torch
+    specs = [(TYPE_INT64, D_UNI, 0, 200, 0),
+             (TYPE_STRING, D_DICT, 30, 0, 0),
+             (TYPE_INT64, D_UNI, -100, 100, 100_000),
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 0)]
+    n = 100_000
+    from baikaldb_amd import QueryPlan
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED)
+        plan = QueryPlan(t.col_types, conjuncts=[(3, ">", 1 << 29)],
+                         group=[0, 1, 2],
+                         aggs=[("count_star", -1), ("sum", 3)],
+                         group_bits=[8, 5, 8], group_base=[0, 0, -100])
+        whole = eng.filter_agg(t, plan, expected_groups=1 << 16)
+        wf = whole.fetch(sorted=True)
+        whole.free()
+        bounds = [0, n // 3, 2 * n // 3, n]
+        shards = [eng.filter_agg(t, plan, row_begin=bounds[i],
+                                 row_end=bounds[i + 1],
+                                 expected_groups=1 << 16)
+                  for i in range(3)]
+        dst = shards[0]
+        for sh in shards[1:]:
+            nb = sh.export_bytes()
+            buf = torch.empty(nb, dtype=torch.uint8, device="cuda")
+            sh.export_to(buf.data_ptr(), nb)
+            dst.merge_blob(buf.data_ptr(), sh.ngroups)
+        mf = dst.fetch(sorted=True)
+        for sh in shards:
+            sh.free()
+    finally:
+        t.free()
+    assert mf["ngroups"] == wf["ngroups"]
+    assert np.array_equal(mf["enc"], wf["enc"])
+    assert np.array_equal(mf["flags"], wf["flags"])
+    assert np.array_equal(mf["agg_i"], wf["agg_i"])
