@@ -238,7 +238,7 @@ class Oracle:
                 "flags": np.ctypeslib.as_array(r.g_flag, shape=(max(ng, 1),))[:ng].copy(),
                 "enc": np.ctypeslib.as_array(
                     r.g_enc, shape=(max(ng, 1) * BK_MAX_GROUP,))[:ng * BK_MAX_GROUP]
-                    .copy().reshape(ng, BK_MAX_GROUP) if ng else np.zeros((0, 2), np.uint64),
+                    .copy().reshape(ng, BK_MAX_GROUP) if ng else np.zeros((0, BK_MAX_GROUP), np.uint64),
                 "agg_i": np.ctypeslib.as_array(
                     r.out_i, shape=(max(na * ng, 1),))[:na * ng].copy().reshape(na, ng)
                     if ng else np.zeros((na, 0), np.int64),
