@@ -1860,9 +1860,34 @@ struct EvTimer {
 
 /* partitioned pipeline for one attempt; returns 0 ok (err flag still to be
  * checked by caller), -1 hard error. */
+#include <sys/time.h>
+static double now_ms() {
+    struct timeval tv;
+    gettimeofday(&tv, nullptr);
+    return tv.tv_sec * 1e3 + tv.tv_usec * 1e-3;
+}
+
+/* pipelined variant of run_partitioned: the row range splits into chunks
+ * and consecutive chunks run on TWO HIP streams, so one chunk's part_agg
+ * (LDS-latency-bound, ~1.2 TB/s — NOT HBM-saturated) overlaps the next
+ * chunk's histo/scatter (HBM-bound). The global table is shared — its claim
+ * protocol is already additive/concurrent (same as inter-block claims), so
+ * results are identical to the single-pass run. Per-slot buffers are reused
+ * only after their stream synchronizes (the pool is not stream-aware). */
+static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
+                                int64_t row_begin, int64_t row_end,
+                                int64_t expected_groups, int nchunks);
+
 static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
                            int64_t row_begin, int64_t row_end,
                            int64_t expected_groups) {
+    {
+        int pipe = 0;
+        if (const char* e = getenv("BK_PIPE")) pipe = atoi(e);
+        if (pipe > 1 && row_end - row_begin >= 4 * pipe)
+            return run_partitioned_pipe(o, t, q, row_begin, row_end,
+                                        expected_groups, pipe);
+    }
     const int stride = SLOT_HDR + 2 * q->n_aggs;
     int64_t range = row_end - row_begin;
     if (range >= (int64_t)UINT32_MAX) { set_err("range too large for one pass"); return -1; }
@@ -2014,16 +2039,148 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     return 0;
 }
 
+static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
+                                int64_t row_begin, int64_t row_end,
+                                int64_t expected_groups, int nchunks) {
+    const int stride = SLOT_HDR + 2 * q->n_aggs;
+    RecLayout lay;
+    build_rec_layout(t, q, &lay);
+    uint32_t P = 64;
+    while ((int64_t)P < expected_groups / 1024 && P < 4096) P <<= 1;
+    if (const char* e = getenv("BK_PART_P")) P = (uint32_t)atoi(e);
+    int nblocks = 2048;
+    if (const char* e = getenv("BK_PART_BLOCKS")) nblocks = atoi(e);
+    int threads = 512;
+    if (const char* e = getenv("BK_PART_THREADS")) threads = atoi(e);
+    if (threads != 512 && threads != 1024) threads = 256;
+    auto histo_fn = k_part_histo<256>;
+    auto scat_fn = k_part_scatter<256>;
+    if (threads == 512) { histo_fn = k_part_histo<512>; scat_fn = k_part_scatter<512>; }
+    else if (threads == 1024) { histo_fn = k_part_histo<1024>; scat_fn = k_part_scatter<1024>; }
+    int at = 1024;
+    if (const char* e = getenv("BK_AGG_THREADS")) at = atoi(e);
+    int ilp = 2;
+    if (const char* e = getenv("BK_AGG_ILP")) ilp = atoi(e);
+    auto agg_fn = k_part_agg<1024, 2>;
+    if (at == 512) agg_fn = ilp == 2 ? k_part_agg<512, 2> : k_part_agg<512, 1>;
+    else if (at == 256) agg_fn = ilp == 2 ? k_part_agg<256, 2> : k_part_agg<256, 1>;
+    else { at = 1024; if (ilp != 2) agg_fn = k_part_agg<1024, 1>; }
+    uint32_t agg_lds_slots = 2048;
+    size_t agg_lds_cap = 135 * 1024;
+    if (const char* e = getenv("BK_AGG_LDS_KB")) agg_lds_cap = (size_t)atoi(e) * 1024;
+    while ((size_t)agg_lds_slots * stride * 8 > agg_lds_cap) agg_lds_slots >>= 1;
+    size_t agg_lds_bytes = ((size_t)agg_lds_slots * stride + 1) * 8;
+
+    uint32_t hot_lds_cap = 50 * 1024;
+    uint32_t hot_slots = 512;
+    while ((size_t)hot_slots * stride * 8 > hot_lds_cap) hot_slots >>= 1;
+    uint32_t hot_cap = hot_slots / 2u, hot_probe = 4, hot_min = 1024;
+    size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
+    size_t sc_lds = (size_t)P * 8;
+
+    static hipStream_t streams[2] = {nullptr, nullptr};
+    if (!streams[0]) {
+        HIP_CHECK(hipStreamCreate(&streams[0]));
+        HIP_CHECK(hipStreamCreate(&streams[1]));
+    }
+    const uint32_t nch = (uint32_t)((nblocks + OFFS_CHUNK - 1) / OFFS_CHUNK);
+    int64_t range = row_end - row_begin;
+    int64_t per = (range + nchunks - 1) / nchunks;
+    if (per >= (int64_t)UINT32_MAX) { set_err("chunk too large"); return -1; }
+
+    struct Slot {
+        uint16_t* bucketid = nullptr;
+        uint32_t *H = nullptr, *totals = nullptr, *base = nullptr, *S = nullptr;
+        uint64_t* total_dev = nullptr;
+        uint64_t* rec = nullptr;
+    } sl[2];
+    auto cleanup = [&]() {
+        for (int i = 0; i < 2; i++) {
+            pool_free(sl[i].bucketid); pool_free(sl[i].H); pool_free(sl[i].totals);
+            pool_free(sl[i].base); pool_free(sl[i].S); pool_free(sl[i].total_dev);
+            pool_free(sl[i].rec);
+        }
+    };
+    #define PPCHECK(x) do { if ((x) != hipSuccess) { \
+        snprintf(g_err, sizeof g_err, "pipe %s:%d %s", __FILE__, __LINE__, \
+                 hipGetErrorString(hipGetLastError())); \
+        (void)hipDeviceSynchronize(); cleanup(); return -1; } } while (0)
+    for (int i = 0; i < 2; i++) {
+        PPCHECK(pool_alloc((void**)&sl[i].bucketid, (size_t)per * 2));
+        PPCHECK(pool_alloc((void**)&sl[i].H, (size_t)nblocks * P * 4));
+        PPCHECK(pool_alloc((void**)&sl[i].totals, (size_t)P * 4));
+        PPCHECK(pool_alloc((void**)&sl[i].base, (size_t)P * 4));
+        PPCHECK(pool_alloc((void**)&sl[i].S, (size_t)nch * P * 4));
+        PPCHECK(pool_alloc((void**)&sl[i].total_dev, 8));
+    }
+    DevCols dc = table_cols(t);
+    PPCHECK(hipDeviceSynchronize());
+    double t0 = now_ms();
+    for (int c = 0; c < nchunks; c++) {
+        int si = c & 1;
+        hipStream_t st = streams[si];
+        int64_t cb = row_begin + (int64_t)c * per;
+        int64_t ce = cb + per < row_end ? cb + per : row_end;
+        if (ce <= cb) break;
+        if (c >= 2) {
+            /* slot reuse: its previous chunk must be fully done before we
+             * recycle the record buffer (pool is not stream-aware) */
+            PPCHECK(hipStreamSynchronize(st));
+            pool_free(sl[si].rec);
+            sl[si].rec = nullptr;
+        }
+        PPCHECK(hipMemsetAsync(sl[si].totals, 0, (size_t)P * 4, st));
+        hipLaunchKernelGGL(histo_fn, dim3(nblocks), dim3(threads), histo_lds, st,
+                           dc, *q, cb, ce, P, sl[si].bucketid, sl[si].H,
+                           o->table, o->nslots - 1, (o->nslots * 7) / 8,
+                           o->ctrs, o->ctrs + 1, o->err, hot_slots,
+                           hot_cap, hot_probe, hot_min);
+        hipLaunchKernelGGL(k_part_totals, dim3((P * nch + 255) / 256), dim3(256),
+                           0, st, sl[si].H, nblocks, P, sl[si].S, sl[si].totals);
+        hipLaunchKernelGGL(k_part_scan, dim3(1), dim3(1024), 0, st,
+                           sl[si].totals, P, sl[si].base, sl[si].total_dev);
+        PPCHECK(hipGetLastError());
+        uint64_t total = 0;
+        PPCHECK(hipMemcpyAsync(&total, sl[si].total_dev, 8,
+                               hipMemcpyDeviceToHost, st));
+        PPCHECK(hipStreamSynchronize(st));   /* other stream keeps running */
+        hipLaunchKernelGGL(k_part_offsets, dim3((P * nch + 255) / 256), dim3(256),
+                           0, st, sl[si].H, nblocks, P, sl[si].base, sl[si].S);
+        if (total > 0) {
+            PPCHECK(pool_alloc((void**)&sl[si].rec,
+                               (size_t)total * lay.nwords * 8));
+            hipLaunchKernelGGL(scat_fn, dim3(nblocks), dim3(threads), sc_lds, st,
+                               dc, *q, lay, cb, ce, P, sl[si].bucketid,
+                               sl[si].H, sl[si].rec, total, 0);
+            uint64_t chunk_sz = std::max<uint64_t>(
+                32768, std::min<uint64_t>(1u << 20, total / 2048));
+            if (const char* e = getenv("BK_AGG_CHUNK")) chunk_sz = (uint64_t)atoll(e);
+            uint64_t nrec_chunks = (total + chunk_sz - 1) / chunk_sz;
+            uint32_t grid = (uint32_t)std::min<uint64_t>(nrec_chunks, 32768);
+            hipLaunchKernelGGL(agg_fn, dim3(grid), dim3(at), agg_lds_bytes, st,
+                               *q, lay, sl[si].rec, total, chunk_sz,
+                               o->table, o->nslots - 1, (o->nslots * 7) / 8,
+                               o->ctrs, o->err, agg_lds_slots);
+        }
+        PPCHECK(hipGetLastError());
+    }
+    PPCHECK(hipStreamSynchronize(streams[0]));
+    PPCHECK(hipStreamSynchronize(streams[1]));
+    double wall = now_ms() - t0;
+    o->kernel_ms = (float)wall;
+    o->n_kernels = 1;
+    o->t_ms[0] = (float)wall;
+    snprintf(o->k_names[0], 16, "pipeline");
+    cleanup();
+    #undef PPCHECK
+    return 0;
+}
+
 /* below this many expected groups the per-WG LDS table absorbs the stream
  * and the single fused kernel wins; above it, partition. */
 #define FUSED_MAX_GROUPS 512
 
-#include <sys/time.h>
-static double now_ms() {
-    struct timeval tv;
-    gettimeofday(&tv, nullptr);
-    return tv.tv_sec * 1e3 + tv.tv_usec * 1e-3;
-}
+
 static int g_debug_timing = -1;
 static bool debug_timing() {
     if (g_debug_timing < 0) g_debug_timing = getenv("BK_DEBUG") ? 1 : 0;
