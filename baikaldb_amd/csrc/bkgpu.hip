@@ -2156,7 +2156,12 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
                 32768, std::min<uint64_t>(1u << 20, total / 2048));
             if (const char* e = getenv("BK_AGG_CHUNK")) chunk_sz = (uint64_t)atoll(e);
             uint64_t nrec_chunks = (total + chunk_sz - 1) / chunk_sz;
-            uint32_t grid = (uint32_t)std::min<uint64_t>(nrec_chunks, 32768);
+            /* cap the aggregate's grid so the OTHER stream's HBM-bound
+             * kernels keep CUs: part_agg blocks hold 135 KB LDS (1/CU) and
+             * a full grid would occupy the whole chip */
+            uint64_t gmax = 32768;
+            if (const char* e = getenv("BK_PIPE_AGG_GRID")) gmax = (uint64_t)atoll(e);
+            uint32_t grid = (uint32_t)std::min<uint64_t>(nrec_chunks, gmax);
             hipLaunchKernelGGL(agg_fn, dim3(grid), dim3(at), agg_lds_bytes, st,
                                *q, lay, sl[si].rec, total, chunk_sz,
                                o->table, o->nslots - 1, (o->nslots * 7) / 8,

@@ -13,8 +13,12 @@ conj = [(0,"<",1<<30),(1,"<",int((1<<31)*0.9)),(7,"!=",63)]
 plan = QueryPlan(t.col_types, conjuncts=conj, group=[2,7],
                  aggs=[("count_star",-1),("sum",3),("sum",4),("avg",5)])
 ref = None
-for pipe in ["0", "2", "3", "4", "6", "8"]:
+CASES = [("0", None), ("2", None), ("2", "192"), ("2", "128"), ("2", "96"),
+         ("3", "128"), ("4", "128"), ("4", "96")]
+for pipe, gmax in CASES:
     os.environ["BK_PIPE"] = pipe
+    if gmax: os.environ["BK_PIPE_AGG_GRID"] = gmax
+    else: os.environ.pop("BK_PIPE_AGG_GRID", None)
     best = None
     for rep in range(3):
         r = eng.filter_agg(t, plan, expected_groups=1<<21)
@@ -29,5 +33,6 @@ for pipe in ["0", "2", "3", "4", "6", "8"]:
     else:
         ok = "PARITY-OK" if (ng == ref[0] and np.array_equal(head, ref[1])
                              and rp == ref[2]) else "PARITY-MISMATCH"
-    print(f"pipe={pipe:2s} kernel={ms:7.2f} ms ng={ng} {ok}", flush=True)
+    print(f"pipe={pipe:2s} agg_grid={gmax or '-':4s} kernel={ms:7.2f} ms "
+          f"ng={ng} {ok}", flush=True)
 t.free()
