@@ -300,3 +300,38 @@ def test_cpp_embedding_example(tmp_path):
     assert expect in out.stdout, f"wanted {expect!r} in:\n{out.stdout[:2000]}"
     assert "top-5 by (v, g):" in out.stdout
     assert "row_number=1" in out.stdout
+
+
+@pytest.mark.gpu
+def test_parquet_to_window(lib, tmp_path):
+    """Ingested parquet table through the window path (rank over a stored
+    VARCHAR partition)."""
+    import torch
+    if torch.cuda.is_available():
+        torch.cuda.init()
+    from baikaldb_amd import GpuEngine
+    from baikaldb_amd.engine import GpuTable
+    rng = np.random.default_rng(31)
+    n = 30_000
+    g = [f"p{i%9}" for i in range(n)]
+    v = rng.integers(0, 1000, n, dtype=np.int64)
+    path = str(tmp_path / "w.parquet")
+    tab = pa.table({"g": pa.array(g), "v": pa.array(v)})
+    pq.write_table(tab, path, compression=None, use_dictionary=True,
+                   data_page_version="1.0", write_statistics=False)
+    eng = GpuEngine()
+    h = lib.bkgpu_table_from_parquet(path.encode())
+    assert h, lib.bkparquet_last_error()
+    t = GpuTable(eng, h, [13, 6], n)
+    got = eng.window(t, [("row_number", -1), ("max", 1)], part_col=0,
+                     order=[(1, 1, 1)])
+    t.free()
+    assert got["n"] == n
+    # per partition: row_number restarts; max = partition max broadcast
+    garr = np.array(g)
+    r = got["rowids"]
+    for p in range(9):
+        sel = garr[r] == f"p{p}"
+        assert np.array_equal(got["out_i"][0][sel],
+                              np.arange(1, sel.sum() + 1))
+        assert np.all(got["out_i"][1][sel] == v[garr == f"p{p}"].max())

@@ -692,3 +692,40 @@ def test_gpu_range_val_parity(eng, orc, vframe):
     d = np.abs(got["out_d"] - exp["out_d"])
     tol = 1e-9 * (np.abs(exp["out_d"]) + 100)
     assert np.all(d[mask] <= tol[mask])
+
+
+@pytest.mark.gpu
+def test_gpu_window_empty_selection(eng, orc):
+    """WHERE eliminates every row: window returns 0 rows cleanly."""
+    from baikaldb_amd import QueryPlan
+    t = eng.create_table([(TYPE_INT64, D_UNI, 0, 100, 0)], 10_000)
+    try:
+        eng.generate(t, 1)
+        plan = QueryPlan(t.col_types, conjuncts=[(0, "<", -5)])
+        got = eng.window(t, [("row_number", -1)], part_col=0, order=[],
+                         plan=plan)
+    finally:
+        t.free()
+    assert got["n"] == 0
+
+
+@pytest.mark.gpu
+def test_gpu_window_single_row_partitions(eng, orc):
+    """Every row its own partition (distinct keys): rank fns all 1,
+    aggregates identity."""
+    specs = [(TYPE_INT64, D_UNI, 0, 1 << 62, 0),
+             (TYPE_INT64, D_UNI, -50, 50, 0)]
+    n = 5000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, 9)
+        got = eng.window(t, [("row_number", -1), ("rank", -1),
+                             ("percent_rank", -1), ("sum", 1), ("lag", 1, 1)],
+                         part_col=0, order=[(1, 1, 1)])
+    finally:
+        t.free()
+    assert got["n"] == n
+    assert np.all(got["out_i"][0] == 1)
+    assert np.all(got["out_i"][1] == 1)
+    assert np.all(got["out_d"][2] == 0.0)
+    assert np.all(got["out_null"][4] == 1)   # lag leaves every 1-row partition
