@@ -2,6 +2,15 @@ import os
 import subprocess
 import sys
 
+# Deterministic import order: arrow's C++ runtime loads BEFORE torch/HIP.
+# (A rare collection-time segfault was observed once with the reverse order;
+# pinning the order here removes the variability.)
+try:
+    import pyarrow  # noqa: F401
+    import pyarrow.parquet  # noqa: F401
+except ImportError:
+    pass
+
 import pytest
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
