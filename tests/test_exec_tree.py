@@ -335,3 +335,39 @@ def test_window_frame_through_exec_surface(eng, orc):
         run[start:end] = _np.cumsum(cols[2][order[start:end]])
         start = end
     assert _np.array_equal(vi[:, 2], run)
+
+
+def test_three_group_keys_through_exec_surface(eng, orc):
+    """3 packed group keys through the ExecNode mirror (group_bits/base ride
+    the plan descriptor, bk_exec.h)."""
+    from baikaldb_amd import exec as bx
+    import numpy as _np
+    t, cols, valids, types = make_table(eng, orc, n=40_000)
+    try:
+        nodes = [bx.agg_node(group=[1, 2, 0],
+                             aggs=[("count_star", -1), ("sum", 2)],
+                             group_bits=[11, 9, 0],
+                             group_base=[0, 0, 0]),
+                 bx.filter_node(types, [(0, "<", int((1 << 31) * 0.6))]),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
+    sel = cols[0] < int((1 << 31) * 0.6)
+    trip = list(zip(cols[1][sel], cols[2][sel], cols[0][sel]))
+    uniq = sorted(set(trip))
+    assert tags.shape[0] == len(uniq)
+    # spot-check a handful of rows (canonical order == sorted tuples)
+    import collections
+    cnt = collections.Counter(trip)
+    sm = collections.defaultdict(int)
+    for k in trip:
+        sm[k] += k[1]
+    for r in (0, len(uniq) // 2, len(uniq) - 1):
+        k = (vi[r, 0], vi[r, 1], vi[r, 2])
+        assert k == uniq[r], (r, k, uniq[r])
+        assert vi[r, 3] == cnt[k]
+        assert vi[r, 4] == sm[k]
