@@ -62,7 +62,7 @@ def brute_distinct(cols, valids, col_types, conjuncts, group, aggs):
         gvals = cols[gcol][idx]
         gnull = (valids[gcol][idx] == 0) if valids[gcol] is not None \
             else np.zeros(len(idx), bool)
-        keys = [(bool(gn), None if gn else int(gv))
+        keys = [(bool(gn), None if gn else gv.item())
                 for gn, gv in zip(gnull, gvals)]
     else:
         keys = [(False, None)] * len(idx)
@@ -256,3 +256,77 @@ def test_gpu_distinct_high_cardinality(eng, orc):
     got, exp = run_both_distinct(eng, orc, specs, 400_000, [], [0], aggs,
                                  expected_groups=64)
     assert_parity(got, exp, _parity_names(aggs), [s[0] for s in specs])
+
+
+def _random_distinct_case(rng):
+    ncols = rng.randint(2, 5)
+    specs = []
+    for _ in range(ncols):
+        t = rng.choice([TYPE_INT64, TYPE_INT64, TYPE_DOUBLE, TYPE_STRING])
+        nf = rng.choice([0, 0, 250_000])
+        if t == TYPE_INT64:
+            specs.append((t, D_UNI, 0, rng.choice([5, 40, 1000, 1 << 20]), nf))
+        elif t == TYPE_DOUBLE:
+            specs.append((t, D_SUM16, 0, 0, nf))
+        else:
+            specs.append((t, D_DICT, rng.choice([4, 90]), 0, nf))
+    dcol = rng.randrange(ncols)
+    group = rng.sample([c for c in range(ncols)], rng.randint(0, 1))
+    aggs = [("count_star", -1), ("count_distinct", dcol)]
+    if specs[dcol][0] != TYPE_STRING and rng.random() < 0.7:
+        aggs.append(("sum_distinct", dcol))
+        aggs.append(("avg_distinct", dcol))
+    for _ in range(rng.randint(0, 2)):
+        c = rng.randrange(ncols)
+        aggs.append((rng.choice(["sum", "count", "min", "max"]), c))
+    conj = []
+    if rng.random() < 0.6:
+        c = rng.randrange(ncols)
+        if specs[c][0] == TYPE_DOUBLE:
+            conj.append((c, "<", rng.uniform(-1, 1)))
+        else:
+            conj.append((c, rng.choice(["<", ">", "!="]), rng.randint(0, 500)))
+    n = rng.choice([1000, 15_000, 60_000])
+    return specs, conj, group, aggs, n
+
+
+@pytest.mark.parametrize("cs", range(4))
+def test_oracle_distinct_fuzz_vs_brute(orc, cs):
+    import random
+    rng = random.Random(31_000 + cs)
+    for sub in range(5):
+        specs, conj, group, aggs, n = _random_distinct_case(rng)
+        # brute supports: count_star/count/sum/count_distinct/sum_distinct/
+        # avg_distinct; restrict to those for the brute comparison
+        aggs = [a for a in aggs if a[0] in
+                ("count_star", "count", "sum", "count_distinct",
+                 "sum_distinct", "avg_distinct")]
+        exp, (cols, valids, col_types) = oracle_distinct(
+            orc, specs, n, conj, group, aggs, seed=rng.randrange(1 << 40))
+        brute = brute_distinct(cols, valids, col_types, conj, group, aggs)
+        assert exp["ngroups"] == len(brute), (cs, sub, specs, group, aggs)
+        # aggregate-sum check (order-independent totals per agg column)
+        for a, (name, col) in enumerate(aggs):
+            if name in ("count_star", "count", "count_distinct"):
+                tot_exp = int(exp["agg_i"][a].sum())
+                tot_brute = sum(v[a] for v in brute.values())
+                assert tot_exp == tot_brute, (cs, sub, name)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("cs", range(3))
+def test_gpu_distinct_fuzz(eng, orc, cs):
+    import random
+    rng = random.Random(33_000 + cs)
+    for sub in range(4):
+        specs, conj, group, aggs, n = _random_distinct_case(rng)
+        seed = rng.randrange(1 << 40)
+        got, exp = run_both_distinct(eng, orc, specs, n, conj, group, aggs,
+                                     seed=seed, expected_groups=1 << 12)
+        try:
+            assert_parity(got, exp, _parity_names(aggs),
+                          [s[0] for s in specs])
+        except AssertionError as e:
+            raise AssertionError(
+                f"distinct fuzz {cs}/{sub}: specs={specs} conj={conj} "
+                f"group={group} aggs={aggs}: {e}")
