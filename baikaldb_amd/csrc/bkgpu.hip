@@ -1887,7 +1887,8 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
          * (measured -1.8 ms on the 1e9-row config-3 shape; smaller ranges
          * don't amortize the extra chunk boundaries). BK_PIPE overrides
          * (0/1 = off, N = chunk count). */
-        int pipe = row_end - row_begin >= 400 * 1000 * 1000 ? 2 : 0;
+        int pipe = (row_end - row_begin >= 400 * 1000 * 1000 &&
+                    expected_groups >= (1 << 20)) ? 2 : 0;
         if (const char* e = getenv("BK_PIPE")) pipe = atoi(e);
         if (pipe > 1 && row_end - row_begin >= 4 * pipe)
             return run_partitioned_pipe(o, t, q, row_begin, row_end,
@@ -1955,7 +1956,12 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     if (hot_cap > hot_slots - hot_slots / 8u) hot_cap = hot_slots - hot_slots / 8u;
     uint32_t hot_probe = 4;
     if (const char* e = getenv("BK_HOT_PROBE")) hot_probe = (uint32_t)atoi(e);
-    uint32_t hot_min = 1024;  /* lock on when >= hot_min hits per 4096 rows */
+    /* default NEVER locks on: at 512-thread histo the per-row LDS probe
+     * costs more than the absorption saves on BOTH baseline key shapes
+     * (Zipf/1e5 with 46% absorbable: 11.9 -> 9.6 ms with the hot path off;
+     * zipfoct x dict: auto-disabled anyway). BK_HOT_MIN=<=4096 re-enables
+     * the adaptive lock-on for workloads where it wins. */
+    uint32_t hot_min = 4097;
     if (const char* e = getenv("BK_HOT_MIN")) hot_min = (uint32_t)atoi(e);
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     EvTimer tm;
@@ -2079,7 +2085,7 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     uint32_t hot_lds_cap = 50 * 1024;
     uint32_t hot_slots = 512;
     while ((size_t)hot_slots * stride * 8 > hot_lds_cap) hot_slots >>= 1;
-    uint32_t hot_cap = hot_slots / 2u, hot_probe = 4, hot_min = 1024;
+    uint32_t hot_cap = hot_slots / 2u, hot_probe = 4, hot_min = 4097;
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     size_t sc_lds = (size_t)P * 8;
 
