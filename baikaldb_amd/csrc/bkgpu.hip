@@ -254,7 +254,9 @@ __device__ __forceinline__ KeyPack pack_group_keys(const DevCols& cols,
         if (!cell_valid(c, r)) {
             kp.flag |= 0x80u >> k;   /* null-flag bit, exec_node.cpp:561 */
         } else {
-            e = enc_value(c, r);
+            e = q.group_fns[k]
+                    ? bk_enc_i64(bk_scalar_fn(q.group_fns[k], cell_i64(c, r)))
+                    : enc_value(c, r);
             if (bits < 64) {
                 uint64_t eb = c.type == BK_STRING
                                   ? (uint64_t)q.group_base[k]
@@ -1781,6 +1783,9 @@ static bool any_group_bits(const BkQuerySpec* q) {
 static int build_rec_layout(BkgTable* t, const BkQuerySpec* q, RecLayout* lay) {
     int w = 1; /* word 0 = k0 (or the fused key word) */
     bool plain2 = q->n_group <= 2 && !any_group_bits(q);
+    for (int32_t k = 0; k < q->n_group; k++)
+        if (q->group_fns[k]) plain2 = false;   /* fn keys: no fused/meta tricks
+                                                  (stats are raw-column) */
     /* a dict-encoded (BK_STRING) second group key is a 32-bit code: pack it
      * into the meta word's high half (k1_word == -2) instead of spending a
      * whole record word — 20% narrower records on the config-3 shape.
@@ -2217,6 +2222,13 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
                 (q->group_bits[k] < 0 || q->group_bits[k] > 63 ||
                  q->group_types[k] == BK_DOUBLE)) {
                 set_err("bad group_bits (1..63; DOUBLE keys need bits==0)");
+                return nullptr;
+            }
+    if (q)
+        for (int32_t k = 0; k < q->n_group; k++)
+            if (q->group_fns[k] && q->group_types[k] != BK_INT64 &&
+                q->group_types[k] != BK_DATETIME) {
+                set_err("group_fns need an int64/DATETIME key column");
                 return nullptr;
             }
     if (ensure_device() != 0) return nullptr;

@@ -64,6 +64,7 @@ class BkQuerySpec(C.Structure):
                 ("group_types", C.c_int32 * BK_MAX_GROUP),
                 ("group_bits", C.c_int32 * BK_MAX_GROUP),
                 ("group_base", C.c_int64 * BK_MAX_GROUP),
+                ("group_fns", C.c_int32 * BK_MAX_GROUP),
                 ("aggs", BkAggSpec * BK_MAX_AGGS),
                 ("agg_in_types", C.c_int32 * BK_MAX_AGGS)]
 
@@ -119,6 +120,9 @@ class QueryPlan:
                 cj.lit_i = int(lit)
         q.n_group = len(self.group)
         for i, col in enumerate(self.group):
+            if isinstance(col, tuple):   # ("year", col): GROUP BY fn(col)
+                q.group_fns[i] = _FNS[col[0]]
+                col = col[1]
             q.group_cols[i] = col
             q.group_types[i] = self.col_types[col]
             if i < len(self.group_bits):

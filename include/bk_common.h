@@ -242,6 +242,11 @@ typedef struct BkQuerySpec {
      * DOUBLE keys require bits == 0. */
     int32_t    group_bits[BK_MAX_GROUP];
     int64_t    group_base[BK_MAX_GROUP];
+    /* optional BkScalarFn per group key (GROUP BY year(c) etc. — the
+     * reference evaluates fn exprs inside encode_exprs_key,
+     * exec_node.cpp:555-571). Only int64/DATETIME keys; the key's output
+     * type is INT64. */
+    int32_t    group_fns[BK_MAX_GROUP];
     BkAggSpec  aggs[BK_MAX_AGGS];
     int32_t    agg_in_types[BK_MAX_AGGS];  /* BkType of each agg input col */
 } BkQuerySpec;

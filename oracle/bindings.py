@@ -56,6 +56,7 @@ class BkQuerySpec(C.Structure):
                 ("group_types", C.c_int32 * BK_MAX_GROUP),
                 ("group_bits", C.c_int32 * BK_MAX_GROUP),
                 ("group_base", C.c_int64 * BK_MAX_GROUP),
+                ("group_fns", C.c_int32 * BK_MAX_GROUP),
                 ("aggs", BkAggSpec * BK_MAX_AGGS),
                 ("agg_in_types", C.c_int32 * BK_MAX_AGGS)]
 
@@ -106,6 +107,9 @@ def make_query(conjuncts=(), group=(), aggs=(), col_types=None,
             cj.lit_d = 0.0
     q.n_group = len(group)
     for i, col in enumerate(group):
+        if isinstance(col, tuple):   # (fn_id, col): GROUP BY fn(col)
+            q.group_fns[i] = col[0]
+            col = col[1]
         q.group_cols[i] = col
         q.group_types[i] = col_types[col]
         if i < len(group_bits):

@@ -349,7 +349,8 @@ static inline void orc_agg_merge(OrcAggState* dst, const OrcAggState* src,
 
 /* encode one group-column value to its order-preserving u64 (same internal
  * key the GPU uses). */
-static inline uint64_t orc_enc_group(const OrcCol* c, int64_t r) {
+static inline uint64_t orc_enc_group(const OrcCol* c, int64_t r, int32_t fn) {
+    if (fn) return bk_enc_i64(bk_scalar_fn(fn, cell_i64(c, r)));
     switch (c->type) {
         case BK_INT64:
         case BK_DATETIME: return bk_enc_i64(((int64_t*)c->data)[r]);
@@ -385,7 +386,7 @@ static void* orc_agg_worker(void* arg) {
         for (int k = 0; k < ng; k++) {
             const OrcCol* c = &cols[q->group_cols[k]];
             if (!cell_is_valid(c, r)) flag |= (uint8_t)(0x01u << (7 - k)); /* exec_node.cpp:561 */
-            else e[k] = orc_enc_group(c, r);
+            else e[k] = orc_enc_group(c, r, q->group_fns[k]);
         }
         int created;
         if (t->map.n * 10 >= t->map.cap * 6) orc_map_grow(&t->map, ng);

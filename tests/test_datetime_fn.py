@@ -85,3 +85,24 @@ def eng():
 def orc():
     from oracle import Oracle
     return Oracle()
+
+
+@pytest.mark.gpu
+def test_gpu_group_by_datetime_fn(eng, orc):
+    """GROUP BY year(c), month(c) — fn exprs as group keys
+    (encode_exprs_key evaluates fn calls, exec_node.cpp:555-571)."""
+    from tests.test_gpu_agg import run_both, assert_parity
+    specs = [(TYPE_DATETIME, D_DT, 0, 0, 120_000),
+             (TYPE_INT64, TYPE_INT64 and 0, 0, 1000, 0),
+             (TYPE_DOUBLE, 3, 0, 0, 0)]
+    aggs = [("count_star", -1), ("sum", 1), ("avg", 2)]
+    got, exp = run_both(eng, orc, specs, 150_000,
+                        [(1, "<", 900)], [("year", 0), ("month", 0)], aggs,
+                        expected_groups=1 << 10)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
+    assert 80 <= got["ngroups"] <= 86   # 7 years x 12 months (+null key rows)
+    # decoded keys are the extraction outputs
+    years = np.array([orc.lib.orc_decode_i64(__import__("ctypes").c_uint64(int(e)))
+                      for e in got["enc"][:, 0]])
+    nn = (got["flags"] & 0x80) == 0
+    assert set(years[nn]) <= set(range(2019, 2026))

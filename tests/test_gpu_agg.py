@@ -79,7 +79,9 @@ def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
                              not isinstance(lit, (list, tuple))) or \
             isinstance(lit, float) else TYPE_INT64
         oconj.append((col, ops[op], ct, lit, fn))
-    q = make_query(oconj, group, [(aggmap[a], c) for a, c in aggs], col_types,
+    from baikaldb_amd.plan import _FNS
+    ogroup = [(_FNS[g[0]], g[1]) if isinstance(g, tuple) else g for g in group]
+    q = make_query(oconj, ogroup, [(aggmap[a], c) for a, c in aggs], col_types,
                    group_bits=group_bits, group_base=group_base)
     exp = orc.filter_agg(cols, valids, col_types, q, nthreads=nthreads,
                          dict_seed=seed)
