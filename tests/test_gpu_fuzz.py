@@ -58,8 +58,11 @@ def random_case(rng):
         else:
             conjuncts.append((c, rng.choice(["<", ">", "=", "!=", ">=", "<="]),
                               rng.randint(-100, 3000)))
-    # group keys
+    # group keys (DATETIME columns may group through an extraction fn:
+    # GROUP BY year(c) etc.)
     group = rng.sample(range(ncols), rng.randint(0, min(2, ncols)))
+    group = [(rng.choice(_DT_FNS), c) if specs[c][0] == TYPE_DATETIME
+             and rng.random() < 0.7 else c for c in group]
     # aggs
     aggs = [("count_star", -1)]
     for _ in range(rng.randint(0, 4)):
