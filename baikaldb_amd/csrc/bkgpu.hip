@@ -246,7 +246,9 @@ __device__ __forceinline__ KeyPack pack_group_keys(const DevCols& cols,
                                                    int64_t r) {
     KeyPack kp{0, 0, 0};
     int shift = 0, word = 0;
-    for (int32_t k = 0; k < q.n_group; k++) {
+    #pragma unroll
+    for (int32_t k = 0; k < BK_MAX_GROUP; k++) {
+        if (k >= q.n_group) break;
         int bits = q.group_bits[k] ? q.group_bits[k] : 64;
         if (shift + bits > 64) { word++; shift = 0; }
         const DevCol& c = cols.c[q.group_cols[k]];
@@ -282,8 +284,9 @@ __device__ __forceinline__ void atomic_add_f64_lds(uint64_t* addr, double v) {
 template <bool LDS>
 __device__ __forceinline__ void agg_update_slot(uint64_t* st, const BkQuerySpec& q,
                                                 const DevCols& cols, int64_t r) {
-    #pragma unroll 4
-    for (int32_t a = 0; a < q.n_aggs; a++) {
+    #pragma unroll
+    for (int32_t a = 0; a < BK_MAX_AGGS; a++) {
+        if (a >= q.n_aggs) break;
         uint64_t* val = st + SLOT_HDR + 2 * a;
         uint64_t* cnt = val + 1;
         const BkAggSpec& as = q.aggs[a];
@@ -343,7 +346,9 @@ __device__ __forceinline__ void agg_update_slot(uint64_t* st, const BkQuerySpec&
 template <bool LDS>
 __device__ __forceinline__ void agg_merge_slot(uint64_t* dst, const uint64_t* src,
                                                const BkQuerySpec& q) {
-    for (int32_t a = 0; a < q.n_aggs; a++) {
+    #pragma unroll
+    for (int32_t a = 0; a < BK_MAX_AGGS; a++) {
+        if (a >= q.n_aggs) break;
         uint64_t* val = dst + SLOT_HDR + 2 * a;
         uint64_t* cnt = val + 1;
         uint64_t sv = src[2 * a];
@@ -560,22 +565,29 @@ k_filter_agg_group(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_e
 }
 
 /* no-GROUP-BY aggregate: register accumulation, one wave-reduce, one merge
- * per wave into the single pre-initialized slot 0 (config-1 COUNT(*) path). */
+ * per wave into the single pre-initialized slot 0 (config-1 COUNT(*) path).
+ * Templated on the agg-count bound NA (4 or 8, host-picked) so every
+ * accumulator index is STATIC: a dynamic acc_v[a] would push all three
+ * arrays into per-lane scratch (144 B/lane measured before this). */
+template <int NA>
 __global__ void __launch_bounds__(256)
 k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
                     uint64_t* gtable, uint64_t* rows_passed) {
     int64_t my_passed = 0;
-    uint64_t acc_v[BK_MAX_AGGS];   /* wrapping-int or cnt accum; doubles separate */
-    uint64_t acc_c[BK_MAX_AGGS];
-    double   acc_d[BK_MAX_AGGS];
-    for (int a = 0; a < q.n_aggs; a++) { acc_v[a] = 0; acc_c[a] = 0; acc_d[a] = 0.0; }
+    uint64_t acc_v[NA];   /* wrapping-int or cnt accum; doubles separate */
+    uint64_t acc_c[NA];
+    double   acc_d[NA];
+    #pragma unroll
+    for (int a = 0; a < NA; a++) { acc_v[a] = 0; acc_c[a] = 0; acc_d[a] = 0.0; }
 
     int64_t gstride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          r < row_end; r += gstride) {
         if (!row_passes(cols, q, r)) continue;
         my_passed++;
-        for (int32_t a = 0; a < q.n_aggs; a++) {
+        #pragma unroll
+        for (int32_t a = 0; a < NA; a++) {
+            if (a >= q.n_aggs) break;
             const BkAggSpec& as = q.aggs[a];
             switch (as.agg_type) {
                 case BK_AGG_COUNT_STAR: acc_v[a]++; break;
@@ -619,7 +631,9 @@ k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_
     }
     /* wave reduce + one atomic merge per wave into slot 0 */
     for (int off = 32; off > 0; off >>= 1) my_passed += __shfl_down(my_passed, off, 64);
-    for (int32_t a = 0; a < q.n_aggs; a++) {
+    #pragma unroll
+    for (int32_t a = 0; a < NA; a++) {
+        if (a >= q.n_aggs) break;
         int at = q.aggs[a].agg_type;
         for (int off = 32; off > 0; off >>= 1) {
             uint64_t ov = __shfl_down(acc_v[a], off, 64);
@@ -632,7 +646,9 @@ k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_
     if ((threadIdx.x & 63) == 0) {
         atomicAdd((unsigned long long*)rows_passed, (unsigned long long)my_passed);
         uint64_t* s = gtable;  /* slot 0, pre-initialized state=2 flag=0 */
-        for (int32_t a = 0; a < q.n_aggs; a++) {
+        #pragma unroll
+        for (int32_t a = 0; a < NA; a++) {
+            if (a >= q.n_aggs) break;
             uint64_t* val = s + SLOT_HDR + 2 * a;
             uint64_t* cnt = val + 1;
             int at = q.aggs[a].agg_type;
@@ -2088,9 +2104,15 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     size_t agg_lds_bytes = ((size_t)agg_lds_slots * stride + 1) * 8;
 
     uint32_t hot_lds_cap = 50 * 1024;
+    if (const char* e = getenv("BK_HOT_LDS_KB")) hot_lds_cap = (uint32_t)atoi(e) * 1024;
     uint32_t hot_slots = 512;
+    if (const char* e = getenv("BK_HOT_SLOTS")) hot_slots = (uint32_t)atoi(e);
     while ((size_t)hot_slots * stride * 8 > hot_lds_cap) hot_slots >>= 1;
     uint32_t hot_cap = hot_slots / 2u, hot_probe = 4, hot_min = 4097;
+    if (const char* e = getenv("BK_HOT_CAP")) hot_cap = (uint32_t)atoi(e);
+    if (hot_cap > hot_slots - hot_slots / 8u) hot_cap = hot_slots - hot_slots / 8u;
+    if (const char* e = getenv("BK_HOT_PROBE")) hot_probe = (uint32_t)atoi(e);
+    if (const char* e = getenv("BK_HOT_MIN")) hot_min = (uint32_t)atoi(e);
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     size_t sc_lds = (size_t)P * 8;
 
@@ -2266,7 +2288,9 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
                 uint64_t one = 1;
                 HIP_CHECK_NULL(hipMemcpy(o->ctrs, &one, 8, hipMemcpyHostToDevice));
                 tm.record();
-                hipLaunchKernelGGL(k_filter_agg_scalar, dim3(blocks), dim3(threads),
+                auto scal_fn = q->n_aggs <= 4 ? k_filter_agg_scalar<4>
+                                              : k_filter_agg_scalar<8>;
+                hipLaunchKernelGGL(scal_fn, dim3(blocks), dim3(threads),
                                    0, 0, dc, *q, row_begin, row_end, o->table,
                                    o->ctrs + 1);
             } else {
