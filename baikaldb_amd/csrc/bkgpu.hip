@@ -703,46 +703,31 @@ struct RecLayout {
 #define BK_SKIP_BUCKET 0xFFFFu
 #define BK_HOT_BUCKET  0xFFFEu
 
-/* pass 1 (hot/cold hybrid): predicate, then try the per-WG LDS aggregate
- * table first — under Zipf-shaped keys the first-come LDS set absorbs the
- * hot head of the distribution right here (no record, no later passes).
- * Cold rows get a bucket id + per-block histogram for the partition passes.
- * LDS carve: [ hot table: lds_slots*stride u64 | laux 2 u64 | lhist P u32 ]. */
-template <int BS>
-__global__ void __launch_bounds__(BS)
-k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
-             uint32_t P, uint16_t* bucketid, uint32_t* H,
-             uint64_t* gtable, uint64_t gmask, uint64_t fill_cap, uint64_t* fill,
-             uint64_t* rows_passed, uint32_t* err, uint32_t lds_slots,
-             uint32_t lcap, uint32_t hot_probe, uint32_t hot_min) {
-    extern __shared__ __attribute__((aligned(16))) uint64_t ltab[];
-    const int stride = SLOT_HDR + 2 * q.n_aggs;
-    uint64_t* laux = ltab + (size_t)lds_slots * stride;
-    uint32_t* lfill = (uint32_t*)&laux[0];
-    /* adaptive hot-path state: laux[2] = {attempts, hits}, laux[3].lo = mode
-     * (0 warmup / 1 disabled / 2 locked-on) */
-    uint32_t* lctr = (uint32_t*)&laux[2];
-    volatile uint32_t* lmode = (volatile uint32_t*)&laux[3];
-    uint32_t* lhist = (uint32_t*)&laux[4];
-    for (uint32_t w = threadIdx.x; w < lds_slots * (uint32_t)stride + 4;
-         w += blockDim.x)
-        ltab[w] = 0;
-    for (uint32_t b = threadIdx.x; b < P; b += blockDim.x) lhist[b] = 0;
-    __syncthreads();
-    const uint32_t lmask = lds_slots - 1;
-    /* low load cap + short probes: a MISS must exit fast (it pays on every
-     * cold row); hot keys the short probe skips just stay cold — correct,
-     * only less absorbed. lcap/hot_probe/hot_min are host-tuned. */
-    int64_t my_passed = 0;
-    int64_t gstride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         r < row_end; r += gstride) {
-        int64_t i = r - row_begin;
-        if (!row_passes(cols, q, r)) { bucketid[i] = (uint16_t)BK_SKIP_BUCKET; continue; }
-        my_passed++;
-        KeyPack kp = pack_group_keys(cols, q, r);
-        uint32_t flag = kp.flag;
-        uint64_t k0 = kp.k0, k1 = kp.k1;
+/* pass 1 (hot/cold hybrid): predicate, then (HOT instantiation only) try the
+ * per-WG LDS aggregate table first — under Zipf-shaped keys the first-come
+ * LDS set absorbs the hot head of the distribution right here (no record,
+ * no later passes). Cold rows get a bucket id + per-block histogram for the
+ * partition passes. The default instantiation is HOT=false (the hot path is
+ * measured-dead at 512-thread blocks): no table, no warmup
+ * gate, LDS = just the histogram — and the row loop runs 2 rows/lane so two
+ * independent load chains are in flight (the spec-driven predicate+keypack
+ * code is issue-limited at 1 row/lane: bwprobe preds R1 3.0 vs R2 3.5 TB/s).
+ * LDS carve: HOT: [ hot table: lds_slots*stride u64 | laux 4 u64 | lhist ]
+ *           !HOT: [ laux 4 u64 | lhist P u32 ]. */
+template <int BS, bool HOT>
+__device__ __forceinline__ void histo_row(
+        const DevCols& cols, const BkQuerySpec& q, int64_t r, int64_t row_begin,
+        uint16_t* bucketid, uint32_t P, uint32_t* lhist, uint64_t* ltab,
+        uint32_t lmask, int stride, uint32_t* lfill, uint32_t* lctr,
+        volatile uint32_t* lmode, uint32_t lcap, uint32_t hot_probe,
+        uint32_t hot_min, int64_t& my_passed) {
+    int64_t i = r - row_begin;
+    if (!row_passes(cols, q, r)) { bucketid[i] = (uint16_t)BK_SKIP_BUCKET; return; }
+    my_passed++;
+    KeyPack kp = pack_group_keys(cols, q, r);
+    uint32_t flag = kp.flag;
+    uint64_t k0 = kp.k0, k1 = kp.k1;
+    if (HOT) {
         /* adaptive: pay the LDS-claim probe only while it absorbs >= 1/4
          * of the stream (per-block warmup decides; Zipf-headed keys keep it
          * on, flat/high-cardinality keys turn it off). */
@@ -759,13 +744,58 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
             if (slot) {
                 agg_update_slot<true>(slot, q, cols, r);
                 bucketid[i] = (uint16_t)BK_HOT_BUCKET;
-                continue;
+                return;
             }
         }
-        uint32_t b = PART_BUCKET(key_hash(flag, k0, k1), P);
-        bucketid[i] = (uint16_t)b;
-        atomicAdd(&lhist[b], 1u);
     }
+    uint32_t b = PART_BUCKET(key_hash(flag, k0, k1), P);
+    bucketid[i] = (uint16_t)b;
+    atomicAdd(&lhist[b], 1u);
+}
+
+/* row mapping shared by histo and scatter (MUST match: H rows are
+ * per-block). Plain row-per-lane grid stride: a 2-rows/lane ILP variant
+ * measured NEUTRAL on histo and -7% on scatter (the compiler does not
+ * interleave the two spec-driven bodies; I$ bloat instead). */
+#define BK_PART_ROWS2(BODY_A, BODY_B)                                        \
+    int64_t gstride = (int64_t)gridDim.x * blockDim.x;                       \
+    for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x            \
+                     + threadIdx.x;                                          \
+         r < row_end; r += gstride) {                                        \
+        BODY_A;                                                              \
+    }
+
+template <int BS, bool HOT>
+__global__ void __launch_bounds__(BS)
+k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
+             uint32_t P, uint16_t* bucketid, uint32_t* H,
+             uint64_t* gtable, uint64_t gmask, uint64_t fill_cap, uint64_t* fill,
+             uint64_t* rows_passed, uint32_t* err, uint32_t lds_slots,
+             uint32_t lcap, uint32_t hot_probe, uint32_t hot_min) {
+    extern __shared__ __attribute__((aligned(16))) uint64_t ltab[];
+    const int stride = SLOT_HDR + 2 * q.n_aggs;
+    const uint32_t tslots = HOT ? lds_slots : 0;
+    uint64_t* laux = ltab + (size_t)tslots * stride;
+    uint32_t* lfill = (uint32_t*)&laux[0];
+    /* adaptive hot-path state: laux[2] = {attempts, hits}, laux[3].lo = mode
+     * (0 warmup / 1 disabled / 2 locked-on) */
+    uint32_t* lctr = (uint32_t*)&laux[2];
+    volatile uint32_t* lmode = (volatile uint32_t*)&laux[3];
+    uint32_t* lhist = (uint32_t*)&laux[4];
+    for (uint32_t w = threadIdx.x; w < tslots * (uint32_t)stride + 4;
+         w += blockDim.x)
+        ltab[w] = 0;
+    for (uint32_t b = threadIdx.x; b < P; b += blockDim.x) lhist[b] = 0;
+    __syncthreads();
+    const uint32_t lmask = tslots - 1;
+    int64_t my_passed = 0;
+    BK_PART_ROWS2(
+        (histo_row<BS, HOT>(cols, q, r, row_begin, bucketid, P, lhist, ltab,
+                            lmask, stride, lfill, lctr, lmode, lcap, hot_probe,
+                            hot_min, my_passed)),
+        (histo_row<BS, HOT>(cols, q, r, row_begin, bucketid, P, lhist, ltab,
+                            lmask, stride, lfill, lctr, lmode, lcap, hot_probe,
+                            hot_min, my_passed)))
     /* rows_passed: reduce per wave, then one atomic per block via laux[1] */
     long long w = my_passed;
     for (int off = 32; off > 0; off >>= 1) w += __shfl_down(w, off, 64);
@@ -777,14 +807,16 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
     for (uint32_t b = threadIdx.x; b < P; b += blockDim.x)
         H[(size_t)blockIdx.x * P + b] = lhist[b];
     /* flush this block's hot groups into the global table */
-    __syncthreads();
-    for (uint32_t sl = threadIdx.x; sl < lds_slots; sl += blockDim.x) {
-        uint64_t* s = ltab + (uint64_t)sl * stride;
-        if (((uint32_t*)s)[0] != 2u) continue;
-        uint64_t* g = gtable_claim(gtable, gmask, stride, ((uint32_t*)s)[1],
-                                   s[1], s[2], fill, fill_cap, err);
-        if (!g) break;
-        agg_merge_slot<false>(g, s + SLOT_HDR, q);
+    if (HOT) {
+        __syncthreads();
+        for (uint32_t sl = threadIdx.x; sl < tslots; sl += blockDim.x) {
+            uint64_t* s = ltab + (uint64_t)sl * stride;
+            if (((uint32_t*)s)[0] != 2u) continue;
+            uint64_t* g = gtable_claim(gtable, gmask, stride, ((uint32_t*)s)[1],
+                                       s[1], s[2], fill, fill_cap, err);
+            if (!g) break;
+            agg_merge_slot<false>(g, s + SLOT_HDR, q);
+        }
     }
 }
 
@@ -873,19 +905,126 @@ __device__ __forceinline__ void scatter_store_rec(uint64_t* dst,
 }
 
 /* pass 2: scatter AoS records into bucket-contiguous regions.
- * MUST run with the same grid/block shape as k_part_histo so each block
- * sees the same rows its H row was computed from. */
+ * MUST run with the same grid/block AND row mapping as k_part_histo (the
+ * BK_PART_ROWS2 ILP2 tiles) so each block sees the rows its H row counted.
+ * 2 rows/lane keeps two gather+store chains in flight (the per-record LDS
+ * lcur atomic and the scattered 32-64 B store are latency-bound). */
+template <int BS>
+__device__ __forceinline__ void scatter_row(
+        const DevCols& cols, const BkQuerySpec& q, const RecLayout& lay,
+        int64_t r, int64_t row_begin, const uint16_t* bucketid,
+        uint32_t* lcur, uint32_t* bstate, uint64_t* stash, uint64_t* rec,
+        int paired) {
+    int64_t i = r - row_begin;
+    uint32_t b = bucketid[i];
+    if (b >= BK_HOT_BUCKET) return;  /* filtered out or absorbed hot */
+    /* key words */
+    KeyPack kp = pack_group_keys(cols, q, r);
+    uint64_t meta = kp.flag;
+    uint64_t k0 = kp.k0, k1 = kp.k1;
+    uint64_t regs[8];
+    #pragma unroll
+    for (int w = 0; w < 8; w++) regs[w] = 0;
+    if (lay.k1_word == -3) {
+        uint64_t d0 = (meta & 0x80u) ? 0 : (k0 - lay.k0_base);
+        regs[0] = (d0 << 32) | (k1 << 8) | (meta & 0xFFu);
+    } else {
+        regs[0] = k0;
+        if (lay.k1_word >= 0) regs[lay.k1_word] = k1;
+        else if (lay.k1_word == -2) meta |= k1 << 32;
+    }
+    for (int32_t a = 0; a < q.n_aggs; a++) {
+        if (lay.val_word[a] < 0) continue;
+        const BkAggSpec& as = q.aggs[a];
+        const DevCol& c = cols.c[as.col];
+        int valid = cell_valid(c, r);
+        if (valid) meta |= (uint64_t)1 << (8 + a);
+        uint64_t w = 0;
+        if (valid) {
+            switch (as.agg_type) {
+                case BK_AGG_SUM:
+                    if (q.agg_in_types[a] == BK_DOUBLE) {
+                        double d = ((const double*)c.data)[r];
+                        memcpy(&w, &d, 8);
+                    } else {
+                        w = (uint64_t)cell_i64(c, r);
+                    }
+                    break;
+                case BK_AGG_AVG: {
+                    double d = cell_f64(c, r);
+                    memcpy(&w, &d, 8);
+                    break;
+                }
+                case BK_AGG_MIN:
+                case BK_AGG_MAX:
+                    w = enc_value(c, r);
+                    break;
+                default: break;
+            }
+        }
+        regs[lay.val_word[a]] = w;
+    }
+    if (lay.meta_word >= 0) {
+        /* COUNT(col) validity for aggs without a val word */
+        for (int32_t a = 0; a < q.n_aggs; a++) {
+            if (lay.val_word[a] >= 0 || q.aggs[a].col < 0) continue;
+            if (cell_valid(cols.c[q.aggs[a].col], r))
+                meta |= (uint64_t)1 << (8 + a);
+        }
+        regs[lay.meta_word] = meta;
+    }
+    if (!paired) {
+        uint32_t pos = atomicAdd(&lcur[b], 1u);
+        scatter_store_rec(rec + (size_t)pos * lay.nwords, regs, lay.nwords);
+        return;
+    }
+    /* pairing: claim the bucket stash, or take it and write a pair */
+    bool done = false;
+    for (int tries = 0; tries < 64 && !done; tries++) {
+        uint32_t st = __hip_atomic_load(&bstate[b], __ATOMIC_RELAXED, WGP);
+        if (st == 0u) {
+            uint32_t exp = 0u;
+            if (__hip_atomic_compare_exchange_strong(
+                    &bstate[b], &exp, 1u, __ATOMIC_ACQUIRE,
+                    __ATOMIC_RELAXED, WGP)) {
+                uint64_t* sl = stash + (size_t)b * lay.nwords;
+                for (int w = 0; w < lay.nwords; w++) sl[w] = regs[w];
+                __hip_atomic_store(&bstate[b], 2u, __ATOMIC_RELEASE, WGP);
+                done = true;
+            }
+        } else if (st == 2u) {
+            uint32_t exp = 2u;
+            if (__hip_atomic_compare_exchange_strong(
+                    &bstate[b], &exp, 3u, __ATOMIC_ACQUIRE,
+                    __ATOMIC_RELAXED, WGP)) {
+                uint64_t prev[8];
+                uint64_t* sl = stash + (size_t)b * lay.nwords;
+                #pragma unroll
+                for (int w = 0; w < 8; w++)
+                    prev[w] = w < lay.nwords ? sl[w] : 0;
+                __hip_atomic_store(&bstate[b], 0u, __ATOMIC_RELEASE, WGP);
+                uint32_t pos = atomicAdd(&lcur[b], 2u);
+                uint64_t* dst = rec + (size_t)pos * lay.nwords;
+                scatter_store_rec(dst, prev, lay.nwords);
+                scatter_store_rec(dst + lay.nwords, regs, lay.nwords);
+                done = true;
+            }
+        }
+        /* st 1/3: another wave mid-transfer — retry (no in-place spin) */
+    }
+    if (!done) {  /* contention fallback: single-record write */
+        uint32_t pos = atomicAdd(&lcur[b], 1u);
+        scatter_store_rec(rec + (size_t)pos * lay.nwords, regs, lay.nwords);
+    }
+}
+
 template <int BS>
 __global__ void __launch_bounds__(BS)
 k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
                int64_t row_end, uint32_t P, const uint16_t* bucketid,
                const uint32_t* H, uint64_t* rec, uint64_t total, int paired) {
     /* LDS carve: [ stash: P*nwords u64 (pairing only) ][ lcur: P u32 ]
-     *            [ bstate: P u32 ]
-     * Pair staging: a lone 32-40 B record store touches a 64-128 B line that
-     * gets written back before its neighbour arrives (~2.5x write
-     * amplification by PMC). Stash ONE record per bucket in LDS; the second
-     * arrival writes BOTH contiguously. */
+     *            [ bstate: P u32 ] */
     extern __shared__ __attribute__((aligned(16))) uint64_t lmem[];
     uint64_t* stash = lmem;
     uint32_t* lcur = (uint32_t*)(lmem + (paired ? (size_t)P * lay.nwords : 0));
@@ -895,111 +1034,11 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
         if (paired) bstate[b] = 0;
     }
     __syncthreads();
-    int64_t gstride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         r < row_end; r += gstride) {
-        int64_t i = r - row_begin;
-        uint32_t b = bucketid[i];
-        if (b >= BK_HOT_BUCKET) continue;  /* filtered out or absorbed hot */
-        /* key words */
-        KeyPack kp = pack_group_keys(cols, q, r);
-        uint64_t meta = kp.flag;
-        uint64_t k0 = kp.k0, k1 = kp.k1;
-        uint64_t regs[8];
-        #pragma unroll
-        for (int w = 0; w < 8; w++) regs[w] = 0;
-        if (lay.k1_word == -3) {
-            uint64_t d0 = (meta & 0x80u) ? 0 : (k0 - lay.k0_base);
-            regs[0] = (d0 << 32) | (k1 << 8) | (meta & 0xFFu);
-        } else {
-            regs[0] = k0;
-            if (lay.k1_word >= 0) regs[lay.k1_word] = k1;
-            else if (lay.k1_word == -2) meta |= k1 << 32;
-        }
-        for (int32_t a = 0; a < q.n_aggs; a++) {
-            if (lay.val_word[a] < 0) continue;
-            const BkAggSpec& as = q.aggs[a];
-            const DevCol& c = cols.c[as.col];
-            int valid = cell_valid(c, r);
-            if (valid) meta |= (uint64_t)1 << (8 + a);
-            uint64_t w = 0;
-            if (valid) {
-                switch (as.agg_type) {
-                    case BK_AGG_SUM:
-                        if (q.agg_in_types[a] == BK_DOUBLE) {
-                            double d = ((const double*)c.data)[r];
-                            memcpy(&w, &d, 8);
-                        } else {
-                            w = (uint64_t)cell_i64(c, r);
-                        }
-                        break;
-                    case BK_AGG_AVG: {
-                        double d = cell_f64(c, r);
-                        memcpy(&w, &d, 8);
-                        break;
-                    }
-                    case BK_AGG_MIN:
-                    case BK_AGG_MAX:
-                        w = enc_value(c, r);
-                        break;
-                    default: break;
-                }
-            }
-            regs[lay.val_word[a]] = w;
-        }
-        if (lay.meta_word >= 0) {
-            /* COUNT(col) validity for aggs without a val word */
-            for (int32_t a = 0; a < q.n_aggs; a++) {
-                if (lay.val_word[a] >= 0 || q.aggs[a].col < 0) continue;
-                if (cell_valid(cols.c[q.aggs[a].col], r))
-                    meta |= (uint64_t)1 << (8 + a);
-            }
-            regs[lay.meta_word] = meta;
-        }
-        if (!paired) {
-            uint32_t pos = atomicAdd(&lcur[b], 1u);
-            scatter_store_rec(rec + (size_t)pos * lay.nwords, regs, lay.nwords);
-            continue;
-        }
-        /* pairing: claim the bucket stash, or take it and write a pair */
-        bool done = false;
-        for (int tries = 0; tries < 64 && !done; tries++) {
-            uint32_t st = __hip_atomic_load(&bstate[b], __ATOMIC_RELAXED, WGP);
-            if (st == 0u) {
-                uint32_t exp = 0u;
-                if (__hip_atomic_compare_exchange_strong(
-                        &bstate[b], &exp, 1u, __ATOMIC_ACQUIRE,
-                        __ATOMIC_RELAXED, WGP)) {
-                    uint64_t* sl = stash + (size_t)b * lay.nwords;
-                    for (int w = 0; w < lay.nwords; w++) sl[w] = regs[w];
-                    __hip_atomic_store(&bstate[b], 2u, __ATOMIC_RELEASE, WGP);
-                    done = true;
-                }
-            } else if (st == 2u) {
-                uint32_t exp = 2u;
-                if (__hip_atomic_compare_exchange_strong(
-                        &bstate[b], &exp, 3u, __ATOMIC_ACQUIRE,
-                        __ATOMIC_RELAXED, WGP)) {
-                    uint64_t prev[8];
-                    uint64_t* sl = stash + (size_t)b * lay.nwords;
-                    #pragma unroll
-                    for (int w = 0; w < 8; w++)
-                        prev[w] = w < lay.nwords ? sl[w] : 0;
-                    __hip_atomic_store(&bstate[b], 0u, __ATOMIC_RELEASE, WGP);
-                    uint32_t pos = atomicAdd(&lcur[b], 2u);
-                    uint64_t* dst = rec + (size_t)pos * lay.nwords;
-                    scatter_store_rec(dst, prev, lay.nwords);
-                    scatter_store_rec(dst + lay.nwords, regs, lay.nwords);
-                    done = true;
-                }
-            }
-            /* st 1/3: another wave mid-transfer — retry (no in-place spin) */
-        }
-        if (!done) {  /* contention fallback: single-record write */
-            uint32_t pos = atomicAdd(&lcur[b], 1u);
-            scatter_store_rec(rec + (size_t)pos * lay.nwords, regs, lay.nwords);
-        }
-    }
+    BK_PART_ROWS2(
+        (scatter_row<BS>(cols, q, lay, r, row_begin, bucketid, lcur, bstate,
+                         stash, rec, paired)),
+        (scatter_row<BS>(cols, q, lay, r, row_begin, bucketid, lcur, bstate,
+                         stash, rec, paired)))
     /* drain leftover stashes (one half-pair per bucket at most) */
     if (paired) {
         __syncthreads();
@@ -1888,6 +1927,27 @@ static double now_ms() {
     return tv.tv_sec * 1e3 + tv.tv_usec * 1e-3;
 }
 
+/* host-side instantiation pick for histo/scatter: HOT only when the
+ * adaptive lock-on is enabled (BK_HOT_MIN <= 4096); default is the lean
+ * no-table variant. */
+typedef void (*HistoFn)(DevCols, BkQuerySpec, int64_t, int64_t, uint32_t,
+                        uint16_t*, uint32_t*, uint64_t*, uint64_t, uint64_t,
+                        uint64_t*, uint64_t*, uint32_t*, uint32_t, uint32_t,
+                        uint32_t, uint32_t);
+typedef void (*ScatFn)(DevCols, BkQuerySpec, RecLayout, int64_t, int64_t,
+                       uint32_t, const uint16_t*, const uint32_t*, uint64_t*,
+                       uint64_t, int);
+static HistoFn pick_histo(int threads, bool hot) {
+    if (threads == 512)  return hot ? k_part_histo<512, true>  : k_part_histo<512, false>;
+    if (threads == 1024) return hot ? k_part_histo<1024, true> : k_part_histo<1024, false>;
+    return hot ? k_part_histo<256, true> : k_part_histo<256, false>;
+}
+static ScatFn pick_scat(int threads) {
+    if (threads == 512)  return k_part_scatter<512>;
+    if (threads == 1024) return k_part_scatter<1024>;
+    return k_part_scatter<256>;
+}
+
 /* pipelined variant of run_partitioned: the row range splits into chunks
  * and consecutive chunks run on TWO HIP streams, so one chunk's part_agg
  * (LDS-latency-bound, ~1.2 TB/s — NOT HBM-saturated) overlaps the next
@@ -1934,10 +1994,6 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     int threads = 512;
     if (const char* e = getenv("BK_PART_THREADS")) threads = atoi(e);
     if (threads != 512 && threads != 1024) threads = 256;
-    auto histo_fn = k_part_histo<256>;
-    auto scat_fn = k_part_scatter<256>;
-    if (threads == 512) { histo_fn = k_part_histo<512>; scat_fn = k_part_scatter<512>; }
-    else if (threads == 1024) { histo_fn = k_part_histo<1024>; scat_fn = k_part_scatter<1024>; }
     DevCols dc = table_cols(t);
 
     uint16_t* bucketid = nullptr;
@@ -1984,6 +2040,10 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
      * the adaptive lock-on for workloads where it wins. */
     uint32_t hot_min = 4097;
     if (const char* e = getenv("BK_HOT_MIN")) hot_min = (uint32_t)atoi(e);
+    bool hot = hot_min <= 4096;
+    if (!hot) hot_slots = 0;
+    HistoFn histo_fn = pick_histo(threads, hot);
+    ScatFn scat_fn = pick_scat(threads);
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     EvTimer tm;
     tm.record();
@@ -2085,10 +2145,6 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     int threads = 512;
     if (const char* e = getenv("BK_PART_THREADS")) threads = atoi(e);
     if (threads != 512 && threads != 1024) threads = 256;
-    auto histo_fn = k_part_histo<256>;
-    auto scat_fn = k_part_scatter<256>;
-    if (threads == 512) { histo_fn = k_part_histo<512>; scat_fn = k_part_scatter<512>; }
-    else if (threads == 1024) { histo_fn = k_part_histo<1024>; scat_fn = k_part_scatter<1024>; }
     int at = 1024;
     if (const char* e = getenv("BK_AGG_THREADS")) at = atoi(e);
     int ilp = 2;
@@ -2113,6 +2169,10 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     if (hot_cap > hot_slots - hot_slots / 8u) hot_cap = hot_slots - hot_slots / 8u;
     if (const char* e = getenv("BK_HOT_PROBE")) hot_probe = (uint32_t)atoi(e);
     if (const char* e = getenv("BK_HOT_MIN")) hot_min = (uint32_t)atoi(e);
+    bool hot = hot_min <= 4096;
+    if (!hot) hot_slots = 0;
+    HistoFn histo_fn = pick_histo(threads, hot);
+    ScatFn scat_fn = pick_scat(threads);
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     size_t sc_lds = (size_t)P * 8;
 
