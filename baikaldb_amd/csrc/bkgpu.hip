@@ -2787,3 +2787,4 @@ extern "C" int bkgpu_gather(BkgTable* t, int col, const int64_t* rowids_host,
 
 #include "bksort.inc"
 #include "bkwin.inc"
+#include "bkdedup.inc"

@@ -68,6 +68,9 @@ def _load():
     lib.bkgpu_agg_rollup.restype = C.c_void_p
     lib.bkgpu_agg_rollup.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
                                      C.POINTER(C.c_int32), C.c_int64]
+    lib.bkgpu_filter_agg_sorted.restype = C.c_void_p
+    lib.bkgpu_filter_agg_sorted.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
+                                            C.c_int64, C.c_int64]
     lib.bkgpu_agg_fetch.restype = C.c_int64
     lib.bkgpu_agg_fetch.argtypes = [C.c_void_p, C.c_int, C.c_int64,
                                     C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
@@ -226,15 +229,54 @@ class GpuEngine:
             raise RuntimeError(f"filter_agg: {self.lib.bkgpu_last_error().decode()}")
         return AggResult(self, h, plan)
 
+    def filter_agg_sorted(self, table, plan: QueryPlan, row_begin=0,
+                          row_end=None):
+        """GROUP BY via sort-based dedup (bkgpu_filter_agg_sorted): for
+        group counts approaching the row count, where the hash table
+        degrades. Raises if the plan's keys don't pack into <=56 declared
+        group_bits."""
+        if row_end is None:
+            row_end = table.nrows
+        q = plan.to_spec()
+        h = self.lib.bkgpu_filter_agg_sorted(table.handle, C.byref(q),
+                                             row_begin, row_end)
+        if not h:
+            raise RuntimeError(
+                f"filter_agg_sorted: {self.lib.bkgpu_last_error().decode()}")
+        return AggResult(self, h, plan)
+
+    # above this expected level-1 cardinality, COUNT/SUM(DISTINCT) uses the
+    # sort-dedup level 1 when the keys qualify (BK_DEDUP_SORT=0/1 overrides)
+    DEDUP_SORT_MIN_L1 = 1 << 22
+
+    def _l1_sorted_eligible(self, l1_plan, expected_l1_groups):
+        import os
+        env = os.environ.get("BK_DEDUP_SORT")
+        if env is not None:
+            return env != "0"
+        if expected_l1_groups < self.DEDUP_SORT_MIN_L1:
+            return False
+        bits = [l1_plan.group_bits[i] if i < len(l1_plan.group_bits) and
+                l1_plan.group_bits[i] else 64 for i in range(len(l1_plan.group))]
+        return sum(bits) <= 56
+
     def filter_agg_distinct(self, table, plan: QueryPlan, row_begin=0,
                             row_end=None, expected_l1_groups=1 << 18,
                             expected_groups=1 << 14):
         """COUNT/SUM(DISTINCT d): level-1 filter_agg grouped by
         (user keys + d), then bkgpu_agg_rollup (the reference's multi-
-        distinct planner rewrite, agg_node.cpp:247-258)."""
+        distinct planner rewrite, agg_node.cpp:247-258). High-cardinality
+        level 1 switches to the sort-dedup path (same results)."""
         l1_plan, q2, src_idx = plan.split_distinct()
-        l1 = self.filter_agg(table, l1_plan, row_begin, row_end,
-                             expected_groups=expected_l1_groups)
+        if self._l1_sorted_eligible(l1_plan, expected_l1_groups):
+            try:
+                l1 = self.filter_agg_sorted(table, l1_plan, row_begin, row_end)
+            except RuntimeError:
+                l1 = self.filter_agg(table, l1_plan, row_begin, row_end,
+                                     expected_groups=expected_l1_groups)
+        else:
+            l1 = self.filter_agg(table, l1_plan, row_begin, row_end,
+                                 expected_groups=expected_l1_groups)
         try:
             h = self.lib.bkgpu_agg_rollup(l1.handle, C.byref(q2), src_idx,
                                           expected_groups)

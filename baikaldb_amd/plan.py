@@ -79,7 +79,8 @@ class QueryPlan:
     """
 
     def __init__(self, col_types, conjuncts=(), group=(), aggs=(),
-                 group_bits=(), group_base=()):
+                 group_bits=(), group_base=(), distinct_bits=0,
+                 distinct_base=0):
         self.col_types = list(col_types)
         self.conjuncts = list(conjuncts)
         self.group = list(group)
@@ -88,6 +89,11 @@ class QueryPlan:
         # base value, declared by the caller (bk_common.h BkQuerySpec)
         self.group_bits = list(group_bits)
         self.group_base = list(group_base)
+        # declared width/base of the DISTINCT column's encoding: lets the
+        # level-1 (keys + d) plan pack into one word so the sort-dedup
+        # level 1 qualifies (engine.filter_agg_sorted)
+        self.distinct_bits = distinct_bits
+        self.distinct_base = distinct_base
 
     def to_spec(self):
         q = BkQuerySpec()
@@ -157,9 +163,16 @@ class QueryPlan:
         dcol = dcols.pop()
         plain = [(n, c) for n, c in self.aggs
                  if (_AGGS[n] if isinstance(n, str) else n) < AGG_COUNT_DISTINCT]
+        nk = len(self.group)
+        gb = [self.group_bits[i] if i < len(self.group_bits) else 0
+              for i in range(nk)]
+        gv = [self.group_base[i] if i < len(self.group_base) else 0
+              for i in range(nk)]
         l1 = QueryPlan(self.col_types, conjuncts=self.conjuncts,
                        group=self.group + [dcol],
-                       aggs=plain or [("count_star", -1)])
+                       aggs=plain or [("count_star", -1)],
+                       group_bits=gb + [self.distinct_bits],
+                       group_base=gv + [self.distinct_base])
         q2 = BkQuerySpec()
         q2.n_conjuncts = 0
         q2.n_group = len(self.group)

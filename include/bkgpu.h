@@ -92,6 +92,16 @@ int  bkgpu_agg_export(const BkgAggOut* o, void* dst, int64_t cap);
  * AggFnCall::merge semantics (src/expr/agg_fn_call.cpp:781-830). */
 int  bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups);
 
+/* Level-1 aggregate via SORT-based dedup: radix-sorts the spec-packed group
+ * key per passing row and emits one DENSE table slot per unique key — the
+ * path for high-cardinality DISTINCT (dedup cardinality ~ rows), where the
+ * hash table degrades. Results identical to bkgpu_filter_agg.
+ * Requires every group key packed into <= 56 declared group_bits and a row
+ * range < 2^32; returns NULL (with bkgpu_last_error) otherwise — callers
+ * fall back to bkgpu_filter_agg. */
+BkgAggOut* bkgpu_filter_agg_sorted(BkgTable* t, const BkQuerySpec* q,
+                                   int64_t row_begin, int64_t row_end);
+
 /* ---- DISTINCT rollup (reference planner rewrite, agg_node.cpp:247-258) ----
  * Fold a level-1 aggregate grouped by (user group keys + distinct col) into
  * the level-2 result grouped by the user keys alone. q2: n_group = level-1

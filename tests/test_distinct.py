@@ -330,3 +330,111 @@ def test_gpu_distinct_fuzz(eng, orc, cs):
             raise AssertionError(
                 f"distinct fuzz {cs}/{sub}: specs={specs} conj={conj} "
                 f"group={group} aggs={aggs}: {e}")
+
+
+# ---------------- sort-based dedup level 1 (bkgpu_filter_agg_sorted) -------
+
+@pytest.mark.gpu
+def test_gpu_sorted_dedup_vs_hash_plain_group(eng):
+    """filter_agg_sorted == filter_agg on the same plain GROUP BY."""
+    from baikaldb_amd import QueryPlan
+    specs = [(TYPE_INT64, D_UNI, 0, 4000, 0),
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 50_000)]
+    t = eng.create_table(specs, 500_000)
+    try:
+        eng.generate(t, SEED + 7)
+        plan = QueryPlan(t.col_types, conjuncts=[(1, "<", 1 << 30)],
+                         group=[0], aggs=[("count_star", -1), ("sum", 1),
+                                          ("avg", 2), ("min", 1)],
+                         group_bits=[13], group_base=[0])
+        a = eng.filter_agg_sorted(t, plan)
+        b = eng.filter_agg(t, plan, expected_groups=8192)
+        try:
+            ga, gb = a.fetch(sorted=True), b.fetch(sorted=True)
+        finally:
+            a.free()
+            b.free()
+    finally:
+        t.free()
+    assert ga["ngroups"] == gb["ngroups"]
+    assert ga["rows_passed"] == gb["rows_passed"]
+    assert np.array_equal(ga["enc"], gb["enc"])
+    assert np.array_equal(ga["flags"], gb["flags"])
+    for i in range(2):
+        assert np.array_equal(ga["agg_i"][i], gb["agg_i"][i])
+    assert np.allclose(ga["agg_d"][2], gb["agg_d"][2], rtol=0, atol=1e-9)
+    assert np.array_equal(ga["agg_i"][3], gb["agg_i"][3])
+
+
+@pytest.mark.gpu
+def test_gpu_sorted_dedup_distinct_parity(eng, orc, monkeypatch):
+    """COUNT/SUM(DISTINCT) through the forced sort-dedup level 1 matches the
+    oracle (declared distinct_bits make the level-1 keys pack)."""
+    from baikaldb_amd import QueryPlan
+    monkeypatch.setenv("BK_DEDUP_SORT", "1")
+    specs = [(TYPE_INT64, D_UNI, 0, 16, 0),
+             (TYPE_INT64, D_UNI, 0, 1 << 40, 0),
+             (TYPE_INT64, D_UNI, 0, 1000, 0)]
+    aggs = [("count_star", -1), ("count_distinct", 1), ("sum", 2),
+            ("sum_distinct", 1)]
+    conjuncts = [(2, "<", 900)]
+    n = 400_000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED)
+        plan = QueryPlan(t.col_types, conjuncts=conjuncts, group=[0],
+                         aggs=aggs, group_bits=[6], group_base=[0],
+                         distinct_bits=41, distinct_base=0)
+        res = eng.filter_agg_distinct(t, plan, expected_l1_groups=1 << 16)
+        try:
+            got = res.fetch(sorted=True)
+        finally:
+            res.free()
+    finally:
+        t.free()
+    exp, _ = oracle_distinct(orc, specs, n, conjuncts, [0], aggs)
+    assert_parity(got, exp, _parity_names(aggs), [s[0] for s in specs])
+
+
+@pytest.mark.gpu
+def test_gpu_sorted_dedup_nullable_key(eng):
+    """nullable group key: the null-flag byte rides above the declared bits
+    and must survive the sort round trip."""
+    from baikaldb_amd import QueryPlan
+    specs = [(TYPE_INT64, D_UNI, 0, 500, 200_000),
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 0)]
+    t = eng.create_table(specs, 300_000)
+    try:
+        eng.generate(t, SEED + 11)
+        plan = QueryPlan(t.col_types, group=[0],
+                         aggs=[("count_star", -1), ("max", 1)],
+                         group_bits=[10], group_base=[0])
+        a = eng.filter_agg_sorted(t, plan)
+        b = eng.filter_agg(t, plan, expected_groups=1024)
+        try:
+            ga, gb = a.fetch(sorted=True), b.fetch(sorted=True)
+        finally:
+            a.free()
+            b.free()
+    finally:
+        t.free()
+    assert ga["ngroups"] == gb["ngroups"]
+    assert np.array_equal(ga["enc"], gb["enc"])
+    assert np.array_equal(ga["flags"], gb["flags"])
+    assert np.array_equal(ga["agg_i"][0], gb["agg_i"][0])
+    assert np.array_equal(ga["agg_i"][1], gb["agg_i"][1])
+
+
+def test_sorted_dedup_rejects_wide_keys():
+    """CPU: the C entry refuses keys that don't pack (falls back upstream)."""
+    from baikaldb_amd import GpuEngine, QueryPlan
+    eng = GpuEngine.__new__(GpuEngine)  # bind lib without device init
+    from baikaldb_amd.engine import _load
+    eng.lib = _load()
+    plan = QueryPlan([TYPE_INT64], group=[0], aggs=[("count_star", -1)])
+    q = plan.to_spec()
+    h = eng.lib.bkgpu_filter_agg_sorted(None, C.byref(q), 0, 100)
+    assert not h
+    err = eng.lib.bkgpu_last_error().decode()
+    assert "sorted" in err or "GROUP BY" in err
