@@ -1366,10 +1366,16 @@ __global__ void k_merge_blob(BkQuerySpec q, const uint32_t* flags, const uint64_
 struct SrcIdx { int32_t v[BK_MAX_AGGS]; };  /* level-1 agg index per level-2
                                                agg; -1 = synthesize from d */
 
+/* Level-1 key layout for the rollup: with no declared group_bits the user
+ * key sits raw in k0 and d raw in k1 (l1_bits* == 0). With declared bits
+ * (the sort-dedup path REQUIRES this; the hash path then packs the same
+ * way) both ride in k0: [d - base1 : bits1][key - base0 : bits0]; eb* are
+ * the host-encoded bases, so the raw encodings are recovered here. */
 __global__ void k_rollup(BkQuerySpec q2, int in_naggs, const uint64_t* in_table,
                          uint64_t in_nslots, uint64_t* out, uint64_t omask,
                          uint64_t fill_cap, uint64_t* fill, uint32_t* err,
-                         SrcIdx src) {
+                         SrcIdx src, int l1_bits0, int l1_bits1,
+                         uint64_t l1_eb0, uint64_t l1_eb1) {
     const int in_stride = SLOT_HDR + 2 * in_naggs;
     const int stride = SLOT_HDR + 2 * q2.n_aggs;
     uint64_t gs = (uint64_t)gridDim.x * blockDim.x;
@@ -1384,10 +1390,22 @@ __global__ void k_rollup(BkQuerySpec q2, int in_naggs, const uint64_t* in_table,
         if (q2.n_group == 0) {
             /* no user group keys: level-1 key IS d; out slot 0 pre-claimed */
             e_d = s[1]; d_null = (f & 0x80u) != 0;
+            if (l1_bits0 && !d_null)
+                e_d = (e_d & ((1ull << l1_bits0) - 1ull)) + l1_eb0;
             g = out;
         } else {
-            e_d = s[2]; d_null = (f & 0x40u) != 0;
-            g = gtable_claim(out, omask, stride, f & 0x80u, s[1], 0,
+            uint64_t e0;
+            d_null = (f & 0x40u) != 0;
+            if (l1_bits0 && l1_bits1 && l1_bits0 + l1_bits1 <= 64) {
+                uint64_t k = s[1];
+                e0 = (f & 0x80u) ? 0
+                     : (k & ((1ull << l1_bits0) - 1ull)) + l1_eb0;
+                e_d = d_null ? 0
+                      : ((k >> l1_bits0) & ((1ull << l1_bits1) - 1ull)) + l1_eb1;
+            } else {
+                e0 = s[1]; e_d = s[2];
+            }
+            g = gtable_claim(out, omask, stride, f & 0x80u, e0, 0,
                              fill, fill_cap, err);
             if (!g) return;
         }
@@ -2385,7 +2403,15 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
     }
     o->dirty = true;
     if (debug_timing()) { (void)hipDeviceSynchronize(); t_pipe = now_ms(); }
-    if (agg_compact(o) != 0) { bkgpu_agg_free(o); return nullptr; }
+    /* compaction is LAZY (bkgpu_agg_ngroups / fetch / export run it on
+     * demand): a level-1 distinct aggregate with ~row-count groups never
+     * pays the multi-GB blob export it would never read. rows_passed is
+     * settled from the device counters here (24 B copy). */
+    {
+        uint64_t ctr_host[3];
+        HIP_CHECK_NULL(hipMemcpy(ctr_host, o->ctrs, 24, hipMemcpyDeviceToHost));
+        o->rows_passed = (int64_t)ctr_host[1];
+    }
     if (debug_timing()) {
         fprintf(stderr, "[bkgpu] alloc %.1f ms, pipeline %.1f ms (events %.1f), "
                 "compact %.1f ms\n", t_alloc - t_start, t_pipe - t_alloc,
@@ -2428,8 +2454,37 @@ extern "C" BkgAggOut* bkgpu_agg_rollup(const BkgAggOut* in, const BkQuerySpec* q
         }
         src.v[a] = src_idx[a];
     }
-    /* ensure in->rows_passed / fill counters are settled */
-    if (agg_compact(const_cast<BkgAggOut*>(in)) != 0) return nullptr;
+    /* settle in->rows_passed from the device counters (a full compact of
+     * the level-1 table would export a blob proportional to its group
+     * count — pure waste here; k_rollup reads table slots directly) */
+    {
+        uint64_t ctr_host[3];
+        HIP_CHECK_NULL(hipMemcpy(ctr_host, in->ctrs, 24, hipMemcpyDeviceToHost));
+        const_cast<BkgAggOut*>(in)->rows_passed = (int64_t)ctr_host[1];
+    }
+    /* packed level-1 key layout (declared group_bits): precompute the
+     * per-key widths and ENCODED bases for the kernel's unpack */
+    const BkQuerySpec& l1q = in->q;
+    int l1_bits0 = 0, l1_bits1 = 0;
+    uint64_t l1_eb0 = 0, l1_eb1 = 0;
+    {
+        int k0i = 0, k1i = l1q.n_group - 1;  /* d is the LAST l1 key */
+        if (l1q.n_group == 1) {
+            l1_bits0 = l1q.group_bits[0];
+            l1_eb0 = l1q.group_types[0] == BK_STRING
+                         ? (uint64_t)l1q.group_base[0]
+                         : bk_enc_i64(l1q.group_base[0]);
+        } else if (l1q.n_group == 2) {
+            l1_bits0 = l1q.group_bits[k0i];
+            l1_bits1 = l1q.group_bits[k1i];
+            l1_eb0 = l1q.group_types[k0i] == BK_STRING
+                         ? (uint64_t)l1q.group_base[k0i]
+                         : bk_enc_i64(l1q.group_base[k0i]);
+            l1_eb1 = l1q.group_types[k1i] == BK_STRING
+                         ? (uint64_t)l1q.group_base[k1i]
+                         : bk_enc_i64(l1q.group_base[k1i]);
+        }
+    }
     BkgAggOut* o = new BkgAggOut();
     o->q = *q2;
     int64_t nslots = next_pow2(std::max<int64_t>(expected_groups * 2, 1024));
@@ -2447,7 +2502,8 @@ extern "C" BkgAggOut* bkgpu_agg_rollup(const BkgAggOut* in, const BkQuerySpec* q
         hipLaunchKernelGGL(k_rollup, dim3(1024), dim3(256), 0, 0,
                            *q2, in->q.n_aggs, in->table, in->nslots,
                            o->table, o->nslots - 1, (o->nslots * 7) / 8,
-                           o->ctrs, o->err, src);
+                           o->ctrs, o->err, src, l1_bits0, l1_bits1,
+                           l1_eb0, l1_eb1);
         tm.record();
         hipError_t lerr = hipGetLastError();
         if (lerr != hipSuccess) {
