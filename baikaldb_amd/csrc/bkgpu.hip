@@ -2553,11 +2553,14 @@ extern "C" int64_t bkgpu_agg_rows_passed(const BkgAggOut* o) { return o->rows_pa
 extern "C" double bkgpu_agg_kernel_ms(const BkgAggOut* o) { return o->kernel_ms; }
 
 extern "C" int64_t bkgpu_agg_export_bytes(const BkgAggOut* o) {
+    /* compaction is lazy since the level-1 distinct change: settle it */
+    if (agg_compact(const_cast<BkgAggOut*>(o)) != 0) return -1;
     return (int64_t)blob_bytes_for(o->q.n_aggs, o->ngroups);
 }
 
 extern "C" int bkgpu_agg_export(const BkgAggOut* o, void* dst, int64_t cap) {
-    int64_t need = bkgpu_agg_export_bytes(o);
+    if (agg_compact(const_cast<BkgAggOut*>(o)) != 0) return -1;
+    int64_t need = (int64_t)blob_bytes_for(o->q.n_aggs, o->ngroups);
     if (cap < need) { set_err("export buffer too small"); return -1; }
     /* blob already compacted with capacity == ngroups layout? The blob's
      * internal layout used blob_groups as the stride; re-pack if they differ. */
