@@ -31,6 +31,8 @@ class BkPlanNodeDesc(C.Structure):
                 ("n_aggs", C.c_int32),
                 ("aggs", BkAggSpec * BK_MAX_AGGS),
                 ("expected_groups", C.c_int64),
+                ("distinct_bits", C.c_int32), ("_pad_d", C.c_int32),
+                ("distinct_base", C.c_int64),
                 ("n_order", C.c_int32),
                 ("order", BkOrderSpec * 4),
                 ("n_out_cols", C.c_int32),
@@ -126,7 +128,8 @@ def window_node(part_col, order, fns, out_cols, num_children=1, limit=-1,
 
 
 def agg_node(group, aggs, expected_groups=1 << 16, merge=False, num_children=1,
-             limit=-1, group_bits=(), group_base=()):
+             limit=-1, group_bits=(), group_base=(), distinct_bits=0,
+             distinct_base=0):
     d = BkPlanNodeDesc()
     d.node_type, d.num_children = (MERGE_AGG if merge else AGG), num_children
     d.limit = limit
@@ -142,6 +145,8 @@ def agg_node(group, aggs, expected_groups=1 << 16, merge=False, num_children=1,
         d.aggs[i].agg_type = _AGGS[name] if isinstance(name, str) else name
         d.aggs[i].col = col
     d.expected_groups = expected_groups
+    d.distinct_bits = distinct_bits
+    d.distinct_base = distinct_base
     return d
 
 

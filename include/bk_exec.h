@@ -49,6 +49,13 @@ typedef struct BkPlanNodeDesc {
     int32_t    n_aggs;
     BkAggSpec  aggs[BK_MAX_AGGS];
     int64_t    expected_groups;
+    int32_t    distinct_bits;      /* declared width/base of DISTINCT column
+                                      encodings: lets the level-1 (keys + d)
+                                      pass pack into one word so the sort-
+                                      dedup level 1 qualifies (bkdedup.inc);
+                                      0 = undeclared (hash path only) */
+    int32_t    _pad_d;
+    int64_t    distinct_base;
     /* SORT_NODE payload */
     int32_t     n_order;
     BkOrderSpec order[4];

@@ -371,3 +371,37 @@ def test_three_group_keys_through_exec_surface(eng, orc):
         assert k == uniq[r], (r, k, uniq[r])
         assert vi[r, 3] == cnt[k]
         assert vi[r, 4] == sm[k]
+
+
+def test_count_distinct_sorted_l1_through_exec_surface(eng, orc):
+    """Same distinct query with declared key widths and a large
+    expected_groups: AggNode routes level 1 through the sort-dedup path
+    (bkgpu_filter_agg_sorted) — results must match the hash route."""
+    from baikaldb_amd import exec as bx
+    import numpy as _np
+    t, cols, valids, types = make_table(eng, orc, n=60_000)
+    try:
+        nodes = [bx.agg_node(group=[1], aggs=[("count_star", -1),
+                                              ("count_distinct", 2),
+                                              ("sum", 2)],
+                             group_bits=[11], group_base=[0],
+                             distinct_bits=32, distinct_base=0,
+                             expected_groups=1 << 20),
+                 bx.filter_node(types, [(0, "<", int((1 << 31) * 0.5))]),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
+    idx = _np.nonzero(cols[0] < int((1 << 31) * 0.5))[0]
+    g = cols[1][idx]
+    d = cols[2][idx]
+    vals = _np.unique(g)
+    assert tags.shape[0] == len(vals)
+    for r, gv in enumerate(vals):
+        sel = d[g == gv]
+        assert vi[r, 1] == (g == gv).sum()
+        assert vi[r, 2] == len(_np.unique(sel))
+        assert vi[r, 3] == sel.sum()
