@@ -73,7 +73,63 @@ for cs in range(SORT_CASES):
         print(f"SORT-FAIL {cs}: specs={specs} order={order} "
               f"limit={limit}: {e}", flush=True)
 print(f"sort soak: {SORT_CASES - sfails}/{SORT_CASES} ok", flush=True)
-total_fails = fails + wfails + sfails
+
+# sorted-dedup vs hash path: identical results on random DISTINCT queries
+from baikaldb_amd import QueryPlan
+DD_CASES = AGG_CASES // 2
+dfails = 0
+for cs in range(DD_CASES):
+    rng = random.Random(130_000 + cs)
+    kbits = rng.choice([4, 8, 12])
+    dbits = rng.choice([16, 30, 41])
+    knull = rng.choice([0, 0, 200_000])
+    dnull = rng.choice([0, 0, 200_000])
+    specs = [(fz.TYPE_INT64, 0, 0, 1 << kbits, knull),
+             (fz.TYPE_INT64, 0, 0, 1 << (dbits - 1), dnull),
+             (fz.TYPE_INT64, 0, 0, 1000, 0),
+             (fz.TYPE_DOUBLE, 3, 0, 0, 0)]
+    aggs = [("count_star", -1), ("count_distinct", 1)]
+    if rng.random() < 0.5:
+        aggs.append(("sum", 2))
+    if rng.random() < 0.5:
+        aggs.append(("sum_distinct", 1))
+    if rng.random() < 0.4:
+        aggs.append(("avg", 3))
+    conj = [(2, "<", rng.randrange(100, 1000))] if rng.random() < 0.6 else []
+    n = rng.choice([20_000, 200_000, 1_000_000])
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, rng.randrange(1 << 40))
+        plan = QueryPlan(t.col_types, conjuncts=conj, group=[0], aggs=aggs,
+                         group_bits=[kbits + 1], group_base=[0],
+                         distinct_bits=dbits, distinct_base=0)
+        outs = {}
+        for mode in ("1", "0"):
+            os.environ["BK_DEDUP_SORT"] = mode
+            r = eng.filter_agg_distinct(t, plan, expected_l1_groups=1 << 28)
+            try:
+                outs[mode] = r.fetch(sorted=True)
+            finally:
+                r.free()
+        os.environ.pop("BK_DEDUP_SORT", None)
+        a, b = outs["1"], outs["0"]
+        assert a["ngroups"] == b["ngroups"], "ngroups"
+        assert np.array_equal(a["enc"], b["enc"]), "enc"
+        assert np.array_equal(a["flags"], b["flags"]), "flags"
+        for i, (name, _) in enumerate(aggs):
+            if name in ("count_star", "count_distinct", "sum"):
+                assert np.array_equal(a["agg_i"][i], b["agg_i"][i]), name
+            else:
+                assert np.allclose(a["agg_d"][i], b["agg_d"][i],
+                                   rtol=0, atol=1e-6), name
+    except Exception as e:
+        dfails += 1
+        print(f"DEDUP-FAIL {cs}: specs={specs} aggs={aggs} conj={conj}: {e}",
+              flush=True)
+    finally:
+        t.free()
+print(f"dedup soak: {DD_CASES - dfails}/{DD_CASES} ok", flush=True)
+total_fails = fails + wfails + sfails + dfails
 print(f"SOAK {'PASS' if total_fails == 0 else 'FAIL'} "
       f"({total_fails} failures)", flush=True)
 sys.exit(1 if total_fails else 0)
