@@ -2002,7 +2002,11 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     while ((int64_t)P < expected_groups / 1024 && P < 4096) P <<= 1;
     const char* envP = getenv("BK_PART_P");
     if (envP) P = (uint32_t)atoi(envP);
-    int nblocks = 2048;
+    /* 8192 blocks: with 3-4 resident blocks/CU the deeper dispatch queue
+     * keeps CUs fed through the drain tail (-2-3% on histo+scatter at 1e9;
+     * all three kernels are ~70-87% wave-parked on memory waits, so more
+     * in-flight blocks are the only free-lunch lever left) */
+    int nblocks = 8192;
     const char* envB = getenv("BK_PART_BLOCKS");
     if (envB) nblocks = atoi(envB);
     /* histo+scatter MUST share grid AND block shape (H rows are per-block).
@@ -2158,7 +2162,7 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     uint32_t P = 64;
     while ((int64_t)P < expected_groups / 1024 && P < 4096) P <<= 1;
     if (const char* e = getenv("BK_PART_P")) P = (uint32_t)atoi(e);
-    int nblocks = 2048;
+    int nblocks = 8192;
     if (const char* e = getenv("BK_PART_BLOCKS")) nblocks = atoi(e);
     int threads = 512;
     if (const char* e = getenv("BK_PART_THREADS")) threads = atoi(e);
