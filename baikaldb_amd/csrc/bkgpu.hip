@@ -1746,6 +1746,80 @@ extern "C" int bkgpu_table_upload(BkgTable* t, int col, const void* data,
     return 0;
 }
 
+/* derived remapped STRING column: newcode[r] = remap[oldcode[r]] — the
+ * engine-side compilation of a unary string scalar fn (upper/lower/substr,
+ * internal_functions.cpp via fn_manager.cpp:97-137) applied to a dict
+ * column: the host transforms the WORD LIST, dedups/sorts it (so the new
+ * codes stay order-preserving) and hands the old->new code map here; the
+ * derived column then acts as a normal dict column in GROUP BY / ORDER BY /
+ * MIN/MAX. NULL cells pass through (validity is shared with the source). */
+__global__ void k_remap_col(const int32_t* src, int64_t n,
+                            const int32_t* remap, int64_t ncodes,
+                            int32_t* dst) {
+    int64_t gs = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += gs) {
+        int32_t c = src[r];
+        dst[r] = (c >= 0 && c < ncodes) ? remap[c] : 0;
+    }
+}
+
+extern "C" int bkgpu_table_derive_remap(BkgTable* t, int src_col,
+                                        const int32_t* remap, int64_t ncodes,
+                                        int64_t new_ncodes) {
+    if (!t || src_col < 0 || src_col >= t->ncols ||
+        t->specs[src_col].col_type != BK_STRING) {
+        set_err("derive_remap: need a BK_STRING source column");
+        return -1;
+    }
+    if (t->ncols >= BK_MAX_COLS) { set_err("derive_remap: table full"); return -1; }
+    if (ensure_device() != 0) return -1;
+    int nc = t->ncols;
+    int32_t* dcol = nullptr;
+    int32_t* dremap = nullptr;
+    HIP_CHECK(hipMalloc((void**)&dcol, (size_t)t->nrows * 4));
+    if (hipMalloc((void**)&dremap, (size_t)ncodes * 4) != hipSuccess) {
+        (void)hipFree(dcol);
+        set_err("derive_remap: oom");
+        return -1;
+    }
+    if (hipMemcpy(dremap, remap, (size_t)ncodes * 4,
+                  hipMemcpyHostToDevice) != hipSuccess) {
+        (void)hipFree(dcol); (void)hipFree(dremap);
+        set_err("derive_remap: remap upload failed");
+        return -1;
+    }
+    hipLaunchKernelGGL(k_remap_col, dim3(2048), dim3(256), 0, 0,
+                       (const int32_t*)t->data[src_col], t->nrows,
+                       dremap, ncodes, dcol);
+    if (hipDeviceSynchronize() != hipSuccess) {
+        (void)hipFree(dcol); (void)hipFree(dremap);
+        set_err("derive_remap: remap failed");
+        return -1;
+    }
+    (void)hipFree(dremap);
+    t->data[nc] = dcol;
+    /* own copy of the validity bytes (table_free frees per column) */
+    t->valid[nc] = nullptr;
+    if (t->valid[src_col]) {
+        if (hipMalloc((void**)&t->valid[nc], (size_t)t->nrows) != hipSuccess ||
+            hipMemcpy(t->valid[nc], t->valid[src_col], (size_t)t->nrows,
+                      hipMemcpyDeviceToDevice) != hipSuccess) {
+            (void)hipFree(dcol);
+            if (t->valid[nc]) (void)hipFree(t->valid[nc]);
+            t->valid[nc] = nullptr;
+            set_err("derive_remap: validity copy failed");
+            return -1;
+        }
+    }
+    t->specs[nc] = t->specs[src_col];
+    t->specs[nc].p0 = new_ncodes;
+    t->dict[nc] = nullptr;
+    t->stat_ok[nc] = 0;
+    t->ncols = nc + 1;
+    return nc;
+}
+
 /* ------------------------------------------------------------------ */
 /* host: aggregation                                                   */
 /* ------------------------------------------------------------------ */

@@ -68,6 +68,10 @@ def _load():
     lib.bkgpu_agg_rollup.restype = C.c_void_p
     lib.bkgpu_agg_rollup.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
                                      C.POINTER(C.c_int32), C.c_int64]
+    lib.bkgpu_table_derive_remap.restype = C.c_int
+    lib.bkgpu_table_derive_remap.argtypes = [C.c_void_p, C.c_int,
+                                             C.POINTER(C.c_int32), C.c_int64,
+                                             C.c_int64]
     lib.bkgpu_filter_agg_sorted.restype = C.c_void_p
     lib.bkgpu_filter_agg_sorted.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
                                             C.c_int64, C.c_int64]
@@ -228,6 +232,40 @@ class GpuEngine:
         if not h:
             raise RuntimeError(f"filter_agg: {self.lib.bkgpu_last_error().decode()}")
         return AggResult(self, h, plan)
+
+    # unary string scalar fns the engine compiles to dict remaps
+    # (internal_functions.cpp upper/lower/substr via fn_manager.cpp:97-137)
+    STRING_FNS = {
+        "upper": lambda w: w.upper(),
+        "lower": lambda w: w.lower(),
+        "reverse": lambda w: w[::-1],
+    }
+
+    def derive_string_fn(self, table, col, fn, words):
+        """GROUP BY fn(varchar_col): transform the column's word list
+        (words[code] for code 0..ncodes), dedup + sort the transformed words
+        (order-preserving new codes), append a derived dict column with
+        newcode = remap[oldcode] (bkgpu_table_derive_remap). Returns
+        (new_col_index, new_words). fn: a STRING_FNS name, a callable, or
+        ("substr", start, len) mirroring internal_functions.cpp substr."""
+        if isinstance(fn, tuple) and fn[0] == "substr":
+            start, ln = fn[1], fn[2]
+            f = lambda w: w[start - 1:start - 1 + ln]  # 1-based, substr()
+        elif callable(fn):
+            f = fn
+        else:
+            f = self.STRING_FNS[fn]
+        tw = [f(w) for w in words]
+        new_words = sorted(set(tw))
+        idx = {w: i for i, w in enumerate(new_words)}
+        remap = (C.c_int32 * len(words))(*[idx[w] for w in tw])
+        nc = self.lib.bkgpu_table_derive_remap(table.handle, col, remap,
+                                               len(words), len(new_words))
+        if nc < 0:
+            raise RuntimeError(
+                f"derive_remap: {self.lib.bkgpu_last_error().decode()}")
+        table.col_types.append(table.col_types[col])
+        return nc, new_words
 
     def filter_agg_sorted(self, table, plan: QueryPlan, row_begin=0,
                           row_end=None):

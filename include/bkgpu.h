@@ -58,6 +58,14 @@ int  bkgpu_table_generate(BkgTable* t, uint64_t seed, int64_t row_begin);
 /* Upload host column data (and optional validity bytes) instead. */
 int  bkgpu_table_upload(BkgTable* t, int col, const void* data, const uint8_t* valid);
 int64_t bkgpu_table_nrows(const BkgTable* t);
+/* Append a derived BK_STRING column: newcode[r] = remap[oldcode[r]] — the
+ * engine-side compilation of a unary string scalar fn (upper/lower/substr,
+ * src/expr/internal_functions.cpp via fn_manager.cpp:97-137) on a dict
+ * column: the caller transforms the word list, dedups/sorts it (codes stay
+ * order-preserving) and passes old->new code map (len ncodes). Returns the
+ * new column index, usable in GROUP BY / ORDER BY / MIN/MAX. */
+int bkgpu_table_derive_remap(BkgTable* t, int src_col, const int32_t* remap,
+                             int64_t ncodes, int64_t new_ncodes);
 int32_t bkgpu_table_col_type(const BkgTable* t, int col);
 int32_t bkgpu_table_ncols(const BkgTable* t);
 void bkgpu_table_free(BkgTable* t);
