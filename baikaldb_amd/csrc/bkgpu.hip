@@ -2195,7 +2195,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
          * chunks x distinct-per-chunk), while >=2048 chunks keep the chip
          * balanced */
         uint64_t chunk = std::max<uint64_t>(
-            32768, std::min<uint64_t>(1u << 20, total / 2048));
+            32768, std::min<uint64_t>(1u << 20, total / 4096));
         const char* envC = getenv("BK_AGG_CHUNK");
         if (envC) chunk = (uint64_t)atoll(envC);
         uint64_t nchunks = (total + chunk - 1) / chunk;
@@ -2347,7 +2347,7 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
                                dc, *q, lay, cb, ce, P, sl[si].bucketid,
                                sl[si].H, sl[si].rec, total, 0);
             uint64_t chunk_sz = std::max<uint64_t>(
-                32768, std::min<uint64_t>(1u << 20, total / 2048));
+                32768, std::min<uint64_t>(1u << 20, total / 4096));
             if (const char* e = getenv("BK_AGG_CHUNK")) chunk_sz = (uint64_t)atoll(e);
             uint64_t nrec_chunks = (total + chunk_sz - 1) / chunk_sz;
             /* cap the aggregate's grid so the OTHER stream's HBM-bound
