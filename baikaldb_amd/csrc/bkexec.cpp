@@ -441,12 +441,12 @@ public:
                 if (_desc.distinct_bits > 0) {
                     q1.group_bits[q.n_group] = _desc.distinct_bits;
                     q1.group_base[q.n_group] = _desc.distinct_base;
-                    int bits_sum = _desc.distinct_bits;
-                    for (int32_t k = 0; k < q.n_group; k++)
-                        bits_sum += q.group_bits[k] ? q.group_bits[k] : 64;
-                    if (bits_sum <= 56 && expected * 16 >= (1ll << 22))
-                        l1 = bkgpu_filter_agg_sorted(t, &q1, 0, nrows_t);
                 }
+                /* high-cardinality dedup: try the sort-based level 1 (packs
+                 * via declared widths or column stats, bkdedup.inc); falls
+                 * back to the hash path when the shape does not qualify */
+                if (expected * 16 >= (1ll << 22))
+                    l1 = bkgpu_filter_agg_sorted(t, &q1, 0, nrows_t);
                 if (!l1)
                     l1 = bkgpu_filter_agg(t, &q1, 0, nrows_t, expected * 16);
                 BkgAggOut* r = l1 ? bkgpu_agg_rollup(l1, &q2, src_idx,

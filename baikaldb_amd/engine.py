@@ -292,11 +292,9 @@ class GpuEngine:
         env = os.environ.get("BK_DEDUP_SORT")
         if env is not None:
             return env != "0"
-        if expected_l1_groups < self.DEDUP_SORT_MIN_L1:
-            return False
-        bits = [l1_plan.group_bits[i] if i < len(l1_plan.group_bits) and
-                l1_plan.group_bits[i] else 64 for i in range(len(l1_plan.group))]
-        return sum(bits) <= 56
+        # no declared-width requirement: bkgpu_filter_agg_sorted auto-packs
+        # from column stats and errors out (-> hash fallback) otherwise
+        return expected_l1_groups >= self.DEDUP_SORT_MIN_L1
 
     def filter_agg_distinct(self, table, plan: QueryPlan, row_begin=0,
                             row_end=None, expected_l1_groups=1 << 18,
