@@ -173,6 +173,53 @@ static int run_window(BkgTable* table) {
     return 0;
 }
 
+static int run_distinct(BkgTable* table) {
+    /* SELECT g, COUNT(*), COUNT(DISTINCT v) FROM t GROUP BY g — the AggNode
+     * applies the reference's multi-distinct rewrite internally
+     * (agg_node.cpp:247-258); a high expected_groups routes level 1 through
+     * the sort-dedup path (bkdedup.inc) with automatic fallback. */
+    BkPlanNodeDesc plan[2];
+    memset(plan, 0, sizeof plan);
+    plan[0].node_type = BK_AGG_NODE;
+    plan[0].num_children = 1;
+    plan[0].limit = 5;
+    plan[0].n_group = 1;
+    plan[0].group_cols[0] = 0;
+    plan[0].n_aggs = 2;
+    plan[0].aggs[0] = {BK_AGG_COUNT_STAR, -1};
+    plan[0].aggs[1] = {BK_AGG_COUNT_DISTINCT, 2};
+    plan[0].expected_groups = 1 << 18;   /* high: sort-dedup level 1 */
+    plan[1].node_type = BK_SCAN_NODE;
+    plan[1].limit = -1;
+    plan[1].table = table;
+
+    BkExecTree* t = bkexec_create_tree(plan, 2);
+    if (!t) { fprintf(stderr, "create_tree: %s\n", bkgpu_last_error()); return 1; }
+    if (bkexec_open(t) < 0) { fprintf(stderr, "distinct open failed\n"); return 1; }
+    int ns = bkexec_n_slots(t);
+    std::vector<int32_t> tag(8 * ns);
+    std::vector<int64_t> vi(8 * ns);
+    std::vector<double> vd(8 * ns);
+    std::vector<uint8_t> nul(8 * ns);
+    int eos = 0;
+    char word[64];
+    printf("count-distinct (GROUP BY g): first rows\n");
+    while (!eos) {
+        int64_t n = bkexec_get_next(t, 8, tag.data(), vi.data(), vd.data(),
+                                    nul.data(), &eos);
+        if (n < 0) { fprintf(stderr, "distinct get_next failed\n"); return 1; }
+        for (int64_t r = 0; r < n; r++) {
+            size_t i0 = (size_t)r * ns;
+            if (bkgpu_table_dict_word(table, 0, vi[i0], word, sizeof word) < 0)
+                snprintf(word, sizeof word, "code:%lld", (long long)vi[i0]);
+            printf("  g=%s count=%lld count_distinct_v=%lld\n",
+                   word, (long long)vi[i0 + 1], (long long)vi[i0 + 2]);
+        }
+    }
+    bkexec_close(t);
+    return 0;
+}
+
 int main(int argc, char** argv) {
     if (argc != 2) {
         fprintf(stderr, "usage: %s <file.parquet>\n", argv[0]);
@@ -192,6 +239,7 @@ int main(int argc, char** argv) {
     int rc = run_agg(table);
     if (rc == 0) rc = run_sort(table);
     if (rc == 0) rc = run_window(table);
+    if (rc == 0) rc = run_distinct(table);
     bkgpu_table_free(table);
     return rc;
 }
