@@ -130,20 +130,25 @@ __device__ __forceinline__ bool in_list_hit(const BkConjunct& cj, int64_t v) {
     return false;
 }
 
+template <bool SIMPLE = false>
 __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpec& q,
                                            int64_t r) {
     /* manually scalarized so the staged values live in registers (indexed
      * locals spill to scratch — guide §5.4 rule 20); the first 4 conjuncts
      * (all bench queries) issue eagerly, the rare rest evaluates lazily */
+    /* SIMPLE instantiation (host-verified: plain compares, non-nullable
+     * int-encoded columns, no fns/IN/doubles, <= 4 conjuncts): the validity
+     * / fn / IN / double branches compile out — the generic machinery costs
+     * ~20% of the streaming ceiling (bwprobe predk vs preds) */
     #define BK_EVAL1(J, VI, VD, OK)                                         \
         int64_t VI = 0; double VD = 0.0; bool OK = true;                    \
         if (q.n_conjuncts > (J)) {                                          \
             const BkConjunct& cj = q.conjuncts[J];                          \
             const DevCol& c = cols.c[cj.col];                               \
-            OK = cell_valid(c, r);                                          \
-            if (cj.cmp_type == BK_DOUBLE) VD = cell_f64(c, r);              \
+            if (!SIMPLE) OK = cell_valid(c, r);                             \
+            if (!SIMPLE && cj.cmp_type == BK_DOUBLE) VD = cell_f64(c, r);   \
             else { VI = cell_i64(c, r);                                     \
-                   if (cj.fn) VI = bk_scalar_fn(cj.fn, VI); }               \
+                   if (!SIMPLE && cj.fn) VI = bk_scalar_fn(cj.fn, VI); }    \
         }
     BK_EVAL1(0, vi0, vd0, ok0)
     BK_EVAL1(1, vi1, vd1, ok1)
@@ -154,16 +159,16 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
         if (q.n_conjuncts > (J)) {                                          \
             const BkConjunct& cj = q.conjuncts[J];                          \
             bool pass;                                                      \
-            if (cj.op >= BK_OP_IN_BITMAP) {                                 \
+            if (!SIMPLE && cj.op >= BK_OP_IN_BITMAP) {                                 \
                 const uint8_t* bm = (const uint8_t*)(uintptr_t)cj.lit_i;    \
                 bool hit = (VI) >= 0 && (VI) < cj.n_in &&                   \
                            ((bm[(VI) >> 3] >> ((VI) & 7)) & 1);             \
                 pass = cj.op == BK_OP_IN_BITMAP ? hit : !hit;               \
-            } else if (cj.op >= BK_OP_IN) {                                 \
+            } else if (!SIMPLE && cj.op >= BK_OP_IN) {                      \
                 bool found = in_list_hit(cj, (VI));                         \
                 pass = cj.op == BK_OP_IN ? found : !found;                  \
             } else {                                                        \
-                int cmp = (cj.cmp_type == BK_DOUBLE)                        \
+                int cmp = (!SIMPLE && cj.cmp_type == BK_DOUBLE)             \
                     ? ((VD > cj.lit_d) - (VD < cj.lit_d))                   \
                     : ((VI > cj.lit_i) - (VI < cj.lit_i));                  \
                 switch (cj.op) {                                            \
@@ -183,7 +188,7 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
     BK_TEST1(2, vi2, vd2, ok2)
     BK_TEST1(3, vi3, vd3, ok3)
     #undef BK_TEST1
-    for (int32_t j = 4; j < q.n_conjuncts && pass_all; j++) {
+    for (int32_t j = 4; !SIMPLE && j < q.n_conjuncts && pass_all; j++) {
         const BkConjunct& cj = q.conjuncts[j];
         const DevCol& c = cols.c[cj.col];
         if (!cell_valid(c, r)) return false;
@@ -241,6 +246,7 @@ __device__ __forceinline__ uint64_t enc_value(const DevCol& c, int64_t r) {
  * keys are identical on every rank and partial blobs merge across GPUs. */
 struct KeyPack { uint64_t k0, k1; uint32_t flag; };
 
+template <bool SIMPLE = false>
 __device__ __forceinline__ KeyPack pack_group_keys(const DevCols& cols,
                                                    const BkQuerySpec& q,
                                                    int64_t r) {
@@ -253,10 +259,10 @@ __device__ __forceinline__ KeyPack pack_group_keys(const DevCols& cols,
         if (shift + bits > 64) { word++; shift = 0; }
         const DevCol& c = cols.c[q.group_cols[k]];
         uint64_t e = 0;
-        if (!cell_valid(c, r)) {
+        if (!SIMPLE && !cell_valid(c, r)) {
             kp.flag |= 0x80u >> k;   /* null-flag bit, exec_node.cpp:561 */
         } else {
-            e = q.group_fns[k]
+            e = (!SIMPLE && q.group_fns[k])
                     ? bk_enc_i64(bk_scalar_fn(q.group_fns[k], cell_i64(c, r)))
                     : enc_value(c, r);
             if (bits < 64) {
@@ -714,7 +720,7 @@ struct RecLayout {
  * code is issue-limited at 1 row/lane: bwprobe preds R1 3.0 vs R2 3.5 TB/s).
  * LDS carve: HOT: [ hot table: lds_slots*stride u64 | laux 4 u64 | lhist ]
  *           !HOT: [ laux 4 u64 | lhist P u32 ]. */
-template <int BS, bool HOT>
+template <int BS, bool HOT, bool SIMPLE>
 __device__ __forceinline__ void histo_row(
         const DevCols& cols, const BkQuerySpec& q, int64_t r, int64_t row_begin,
         uint16_t* bucketid, uint32_t P, uint32_t* lhist, uint64_t* ltab,
@@ -722,9 +728,9 @@ __device__ __forceinline__ void histo_row(
         volatile uint32_t* lmode, uint32_t lcap, uint32_t hot_probe,
         uint32_t hot_min, int64_t& my_passed) {
     int64_t i = r - row_begin;
-    if (!row_passes(cols, q, r)) { bucketid[i] = (uint16_t)BK_SKIP_BUCKET; return; }
+    if (!row_passes<SIMPLE>(cols, q, r)) { bucketid[i] = (uint16_t)BK_SKIP_BUCKET; return; }
     my_passed++;
-    KeyPack kp = pack_group_keys(cols, q, r);
+    KeyPack kp = pack_group_keys<SIMPLE>(cols, q, r);
     uint32_t flag = kp.flag;
     uint64_t k0 = kp.k0, k1 = kp.k1;
     if (HOT) {
@@ -765,7 +771,7 @@ __device__ __forceinline__ void histo_row(
         BODY_A;                                                              \
     }
 
-template <int BS, bool HOT>
+template <int BS, bool HOT, bool SIMPLE = false>
 __global__ void __launch_bounds__(BS)
 k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
              uint32_t P, uint16_t* bucketid, uint32_t* H,
@@ -790,12 +796,12 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
     const uint32_t lmask = tslots - 1;
     int64_t my_passed = 0;
     BK_PART_ROWS2(
-        (histo_row<BS, HOT>(cols, q, r, row_begin, bucketid, P, lhist, ltab,
-                            lmask, stride, lfill, lctr, lmode, lcap, hot_probe,
-                            hot_min, my_passed)),
-        (histo_row<BS, HOT>(cols, q, r, row_begin, bucketid, P, lhist, ltab,
-                            lmask, stride, lfill, lctr, lmode, lcap, hot_probe,
-                            hot_min, my_passed)))
+        (histo_row<BS, HOT, SIMPLE>(cols, q, r, row_begin, bucketid, P, lhist,
+                            ltab, lmask, stride, lfill, lctr, lmode, lcap,
+                            hot_probe, hot_min, my_passed)),
+        (histo_row<BS, HOT, SIMPLE>(cols, q, r, row_begin, bucketid, P, lhist,
+                            ltab, lmask, stride, lfill, lctr, lmode, lcap,
+                            hot_probe, hot_min, my_passed)))
     /* rows_passed: reduce per wave, then one atomic per block via laux[1] */
     long long w = my_passed;
     for (int off = 32; off > 0; off >>= 1) w += __shfl_down(w, off, 64);
@@ -909,7 +915,7 @@ __device__ __forceinline__ void scatter_store_rec(uint64_t* dst,
  * BK_PART_ROWS2 ILP2 tiles) so each block sees the rows its H row counted.
  * 2 rows/lane keeps two gather+store chains in flight (the per-record LDS
  * lcur atomic and the scattered 32-64 B store are latency-bound). */
-template <int BS>
+template <int BS, bool SIMPLE>
 __device__ __forceinline__ void scatter_row(
         const DevCols& cols, const BkQuerySpec& q, const RecLayout& lay,
         int64_t r, int64_t row_begin, const uint16_t* bucketid,
@@ -919,7 +925,7 @@ __device__ __forceinline__ void scatter_row(
     uint32_t b = bucketid[i];
     if (b >= BK_HOT_BUCKET) return;  /* filtered out or absorbed hot */
     /* key words */
-    KeyPack kp = pack_group_keys(cols, q, r);
+    KeyPack kp = pack_group_keys<SIMPLE>(cols, q, r);
     uint64_t meta = kp.flag;
     uint64_t k0 = kp.k0, k1 = kp.k1;
     uint64_t regs[8];
@@ -1018,7 +1024,7 @@ __device__ __forceinline__ void scatter_row(
     }
 }
 
-template <int BS>
+template <int BS, bool SIMPLE = false>
 __global__ void __launch_bounds__(BS)
 k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
                int64_t row_end, uint32_t P, const uint16_t* bucketid,
@@ -1035,10 +1041,10 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
     }
     __syncthreads();
     BK_PART_ROWS2(
-        (scatter_row<BS>(cols, q, lay, r, row_begin, bucketid, lcur, bstate,
-                         stash, rec, paired)),
-        (scatter_row<BS>(cols, q, lay, r, row_begin, bucketid, lcur, bstate,
-                         stash, rec, paired)))
+        (scatter_row<BS, SIMPLE>(cols, q, lay, r, row_begin, bucketid, lcur,
+                         bstate, stash, rec, paired)),
+        (scatter_row<BS, SIMPLE>(cols, q, lay, r, row_begin, bucketid, lcur,
+                         bstate, stash, rec, paired)))
     /* drain leftover stashes (one half-pair per bucket at most) */
     if (paired) {
         __syncthreads();
@@ -2029,12 +2035,40 @@ typedef void (*HistoFn)(DevCols, BkQuerySpec, int64_t, int64_t, uint32_t,
 typedef void (*ScatFn)(DevCols, BkQuerySpec, RecLayout, int64_t, int64_t,
                        uint32_t, const uint16_t*, const uint32_t*, uint64_t*,
                        uint64_t, int);
-static HistoFn pick_histo(int threads, bool hot) {
+/* SIMPLE shape: plain int compares on non-nullable columns, no fns/IN/
+ * doubles, <= 4 conjuncts, <= 2 fn-free non-nullable group keys — the
+ * sysbench/BASELINE shapes. The SIMPLE instantiations compile the unused
+ * branches out of the per-row machinery. */
+static bool query_simple(const BkgTable* t, const BkQuerySpec* q) {
+    if (q->n_conjuncts > 4 || q->n_group > 2) return false;
+    for (int32_t j = 0; j < q->n_conjuncts; j++) {
+        const BkConjunct& cj = q->conjuncts[j];
+        if (cj.op >= BK_OP_IN || cj.fn || cj.cmp_type == BK_DOUBLE)
+            return false;
+        if (t->valid[cj.col]) return false;
+    }
+    for (int32_t k = 0; k < q->n_group; k++) {
+        if (q->group_fns[k]) return false;
+        if (t->valid[q->group_cols[k]]) return false;
+    }
+    return true;
+}
+static HistoFn pick_histo(int threads, bool hot, bool simple) {
+    if (simple && !hot) {
+        if (threads == 512)  return k_part_histo<512, false, true>;
+        if (threads == 1024) return k_part_histo<1024, false, true>;
+        return k_part_histo<256, false, true>;
+    }
     if (threads == 512)  return hot ? k_part_histo<512, true>  : k_part_histo<512, false>;
     if (threads == 1024) return hot ? k_part_histo<1024, true> : k_part_histo<1024, false>;
     return hot ? k_part_histo<256, true> : k_part_histo<256, false>;
 }
-static ScatFn pick_scat(int threads) {
+static ScatFn pick_scat(int threads, bool simple) {
+    if (simple) {
+        if (threads == 512)  return k_part_scatter<512, true>;
+        if (threads == 1024) return k_part_scatter<1024, true>;
+        return k_part_scatter<256, true>;
+    }
     if (threads == 512)  return k_part_scatter<512>;
     if (threads == 1024) return k_part_scatter<1024>;
     return k_part_scatter<256>;
@@ -2138,8 +2172,9 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     if (const char* e = getenv("BK_HOT_MIN")) hot_min = (uint32_t)atoi(e);
     bool hot = hot_min <= 4096;
     if (!hot) hot_slots = 0;
-    HistoFn histo_fn = pick_histo(threads, hot);
-    ScatFn scat_fn = pick_scat(threads);
+    bool simple = query_simple(t, q) && getenv("BK_NO_SIMPLE") == nullptr;
+    HistoFn histo_fn = pick_histo(threads, hot, simple);
+    ScatFn scat_fn = pick_scat(threads, simple);
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     EvTimer tm;
     tm.record();
@@ -2267,8 +2302,9 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     if (const char* e = getenv("BK_HOT_MIN")) hot_min = (uint32_t)atoi(e);
     bool hot = hot_min <= 4096;
     if (!hot) hot_slots = 0;
-    HistoFn histo_fn = pick_histo(threads, hot);
-    ScatFn scat_fn = pick_scat(threads);
+    bool simple = query_simple(t, q) && getenv("BK_NO_SIMPLE") == nullptr;
+    HistoFn histo_fn = pick_histo(threads, hot, simple);
+    ScatFn scat_fn = pick_scat(threads, simple);
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     size_t sc_lds = (size_t)P * 8;
 
