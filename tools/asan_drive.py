@@ -27,6 +27,9 @@ for cs in range(25):
             fn = _FNS[col[0]]; col = col[1]
         ct = 12 if (types[col]==12 and not isinstance(lit,(list,tuple))) or isinstance(lit,float) else 6
         oc.append((col, ops[op], ct, lit, fn))
+    from baikaldb_amd.plan import _FNS as _GF
+    group = [(_GF[g[0]], g[1]) if isinstance(g, tuple) and isinstance(g[0], str)
+             else g for g in group]
     q = make_query(oc, group, [(am[a], c) for a, c in aggs], types)
     orc.filter_agg(cols, valids, types, q, nthreads=3, dict_seed=7)
     # sort + window paths
@@ -36,7 +39,8 @@ for cs in range(25):
         orc.sort_topk(cols, valids, types, order, 50)
         orc.window(cols, valids, types,
                    [(10, -1, 0), (2, icols[0], 0), (19, -1, 0), (20, -1, 3)],
-                   part_col=group[0] if group else -1, order=order)
+                   part_col=group[0] if group and isinstance(group[0], int) else -1,
+                   order=order)
         orc.window(cols, valids, types, [(2, icols[0], 0), (1, icols[0], 0)],
                    part_col=-1, order=order, frame=(2, 1))
 print("ASAN DRIVE OK")
