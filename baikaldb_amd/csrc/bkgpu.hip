@@ -2465,11 +2465,19 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
 
     /* small/mid ranges: the sort-dedup path beats the partitioned pipeline
      * (histo+scatter+agg fixed passes dominate at low survivor volume; a
-     * 1e8-row config2-shaped query measured ~3x faster sorted). The sorted
-     * path auto-packs keys from column stats and falls through here when
-     * the shape does not qualify. BK_SORTED_RANGE overrides (0 = off). */
+     * 1e8-row config2-shaped query measured ~3x faster sorted). An EQUALITY
+     * conjunct raises the range bound: it marks a selective scan (the
+     * reference's index selector uses exactly this signal,
+     * src/physical_plan/index_selector.cpp) and the sorted path's cost
+     * scales with SURVIVORS, not rows (1e9-row 8-int64 c1<K AND c2=K2:
+     * sorted 7.6 ms vs partitioned 12.4 ms). The sorted path auto-packs
+     * keys from column stats and falls through here when the shape does
+     * not qualify. BK_SORTED_RANGE overrides (0 = off). */
     if (partitioned) {
-        int64_t smax = 200 * 1000 * 1000;
+        bool has_eq = false;
+        for (int32_t j = 0; j < q->n_conjuncts; j++)
+            if (q->conjuncts[j].op == BK_OP_EQ) has_eq = true;
+        int64_t smax = has_eq ? 1200 * 1000 * 1000ll : 200 * 1000 * 1000;
         if (const char* e = getenv("BK_SORTED_RANGE")) smax = atoll(e);
         if (row_end - row_begin <= smax) {
             BkgAggOut* so = bkgpu_filter_agg_sorted(t, q, row_begin, row_end);

@@ -47,6 +47,20 @@ CONFIGS = {
         # algorithmic bytes/row: c0,c1 for every row (16 B); group c2 + agg
         # c3 only for the 0.5 * 0.05 surviving fraction
         bytes_per_row=16 + 16 * 0.5 * 0.05),
+    # the north_star target shape: "1e9-row 8-INT64-col scan+filter+GROUP BY
+    # at 1 GPU, >= 40% of HBM3E peak read bandwidth, >= 10x CPU" — config2's
+    # query at 1e9 rows
+    "config2_1e9_8int64": dict(
+        nrows=1_000_000_000,
+        specs=[(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+               (TYPE_INT64, D_UNI, 0, 20, 0)]
+            + [(TYPE_INT64, 4, 100_000, 0, 0)]
+            + [(TYPE_INT64, D_UNI, 0, 1000, 0)]
+            + [(TYPE_INT64, D_UNI, 0, 1 << 31, 0)] * 4,
+        conjuncts=[(0, "<", 1 << 30), (1, "=", 7)],
+        group=[2], aggs=[("sum", 3)],
+        expected_groups=1 << 18,
+        bytes_per_row=16 + 16 * 0.5 * 0.05),
     "config3_1e9_mixed": dict(
         nrows=1_000_000_000,
         specs=[(TYPE_INT64, D_UNI, 0, 1 << 31, 0),      # c0 predicate
