@@ -910,25 +910,14 @@ __device__ __forceinline__ void scatter_store_rec(uint64_t* dst,
     }
 }
 
-/* pass 2: scatter AoS records into bucket-contiguous regions.
- * MUST run with the same grid/block AND row mapping as k_part_histo (the
- * BK_PART_ROWS2 ILP2 tiles) so each block sees the rows its H row counted.
- * 2 rows/lane keeps two gather+store chains in flight (the per-record LDS
- * lcur atomic and the scattered 32-64 B store are latency-bound). */
-template <int BS, bool SIMPLE>
-__device__ __forceinline__ void scatter_row(
+/* materialize one partition/sort record from a row: key word(s) + agg
+ * inputs per RecLayout (shared by k_part_scatter and the sort path's
+ * record-carrying mat) */
+__device__ __forceinline__ void build_record(
         const DevCols& cols, const BkQuerySpec& q, const RecLayout& lay,
-        int64_t r, int64_t row_begin, const uint16_t* bucketid,
-        uint32_t* lcur, uint32_t* bstate, uint64_t* stash, uint64_t* rec,
-        int paired) {
-    int64_t i = r - row_begin;
-    uint32_t b = bucketid[i];
-    if (b >= BK_HOT_BUCKET) return;  /* filtered out or absorbed hot */
-    /* key words */
-    KeyPack kp = pack_group_keys<SIMPLE>(cols, q, r);
+        int64_t r, const KeyPack& kp, uint64_t* regs) {
     uint64_t meta = kp.flag;
     uint64_t k0 = kp.k0, k1 = kp.k1;
-    uint64_t regs[8];
     #pragma unroll
     for (int w = 0; w < 8; w++) regs[w] = 0;
     if (lay.k1_word == -3) {
@@ -979,6 +968,26 @@ __device__ __forceinline__ void scatter_row(
         }
         regs[lay.meta_word] = meta;
     }
+}
+
+/* pass 2: scatter AoS records into bucket-contiguous regions.
+ * MUST run with the same grid/block AND row mapping as k_part_histo (the
+ * BK_PART_ROWS2 ILP2 tiles) so each block sees the rows its H row counted.
+ * 2 rows/lane keeps two gather+store chains in flight (the per-record LDS
+ * lcur atomic and the scattered 32-64 B store are latency-bound). */
+template <int BS, bool SIMPLE>
+__device__ __forceinline__ void scatter_row(
+        const DevCols& cols, const BkQuerySpec& q, const RecLayout& lay,
+        int64_t r, int64_t row_begin, const uint16_t* bucketid,
+        uint32_t* lcur, uint32_t* bstate, uint64_t* stash, uint64_t* rec,
+        int paired) {
+    int64_t i = r - row_begin;
+    uint32_t b = bucketid[i];
+    if (b >= BK_HOT_BUCKET) return;  /* filtered out or absorbed hot */
+    /* key words */
+    KeyPack kp = pack_group_keys<SIMPLE>(cols, q, r);
+    uint64_t regs[8];
+    build_record(cols, q, lay, r, kp, regs);
     if (!paired) {
         uint32_t pos = atomicAdd(&lcur[b], 1u);
         scatter_store_rec(rec + (size_t)pos * lay.nwords, regs, lay.nwords);
