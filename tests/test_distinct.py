@@ -438,3 +438,42 @@ def test_sorted_dedup_rejects_wide_keys():
     assert not h
     err = eng.lib.bkgpu_last_error().decode()
     assert "sorted" in err or "GROUP BY" in err
+
+
+@pytest.mark.gpu
+def test_gpu_sorted_rec_mode_vs_partitioned(eng):
+    """REC-mode sorted aggregation (>= 2 agg inputs -> record-carrying sort)
+    must equal the partitioned hash path on the same query."""
+    import os
+    from baikaldb_amd import QueryPlan
+    specs = [(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+             (TYPE_INT64, D_UNI, 0, 4000, 0),
+             (TYPE_INT64, D_UNI, 0, 1000, 0),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 120_000),
+             (TYPE_STRING, D_DICT, 300, 0, 0)]
+    t = eng.create_table(specs, 800_000)
+    try:
+        eng.generate(t, SEED + 21)
+        plan = QueryPlan(t.col_types, conjuncts=[(0, "<", 1 << 30)],
+                         group=[1, 4],
+                         aggs=[("count_star", -1), ("sum", 2), ("avg", 3),
+                               ("min", 2)])
+        a = eng.filter_agg_sorted(t, plan)            # auto-pack + REC
+        os.environ["BK_SORTED_RANGE"] = "0"
+        try:
+            b = eng.filter_agg(t, plan, expected_groups=1 << 20)  # partitioned
+        finally:
+            os.environ.pop("BK_SORTED_RANGE", None)
+        try:
+            ga, gb = a.fetch(sorted=True), b.fetch(sorted=True)
+        finally:
+            a.free()
+            b.free()
+    finally:
+        t.free()
+    assert ga["ngroups"] == gb["ngroups"]
+    assert np.array_equal(ga["enc"], gb["enc"])
+    assert np.array_equal(ga["flags"], gb["flags"])
+    for i in (0, 1, 3):
+        assert np.array_equal(ga["agg_i"][i], gb["agg_i"][i])
+    assert np.allclose(ga["agg_d"][2], gb["agg_d"][2], rtol=0, atol=1e-9)
