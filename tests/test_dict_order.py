@@ -92,3 +92,42 @@ def test_group_by_string_fn_derived_column(eng, oracle):
         sel = cols[1][g == gv]
         assert got["agg_i"][0][r_i] == len(sel)
         assert got["agg_i"][1][r_i] == sel.sum()
+
+
+@pytest.mark.gpu
+def test_order_by_string_fn_derived_column(eng, oracle):
+    """ORDER BY substr(s,1,3): derived dict-remap codes are order-preserving
+    over the transformed words, so top-N on the derived column equals a
+    numpy sort of the transformed strings."""
+    import numpy as np
+    from baikaldb_amd import QueryPlan
+    lib = oracle.lib
+    SEED3 = 88_002
+    NCODES, N = 250, 80_000
+    T_I, T_S = 6, 13
+    specs = [(T_S, 2, NCODES, 0, 0), (T_I, 0, 0, 1 << 31, 0)]
+    t = eng.create_table(specs, N)
+    try:
+        eng.generate(t, SEED3)
+        buf = C.create_string_buffer(64)
+        words = []
+        for code in range(NCODES):
+            lib.orc_dict_word(C.c_uint64(SEED3), C.c_int64(code), buf, 64)
+            words.append(buf.value.decode())
+        nc, new_words = eng.derive_string_fn(t, 0, ("substr", 1, 3), words)
+        limit = 500
+        rowids = eng.sort_topk(t, [(nc, 1, 1)], limit)
+    finally:
+        t.free()
+    from oracle.bindings import BkColSpec
+    arr = (BkColSpec * len(specs))()
+    for i, s in enumerate(specs):
+        (arr[i].col_type, arr[i].dist, arr[i].p0, arr[i].p1,
+         arr[i].null_frac_x1e6) = s
+    cols, valids = oracle.generate_table(list(arr), N, SEED3)
+    tw = np.array([new_words.index(words[c][:3]) for c in range(NCODES)])
+    derived = tw[cols[0]]
+    order = np.lexsort((np.arange(N), derived))[:limit]
+    assert np.array_equal(np.sort(rowids), np.sort(order))
+    # arrival-order ties within equal keys (TopNSorter semantics)
+    assert np.array_equal(derived[rowids], derived[order])
