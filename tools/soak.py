@@ -19,9 +19,10 @@ eng = GpuEngine()
 orc = Oracle()
 
 AGG_CASES = int(sys.argv[1]) if len(sys.argv) > 1 else 40
+SEED_OFF = int(sys.argv[2]) if len(sys.argv) > 2 else 0
 fails = 0
 for cs in range(AGG_CASES):
-    rng = random.Random(90_000 + cs)
+    rng = random.Random(90_000 + SEED_OFF + cs)
     specs, conjuncts, group, aggs = fz.random_case(rng)
     n = rng.choice([1000, 20_000, 120_000])
     try:
@@ -38,9 +39,9 @@ WIN_CASES = AGG_CASES // 2
 wfails = 0
 for cs in range(WIN_CASES):
     try:
-        tw.test_gpu_window_fuzz.__wrapped__(eng, orc, 100_000 + cs) \
+        tw.test_gpu_window_fuzz.__wrapped__(eng, orc, 100_000 + SEED_OFF + cs) \
             if hasattr(tw.test_gpu_window_fuzz, "__wrapped__") else \
-            tw.test_gpu_window_fuzz(eng, orc, 100_000 + cs)
+            tw.test_gpu_window_fuzz(eng, orc, 100_000 + SEED_OFF + cs)
     except Exception as e:
         wfails += 1
         print(f"WIN-FAIL {cs}: {e}", flush=True)
@@ -49,7 +50,7 @@ print(f"window soak: {WIN_CASES - wfails}/{WIN_CASES} ok", flush=True)
 SORT_CASES = AGG_CASES // 2
 sfails = 0
 for cs in range(SORT_CASES):
-    rng = random.Random(110_000 + cs)
+    rng = random.Random(110_000 + SEED_OFF + cs)
     ncols = rng.randint(2, 4)
     specs = []
     for _ in range(ncols):
@@ -79,7 +80,7 @@ from baikaldb_amd import QueryPlan
 DD_CASES = AGG_CASES // 2
 dfails = 0
 for cs in range(DD_CASES):
-    rng = random.Random(130_000 + cs)
+    rng = random.Random(130_000 + SEED_OFF + cs)
     kbits = rng.choice([4, 8, 12])
     dbits = rng.choice([16, 30, 41])
     knull = rng.choice([0, 0, 200_000])
