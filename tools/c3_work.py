@@ -25,7 +25,20 @@ if which == "c3":
     for _ in range(reps):
         r = eng.filter_agg(t, plan, expected_groups=1 << 21)
         r.free()
-else:
+elif which == "c2b":
+    specs = [(T_I, 0, 0, 1 << 31, 0), (T_I, 0, 0, 20, 0),
+             (T_I, 4, 100_000, 0, 0), (T_I, 0, 0, 1000, 0)] + \
+            [(T_I, 0, 0, 1 << 31, 0)] * 4
+    t = eng.create_table(specs, 1_000_000_000)
+    eng.generate(t, 20260915)
+    eng.sync()
+    plan = QueryPlan(t.col_types,
+                     conjuncts=[(0, "<", 1 << 30), (1, "=", 7)],
+                     group=[2], aggs=[("sum", 3)])
+    for _ in range(reps):
+        r = eng.filter_agg(t, plan, expected_groups=1 << 18)
+        r.free()
+elif which == "c5":
     specs = [(T_I, 0, 0, 1 << 31, 0)] * 3
     t = eng.create_table(specs, 1_000_000_000)
     eng.generate(t, 20260915)

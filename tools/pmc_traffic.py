@@ -76,5 +76,11 @@ def main(config, dominant):
 
 
 if __name__ == "__main__":
-    main("config3_1e9_mixed", "k_part_scatter")
-    main("config5_1e9_sort", "k_topk_scan")
+    import sys as _sys
+    which = _sys.argv[1] if len(_sys.argv) > 1 else "all"
+    if which in ("all", "c3"):
+        main("config3_1e9_mixed", "k_part_scatter")
+    if which in ("all", "c5"):
+        main("config5_1e9_sort", "k_topk_scan")
+    if which in ("all", "c2b"):
+        main("config2_1e9_8int64", "k_dedup_mat")
