@@ -85,7 +85,8 @@ int64_t bkgpu_agg_rows_passed(const BkgAggOut* o);
 double  bkgpu_agg_kernel_ms(const BkgAggOut* o);
 /* per-kernel breakdown: fills ms[0..n) and 16-byte names; returns n.
  * Fused path: {fused_agg}; partitioned path: {histo, totals, scan, offsets,
- * scatter, part_agg}. */
+ * scatter, part_agg} (pipelined large ranges report one {pipeline} entry);
+ * sort-dedup path: {dedup_mat, sort, scan, emit}. */
 int     bkgpu_agg_breakdown(const BkgAggOut* o, char* names, double* ms, int cap);
 
 /* ---- partial-aggregate exchange (multi-GPU merge over RCCL) ----
