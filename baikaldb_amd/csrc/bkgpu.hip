@@ -1587,10 +1587,16 @@ extern "C" int bkgpu_table_dict_word(const BkgTable* t, int col, int64_t code,
  * exact size. Host driving is single-threaded (one bthread drives the exec
  * tree in the reference too — SURVEY §8b threading contract). ---- */
 #include <map>
+#include <mutex>
 static std::multimap<size_t, void*> g_pool_free;
 static std::map<void*, size_t> g_pool_sizes;
+/* a store drives MANY region queries concurrently (different exec trees on
+ * different bthreads; only each TREE is single-threaded) — the
+ * pool maps are the engine's only mutable cross-query state */
+static std::mutex g_pool_mu;
 
 static hipError_t pool_alloc(void** p, size_t bytes) {
+    std::lock_guard<std::mutex> lk(g_pool_mu);
     /* round to a size class so runs whose buffer sizes wobble slightly
      * (e.g. the adaptive hot path makes cold-record counts timing-dependent)
      * still hit the cache: 4 KiB classes below 64 MiB, 64 MiB classes above */
@@ -1613,12 +1619,14 @@ static hipError_t pool_alloc(void** p, size_t bytes) {
 
 static void pool_free(void* p) {
     if (!p) return;
+    std::lock_guard<std::mutex> lk(g_pool_mu);
     auto it = g_pool_sizes.find(p);
     if (it == g_pool_sizes.end()) { (void)hipFree(p); return; }
     g_pool_free.insert({it->second, p});
 }
 
 extern "C" void bkgpu_pool_trim(void) {
+    std::lock_guard<std::mutex> lk(g_pool_mu);
     for (auto& kv : g_pool_free) {
         (void)hipFree(kv.second);
         g_pool_sizes.erase(kv.second);
