@@ -169,3 +169,17 @@ def test_nullable_order_with_filter(eng, orc):
     got, exp = run_both(eng, orc, specs, 50_000, [(0, 1, 1), (1, 1, 1)], 1500,
                         conjuncts=[(1, "<", int((1 << 31) * 0.6))])
     assert np.array_equal(got, exp)
+
+
+@pytest.mark.gpu
+def test_topk_dense_collect_stage(eng, orc):
+    """Few distinct keys + limit == n: the collect passes see very high
+    per-block match density — the shape whose LDS stage overflow the round-1
+    soak caught (k_topk_scan<TK> flush-guard reserve)."""
+    specs = [(TYPE_INT64, 0, 0, 3, 0),        # 3 distinct values
+             (TYPE_INT64, 0, 0, 1 << 40, 200_000)]
+    n = 150_000
+    got, exp = run_both(eng, orc, specs, n,
+                        [(0, 1, 1), (1, 0, 0)], n)
+    import numpy as np
+    assert np.array_equal(got, exp)
