@@ -67,6 +67,12 @@ public:
     bool is_traverse_over() const { return _idx >= _rows.size(); }
     void clear() { _rows.clear(); _idx = 0; }
     void truncate(size_t n) { if (n < _rows.size()) _rows.resize(n); }
+    void drop_range(size_t pos, size_t k) {
+        if (k == 0 || pos >= _rows.size()) return;
+        size_t end = pos + k < _rows.size() ? pos + k : _rows.size();
+        _rows.erase(_rows.begin() + (ptrdiff_t)pos,
+                    _rows.begin() + (ptrdiff_t)end);
+    }
 private:
     std::vector<std::unique_ptr<MemRow>> _rows;
     size_t _idx = 0;
@@ -789,13 +795,38 @@ private:
 };
 
 class LimitNode : public ExecNode {
+    int64_t _offset = 0;
+    int64_t _num_rows_skipped = 0;
 public:
+    int init(const BkPlanNodeDesc& node) override {
+        ExecNode::init(node);
+        _offset = node.offset;
+        return 0;
+    }
+    void close(RuntimeState* state) override {
+        ExecNode::close(state);
+        _num_rows_skipped = 0;   /* idempotent reset, limit_node.h:16-18 */
+    }
     int get_next(RuntimeState* state, RowBatch* batch, bool* eos) override {
         if (_children.empty()) { *eos = true; return 0; }
         if (reached_limit()) { *eos = true; return 0; }
         size_t before = batch->size();
         int ret = _children[0]->get_next(state, batch, eos);
         if (ret < 0) return ret;
+        /* OFFSET: drop leading rows until _offset are skipped
+         * (limit_node.cpp _num_rows_skipped loop) */
+        while (_num_rows_skipped < _offset && batch->size() > before) {
+            size_t have = batch->size() - before;
+            size_t need = (size_t)(_offset - _num_rows_skipped);
+            size_t drop = have < need ? have : need;
+            batch->drop_range(before, drop);   /* leading rows of THIS
+                                                  child batch only */
+            _num_rows_skipped += (int64_t)drop;
+            if (batch->size() == before && !*eos) {
+                ret = _children[0]->get_next(state, batch, eos);
+                if (ret < 0) return ret;
+            }
+        }
         _num_rows_returned += (int64_t)(batch->size() - before);
         if (_limit > 0 && _num_rows_returned > _limit) {
             /* truncate the overshoot so exactly _limit rows are emitted */

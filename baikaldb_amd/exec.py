@@ -20,7 +20,7 @@ WINDOW = 42
 
 class BkPlanNodeDesc(C.Structure):
     _fields_ = [("node_type", C.c_int32), ("num_children", C.c_int32),
-                ("limit", C.c_int64),
+                ("limit", C.c_int64), ("offset", C.c_int64),
                 ("table", C.c_void_p),
                 ("n_conjuncts", C.c_int32),
                 ("conjuncts", BkConjunct * BK_MAX_CONJ),
@@ -163,10 +163,11 @@ def sort_node(order, out_cols, limit, num_children=1):
     return d
 
 
-def limit_node(limit, num_children=1):
+def limit_node(limit, num_children=1, offset=0):
     d = BkPlanNodeDesc()
     d.node_type, d.num_children = LIMIT, num_children
     d.limit = limit
+    d.offset = offset
     return d
 
 

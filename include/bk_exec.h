@@ -36,6 +36,8 @@ typedef struct BkPlanNodeDesc {
     int32_t node_type;     /* BkNodeType */
     int32_t num_children;  /* pre-order flattening, like pb::Plan */
     int64_t limit;         /* LIMIT_NODE / SORT_NODE limit, -1 none */
+    int64_t offset;        /* LIMIT_NODE rows skipped before emitting
+                              (limit_node.h:21-41 _offset semantics) */
     /* SCAN_NODE payload */
     BkgTable* table;       /* the region's columnar source */
     /* WHERE/TABLE_FILTER_NODE payload */

@@ -405,3 +405,25 @@ def test_count_distinct_sorted_l1_through_exec_surface(eng, orc):
         assert vi[r, 1] == (g == gv).sum()
         assert vi[r, 2] == len(_np.unique(sel))
         assert vi[r, 3] == sel.sum()
+
+
+def test_limit_offset_through_exec_surface(eng, orc):
+    """LIMIT with OFFSET (limit_node.h:21-41 _offset/_num_rows_skipped):
+    skip the first `offset` sorted rows, emit the next `limit`."""
+    from baikaldb_amd import exec as bx
+    import numpy as _np
+    t, cols, valids, types = make_table(eng, orc, n=50_000)
+    try:
+        nodes = [bx.limit_node(limit=7, offset=12),
+                 bx.sort_node([(0, 1, 1)], [0, 2], limit=-1),
+                 bx.scan_node(t)]
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all()
+        tree.close()
+    finally:
+        t.free()
+    order = _np.lexsort((_np.arange(len(cols[0])), cols[0]))
+    assert tags.shape[0] == 7
+    assert _np.array_equal(vi[:, 0], cols[0][order[12:19]])
+    assert _np.array_equal(vi[:, 1], cols[2][order[12:19]])
