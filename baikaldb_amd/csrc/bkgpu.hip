@@ -155,6 +155,17 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
     BK_EVAL1(2, vi2, vd2, ok2)
     BK_EVAL1(3, vi3, vd3, ok3)
     #undef BK_EVAL1
+    /* clause combine (CNF, BkConjunct.or_group): standalone terms AND with
+     * early short-circuit via pass_all; OR members set per-clause bits —
+     * SIMPLE instantiations are host-guaranteed all-standalone */
+    #define BK_CLAUSE(J, P)                                                 \
+        if (SIMPLE || q.conjuncts[J].or_group == 0) {                       \
+            pass_all = pass_all && (P);                                     \
+        } else {                                                            \
+            uint32_t gbit = 1u << (q.conjuncts[J].or_group & 31);           \
+            or_seen |= gbit;                                                \
+            if (P) or_sat |= gbit;                                          \
+        }
     #define BK_TEST1(J, VI, VD, OK)                                          \
         if (q.n_conjuncts > (J)) {                                          \
             const BkConjunct& cj = q.conjuncts[J];                          \
@@ -180,9 +191,10 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
                     default:       pass = (cmp <= 0); break;                \
                 }                                                           \
             }                                                               \
-            pass_all = pass_all && (OK) && pass;                            \
+            BK_CLAUSE(J, (OK) && pass)                                      \
         }
     bool pass_all = true;
+    uint32_t or_seen = 0, or_sat = 0;
     BK_TEST1(0, vi0, vd0, ok0)
     BK_TEST1(1, vi1, vd1, ok1)
     BK_TEST1(2, vi2, vd2, ok2)
@@ -191,7 +203,11 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
     for (int32_t j = 4; !SIMPLE && j < q.n_conjuncts && pass_all; j++) {
         const BkConjunct& cj = q.conjuncts[j];
         const DevCol& c = cols.c[cj.col];
-        if (!cell_valid(c, r)) return false;
+        if (!cell_valid(c, r)) {
+            if (cj.or_group == 0) return false;
+            or_seen |= 1u << (cj.or_group & 31);
+            continue;
+        }
         bool pass;
         if (cj.op >= BK_OP_IN_BITMAP) {
             int64_t v = cell_i64(c, r);
@@ -223,9 +239,10 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
                 default:       pass = (cmp <= 0); break;
             }
         }
-        pass_all = pass_all && pass;
+        BK_CLAUSE(j, pass)
     }
-    return pass_all;
+    #undef BK_CLAUSE
+    return pass_all && (or_sat & or_seen) == or_seen;
 }
 
 /* order-preserving u64 encode of a group/minmax value (bk_keyenc.h) */
@@ -2060,7 +2077,8 @@ static bool query_simple(const BkgTable* t, const BkQuerySpec* q) {
     if (q->n_conjuncts > 4 || q->n_group > 2) return false;
     for (int32_t j = 0; j < q->n_conjuncts; j++) {
         const BkConjunct& cj = q->conjuncts[j];
-        if (cj.op >= BK_OP_IN || cj.fn || cj.cmp_type == BK_DOUBLE)
+        if (cj.op >= BK_OP_IN || cj.fn || cj.cmp_type == BK_DOUBLE ||
+            cj.or_group)
             return false;
         if (t->valid[cj.col]) return false;
     }

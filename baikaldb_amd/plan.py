@@ -30,7 +30,7 @@ class BkConjunct(C.Structure):
                 ("cmp_type", C.c_int32), ("n_in", C.c_int32),
                 ("lit_i", C.c_int64), ("lit_d", C.c_double),
                 ("in_list", C.c_int64 * 16),
-                ("fn", C.c_int32), ("_pad2", C.c_int32)]
+                ("fn", C.c_int32), ("or_group", C.c_int32)]
 
 
 class BkAggSpec(C.Structure):
@@ -98,8 +98,13 @@ class QueryPlan:
     def to_spec(self):
         q = BkQuerySpec()
         q.n_conjuncts = len(self.conjuncts)
-        for i, (col, op, lit) in enumerate(self.conjuncts):
+        for i, cjt in enumerate(self.conjuncts):
+            # (col, op, lit) or (col, op, lit, or_group): members sharing an
+            # or_group > 0 OR together, clauses AND together (CNF pushdown of
+            # an OR expr tree, filter_node.cpp:726-734)
+            col, op, lit = cjt[0], cjt[1], cjt[2]
             cj = q.conjuncts[i]
+            cj.or_group = cjt[3] if len(cjt) > 3 else 0
             if isinstance(col, tuple):   # ("hour", col): pushed-down scalar fn
                 cj.fn = _FNS[col[0]]
                 col = col[1]

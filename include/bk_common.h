@@ -158,7 +158,12 @@ typedef struct BkConjunct {
     double  lit_d;
     int64_t in_list[BK_MAX_INLIST];  /* int64 / dict-code IN literals */
     int32_t fn;        /* BkScalarFn on the column value (int64 paths only) */
-    int32_t _pad2;
+    int32_t or_group;  /* 0 = standalone AND term; >0 = OR-clause id: members
+                          sharing an id OR together, clauses AND together —
+                          how an OR expr tree pushed into FilterNode::
+                          need_copy (filter_node.cpp:726-734) evaluates in
+                          CNF. SQL ternary: a NULL member is simply not
+                          true. */
 } BkConjunct;
 
 /* One aggregate call (reference: src/expr/agg_fn_call.cpp:496-555 update,

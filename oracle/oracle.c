@@ -125,12 +125,20 @@ static inline double cell_f64_cast(const OrcCol* c, int64_t r) {
     return (double)cell_i64(c, r);
 }
 
-/* returns 1 iff row passes every conjunct (NULL or false rejects). */
+/* returns 1 iff row passes every clause: standalone conjuncts AND
+ * together; or_group members OR within the clause (CNF; see BkConjunct).
+ * NULL or false rejects a standalone term; a NULL OR-member is just not
+ * true. */
 static int orc_row_passes(const OrcCol* cols, const BkQuerySpec* q, int64_t r) {
+    uint32_t or_seen = 0, or_sat = 0;
     for (int32_t j = 0; j < q->n_conjuncts; j++) {
         const BkConjunct* cj = &q->conjuncts[j];
         const OrcCol* c = &cols[cj->col];
-        if (!cell_is_valid(c, r)) return 0;          /* NULL operand => NULL => reject */
+        if (!cell_is_valid(c, r)) {
+            if (cj->or_group == 0) return 0;  /* NULL operand => reject */
+            or_seen |= 1u << (cj->or_group & 31);
+            continue;
+        }
         int cmp; /* sign of (col - lit) */
         if (cj->cmp_type == BK_DOUBLE) {
             double v = cell_f64_cast(c, r);
@@ -177,9 +185,14 @@ static int orc_row_passes(const OrcCol* cols, const BkQuerySpec* q, int64_t r) {
                 default: pass = 0;
             }
         }
-        if (!pass) return 0;
+        if (cj->or_group == 0) {
+            if (!pass) return 0;
+        } else {
+            or_seen |= 1u << (cj->or_group & 31);
+            if (pass) or_sat |= 1u << (cj->or_group & 31);
+        }
     }
-    return 1;
+    return (or_sat & or_seen) == or_seen;
 }
 
 /* ------------------------------------------------------------------ */

@@ -34,9 +34,9 @@ def oracle_distinct(orc, specs, n, conjuncts, group, aggs, seed=SEED,
     plan = QueryPlan(col_types, conjuncts=conjuncts, group=group, aggs=aggs)
     l1_plan, _, src_idx = plan.split_distinct()
     oconj = []
-    for col, op, lit in conjuncts:
+    for col, op, lit, *og in conjuncts:
         ct = TYPE_DOUBLE if isinstance(lit, float) else TYPE_INT64
-        oconj.append((col, OPS[op], ct, lit))
+        oconj.append((col, OPS[op], ct, lit, 0, og[0] if og else 0))
     q1 = make_query(oconj, l1_plan.group,
                     [(AGGMAP[a], c) for a, c in l1_plan.aggs], col_types)
     q2 = make_query((), group, [(AGGMAP[a], c) for a, c in aggs], col_types)
@@ -49,7 +49,7 @@ def brute_distinct(cols, valids, col_types, conjuncts, group, aggs):
     """Numpy recompute (independent of both engine and oracle code paths)."""
     n = len(cols[0])
     mask = np.ones(n, dtype=bool)
-    for col, op, lit in conjuncts:
+    for col, op, lit, *og in conjuncts:
         v = cols[col]
         ok = {"=": v == lit, "!=": v != lit, "<": v < lit, "<=": v <= lit,
               ">": v > lit, ">=": v >= lit}[op]

@@ -69,7 +69,9 @@ def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
            "in": 6, "not_in": 7}
     aggmap = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5}
     oconj = []
-    for col, op, lit in conjuncts:
+    for cjt in conjuncts:
+        col, op, lit = cjt[0], cjt[1], cjt[2]
+        og = cjt[3] if len(cjt) > 3 else 0
         fn = 0
         if isinstance(col, tuple):   # ("hour", col) scalar-fn pushdown
             from baikaldb_amd.plan import _FNS
@@ -78,7 +80,7 @@ def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
         ct = TYPE_DOUBLE if (col_types[col] == TYPE_DOUBLE and
                              not isinstance(lit, (list, tuple))) or \
             isinstance(lit, float) else TYPE_INT64
-        oconj.append((col, ops[op], ct, lit, fn))
+        oconj.append((col, ops[op], ct, lit, fn, og))
     from baikaldb_amd.plan import _FNS
     ogroup = [(_FNS[g[0]], g[1]) if isinstance(g, tuple) else g for g in group]
     q = make_query(oconj, ogroup, [(aggmap[a], c) for a, c in aggs], col_types,
@@ -504,3 +506,40 @@ def test_packed_keys_shard_merge_equals_whole(eng, orc):
     assert np.array_equal(mf["enc"], wf["enc"])
     assert np.array_equal(mf["flags"], wf["flags"])
     assert np.array_equal(mf["agg_i"], wf["agg_i"])
+
+
+@pytest.mark.gpu
+def test_or_clauses(eng, orc):
+    """CNF OR clauses (BkConjunct.or_group): (c0<K1 OR c1>K2) AND c2<=K3,
+    with a nullable OR member (NULL member is just not-true)."""
+    specs = [(TYPE_INT64, D_UNI, 0, 1000, 0),
+             (TYPE_INT64, D_UNI, 0, 1000, 250_000),
+             (TYPE_INT64, D_UNI, 0, 1000, 0),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 0)]
+    conj = [(0, "<", 200, 1), (1, ">", 800, 1), (2, "<=", 900)]
+    aggs = [("count_star", -1), ("sum", 0), ("avg", 3)]
+    got, exp = run_both(eng, orc, specs, 300_000, conj, [2], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
+    # numpy brute on the same generated inputs
+    import numpy as np
+    specs_c = (BkColSpec * len(specs))()
+    for i, s in enumerate(specs):
+        (specs_c[i].col_type, specs_c[i].dist, specs_c[i].p0, specs_c[i].p1,
+         specs_c[i].null_frac_x1e6) = s
+    cols, valids = orc.generate_table(list(specs_c), 300_000, SEED)
+    m0 = cols[0] < 200
+    m1 = (cols[1] > 800) & (valids[1] != 0)
+    keep = (m0 | m1) & (cols[2] <= 900)
+    assert got["rows_passed"] == int(keep.sum())
+
+
+@pytest.mark.gpu
+def test_two_or_clauses(eng, orc):
+    """two independent OR clauses AND a standalone term."""
+    specs = [(TYPE_INT64, D_UNI, 0, 500, 0)] * 4
+    conj = [(0, "<", 100, 1), (1, ">", 400, 1),
+            (2, "=", 7, 2), (2, "=", 9, 2),
+            (3, "!=", 499)]
+    aggs = [("count_star", -1), ("min", 0), ("max", 1)]
+    got, exp = run_both(eng, orc, specs, 200_000, conj, [], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])

@@ -58,6 +58,12 @@ def random_case(rng):
         else:
             conjuncts.append((c, rng.choice(["<", ">", "=", "!=", ">=", "<="]),
                               rng.randint(-100, 3000)))
+    # OR clauses (BkConjunct.or_group CNF): ~25% of cases wrap 2+ of the
+    # conjuncts into one OR clause
+    if len(conjuncts) >= 2 and rng.random() < 0.25:
+        k = rng.randint(2, len(conjuncts))
+        tail = conjuncts[-k:]
+        conjuncts = conjuncts[:-k] + [tuple(cj) + (1,) for cj in tail]
     # group keys (DATETIME columns may group through an extraction fn:
     # GROUP BY year(c) etc.)
     group = rng.sample(range(ncols), rng.randint(0, min(2, ncols)))

@@ -44,7 +44,7 @@ def run_both(eng, orc, spec_rows, n, order, limit, conjuncts=()):
     types = [s[0] for s in spec_rows]
     ops = {"=": 0, "!=": 1, ">": 2, ">=": 3, "<": 4, "<=": 5}
     oconj = []
-    for col, op, lit in conjuncts:
+    for col, op, lit, *og in conjuncts:
         ct = TYPE_DOUBLE if (types[col] == TYPE_DOUBLE or isinstance(lit, float)) \
             else TYPE_INT64
         oconj.append((col, ops[op], ct, lit))

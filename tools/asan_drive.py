@@ -20,13 +20,13 @@ for cs in range(25):
     ops = {"=":0,"!=":1,">":2,">=":3,"<":4,"<=":5,"in":6,"not_in":7}
     am = {"count_star":0,"count":1,"sum":2,"avg":3,"min":4,"max":5}
     oc = []
-    for col, op, lit in conjuncts:
+    for col, op, lit, *og in conjuncts:
         fn = 0
         if isinstance(col, tuple):
             from baikaldb_amd.plan import _FNS
             fn = _FNS[col[0]]; col = col[1]
         ct = 12 if (types[col]==12 and not isinstance(lit,(list,tuple))) or isinstance(lit,float) else 6
-        oc.append((col, ops[op], ct, lit, fn))
+        oc.append((col, ops[op], ct, lit, fn, og[0] if og else 0))
     from baikaldb_amd.plan import _FNS as _GF
     group = [(_GF[g[0]], g[1]) if isinstance(g, tuple) and isinstance(g[0], str)
              else g for g in group]
