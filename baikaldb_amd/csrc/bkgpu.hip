@@ -149,6 +149,20 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
             if (!SIMPLE && cj.cmp_type == BK_DOUBLE) VD = cell_f64(c, r);   \
             else { VI = cell_i64(c, r);                                     \
                    if (!SIMPLE && cj.fn) VI = bk_scalar_fn(cj.fn, VI); }    \
+            if (!SIMPLE && cj.arith) {                                      \
+                const DevCol& c2 = cols.c[cj.col2];                         \
+                OK = OK && cell_valid(c2, r);                               \
+                if (cj.cmp_type == BK_DOUBLE) {                             \
+                    double b = cell_f64(c2, r);                             \
+                    VD = cj.arith == BK_ARITH_ADD ? VD + b                  \
+                         : cj.arith == BK_ARITH_SUB ? VD - b : VD * b;      \
+                } else {                                                    \
+                    int64_t b = cell_i64(c2, r);                            \
+                    uint64_t ua = (uint64_t)VI, ub = (uint64_t)b;           \
+                    VI = (int64_t)(cj.arith == BK_ARITH_ADD ? ua + ub       \
+                         : cj.arith == BK_ARITH_SUB ? ua - ub : ua * ub);   \
+                }                                                           \
+            }                                                               \
         }
     BK_EVAL1(0, vi0, vd0, ok0)
     BK_EVAL1(1, vi1, vd1, ok1)
@@ -224,10 +238,32 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
             int cmp;
             if (cj.cmp_type == BK_DOUBLE) {
                 double v = cell_f64(c, r);
+                if (cj.arith) {
+                    const DevCol& c2 = cols.c[cj.col2];
+                    if (!cell_valid(c2, r)) {
+                        if (cj.or_group == 0) return false;
+                        or_seen |= 1u << (cj.or_group & 31);
+                        continue;
+                    }
+                    double b = cell_f64(c2, r);
+                    v = cj.arith == BK_ARITH_ADD ? v + b
+                        : cj.arith == BK_ARITH_SUB ? v - b : v * b;
+                }
                 cmp = (v > cj.lit_d) - (v < cj.lit_d);
             } else {
                 int64_t v = cell_i64(c, r);
                 if (cj.fn) v = bk_scalar_fn(cj.fn, v);
+                if (cj.arith) {
+                    const DevCol& c2 = cols.c[cj.col2];
+                    if (!cell_valid(c2, r)) {
+                        if (cj.or_group == 0) return false;
+                        or_seen |= 1u << (cj.or_group & 31);
+                        continue;
+                    }
+                    uint64_t ua = (uint64_t)v, ub = (uint64_t)cell_i64(c2, r);
+                    v = (int64_t)(cj.arith == BK_ARITH_ADD ? ua + ub
+                        : cj.arith == BK_ARITH_SUB ? ua - ub : ua * ub);
+                }
                 cmp = (v > cj.lit_i) - (v < cj.lit_i);
             }
             switch (cj.op) {
@@ -2078,7 +2114,7 @@ static bool query_simple(const BkgTable* t, const BkQuerySpec* q) {
     for (int32_t j = 0; j < q->n_conjuncts; j++) {
         const BkConjunct& cj = q->conjuncts[j];
         if (cj.op >= BK_OP_IN || cj.fn || cj.cmp_type == BK_DOUBLE ||
-            cj.or_group)
+            cj.or_group || cj.arith)
             return false;
         if (t->valid[cj.col]) return false;
     }

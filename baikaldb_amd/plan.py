@@ -17,6 +17,7 @@ BK_MAX_GROUP, BK_MAX_CONJ, BK_MAX_AGGS = 4, 8, 8
 _OPS = {"=": OP_EQ, "!=": OP_NE, ">": OP_GT, ">=": OP_GE, "<": OP_LT,
         "<=": OP_LE, "in": 6, "not_in": 7,
         "in_bitmap": 8, "not_in_bitmap": 9}
+_ARITH = {"add": 1, "sub": 2, "mul": 3}
 _FNS = {"year": 1, "month": 2, "day": 3, "dayofmonth": 3, "hour": 4,
         "minute": 5, "second": 6}
 _AGGS = {"count_star": AGG_COUNT_STAR, "count": AGG_COUNT, "sum": AGG_SUM,
@@ -30,7 +31,8 @@ class BkConjunct(C.Structure):
                 ("cmp_type", C.c_int32), ("n_in", C.c_int32),
                 ("lit_i", C.c_int64), ("lit_d", C.c_double),
                 ("in_list", C.c_int64 * 16),
-                ("fn", C.c_int32), ("or_group", C.c_int32)]
+                ("fn", C.c_int32), ("or_group", C.c_int32),
+                ("col2", C.c_int32), ("arith", C.c_int32)]
 
 
 class BkAggSpec(C.Structure):
@@ -105,7 +107,17 @@ class QueryPlan:
             col, op, lit = cjt[0], cjt[1], cjt[2]
             cj = q.conjuncts[i]
             cj.or_group = cjt[3] if len(cjt) > 3 else 0
-            if isinstance(col, tuple):   # ("hour", col): pushed-down scalar fn
+            cj.col2 = -1
+            if isinstance(col, tuple) and col[0] in _ARITH:
+                # ("add"|"sub"|"mul", c1, c2): binary-arith predicate; the
+                # compare domain is DOUBLE iff either column is DOUBLE
+                cj.arith = _ARITH[col[0]]
+                cj.col2 = col[2]
+                if (self.col_types[col[1]] == TYPE_DOUBLE or
+                        self.col_types[col[2]] == TYPE_DOUBLE):
+                    lit = float(lit)
+                col = col[1]
+            elif isinstance(col, tuple):  # ("hour", col): pushed-down scalar fn
                 cj.fn = _FNS[col[0]]
                 col = col[1]
             cj.col = col

@@ -164,7 +164,20 @@ typedef struct BkConjunct {
                           need_copy (filter_node.cpp:726-734) evaluates in
                           CNF. SQL ternary: a NULL member is simply not
                           true. */
+    int32_t col2;      /* binary-arith predicate (a OP b <cmp> lit): second
+                          column; -1 = none. operators.cpp add/minus/
+                          multiply semantics: int64 wraps, mixed/double in
+                          IEEE f64 (cmp_type decides the domain). Either
+                          operand NULL => NULL. */
+    int32_t arith;     /* BkArith */
 } BkConjunct;
+
+typedef enum BkArith {
+    BK_ARITH_NONE = 0,
+    BK_ARITH_ADD = 1,
+    BK_ARITH_SUB = 2,
+    BK_ARITH_MUL = 3,
+} BkArith;
 
 /* One aggregate call (reference: src/expr/agg_fn_call.cpp:496-555 update,
  * 719-830 merge, 927-975 finalize). col == -1 for COUNT(*). */

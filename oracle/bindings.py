@@ -29,7 +29,8 @@ class BkConjunct(C.Structure):
                 ("cmp_type", C.c_int32), ("n_in", C.c_int32),
                 ("lit_i", C.c_int64), ("lit_d", C.c_double),
                 ("in_list", C.c_int64 * 16),
-                ("fn", C.c_int32), ("or_group", C.c_int32)]
+                ("fn", C.c_int32), ("or_group", C.c_int32),
+                ("col2", C.c_int32), ("arith", C.c_int32)]
 
 
 class BkAggSpec(C.Structure):
@@ -90,6 +91,8 @@ def make_query(conjuncts=(), group=(), aggs=(), col_types=None,
         cj.col, cj.op, cj.cmp_type = col, op, cmp_type
         cj.fn = cjt[4] if len(cjt) > 4 else 0
         cj.or_group = cjt[5] if len(cjt) > 5 else 0
+        cj.col2 = cjt[6] if len(cjt) > 6 else -1
+        cj.arith = cjt[7] if len(cjt) > 7 else 0
         if op >= 8:  # bitmap membership: lit = (host_ptr, n_bits)
             cj.lit_i, cj.n_in = int(lit[0]), int(lit[1])
         elif op >= 6:  # IN / NOT IN: small lists inline; big lists as

@@ -139,13 +139,33 @@ static int orc_row_passes(const OrcCol* cols, const BkQuerySpec* q, int64_t r) {
             or_seen |= 1u << (cj->or_group & 31);
             continue;
         }
-        int cmp; /* sign of (col - lit) */
+        if (cj->arith) {
+            /* binary-arith predicate: either operand NULL => NULL */
+            const OrcCol* c2 = &cols[cj->col2];
+            if (!cell_is_valid(c2, r)) {
+                if (cj->or_group == 0) return 0;
+                or_seen |= 1u << (cj->or_group & 31);
+                continue;
+            }
+        }
+        int cmp; /* sign of (expr - lit) */
         if (cj->cmp_type == BK_DOUBLE) {
             double v = cell_f64_cast(c, r);
+            if (cj->arith) {
+                double b = cell_f64_cast(&cols[cj->col2], r);
+                v = cj->arith == BK_ARITH_ADD ? v + b
+                    : cj->arith == BK_ARITH_SUB ? v - b : v * b;
+            }
             cmp = (v > cj->lit_d) - (v < cj->lit_d);
         } else { /* BK_INT64 or BK_STRING dict-code compare */
             int64_t v = cell_i64(c, r);
             if (cj->fn) v = bk_scalar_fn(cj->fn, v);
+            if (cj->arith) {
+                uint64_t ua = (uint64_t)v;
+                uint64_t ub = (uint64_t)cell_i64(&cols[cj->col2], r);
+                v = (int64_t)(cj->arith == BK_ARITH_ADD ? ua + ub
+                    : cj->arith == BK_ARITH_SUB ? ua - ub : ua * ub);
+            }
             cmp = (v > cj->lit_i) - (v < cj->lit_i);
         }
         int pass;

@@ -69,18 +69,22 @@ def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
            "in": 6, "not_in": 7}
     aggmap = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5}
     oconj = []
+    from baikaldb_amd.plan import _FNS, _ARITH
     for cjt in conjuncts:
         col, op, lit = cjt[0], cjt[1], cjt[2]
         og = cjt[3] if len(cjt) > 3 else 0
-        fn = 0
-        if isinstance(col, tuple):   # ("hour", col) scalar-fn pushdown
-            from baikaldb_amd.plan import _FNS
+        fn, col2, arith = 0, -1, 0
+        if isinstance(col, tuple) and col[0] in _ARITH:
+            arith, col2, col = _ARITH[col[0]], col[2], col[1]
+            if col_types[col] == TYPE_DOUBLE or col_types[col2] == TYPE_DOUBLE:
+                lit = float(lit)
+        elif isinstance(col, tuple):   # ("hour", col) scalar-fn pushdown
             fn = _FNS[col[0]]
             col = col[1]
         ct = TYPE_DOUBLE if (col_types[col] == TYPE_DOUBLE and
                              not isinstance(lit, (list, tuple))) or \
             isinstance(lit, float) else TYPE_INT64
-        oconj.append((col, ops[op], ct, lit, fn, og))
+        oconj.append((col, ops[op], ct, lit, fn, og, col2, arith))
     from baikaldb_amd.plan import _FNS
     ogroup = [(_FNS[g[0]], g[1]) if isinstance(g, tuple) else g for g in group]
     q = make_query(oconj, ogroup, [(aggmap[a], c) for a, c in aggs], col_types,
@@ -541,5 +545,42 @@ def test_two_or_clauses(eng, orc):
             (2, "=", 7, 2), (2, "=", 9, 2),
             (3, "!=", 499)]
     aggs = [("count_star", -1), ("min", 0), ("max", 1)]
+    got, exp = run_both(eng, orc, specs, 200_000, conj, [], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
+
+
+@pytest.mark.gpu
+def test_arith_predicates(eng, orc):
+    """binary-arith predicates (operators.cpp add/minus/multiply): int64
+    wraps, mixed int/double compares in f64, NULL operand => not true."""
+    specs = [(TYPE_INT64, D_UNI, 0, 1000, 0),
+             (TYPE_INT64, D_UNI, 0, 1000, 200_000),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 0),
+             (TYPE_INT64, D_UNI, 0, 100, 0)]
+    conj = [(("add", 0, 1), "<", 900),
+            (("mul", 0, 3), ">", 5000),
+            (("sub", 2, 2), "=", 0.0)]
+    aggs = [("count_star", -1), ("sum", 0), ("min", 3)]
+    got, exp = run_both(eng, orc, specs, 250_000, conj, [3], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
+    import numpy as np
+    specs_c = (BkColSpec * len(specs))()
+    for i, sp in enumerate(specs):
+        (specs_c[i].col_type, specs_c[i].dist, specs_c[i].p0, specs_c[i].p1,
+         specs_c[i].null_frac_x1e6) = sp
+    cols, valids = orc.generate_table(list(specs_c), 250_000, SEED)
+    keep = ((cols[0] + cols[1] < 900) & (valids[1] != 0) &
+            (cols[0] * cols[3] > 5000))
+    assert got["rows_passed"] == int(keep.sum())
+
+
+@pytest.mark.gpu
+def test_arith_mixed_double(eng, orc):
+    """int + double arith compares in the double domain (get_numberic cast,
+    expr_value.h:341)."""
+    specs = [(TYPE_INT64, D_UNI, 0, 100, 0),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 100_000)]
+    conj = [(("add", 0, 1), ">", 50.0)]
+    aggs = [("count_star", -1), ("avg", 1)]
     got, exp = run_both(eng, orc, specs, 200_000, conj, [], aggs)
     assert_parity(got, exp, aggs, [s[0] for s in specs])

@@ -55,6 +55,16 @@ def random_case(rng):
         elif rng.random() < 0.3:
             vals = [rng.randint(0, 3000) for _ in range(rng.randint(1, 8))]
             conjuncts.append((c, rng.choice(["in", "not_in"]), vals))
+        elif rng.random() < 0.25 and ncols >= 2:
+            # binary-arith predicate (add/sub/mul); lit domain follows the
+            # operand types (plan casts to f64 when either col is DOUBLE)
+            c2 = rng.randrange(ncols)
+            while specs[c2][0] == TYPE_DATETIME or specs[c][0] == TYPE_DATETIME:
+                c = rng.randrange(ncols); c2 = rng.randrange(ncols)
+            arith = rng.choice(["add", "sub", "mul"])
+            lit = rng.randint(-2000, 6000)
+            conjuncts.append(((arith, c, c2),
+                              rng.choice(["<", ">", ">=", "<="]), lit))
         else:
             conjuncts.append((c, rng.choice(["<", ">", "=", "!=", ">=", "<="]),
                               rng.randint(-100, 3000)))
