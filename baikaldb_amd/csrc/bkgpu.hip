@@ -291,6 +291,48 @@ __device__ __forceinline__ uint64_t enc_value(const DevCol& c, int64_t r) {
     }
 }
 
+/* expression-capable aggregate input (BkAggSpec.col2/arith): value in the
+ * agg_in_type domain; NULL if either operand NULL (agg_fn_call.cpp input
+ * cast semantics) */
+struct AggIn { bool valid; int64_t i; double d; };
+__device__ __forceinline__ AggIn agg_input(const DevCols& cols,
+                                           const BkAggSpec& as,
+                                           int32_t in_type, int64_t r) {
+    AggIn o{true, 0, 0.0};
+    const DevCol& c = cols.c[as.col];
+    o.valid = cell_valid(c, r);
+    if (as.arith) {
+        const DevCol& c2 = cols.c[as.col2];
+        o.valid = o.valid && cell_valid(c2, r);
+        if (!o.valid) return o;
+        if (in_type == BK_DOUBLE) {
+            double a = cell_f64(c, r), b = cell_f64(c2, r);
+            o.d = as.arith == BK_ARITH_ADD ? a + b
+                  : as.arith == BK_ARITH_SUB ? a - b : a * b;
+        } else {
+            uint64_t a = (uint64_t)cell_i64(c, r);
+            uint64_t b = (uint64_t)cell_i64(c2, r);
+            o.i = (int64_t)(as.arith == BK_ARITH_ADD ? a + b
+                  : as.arith == BK_ARITH_SUB ? a - b : a * b);
+            o.d = (double)o.i;
+        }
+        return o;
+    }
+    if (!o.valid) return o;
+    if (in_type == BK_DOUBLE) o.d = cell_f64(c, r);
+    else { o.i = cell_i64(c, r); o.d = (double)o.i; }
+    return o;
+}
+
+/* order-preserving encoding of an aggregate input (MIN/MAX) */
+__device__ __forceinline__ uint64_t agg_enc(const DevCols& cols,
+                                            const BkAggSpec& as,
+                                            int32_t in_type, const AggIn& v,
+                                            int64_t r) {
+    if (!as.arith) return enc_value(cols.c[as.col], r);
+    return in_type == BK_DOUBLE ? bk_enc_f64(v.d) : bk_enc_i64(v.i);
+}
+
 /* spec-driven group-key packing (bk_common.h group_bits/group_base):
  * per key, (enc - enc(base)) packs into `bits` bits; bits == 0 = a full
  * 64-bit word. With <= 2 keys and all-zero bits this degenerates to the
@@ -354,45 +396,46 @@ __device__ __forceinline__ void agg_update_slot(uint64_t* st, const BkQuerySpec&
                 atomicAdd((unsigned long long*)val, 1ull);
                 break;
             case BK_AGG_COUNT: {
-                if (cell_valid(cols.c[as.col], r))
+                if (agg_input(cols, as, q.agg_in_types[a], r).valid)
                     atomicAdd((unsigned long long*)val, 1ull);
                 break;
             }
             case BK_AGG_SUM: {
-                const DevCol& c = cols.c[as.col];
-                if (!cell_valid(c, r)) break;
+                AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                if (!v.valid) break;
                 if (q.agg_in_types[a] == BK_DOUBLE) {
-                    if (LDS) atomic_add_f64_lds(val, ((const double*)c.data)[r]);
-                    else     atomic_add_f64_global(val, ((const double*)c.data)[r]);
+                    if (LDS) atomic_add_f64_lds(val, v.d);
+                    else     atomic_add_f64_global(val, v.d);
                 } else {
                     atomicAdd((unsigned long long*)val,
-                              (unsigned long long)cell_i64(c, r)); /* int64 wrap */
+                              (unsigned long long)v.i); /* int64 wrap */
                 }
                 atomicAdd((unsigned long long*)cnt, 1ull);
                 break;
             }
             case BK_AGG_AVG: {
-                const DevCol& c = cols.c[as.col];
-                if (!cell_valid(c, r)) break;
-                double v = cell_f64(c, r);
-                if (LDS) atomic_add_f64_lds(val, v);
-                else     atomic_add_f64_global(val, v);
+                AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                if (!v.valid) break;
+                if (LDS) atomic_add_f64_lds(val, v.d);
+                else     atomic_add_f64_global(val, v.d);
                 atomicAdd((unsigned long long*)cnt, 1ull);
                 break;
             }
             case BK_AGG_MIN: {
-                const DevCol& c = cols.c[as.col];
-                if (!cell_valid(c, r)) break;
+                AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                if (!v.valid) break;
                 atomicMax((unsigned long long*)val,
-                          (unsigned long long)(~enc_value(c, r)));
+                          (unsigned long long)(~agg_enc(cols, as,
+                                                        q.agg_in_types[a], v, r)));
                 atomicAdd((unsigned long long*)cnt, 1ull);
                 break;
             }
             case BK_AGG_MAX: {
-                const DevCol& c = cols.c[as.col];
-                if (!cell_valid(c, r)) break;
+                AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                if (!v.valid) break;
                 atomicMax((unsigned long long*)val,
-                          (unsigned long long)enc_value(c, r));
+                          (unsigned long long)agg_enc(cols, as,
+                                                      q.agg_in_types[a], v, r));
                 atomicAdd((unsigned long long*)cnt, 1ull);
                 break;
             }
@@ -651,35 +694,36 @@ k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_
             switch (as.agg_type) {
                 case BK_AGG_COUNT_STAR: acc_v[a]++; break;
                 case BK_AGG_COUNT:
-                    if (cell_valid(cols.c[as.col], r)) acc_v[a]++;
+                    if (agg_input(cols, as, q.agg_in_types[a], r).valid)
+                        acc_v[a]++;
                     break;
                 case BK_AGG_SUM: {
-                    const DevCol& c = cols.c[as.col];
-                    if (!cell_valid(c, r)) break;
-                    if (q.agg_in_types[a] == BK_DOUBLE) acc_d[a] += ((const double*)c.data)[r];
-                    else acc_v[a] += (uint64_t)cell_i64(c, r);
+                    AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                    if (!v.valid) break;
+                    if (q.agg_in_types[a] == BK_DOUBLE) acc_d[a] += v.d;
+                    else acc_v[a] += (uint64_t)v.i;
                     acc_c[a]++;
                     break;
                 }
                 case BK_AGG_AVG: {
-                    const DevCol& c = cols.c[as.col];
-                    if (!cell_valid(c, r)) break;
-                    acc_d[a] += cell_f64(c, r);
+                    AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                    if (!v.valid) break;
+                    acc_d[a] += v.d;
                     acc_c[a]++;
                     break;
                 }
                 case BK_AGG_MIN: {
-                    const DevCol& c = cols.c[as.col];
-                    if (!cell_valid(c, r)) break;
-                    uint64_t e = ~enc_value(c, r);
+                    AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                    if (!v.valid) break;
+                    uint64_t e = ~agg_enc(cols, as, q.agg_in_types[a], v, r);
                     if (e > acc_v[a]) acc_v[a] = e;
                     acc_c[a]++;
                     break;
                 }
                 case BK_AGG_MAX: {
-                    const DevCol& c = cols.c[as.col];
-                    if (!cell_valid(c, r)) break;
-                    uint64_t e = enc_value(c, r);
+                    AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                    if (!v.valid) break;
+                    uint64_t e = agg_enc(cols, as, q.agg_in_types[a], v, r);
                     if (e > acc_v[a]) acc_v[a] = e;
                     acc_c[a]++;
                     break;
@@ -984,28 +1028,21 @@ __device__ __forceinline__ void build_record(
     for (int32_t a = 0; a < q.n_aggs; a++) {
         if (lay.val_word[a] < 0) continue;
         const BkAggSpec& as = q.aggs[a];
-        const DevCol& c = cols.c[as.col];
-        int valid = cell_valid(c, r);
-        if (valid) meta |= (uint64_t)1 << (8 + a);
+        AggIn v = agg_input(cols, q.aggs[a], q.agg_in_types[a], r);
+        if (v.valid) meta |= (uint64_t)1 << (8 + a);
         uint64_t w = 0;
-        if (valid) {
+        if (v.valid) {
             switch (as.agg_type) {
                 case BK_AGG_SUM:
-                    if (q.agg_in_types[a] == BK_DOUBLE) {
-                        double d = ((const double*)c.data)[r];
-                        memcpy(&w, &d, 8);
-                    } else {
-                        w = (uint64_t)cell_i64(c, r);
-                    }
+                    if (q.agg_in_types[a] == BK_DOUBLE) memcpy(&w, &v.d, 8);
+                    else w = (uint64_t)v.i;
                     break;
-                case BK_AGG_AVG: {
-                    double d = cell_f64(c, r);
-                    memcpy(&w, &d, 8);
+                case BK_AGG_AVG:
+                    memcpy(&w, &v.d, 8);
                     break;
-                }
                 case BK_AGG_MIN:
                 case BK_AGG_MAX:
-                    w = enc_value(c, r);
+                    w = agg_enc(cols, as, q.agg_in_types[a], v, r);
                     break;
                 default: break;
             }
@@ -1016,7 +1053,7 @@ __device__ __forceinline__ void build_record(
         /* COUNT(col) validity for aggs without a val word */
         for (int32_t a = 0; a < q.n_aggs; a++) {
             if (lay.val_word[a] >= 0 || q.aggs[a].col < 0) continue;
-            if (cell_valid(cols.c[q.aggs[a].col], r))
+            if (agg_input(cols, q.aggs[a], q.agg_in_types[a], r).valid)
                 meta |= (uint64_t)1 << (8 + a);
         }
         regs[lay.meta_word] = meta;
@@ -2020,6 +2057,8 @@ static int build_rec_layout(BkgTable* t, const BkQuerySpec* q, RecLayout* lay) {
     for (int a = 0; a < q->n_aggs; a++) {
         int col = q->aggs[a].col;
         if (col >= 0 && t->valid[col]) agg_meta = true;
+        int col2 = q->aggs[a].col2;
+        if (q->aggs[a].arith && col2 >= 0 && t->valid[col2]) agg_meta = true;
     }
     /* FUSED narrow-key mode (adaptive, from cached column stats): when the
      * first key's valid-cell encoding span fits 32 bits, the second key (if

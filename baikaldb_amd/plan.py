@@ -36,7 +36,8 @@ class BkConjunct(C.Structure):
 
 
 class BkAggSpec(C.Structure):
-    _fields_ = [("agg_type", C.c_int32), ("col", C.c_int32)]
+    _fields_ = [("agg_type", C.c_int32), ("col", C.c_int32),
+                ("col2", C.c_int32), ("arith", C.c_int32)]
 
 
 _WINFNS = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4,
@@ -155,6 +156,23 @@ class QueryPlan:
         q.n_aggs = len(self.aggs)
         for i, (name, col) in enumerate(self.aggs):
             q.aggs[i].agg_type = _AGGS[name] if isinstance(name, str) else name
+            q.aggs[i].col2 = -1
+            if isinstance(col, tuple):
+                # expression input: ("add"|"sub"|"mul", a, b) — the domain is
+                # DOUBLE iff either operand is DOUBLE (AggFnCall input cast,
+                # agg_fn_call.cpp:496-555)
+                arith, a_c, b_c = _ARITH[col[0]], col[1], col[2]
+                if (self.col_types[a_c] == TYPE_STRING or
+                        self.col_types[b_c] == TYPE_STRING):
+                    raise ValueError("arith agg input needs numeric columns")
+                q.aggs[i].arith = arith
+                q.aggs[i].col = a_c
+                q.aggs[i].col2 = b_c
+                q.agg_in_types[i] = (TYPE_DOUBLE
+                                     if TYPE_DOUBLE in (self.col_types[a_c],
+                                                        self.col_types[b_c])
+                                     else TYPE_INT64)
+                continue
             q.aggs[i].col = col
             q.agg_in_types[i] = self.col_types[col] if col >= 0 else TYPE_INT64
         return q

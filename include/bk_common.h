@@ -184,6 +184,13 @@ typedef enum BkArith {
 typedef struct BkAggSpec {
     int32_t agg_type;  /* BkAggType */
     int32_t col;       /* input column, -1 for COUNT_STAR */
+    int32_t col2;      /* expression input (col ARITH col2): -1 = plain
+                          column. agg_in_types[] carries the result domain
+                          (DOUBLE iff either operand is DOUBLE — the
+                          reference casts AggFnCall inputs the same way,
+                          agg_fn_call.cpp:496-555). NULL if either operand
+                          NULL. */
+    int32_t arith;     /* BkArith */
 } BkAggSpec;
 
 /* ---- window functions (reference src/expr/window_fn_call.cpp:20-38 name

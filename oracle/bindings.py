@@ -34,7 +34,8 @@ class BkConjunct(C.Structure):
 
 
 class BkAggSpec(C.Structure):
-    _fields_ = [("agg_type", C.c_int32), ("col", C.c_int32)]
+    _fields_ = [("agg_type", C.c_int32), ("col", C.c_int32),
+                ("col2", C.c_int32), ("arith", C.c_int32)]
 
 
 class BkWindowFn(C.Structure):
@@ -123,6 +124,18 @@ def make_query(conjuncts=(), group=(), aggs=(), col_types=None,
     q.n_aggs = len(aggs)
     for i, (at, col) in enumerate(aggs):
         q.aggs[i].agg_type = at
+        q.aggs[i].col2 = -1
+        if isinstance(col, tuple):
+            # expression input: (arith_code, a, b); DOUBLE domain iff either
+            # operand is DOUBLE (mirrors plan.py to_spec)
+            q.aggs[i].arith, a_c, b_c = col
+            q.aggs[i].col = a_c
+            q.aggs[i].col2 = b_c
+            q.agg_in_types[i] = (TYPE_DOUBLE
+                                 if TYPE_DOUBLE in (col_types[a_c],
+                                                    col_types[b_c])
+                                 else TYPE_INT64)
+            continue
         q.aggs[i].col = col
         q.agg_in_types[i] = col_types[col] if col >= 0 else TYPE_INT64
     return q

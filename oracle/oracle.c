@@ -432,7 +432,27 @@ static void* orc_agg_worker(void* arg) {
             if (as->col >= 0) {
                 const OrcCol* c = &cols[as->col];
                 in_valid = cell_is_valid(c, r);
-                if (in_valid) {
+                if (as->arith) {
+                    /* expression input (col ARITH col2): NULL if either
+                     * operand NULL; computed in the agg_in_type domain */
+                    const OrcCol* c2 = &cols[as->col2];
+                    in_valid = in_valid && cell_is_valid(c2, r);
+                    if (in_valid) {
+                        if (vtype == BK_DOUBLE) {
+                            double av = cell_f64_cast(c, r);
+                            double bv = cell_f64_cast(c2, r);
+                            vd = as->arith == BK_ARITH_ADD ? av + bv
+                                 : as->arith == BK_ARITH_SUB ? av - bv
+                                 : av * bv;
+                        } else {
+                            uint64_t av = (uint64_t)cell_i64(c, r);
+                            uint64_t bv = (uint64_t)cell_i64(c2, r);
+                            vi = (int64_t)(as->arith == BK_ARITH_ADD ? av + bv
+                                 : as->arith == BK_ARITH_SUB ? av - bv
+                                 : av * bv);
+                        }
+                    }
+                } else if (in_valid) {
                     if (c->type == BK_DOUBLE) vd = ((double*)c->data)[r];
                     else vi = cell_i64(c, r);
                 }

@@ -87,7 +87,12 @@ def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
         oconj.append((col, ops[op], ct, lit, fn, og, col2, arith))
     from baikaldb_amd.plan import _FNS
     ogroup = [(_FNS[g[0]], g[1]) if isinstance(g, tuple) else g for g in group]
-    q = make_query(oconj, ogroup, [(aggmap[a], c) for a, c in aggs], col_types,
+    oaggs = []
+    for a, c in aggs:
+        if isinstance(c, tuple):
+            c = (_ARITH[c[0]], c[1], c[2])
+        oaggs.append((aggmap[a], c))
+    q = make_query(oconj, ogroup, oaggs, col_types,
                    group_bits=group_bits, group_base=group_base)
     exp = orc.filter_agg(cols, valids, col_types, q, nthreads=nthreads,
                          dict_seed=seed)
@@ -101,6 +106,9 @@ def assert_parity(got, exp, aggs, col_types):
     assert np.array_equal(got["enc"], exp["enc"])
     assert np.array_equal(got["agg_has"], exp["agg_has"])
     for a, (name, col) in enumerate(aggs):
+        if isinstance(col, tuple):   # expression input: the compute DOMAIN
+            col = (col[1] if col_types[col[1]] == TYPE_DOUBLE else
+                   col[2] if col_types[col[2]] == TYPE_DOUBLE else col[1])
         is_double = col >= 0 and col_types[col] == TYPE_DOUBLE
         if name in ("count_star", "count") or not is_double:
             assert np.array_equal(got["agg_i"][a], exp["agg_i"][a]), f"agg {a} {name}"
@@ -584,3 +592,24 @@ def test_arith_mixed_double(eng, orc):
     aggs = [("count_star", -1), ("avg", 1)]
     got, exp = run_both(eng, orc, specs, 200_000, conj, [], aggs)
     assert_parity(got, exp, aggs, [s[0] for s in specs])
+
+
+@pytest.mark.gpu
+def test_expr_agg_inputs(eng, orc):
+    """expression-valued aggregate inputs (SUM(a*b), AVG(a+d), MIN(a-b)):
+    int64 sums wrap, mixed domains compute in f64, NULL operand skips the
+    row (AggFnCall input semantics, agg_fn_call.cpp:496-555)."""
+    specs = [(TYPE_INT64, D_UNI, 0, 1000, 0),
+             (TYPE_INT64, D_UNI, 0, 1000, 150_000),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 0),
+             (TYPE_INT64, D_UNI, 0, 50, 0)]
+    aggs = [("count_star", -1),
+            ("sum", ("mul", 0, 1)),
+            ("avg", ("add", 0, 2)),
+            ("min", ("sub", 0, 1)),
+            ("count", ("add", 1, 1)),
+            ("max", ("mul", 2, 2))]
+    got, exp = run_both(eng, orc, specs, 250_000, [(3, "<", 40)], [3], aggs)
+    names = [("count_star", -1), ("sum", 0), ("avg", 2), ("min", 0),
+             ("count", 1), ("max", 2)]
+    assert_parity(got, exp, names, [s[0] for s in specs])

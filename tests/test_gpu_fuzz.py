@@ -79,12 +79,22 @@ def random_case(rng):
     group = rng.sample(range(ncols), rng.randint(0, min(2, ncols)))
     group = [(rng.choice(_DT_FNS), c) if specs[c][0] == TYPE_DATETIME
              and rng.random() < 0.7 else c for c in group]
+    # ~20%: one aggregate over a two-column arithmetic expression
+    arith_agg = None
+    num_cols = [c for c in range(ncols)
+                if specs[c][0] in (TYPE_INT64, TYPE_DOUBLE)]
+    if len(num_cols) >= 2 and rng.random() < 0.2:
+        a_c, b_c = rng.sample(num_cols, 2)
+        arith_agg = (rng.choice(["sum", "avg", "min", "max", "count"]),
+                     (rng.choice(["add", "sub", "mul"]), a_c, b_c))
     # aggs
     aggs = [("count_star", -1)]
     for _ in range(rng.randint(0, 4)):
         c = rng.randrange(ncols)
         name = rng.choice(["count", "sum", "avg", "min", "max"])
         aggs.append((name, c))  # string min/max OK: dict is order-preserving
+    if arith_agg is not None:
+        aggs.append(arith_agg)
     return specs, conjuncts, group, aggs
 
 
