@@ -359,8 +359,18 @@ public:
         q.n_aggs = _desc.n_aggs;
         for (int a = 0; a < q.n_aggs; a++) {
             q.aggs[a] = _desc.aggs[a];
-            q.agg_in_types[a] = _desc.aggs[a].col >= 0
-                ? bkgpu_table_col_type(t, _desc.aggs[a].col) : BK_INT64;
+            const BkAggSpec& as = _desc.aggs[a];
+            if (as.arith && as.col2 >= 0) {
+                /* expression input: DOUBLE domain iff either operand is
+                 * DOUBLE (AggFnCall input cast, agg_fn_call.cpp:496-555) */
+                int t1 = bkgpu_table_col_type(t, as.col);
+                int t2 = bkgpu_table_col_type(t, as.col2);
+                q.agg_in_types[a] = (t1 == BK_DOUBLE || t2 == BK_DOUBLE)
+                                        ? BK_DOUBLE : BK_INT64;
+            } else {
+                q.agg_in_types[a] = as.col >= 0
+                    ? bkgpu_table_col_type(t, as.col) : BK_INT64;
+            }
         }
         _q = q;
         int64_t expected = _desc.expected_groups > 0 ? _desc.expected_groups : 65536;
