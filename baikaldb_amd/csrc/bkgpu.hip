@@ -540,8 +540,12 @@ __device__ uint64_t* gtable_claim(uint64_t* table, uint64_t mask, int stride,
 __device__ uint64_t* ltable_claim(uint64_t* ltab, uint32_t lmask, int stride,
                                   uint32_t flag, uint64_t k0, uint64_t k1,
                                   uint32_t* lfill, uint32_t lcap,
-                                  int max_probe = 64) {
-    uint64_t slot = (key_hash(flag, k0, k1) >> 32) & lmask;
+                                  int max_probe = 64, uint32_t salt = 0) {
+    /* salt: slot REPLICATION for low-cardinality tables — each lane class
+     * lands the same key in a different slot, cutting same-address LDS
+     * atomic serialization R-fold; the flush's additive global merge
+     * recombines replicas */
+    uint64_t slot = ((key_hash(flag, k0, k1) >> 32) + salt * 0x9E3779B9u) & lmask;
     for (int it = 0; it < max_probe; ++it) {
         uint64_t* s = ltab + slot * (uint64_t)stride;
         uint32_t* statep = (uint32_t*)s;
@@ -605,11 +609,11 @@ __global__ void k_generate(DevCols cols, int ncols, DevSpecs specs,
 
 /* fused filter + grouped hash aggregate.
  * dynamic LDS: LDS_SLOTS * stride u64 words (per-workgroup pre-agg table). */
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(1024)
 k_filter_agg_group(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
                    uint64_t* gtable, uint64_t gmask, uint64_t fill_cap,
                    uint64_t* fill, uint64_t* rows_passed, uint32_t* err,
-                   uint32_t lds_slots) {
+                   uint32_t lds_slots, uint32_t rep_mask) {
     /* ALL LDS in one dynamic carve (guide G17: a static __shared__ ahead of
      * the dynamic region misaligns the u64 table -> 64-cycle replays):
      *   [ lds_slots*stride table ][ laux[0]=rows_passed ][ laux[1].lo=fill ] */
@@ -636,7 +640,9 @@ k_filter_agg_group(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_e
         KeyPack kp = pack_group_keys(cols, q, r);
         uint32_t flag = kp.flag;
         uint64_t k0 = kp.k0, k1 = kp.k1;
-        uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1, lfill, lcap);
+        uint64_t* slot = ltable_claim(ltab, lmask, stride, flag, k0, k1,
+                                      lfill, lcap, 64,
+                                      threadIdx.x & rep_mask);
         if (slot) {
             agg_update_slot<true>(slot, q, cols, r);
         } else {
@@ -2608,10 +2614,13 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
         } else {
             static const char* NAMES1[] = {"fused_agg"};
             EvTimer tm;
-            uint32_t lds_slots = 512;
-            /* keep >=2 blocks/CU: LDS per block <= 64 KiB */
-            while ((size_t)lds_slots * stride * 8 > 63 * 1024) lds_slots >>= 1;
+            uint32_t lds_slots = 2048;
+            size_t fcap = 135 * 1024;
+            if (const char* e = getenv("BK_FUSED_LDS_KB"))
+                fcap = (size_t)atoi(e) * 1024;
+            while ((size_t)lds_slots * stride * 8 > fcap) lds_slots >>= 1;
             size_t lds_bytes = ((size_t)lds_slots * stride + 2) * 8;
+            if (q->n_group > 0) threads = 1024;   /* 16 waves at 1 block/CU */
             if (q->n_group == 0) {
                 /* pre-initialize slot 0 as the single group (state=2, flag 0):
                  * mirrors agg_node.cpp:490-505's always-present row */
@@ -2626,12 +2635,21 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
                                    0, 0, dc, *q, row_begin, row_end, o->table,
                                    o->ctrs + 1);
             } else {
+                /* low-cardinality: 1024-thread blocks (16 waves/CU at the
+                 * 135 KB table) took the 300-group 1e9-row shape from 28.7
+                 * to 23.3 ms. Per-lane slot REPLICATION is measured DEAD
+                 * (R=2 +11%, R=4 +40%: replica probe chains cost more than
+                 * the contention they remove); BK_FUSED_REP re-enables it
+                 * for experiments. */
+                uint32_t rep = 1;
+                if (const char* e = getenv("BK_FUSED_REP")) rep = atoi(e);
                 tm.record();
                 hipLaunchKernelGGL(k_filter_agg_group, dim3(blocks), dim3(threads),
                                    lds_bytes, 0,
                                    dc, *q, row_begin, row_end, o->table,
                                    o->nslots - 1, (o->nslots * 7) / 8,
-                                   o->ctrs, o->ctrs + 1, o->err, lds_slots);
+                                   o->ctrs, o->ctrs + 1, o->err, lds_slots,
+                                   rep - 1);
             }
             tm.record();
             hipError_t lerr = hipGetLastError();
