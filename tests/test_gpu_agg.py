@@ -613,3 +613,41 @@ def test_expr_agg_inputs(eng, orc):
     names = [("count_star", -1), ("sum", 0), ("avg", 2), ("min", 0),
              ("count", 1), ("max", 2)]
     assert_parity(got, exp, names, [s[0] for s in specs])
+
+
+@pytest.mark.gpu
+def test_fused_low_cardinality_full_mix(eng, orc):
+    """the fused <=512-group kernel (k_filter_agg_group) under the full
+    feature mix: nullable dict/double cols, OR clause, arith predicate and
+    expression agg input — vs the oracle, and vs the sorted path."""
+    specs = [(TYPE_INT64, D_UNI, 0, 300, 120_000),   # group key, nullable
+             (TYPE_INT64, D_UNI, 0, 1000, 0),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 150_000),
+             (TYPE_STRING, D_DICT, 40, 0, 0),
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 0)]
+    conj = [(4, "<", 1 << 30), (1, ">", 100, 1), (3, "!=", 7, 1),
+            (("add", 1, 4), ">", 500)]
+    aggs = [("count_star", -1), ("sum", ("mul", 1, 1)), ("avg", 2),
+            ("min", 3), ("max", 1), ("count", 2)]
+    got, exp = run_both(eng, orc, specs, 400_000, conj, [0], aggs,
+                        expected_groups=400)   # <= 512: fused kernel
+    names = [("count_star", -1), ("sum", 1), ("avg", 2), ("min", 3),
+             ("max", 1), ("count", 2)]
+    assert_parity(got, exp, names, [s[0] for s in specs])
+    # cross-path: the sorted engine must agree bit-for-bit on int outputs
+    from baikaldb_amd import QueryPlan
+    t = eng.create_table(specs, 400_000)
+    try:
+        eng.generate(t, SEED)
+        plan = QueryPlan(t.col_types, conjuncts=conj, group=[0], aggs=aggs)
+        r = eng.filter_agg_sorted(t, plan)
+        try:
+            gs = r.fetch(sorted=True)
+        finally:
+            r.free()
+    finally:
+        t.free()
+    assert gs["ngroups"] == got["ngroups"]
+    assert np.array_equal(gs["enc"], got["enc"])
+    for i in (0, 1, 4, 5):
+        assert np.array_equal(gs["agg_i"][i], got["agg_i"][i])
