@@ -381,6 +381,45 @@ __device__ __forceinline__ void atomic_add_f64_lds(uint64_t* addr, double v) {
     unsafeAtomicAdd((double*)addr, v);   /* ds_add_f64 */
 }
 
+__device__ __forceinline__ void dedup_flush_run_arrays(
+        uint64_t* slot, const BkQuerySpec& q, const uint64_t* accv,
+        const uint64_t* accc, const double* accd, int na) {
+    for (int32_t a = 0; a < na; a++) {
+        if (a >= q.n_aggs) break;
+        uint64_t* val = slot + SLOT_HDR + 2 * a;
+        uint64_t* cnt = val + 1;
+        int at = q.aggs[a].agg_type;
+        switch (at) {
+            case BK_AGG_COUNT_STAR:
+            case BK_AGG_COUNT:
+                if (accv[a])
+                    atomicAdd((unsigned long long*)val,
+                              (unsigned long long)accv[a]);
+                break;
+            case BK_AGG_MIN:
+            case BK_AGG_MAX:
+                if (accc[a]) {
+                    atomicMax((unsigned long long*)val,
+                              (unsigned long long)accv[a]);
+                    atomicAdd((unsigned long long*)cnt,
+                              (unsigned long long)accc[a]);
+                }
+                break;
+            default:  /* SUM / AVG */
+                if (accc[a]) {
+                    if (q.agg_in_types[a] == BK_DOUBLE || at == BK_AGG_AVG)
+                        atomic_add_f64_global(val, accd[a]);
+                    else
+                        atomicAdd((unsigned long long*)val,
+                                  (unsigned long long)accv[a]);
+                    atomicAdd((unsigned long long*)cnt,
+                              (unsigned long long)accc[a]);
+                }
+                break;
+        }
+    }
+}
+
 /* ---- per-agg atomic update into a slot (LDS or global templated) ---- */
 template <bool LDS>
 __device__ __forceinline__ void agg_update_slot(uint64_t* st, const BkQuerySpec& q,
@@ -778,6 +817,160 @@ k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_
                 }
             }
         }
+    }
+}
+
+/* very-low-cardinality GROUP BY (expected <= 32): per-WAVE ballot loop
+ * over the distinct keys present in each 64-row batch — one masked
+ * full-wave reduction and ONE leader atomic set per distinct key, instead
+ * of per-row same-address LDS atomics (which bound k_filter_agg_group at
+ * ~0.9 TB/s). NA = agg-count bound (4/8) so accumulators stay in
+ * registers. */
+template <int NA>
+__global__ void __launch_bounds__(512)
+k_filter_agg_wcomb(DevCols cols, BkQuerySpec q, int64_t row_begin,
+                   int64_t row_end, uint64_t* gtable, uint64_t gmask,
+                   uint64_t fill_cap, uint64_t* fill, uint64_t* rows_passed,
+                   uint32_t* err, uint32_t lds_slots) {
+    extern __shared__ __attribute__((aligned(16))) uint64_t ltab[];
+    const int stride = SLOT_HDR + 2 * q.n_aggs;
+    uint64_t* laux = ltab + (size_t)lds_slots * stride;
+    uint32_t* lfill = (uint32_t*)&laux[1];
+    for (uint32_t w = threadIdx.x; w < lds_slots * (uint32_t)stride + 2;
+         w += blockDim.x)
+        ltab[w] = 0;
+    __syncthreads();
+    const uint32_t lmask = lds_slots - 1;
+    const uint32_t lcap = (lds_slots * 3u) / 4u;
+    const int lane = threadIdx.x & 63;
+    int64_t my_passed = 0;
+    int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         ; r += gstride) {
+        bool live = r < row_end;
+        if (!__any(live)) break;
+        bool pass = live && row_passes(cols, q, r);
+        if (pass) my_passed++;
+        KeyPack kp{0, 0, 0};
+        if (pass) kp = pack_group_keys(cols, q, r);
+        uint64_t pending = __ballot(pass);
+        while (pending) {
+            int leader = __ffsll((unsigned long long)pending) - 1;
+            uint64_t lk0 = __shfl((unsigned long long)kp.k0, leader, 64);
+            uint64_t lk1 = __shfl((unsigned long long)kp.k1, leader, 64);
+            uint32_t lf = (uint32_t)__shfl((int)kp.flag, leader, 64);
+            bool mine = pass && kp.k0 == lk0 && kp.k1 == lk1 && kp.flag == lf;
+            uint64_t mmask = __ballot(mine);
+            uint64_t* slot = nullptr;
+            if (lane == leader) {
+                slot = ltable_claim(ltab, lmask, stride, lf, lk0, lk1,
+                                    lfill, lcap);
+                if (!slot)
+                    slot = gtable_claim(gtable, gmask, stride, lf, lk0, lk1,
+                                        fill, fill_cap, err);
+            }
+            /* per agg: contributions recomputed from L1-hot column loads
+             * (register ARRAYS held across this loop spill to 150-210 B/lane
+             * scratch that gets re-read every key — measured 10x slower),
+             * one masked full-wave reduce, leader flushes inline */
+            #pragma unroll
+            for (int32_t a = 0; a < NA; a++) {
+                if (a >= q.n_aggs) break;
+                const BkAggSpec& as = q.aggs[a];
+                int at = as.agg_type;
+                bool im = (at == BK_AGG_MIN || at == BK_AGG_MAX);
+                uint64_t tv = 0;
+                uint64_t tc = 0;
+                double td = 0.0;
+                if (mine) {
+                    switch (at) {
+                        case BK_AGG_COUNT_STAR: tv = 1; break;
+                        case BK_AGG_COUNT:
+                            tv = agg_input(cols, as, q.agg_in_types[a], r)
+                                     .valid ? 1 : 0;
+                            break;
+                        case BK_AGG_SUM: {
+                            AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                            if (!v.valid) break;
+                            if (q.agg_in_types[a] == BK_DOUBLE) td = v.d;
+                            else tv = (uint64_t)v.i;
+                            tc = 1;
+                            break;
+                        }
+                        case BK_AGG_AVG: {
+                            AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                            if (!v.valid) break;
+                            td = v.d;
+                            tc = 1;
+                            break;
+                        }
+                        case BK_AGG_MIN: {
+                            AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                            if (!v.valid) break;
+                            tv = ~agg_enc(cols, as, q.agg_in_types[a], v, r);
+                            tc = 1;
+                            break;
+                        }
+                        case BK_AGG_MAX: {
+                            AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                            if (!v.valid) break;
+                            tv = agg_enc(cols, as, q.agg_in_types[a], v, r);
+                            tc = 1;
+                            break;
+                        }
+                        default: break;
+                    }
+                }
+                #pragma unroll
+                for (int off = 32; off > 0; off >>= 1) {
+                    uint64_t ov = __shfl_xor((unsigned long long)tv, off, 64);
+                    tc += __shfl_xor((unsigned long long)tc, off, 64);
+                    td += __shfl_xor(td, off, 64);
+                    tv = im ? (tv > ov ? tv : ov) : tv + ov;
+                }
+                if (lane == leader && slot) {
+                    uint64_t* val = slot + SLOT_HDR + 2 * a;
+                    uint64_t* cnt = val + 1;
+                    if (at == BK_AGG_COUNT_STAR || at == BK_AGG_COUNT) {
+                        if (tv) atomicAdd((unsigned long long*)val,
+                                          (unsigned long long)tv);
+                    } else if (im) {
+                        if (tc) {
+                            atomicMax((unsigned long long*)val,
+                                      (unsigned long long)tv);
+                            atomicAdd((unsigned long long*)cnt,
+                                      (unsigned long long)tc);
+                        }
+                    } else if (tc) {
+                        if (q.agg_in_types[a] == BK_DOUBLE || at == BK_AGG_AVG)
+                            atomic_add_f64_global(val, td);
+                        else
+                            atomicAdd((unsigned long long*)val,
+                                      (unsigned long long)tv);
+                        atomicAdd((unsigned long long*)cnt,
+                                  (unsigned long long)tc);
+                    }
+                }
+            }
+            pending &= ~mmask;
+        }
+    }
+    /* rows_passed + final LDS flush (same as k_filter_agg_group) */
+    long long w = my_passed;
+    for (int off = 32; off > 0; off >>= 1) w += __shfl_down(w, off, 64);
+    if ((threadIdx.x & 63) == 0)
+        atomicAdd((unsigned long long*)&laux[0], (unsigned long long)w);
+    __syncthreads();
+    if (threadIdx.x == 0)
+        atomicAdd((unsigned long long*)rows_passed, (unsigned long long)laux[0]);
+    for (uint32_t sl = threadIdx.x; sl < lds_slots; sl += blockDim.x) {
+        uint64_t* s = ltab + (uint64_t)sl * stride;
+        uint32_t st = ((uint32_t*)s)[0];
+        if (st != 2u) continue;
+        uint64_t* g = gtable_claim(gtable, gmask, stride, ((uint32_t*)s)[1],
+                                   s[1], s[2], fill, fill_cap, err);
+        if (!g) break;
+        agg_merge_slot<false>(g, s + SLOT_HDR, q);
     }
 }
 
@@ -2640,16 +2833,36 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
                  * to 23.3 ms. Per-lane slot REPLICATION is measured DEAD
                  * (R=2 +11%, R=4 +40%: replica probe chains cost more than
                  * the contention they remove); BK_FUSED_REP re-enables it
-                 * for experiments. */
+                 * for experiments. BK_WCOMB_MAX=N opts tiny-cardinality
+                 * queries (<= N expected groups) into the WAVE-COMBINE
+                 * variant (one masked wave reduction + one leader atomic per
+                 * distinct key per 64-row batch) — default OFF: the claim +
+                 * shuffle state held across the ballot loop spills 180-240
+                 * B/lane scratch and measured 24x SLOWER than the plain
+                 * same-address LDS-atomic kernel. */
                 uint32_t rep = 1;
                 if (const char* e = getenv("BK_FUSED_REP")) rep = atoi(e);
+                int64_t wmax = 0;
+                if (const char* e = getenv("BK_WCOMB_MAX")) wmax = atoll(e);
                 tm.record();
-                hipLaunchKernelGGL(k_filter_agg_group, dim3(blocks), dim3(threads),
-                                   lds_bytes, 0,
-                                   dc, *q, row_begin, row_end, o->table,
-                                   o->nslots - 1, (o->nslots * 7) / 8,
-                                   o->ctrs, o->ctrs + 1, o->err, lds_slots,
-                                   rep - 1);
+                if (expected_groups <= wmax) {
+                    uint32_t wslots = 512;
+                    while ((size_t)wslots * stride * 8 > 60 * 1024) wslots >>= 1;
+                    size_t wbytes = ((size_t)wslots * stride + 2) * 8;
+                    auto wfn = q->n_aggs <= 4 ? k_filter_agg_wcomb<4>
+                                              : k_filter_agg_wcomb<8>;
+                    hipLaunchKernelGGL(wfn, dim3(8192), dim3(512), wbytes, 0,
+                                       dc, *q, row_begin, row_end, o->table,
+                                       o->nslots - 1, (o->nslots * 7) / 8,
+                                       o->ctrs, o->ctrs + 1, o->err, wslots);
+                } else {
+                    hipLaunchKernelGGL(k_filter_agg_group, dim3(blocks),
+                                       dim3(threads), lds_bytes, 0,
+                                       dc, *q, row_begin, row_end, o->table,
+                                       o->nslots - 1, (o->nslots * 7) / 8,
+                                       o->ctrs, o->ctrs + 1, o->err, lds_slots,
+                                       rep - 1);
+                }
             }
             tm.record();
             hipError_t lerr = hipGetLastError();

@@ -651,3 +651,33 @@ def test_fused_low_cardinality_full_mix(eng, orc):
     assert np.array_equal(gs["enc"], got["enc"])
     for i in (0, 1, 4, 5):
         assert np.array_equal(gs["agg_i"][i], got["agg_i"][i])
+
+
+@pytest.mark.gpu
+def test_wave_combine_tiny_cardinality(eng, orc):
+    """Opt-in <=32-group wave-combine kernel (BK_WCOMB_MAX, default off —
+    measured-dead lever) vs oracle and vs the default
+    generic fused kernel, full feature mix incl nullable key."""
+    import os
+    specs = [(TYPE_INT64, D_UNI, 0, 12, 150_000),
+             (TYPE_INT64, D_UNI, 0, 1000, 0),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 120_000),
+             (TYPE_STRING, D_DICT, 50, 0, 0)]
+    conj = [(1, ">", 50)]
+    aggs = [("count_star", -1), ("sum", 1), ("avg", 2), ("min", 3),
+            ("max", ("add", 1, 1)), ("count", 2)]
+    os.environ["BK_WCOMB_MAX"] = "32"
+    try:
+        got, exp = run_both(eng, orc, specs, 300_000, conj, [0], aggs,
+                            expected_groups=16)
+    finally:
+        os.environ.pop("BK_WCOMB_MAX", None)
+    names = [("count_star", -1), ("sum", 1), ("avg", 2), ("min", 3),
+             ("max", 1), ("count", 2)]
+    assert_parity(got, exp, names, [s[0] for s in specs])
+    got2, _ = run_both(eng, orc, specs, 300_000, conj, [0], aggs,
+                       expected_groups=16)
+    assert got2["ngroups"] == got["ngroups"]
+    assert np.array_equal(got2["enc"], got["enc"])
+    for i in (0, 1, 4, 5):
+        assert np.array_equal(got2["agg_i"][i], got["agg_i"][i])
