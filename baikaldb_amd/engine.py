@@ -98,6 +98,10 @@ def _load():
                                  C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
     lib.bkgpu_table_col_type.restype = C.c_int32
     lib.bkgpu_table_col_type.argtypes = [C.c_void_p, C.c_int]
+    lib.bkgpu_table_compact.restype = C.c_int
+    lib.bkgpu_table_compact.argtypes = [C.c_void_p, C.c_int]
+    lib.bkgpu_table_col_width.restype = C.c_int
+    lib.bkgpu_table_col_width.argtypes = [C.c_void_p, C.c_int]
     lib.bkgpu_upload_bytes.restype = C.c_void_p
     lib.bkgpu_upload_bytes.argtypes = [C.c_void_p, C.c_int64]
     lib.bkgpu_free_ptr.argtypes = [C.c_void_p]
@@ -213,9 +217,24 @@ class GpuEngine:
             raise RuntimeError(f"table_create: {self.lib.bkgpu_last_error().decode()}")
         return GpuTable(self, h, types, nrows)
 
-    def generate(self, table, seed, row_begin=0):
+    def generate(self, table, seed, row_begin=0, compact=None):
         self._check(self.lib.bkgpu_table_generate(table.handle, seed, row_begin),
                     "table_generate")
+        # narrow physical column encoding (bkgpu_table_compact): on by
+        # default, BK_NARROW=0 disables, compact=False per call
+        if compact is None:
+            compact = os.environ.get("BK_NARROW", "1") != "0"
+        if compact:
+            self.compact(table)
+
+    def compact(self, table, col=-1):
+        """Narrow integer columns to frame-of-reference u8/u16/u32 storage
+        (value-preserving; see bkgpu_table_compact in include/bkgpu.h)."""
+        self._check(self.lib.bkgpu_table_compact(table.handle, col),
+                    "table_compact")
+
+    def col_width(self, table, col):
+        return self.lib.bkgpu_table_col_width(table.handle, col)
 
     def upload(self, table, col, data, valid=None):
         vptr = valid.ctypes.data_as(C.c_void_p) if valid is not None else None

@@ -58,6 +58,13 @@ int  bkgpu_table_generate(BkgTable* t, uint64_t seed, int64_t row_begin);
 /* Upload host column data (and optional validity bytes) instead. */
 int  bkgpu_table_upload(BkgTable* t, int col, const void* data, const uint8_t* valid);
 int64_t bkgpu_table_nrows(const BkgTable* t);
+/* Narrow the physical storage of integer-typed columns (col = -1: all)
+ * whose all-rows value range fits 1/2/4 bytes to frame-of-reference
+ * deltas — value-preserving (kernels reconstruct exact int64s), cuts the
+ * column's HBM traffic 2-8x. Re-generate/upload transparently re-widens. */
+int  bkgpu_table_compact(BkgTable* t, int col);
+/* Physical bytes/elem of a column after compaction (introspection). */
+int  bkgpu_table_col_width(const BkgTable* t, int col);
 /* Append a derived BK_STRING column: newcode[r] = remap[oldcode[r]] — the
  * engine-side compilation of a unary string scalar fn (upper/lower/substr,
  * src/expr/internal_functions.cpp via fn_manager.cpp:97-137) on a dict

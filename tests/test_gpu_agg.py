@@ -681,3 +681,61 @@ def test_wave_combine_tiny_cardinality(eng, orc):
     assert np.array_equal(got2["enc"], got["enc"])
     for i in (0, 1, 4, 5):
         assert np.array_equal(got2["agg_i"][i], got["agg_i"][i])
+
+
+@pytest.mark.gpu
+def test_narrow_storage_widths_and_parity(eng):
+    """Narrow frame-of-reference column storage (bkgpu_table_compact):
+    chosen widths match the all-rows value ranges, every query result is
+    bit-identical to the same query on wide storage, and re-upload
+    transparently re-widens."""
+    from baikaldb_amd import QueryPlan
+    specs = [(TYPE_INT64, D_UNI, 0, 20, 0),            # range 20 -> u8
+             (TYPE_INT64, D_UNI, 0, 40_000, 0),        # -> u16
+             (TYPE_INT64, D_UNI, 0, 1 << 31, 100_000),  # -> u32, nullable
+             (TYPE_INT64, D_SUM16, 0, 0, 0),           # wide (full range)
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 0),          # double: never narrow
+             (TYPE_STRING, D_DICT, 50, 0, 0)]          # 50 codes -> u8
+    n = 200_000
+    t = eng.create_table(specs, n)
+    tw = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED)                      # auto-compacts
+        eng.generate(tw, SEED, compact=False)      # stays wide
+        assert eng.col_width(t, 0) == 1
+        assert eng.col_width(t, 1) == 2
+        assert eng.col_width(t, 2) == 4
+        assert eng.col_width(t, 4) == 8
+        assert eng.col_width(t, 5) == 1
+        assert all(eng.col_width(tw, c) in (4, 8) for c in range(6))
+        plan = QueryPlan(t.col_types,
+                         conjuncts=[(2, "<", 1 << 30), (0, "!=", 3)],
+                         group=[1, 5],
+                         aggs=[("count_star", -1), ("sum", 0), ("sum", 3),
+                               ("avg", 4), ("min", 2), ("max", 1)])
+        def run(tab):
+            r = eng.filter_agg(tab, plan, expected_groups=1 << 16)
+            try:
+                return r.fetch(sorted=True)
+            finally:
+                r.free()
+        a, b = run(t), run(tw)
+        assert a["rows_passed"] == b["rows_passed"]
+        assert a["ngroups"] == b["ngroups"]
+        assert np.array_equal(a["enc"], b["enc"])
+        for i in range(6):
+            assert np.array_equal(a["agg_i"][i], b["agg_i"][i]), i
+            assert np.array_equal(a["agg_d"][i], b["agg_d"][i]), i
+        # ORDER BY on narrow keys matches wide
+        sa = eng.sort_topk(t, [(1, 1, 1), (2, 0, 1)], 500)
+        sb = eng.sort_topk(tw, [(1, 1, 1), (2, 0, 1)], 500)
+        assert np.array_equal(sa["rowids"], sb["rowids"])
+        # re-upload re-widens and preserves values
+        host = np.arange(n, dtype=np.int64) % 7
+        eng.upload(t, 0, host)
+        assert eng.col_width(t, 0) == 8
+        eng.compact(t, 0)
+        assert eng.col_width(t, 0) == 1
+    finally:
+        t.free()
+        tw.free()
