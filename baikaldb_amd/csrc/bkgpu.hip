@@ -66,12 +66,14 @@ extern "C" const char* bkgpu_last_error(void) { return g_err; }
 
 struct DevCol {
     int32_t type;            /* BkType */
-    int32_t width;           /* PHYSICAL bytes/elem: 8|4 natural, or narrowed
-                              * 4|2|1 (frame-of-reference u32/u16/u8 deltas
-                              * from `base` — bkgpu_table_compact) */
+    uint32_t lshift;         /* log2(PHYSICAL bytes/elem): natural 3 (2 for
+                              * BK_STRING codes) or narrowed 2|1|0 — frame-of-
+                              * reference deltas from `base`
+                              * (bkgpu_table_compact) */
     const void* data;
     const uint8_t* valid;    /* null => all valid */
-    int64_t base;            /* frame-of-reference base (narrow widths only) */
+    int64_t base;            /* frame-of-reference base (0 when wide) */
+    uint64_t mask;           /* (1 << 8*width) - 1; ~0 for width 8 */
 };
 
 struct DevCols {
@@ -101,22 +103,19 @@ __device__ __forceinline__ bool cell_valid(const DevCol& c, int64_t r) {
     return c.valid == nullptr || c.valid[r];
 }
 __device__ __forceinline__ int64_t cell_i64(const DevCol& c, int64_t r) {
-    /* width is wave-uniform (kernarg) — the branches cost scalar ops only;
-     * narrow widths are frame-of-reference deltas from c.base and cut the
-     * column's HBM line traffic 2-8x (bkgpu_table_compact) */
-    if (c.type == BK_STRING) {
-        if (__builtin_expect(c.width == 4, 1))
-            return (int64_t)((const int32_t*)c.data)[r];
-        uint64_t d = c.width == 2 ? ((const uint16_t*)c.data)[r]
-                                  : ((const uint8_t*)c.data)[r];
-        return (int64_t)((uint64_t)c.base + d);
-    }
-    if (__builtin_expect(c.width == 8, 1))
-        return ((const int64_t*)c.data)[r];
-    uint64_t d = c.width == 4 ? ((const uint32_t*)c.data)[r]
-                 : c.width == 2 ? ((const uint16_t*)c.data)[r]
-                                : ((const uint8_t*)c.data)[r];
-    return (int64_t)((uint64_t)c.base + d);
+    /* BRANCHLESS width decode: one (possibly unaligned) 8-byte load at
+     * r << lshift, then base + (raw & mask). gfx950 global loads handle
+     * any alignment in hardware (single global_load_dwordx2); column
+     * buffers carry an 8-byte tail pad for the overread. Width-dependent
+     * branches in this accessor were measured to spill 200-700 B/lane of
+     * scratch in the batched kernels (k_dedup_mat/k_topk_scan) and erase
+     * the narrow-storage win. Wide columns have lshift 3, base 0, mask ~0
+     * (BK_STRING: lshift 2, mask 2^32-1 — codes are non-negative, so the
+     * zero-extend equals the reference's sign-extended int32 codes). */
+    uint64_t raw;
+    __builtin_memcpy(&raw,
+                     (const char*)c.data + ((uint64_t)r << c.lshift), 8);
+    return (int64_t)((uint64_t)c.base + (raw & c.mask));
 }
 __device__ __forceinline__ double cell_f64(const DevCol& c, int64_t r) {
     if (c.type == BK_DOUBLE) return ((const double*)c.data)[r];
@@ -1930,7 +1929,7 @@ static int widen_col(BkgTable* t, int c) {
         return 0;
     }
     void* nd = nullptr;
-    if (hipMalloc(&nd, (size_t)t->nrows * es) != hipSuccess) {
+    if (hipMalloc(&nd, (size_t)t->nrows * es + 8) != hipSuccess) {
         set_err("widen_col: oom");
         return -1;
     }
@@ -1985,7 +1984,7 @@ extern "C" int bkgpu_table_compact(BkgTable* t, int col) {
                   : range <= 0xFFFFFFFFull ? 4 : 8;
         if (tgt >= cur) continue;
         void* nd = nullptr;
-        if (hipMalloc(&nd, (size_t)t->nrows * tgt) != hipSuccess) continue;
+        if (hipMalloc(&nd, (size_t)t->nrows * tgt + 8) != hipSuccess) continue;
         switch (tgt) {
             case 1:
                 hipLaunchKernelGGL(k_narrow_col<uint8_t>, dim3(2048),
@@ -2142,7 +2141,7 @@ extern "C" BkgTable* bkgpu_table_create(int ncols, const BkColSpec* specs,
         size_t es = elem_size(specs[c].col_type);
         if (es == 0) { set_err("unsupported col type"); delete t; return nullptr; }
         t->width[c] = (uint8_t)es;
-        if (hipMalloc(&t->data[c], (size_t)nrows * es) != hipSuccess) {
+        if (hipMalloc(&t->data[c], (size_t)nrows * es + 8) != hipSuccess) {
             set_err("hipMalloc column failed");
             bkgpu_table_free(t);
             return nullptr;
@@ -2199,12 +2198,14 @@ extern "C" void bkgpu_table_free(BkgTable* t) {
 static DevCols table_cols(const BkgTable* t) {
     DevCols dc{};
     for (int c = 0; c < t->ncols; c++) {
+        int w = t->width[c] ? t->width[c]
+                            : (int)elem_size(t->specs[c].col_type);
         dc.c[c].type = t->specs[c].col_type;
-        dc.c[c].width = t->width[c] ? t->width[c]
-                                    : (int32_t)elem_size(t->specs[c].col_type);
+        dc.c[c].lshift = w == 8 ? 3 : w == 4 ? 2 : w == 2 ? 1 : 0;
         dc.c[c].data = t->data[c];
         dc.c[c].valid = t->valid[c];
         dc.c[c].base = t->base[c];
+        dc.c[c].mask = w == 8 ? ~0ull : (1ull << (8 * w)) - 1;
     }
     return dc;
 }
@@ -2272,7 +2273,7 @@ extern "C" int bkgpu_table_derive_remap(BkgTable* t, int src_col,
     int nc = t->ncols;
     int32_t* dcol = nullptr;
     int32_t* dremap = nullptr;
-    HIP_CHECK(hipMalloc((void**)&dcol, (size_t)t->nrows * 4));
+    HIP_CHECK(hipMalloc((void**)&dcol, (size_t)t->nrows * 4 + 8));
     if (hipMalloc((void**)&dremap, (size_t)ncodes * 4) != hipSuccess) {
         (void)hipFree(dcol);
         set_err("derive_remap: oom");

@@ -729,7 +729,7 @@ def test_narrow_storage_widths_and_parity(eng):
         # ORDER BY on narrow keys matches wide
         sa = eng.sort_topk(t, [(1, 1, 1), (2, 0, 1)], 500)
         sb = eng.sort_topk(tw, [(1, 1, 1), (2, 0, 1)], 500)
-        assert np.array_equal(sa["rowids"], sb["rowids"])
+        assert np.array_equal(sa, sb)
         # re-upload re-widens and preserves values
         host = np.arange(n, dtype=np.int64) % 7
         eng.upload(t, 0, host)

@@ -36,10 +36,10 @@ for narrow in (False, True):
         eng.sync()
         dt = (time.time() - t0) * 1e3
         tag = "narrow" if narrow else "wide"
-        print(f"{tag} rep{rep} wall_ms={dt:.2f} kernel_ms={r.kernel_ms():.2f} "
-              f"ngroups={r.ngroups()} rows_passed={r.rows_passed()}")
+        print(f"{tag} rep{rep} wall_ms={dt:.2f} kernel_ms={r.kernel_ms:.2f} "
+              f"ngroups={r.ngroups} rows_passed={r.rows_passed}")
         if rep == 2:
-            base[tag] = (r.ngroups(), r.rows_passed())
+            base[tag] = (r.ngroups, r.rows_passed)
         r.free()
 assert base["wide"] == base["narrow"], base
 print("RESULTS MATCH", base)
