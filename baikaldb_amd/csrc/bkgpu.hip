@@ -159,7 +159,7 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
      * ~20% of the streaming ceiling (bwprobe predk vs preds) */
     #define BK_EVAL1(J, VI, VD, OK)                                         \
         int64_t VI = 0; double VD = 0.0; bool OK = true;                    \
-        if (q.n_conjuncts > (J)) {                                          \
+        if (SIMPLE || q.n_conjuncts > (J)) {                                \
             const BkConjunct& cj = q.conjuncts[J];                          \
             const DevCol& c = cols.c[cj.col];                               \
             if (!SIMPLE) OK = cell_valid(c, r);                             \
@@ -190,27 +190,38 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
      * early short-circuit via pass_all; OR members set per-clause bits —
      * SIMPLE instantiations are host-guaranteed all-standalone */
     #define BK_CLAUSE(J, P)                                                 \
-        if (SIMPLE || q.conjuncts[J].or_group == 0) {                       \
+        if (SIMPLE) {                                                       \
+            pass_all = pass_all & (bool)(P);                                \
+        } else if (q.conjuncts[J].or_group == 0) {                          \
             pass_all = pass_all && (P);                                     \
         } else {                                                            \
             uint32_t gbit = 1u << (q.conjuncts[J].or_group & 31);           \
             or_seen |= gbit;                                                \
             if (P) or_sat |= gbit;                                          \
         }
+    /* SIMPLE compare: 3-bit truth table over sign(cmp)+1, indexed by op
+     * (EQ 010, NE 101, GT 100, GE 110, LT 001, LE 011 packed little-endian
+     * at 3*op) — one shift+and, no switch, so the T-row materialize loops
+     * compile straight-line and the scheduler batches the column loads
+     * (the op switch split every load into its own basic block: measured
+     * 1-deep MLP, s_waitcnt vmcnt(0) after every load). */
     #define BK_TEST1(J, VI, VD, OK)                                          \
-        if (q.n_conjuncts > (J)) {                                          \
+        if (SIMPLE || q.n_conjuncts > (J)) {                                \
             const BkConjunct& cj = q.conjuncts[J];                          \
             bool pass;                                                      \
-            if (!SIMPLE && cj.op >= BK_OP_IN_BITMAP) {                                 \
+            if (SIMPLE) {                                                   \
+                int cmp = (VI > cj.lit_i) - (VI < cj.lit_i);                \
+                pass = (0x19D2Au >> (3 * cj.op + cmp + 1)) & 1u;            \
+            } else if (cj.op >= BK_OP_IN_BITMAP) {                          \
                 const uint8_t* bm = (const uint8_t*)(uintptr_t)cj.lit_i;    \
                 bool hit = (VI) >= 0 && (VI) < cj.n_in &&                   \
                            ((bm[(VI) >> 3] >> ((VI) & 7)) & 1);             \
                 pass = cj.op == BK_OP_IN_BITMAP ? hit : !hit;               \
-            } else if (!SIMPLE && cj.op >= BK_OP_IN) {                      \
+            } else if (cj.op >= BK_OP_IN) {                                 \
                 bool found = in_list_hit(cj, (VI));                         \
                 pass = cj.op == BK_OP_IN ? found : !found;                  \
             } else {                                                        \
-                int cmp = (!SIMPLE && cj.cmp_type == BK_DOUBLE)             \
+                int cmp = cj.cmp_type == BK_DOUBLE                          \
                     ? ((VD > cj.lit_d) - (VD < cj.lit_d))                   \
                     : ((VI > cj.lit_i) - (VI < cj.lit_i));                  \
                 switch (cj.op) {                                            \
@@ -308,6 +319,14 @@ __device__ __forceinline__ uint64_t enc_value(const DevCol& c, int64_t r) {
     }
 }
 
+/* branchless enc_value for SIMPLE contexts (query_simple excludes
+ * doubles): INT64/DATETIME flip the sign bit, STRING codes pass through
+ * (uniform ternary -> s_cselect, no control flow) */
+__device__ __forceinline__ uint64_t enc_value_nf(const DevCol& c, int64_t r) {
+    uint64_t flip = c.type == BK_STRING ? 0 : (1ull << 63);
+    return (uint64_t)cell_i64(c, r) ^ flip;
+}
+
 /* expression-capable aggregate input (BkAggSpec.col2/arith): value in the
  * agg_in_type domain; NULL if either operand NULL (agg_fn_call.cpp input
  * cast semantics) */
@@ -374,7 +393,8 @@ __device__ __forceinline__ KeyPack pack_group_keys(const DevCols& cols,
         if (!SIMPLE && !cell_valid(c, r)) {
             kp.flag |= 0x80u >> k;   /* null-flag bit, exec_node.cpp:561 */
         } else {
-            e = (!SIMPLE && q.group_fns[k])
+            e = SIMPLE ? enc_value_nf(c, r)
+                : q.group_fns[k]
                     ? bk_enc_i64(bk_scalar_fn(q.group_fns[k], cell_i64(c, r)))
                     : enc_value(c, r);
             if (bits < 64) {
@@ -2545,9 +2565,37 @@ static bool query_simple(const BkgTable* t, const BkQuerySpec* q) {
     for (int32_t k = 0; k < q->n_group; k++) {
         if (q->group_fns[k]) return false;
         if (t->valid[q->group_cols[k]]) return false;
+        /* SIMPLE key packing uses the branchless integer encode
+         * (enc_value_nf) — doubles need bk_enc_f64's conditional form */
+        if (t->specs[q->group_cols[k]].col_type == BK_DOUBLE) return false;
     }
     return true;
 }
+/* padded query copy for SIMPLE kernel launches: conjuncts replicated to
+ * exactly 4 (AND-idempotent; a GE INT64_MIN tautology on a referenced
+ * column when there are none), so row_passes<SIMPLE> compiles with no
+ * per-conjunct count branches — the duplicate loads hit the same
+ * addresses (L1) and cost no HBM traffic. */
+static BkQuerySpec pad_simple_q(const BkQuerySpec* q) {
+    BkQuerySpec o = *q;
+    BkConjunct c0;
+    if (o.n_conjuncts == 0) {
+        memset(&c0, 0, sizeof c0);
+        c0.col = o.n_group ? o.group_cols[0]
+                 : (o.n_aggs && o.aggs[0].col >= 0 ? o.aggs[0].col : 0);
+        c0.op = BK_OP_GE;
+        c0.cmp_type = BK_INT64;
+        c0.lit_i = INT64_MIN;
+        c0.col2 = -1;
+        o.conjuncts[0] = c0;
+    } else {
+        c0 = o.conjuncts[0];
+    }
+    for (int j = o.n_conjuncts; j < 4; j++) o.conjuncts[j] = c0;
+    o.n_conjuncts = 4;
+    return o;
+}
+
 static HistoFn pick_histo(int threads, bool hot, bool simple) {
     if (simple && !hot) {
         if (threads == 512)  return k_part_histo<512, false, true>;
@@ -2668,13 +2716,16 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     bool hot = hot_min <= 4096;
     if (!hot) hot_slots = 0;
     bool simple = query_simple(t, q) && getenv("BK_NO_SIMPLE") == nullptr;
+    BkQuerySpec qpad;
+    const BkQuerySpec* qk = q;     /* SIMPLE kernels get the padded copy */
+    if (simple) { qpad = pad_simple_q(q); qk = &qpad; }
     HistoFn histo_fn = pick_histo(threads, hot, simple);
     ScatFn scat_fn = pick_scat(threads, simple);
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
     EvTimer tm;
     tm.record();
     hipLaunchKernelGGL(histo_fn, dim3(nblocks), dim3(threads), histo_lds, 0,
-                       dc, *q, row_begin, row_end, P, bucketid, H,
+                       dc, *qk, row_begin, row_end, P, bucketid, H,
                        o->table, o->nslots - 1, (o->nslots * 7) / 8,
                        o->ctrs, o->ctrs + 1, o->err, hot_slots,
                        hot_cap, hot_probe, hot_min);
@@ -2710,7 +2761,7 @@ static int run_partitioned(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
         size_t sc_lds = paired ? pair_lds : (size_t)P * 8;
         hipLaunchKernelGGL(scat_fn, dim3(nblocks), dim3(threads),
                            sc_lds, 0,
-                           dc, *q, lay, row_begin, row_end, P, bucketid, H,
+                           dc, *qk, lay, row_begin, row_end, P, bucketid, H,
                            rec, total, paired);
     }
     tm.record();
@@ -2798,6 +2849,9 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
     bool hot = hot_min <= 4096;
     if (!hot) hot_slots = 0;
     bool simple = query_simple(t, q) && getenv("BK_NO_SIMPLE") == nullptr;
+    BkQuerySpec qpad;
+    const BkQuerySpec* qk = q;     /* SIMPLE kernels get the padded copy */
+    if (simple) { qpad = pad_simple_q(q); qk = &qpad; }
     HistoFn histo_fn = pick_histo(threads, hot, simple);
     ScatFn scat_fn = pick_scat(threads, simple);
     size_t histo_lds = ((size_t)hot_slots * stride + 4) * 8 + (size_t)P * 4;
@@ -2856,7 +2910,7 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
         }
         PPCHECK(hipMemsetAsync(sl[si].totals, 0, (size_t)P * 4, st));
         hipLaunchKernelGGL(histo_fn, dim3(nblocks), dim3(threads), histo_lds, st,
-                           dc, *q, cb, ce, P, sl[si].bucketid, sl[si].H,
+                           dc, *qk, cb, ce, P, sl[si].bucketid, sl[si].H,
                            o->table, o->nslots - 1, (o->nslots * 7) / 8,
                            o->ctrs, o->ctrs + 1, o->err, hot_slots,
                            hot_cap, hot_probe, hot_min);
@@ -2875,7 +2929,7 @@ static int run_partitioned_pipe(BkgAggOut* o, BkgTable* t, const BkQuerySpec* q,
             PPCHECK(pool_alloc((void**)&sl[si].rec,
                                (size_t)total * lay.nwords * 8));
             hipLaunchKernelGGL(scat_fn, dim3(nblocks), dim3(threads), sc_lds, st,
-                               dc, *q, lay, cb, ce, P, sl[si].bucketid,
+                               dc, *qk, lay, cb, ce, P, sl[si].bucketid,
                                sl[si].H, sl[si].rec, total, 0);
             uint64_t chunk_sz = std::max<uint64_t>(
                 32768, std::min<uint64_t>(1u << 20, total / 4096));

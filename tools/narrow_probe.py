@@ -40,6 +40,7 @@ for narrow in (False, True):
               f"ngroups={r.ngroups} rows_passed={r.rows_passed}")
         if rep == 2:
             base[tag] = (r.ngroups, r.rows_passed)
+            print(tag, "breakdown", r.breakdown())
         r.free()
 assert base["wide"] == base["narrow"], base
 print("RESULTS MATCH", base)
