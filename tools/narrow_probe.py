@@ -23,6 +23,21 @@ else:
     conj = [(0, "<", 1 << 30), (1, "<", int((1 << 31) * 0.9)), (7, "!=", 63)]
     group, aggs, eg = [2, 7], [("count_star", -1), ("sum", 3), ("sum", 4),
                                ("avg", 5)], 1 << 21
+if which == "c5":
+    specs = [(T_I, 0, 0, 1 << 31, 0)] * 3
+    t = eng.create_table(specs, 1_000_000_000)
+    for narrow in (False, True):
+        eng.generate(t, 20260915, compact=narrow)
+        eng.sync()
+        print("widths", [eng.col_width(t, c) for c in range(3)])
+        for rep in range(3):
+            t0 = time.time()
+            ids = eng.sort_topk(t, [(0, 1, 1), (1, 1, 1)], 1_000_000)
+            dt = (time.time() - t0) * 1e3
+            print(("narrow" if narrow else "wide"), rep,
+                  "wall_ms=%.2f" % dt, "kernel_ms=%.2f" % eng.topk_kernel_ms())
+    t.free()
+    sys.exit(0)
 t = eng.create_table(specs, 1_000_000_000)
 plan = QueryPlan(t.col_types, conjuncts=conj, group=group, aggs=aggs)
 base = {}
