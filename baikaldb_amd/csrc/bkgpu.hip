@@ -106,7 +106,8 @@ __device__ __forceinline__ int64_t cell_i64(const DevCol& c, int64_t r) {
     /* BRANCHLESS width decode: one (possibly unaligned) 8-byte load at
      * r << lshift, then base + (raw & mask). gfx950 global loads handle
      * any alignment in hardware (single global_load_dwordx2); column
-     * buffers carry an 8-byte tail pad for the overread. Width-dependent
+     * buffers carry a 16-byte tail pad (k_dedup_mat_vec reads 16 B
+     * per conjunct per 4-row sub-chunk; worst overread 12 B at u8). Width-dependent
      * branches in this accessor were measured to spill 200-700 B/lane of
      * scratch in the batched kernels (k_dedup_mat/k_topk_scan) and erase
      * the narrow-storage win. Wide columns have lshift 3, base 0, mask ~0
@@ -1949,7 +1950,7 @@ static int widen_col(BkgTable* t, int c) {
         return 0;
     }
     void* nd = nullptr;
-    if (hipMalloc(&nd, (size_t)t->nrows * es + 8) != hipSuccess) {
+    if (hipMalloc(&nd, (size_t)t->nrows * es + 16) != hipSuccess) {
         set_err("widen_col: oom");
         return -1;
     }
@@ -2004,7 +2005,7 @@ extern "C" int bkgpu_table_compact(BkgTable* t, int col) {
                   : range <= 0xFFFFFFFFull ? 4 : 8;
         if (tgt >= cur) continue;
         void* nd = nullptr;
-        if (hipMalloc(&nd, (size_t)t->nrows * tgt + 8) != hipSuccess) continue;
+        if (hipMalloc(&nd, (size_t)t->nrows * tgt + 16) != hipSuccess) continue;
         switch (tgt) {
             case 1:
                 hipLaunchKernelGGL(k_narrow_col<uint8_t>, dim3(2048),
@@ -2161,7 +2162,7 @@ extern "C" BkgTable* bkgpu_table_create(int ncols, const BkColSpec* specs,
         size_t es = elem_size(specs[c].col_type);
         if (es == 0) { set_err("unsupported col type"); delete t; return nullptr; }
         t->width[c] = (uint8_t)es;
-        if (hipMalloc(&t->data[c], (size_t)nrows * es + 8) != hipSuccess) {
+        if (hipMalloc(&t->data[c], (size_t)nrows * es + 16) != hipSuccess) {
             set_err("hipMalloc column failed");
             bkgpu_table_free(t);
             return nullptr;
@@ -2293,7 +2294,7 @@ extern "C" int bkgpu_table_derive_remap(BkgTable* t, int src_col,
     int nc = t->ncols;
     int32_t* dcol = nullptr;
     int32_t* dremap = nullptr;
-    HIP_CHECK(hipMalloc((void**)&dcol, (size_t)t->nrows * 4 + 8));
+    HIP_CHECK(hipMalloc((void**)&dcol, (size_t)t->nrows * 4 + 16));
     if (hipMalloc((void**)&dremap, (size_t)ncodes * 4) != hipSuccess) {
         (void)hipFree(dcol);
         set_err("derive_remap: oom");
