@@ -23,7 +23,7 @@ def run_pair(nrows, reps):
                      conjuncts=[(0, "<", 1 << 30), (1, "=", 7)],
                      group=[2], aggs=[("sum", 3)])
     res = {}
-    for mode in ("0", "1"):
+    for mode in ("0", "1", "2"):
         os.environ["BK_MAT_VEC"] = mode
         best = None
         for rep in range(reps):
@@ -35,16 +35,17 @@ def run_pair(nrows, reps):
             if rep == 0:
                 res[mode] = r.fetch(sorted=True)
             r.free()
-        print(("vec    " if mode == "1" else "strided"), f"n={nrows}",
+        print({"0": "strided", "1": "vec    ", "2": "vecukey"}[mode], f"n={nrows}",
               "kernel_ms=%.2f" % best[0],
               " ".join(f"{k}={v:.2f}" for k, v in best[1].items()
                        if v >= 0.05), flush=True)
-    a, b = res["0"], res["1"]
-    assert a["rows_passed"] == b["rows_passed"]
-    assert a["ngroups"] == b["ngroups"]
-    assert np.array_equal(a["enc"], b["enc"])
-    for i in range(len(a["agg_i"])):
-        assert np.array_equal(a["agg_i"][i], b["agg_i"][i]), i
+    a = res["0"]
+    for b in (res["1"], res["2"]):
+        assert a["rows_passed"] == b["rows_passed"]
+        assert a["ngroups"] == b["ngroups"]
+        assert np.array_equal(a["enc"], b["enc"])
+        for i in range(len(a["agg_i"])):
+            assert np.array_equal(a["agg_i"][i], b["agg_i"][i]), i
     print(f"parity OK n={nrows} rows={a['rows_passed']} "
           f"groups={a['ngroups']}", flush=True)
     t.free()
