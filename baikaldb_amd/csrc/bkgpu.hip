@@ -2618,6 +2618,8 @@ static ScatFn pick_scat(int threads, bool simple) {
     return k_part_scatter<256>;
 }
 
+#include "bkdpart.inc"
+
 /* pipelined variant of run_partitioned: the row range splits into chunks
  * and consecutive chunks run on TWO HIP streams, so one chunk's part_agg
  * (LDS-latency-bound, ~1.2 TB/s — NOT HBM-saturated) overlaps the next
@@ -3013,6 +3015,17 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
     bool partitioned = q->n_group > 0 && expected_groups > FUSED_MAX_GROUPS &&
                        row_end > row_begin;
 
+    /* dense-span plan (bkdpart.inc): a small packed key domain turns the
+     * aggregate into a range partition + direct-indexed LDS accumulate (no
+     * hash probes / claim chains). Preferred at large ranges; BK_DENSE=0
+     * disables, BK_DENSE=2 prefers it at every range. */
+    bool use_dense = false;
+    DenseSpec dsp;
+    DRecLayout dlay;
+    if (partitioned && dense_eligible(t, q, &dsp, &dlay)) use_dense = true;
+    int dense_pref = 1;
+    if (const char* e = getenv("BK_DENSE")) dense_pref = atoi(e);
+
     /* small/mid ranges: the sort-dedup path beats the partitioned pipeline
      * (histo+scatter+agg fixed passes dominate at low survivor volume; a
      * 1e8-row config2-shaped query measured ~3x faster sorted). An EQUALITY
@@ -3029,7 +3042,10 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
             if (q->conjuncts[j].op == BK_OP_EQ) has_eq = true;
         int64_t smax = has_eq ? 1200 * 1000 * 1000ll : 200 * 1000 * 1000;
         if (const char* e = getenv("BK_SORTED_RANGE")) smax = atoll(e);
-        if (row_end - row_begin <= smax) {
+        bool dense_first = use_dense &&
+                           (dense_pref >= 2 ||
+                            row_end - row_begin > 200 * 1000 * 1000);
+        if (!dense_first && row_end - row_begin <= smax) {
             BkgAggOut* so = bkgpu_filter_agg_sorted(t, q, row_begin, row_end);
             if (so) { delete o; return so; }
             g_err[0] = 0;   /* shape did not qualify — partitioned path */
@@ -3040,7 +3056,12 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
         if (agg_alloc(o, nslots) != 0) { bkgpu_agg_free(o); return nullptr; }
         if (debug_timing()) { (void)hipDeviceSynchronize(); t_alloc = now_ms(); }
         int blocks = 2048, threads = 256;
-        if (partitioned) {
+        if (partitioned && use_dense) {
+            if (run_dense(o, t, q, dsp, dlay, row_begin, row_end) != 0) {
+                bkgpu_agg_free(o);
+                return nullptr;
+            }
+        } else if (partitioned) {
             if (run_partitioned(o, t, q, row_begin, row_end, expected_groups) != 0) {
                 bkgpu_agg_free(o);
                 return nullptr;
