@@ -78,9 +78,13 @@ def main(config, dominant):
 if __name__ == "__main__":
     import sys as _sys
     which = _sys.argv[1] if len(_sys.argv) > 1 else "all"
+    # dominant kernels for the round-2 dense-span pipeline (bkdpart.inc);
+    # pass a third arg to override (old hash-path names: k_part_scatter /
+    # k_dedup_mat)
+    dom = _sys.argv[2] if len(_sys.argv) > 2 else None
     if which in ("all", "c3"):
-        main("config3_1e9_mixed", "k_part_scatter")
+        main("config3_1e9_mixed", dom or "k_dscatter")
     if which in ("all", "c5"):
-        main("config5_1e9_sort", "k_topk_scan")
+        main("config5_1e9_sort", dom or "k_topk_scan")
     if which in ("all", "c2b"):
-        main("config2_1e9_8int64", "k_dedup_mat")
+        main("config2_1e9_8int64", dom or "k_dhisto")
