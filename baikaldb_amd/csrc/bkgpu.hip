@@ -2359,7 +2359,18 @@ struct BkgAggOut {
     float  t_ms[8] = {};           /* per-kernel breakdown */
     char   k_names[8][16] = {};
     bool dirty = true;             /* table modified since last compact */
+    /* dense-span result mode (bkdpart.inc): the result lives in the dense
+     * accumulator arrays; compact fills the blob DIRECTLY from them, and
+     * merge/rollup rebuild the hash table lazily (dense_to_hash) only when
+     * they actually need slot probing. */
+    bool dense_mode = false;
+    void* dense_spec = nullptr;    /* heap DenseSpec */
+    uint64_t* dvals = nullptr;     /* device: span * nv value words */
+    uint64_t* dtouch = nullptr;    /* device: span touched counters */
 };
+
+static int dense_compact(BkgAggOut* o);        /* bkdpart.inc */
+static int dense_to_hash(BkgAggOut* o);        /* bkdpart.inc */
 
 static int64_t next_pow2(int64_t x) {
     int64_t p = 1;
@@ -2395,12 +2406,16 @@ extern "C" void bkgpu_agg_free(BkgAggOut* o) {
     if (o->ctrs) (void)hipFree(o->ctrs);
     if (o->err) (void)hipFree(o->err);
     if (o->blob) pool_free(o->blob);
+    if (o->dvals) pool_free(o->dvals);
+    if (o->dtouch) pool_free(o->dtouch);
+    free(o->dense_spec);
     delete o;
 }
 
 /* run compact if table changed; updates o->ngroups and o->blob */
 static int agg_compact(BkgAggOut* o) {
     if (!o->dirty && o->ngroups >= 0) return 0;
+    if (o->dense_mode) return dense_compact(o);
     const int naggs = o->q.n_aggs;
     int64_t cap = (int64_t)o->nslots;
     /* read fill count to size the blob */
@@ -3053,7 +3068,13 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
     }
 
     for (int attempt = 0; attempt < 8; attempt++) {
-        if (agg_alloc(o, nslots) != 0) { bkgpu_agg_free(o); return nullptr; }
+        /* dense mode never touches the hash table during aggregation (a
+         * 16-slot stub keeps downstream code paths non-null; dense_to_hash
+         * re-allocates at the real size if merge/rollup need it) */
+        if (agg_alloc(o, partitioned && use_dense ? 16 : nslots) != 0) {
+            bkgpu_agg_free(o);
+            return nullptr;
+        }
         if (debug_timing()) { (void)hipDeviceSynchronize(); t_alloc = now_ms(); }
         int blocks = 2048, threads = 256;
         if (partitioned && use_dense) {
@@ -3178,6 +3199,9 @@ extern "C" BkgAggOut* bkgpu_agg_rollup(const BkgAggOut* in, const BkQuerySpec* q
                                        const int32_t* src_idx,
                                        int64_t expected_groups) {
     if (!in || !q2 || !src_idx) { set_err("rollup: null arg"); return nullptr; }
+    if (in->dense_mode &&
+        dense_to_hash(const_cast<BkgAggOut*>(in)) != 0)
+        return nullptr;
     if (in->q.n_group != q2->n_group + 1 || q2->n_group > 1) {
         set_err("rollup: level-1 must group by (user keys + distinct col), "
                 "user keys <= 1");
@@ -3346,6 +3370,9 @@ static int merge_blob_into(BkgAggOut* o, const void* blob, int64_t n_groups) {
 }
 
 extern "C" int bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups) {
+    /* a dense-mode result first materializes as a hash table (lazy: only
+     * merges/rollups need slot probing) */
+    if (o->dense_mode && dense_to_hash(o) != 0) return -1;
     /* k_merge_blob mutates the table as it goes, so an overflow mid-merge is
      * unrecoverable — ensure capacity FIRST: if fill+n could cross the cap,
      * rebuild the table at a larger size from our own compact blob (additive
