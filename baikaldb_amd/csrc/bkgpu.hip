@@ -3403,6 +3403,109 @@ extern "C" int bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups)
     return 0;
 }
 
+/* ---- hash-partitioned exchange (ExchangeSenderNode::repartition's role,
+ * exchange_sender_node.h:228-235, feeding RCCL all-to-all over xGMI) ---- */
+
+__global__ void k_blob_part_hist(const uint32_t* flags, const uint64_t* k0,
+                                 const uint64_t* k1, int64_t n, int nparts,
+                                 unsigned long long* counts) {
+    int64_t gs = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gs)
+        atomicAdd(&counts[key_hash(flags[i], k0[i], k1[i])
+                          % (uint64_t)nparts], 1ull);
+}
+
+__global__ void k_blob_export_part(const uint32_t* flags, const uint64_t* k0,
+                                   const uint64_t* k1, const uint64_t* states,
+                                   int64_t n, int naggs, int nparts, int part,
+                                   uint64_t* cursor, uint32_t* of,
+                                   uint64_t* ok0, uint64_t* ok1, uint64_t* ost,
+                                   uint64_t cap) {
+    int64_t gs = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gs) {
+        if (key_hash(flags[i], k0[i], k1[i]) % (uint64_t)nparts
+            != (uint64_t)part)
+            continue;
+        uint64_t j = atomicAdd((unsigned long long*)cursor, 1ull);
+        if (j >= cap) continue;
+        of[j] = flags[i];
+        ok0[j] = k0[i];
+        ok1[j] = k1[i];
+        for (int w = 0; w < 2 * naggs; w++)
+            ost[j * (uint64_t)(2 * naggs) + w]
+                = states[i * (uint64_t)(2 * naggs) + w];
+    }
+}
+
+extern "C" int bkgpu_agg_part_counts(const BkgAggOut* o_, int nparts,
+                                     int64_t* counts) {
+    BkgAggOut* o = const_cast<BkgAggOut*>(o_);
+    if (nparts < 1 || nparts > 65536) { set_err("bad nparts"); return -1; }
+    if (agg_compact(o) != 0) return -1;
+    unsigned long long* d = nullptr;
+    HIP_CHECK(pool_alloc((void**)&d, (size_t)nparts * 8));
+    HIP_CHECK(hipMemset(d, 0, (size_t)nparts * 8));
+    if (o->ngroups > 0) {
+        int64_t c = o->blob_groups;
+        const uint8_t* b = o->blob;
+        hipLaunchKernelGGL(k_blob_part_hist, dim3(512), dim3(256), 0, 0,
+                           (const uint32_t*)b, (const uint64_t*)(b + c * 4),
+                           (const uint64_t*)(b + c * 12), o->ngroups, nparts,
+                           d);
+        HIP_CHECK(hipGetLastError());
+    }
+    HIP_CHECK(hipMemcpy(counts, d, (size_t)nparts * 8,
+                        hipMemcpyDeviceToHost));
+    pool_free(d);
+    return 0;
+}
+
+extern "C" int bkgpu_agg_export_part(const BkgAggOut* o_, int nparts,
+                                     int part, void* dst,
+                                     int64_t part_groups) {
+    BkgAggOut* o = const_cast<BkgAggOut*>(o_);
+    if (nparts < 1 || part < 0 || part >= nparts) {
+        set_err("bad part");
+        return -1;
+    }
+    if (agg_compact(o) != 0) return -1;
+    if (part_groups <= 0) return 0;
+    uint64_t* cursor = nullptr;
+    HIP_CHECK(pool_alloc((void**)&cursor, 8));
+    HIP_CHECK(hipMemset(cursor, 0, 8));
+    int64_t c = o->blob_groups;
+    const uint8_t* b = o->blob;
+    uint8_t* dp = (uint8_t*)dst;
+    uint32_t* of = (uint32_t*)dp;
+    uint64_t* ok0 = (uint64_t*)(dp + part_groups * 4);
+    uint64_t* ok1 = (uint64_t*)(dp + part_groups * 12);
+    uint64_t* ost = (uint64_t*)(dp + part_groups * 20);
+    hipLaunchKernelGGL(k_blob_export_part, dim3(512), dim3(256), 0, 0,
+                       (const uint32_t*)b, (const uint64_t*)(b + c * 4),
+                       (const uint64_t*)(b + c * 12),
+                       (const uint64_t*)(b + c * 20), o->ngroups,
+                       o->q.n_aggs, nparts, part, cursor, of, ok0, ok1, ost,
+                       (uint64_t)part_groups);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipDeviceSynchronize());
+    pool_free(cursor);
+    return 0;
+}
+
+extern "C" BkgAggOut* bkgpu_agg_empty(const BkQuerySpec* q,
+                                      int64_t expected_groups) {
+    if (!q) { set_err("agg_empty: null query"); return nullptr; }
+    if (ensure_device() != 0) return nullptr;
+    BkgAggOut* o = new BkgAggOut();
+    o->q = *q;
+    int64_t nslots = next_pow2(std::max<int64_t>(expected_groups * 2, 1024));
+    if (agg_alloc(o, nslots) != 0) { bkgpu_agg_free(o); return nullptr; }
+    o->dirty = true;
+    return o;
+}
+
 /* ---- fetch: download + finalize (agg_fn_call.cpp:927-975) ---- */
 
 struct HostGroups {

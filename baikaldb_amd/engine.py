@@ -57,6 +57,14 @@ def _load():
     lib.bkgpu_agg_export_bytes.argtypes = [C.c_void_p]
     lib.bkgpu_agg_export.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
     lib.bkgpu_agg_merge.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
+    lib.bkgpu_agg_part_counts.restype = C.c_int
+    lib.bkgpu_agg_part_counts.argtypes = [C.c_void_p, C.c_int,
+                                          C.POINTER(C.c_int64)]
+    lib.bkgpu_agg_export_part.restype = C.c_int
+    lib.bkgpu_agg_export_part.argtypes = [C.c_void_p, C.c_int, C.c_int,
+                                          C.c_void_p, C.c_int64]
+    lib.bkgpu_agg_empty.restype = C.c_void_p
+    lib.bkgpu_agg_empty.argtypes = [C.POINTER(BkQuerySpec), C.c_int64]
     lib.bkgpu_window.restype = C.c_int64
     lib.bkgpu_window.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec), C.c_int32,
                                  C.POINTER(BkOrderSpec), C.c_int,
@@ -157,6 +165,19 @@ class AggResult:
         rc = self.engine.lib.bkgpu_agg_merge(self.handle, C.c_void_p(dev_ptr), n_groups)
         self.engine._check(rc, "agg_merge")
 
+    def part_counts(self, nparts):
+        """Per-part group counts for the hash-partitioned exchange (the
+        repartition ExchangeSenderNode does, exchange_sender_node.h:228)."""
+        out = (C.c_int64 * nparts)()
+        rc = self.engine.lib.bkgpu_agg_part_counts(self.handle, nparts, out)
+        self.engine._check(rc, "agg_part_counts")
+        return list(out)
+
+    def export_part(self, nparts, part, dev_ptr, part_groups):
+        rc = self.engine.lib.bkgpu_agg_export_part(
+            self.handle, nparts, part, C.c_void_p(dev_ptr), part_groups)
+        self.engine._check(rc, "agg_export_part")
+
     def fetch(self, sorted=True, max_groups=None):
         n = self.ngroups if max_groups is None else min(max_groups, self.ngroups)
         n = max(n, 0)
@@ -250,6 +271,15 @@ class GpuEngine:
                                       expected_groups)
         if not h:
             raise RuntimeError(f"filter_agg: {self.lib.bkgpu_last_error().decode()}")
+        return AggResult(self, h, plan)
+
+    def agg_empty(self, plan: QueryPlan, expected_groups=1 << 16):
+        """Fresh empty aggregate result (merge target for the partitioned
+        exchange's received blobs)."""
+        q = plan.to_spec()
+        h = self.lib.bkgpu_agg_empty(C.byref(q), expected_groups)
+        if not h:
+            raise RuntimeError(f"agg_empty: {self.lib.bkgpu_last_error().decode()}")
         return AggResult(self, h, plan)
 
     # unary string scalar fns the engine compiles to dict remaps

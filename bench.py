@@ -253,29 +253,50 @@ def main():
             log(f"[dbg] breakdown { {k: round(v, 1) for k, v in res.breakdown().items()} }")
         if dist is not None:
             eng.sync()
-            ng = torch.tensor([res.ngroups], dtype=torch.int64, device="cuda")
-            sizes = [torch.zeros(1, dtype=torch.int64, device="cuda")
-                     for _ in range(world)]
-            dist.all_gather(sizes, ng)
-            sizes = [int(s.item()) for s in sizes]
-            # wire-blob bytes per group (flags4 + k0/k1 16 + states 16*naggs)
-            # — computed analytically so every rank allocates identically
-            # even when its own result is empty
+            # hash-partitioned all-to-all exchange — the repartition the
+            # reference's ExchangeSenderNode does over brpc
+            # (exchange_sender_node.h:228-235), here over RCCL/xGMI: rank r
+            # keeps hash-part r and merges every peer's part-r blob, so the
+            # merge work spreads across ranks and the traffic across all 7
+            # xGMI links (replaces the serialized gather-to-rank-0 merge).
             per_group = 20 + 16 * len(cfg["aggs"])
-            maxg = max(sizes)
-            buf = torch.zeros(max(maxg * per_group, 1), dtype=torch.uint8,
-                              device="cuda")
-            if res.ngroups > 0:
-                res.export_to(buf.data_ptr(), maxg * per_group)
-            gath = [torch.zeros_like(buf) for _ in range(world)]
-            dist.all_gather(gath, buf)
+            counts = res.part_counts(world)
+            scounts = torch.tensor(counts, dtype=torch.int64, device="cuda")
+            rcounts = torch.zeros_like(scounts)
+            dist.all_to_all_single(rcounts, scounts)
+            in_splits = [c * per_group for c in counts]
+            sbuf = torch.zeros(max(sum(in_splits), 1), dtype=torch.uint8,
+                               device="cuda")
+            off = 0
+            for p in range(world):
+                if counts[p] > 0:
+                    res.export_part(world, p, sbuf.data_ptr() + off,
+                                    counts[p])
+                off += in_splits[p]
+            rlist = [int(x) for x in rcounts.cpu()]
+            out_splits = [c * per_group for c in rlist]
+            rbuf = torch.zeros(max(sum(out_splits), 1), dtype=torch.uint8,
+                               device="cuda")
+            dist.all_to_all_single(rbuf, sbuf, out_splits, in_splits)
             torch.cuda.synchronize()
-            if rank == 0:
-                for peer in range(1, world):
-                    if sizes[peer] > 0:
-                        # repack: peer blob was laid out for sizes[peer] groups
-                        res.merge_blob(gath[peer].data_ptr(), sizes[peer])
-                eng.sync()
+            # local merge of the received part-`rank` blobs: the final
+            # result is SHARDED by key hash (disjoint across ranks)
+            merged = eng.agg_empty(plan,
+                                   expected_groups=cfg["expected_groups"])
+            off = 0
+            for p in range(world):
+                if rlist[p] > 0:
+                    merged.merge_blob(rbuf.data_ptr() + off, rlist[p])
+                off += out_splits[p]
+            eng.sync()
+            tot = torch.tensor([merged.ngroups], dtype=torch.int64,
+                               device="cuda")
+            dist.all_reduce(tot)
+            merged.free()
+            rp = res.rows_passed
+            ng = int(tot.item())
+            res.free()
+            return rp, ng
         rp = res.rows_passed
         ng = res.ngroups
         res.free()

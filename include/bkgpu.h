@@ -108,6 +108,22 @@ int  bkgpu_agg_export(const BkgAggOut* o, void* dst, int64_t cap);
  * AggFnCall::merge semantics (src/expr/agg_fn_call.cpp:781-830). */
 int  bkgpu_agg_merge(BkgAggOut* o, const void* blob, int64_t n_groups);
 
+/* ---- hash-partitioned exchange (the MPP repartition the reference's
+ * ExchangeSenderNode does over brpc, exchange_sender_node.h:228-235;
+ * here it feeds an RCCL all-to-all over xGMI): the partial result's
+ * groups split by key hash into `nparts` disjoint wire blobs, every rank
+ * merges the blobs of ITS part in parallel — replacing the serialized
+ * gather-to-rank-0 merge. The hash is the engine's key_hash, identical
+ * on every rank by construction. ---- */
+/* per-part group counts (runs compact if needed) */
+int  bkgpu_agg_part_counts(const BkgAggOut* o, int nparts, int64_t* counts);
+/* write part `part`'s blob (DEVICE dst; layout = the standard wire blob
+ * with part_groups groups — pass the count bkgpu_agg_part_counts gave) */
+int  bkgpu_agg_export_part(const BkgAggOut* o, int nparts, int part,
+                           void* dst, int64_t part_groups);
+/* fresh empty result for `q` (merge target for received part blobs) */
+BkgAggOut* bkgpu_agg_empty(const BkQuerySpec* q, int64_t expected_groups);
+
 /* Level-1 aggregate via SORT-based dedup: radix-sorts the spec-packed group
  * key per passing row and emits one DENSE table slot per unique key — the
  * path for high-cardinality DISTINCT (dedup cardinality ~ rows), where the

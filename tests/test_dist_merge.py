@@ -132,3 +132,80 @@ def test_gloo_shard_merge_matches_whole():
         # min/max exact
         assert m[4][0] == w[4][0]
         assert m[5][1] == w[5][1]
+
+
+def _bk_mix64(x):
+    # bk_datagen.h bk_mix64 (splitmix64 finalizer) — restated exactly, so
+    # the CPU test partitions with the SAME hash the engine's key_hash uses
+    # on device (a key must land on the same part on every rank)
+    M = 2**64 - 1
+    x = (int(x) + 0x9E3779B97F4A7C15) & M
+    x = (((x ^ (x >> 30)) * 0xBF58476D1CE4E5B9) & M)
+    x = (((x ^ (x >> 27)) * 0x94D049BB133111EB) & M)
+    return np.uint64(x ^ (x >> 31))
+
+
+def _key_part(key, nparts):
+    flag, e0, e1 = key
+    h = _bk_mix64(np.uint64(e0) ^ np.uint64(0x9E3779B97F4A7C15))
+    h = _bk_mix64(h ^ np.uint64(e1))
+    h = _bk_mix64(h ^ np.uint64(flag))
+    return int(h) % nparts
+
+
+def _worker_partitioned(rank, world, rendezvous, results):
+    """Hash-partitioned exchange protocol (the all-to-all bench.py drives
+    over RCCL; gloo has no all_to_all, so parts travel via
+    all_gather_object): every rank keeps only part `rank`, merges the
+    peers' part-`rank` blobs, and the SHARDS must be disjoint and union to
+    the whole result."""
+    torch.distributed.init_process_group(
+        "gloo", init_method=rendezvous, rank=rank, world_size=world)
+    part, _ = oracle_partial(rank, world, 40_000)
+    split = [dict() for _ in range(world)]
+    for key, states in part.items():
+        split[_key_part(key, world)][key] = states
+    gathered = [None] * world
+    torch.distributed.all_gather_object(gathered, split)
+    agg_types = ["count_star", "sum", "sum", "avg", "min", "max"]
+    is_dbl = [False, False, True, True, False, True]
+    mine = merge_partials([g[rank] for g in gathered], 6, agg_types, is_dbl)
+    shards = [None] * world
+    torch.distributed.all_gather_object(shards, mine)
+    if rank == 0:
+        results.put(shards)
+    torch.distributed.destroy_process_group()
+
+
+def test_gloo_partitioned_exchange_matches_whole():
+    import tempfile
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    rdvfile = tempfile.NamedTemporaryFile(delete=False)
+    rendezvous = f"file://{rdvfile.name}"
+    procs = [ctx.Process(target=_worker_partitioned,
+                         args=(r, 2, rendezvous, results))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    shards = results.get(timeout=180)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    whole, _ = oracle_partial(-1, 1, 40_000)
+    # shards are disjoint and union to the whole key set
+    assert not (set(shards[0]) & set(shards[1]))
+    assert set(shards[0]) | set(shards[1]) == set(whole.keys())
+    # every key landed on the part its hash names
+    for p, shard in enumerate(shards):
+        for key in shard:
+            assert _key_part(key, 2) == p
+    merged = {**shards[0], **shards[1]}
+    for key in whole:
+        w, m = whole[key], merged[key]
+        assert m[0][0] == w[0][0]                       # COUNT(*)
+        assert m[1][0] == w[1][0]                       # SUM int64
+        assert m[2][1] == pytest.approx(w[2][1], rel=1e-9, abs=1e-9)
+        assert m[4][0] == w[4][0]                       # MIN
+        assert m[5][1] == w[5][1]                       # MAX

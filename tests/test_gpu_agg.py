@@ -792,3 +792,71 @@ def test_vec_mat_parity(eng):
             assert np.array_equal(a["agg_i"][i], b["agg_i"][i]), i
     finally:
         t.free()
+
+
+def test_partitioned_exchange_equals_whole(eng, orc):
+    """Hash-partitioned exchange (bkgpu_agg_part_counts / export_part +
+    agg_empty + merge — ExchangeSenderNode::repartition's role,
+    exchange_sender_node.h:228-235), simulated with 2 'ranks' on one GPU:
+    each shard result splits into 2 disjoint part blobs by key hash;
+    'rank p' merges both shards' part-p blobs into an empty result; the
+    union of the merged shards must equal the whole-range result."""
+    import torch
+    from baikaldb_amd import QueryPlan
+    n = 400_000
+    aggs = [("count_star", -1), ("sum", 2), ("avg", 3), ("min", 0)]
+    conj = [(0, "<", int((1 << 31) * 0.7))]
+    t = eng.create_table(BASE5, n)
+    try:
+        eng.generate(t, SEED)
+        plan = QueryPlan(t.col_types, conjuncts=conj, group=[1, 4], aggs=aggs)
+        whole = eng.filter_agg(t, plan, expected_groups=1 << 14)
+        expect = whole.fetch(sorted=True)
+        whole.free()
+        world = 2
+        per_group = 20 + 16 * len(aggs)
+        shards = [eng.filter_agg(t, plan, row_end=n // 2,
+                                 expected_groups=1 << 14),
+                  eng.filter_agg(t, plan, row_begin=n // 2,
+                                 expected_groups=1 << 14)]
+        parts = {}
+        for s, res in enumerate(shards):
+            counts = res.part_counts(world)
+            assert sum(counts) == res.ngroups
+            for p in range(world):
+                buf = torch.zeros(max(counts[p] * per_group, 1),
+                                  dtype=torch.uint8, device="cuda")
+                if counts[p]:
+                    res.export_part(world, p, buf.data_ptr(), counts[p])
+                parts[(s, p)] = (buf, counts[p])
+        fetches = []
+        total_groups = 0
+        for p in range(world):
+            m = eng.agg_empty(plan, expected_groups=1 << 14)
+            for s in range(world):
+                buf, cnt = parts[(s, p)]
+                if cnt:
+                    m.merge_blob(buf.data_ptr(), cnt)
+            fetches.append(m.fetch(sorted=True))
+            total_groups += m.ngroups
+            m.free()
+        for s in shards:
+            s.free()
+    finally:
+        t.free()
+    # disjoint parts covering exactly the whole result
+    assert total_groups == expect["ngroups"]
+    enc = np.concatenate([f["enc"] for f in fetches])
+    flags = np.concatenate([f["flags"] for f in fetches])
+    agg_i = np.concatenate([f["agg_i"] for f in fetches], axis=1)
+    agg_d = np.concatenate([f["agg_d"] for f in fetches], axis=1)
+    order = np.lexsort(tuple(enc[:, k] for k in range(enc.shape[1] - 1, -1, -1)))
+    assert np.array_equal(enc[order], expect["enc"])
+    assert np.array_equal(flags[order], expect["flags"])
+    # COUNT(*), SUM int64, MIN int bit-exact; AVG double within tolerance
+    assert np.array_equal(agg_i[0][order], expect["agg_i"][0])
+    assert np.array_equal(agg_i[1][order], expect["agg_i"][1])
+    assert np.array_equal(agg_i[3][order], expect["agg_i"][3])
+    denom = np.abs(expect["agg_d"][2]) + np.maximum(expect["agg_i"][0], 1)
+    assert np.all(np.abs(agg_d[2][order] - expect["agg_d"][2])
+                  <= DTOL_REL * denom)
