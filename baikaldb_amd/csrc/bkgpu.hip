@@ -3321,6 +3321,17 @@ extern "C" int64_t bkgpu_agg_ngroups(const BkgAggOut* o) {
     return o->ngroups;
 }
 extern "C" int64_t bkgpu_agg_rows_passed(const BkgAggOut* o) { return o->rows_passed; }
+
+/* distinct slots claimed in the hash table — equals the group count for
+ * insert-only tables (every distinct key claims exactly once) WITHOUT
+ * paying a compact; the exchange path reads this after its merges. Dense
+ * results fall back to the compacted count. */
+extern "C" int64_t bkgpu_agg_nfilled(const BkgAggOut* o) {
+    if (o->dense_mode) return bkgpu_agg_ngroups(o);
+    uint64_t fill = 0;
+    HIP_CHECK(hipMemcpy(&fill, o->ctrs, 8, hipMemcpyDeviceToHost));
+    return (int64_t)fill;
+}
 extern "C" double bkgpu_agg_kernel_ms(const BkgAggOut* o) { return o->kernel_ms; }
 
 extern "C" int64_t bkgpu_agg_export_bytes(const BkgAggOut* o) {

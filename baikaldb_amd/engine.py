@@ -65,6 +65,8 @@ def _load():
                                           C.c_void_p, C.c_int64]
     lib.bkgpu_agg_empty.restype = C.c_void_p
     lib.bkgpu_agg_empty.argtypes = [C.POINTER(BkQuerySpec), C.c_int64]
+    lib.bkgpu_agg_nfilled.restype = C.c_int64
+    lib.bkgpu_agg_nfilled.argtypes = [C.c_void_p]
     lib.bkgpu_window.restype = C.c_int64
     lib.bkgpu_window.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec), C.c_int32,
                                  C.POINTER(BkOrderSpec), C.c_int,
@@ -142,6 +144,12 @@ class AggResult:
     @property
     def rows_passed(self):
         return self.engine.lib.bkgpu_agg_rows_passed(self.handle)
+
+    @property
+    def nfilled(self):
+        """Group count via the table fill counter (no compact) — valid
+        for insert-only hash results (the exchange merge target)."""
+        return self.engine.lib.bkgpu_agg_nfilled(self.handle)
 
     @property
     def kernel_ms(self):

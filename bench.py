@@ -260,7 +260,10 @@ def main():
             # merge work spreads across ranks and the traffic across all 7
             # xGMI links (replaces the serialized gather-to-rank-0 merge).
             per_group = 20 + 16 * len(cfg["aggs"])
+            t_p0 = time.perf_counter()
             counts = res.part_counts(world)
+            if os.environ.get("BK_BENCH_DEBUG"):
+                log(f"[dbg] part_counts {(time.perf_counter()-t_p0)*1e3:.1f} ms")
             scounts = torch.tensor(counts, dtype=torch.int64, device="cuda")
             rcounts = torch.zeros_like(scounts)
             dist.all_to_all_single(rcounts, scounts)
@@ -280,19 +283,26 @@ def main():
             dist.all_to_all_single(rbuf, sbuf, out_splits, in_splits)
             torch.cuda.synchronize()
             # local merge of the received part-`rank` blobs: the final
-            # result is SHARDED by key hash (disjoint across ranks)
+            # result is SHARDED by key hash (disjoint across ranks); the
+            # target is sized exactly (sum of received counts is an upper
+            # bound: keys dedup across peers)
+            t_x0 = time.perf_counter()
             merged = eng.agg_empty(plan,
-                                   expected_groups=cfg["expected_groups"])
+                                   expected_groups=max(sum(rlist), 1))
             off = 0
             for p in range(world):
                 if rlist[p] > 0:
                     merged.merge_blob(rbuf.data_ptr() + off, rlist[p])
                 off += out_splits[p]
             eng.sync()
-            tot = torch.tensor([merged.ngroups], dtype=torch.int64,
+            # group count from the fill counter (insert-only target: every
+            # distinct key claimed exactly once) — no compact needed
+            tot = torch.tensor([merged.nfilled], dtype=torch.int64,
                                device="cuda")
             dist.all_reduce(tot)
             merged.free()
+            if os.environ.get("BK_BENCH_DEBUG"):
+                log(f"[dbg] exchange merge leg {(time.perf_counter()-t_x0)*1e3:.1f} ms")
             rp = res.rows_passed
             ng = int(tot.item())
             res.free()
