@@ -28,6 +28,7 @@
 #include <vector>
 
 #include "../../include/bk_common.h"
+#include "../../include/bk_like.h"
 #include "../../include/bk_keyenc.h"
 #include "../../include/bk_datagen.h"
 #include "../../include/bkgpu.h"
@@ -981,4 +982,20 @@ extern "C" void bkexec_close(BkExecTree* t) {
 extern "C" int bkexec_dict_word(uint64_t dict_seed, int64_t code, char* out,
                                 int cap) {
     return bk_dict_word(dict_seed, code, out, cap);
+}
+
+/* SQL LIKE export (include/bk_like.h; LikePredicate::like restated) for
+ * binding-level pattern->dict-bitmap compilation and the golden tests.
+ * Returns 1/0/-1 (-1 = invalid sequence, the reference's boost::none). */
+extern "C" int bkgpu_like_match(const char* target, int64_t tlen,
+                                const char* pattern, int64_t plen,
+                                int charset, char escape_char) {
+    return bk_like_match(target, (size_t)tlen, pattern, (size_t)plen,
+                         charset, escape_char);
+}
+extern "C" int bkgpu_like_one(const char* target, int64_t tlen,
+                              const char* pattern, int64_t plen,
+                              int charset, char escape_char) {
+    return bk_like_one(target, (size_t)tlen, pattern, (size_t)plen,
+                       charset, escape_char);
 }

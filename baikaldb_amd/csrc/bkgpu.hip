@@ -1094,15 +1094,16 @@ __device__ __forceinline__ void histo_row(
 }
 
 /* row mapping shared by histo and scatter (MUST match: H rows are
- * per-block). Plain row-per-lane grid stride: a 2-rows/lane ILP variant
- * measured NEUTRAL on histo and -7% on scatter (the compiler does not
- * interleave the two spec-driven bodies; I$ bloat instead). */
-#define BK_PART_ROWS2(BODY_A, BODY_B)                                        \
+ * per-block). Plain row-per-lane grid stride — a 2-rows/lane ILP variant
+ * was measured NEUTRAL on histo and -7% on scatter (the compiler does not
+ * interleave the two spec-driven bodies; I$ bloat instead), so the macro
+ * takes one body. */
+#define BK_PART_ROWS(BODY)                                                   \
     int64_t gstride = (int64_t)gridDim.x * blockDim.x;                       \
     for (int64_t r = row_begin + (int64_t)blockIdx.x * blockDim.x            \
                      + threadIdx.x;                                          \
          r < row_end; r += gstride) {                                        \
-        BODY_A;                                                              \
+        BODY;                                                                \
     }
 
 template <int BS, bool HOT, bool SIMPLE = false>
@@ -1129,10 +1130,7 @@ k_part_histo(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_end,
     __syncthreads();
     const uint32_t lmask = tslots - 1;
     int64_t my_passed = 0;
-    BK_PART_ROWS2(
-        (histo_row<BS, HOT, SIMPLE>(cols, q, r, row_begin, bucketid, P, lhist,
-                            ltab, lmask, stride, lfill, lctr, lmode, lcap,
-                            hot_probe, hot_min, my_passed)),
+    BK_PART_ROWS(
         (histo_row<BS, HOT, SIMPLE>(cols, q, r, row_begin, bucketid, P, lhist,
                             ltab, lmask, stride, lfill, lctr, lmode, lcap,
                             hot_probe, hot_min, my_passed)))
@@ -1376,9 +1374,7 @@ k_part_scatter(DevCols cols, BkQuerySpec q, RecLayout lay, int64_t row_begin,
         if (paired) bstate[b] = 0;
     }
     __syncthreads();
-    BK_PART_ROWS2(
-        (scatter_row<BS, SIMPLE>(cols, q, lay, r, row_begin, bucketid, lcur,
-                         bstate, stash, rec, paired)),
+    BK_PART_ROWS(
         (scatter_row<BS, SIMPLE>(cols, q, lay, r, row_begin, bucketid, lcur,
                          bstate, stash, rec, paired)))
     /* drain leftover stashes (one half-pair per bucket at most) */
@@ -2101,6 +2097,16 @@ static hipError_t pool_alloc(void** p, size_t bytes) {
     return e;
 }
 
+/* INVARIANT (cross-stream reuse): the pool is not stream-aware — a buffer
+ * returned here may be handed to the next pool_alloc immediately. That is
+ * safe only while every engine stream is BLOCKING w.r.t. the null stream
+ * (hipStreamCreate default) AND callers synchronize a stream (or an event
+ * recorded after the last consumer, EvTimer::finish) before freeing
+ * buffers a kernel on that stream still references — run_dense /
+ * run_partitioned_pipe do exactly that before recycling per-chunk slots.
+ * Switching to hipStreamNonBlocking or a stream-ordered allocator would
+ * turn this reuse into a device use-after-free; make the pool
+ * stream-aware first. */
 static void pool_free(void* p) {
     if (!p) return;
     std::lock_guard<std::mutex> lk(g_pool_mu);
@@ -3017,6 +3023,14 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
             if (q->group_fns[k] && q->group_types[k] != BK_INT64 &&
                 q->group_types[k] != BK_DATETIME) {
                 set_err("group_fns need an int64/DATETIME key column");
+                return nullptr;
+            }
+    if (q)
+        for (int32_t j = 0; j < q->n_conjuncts; j++)
+            if (q->conjuncts[j].or_group < 0 || q->conjuncts[j].or_group > 31) {
+                /* kernels and oracle fold clause ids with & 31 — ids
+                 * congruent mod 32 would silently merge into one clause */
+                set_err("or_group out of range (0..31)");
                 return nullptr;
             }
     if (ensure_device() != 0) return nullptr;

@@ -108,6 +108,10 @@ class QueryPlan:
             col, op, lit = cjt[0], cjt[1], cjt[2]
             cj = q.conjuncts[i]
             cj.or_group = cjt[3] if len(cjt) > 3 else 0
+            if not (0 <= cj.or_group <= 31):
+                # engine + oracle fold clause ids with & 31: out-of-range
+                # ids would silently merge clauses congruent mod 32
+                raise ValueError(f"or_group {cj.or_group} out of range 0..31")
             cj.col2 = -1
             if isinstance(col, tuple) and col[0] in _ARITH:
                 # ("add"|"sub"|"mul", c1, c2): binary-arith predicate; the

@@ -124,6 +124,17 @@ int  bkgpu_agg_export_part(const BkgAggOut* o, int nparts, int part,
 /* fresh empty result for `q` (merge target for received part blobs) */
 BkgAggOut* bkgpu_agg_empty(const BkQuerySpec* q, int64_t expected_groups);
 
+/* SQL LIKE (include/bk_like.h: LikePredicate::like + like_one restated,
+ * predicate.h:502-573 / predicate.cpp:509-530) — the host compiles LIKE
+ * patterns against dictionary words into BK_OP_IN_BITMAP pushdowns with
+ * these. charset: 0 Binary, 1 UTF8, 2 GBK. bkgpu_like_match returns
+ * 1/0/-1 (-1 = invalid sequence); bkgpu_like_one applies the reference's
+ * GBK->Binary retry and returns 1/0. */
+int bkgpu_like_match(const char* target, int64_t tlen, const char* pattern,
+                     int64_t plen, int charset, char escape_char);
+int bkgpu_like_one(const char* target, int64_t tlen, const char* pattern,
+                   int64_t plen, int charset, char escape_char);
+
 /* Level-1 aggregate via SORT-based dedup: radix-sorts the spec-packed group
  * key per passing row and emits one DENSE table slot per unique key — the
  * path for high-cardinality DISTINCT (dedup cardinality ~ rows), where the

@@ -331,11 +331,14 @@ def test_in_predicates(eng, orc):
 
 def test_like_pushdown_via_dict_bitmap(eng, orc):
     """LIKE on a dictionary VARCHAR pushes down as a code-accept bitmap
-    (BK_OP_IN_BITMAP): the host matches the pattern against the dictionary
-    once, the engine filters by code membership (cstore-dict pushdown)."""
-    import fnmatch
+    (BK_OP_IN_BITMAP): the host compiles the SQL LIKE pattern ('%'/'_'/
+    escape — LikePredicate::like semantics, include/bk_like.h, pinned
+    against test_predicate.cpp's vectors in test_like_golden.py) against
+    the dictionary once, the engine filters by code membership (the
+    cstore-dict pushdown)."""
     import numpy as np
     from baikaldb_amd import QueryPlan
+    from baikaldb_amd.like import like_accept_codes
     from oracle.bindings import make_query as mq
 
     n = 200_000
@@ -343,14 +346,14 @@ def test_like_pushdown_via_dict_bitmap(eng, orc):
     specs = [(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
              (TYPE_STRING, D_DICT, nwords, 0, 0),
              (TYPE_INT64, D_UNI, 0, 1000, 0)]
-    # dict words are deterministic from the seed: LIKE 'w%3_%' etc.
-    pattern = "w*1*"  # SQL: LIKE 'w%1%'
+    # dict words are deterministic from the seed: SQL LIKE 'w%1%'
+    pattern = b"w%1%"
+    words = [orc.dict_word(SEED, c) for c in range(nwords)]
+    codes = like_accept_codes(words, pattern)
     accept = bytearray((nwords + 7) // 8)
-    naccept = 0
-    for c in range(nwords):
-        if fnmatch.fnmatch(orc.dict_word(SEED, c), pattern):
-            accept[c >> 3] |= 1 << (c & 7)
-            naccept += 1
+    for c in codes:
+        accept[c >> 3] |= 1 << (c & 7)
+    naccept = len(codes)
     assert 0 < naccept < nwords
     accept = bytes(accept)
 
