@@ -11,9 +11,11 @@
 // this is the product path an embedder links (libbkgpu.so + headers).
 #include <cstdio>
 #include <cstring>
+#include <string>
 #include <vector>
 
 #include "../include/bk_exec.h"
+#include "../include/bk_arrow.h"
 
 static int run_agg(BkgTable* table) {
     BkPlanNodeDesc plan[3];
@@ -220,9 +222,129 @@ static int run_distinct(BkgTable* table) {
     return 0;
 }
 
+/* run the same GROUP BY tree and serialize the result batch as an Arrow
+ * IPC stream — the vectorized-result bytes Region::select returns to the
+ * frontend (region.cpp:2905-2918), written here by the from-scratch
+ * serializer (include/bk_arrow.h); pyarrow round-trips the file in
+ * tests/test_parquet.py. */
+static int run_agg_arrow(BkgTable* table, const char* out_path) {
+    BkPlanNodeDesc plan[3];
+    memset(plan, 0, sizeof plan);
+    plan[0].node_type = BK_AGG_NODE;
+    plan[0].num_children = 1;
+    plan[0].limit = -1;
+    plan[0].n_group = 1;
+    plan[0].group_cols[0] = 0;
+    plan[0].n_aggs = 3;
+    plan[0].aggs[0] = {BK_AGG_COUNT_STAR, -1};
+    plan[0].aggs[1] = {BK_AGG_SUM, 2};
+    plan[0].aggs[2] = {BK_AGG_MIN, 1};
+    plan[0].expected_groups = 1 << 8;
+    plan[1].node_type = BK_WHERE_FILTER_NODE;
+    plan[1].num_children = 1;
+    plan[1].limit = -1;
+    plan[1].n_conjuncts = 1;
+    plan[1].conjuncts[0].col = 2;
+    plan[1].conjuncts[0].op = BK_OP_LT;
+    plan[1].conjuncts[0].cmp_type = BK_INT64;
+    plan[1].conjuncts[0].lit_i = 80;
+    plan[2].node_type = BK_SCAN_NODE;
+    plan[2].limit = -1;
+    plan[2].table = table;
+
+    BkExecTree* t = bkexec_create_tree(plan, 3);
+    if (!t) { fprintf(stderr, "create_tree: %s\n", bkgpu_last_error()); return 1; }
+    if (bkexec_open(t) < 0) { fprintf(stderr, "open failed\n"); return 1; }
+    int ns = bkexec_n_slots(t);
+    std::vector<int32_t> tag(64 * ns);
+    std::vector<int64_t> vi(64 * ns);
+    std::vector<double> vd(64 * ns);
+    std::vector<uint8_t> nul(64 * ns);
+    /* collect the whole result columnar (the batch the store would hand to
+     * SerializeRecordBatch) */
+    std::vector<std::vector<int64_t>> ci(ns);
+    std::vector<std::vector<double>> cd(ns);
+    std::vector<std::vector<uint8_t>> cv(ns);
+    std::vector<int32_t> ctype(ns, BK_INT64);
+    int eos = 0;
+    while (!eos) {
+        int64_t n = bkexec_get_next(t, 64, tag.data(), vi.data(), vd.data(),
+                                    nul.data(), &eos);
+        if (n < 0) { fprintf(stderr, "get_next failed\n"); return 1; }
+        for (int64_t r = 0; r < n; r++)
+            for (int s = 0; s < ns; s++) {
+                size_t i = (size_t)r * ns + s;
+                if (!nul[i]) ctype[s] = tag[i];
+                ci[s].push_back(vi[i]);
+                cd[s].push_back(vd[i]);
+                cv[s].push_back(nul[i] ? 0 : 1);
+            }
+    }
+    int64_t nrows = ci.empty() ? 0 : (int64_t)ci[0].size();
+    /* dictionary words for string slots (codes -> words) */
+    std::vector<std::vector<std::string>> wstore(ns);
+    std::vector<std::vector<const char*>> wptr(ns);
+    std::vector<BkArrowCol> cols(ns);
+    std::vector<std::string> names(ns);
+    std::vector<const char*> nameptr(ns);
+    std::vector<std::vector<int32_t>> codes32(ns);
+    for (int s = 0; s < ns; s++) {
+        names[s] = "c" + std::to_string(s);
+        nameptr[s] = names[s].c_str();
+        cols[s].col_type = ctype[s];
+        cols[s].valid = cv[s].data();
+        cols[s].words = nullptr;
+        cols[s].nwords = 0;
+        if (ctype[s] == BK_DOUBLE) {
+            cols[s].data = cd[s].data();
+        } else if (ctype[s] == BK_STRING) {
+            int64_t maxc = -1;
+            for (int64_t r = 0; r < nrows; r++)
+                if (cv[s][r] && ci[s][r] > maxc) maxc = ci[s][r];
+            char word[256];
+            /* slot 0 = the group key (table col 0); other string slots in
+             * this plan come from MIN(w) on table col 1 (same mapping the
+             * row printer uses) */
+            int src = s == 0 ? 0 : 1;
+            for (int64_t c = 0; c <= maxc; c++) {
+                if (bkgpu_table_dict_word(table, src, c, word,
+                                          sizeof word) >= 0)
+                    wstore[s].push_back(word);
+                else
+                    wstore[s].push_back("");
+            }
+            for (auto& w : wstore[s]) wptr[s].push_back(w.c_str());
+            codes32[s].resize(nrows);
+            for (int64_t r = 0; r < nrows; r++)
+                codes32[s][r] = (int32_t)ci[s][r];
+            cols[s].data = codes32[s].data();
+            cols[s].words = wptr[s].data();
+            cols[s].nwords = (int64_t)wstore[s].size();
+        } else {
+            cols[s].data = ci[s].data();
+        }
+    }
+    void* buf = nullptr;
+    int64_t blen = 0;
+    if (bk_arrow_ipc_stream(ns, cols.data(), nrows, nameptr.data(), &buf,
+                            &blen) != 0) {
+        fprintf(stderr, "arrow serialize failed\n");
+        return 1;
+    }
+    FILE* f = fopen(out_path, "wb");
+    if (!f) { fprintf(stderr, "cannot write %s\n", out_path); return 1; }
+    fwrite(buf, 1, (size_t)blen, f);
+    fclose(f);
+    bk_arrow_free(buf);
+    printf("arrow ipc: %lld rows, %lld bytes -> %s\n", (long long)nrows,
+           (long long)blen, out_path);
+    bkexec_close(t);
+    return 0;
+}
+
 int main(int argc, char** argv) {
-    if (argc != 2) {
-        fprintf(stderr, "usage: %s <file.parquet>\n", argv[0]);
+    if (argc != 2 && argc != 3) {
+        fprintf(stderr, "usage: %s <file.parquet> [out.arrow]\n", argv[0]);
         return 2;
     }
     if (bkgpu_device_count() < 1) {
@@ -240,6 +362,7 @@ int main(int argc, char** argv) {
     if (rc == 0) rc = run_sort(table);
     if (rc == 0) rc = run_window(table);
     if (rc == 0) rc = run_distinct(table);
+    if (rc == 0 && argc == 3) rc = run_agg_arrow(table, argv[2]);
     bkgpu_table_free(table);
     return rc;
 }

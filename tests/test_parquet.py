@@ -287,8 +287,9 @@ def test_cpp_embedding_example(tmp_path):
     tab = pa.table({"g": pa.array(g), "w": pa.array(w), "v": pa.array(v)})
     pq.write_table(tab, path, compression=None, use_dictionary=True,
                    data_page_version="1.0", write_statistics=False)
-    out = subprocess.run([exe, path], capture_output=True, text=True,
-                         timeout=120)
+    arrow_out = str(tmp_path / "result.arrow")
+    out = subprocess.run([exe, path, arrow_out], capture_output=True,
+                         text=True, timeout=120)
     assert out.returncode == 0, out.stderr
     assert f"ingested {n} rows x 3 cols" in out.stdout
     # verify one group row against a brute-force recompute
@@ -300,6 +301,21 @@ def test_cpp_embedding_example(tmp_path):
     assert expect in out.stdout, f"wanted {expect!r} in:\n{out.stdout[:2000]}"
     assert "top-5 by (v, g):" in out.stdout
     assert "row_number=1" in out.stdout
+    # the example also serializes the GROUP BY result as an Arrow IPC
+    # stream (the region.cpp:2905-2918 response bytes, written by the
+    # from-scratch serializer) — pyarrow must round-trip it and agree
+    # with a pandas-free brute-force recompute
+    assert "arrow ipc:" in out.stdout
+    t = pa.ipc.open_stream(open(arrow_out, "rb").read()).read_all()
+    t.validate(full=True)
+    got = {t.column(0)[i].as_py(): (t.column(1)[i].as_py(),
+                                    t.column(2)[i].as_py(),
+                                    t.column(3)[i].as_py())
+           for i in range(t.num_rows)}
+    for grp in set(garr[sel]):
+        rows_g = sel & (garr == grp)
+        assert got[grp] == (int(rows_g.sum()), int(v[rows_g].sum()),
+                            min(warr[rows_g])), grp
 
 
 @pytest.mark.gpu
