@@ -148,6 +148,84 @@ __device__ __forceinline__ bool in_list_hit(const BkConjunct& cj, int64_t v) {
     return false;
 }
 
+
+/* ---- postfix expression programs (bk_common.h BkExprOp) ----
+ * ScalarFnCall::get_value's arbitrary trees (scalar_fn_call.cpp:194-225),
+ * flattened by the planner; ARITH ops compute in their declared domain
+ * (the reference's arg-cast rule, scalar_fn_call.cpp:219-225): int64
+ * wraps like operators.cpp, DOUBLE in IEEE f64; any NULL operand => NULL.
+ * __noinline__: the runtime-indexed operand stack lives in scratch — kept
+ * out of line so it cannot drag the callers' hot loops with it (generic
+ * shapes only; SIMPLE and the eager/dense fast paths exclude programs). */
+struct PVal { bool valid; int64_t i; double d; };
+__device__ __noinline__ PVal eval_prog(const DevCols& cols,
+                                       const BkQuerySpec& q, int32_t begin,
+                                       int32_t len, int64_t r) {
+    int64_t si[BK_MAX_PROG_DEPTH];
+    double sd[BK_MAX_PROG_DEPTH];
+    bool sv[BK_MAX_PROG_DEPTH];
+    int sp = 0;
+    for (int32_t k = 0; k < len; k++) {
+        const BkExprOp& e = q.prog[begin + k];
+        switch (e.op) {
+            case BK_PROG_COL: {
+                const DevCol& c = cols.c[e.arg];
+                sv[sp] = cell_valid(c, r);
+                if (c.type == BK_DOUBLE) {
+                    sd[sp] = cell_f64(c, r);
+                    si[sp] = (int64_t)sd[sp];
+                } else {
+                    si[sp] = cell_i64(c, r);
+                    sd[sp] = (double)si[sp];
+                }
+                sp++;
+            } break;
+            case BK_PROG_LIT_I:
+                si[sp] = e.lit_i;
+                sd[sp] = (double)e.lit_i;
+                sv[sp] = true;
+                sp++;
+                break;
+            case BK_PROG_LIT_D:
+                sd[sp] = e.lit_d;
+                si[sp] = (int64_t)e.lit_d;
+                sv[sp] = true;
+                sp++;
+                break;
+            case BK_PROG_ARITH: {
+                sp--;
+                bool v = sv[sp - 1] && sv[sp];
+                if (e.domain == BK_DOUBLE) {
+                    double a = sd[sp - 1], b = sd[sp];
+                    double o = e.arg == BK_ARITH_ADD   ? a + b
+                               : e.arg == BK_ARITH_SUB ? a - b
+                                                       : a * b;
+                    sd[sp - 1] = o;
+                    si[sp - 1] = (int64_t)o;
+                } else {
+                    uint64_t a = (uint64_t)si[sp - 1], b = (uint64_t)si[sp];
+                    int64_t o = (int64_t)(e.arg == BK_ARITH_ADD   ? a + b
+                                          : e.arg == BK_ARITH_SUB ? a - b
+                                                                  : a * b);
+                    si[sp - 1] = o;
+                    sd[sp - 1] = (double)o;
+                }
+                sv[sp - 1] = v;
+            } break;
+            default: {                      /* BK_PROG_FN (int64 domain) */
+                int64_t o = bk_scalar_fn(e.arg, si[sp - 1]);
+                si[sp - 1] = o;
+                sd[sp - 1] = (double)o;
+            } break;
+        }
+    }
+    PVal out;
+    out.valid = sv[0];
+    out.i = si[0];
+    out.d = sd[0];
+    return out;
+}
+
 template <bool SIMPLE = false>
 __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpec& q,
                                            int64_t r) {
@@ -160,7 +238,12 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
      * ~20% of the streaming ceiling (bwprobe predk vs preds) */
     #define BK_EVAL1(J, VI, VD, OK)                                         \
         int64_t VI = 0; double VD = 0.0; bool OK = true;                    \
-        if (SIMPLE || q.n_conjuncts > (J)) {                                \
+        if ((SIMPLE || q.n_conjuncts > (J)) && !SIMPLE &&                   \
+            q.conjuncts[J].prog_len > 0) {                                  \
+            PVal p_ = eval_prog(cols, q, q.conjuncts[J].prog_begin,         \
+                                q.conjuncts[J].prog_len, r);                \
+            OK = p_.valid; VI = p_.i; VD = p_.d;                            \
+        } else if (SIMPLE || q.n_conjuncts > (J)) {                         \
             const BkConjunct& cj = q.conjuncts[J];                          \
             const DevCol& c = cols.c[cj.col];                               \
             if (!SIMPLE) OK = cell_valid(c, r);                             \
@@ -245,6 +328,18 @@ __device__ __forceinline__ bool row_passes(const DevCols& cols, const BkQuerySpe
     #undef BK_TEST1
     for (int32_t j = 4; !SIMPLE && j < q.n_conjuncts && pass_all; j++) {
         const BkConjunct& cj = q.conjuncts[j];
+        if (cj.prog_len > 0) {
+            PVal p_ = eval_prog(cols, q, cj.prog_begin, cj.prog_len, r);
+            bool pass = false;
+            if (p_.valid) {
+                int cmp = cj.cmp_type == BK_DOUBLE
+                              ? ((p_.d > cj.lit_d) - (p_.d < cj.lit_d))
+                              : ((p_.i > cj.lit_i) - (p_.i < cj.lit_i));
+                pass = (0x19D2Au >> (3 * cj.op + cmp + 1)) & 1u;
+            }
+            BK_CLAUSE(j, pass)
+            continue;
+        }
         const DevCol& c = cols.c[cj.col];
         if (!cell_valid(c, r)) {
             if (cj.or_group == 0) return false;
@@ -333,9 +428,17 @@ __device__ __forceinline__ uint64_t enc_value_nf(const DevCol& c, int64_t r) {
  * cast semantics) */
 struct AggIn { bool valid; int64_t i; double d; };
 __device__ __forceinline__ AggIn agg_input(const DevCols& cols,
+                                           const BkQuerySpec& q,
                                            const BkAggSpec& as,
                                            int32_t in_type, int64_t r) {
     AggIn o{true, 0, 0.0};
+    if (as.prog_len > 0) {                 /* postfix expression input */
+        PVal p = eval_prog(cols, q, as.prog_begin, as.prog_len, r);
+        o.valid = p.valid;
+        o.i = p.i;
+        o.d = p.d;
+        return o;
+    }
     const DevCol& c = cols.c[as.col];
     o.valid = cell_valid(c, r);
     if (as.arith) {
@@ -366,7 +469,7 @@ __device__ __forceinline__ uint64_t agg_enc(const DevCols& cols,
                                             const BkAggSpec& as,
                                             int32_t in_type, const AggIn& v,
                                             int64_t r) {
-    if (!as.arith) return enc_value(cols.c[as.col], r);
+    if (!as.arith && as.prog_len == 0) return enc_value(cols.c[as.col], r);
     return in_type == BK_DOUBLE ? bk_enc_f64(v.d) : bk_enc_i64(v.i);
 }
 
@@ -473,12 +576,12 @@ __device__ __forceinline__ void agg_update_slot(uint64_t* st, const BkQuerySpec&
                 atomicAdd((unsigned long long*)val, 1ull);
                 break;
             case BK_AGG_COUNT: {
-                if (agg_input(cols, as, q.agg_in_types[a], r).valid)
+                if (agg_input(cols, q, as, q.agg_in_types[a], r).valid)
                     atomicAdd((unsigned long long*)val, 1ull);
                 break;
             }
             case BK_AGG_SUM: {
-                AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                 if (!v.valid) break;
                 if (q.agg_in_types[a] == BK_DOUBLE) {
                     if (LDS) atomic_add_f64_lds(val, v.d);
@@ -491,7 +594,7 @@ __device__ __forceinline__ void agg_update_slot(uint64_t* st, const BkQuerySpec&
                 break;
             }
             case BK_AGG_AVG: {
-                AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                 if (!v.valid) break;
                 if (LDS) atomic_add_f64_lds(val, v.d);
                 else     atomic_add_f64_global(val, v.d);
@@ -499,7 +602,7 @@ __device__ __forceinline__ void agg_update_slot(uint64_t* st, const BkQuerySpec&
                 break;
             }
             case BK_AGG_MIN: {
-                AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                 if (!v.valid) break;
                 atomicMax((unsigned long long*)val,
                           (unsigned long long)(~agg_enc(cols, as,
@@ -508,7 +611,7 @@ __device__ __forceinline__ void agg_update_slot(uint64_t* st, const BkQuerySpec&
                 break;
             }
             case BK_AGG_MAX: {
-                AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                 if (!v.valid) break;
                 atomicMax((unsigned long long*)val,
                           (unsigned long long)agg_enc(cols, as,
@@ -777,11 +880,11 @@ k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_
             switch (as.agg_type) {
                 case BK_AGG_COUNT_STAR: acc_v[a]++; break;
                 case BK_AGG_COUNT:
-                    if (agg_input(cols, as, q.agg_in_types[a], r).valid)
+                    if (agg_input(cols, q, as, q.agg_in_types[a], r).valid)
                         acc_v[a]++;
                     break;
                 case BK_AGG_SUM: {
-                    AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                    AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                     if (!v.valid) break;
                     if (q.agg_in_types[a] == BK_DOUBLE) acc_d[a] += v.d;
                     else acc_v[a] += (uint64_t)v.i;
@@ -789,14 +892,14 @@ k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_
                     break;
                 }
                 case BK_AGG_AVG: {
-                    AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                    AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                     if (!v.valid) break;
                     acc_d[a] += v.d;
                     acc_c[a]++;
                     break;
                 }
                 case BK_AGG_MIN: {
-                    AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                    AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                     if (!v.valid) break;
                     uint64_t e = ~agg_enc(cols, as, q.agg_in_types[a], v, r);
                     if (e > acc_v[a]) acc_v[a] = e;
@@ -804,7 +907,7 @@ k_filter_agg_scalar(DevCols cols, BkQuerySpec q, int64_t row_begin, int64_t row_
                     break;
                 }
                 case BK_AGG_MAX: {
-                    AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                    AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                     if (!v.valid) break;
                     uint64_t e = agg_enc(cols, as, q.agg_in_types[a], v, r);
                     if (e > acc_v[a]) acc_v[a] = e;
@@ -924,11 +1027,11 @@ k_filter_agg_wcomb(DevCols cols, BkQuerySpec q, int64_t row_begin,
                     switch (at) {
                         case BK_AGG_COUNT_STAR: tv = 1; break;
                         case BK_AGG_COUNT:
-                            tv = agg_input(cols, as, q.agg_in_types[a], r)
+                            tv = agg_input(cols, q, as, q.agg_in_types[a], r)
                                      .valid ? 1 : 0;
                             break;
                         case BK_AGG_SUM: {
-                            AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                            AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                             if (!v.valid) break;
                             if (q.agg_in_types[a] == BK_DOUBLE) td = v.d;
                             else tv = (uint64_t)v.i;
@@ -936,21 +1039,21 @@ k_filter_agg_wcomb(DevCols cols, BkQuerySpec q, int64_t row_begin,
                             break;
                         }
                         case BK_AGG_AVG: {
-                            AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                            AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                             if (!v.valid) break;
                             td = v.d;
                             tc = 1;
                             break;
                         }
                         case BK_AGG_MIN: {
-                            AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                            AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                             if (!v.valid) break;
                             tv = ~agg_enc(cols, as, q.agg_in_types[a], v, r);
                             tc = 1;
                             break;
                         }
                         case BK_AGG_MAX: {
-                            AggIn v = agg_input(cols, as, q.agg_in_types[a], r);
+                            AggIn v = agg_input(cols, q, as, q.agg_in_types[a], r);
                             if (!v.valid) break;
                             tv = agg_enc(cols, as, q.agg_in_types[a], v, r);
                             tc = 1;
@@ -1263,7 +1366,7 @@ __device__ __forceinline__ void build_record(
     for (int32_t a = 0; a < q.n_aggs; a++) {
         if (lay.val_word[a] < 0) continue;
         const BkAggSpec& as = q.aggs[a];
-        AggIn v = agg_input(cols, q.aggs[a], q.agg_in_types[a], r);
+        AggIn v = agg_input(cols, q, q.aggs[a], q.agg_in_types[a], r);
         if (v.valid) meta |= (uint64_t)1 << (8 + a);
         uint64_t w = 0;
         if (v.valid) {
@@ -1288,7 +1391,7 @@ __device__ __forceinline__ void build_record(
         /* COUNT(col) validity for aggs without a val word */
         for (int32_t a = 0; a < q.n_aggs; a++) {
             if (lay.val_word[a] >= 0 || q.aggs[a].col < 0) continue;
-            if (agg_input(cols, q.aggs[a], q.agg_in_types[a], r).valid)
+            if (agg_input(cols, q, q.aggs[a], q.agg_in_types[a], r).valid)
                 meta |= (uint64_t)1 << (8 + a);
         }
         regs[lay.meta_word] = meta;
@@ -2580,7 +2683,7 @@ static bool query_simple(const BkgTable* t, const BkQuerySpec* q) {
     for (int32_t j = 0; j < q->n_conjuncts; j++) {
         const BkConjunct& cj = q->conjuncts[j];
         if (cj.op >= BK_OP_IN || cj.fn || cj.cmp_type == BK_DOUBLE ||
-            cj.or_group || cj.arith)
+            cj.or_group || cj.arith || cj.prog_len)
             return false;
         if (t->valid[cj.col]) return false;
     }
@@ -3033,6 +3136,50 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
                 set_err("or_group out of range (0..31)");
                 return nullptr;
             }
+    if (q) {
+        /* postfix-program validation: indices in the pool, operand stack
+         * within [1, BK_MAX_PROG_DEPTH], exactly one result */
+        auto bad_prog = [&](int32_t begin, int32_t len) -> bool {
+            if (len == 0) return false;
+            if (begin < 0 || len < 0 || q->n_prog > BK_MAX_PROG_POOL ||
+                begin + len > q->n_prog)
+                return true;
+            int sp = 0;
+            for (int32_t k = 0; k < len; k++) {
+                const BkExprOp& e = q->prog[begin + k];
+                switch (e.op) {
+                    case BK_PROG_COL:
+                        if (e.arg < 0 || e.arg >= t->ncols) return true;
+                        /* fallthrough */
+                    case BK_PROG_LIT_I:
+                    case BK_PROG_LIT_D:
+                        if (++sp > BK_MAX_PROG_DEPTH) return true;
+                        break;
+                    case BK_PROG_ARITH:
+                        if (sp < 2) return true;
+                        sp--;
+                        break;
+                    case BK_PROG_FN:
+                        if (sp < 1) return true;
+                        break;
+                    default:
+                        return true;
+                }
+            }
+            return sp != 1;
+        };
+        for (int32_t j = 0; j < q->n_conjuncts; j++)
+            if (bad_prog(q->conjuncts[j].prog_begin,
+                         q->conjuncts[j].prog_len)) {
+                set_err("bad conjunct expression program");
+                return nullptr;
+            }
+        for (int32_t a = 0; a < q->n_aggs; a++)
+            if (bad_prog(q->aggs[a].prog_begin, q->aggs[a].prog_len)) {
+                set_err("bad aggregate expression program");
+                return nullptr;
+            }
+    }
     if (ensure_device() != 0) return nullptr;
     double t_start = debug_timing() ? now_ms() : 0;
     double t_alloc = 0, t_pipe = 0;

@@ -32,12 +32,89 @@ class BkConjunct(C.Structure):
                 ("lit_i", C.c_int64), ("lit_d", C.c_double),
                 ("in_list", C.c_int64 * 16),
                 ("fn", C.c_int32), ("or_group", C.c_int32),
-                ("col2", C.c_int32), ("arith", C.c_int32)]
+                ("col2", C.c_int32), ("arith", C.c_int32),
+                ("prog_begin", C.c_int32), ("prog_len", C.c_int32)]
 
 
 class BkAggSpec(C.Structure):
     _fields_ = [("agg_type", C.c_int32), ("col", C.c_int32),
-                ("col2", C.c_int32), ("arith", C.c_int32)]
+                ("col2", C.c_int32), ("arith", C.c_int32),
+                ("prog_begin", C.c_int32), ("prog_len", C.c_int32)]
+
+
+class BkExprOp(C.Structure):
+    _fields_ = [("op", C.c_int32), ("arg", C.c_int32),
+                ("domain", C.c_int32), ("_pad", C.c_int32),
+                ("lit_i", C.c_int64), ("lit_d", C.c_double)]
+
+
+BK_MAX_PROG_POOL = 24
+PROG_COL, PROG_LIT_I, PROG_LIT_D, PROG_ARITH, PROG_FN = 0, 1, 2, 3, 4
+
+
+def compile_expr(e, col_types, pool):
+    """RPN-compile a nested expression tree into `pool` (list of op dicts);
+    returns the expression's domain type. This is the planner half of the
+    engine's postfix programs: ScalarFnCall::get_value walks such trees per
+    row (scalar_fn_call.cpp:194-225); the compute domain of every ARITH
+    node follows the reference's arg-cast rule (children cast to the fn's
+    arg types, scalar_fn_call.cpp:219-225 — DOUBLE iff either child is).
+
+    Operands: int = column index; float = double literal;
+    ("liti", v) / ("litf", v) explicit literals;
+    ("add"|"sub"|"mul", a, b); scalar fns ("year"|..., a)."""
+    if isinstance(e, bool):
+        raise ValueError("bool is not an expression operand")
+    if isinstance(e, int):
+        pool.append(dict(op=PROG_COL, arg=e))
+        return TYPE_DOUBLE if col_types[e] == TYPE_DOUBLE else TYPE_INT64
+    if isinstance(e, float):
+        pool.append(dict(op=PROG_LIT_D, lit_d=e))
+        return TYPE_DOUBLE
+    tag = e[0]
+    if tag == "liti":
+        pool.append(dict(op=PROG_LIT_I, lit_i=int(e[1])))
+        return TYPE_INT64
+    if tag == "litf":
+        pool.append(dict(op=PROG_LIT_D, lit_d=float(e[1])))
+        return TYPE_DOUBLE
+    if tag in _ARITH:
+        d1 = compile_expr(e[1], col_types, pool)
+        d2 = compile_expr(e[2], col_types, pool)
+        dom = TYPE_DOUBLE if TYPE_DOUBLE in (d1, d2) else TYPE_INT64
+        pool.append(dict(op=PROG_ARITH, arg=_ARITH[tag], domain=dom))
+        return dom
+    if tag in _FNS:
+        compile_expr(e[1], col_types, pool)
+        pool.append(dict(op=PROG_FN, arg=_FNS[tag]))
+        return TYPE_INT64
+    raise ValueError(f"bad expression node: {e!r}")
+
+
+def expr_is_deep(e):
+    """True when the tuple expression needs a postfix program (anything
+    beyond the legacy one-arith / one-fn shapes with plain column refs)."""
+    if not isinstance(e, tuple):
+        return False
+    if e[0] in ("liti", "litf"):
+        return True
+    return any(isinstance(x, (tuple, float)) for x in e[1:])
+
+
+def spec_add_prog(q, ops):
+    """Write compiled ops into q.prog; returns (begin, len)."""
+    begin = q.n_prog
+    if begin + len(ops) > BK_MAX_PROG_POOL:
+        raise ValueError("expression program pool overflow")
+    for i, o in enumerate(ops):
+        dst = q.prog[begin + i]
+        dst.op = o.get("op", 0)
+        dst.arg = o.get("arg", 0)
+        dst.domain = o.get("domain", TYPE_INT64)
+        dst.lit_i = o.get("lit_i", 0)
+        dst.lit_d = o.get("lit_d", 0.0)
+    q.n_prog = begin + len(ops)
+    return begin, len(ops)
 
 
 _WINFNS = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4,
@@ -69,7 +146,9 @@ class BkQuerySpec(C.Structure):
                 ("group_base", C.c_int64 * BK_MAX_GROUP),
                 ("group_fns", C.c_int32 * BK_MAX_GROUP),
                 ("aggs", BkAggSpec * BK_MAX_AGGS),
-                ("agg_in_types", C.c_int32 * BK_MAX_AGGS)]
+                ("agg_in_types", C.c_int32 * BK_MAX_AGGS),
+                ("n_prog", C.c_int32), ("_pad2", C.c_int32),
+                ("prog", BkExprOp * BK_MAX_PROG_POOL)]
 
 
 class QueryPlan:
@@ -113,6 +192,19 @@ class QueryPlan:
                 # ids would silently merge clauses congruent mod 32
                 raise ValueError(f"or_group {cj.or_group} out of range 0..31")
             cj.col2 = -1
+            if expr_is_deep(col):
+                ops = []
+                dom = compile_expr(col, self.col_types, ops)
+                cj.prog_begin, cj.prog_len = spec_add_prog(q, ops)
+                cj.col = 0
+                cj.op = _OPS[op] if isinstance(op, str) else op
+                if dom == TYPE_DOUBLE or isinstance(lit, float):
+                    cj.cmp_type = TYPE_DOUBLE
+                    cj.lit_d = float(lit)
+                else:
+                    cj.cmp_type = TYPE_INT64
+                    cj.lit_i = int(lit)
+                continue
             if isinstance(col, tuple) and col[0] in _ARITH:
                 # ("add"|"sub"|"mul", c1, c2): binary-arith predicate; the
                 # compare domain is DOUBLE iff either column is DOUBLE
@@ -161,6 +253,14 @@ class QueryPlan:
         for i, (name, col) in enumerate(self.aggs):
             q.aggs[i].agg_type = _AGGS[name] if isinstance(name, str) else name
             q.aggs[i].col2 = -1
+            if expr_is_deep(col):
+                ops = []
+                dom = compile_expr(col, self.col_types, ops)
+                (q.aggs[i].prog_begin,
+                 q.aggs[i].prog_len) = spec_add_prog(q, ops)
+                q.aggs[i].col = 0
+                q.agg_in_types[i] = dom
+                continue
             if isinstance(col, tuple):
                 # expression input: ("add"|"sub"|"mul", a, b) — the domain is
                 # DOUBLE iff either operand is DOUBLE (AggFnCall input cast,

@@ -170,6 +170,10 @@ typedef struct BkConjunct {
                           IEEE f64 (cmp_type decides the domain). Either
                           operand NULL => NULL. */
     int32_t arith;     /* BkArith */
+    /* deeper expression LHS: a postfix program in BkQuerySpec.prog
+     * (prog_len > 0 replaces col/col2/fn; the compare stays cmp_type) */
+    int32_t prog_begin;
+    int32_t prog_len;
 } BkConjunct;
 
 typedef enum BkArith {
@@ -191,7 +195,39 @@ typedef struct BkAggSpec {
                           agg_fn_call.cpp:496-555). NULL if either operand
                           NULL. */
     int32_t arith;     /* BkArith */
+    /* deeper expression input: postfix program (replaces col/col2 when
+     * prog_len > 0; agg_in_types[] still carries the result domain) */
+    int32_t prog_begin;
+    int32_t prog_len;
 } BkAggSpec;
+
+/* ---- postfix (RPN) expression programs ----
+ * ScalarFnCall::get_value walks arbitrary expression trees per row
+ * (src/expr/scalar_fn_call.cpp:194-225); the planner flattens such trees
+ * into these programs. Each ARITH op declares its COMPUTE DOMAIN,
+ * mirroring the reference's arg-cast rule (children cast to the fn's
+ * declared arg types, scalar_fn_call.cpp:219-225): INT64 ops wrap like
+ * operators.cpp, DOUBLE ops compute IEEE f64 with int operands cast.
+ * Any NULL operand makes the result NULL. */
+typedef enum BkProgOp {
+    BK_PROG_COL   = 0,   /* push column value (arg = column index) */
+    BK_PROG_LIT_I = 1,   /* push int64 literal */
+    BK_PROG_LIT_D = 2,   /* push double literal */
+    BK_PROG_ARITH = 3,   /* pop b, a -> push a OP b (arg = BkArith) */
+    BK_PROG_FN    = 4,   /* pop a -> push fn(a) (arg = BkScalarFn, int64) */
+} BkProgOp;
+
+typedef struct BkExprOp {
+    int32_t op;        /* BkProgOp */
+    int32_t arg;
+    int32_t domain;    /* BK_INT64 or BK_DOUBLE (ARITH compute domain) */
+    int32_t _pad;
+    int64_t lit_i;
+    double  lit_d;
+} BkExprOp;
+
+#define BK_MAX_PROG_POOL 24   /* ops shared by all programs of one query */
+#define BK_MAX_PROG_DEPTH 6   /* max operand-stack depth */
 
 /* ---- window functions (reference src/expr/window_fn_call.cpp:20-38 name
  * map; executed in the reference's NON-FRAME mode, window_node.cpp:39-41 —
@@ -274,6 +310,10 @@ typedef struct BkQuerySpec {
     int32_t    group_fns[BK_MAX_GROUP];
     BkAggSpec  aggs[BK_MAX_AGGS];
     int32_t    agg_in_types[BK_MAX_AGGS];  /* BkType of each agg input col */
+    /* shared postfix-program pool (BkConjunct/BkAggSpec prog_begin/len) */
+    int32_t    n_prog;
+    int32_t    _pad2;
+    BkExprOp   prog[BK_MAX_PROG_POOL];
 } BkQuerySpec;
 
 #ifdef __cplusplus

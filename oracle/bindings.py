@@ -30,12 +30,23 @@ class BkConjunct(C.Structure):
                 ("lit_i", C.c_int64), ("lit_d", C.c_double),
                 ("in_list", C.c_int64 * 16),
                 ("fn", C.c_int32), ("or_group", C.c_int32),
-                ("col2", C.c_int32), ("arith", C.c_int32)]
+                ("col2", C.c_int32), ("arith", C.c_int32),
+                ("prog_begin", C.c_int32), ("prog_len", C.c_int32)]
 
 
 class BkAggSpec(C.Structure):
     _fields_ = [("agg_type", C.c_int32), ("col", C.c_int32),
-                ("col2", C.c_int32), ("arith", C.c_int32)]
+                ("col2", C.c_int32), ("arith", C.c_int32),
+                ("prog_begin", C.c_int32), ("prog_len", C.c_int32)]
+
+
+class BkExprOp(C.Structure):
+    _fields_ = [("op", C.c_int32), ("arg", C.c_int32),
+                ("domain", C.c_int32), ("_pad", C.c_int32),
+                ("lit_i", C.c_int64), ("lit_d", C.c_double)]
+
+
+BK_MAX_PROG_POOL = 24
 
 
 class BkWindowFn(C.Structure):
@@ -60,7 +71,9 @@ class BkQuerySpec(C.Structure):
                 ("group_base", C.c_int64 * BK_MAX_GROUP),
                 ("group_fns", C.c_int32 * BK_MAX_GROUP),
                 ("aggs", BkAggSpec * BK_MAX_AGGS),
-                ("agg_in_types", C.c_int32 * BK_MAX_AGGS)]
+                ("agg_in_types", C.c_int32 * BK_MAX_AGGS),
+                ("n_prog", C.c_int32), ("_pad2", C.c_int32),
+                ("prog", BkExprOp * BK_MAX_PROG_POOL)]
 
 
 class _OrcCol(C.Structure):
@@ -84,11 +97,23 @@ def make_query(conjuncts=(), group=(), aggs=(), col_types=None,
     aggs:      list of (agg_type, col)  (col=-1 for COUNT(*))
     col_types: list of BkType per table column (needed for group/agg typing)
     """
+    from baikaldb_amd.plan import (compile_expr, expr_is_deep,
+                                   spec_add_prog)
     q = BkQuerySpec()
     q.n_conjuncts = len(conjuncts)
     for i, cjt in enumerate(conjuncts):
         col, op, cmp_type, lit = cjt[:4]
         cj = q.conjuncts[i]
+        if expr_is_deep(col):
+            ops = []
+            compile_expr(col, col_types, ops)
+            cj.prog_begin, cj.prog_len = spec_add_prog(q, ops)
+            cj.op, cj.cmp_type = op, cmp_type
+            if cmp_type == TYPE_DOUBLE:
+                cj.lit_d = float(lit)
+            else:
+                cj.lit_i = int(lit)
+            continue
         cj.col, cj.op, cj.cmp_type = col, op, cmp_type
         cj.fn = cjt[4] if len(cjt) > 4 else 0
         cj.or_group = cjt[5] if len(cjt) > 5 else 0
@@ -125,6 +150,13 @@ def make_query(conjuncts=(), group=(), aggs=(), col_types=None,
     for i, (at, col) in enumerate(aggs):
         q.aggs[i].agg_type = at
         q.aggs[i].col2 = -1
+        if expr_is_deep(col):
+            ops = []
+            dom = compile_expr(col, col_types, ops)
+            q.aggs[i].prog_begin, q.aggs[i].prog_len = spec_add_prog(q, ops)
+            q.aggs[i].col = 0
+            q.agg_in_types[i] = dom
+            continue
         if isinstance(col, tuple):
             # expression input: (arith_code, a, b); DOUBLE domain iff either
             # operand is DOUBLE (mirrors plan.py to_spec)

@@ -125,6 +125,76 @@ static inline double cell_f64_cast(const OrcCol* c, int64_t r) {
     return (double)cell_i64(c, r);
 }
 
+/* postfix expression programs (bk_common.h BkExprOp): restates the same
+ * flattening of ScalarFnCall::get_value trees the engine evaluates
+ * (scalar_fn_call.cpp:194-225; arg-cast rule 219-225). */
+typedef struct { int valid; int64_t i; double d; } OrcPVal;
+static OrcPVal orc_eval_prog(const OrcCol* cols, const BkQuerySpec* q,
+                             int32_t begin, int32_t len, int64_t r) {
+    int64_t si[BK_MAX_PROG_DEPTH];
+    double sd[BK_MAX_PROG_DEPTH];
+    int sv[BK_MAX_PROG_DEPTH];
+    int sp = 0;
+    for (int32_t k = 0; k < len; k++) {
+        const BkExprOp* e = &q->prog[begin + k];
+        switch ((BkProgOp)e->op) {
+            case BK_PROG_COL: {
+                const OrcCol* c = &cols[e->arg];
+                sv[sp] = cell_is_valid(c, r);
+                if (c->type == BK_DOUBLE) {
+                    sd[sp] = ((double*)c->data)[r];
+                    si[sp] = (int64_t)sd[sp];
+                } else {
+                    si[sp] = cell_i64(c, r);
+                    sd[sp] = (double)si[sp];
+                }
+                sp++;
+            } break;
+            case BK_PROG_LIT_I:
+                si[sp] = e->lit_i;
+                sd[sp] = (double)e->lit_i;
+                sv[sp] = 1;
+                sp++;
+                break;
+            case BK_PROG_LIT_D:
+                sd[sp] = e->lit_d;
+                si[sp] = (int64_t)e->lit_d;
+                sv[sp] = 1;
+                sp++;
+                break;
+            case BK_PROG_ARITH: {
+                sp--;
+                int v = sv[sp - 1] && sv[sp];
+                if (e->domain == BK_DOUBLE) {
+                    double a = sd[sp - 1], b = sd[sp];
+                    double o = e->arg == BK_ARITH_ADD   ? a + b
+                               : e->arg == BK_ARITH_SUB ? a - b
+                                                        : a * b;
+                    sd[sp - 1] = o;
+                    si[sp - 1] = (int64_t)o;
+                } else {
+                    uint64_t a = (uint64_t)si[sp - 1], b = (uint64_t)si[sp];
+                    int64_t o = (int64_t)(e->arg == BK_ARITH_ADD   ? a + b
+                                          : e->arg == BK_ARITH_SUB ? a - b
+                                                                   : a * b);
+                    si[sp - 1] = o;
+                    sd[sp - 1] = (double)o;
+                }
+                sv[sp - 1] = v;
+            } break;
+            default:                           /* BK_PROG_FN */
+                si[sp - 1] = bk_scalar_fn(e->arg, si[sp - 1]);
+                sd[sp - 1] = (double)si[sp - 1];
+                break;
+        }
+    }
+    OrcPVal out;
+    out.valid = sv[0];
+    out.i = si[0];
+    out.d = sd[0];
+    return out;
+}
+
 /* returns 1 iff row passes every clause: standalone conjuncts AND
  * together; or_group members OR within the clause (CNF; see BkConjunct).
  * NULL or false rejects a standalone term; a NULL OR-member is just not
@@ -133,6 +203,31 @@ static int orc_row_passes(const OrcCol* cols, const BkQuerySpec* q, int64_t r) {
     uint32_t or_seen = 0, or_sat = 0;
     for (int32_t j = 0; j < q->n_conjuncts; j++) {
         const BkConjunct* cj = &q->conjuncts[j];
+        if (cj->prog_len > 0) {
+            OrcPVal p = orc_eval_prog(cols, q, cj->prog_begin, cj->prog_len,
+                                      r);
+            int pass = 0;
+            if (p.valid) {
+                int cmp = cj->cmp_type == BK_DOUBLE
+                              ? ((p.d > cj->lit_d) - (p.d < cj->lit_d))
+                              : ((p.i > cj->lit_i) - (p.i < cj->lit_i));
+                switch ((BkCmpOp)cj->op) {
+                    case BK_OP_EQ: pass = (cmp == 0); break;
+                    case BK_OP_NE: pass = (cmp != 0); break;
+                    case BK_OP_GT: pass = (cmp > 0);  break;
+                    case BK_OP_GE: pass = (cmp >= 0); break;
+                    case BK_OP_LT: pass = (cmp < 0);  break;
+                    default:       pass = (cmp <= 0); break;
+                }
+            }
+            if (cj->or_group == 0) {
+                if (!pass) return 0;
+            } else {
+                or_seen |= 1u << (cj->or_group & 31);
+                if (pass) or_sat |= 1u << (cj->or_group & 31);
+            }
+            continue;
+        }
         const OrcCol* c = &cols[cj->col];
         if (!cell_is_valid(c, r)) {
             if (cj->or_group == 0) return 0;  /* NULL operand => reject */
@@ -429,7 +524,14 @@ static void* orc_agg_worker(void* arg) {
             int vtype = q->agg_in_types[a];
             int in_valid = 1;
             int64_t vi = 0; double vd = 0.0;
-            if (as->col >= 0) {
+            if (as->prog_len > 0) {
+                /* postfix expression input */
+                OrcPVal p = orc_eval_prog(cols, q, as->prog_begin,
+                                          as->prog_len, r);
+                in_valid = p.valid;
+                vi = p.i;
+                vd = p.d;
+            } else if (as->col >= 0) {
                 const OrcCol* c = &cols[as->col];
                 in_valid = cell_is_valid(c, r);
                 if (as->arith) {

@@ -69,11 +69,22 @@ def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
            "in": 6, "not_in": 7}
     aggmap = {"count_star": 0, "count": 1, "sum": 2, "avg": 3, "min": 4, "max": 5}
     oconj = []
-    from baikaldb_amd.plan import _FNS, _ARITH
+    from baikaldb_amd.plan import _FNS, _ARITH, expr_is_deep, compile_expr
     for cjt in conjuncts:
         col, op, lit = cjt[0], cjt[1], cjt[2]
         og = cjt[3] if len(cjt) > 3 else 0
         fn, col2, arith = 0, -1, 0
+        if expr_is_deep(col):
+            # deep expression LHS: both sides compile it with the same
+            # planner (make_query handles the tuple); cmp domain = root
+            # domain or a float literal
+            dom = compile_expr(col, col_types, [])
+            ct = TYPE_DOUBLE if (dom == TYPE_DOUBLE or
+                                 isinstance(lit, float)) else TYPE_INT64
+            if ct == TYPE_DOUBLE:
+                lit = float(lit)
+            oconj.append((col, ops[op], ct, lit, 0, og, -1, 0))
+            continue
         if isinstance(col, tuple) and col[0] in _ARITH:
             arith, col2, col = _ARITH[col[0]], col[2], col[1]
             if col_types[col] == TYPE_DOUBLE or col_types[col2] == TYPE_DOUBLE:
@@ -89,9 +100,9 @@ def run_both(eng, orc, spec_rows, n, conjuncts, group, aggs, nthreads=4,
     ogroup = [(_FNS[g[0]], g[1]) if isinstance(g, tuple) else g for g in group]
     oaggs = []
     for a, c in aggs:
-        if isinstance(c, tuple):
-            c = (_ARITH[c[0]], c[1], c[2])
-        oaggs.append((aggmap[a], c))
+        if isinstance(c, tuple) and not expr_is_deep(c):
+            c = (_ARITH[c[0]], c[1], c[2])   # legacy one-arith shape
+        oaggs.append((aggmap[a], c))         # deep exprs pass through
     q = make_query(oconj, ogroup, oaggs, col_types,
                    group_bits=group_bits, group_base=group_base)
     exp = orc.filter_agg(cols, valids, col_types, q, nthreads=nthreads,
@@ -107,9 +118,18 @@ def assert_parity(got, exp, aggs, col_types):
     assert np.array_equal(got["agg_has"], exp["agg_has"])
     for a, (name, col) in enumerate(aggs):
         if isinstance(col, tuple):   # expression input: the compute DOMAIN
-            col = (col[1] if col_types[col[1]] == TYPE_DOUBLE else
-                   col[2] if col_types[col[2]] == TYPE_DOUBLE else col[1])
-        is_double = col >= 0 and col_types[col] == TYPE_DOUBLE
+            from baikaldb_amd.plan import expr_is_deep, compile_expr
+            if expr_is_deep(col):
+                dom = compile_expr(col, col_types, [])
+                is_double = dom == TYPE_DOUBLE
+                col = 0
+            else:
+                col = (col[1] if col_types[col[1]] == TYPE_DOUBLE else
+                       col[2] if col_types[col[2]] == TYPE_DOUBLE else
+                       col[1])
+                is_double = col_types[col] == TYPE_DOUBLE
+        else:
+            is_double = col >= 0 and col_types[col] == TYPE_DOUBLE
         if name in ("count_star", "count") or not is_double:
             assert np.array_equal(got["agg_i"][a], exp["agg_i"][a]), f"agg {a} {name}"
         elif name in ("min", "max"):
@@ -863,3 +883,47 @@ def test_partitioned_exchange_equals_whole(eng, orc):
     denom = np.abs(expect["agg_d"][2]) + np.maximum(expect["agg_i"][0], 1)
     assert np.all(np.abs(agg_d[2][order] - expect["agg_d"][2])
                   <= DTOL_REL * denom)
+
+
+def test_expr_programs_deep_trees(eng, orc):
+    """Postfix expression programs (BkExprOp, the planner-flattened form of
+    ScalarFnCall::get_value's arbitrary trees, scalar_fn_call.cpp:194-225):
+    3-deep mixed int/double trees as conjunct LHS and aggregate inputs,
+    parity vs the oracle evaluating the identically-compiled programs."""
+    aggs = [("count_star", -1),
+            ("sum", ("mul", ("add", 0, 2), 2)),              # int64 3-deep
+            ("sum", ("add", ("mul", 3, 3), ("mul", 2, 2))),  # mixed double
+            ("avg", ("sub", 3, ("mul", 2, 0.5))),            # double w/ lit
+            ("max", ("add", ("liti", 10), ("mul", 2, 2)))]
+    conj = [(("add", ("mul", 0, ("liti", 3)), 2), "<", (3 << 31)),
+            (("sub", 3, ("mul", 3, 0.25)), ">", -1.0)]
+    got, exp = run_both(eng, orc, BASE5, 300_000, conj, [1], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in BASE5])
+
+
+def test_expr_programs_nulls_propagate(eng, orc):
+    """Any NULL operand makes the whole expression NULL (the reference's
+    arg-cast + null propagation through the tree)."""
+    specs = [(TYPE_INT64, D_UNI, 0, 1 << 31, 200_000),
+             (TYPE_INT64, D_SKEW, 500, 0, 0),
+             (TYPE_INT64, D_UNI, 0, 100, 300_000),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 150_000)]
+    aggs = [("count_star", -1),
+            ("sum", ("mul", ("add", 0, 2), 2)),
+            ("count", ("add", 3, ("mul", 0, 2)))]
+    conj = [(("add", ("mul", 2, ("liti", 2)), 0), "<", (1 << 32))]
+    got, exp = run_both(eng, orc, specs, 250_000, conj, [1], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
+
+
+def test_expr_programs_with_datetime_fn(eng, orc):
+    """Scalar fns inside programs (GROUP-BY-side fns already existed; here
+    year(c)*100+month(c)-style predicate trees)."""
+    specs = [(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+             (14, 5, 0, 0, 0),                  # DATETIME col
+             (TYPE_INT64, D_UNI, 0, 100, 0)]
+    aggs = [("count_star", -1), ("sum", 2)]
+    conj = [(("add", ("mul", ("year", 1), ("liti", 100)), ("month", 1)),
+             ">=", 2021 * 100 + 3)]
+    got, exp = run_both(eng, orc, specs, 200_000, conj, [2], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
