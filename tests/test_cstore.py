@@ -266,4 +266,7 @@ def test_cstore_to_hbm_group_by(lib):
     assert g1["ngroups"] == g2["ngroups"]
     assert np.array_equal(g1["enc"], g2["enc"])
     assert np.array_equal(g1["agg_i"], g2["agg_i"])
-    np.testing.assert_allclose(g1["agg_d"], g2["agg_d"], rtol=0, atol=0)
+    # decode is bit-exact; the AVG(double) residual is only the f64 atomic
+    # reduction ORDER differing between the two GPU runs
+    np.testing.assert_allclose(g1["agg_d"], g2["agg_d"], rtol=1e-10,
+                               atol=1e-12)
