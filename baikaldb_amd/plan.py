@@ -48,7 +48,7 @@ class BkExprOp(C.Structure):
                 ("lit_i", C.c_int64), ("lit_d", C.c_double)]
 
 
-BK_MAX_PROG_POOL = 24
+BK_MAX_PROG_POOL = 32
 PROG_COL, PROG_LIT_I, PROG_LIT_D, PROG_ARITH, PROG_FN = 0, 1, 2, 3, 4
 
 

@@ -226,7 +226,7 @@ typedef struct BkExprOp {
     double  lit_d;
 } BkExprOp;
 
-#define BK_MAX_PROG_POOL 24   /* ops shared by all programs of one query */
+#define BK_MAX_PROG_POOL 32   /* ops shared by all programs of one query */
 #define BK_MAX_PROG_DEPTH 6   /* max operand-stack depth */
 
 /* ---- window functions (reference src/expr/window_fn_call.cpp:20-38 name

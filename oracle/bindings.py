@@ -46,7 +46,7 @@ class BkExprOp(C.Structure):
                 ("lit_i", C.c_int64), ("lit_d", C.c_double)]
 
 
-BK_MAX_PROG_POOL = 24
+BK_MAX_PROG_POOL = 32
 
 
 class BkWindowFn(C.Structure):
