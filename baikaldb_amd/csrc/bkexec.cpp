@@ -406,8 +406,15 @@ public:
             outs.push_back(out);
             for (int a = 0; a < q.n_aggs; a++) src[a] = {0, a};
         } else {
-            if (q.n_group > 1) {
-                state->error_msg = "DISTINCT aggs support <= 1 group key";
+            if (q.n_group > 2) {
+                state->error_msg = "DISTINCT aggs support <= 2 group keys";
+                return -1;
+            }
+            if (q.n_group == 2 && (!q.group_bits[0] || !q.group_bits[1])) {
+                /* 2 user keys + d = 3 level-1 keys: they must pack into
+                 * the two 64-bit key words via declared widths */
+                state->error_msg = "DISTINCT with 2 group keys needs "
+                                   "group_bits declared for both";
                 return -1;
             }
             /* pass 0: the plain aggs (always run — it also anchors the

@@ -1807,6 +1807,19 @@ __global__ void k_merge_blob(BkQuerySpec q, const uint32_t* flags, const uint64_
 struct SrcIdx { int32_t v[BK_MAX_AGGS]; };  /* level-1 agg index per level-2
                                                agg; -1 = synthesize from d */
 
+/* level-1 -> level-2 key plumbing: unpack the level-1 packed keys
+ * (pack_group_keys layout: sequential shift packing with word overflow),
+ * then REPACK the user keys per the level-2 spec's own declared widths —
+ * the word splits can differ between the (keys + d) and keys-only
+ * packings. eb = host-encoded bases. */
+struct RollK {
+    int32_t l1n;            /* level-1 key count (user keys + d), <= 3 */
+    int32_t l1bits[3];
+    uint64_t l1eb[3];
+    int32_t q2bits[2];
+    uint64_t q2eb[2];
+};
+
 /* Level-1 key layout for the rollup: with no declared group_bits the user
  * key sits raw in k0 and d raw in k1 (l1_bits* == 0). With declared bits
  * (the sort-dedup path REQUIRES this; the hash path then packs the same
@@ -1815,38 +1828,52 @@ struct SrcIdx { int32_t v[BK_MAX_AGGS]; };  /* level-1 agg index per level-2
 __global__ void k_rollup(BkQuerySpec q2, int in_naggs, const uint64_t* in_table,
                          uint64_t in_nslots, uint64_t* out, uint64_t omask,
                          uint64_t fill_cap, uint64_t* fill, uint32_t* err,
-                         SrcIdx src, int l1_bits0, int l1_bits1,
-                         uint64_t l1_eb0, uint64_t l1_eb1) {
+                         SrcIdx src, RollK rk) {
     const int in_stride = SLOT_HDR + 2 * in_naggs;
     const int stride = SLOT_HDR + 2 * q2.n_aggs;
+    const int n_user = rk.l1n - 1;
     uint64_t gs = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t sl = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
          sl < in_nslots; sl += gs) {
         const uint64_t* s = in_table + sl * in_stride;
         if (((const uint32_t*)s)[0] != 2u) continue;
         uint32_t f = ((const uint32_t*)s)[1];
-        uint64_t e_d;
-        int d_null;
+        /* unpack ALL level-1 keys (raw encodings; null -> 0) */
+        uint64_t e[3] = {0, 0, 0};
+        {
+            int shift = 0, word = 0;
+            for (int k = 0; k < rk.l1n; k++) {
+                int bits = rk.l1bits[k] ? rk.l1bits[k] : 64;
+                if (shift + bits > 64) { word++; shift = 0; }
+                uint64_t raw = (word == 0 ? s[1] : s[2]) >> shift;
+                if (bits < 64) raw &= (1ull << bits) - 1ull;
+                int isnull = (f >> (7 - k)) & 1;
+                e[k] = isnull ? 0 : (rk.l1bits[k] ? raw + rk.l1eb[k] : raw);
+                shift += bits;
+            }
+        }
+        int d_null = (f >> (7 - n_user)) & 1;
+        uint64_t e_d = e[n_user];
         uint64_t* g;
         if (q2.n_group == 0) {
-            /* no user group keys: level-1 key IS d; out slot 0 pre-claimed */
-            e_d = s[1]; d_null = (f & 0x80u) != 0;
-            if (l1_bits0 && !d_null)
-                e_d = (e_d & ((1ull << l1_bits0) - 1ull)) + l1_eb0;
-            g = out;
+            g = out;        /* out slot 0 pre-claimed */
         } else {
-            uint64_t e0;
-            d_null = (f & 0x40u) != 0;
-            if (l1_bits0 && l1_bits1 && l1_bits0 + l1_bits1 <= 64) {
-                uint64_t k = s[1];
-                e0 = (f & 0x80u) ? 0
-                     : (k & ((1ull << l1_bits0) - 1ull)) + l1_eb0;
-                e_d = d_null ? 0
-                      : ((k >> l1_bits0) & ((1ull << l1_bits1) - 1ull)) + l1_eb1;
-            } else {
-                e0 = s[1]; e_d = s[2];
+            /* repack the user keys per the LEVEL-2 spec's widths (the
+             * layout a direct GROUP BY with q2 would produce) */
+            uint64_t ok0 = 0, ok1 = 0;
+            int shift = 0, word = 0;
+            for (int k = 0; k < n_user; k++) {
+                int bits = rk.q2bits[k] ? rk.q2bits[k] : 64;
+                if (shift + bits > 64) { word++; shift = 0; }
+                int isnull = (f >> (7 - k)) & 1;
+                uint64_t enc = isnull ? 0 : e[k];
+                if (!isnull && bits < 64)
+                    enc = (enc - rk.q2eb[k]) & ((1ull << bits) - 1ull);
+                if (word == 0) ok0 |= enc << shift; else ok1 |= enc << shift;
+                shift += bits;
             }
-            g = gtable_claim(out, omask, stride, f & 0x80u, e0, 0,
+            uint32_t uflag = f & (uint32_t)((0xFFu << (8 - n_user)) & 0xFFu);
+            g = gtable_claim(out, omask, stride, uflag, ok0, ok1,
                              fill, fill_cap, err);
             if (!g) return;
         }
@@ -3363,9 +3390,9 @@ extern "C" BkgAggOut* bkgpu_agg_rollup(const BkgAggOut* in, const BkQuerySpec* q
     if (in->dense_mode &&
         dense_to_hash(const_cast<BkgAggOut*>(in)) != 0)
         return nullptr;
-    if (in->q.n_group != q2->n_group + 1 || q2->n_group > 1) {
+    if (in->q.n_group != q2->n_group + 1 || q2->n_group > 2) {
         set_err("rollup: level-1 must group by (user keys + distinct col), "
-                "user keys <= 1");
+                "user keys <= 2");
         return nullptr;
     }
     if (q2->n_aggs < 1 || q2->n_aggs > BK_MAX_AGGS) {
@@ -3394,28 +3421,23 @@ extern "C" BkgAggOut* bkgpu_agg_rollup(const BkgAggOut* in, const BkQuerySpec* q
         HIP_CHECK_NULL(hipMemcpy(ctr_host, in->ctrs, 24, hipMemcpyDeviceToHost));
         const_cast<BkgAggOut*>(in)->rows_passed = (int64_t)ctr_host[1];
     }
-    /* packed level-1 key layout (declared group_bits): precompute the
-     * per-key widths and ENCODED bases for the kernel's unpack */
+    /* packed key layouts: per-key widths and ENCODED bases for the
+     * kernel's unpack (level 1: user keys + d) and repack (level 2) */
     const BkQuerySpec& l1q = in->q;
-    int l1_bits0 = 0, l1_bits1 = 0;
-    uint64_t l1_eb0 = 0, l1_eb1 = 0;
-    {
-        int k0i = 0, k1i = l1q.n_group - 1;  /* d is the LAST l1 key */
-        if (l1q.n_group == 1) {
-            l1_bits0 = l1q.group_bits[0];
-            l1_eb0 = l1q.group_types[0] == BK_STRING
-                         ? (uint64_t)l1q.group_base[0]
-                         : bk_enc_i64(l1q.group_base[0]);
-        } else if (l1q.n_group == 2) {
-            l1_bits0 = l1q.group_bits[k0i];
-            l1_bits1 = l1q.group_bits[k1i];
-            l1_eb0 = l1q.group_types[k0i] == BK_STRING
-                         ? (uint64_t)l1q.group_base[k0i]
-                         : bk_enc_i64(l1q.group_base[k0i]);
-            l1_eb1 = l1q.group_types[k1i] == BK_STRING
-                         ? (uint64_t)l1q.group_base[k1i]
-                         : bk_enc_i64(l1q.group_base[k1i]);
-        }
+    RollK rk;
+    memset(&rk, 0, sizeof rk);
+    rk.l1n = l1q.n_group;
+    for (int k = 0; k < l1q.n_group && k < 3; k++) {
+        rk.l1bits[k] = l1q.group_bits[k];
+        rk.l1eb[k] = l1q.group_types[k] == BK_STRING
+                         ? (uint64_t)l1q.group_base[k]
+                         : bk_enc_i64(l1q.group_base[k]);
+    }
+    for (int k = 0; k < q2->n_group && k < 2; k++) {
+        rk.q2bits[k] = q2->group_bits[k];
+        rk.q2eb[k] = q2->group_types[k] == BK_STRING
+                         ? (uint64_t)q2->group_base[k]
+                         : bk_enc_i64(q2->group_base[k]);
     }
     BkgAggOut* o = new BkgAggOut();
     o->q = *q2;
@@ -3434,8 +3456,7 @@ extern "C" BkgAggOut* bkgpu_agg_rollup(const BkgAggOut* in, const BkQuerySpec* q
         hipLaunchKernelGGL(k_rollup, dim3(1024), dim3(256), 0, 0,
                            *q2, in->q.n_aggs, in->table, in->nslots,
                            o->table, o->nslots - 1, (o->nslots * 7) / 8,
-                           o->ctrs, o->err, src, l1_bits0, l1_bits1,
-                           l1_eb0, l1_eb1);
+                           o->ctrs, o->err, src, rk);
         tm.record();
         hipError_t lerr = hipGetLastError();
         if (lerr != hipSuccess) {

@@ -75,6 +75,15 @@ def _load():
                                  C.c_int64, C.c_int64,
                                  C.POINTER(C.c_int64), C.POINTER(C.c_int64),
                                  C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
+    lib.bkgpu_window_multi.restype = C.c_int64
+    lib.bkgpu_window_multi.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
+                                 C.POINTER(C.c_int32), C.c_int32,
+                                 C.POINTER(BkOrderSpec), C.c_int,
+                                 C.POINTER(BkWindowFn), C.c_int,
+                                 C.c_int32, C.c_int64, C.c_int64,
+                                 C.c_int64, C.c_int64,
+                                 C.POINTER(C.c_int64), C.POINTER(C.c_int64),
+                                 C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
     lib.bkgpu_agg_rollup.restype = C.c_void_p
     lib.bkgpu_agg_rollup.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
                                      C.POINTER(C.c_int32), C.c_int64]
@@ -436,9 +445,20 @@ class GpuEngine:
             fr, fpre, ffol = 4, frame[1], frame[2]
         else:
             fr, fpre, ffol = 1, frame[0], frame[1]
-        n = self.lib.bkgpu_window(
-            table.handle, C.byref(q), part_col, oarr, len(order),
-            farr, len(fns), fr, fpre, ffol, row_begin, row_end,
+        if isinstance(part_col, (list, tuple)):
+            parr = (C.c_int32 * max(len(part_col), 1))(*part_col)
+            n = self.lib.bkgpu_window_multi(
+                table.handle, C.byref(q), parr, len(part_col), oarr,
+                len(order), farr, len(fns), fr, fpre, ffol, row_begin,
+                row_end,
+                rowids.ctypes.data_as(C.POINTER(C.c_int64)),
+                out_i.ctypes.data_as(C.POINTER(C.c_int64)),
+                out_d.ctypes.data_as(C.POINTER(C.c_double)),
+                out_null.ctypes.data_as(C.POINTER(C.c_uint8)))
+        else:
+            n = self.lib.bkgpu_window(
+                table.handle, C.byref(q), part_col, oarr, len(order),
+                farr, len(fns), fr, fpre, ffol, row_begin, row_end,
             rowids.ctypes.data_as(C.POINTER(C.c_int64)),
             out_i.ctypes.data_as(C.POINTER(C.c_int64)),
             out_d.ctypes.data_as(C.POINTER(C.c_double)),

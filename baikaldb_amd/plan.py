@@ -290,15 +290,21 @@ class QueryPlan:
         247-258): return (level1_plan, level2_spec, src_idx) where level 1
         groups by (user keys + distinct col) and carries the plain aggs, and
         level 2 (bkgpu_agg_rollup / orc_filter_agg_distinct) folds the dedup
-        key back out. v1 envelope: one distinct column, <= 1 user group key.
+        key back out. Envelope: one distinct column, <= 2 user group keys
+        (2 keys need group_bits declared so (keys + d) packs into the two
+        64-bit key words).
         """
         dist = [(i, n, c) for i, (n, c) in enumerate(self.aggs)
                 if (_AGGS[n] if isinstance(n, str) else n) >= AGG_COUNT_DISTINCT]
         dcols = {c for _, _, c in dist}
         if len(dcols) != 1:
             raise ValueError("exactly one DISTINCT column supported")
-        if len(self.group) > 1:
-            raise ValueError("DISTINCT aggs support <= 1 group key")
+        if len(self.group) > 2:
+            raise ValueError("DISTINCT aggs support <= 2 group keys")
+        if len(self.group) == 2 and (len(self.group_bits) < 2 or
+                                     not all(self.group_bits[:2])):
+            raise ValueError("DISTINCT with 2 group keys needs group_bits "
+                             "declared for both")
         dcol = dcols.pop()
         plain = [(n, c) for n, c in self.aggs
                  if (_AGGS[n] if isinstance(n, str) else n) < AGG_COUNT_DISTINCT]
@@ -318,6 +324,10 @@ class QueryPlan:
         for i, col in enumerate(self.group):
             q2.group_cols[i] = col
             q2.group_types[i] = self.col_types[col]
+            if i < len(self.group_bits):
+                q2.group_bits[i] = self.group_bits[i]
+            if i < len(self.group_base):
+                q2.group_base[i] = self.group_base[i]
         q2.n_aggs = len(self.aggs)
         src_idx = (C.c_int32 * len(self.aggs))()
         next_plain = 0

@@ -190,6 +190,17 @@ int64_t bkgpu_window(BkgTable* t, const BkQuerySpec* q, int32_t part_col,
                      int64_t row_begin, int64_t row_end,
                      int64_t* out_rowids, int64_t* out_i, double* out_d,
                      uint8_t* out_null);
+/* PARTITION BY multiple columns (window_node.cpp evaluates every
+ * partition expr; a new partition starts when any changes). n_part <= 3,
+ * n_part + norder <= 4 sort keys. bkgpu_window == n_part <= 1 case. */
+int64_t bkgpu_window_multi(BkgTable* t, const BkQuerySpec* q,
+                     const int32_t* part_cols, int32_t n_part,
+                     const BkOrderSpec* order, int norder,
+                     const BkWindowFn* fns, int nfns,
+                     int32_t frame_rows, int64_t frame_pre, int64_t frame_fol,
+                     int64_t row_begin, int64_t row_end,
+                     int64_t* out_rowids, int64_t* out_i, double* out_d,
+                     uint8_t* out_null);
 
 /* ---- SELECT without GROUP BY (FilterNode row emission, filter_node.cpp:
  * 736-795): collect up to limit passing global row ids (order unspecified);

@@ -203,6 +203,14 @@ class Oracle:
             C.c_int32, C.c_int64, C.c_int64,
             C.c_int64, C.c_int64, C.POINTER(C.c_int64), C.POINTER(C.c_int64),
             C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
+        lib.orc_window_multi.restype = C.c_int64
+        lib.orc_window_multi.argtypes = [
+            C.POINTER(_OrcCol), C.c_int, C.POINTER(BkQuerySpec),
+            C.POINTER(C.c_int32), C.c_int32,
+            C.POINTER(BkOrderSpec), C.c_int, C.POINTER(BkWindowFn), C.c_int,
+            C.c_int32, C.c_int64, C.c_int64,
+            C.c_int64, C.c_int64, C.POINTER(C.c_int64), C.POINTER(C.c_int64),
+            C.POINTER(C.c_double), C.POINTER(C.c_uint8)]
         lib.orc_dict_word.restype = C.c_int
         lib.orc_dict_word.argtypes = [C.c_uint64, C.c_int64, C.c_char_p, C.c_int]
         for f in ("orc_encode_i64", "orc_decode_i64"):
@@ -349,9 +357,20 @@ class Oracle:
             fr, fpre, ffol = 4, frame[1], frame[2]
         else:
             fr, fpre, ffol = 1, frame[0], frame[1]
-        n = self.lib.orc_window(
-            carr, len(cols), C.byref(q), part_col, oarr, len(order),
-            farr, len(fns), fr, fpre, ffol, row_begin, row_end,
+        if isinstance(part_col, (list, tuple)):
+            parr = (C.c_int32 * max(len(part_col), 1))(*part_col)
+            n = self.lib.orc_window_multi(
+                carr, len(cols), C.byref(q), parr, len(part_col), oarr,
+                len(order), farr, len(fns), fr, fpre, ffol, row_begin,
+                row_end,
+                rowids.ctypes.data_as(C.POINTER(C.c_int64)),
+                out_i.ctypes.data_as(C.POINTER(C.c_int64)),
+                out_d.ctypes.data_as(C.POINTER(C.c_double)),
+                out_null.ctypes.data_as(C.POINTER(C.c_uint8)))
+        else:
+            n = self.lib.orc_window(
+                carr, len(cols), C.byref(q), part_col, oarr, len(order),
+                farr, len(fns), fr, fpre, ffol, row_begin, row_end,
             rowids.ctypes.data_as(C.POINTER(C.c_int64)),
             out_i.ctypes.data_as(C.POINTER(C.c_int64)),
             out_d.ctypes.data_as(C.POINTER(C.c_double)),

@@ -646,7 +646,7 @@ ORC_EXPORT OrcAggResult* orc_filter_agg_distinct(
         int64_t row_begin, int64_t row_end, int nthreads,
         uint64_t dict_seed, int sort_keys) {
     (void)ncols;
-    if (q1->n_group != q2->n_group + 1 || q2->n_group > 1) return NULL;
+    if (q1->n_group != q2->n_group + 1 || q2->n_group > 2) return NULL;
     int ng2 = q2->n_group;
     if (nthreads < 1) nthreads = 1;
     if (nthreads > 128) nthreads = 128;
@@ -696,7 +696,7 @@ ORC_EXPORT OrcAggResult* orc_filter_agg_distinct(
     for (uint64_t i = 0; i < l1.cap; i++) {
         if (!l1.hashes[i]) continue;
         OrcGroup* gi = &l1.groups[i];
-        uint8_t f2 = ng2 ? (uint8_t)(gi->flag & 0x80u) : 0;
+        uint8_t f2 = (uint8_t)(gi->flag & ((0xFFu << (8 - ng2)) & 0xFFu));
         int d_null = (gi->flag >> (7 - ng2)) & 1;
         uint64_t e_d = gi->e[ng2];
         uint64_t e2[BK_MAX_GROUP] = {0, 0};
@@ -1050,8 +1050,9 @@ static void orc_win_value(const OrcCol* c, int64_t r, int64_t idx,
     else out_i[idx] = cell_i64(c, r);
 }
 
-ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
-                              const BkQuerySpec* q, int32_t part_col,
+ORC_EXPORT int64_t orc_window_multi(const OrcCol* cols, int ncols,
+                              const BkQuerySpec* q,
+                              const int32_t* part_cols, int32_t n_part,
                               const BkOrderSpec* order, int norder,
                               const BkWindowFn* fns, int nfns,
                               int32_t frame_rows, int64_t f_pre, int64_t f_fol,
@@ -1060,8 +1061,8 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
                               double* out_d, uint8_t* out_null) {
     BkOrderSpec full[4];
     int nf = 0;
-    if (part_col >= 0) {
-        full[nf].col = part_col; full[nf].is_asc = 1;
+    for (int k = 0; k < n_part && nf < 4; k++) {
+        full[nf].col = part_cols[k]; full[nf].is_asc = 1;
         full[nf].is_null_first = 1; nf++;
     }
     for (int k = 0; k < norder && nf < 4; k++) full[nf++] = order[k];
@@ -1072,9 +1073,13 @@ ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
     int64_t ps = 0;
     while (ps < n) {
         int64_t pe = ps + 1;
-        while (pe < n && (part_col < 0 ||
-               orc_cell_eq(&cols[part_col], out_rows[pe], out_rows[pe - 1])))
-            pe++;
+        for (; pe < n; pe++) {
+            int eq = 1;
+            for (int k = 0; k < n_part && eq; k++)
+                eq = orc_cell_eq(&cols[part_cols[k]], out_rows[pe],
+                                 out_rows[pe - 1]);
+            if (!eq) break;
+        }
         int64_t pn = pe - ps;
         /* per-partition aggregate states */
         for (int f = 0; f < nfns; f++) {
@@ -1365,4 +1370,20 @@ ORC_EXPORT double   orc_decode_f64(uint64_t u) { return bk_dec_f64(u); }
 ORC_EXPORT uint64_t orc_mix64(uint64_t x)      { return bk_mix64(x); }
 ORC_EXPORT uint64_t orc_cell_bits(uint64_t seed, uint64_t row, uint32_t col) {
     return bk_cell_bits(seed, row, col);
+}
+
+/* single-partition-column compatibility entry */
+ORC_EXPORT int64_t orc_window(const OrcCol* cols, int ncols,
+                              const BkQuerySpec* q, int32_t part_col,
+                              const BkOrderSpec* order, int norder,
+                              const BkWindowFn* fns, int nfns,
+                              int32_t frame_rows, int64_t f_pre, int64_t f_fol,
+                              int64_t row_begin, int64_t row_end,
+                              int64_t* out_rows, int64_t* out_i,
+                              double* out_d, uint8_t* out_null) {
+    int32_t pc[1] = {part_col};
+    return orc_window_multi(cols, ncols, q, pc, part_col >= 0 ? 1 : 0,
+                            order, norder, fns, nfns, frame_rows, f_pre,
+                            f_fol, row_begin, row_end, out_rows, out_i,
+                            out_d, out_null);
 }

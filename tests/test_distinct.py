@@ -477,3 +477,71 @@ def test_gpu_sorted_rec_mode_vs_partitioned(eng):
     for i in (0, 1, 3):
         assert np.array_equal(ga["agg_i"][i], gb["agg_i"][i])
     assert np.allclose(ga["agg_d"][2], gb["agg_d"][2], rtol=0, atol=1e-9)
+
+
+@pytest.mark.gpu
+def test_gpu_distinct_two_group_keys(eng, orc):
+    """DISTINCT aggregates under TWO user group keys (the round-1 cap was
+    one): level 1 groups by (k0, k1, d) packed via declared group_bits into
+    the two 64-bit key words; k_rollup unpacks all three and repacks the
+    user keys per the level-2 spec."""
+    from baikaldb_amd import QueryPlan
+    specs = [(TYPE_INT64, D_UNI, 0, 40, 0),       # k0
+             (TYPE_INT64, D_UNI, 0, 25, 0),       # k1
+             (TYPE_INT64, D_UNI, 0, 500, 100_000),  # d (nullable)
+             (TYPE_INT64, D_UNI, 0, 1000, 0)]     # plain agg input
+    aggs = [("count_star", -1), ("count_distinct", 2), ("sum", 3),
+            ("sum_distinct", 2)]
+    conj = [(3, "<", 900)]
+    group = [0, 1]
+    n = 150_000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED)
+        plan = QueryPlan(t.col_types, conjuncts=conj, group=group,
+                         aggs=aggs, group_bits=[8, 8], group_base=[0, 0],
+                         distinct_bits=10, distinct_base=0)
+        res = eng.filter_agg_distinct(t, plan, expected_l1_groups=1 << 16,
+                                      expected_groups=1 << 12)
+        try:
+            got = res.fetch(sorted=True)
+        finally:
+            res.free()
+    finally:
+        t.free()
+    exp, _ = oracle_distinct(orc, specs, n, conj, group, aggs, seed=SEED)
+    assert got["ngroups"] == exp["ngroups"]
+    assert np.array_equal(got["flags"], exp["flags"])
+    assert np.array_equal(got["enc"], exp["enc"])
+    assert np.array_equal(got["agg_i"], exp["agg_i"])
+    assert np.array_equal(got["agg_has"], exp["agg_has"])
+
+
+@pytest.mark.gpu
+def test_gpu_distinct_two_keys_with_nullable_key(eng, orc):
+    from baikaldb_amd import QueryPlan
+    specs = [(TYPE_INT64, D_UNI, 0, 30, 200_000),   # k0 nullable
+             (TYPE_STRING, D_DICT, 16, 0, 0),       # k1 dict
+             (TYPE_INT64, D_UNI, 0, 200, 0)]        # d
+    aggs = [("count_star", -1), ("count_distinct", 2)]
+    group = [0, 1]
+    n = 100_000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED + 1)
+        plan = QueryPlan(t.col_types, group=group, aggs=aggs,
+                         group_bits=[8, 6], group_base=[0, 0],
+                         distinct_bits=9, distinct_base=0)
+        res = eng.filter_agg_distinct(t, plan, expected_l1_groups=1 << 14,
+                                      expected_groups=1 << 10)
+        try:
+            got = res.fetch(sorted=True)
+        finally:
+            res.free()
+    finally:
+        t.free()
+    exp, _ = oracle_distinct(orc, specs, n, [], group, aggs, seed=SEED + 1)
+    assert got["ngroups"] == exp["ngroups"]
+    assert np.array_equal(got["flags"], exp["flags"])
+    assert np.array_equal(got["enc"], exp["enc"])
+    assert np.array_equal(got["agg_i"], exp["agg_i"])

@@ -729,3 +729,64 @@ def test_gpu_window_single_row_partitions(eng, orc):
     assert np.all(got["out_i"][1] == 1)
     assert np.all(got["out_d"][2] == 0.0)
     assert np.all(got["out_null"][4] == 1)   # lag leaves every 1-row partition
+
+
+@pytest.mark.gpu
+def test_gpu_window_two_partition_cols(eng, orc):
+    """PARTITION BY two columns (round-1 cap was one): a new partition
+    starts when ANY partition expr changes (window_node.cpp evaluates all
+    of them); engine vs oracle on identical inputs."""
+    from baikaldb_amd import QueryPlan
+    specs = [(TYPE_INT64, D_UNI, 0, 12, 30_000),    # p0 (nullable)
+             (TYPE_STRING, D_DICT, 8, 0, 0),        # p1 dict
+             (TYPE_INT64, D_UNI, 0, 40, 0),         # order
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 100_000)]  # value
+    fns = [("row_number", -1), ("rank", -1), ("count_star", -1),
+           ("sum", 3), ("max", 2), ("dense_rank", -1)]
+    n = 120_000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED)
+        plan = QueryPlan(t.col_types, conjuncts=[(2, "<", 35)])
+        got = eng.window(t, fns, part_col=[0, 1], order=[(2, 1, 1)],
+                         plan=plan)
+    finally:
+        t.free()
+    cols, valids, types = gen(orc, specs, n)
+    from oracle.bindings import make_query
+    q = make_query([(2, 4, TYPE_INT64, 35)], (), ((0, -1),), types)
+    q.n_aggs = 0
+    exp = orc.window(cols, valids, types,
+                     [(W[f[0]], f[1]) for f in fns],
+                     part_col=[0, 1], order=[(2, 1, 1)], q=q)
+    assert got["n"] == exp["n"]
+    assert np.array_equal(got["rowids"], exp["rowids"])
+    assert np.array_equal(got["out_null"], exp["out_null"])
+    assert np.array_equal(got["out_i"], exp["out_i"])
+    mask = exp["out_null"] == 0
+    d = np.abs(got["out_d"] - exp["out_d"])
+    tol = 1e-10 * (np.abs(exp["out_d"]) + 100)
+    assert np.all(d[mask] <= tol[mask])
+
+
+@pytest.mark.gpu
+def test_gpu_window_three_partition_cols(eng, orc):
+    from baikaldb_amd import QueryPlan
+    specs = [(TYPE_INT64, D_UNI, 0, 5, 0),
+             (TYPE_INT64, D_UNI, 0, 7, 0),
+             (TYPE_INT64, D_UNI, 0, 3, 0),
+             (TYPE_INT64, D_UNI, 0, 1000, 0)]
+    fns = [("row_number", -1), ("count_star", -1), ("min", 3)]
+    n = 60_000
+    t = eng.create_table(specs, n)
+    try:
+        eng.generate(t, SEED + 2)
+        got = eng.window(t, fns, part_col=[0, 1, 2], order=[(3, 1, 1)])
+    finally:
+        t.free()
+    cols, valids, types = gen(orc, specs, n)
+    exp = orc.window(cols, valids, types, [(W[f[0]], f[1]) for f in fns],
+                     part_col=[0, 1, 2], order=[(3, 1, 1)])
+    assert got["n"] == exp["n"]
+    assert np.array_equal(got["rowids"], exp["rowids"])
+    assert np.array_equal(got["out_i"], exp["out_i"])
