@@ -19,8 +19,11 @@ OPS = {"=": 0, "!=": 1, ">": 2, ">=": 3, "<": 4, "<=": 5}
 
 
 def oracle_distinct(orc, specs, n, conjuncts, group, aggs, seed=SEED,
-                    nthreads=4):
-    """Run the oracle's two-level distinct path; returns the fetch dict."""
+                    nthreads=4, group_bits=(), group_base=(),
+                    distinct_bits=0, distinct_base=0):
+    """Run the oracle's two-level distinct path; returns the fetch dict.
+    (The oracle's maps hold raw per-key encodings — the packing bits only
+    matter to the ENGINE's key words, so they default off here.)"""
     from oracle.bindings import make_query, BkColSpec
     from baikaldb_amd.plan import QueryPlan
 
@@ -31,7 +34,9 @@ def oracle_distinct(orc, specs, n, conjuncts, group, aggs, seed=SEED,
     cols, valids = orc.generate_table(list(arr), n, seed)
     col_types = [s[0] for s in specs]
 
-    plan = QueryPlan(col_types, conjuncts=conjuncts, group=group, aggs=aggs)
+    plan = QueryPlan(col_types, conjuncts=conjuncts, group=group, aggs=aggs,
+                     group_bits=group_bits, group_base=group_base,
+                     distinct_bits=distinct_bits, distinct_base=distinct_base)
     l1_plan, _, src_idx = plan.split_distinct()
     oconj = []
     for col, op, lit, *og in conjuncts:
@@ -509,7 +514,9 @@ def test_gpu_distinct_two_group_keys(eng, orc):
             res.free()
     finally:
         t.free()
-    exp, _ = oracle_distinct(orc, specs, n, conj, group, aggs, seed=SEED)
+    exp, _ = oracle_distinct(orc, specs, n, conj, group, aggs, seed=SEED,
+                             group_bits=[8, 8], group_base=[0, 0],
+                             distinct_bits=10, distinct_base=0)
     assert got["ngroups"] == exp["ngroups"]
     assert np.array_equal(got["flags"], exp["flags"])
     assert np.array_equal(got["enc"], exp["enc"])
@@ -540,7 +547,9 @@ def test_gpu_distinct_two_keys_with_nullable_key(eng, orc):
             res.free()
     finally:
         t.free()
-    exp, _ = oracle_distinct(orc, specs, n, [], group, aggs, seed=SEED + 1)
+    exp, _ = oracle_distinct(orc, specs, n, [], group, aggs, seed=SEED + 1,
+                             group_bits=[8, 6], group_base=[0, 0],
+                             distinct_bits=9, distinct_base=0)
     assert got["ngroups"] == exp["ngroups"]
     assert np.array_equal(got["flags"], exp["flags"])
     assert np.array_equal(got["enc"], exp["enc"])

@@ -784,7 +784,7 @@ def test_gpu_window_three_partition_cols(eng, orc):
         got = eng.window(t, fns, part_col=[0, 1, 2], order=[(3, 1, 1)])
     finally:
         t.free()
-    cols, valids, types = gen(orc, specs, n)
+    cols, valids, types = gen(orc, specs, n, seed=SEED + 2)
     exp = orc.window(cols, valids, types, [(W[f[0]], f[1]) for f in fns],
                      part_col=[0, 1, 2], order=[(3, 1, 1)])
     assert got["n"] == exp["n"]
