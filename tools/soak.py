@@ -131,7 +131,43 @@ for cs in range(DD_CASES):
     finally:
         t.free()
 print(f"dedup soak: {DD_CASES - dfails}/{DD_CASES} ok", flush=True)
-total_fails = fails + wfails + sfails + dfails
+# ---- round-2 paths: dense-span pipeline + postfix expression programs ----
+DN_CASES = AGG_CASES
+dnfails = 0
+for cs in range(DN_CASES):
+    rng = random.Random(130_000 + SEED_OFF + cs)
+    span1 = rng.choice([50, 3000, 200_000])
+    span2 = rng.choice([4, 100, 4000])
+    specs = [(fz.TYPE_INT64, 0, 0, 1 << 31, 0),
+             (fz.TYPE_INT64, 1, span1, 0, 0),
+             (fz.TYPE_INT64, 0, 0, span2, 0),
+             (fz.TYPE_DOUBLE, 3, 0, 0, 0)]
+    group = [1] if rng.random() < 0.5 else [1, 2]
+    aggs = [("count_star", -1)]
+    for _ in range(rng.randint(1, 4)):
+        aggs.append((rng.choice(["sum", "avg", "min", "max", "count"]),
+                     rng.randint(2, 3)))
+    conj = [(0, "<", int((1 << 31) * rng.random()))]
+    if rng.random() < 0.4:   # deep expression program LHS / agg input
+        conj.append((("add", ("mul", 2, ("liti", rng.randint(1, 9))), 0),
+                     "<", rng.randint(0, 1 << 33)))
+        aggs.append(("sum", ("mul", ("add", 2, 2), 2)))
+    os.environ["BK_DENSE"] = "2"
+    try:
+        got, exp = run_both(eng, orc, specs, rng.choice([5000, 150_000]),
+                            conj, group, aggs,
+                            seed=rng.randrange(1 << 40),
+                            expected_groups=1 << 13)
+        assert_parity(got, exp, aggs, [s[0] for s in specs])
+    except Exception as e:
+        dnfails += 1
+        print(f"DENSE-FAIL {cs}: group={group} aggs={aggs} conj={conj}: {e}",
+              flush=True)
+    finally:
+        os.environ.pop("BK_DENSE", None)
+print(f"dense soak: {DN_CASES - dnfails}/{DN_CASES} ok", flush=True)
+
+total_fails = fails + wfails + sfails + dfails + dnfails
 print(f"SOAK {'PASS' if total_fails == 0 else 'FAIL'} "
       f"({total_fails} failures)", flush=True)
 sys.exit(1 if total_fails else 0)
