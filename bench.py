@@ -3,12 +3,14 @@
 plus achieved HBM GB/s on the 1e9-row filter+GROUP BY workload (config 3 of
 BASELINE.json, the configuration the metric is quoted on; it fits one GPU).
 
-One "step" = one pass of the fused scan+filter+hash-aggregate pipeline over
+One "step" = one pass of the fused scan+filter+aggregate pipeline over
 the full synthetic table resident in HBM (data generated on device before the
 timed region; `data: synthetic`). With N>1 ranks each rank owns its region
 set (weak scaling: per-GPU rows fixed) and the step includes the RCCL
-merge-aggregate of partial group tables (the MERGE_AGG exchange the reference
-does over brpc, exchange_sender_node.h:228-235 — here all-gather over xGMI).
+merge-aggregate: a hash-partitioned all-to-all of part blobs (the
+repartition the reference's ExchangeSenderNode does over brpc,
+exchange_sender_node.h:228-235), every rank merging its key-hash shard in
+parallel.
 
 Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--config NAME]
 The driver launches N>1 via torch.distributed.run; ranks read RANK/WORLD_SIZE.
