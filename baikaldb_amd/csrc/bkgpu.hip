@@ -39,6 +39,7 @@
 #include <string>
 #include <vector>
 #include <algorithm>
+#include <map>
 
 #include "../../include/bk_common.h"
 #include "../../include/bk_datagen.h"
@@ -2397,6 +2398,72 @@ extern "C" int bkgpu_table_upload(BkgTable* t, int col, const void* data,
                             hipMemcpyHostToDevice));
     }
     return 0;
+}
+
+/* Upload ARBITRARY strings into a BK_STRING column: the host builds the
+ * order-preserving dictionary (sorted unique byte strings -> codes, the
+ * same policy as the parquet/cstore ingests) and uploads int32 codes.
+ * This is how a drop-in embedder hands the engine non-dictionary VARCHAR
+ * (the reference's ExprValue STRING comparisons, expr_value.h, and
+ * MutTableKey string keys, mut_table_key.h:196-208): code order == byte
+ * order, so GROUP BY / MIN / MAX / ORDER BY and RANGE predicates on the
+ * codes reproduce string semantics exactly; equality literals map through
+ * bkgpu_table_dict_code. offs[r]..offs[r+1] delimit row r's bytes in
+ * `bytes` (offs has nrows+1 entries); NULL rows (valid[r]==0) contribute
+ * no word. */
+extern "C" int bkgpu_table_upload_strings(BkgTable* t, int col,
+                                          const char* bytes,
+                                          const int64_t* offs,
+                                          const uint8_t* valid) {
+    if (!t || col < 0 || col >= t->ncols ||
+        t->specs[col].col_type != BK_STRING) {
+        set_err("upload_strings: not a BK_STRING column");
+        return -1;
+    }
+    std::map<std::string, int32_t> dict;
+    int64_t n = t->nrows;
+    for (int64_t r = 0; r < n; r++) {
+        if (valid && !valid[r]) continue;
+        dict.emplace(std::string(bytes + offs[r],
+                                 (size_t)(offs[r + 1] - offs[r])), 0);
+    }
+    int32_t next = 0;
+    for (auto& kv : dict) kv.second = next++;
+    std::vector<int32_t> codes((size_t)n, 0);
+    for (int64_t r = 0; r < n; r++) {
+        if (valid && !valid[r]) continue;
+        codes[(size_t)r] = dict[std::string(bytes + offs[r],
+                                            (size_t)(offs[r + 1] - offs[r]))];
+    }
+    if (bkgpu_table_upload(t, col, codes.data(), valid) != 0) return -1;
+    auto* d = new std::vector<std::string>();
+    d->reserve(dict.size());
+    for (auto& kv : dict) d->push_back(kv.first);
+    delete t->dict[col];
+    t->dict[col] = d;
+    return 0;
+}
+
+/* literal -> dict code mapping for predicates on string columns:
+ * mode 0 = exact (code of the word, -1 if absent);
+ * mode 1 = lower_bound (first code whose word >= the literal) — with
+ * order-preserving codes this turns any string RANGE predicate into an
+ * integer compare on codes: `s < L` <=> `code < lower_bound(L)`,
+ * `s >= L` <=> `code >= lower_bound(L)` (operators.cpp string compare
+ * semantics carried through the encoding). */
+extern "C" int64_t bkgpu_table_dict_code(const BkgTable* t, int col,
+                                         const char* word, int64_t wlen,
+                                         int mode) {
+    if (!t || col < 0 || col >= t->ncols || !t->dict[col]) {
+        set_err("dict_code: no dictionary on column");
+        return -2;
+    }
+    const auto& d = *t->dict[col];
+    std::string w(word, (size_t)(wlen < 0 ? strlen(word) : (size_t)wlen));
+    auto it = std::lower_bound(d.begin(), d.end(), w);
+    if (mode == 1) return (int64_t)(it - d.begin());
+    if (it != d.end() && *it == w) return (int64_t)(it - d.begin());
+    return -1;
 }
 
 /* derived remapped STRING column: newcode[r] = remap[oldcode[r]] — the

@@ -38,6 +38,14 @@ def _load():
     lib.bkgpu_table_create.argtypes = [C.c_int, C.POINTER(_BkColSpec), C.c_int64]
     lib.bkgpu_table_generate.argtypes = [C.c_void_p, C.c_uint64, C.c_int64]
     lib.bkgpu_table_upload.argtypes = [C.c_void_p, C.c_int, C.c_void_p, C.c_void_p]
+    lib.bkgpu_table_upload_strings.restype = C.c_int
+    lib.bkgpu_table_upload_strings.argtypes = [C.c_void_p, C.c_int,
+                                               C.c_char_p,
+                                               C.POINTER(C.c_int64),
+                                               C.c_void_p]
+    lib.bkgpu_table_dict_code.restype = C.c_int64
+    lib.bkgpu_table_dict_code.argtypes = [C.c_void_p, C.c_int, C.c_char_p,
+                                          C.c_int64, C.c_int]
     lib.bkgpu_table_nrows.restype = C.c_int64
     lib.bkgpu_table_nrows.argtypes = [C.c_void_p]
     lib.bkgpu_table_free.argtypes = [C.c_void_p]
@@ -278,6 +286,33 @@ class GpuEngine:
         vptr = valid.ctypes.data_as(C.c_void_p) if valid is not None else None
         self._check(self.lib.bkgpu_table_upload(
             table.handle, col, data.ctypes.data_as(C.c_void_p), vptr), "table_upload")
+
+    def upload_strings(self, table, col, strings, valid=None):
+        """Arbitrary (non-dictionary) VARCHAR: the engine builds the
+        order-preserving dictionary and stores int32 codes — string
+        GROUP BY / MIN / MAX / ORDER BY and range predicates then run as
+        integer code ops with identical semantics."""
+        byts = bytearray()
+        offs = np.zeros(len(strings) + 1, dtype=np.int64)
+        for i, w in enumerate(strings):
+            if valid is None or valid[i]:
+                byts += w.encode() if isinstance(w, str) else bytes(w)
+            offs[i + 1] = len(byts)
+        vptr = valid.ctypes.data_as(C.c_void_p) if valid is not None else None
+        self._check(self.lib.bkgpu_table_upload_strings(
+            table.handle, col, bytes(byts),
+            offs.ctypes.data_as(C.POINTER(C.c_int64)), vptr),
+            "upload_strings")
+
+    def dict_code(self, table, col, word, mode=0):
+        """mode 0: exact code (-1 absent); mode 1: lower_bound — turns a
+        string range literal into a code literal."""
+        w = word.encode() if isinstance(word, str) else bytes(word)
+        r = self.lib.bkgpu_table_dict_code(table.handle, col, w, len(w),
+                                           mode)
+        if r == -2:
+            raise RuntimeError(self.lib.bkgpu_last_error().decode())
+        return int(r)
 
     def filter_agg(self, table, plan: QueryPlan, row_begin=0, row_end=None,
                    expected_groups=1 << 16):

@@ -57,6 +57,18 @@ BkgTable* bkgpu_table_create(int ncols, const BkColSpec* specs, int64_t nrows);
 int  bkgpu_table_generate(BkgTable* t, uint64_t seed, int64_t row_begin);
 /* Upload host column data (and optional validity bytes) instead. */
 int  bkgpu_table_upload(BkgTable* t, int col, const void* data, const uint8_t* valid);
+/* Upload ARBITRARY strings into a BK_STRING column: builds the
+ * order-preserving dictionary host-side (code order == byte order, the
+ * parquet/cstore ingest policy) and uploads int32 codes — the drop-in
+ * path for non-dictionary VARCHAR (ExprValue STRING compares /
+ * mut_table_key.h:196-208 string keys map onto integer code compares).
+ * offs[nrows+1] delimits each row's bytes; NULL rows need valid[r]==0. */
+int  bkgpu_table_upload_strings(BkgTable* t, int col, const char* bytes,
+                                const int64_t* offs, const uint8_t* valid);
+/* literal -> code for string predicates: mode 0 exact (-1 absent),
+ * mode 1 lower_bound (s < L <=> code < lower_bound(L), etc.) */
+int64_t bkgpu_table_dict_code(const BkgTable* t, int col, const char* word,
+                              int64_t wlen, int mode);
 int64_t bkgpu_table_nrows(const BkgTable* t);
 /* Narrow the physical storage of integer-typed columns (col = -1: all)
  * whose all-rows value range fits 1/2/4 bytes to frame-of-reference
