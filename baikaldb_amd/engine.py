@@ -46,6 +46,9 @@ def _load():
     lib.bkgpu_table_dict_code.restype = C.c_int64
     lib.bkgpu_table_dict_code.argtypes = [C.c_void_p, C.c_int, C.c_char_p,
                                           C.c_int64, C.c_int]
+    lib.bkgpu_table_dict_word.restype = C.c_int
+    lib.bkgpu_table_dict_word.argtypes = [C.c_void_p, C.c_int, C.c_int64,
+                                          C.c_char_p, C.c_int]
     lib.bkgpu_table_nrows.restype = C.c_int64
     lib.bkgpu_table_nrows.argtypes = [C.c_void_p]
     lib.bkgpu_table_free.argtypes = [C.c_void_p]
@@ -303,6 +306,13 @@ class GpuEngine:
             table.handle, col, bytes(byts),
             offs.ctypes.data_as(C.POINTER(C.c_int64)), vptr),
             "upload_strings")
+
+    def dict_word(self, table, col, code, cap=256):
+        buf = C.create_string_buffer(cap)
+        if self.lib.bkgpu_table_dict_word(table.handle, col, code, buf,
+                                          cap) < 0:
+            return None
+        return buf.value.decode()
 
     def dict_code(self, table, col, word, mode=0):
         """mode 0: exact code (-1 absent); mode 1: lower_bound — turns a

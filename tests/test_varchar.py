@@ -53,14 +53,9 @@ def test_group_by_arbitrary_strings(eng):
         got = res.fetch(sorted=True)
         res.free()
         # decode group keys back to words
-        words = [None] * got["ngroups"]
-        import ctypes as C
-        buf = C.create_string_buffer(64)
-        for g in range(got["ngroups"]):
-            code = int(got["enc"][g][0])
-            assert eng.lib.bkgpu_table_dict_word(t.handle, 0, code, buf,
-                                                 64) >= 0
-            words[g] = buf.value.decode()
+        words = [eng.dict_word(t, 0, int(got["enc"][g][0]))
+                 for g in range(got["ngroups"])]
+        assert all(w is not None for w in words)
     finally:
         t.free()
     # brute force
@@ -121,14 +116,8 @@ def test_string_minmax_and_order_by(eng):
         res = eng.filter_agg(t, plan, expected_groups=1)
         got = res.fetch()
         res.free()
-        import ctypes as C
-        buf = C.create_string_buffer(64)
-        eng.lib.bkgpu_table_dict_word(t.handle, 0, int(got["agg_i"][0][0]),
-                                      buf, 64)
-        assert buf.value.decode() == min(strings)
-        eng.lib.bkgpu_table_dict_word(t.handle, 0, int(got["agg_i"][1][0]),
-                                      buf, 64)
-        assert buf.value.decode() == max(strings)
+        assert eng.dict_word(t, 0, int(got["agg_i"][0][0])) == min(strings)
+        assert eng.dict_word(t, 0, int(got["agg_i"][1][0])) == max(strings)
         # ORDER BY s LIMIT 10: rowids of the byte-wise smallest strings
         rowids = eng.sort_topk(t, [(0, 1, 1)], 10)
         expect = sorted(range(n), key=lambda i: (strings[i], i))[:10]
@@ -145,13 +134,13 @@ def test_string_nulls_and_like(eng):
     t, strings, v, valid = make_table(eng, n, rng, null_frac=0.25)
     try:
         # LIKE 'm%' via the real pattern compiler against the built dict
-        nw = 0
-        import ctypes as C
-        buf = C.create_string_buffer(64)
         words = []
-        while eng.lib.bkgpu_table_dict_word(t.handle, 0, nw, buf, 64) >= 0:
-            words.append(buf.value.decode())
-            nw += 1
+        while True:
+            w = eng.dict_word(t, 0, len(words))
+            if w is None:
+                break
+            words.append(w)
+        nw = len(words)
         codes = like_accept_codes(words, b"m%")
         accept = bytearray((nw + 7) // 8)
         for c in codes:
