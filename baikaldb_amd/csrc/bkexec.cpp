@@ -20,8 +20,10 @@
 // NO row-at-a-time CPU fallback: a plan that reaches ScanNode::get_next in
 // row mode returns an error instead of silently computing on the host.
 
+#include <algorithm>
 #include <cstdint>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <memory>
 #include <string>
@@ -179,30 +181,39 @@ public:
     int n_conjuncts() const { return _n_conjuncts; }
     const BkConjunct* conjuncts() const { return _conjuncts; }
     /* As the EFFECTIVE ROOT (SELECT without GROUP BY/ORDER BY) FilterNode
-     * emits the passing rows itself (filter_node.cpp:736-795): one GPU
-     * filter_collect for the row ids, then column gathers. All table
-     * columns become the row's slots (the child scan's tuple). */
+     * emits the passing rows itself (filter_node.cpp:736-795): the GPU
+     * filter runs over BOUNDED row-range chunks (BK_FETCH_CHUNK rows,
+     * default 4M) and get_next() streams each chunk's survivors before
+     * scanning the next range, so host memory stays O(chunk) for a
+     * filter-only SELECT over any table size — the streamed analogue of
+     * the reference scan's batch iterator (rocksdb_scan_node.cpp
+     * get_next loop), not a whole-result materialization. */
     int get_next(RuntimeState* state, RowBatch* batch, bool* eos) override;
     int n_slots();  /* lazily resolved from the scan's table */
     void close(RuntimeState* state) override {
         ExecNode::close(state);
-        _materialized = false;
+        _opened = false;
+        _eos_source = false;
+        _scan_pos = 0;
         _rowids.clear();
         _cols_i.clear(); _cols_d.clear(); _cols_n.clear();
         _iter = 0;
     }
 private:
-    int materialize(RuntimeState* state);
+    int fetch_chunk(RuntimeState* state);  /* 1 = chunk ready, 0 = drained */
     int32_t _n_conjuncts = 0;
     BkConjunct _conjuncts[BK_MAX_CONJUNCTS] = {};
-    bool _materialized = false;
+    bool _opened = false;
+    bool _eos_source = false;
     int _ncols = 0;
-    std::vector<int64_t> _rowids;
+    int64_t _nrows = 0;
+    int64_t _scan_pos = 0;
+    std::vector<int64_t> _rowids;          /* current chunk's survivors */
     std::vector<std::vector<int64_t>> _cols_i;
     std::vector<std::vector<double>> _cols_d;
     std::vector<std::vector<uint8_t>> _cols_n;
     std::vector<int32_t> _col_types;
-    int64_t _iter = 0;
+    int64_t _iter = 0;                     /* cursor within the chunk */
 };
 
 /* helpers to locate the pipeline pieces below a blocking node */
@@ -216,55 +227,76 @@ int FilterNode::n_slots() {
     return _ncols;
 }
 
-int FilterNode::materialize(RuntimeState* state) {
+static int64_t fetch_chunk_rows() {
+    /* read per chunk (once every few million rows), not cached, so a
+     * long-lived process can be re-tuned between queries */
+    const char* e = getenv("BK_FETCH_CHUNK");
+    int64_t n = e ? atoll(e) : 0;
+    return n > 0 ? n : (int64_t)(4 << 20);  /* 4M rows per chunk */
+}
+
+int FilterNode::fetch_chunk(RuntimeState* state) {
     ScanNode* scan = find_scan(this);
     if (!scan) { state->error_msg = "FilterNode: no scan below"; return -1; }
     BkgTable* t = scan->table();
+    if (!_opened) {
+        _nrows = bkgpu_table_nrows(t);
+        _ncols = bkgpu_table_ncols(t);
+        _col_types.resize(_ncols);
+        _cols_i.assign(_ncols, {});
+        _cols_d.assign(_ncols, {});
+        _cols_n.assign(_ncols, {});
+        for (int c = 0; c < _ncols; c++)
+            _col_types[c] = bkgpu_table_col_type(t, c);
+        _scan_pos = 0;
+        _opened = true;
+    }
     BkQuerySpec q{};
     q.n_conjuncts = _n_conjuncts;
     memcpy(q.conjuncts, _conjuncts, sizeof(q.conjuncts));
-    int64_t nrows = bkgpu_table_nrows(t);
-    int64_t limit = _limit > 0 ? _limit : nrows;
-    _rowids.resize(limit > 0 ? limit : 1);
-    int64_t got = bkgpu_filter_collect(t, &q, 0, nrows, limit, _rowids.data());
-    if (got < 0) { state->error_msg = bkgpu_last_error(); return -1; }
-    _rowids.resize(got);
-    /* arrival order: the reference scan emits rows in iterator order */
-    std::sort(_rowids.begin(), _rowids.end());
-    state->inc_num_scan_rows(nrows);
-    state->inc_num_filter_rows(nrows - got);
-    _ncols = bkgpu_table_ncols(t);
-    _cols_i.assign(_ncols, {});
-    _cols_d.assign(_ncols, {});
-    _cols_n.assign(_ncols, {});
-    _col_types.resize(_ncols);
-    for (int c = 0; c < _ncols; c++) {
-        _col_types[c] = bkgpu_table_col_type(t, c);
-        _cols_i[c].resize(got ? got : 1);
-        _cols_d[c].resize(got ? got : 1);
-        _cols_n[c].resize(got ? got : 1);
-        if (got > 0 && bkgpu_gather(t, c, _rowids.data(), got,
-                                    _cols_i[c].data(), _cols_d[c].data(),
-                                    _cols_n[c].data()) != 0) {
-            state->error_msg = bkgpu_last_error();
-            return -1;
+    const int64_t chunk = fetch_chunk_rows();
+    while (_scan_pos < _nrows) {
+        int64_t end = std::min(_nrows, _scan_pos + chunk);
+        int64_t span = end - _scan_pos;
+        _rowids.resize(span);
+        int64_t got = bkgpu_filter_collect(t, &q, _scan_pos, end, span,
+                                           _rowids.data());
+        if (got < 0) { state->error_msg = bkgpu_last_error(); return -1; }
+        _rowids.resize(got);
+        /* arrival order: the reference scan emits rows in iterator order;
+         * chunks advance in row order, so sorting within the chunk keeps
+         * the global stream ordered. */
+        std::sort(_rowids.begin(), _rowids.end());
+        state->inc_num_scan_rows(span);
+        state->inc_num_filter_rows(span - got);
+        _scan_pos = end;
+        if (got == 0) continue;
+        for (int c = 0; c < _ncols; c++) {
+            _cols_i[c].resize(got);
+            _cols_d[c].resize(got);
+            _cols_n[c].resize(got);
+            if (bkgpu_gather(t, c, _rowids.data(), got,
+                             _cols_i[c].data(), _cols_d[c].data(),
+                             _cols_n[c].data()) != 0) {
+                state->error_msg = bkgpu_last_error();
+                return -1;
+            }
         }
+        _iter = 0;
+        return 1;
     }
-    _iter = 0;
-    _materialized = true;
     return 0;
 }
 
 int FilterNode::get_next(RuntimeState* state, RowBatch* batch, bool* eos) {
-    if (!_materialized) {
-        int ret = materialize(state);
-        if (ret < 0) return ret;
-    }
     while (true) {
         if (state->is_cancelled()) { *eos = true; return 0; }
-        if (reached_limit() || _iter >= (int64_t)_rowids.size()) {
-            *eos = true;
-            return 0;
+        if (reached_limit()) { *eos = true; return 0; }
+        if (_iter >= (int64_t)_rowids.size()) {
+            if (_eos_source) { *eos = true; return 0; }
+            int r = fetch_chunk(state);
+            if (r < 0) return r;
+            if (r == 0) { _eos_source = true; *eos = true; return 0; }
         }
         if (batch->is_full()) return 0;
         auto row = std::make_unique<MemRow>(_ncols);
