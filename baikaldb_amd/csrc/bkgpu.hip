@@ -2542,6 +2542,106 @@ extern "C" int bkgpu_table_derive_remap(BkgTable* t, int src_col,
     return nc;
 }
 
+/* derived EXPRESSION column: dst[r] = eval_prog(row r) — the engine-side
+ * projection of an arbitrary-depth expression tree (the reference walks
+ * ScalarFnCall trees per row, scalar_fn_call.cpp:194-225) into a real
+ * table column, so WINDOW fn inputs, ORDER BY keys and out_cols can be
+ * expressions with zero changes to the window/sort kernels. One pass over
+ * the table; either-input-NULL => NULL (validity materialized only when
+ * some referenced column is nullable). */
+__global__ void k_project_prog(DevCols cols, BkQuerySpec q, int64_t n,
+                               int32_t out_type, void* dst, uint8_t* vout) {
+    int64_t gs = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += gs) {
+        PVal p = eval_prog(cols, q, 0, q.n_prog, r);
+        if (vout) vout[r] = p.valid ? 1 : 0;
+        if (out_type == BK_DOUBLE) ((double*)dst)[r] = p.valid ? p.d : 0.0;
+        else ((int64_t*)dst)[r] = p.valid ? p.i : 0;
+    }
+}
+
+extern "C" int bkgpu_table_derive_prog(BkgTable* t, const BkExprOp* prog,
+                                       int32_t len, int32_t out_type) {
+    if (!t || !prog || len <= 0 || len > BK_MAX_PROG_POOL) {
+        set_err("derive_prog: bad program");
+        return -1;
+    }
+    if (out_type != BK_INT64 && out_type != BK_DOUBLE) {
+        set_err("derive_prog: out_type must be INT64 or DOUBLE");
+        return -1;
+    }
+    if (t->ncols >= BK_MAX_COLS) { set_err("derive_prog: table full"); return -1; }
+    /* validate: stack discipline + column refs (same rules as the query-
+     * spec prog validator in bkgpu_filter_agg) */
+    bool nullable = false;
+    {
+        int sp = 0;
+        for (int32_t k = 0; k < len; k++) {
+            const BkExprOp& e = prog[k];
+            switch (e.op) {
+                case BK_PROG_COL:
+                    if (e.arg < 0 || e.arg >= t->ncols) {
+                        set_err("derive_prog: bad column ref");
+                        return -1;
+                    }
+                    if (t->valid[e.arg]) nullable = true;
+                    /* fallthrough */
+                case BK_PROG_LIT_I:
+                case BK_PROG_LIT_D:
+                    if (++sp > BK_MAX_PROG_DEPTH) {
+                        set_err("derive_prog: program too deep");
+                        return -1;
+                    }
+                    break;
+                case BK_PROG_ARITH:
+                    if (sp < 2) { set_err("derive_prog: stack underflow"); return -1; }
+                    sp--;
+                    break;
+                case BK_PROG_FN:
+                    if (sp < 1) { set_err("derive_prog: stack underflow"); return -1; }
+                    break;
+                default:
+                    set_err("derive_prog: bad opcode");
+                    return -1;
+            }
+        }
+        if (sp != 1) { set_err("derive_prog: bad program"); return -1; }
+    }
+    if (ensure_device() != 0) return -1;
+    int nc = t->ncols;
+    void* dcol = nullptr;
+    uint8_t* dval = nullptr;
+    HIP_CHECK(hipMalloc(&dcol, (size_t)t->nrows * 8 + 16));
+    if (nullable &&
+        hipMalloc((void**)&dval, (size_t)t->nrows) != hipSuccess) {
+        (void)hipFree(dcol);
+        set_err("derive_prog: oom");
+        return -1;
+    }
+    BkQuerySpec q{};
+    q.n_prog = len;
+    memcpy(q.prog, prog, sizeof(BkExprOp) * (size_t)len);
+    hipLaunchKernelGGL(k_project_prog, dim3(2048), dim3(256), 0, 0,
+                       table_cols(t), q, t->nrows, out_type, dcol, dval);
+    if (hipDeviceSynchronize() != hipSuccess) {
+        (void)hipFree(dcol);
+        if (dval) (void)hipFree(dval);
+        set_err("derive_prog: projection failed");
+        return -1;
+    }
+    t->data[nc] = dcol;
+    t->valid[nc] = dval;
+    memset(&t->specs[nc], 0, sizeof(t->specs[nc]));
+    t->specs[nc].col_type = out_type;
+    t->dict[nc] = nullptr;
+    t->stat_ok[nc] = 0;
+    t->width[nc] = 8;
+    t->base[nc] = 0;
+    t->ncols = nc + 1;
+    return nc;
+}
+
 /* ------------------------------------------------------------------ */
 /* host: aggregation                                                   */
 /* ------------------------------------------------------------------ */

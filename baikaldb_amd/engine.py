@@ -102,6 +102,9 @@ def _load():
     lib.bkgpu_table_derive_remap.argtypes = [C.c_void_p, C.c_int,
                                              C.POINTER(C.c_int32), C.c_int64,
                                              C.c_int64]
+    lib.bkgpu_table_derive_prog.restype = C.c_int
+    lib.bkgpu_table_derive_prog.argtypes = [C.c_void_p, C.c_void_p,
+                                            C.c_int32, C.c_int32]
     lib.bkgpu_filter_agg_sorted.restype = C.c_void_p
     lib.bkgpu_filter_agg_sorted.argtypes = [C.c_void_p, C.POINTER(BkQuerySpec),
                                             C.c_int64, C.c_int64]
@@ -377,6 +380,30 @@ class GpuEngine:
                 f"derive_remap: {self.lib.bkgpu_last_error().decode()}")
         table.col_types.append(table.col_types[col])
         return nc, new_words
+
+    def derive_expr(self, table, expr):
+        """Append a derived expression column = eval(expr) per row (the
+        ScalarFnCall projection, scalar_fn_call.cpp:194-225 compiled to a
+        postfix program): makes WINDOW fn inputs / ORDER BY keys / out_cols
+        expression-valued with no kernel changes. expr uses the
+        plan.compile_expr operand grammar. Returns the new column index."""
+        from .plan import BkExprOp, compile_expr, TYPE_INT64, TYPE_DOUBLE
+        pool = []
+        dom = compile_expr(expr, table.col_types, pool)
+        arr = (BkExprOp * len(pool))()
+        for i, o in enumerate(pool):
+            arr[i].op = o.get("op", 0)
+            arr[i].arg = o.get("arg", 0)
+            arr[i].domain = o.get("domain", TYPE_INT64)
+            arr[i].lit_i = o.get("lit_i", 0)
+            arr[i].lit_d = o.get("lit_d", 0.0)
+        nc = self.lib.bkgpu_table_derive_prog(
+            table.handle, C.cast(arr, C.c_void_p), len(pool), dom)
+        if nc < 0:
+            raise RuntimeError(
+                f"derive_prog: {self.lib.bkgpu_last_error().decode()}")
+        table.col_types.append(dom)
+        return nc
 
     def filter_agg_sorted(self, table, plan: QueryPlan, row_begin=0,
                           row_end=None):

@@ -85,6 +85,15 @@ int  bkgpu_table_col_width(const BkgTable* t, int col);
  * new column index, usable in GROUP BY / ORDER BY / MIN/MAX. */
 int bkgpu_table_derive_remap(BkgTable* t, int src_col, const int32_t* remap,
                              int64_t ncodes, int64_t new_ncodes);
+/* Append a derived EXPRESSION column: dst[r] = eval(prog, row r) — the
+ * projection of an arbitrary-depth expression tree (ScalarFnCall::get_value,
+ * src/expr/scalar_fn_call.cpp:194-225) compiled to a postfix program (same
+ * BkExprOp encoding as BkQuerySpec.prog). out_type is the program's result
+ * domain (BK_INT64/BK_DOUBLE, reference arg-cast rule). The new column is a
+ * first-class WINDOW fn input / ORDER BY key / out_col. Returns the new
+ * column index. */
+int bkgpu_table_derive_prog(BkgTable* t, const BkExprOp* prog, int32_t len,
+                            int32_t out_type);
 int32_t bkgpu_table_col_type(const BkgTable* t, int col);
 int32_t bkgpu_table_ncols(const BkgTable* t);
 void bkgpu_table_free(BkgTable* t);
