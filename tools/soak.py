@@ -167,7 +167,84 @@ for cs in range(DN_CASES):
         os.environ.pop("BK_DENSE", None)
 print(f"dense soak: {DN_CASES - dnfails}/{DN_CASES} ok", flush=True)
 
-total_fails = fails + wfails + sfails + dfails + dnfails
+# ---- derived expression columns (projection parity vs host numpy eval) ----
+DE_CASES = AGG_CASES // 2
+defails = 0
+for cs in range(DE_CASES):
+    rng = random.Random(150_000 + SEED_OFF + cs)
+    nrng = np.random.default_rng(150_000 + SEED_OFF + cs)
+    n = rng.choice([5000, 80_000])
+    a = nrng.integers(-(1 << 40), 1 << 40, n).astype(np.int64)
+    b = nrng.integers(-1000, 1000, n).astype(np.int64)
+    K = rng.randrange(1 << 20)
+    host = (a.astype(np.uint64) * b.astype(np.uint64)
+            + np.uint64(K)).astype(np.int64)
+    t = eng.create_table([(fz.TYPE_INT64, 0, 0, 1 << 31, 0)] * 3, n)
+    try:
+        eng.upload(t, 0, a); eng.upload(t, 1, b); eng.upload(t, 2, host)
+        nc = eng.derive_expr(t, ("add", ("mul", 0, 1), ("liti", K)))
+        plan = QueryPlan(t.col_types,
+                         aggs=[("min", 2), ("min", nc), ("max", 2),
+                               ("max", nc), ("sum", 2), ("sum", nc)])
+        r = eng.filter_agg(t, plan)
+        got = r.fetch(); r.free()
+        for j in (0, 2, 4):
+            assert got["agg_i"][j][0] == got["agg_i"][j + 1][0], j
+        rid_e = eng.sort_topk(t, [(nc, 1, 1)], 50)
+        rid_h = eng.sort_topk(t, [(2, 1, 1)], 50)
+        assert np.array_equal(np.asarray(rid_e), np.asarray(rid_h))
+    except Exception as e:
+        defails += 1
+        print(f"DERIVE-FAIL {cs}: {e}", flush=True)
+    finally:
+        t.free()
+print(f"derive soak: {DE_CASES - defails}/{DE_CASES} ok", flush=True)
+
+# ---- chunked exec-tree streaming (FilterNode BK_FETCH_CHUNK) ----
+from baikaldb_amd import exec as bx
+CE_CASES = AGG_CASES // 2
+cefails = 0
+for cs in range(CE_CASES):
+    rng = random.Random(160_000 + SEED_OFF + cs)
+    n = rng.choice([10_000, 150_000])
+    specs = [(fz.TYPE_INT64, 0, 0, 1 << 31, 0),
+             (fz.TYPE_INT64, 0, 0, 500, 0)]
+    t = eng.create_table(specs, n)
+    os.environ["BK_FETCH_CHUNK"] = str(rng.choice([997, 8192, 1 << 20]))
+    try:
+        eng.generate(t, rng.randrange(1 << 40))
+        lim = rng.choice([-1, 177, 5000])
+        lit = rng.randrange(1, 500)
+        nodes = [bx.filter_node([s[0] for s in specs], [(1, "<", lit)]),
+                 bx.scan_node(t)]
+        if lim > 0:
+            nodes.insert(0, bx.limit_node(lim))
+        tree = bx.ExecTree(nodes)
+        tree.open()
+        tags, vi, vd, nulls = tree.fetch_all(batch=rng.choice([64, 1024]))
+        tree.close()
+        # cross-check against the fused GPU aggregate of the same predicate
+        plan = QueryPlan(t.col_types, conjuncts=[(1, "<", lit)],
+                         aggs=[("count_star", -1), ("sum", 0), ("sum", 1)])
+        r = eng.filter_agg(t, plan)
+        agg = r.fetch(); r.free()
+        nmatch = int(agg["agg_i"][0][0])
+        expect_rows = min(nmatch, lim) if lim > 0 else nmatch
+        assert vi.shape[0] == expect_rows, (vi.shape, expect_rows, lim)
+        if lim <= 0:
+            with np.errstate(over="ignore"):
+                assert int(vi[:, 0].astype(np.uint64).sum()
+                           .astype(np.int64)) == int(agg["agg_i"][1][0])
+                assert int(vi[:, 1].sum()) == int(agg["agg_i"][2][0])
+    except Exception as e:
+        cefails += 1
+        print(f"CHUNK-FAIL {cs}: {e}", flush=True)
+    finally:
+        os.environ.pop("BK_FETCH_CHUNK", None)
+        t.free()
+print(f"chunk soak: {CE_CASES - cefails}/{CE_CASES} ok", flush=True)
+
+total_fails = fails + wfails + sfails + dfails + dnfails + defails + cefails
 print(f"SOAK {'PASS' if total_fails == 0 else 'FAIL'} "
       f"({total_fails} failures)", flush=True)
 sys.exit(1 if total_fails else 0)
