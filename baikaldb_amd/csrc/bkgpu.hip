@@ -3408,10 +3408,28 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
      * keys from column stats and falls through here when the shape does
      * not qualify. BK_SORTED_RANGE overrides (0 = off). */
     if (partitioned) {
+        /* the raised cap applies only when the EQ plausibly SELECTS: a
+         * low-selectivity EQ (2-valued column) over 1e9 rows would
+         * materialize ~5e8 key/rowid records before falling back. Estimate
+         * survivors from column stats (uniform assumption over the encoded
+         * span, EQ conjuncts only — range conjuncts ignored, conservative)
+         * and keep the default cap when the estimate stays huge. */
+        double est = (double)(row_end - row_begin);
         bool has_eq = false;
         for (int32_t j = 0; j < q->n_conjuncts; j++)
-            if (q->conjuncts[j].op == BK_OP_EQ) has_eq = true;
-        int64_t smax = has_eq ? 1200 * 1000 * 1000ll : 200 * 1000 * 1000;
+            if (q->conjuncts[j].op == BK_OP_EQ) {
+                has_eq = true;
+                int c = q->conjuncts[j].col;
+                if (c >= 0 && c < t->ncols && !q->conjuncts[j].arith &&
+                    !q->conjuncts[j].fn && ensure_stats(t, c) == 0 &&
+                    t->stat_ok[c]) {
+                    uint64_t span = t->stat_max[c] - t->stat_min[c];
+                    est /= (double)span + 1.0;
+                }
+            }
+        bool eq_selective = has_eq && est <= 200 * 1000 * 1000.0;
+        int64_t smax = eq_selective ? 1200 * 1000 * 1000ll
+                                    : 200 * 1000 * 1000;
         if (const char* e = getenv("BK_SORTED_RANGE")) smax = atoll(e);
         bool dense_first = use_dense &&
                            (dense_pref >= 2 ||
