@@ -1401,7 +1401,7 @@ __device__ __forceinline__ void build_record(
 
 /* pass 2: scatter AoS records into bucket-contiguous regions.
  * MUST run with the same grid/block AND row mapping as k_part_histo (the
- * BK_PART_ROWS2 ILP2 tiles) so each block sees the rows its H row counted.
+ * BK_PART_ROWS tiles) so each block sees the rows its H row counted.
  * 2 rows/lane keeps two gather+store chains in flight (the per-record LDS
  * lcur atomic and the scattered 32-64 B store are latency-bound). */
 template <int BS, bool SIMPLE>

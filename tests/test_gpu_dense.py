@@ -219,3 +219,77 @@ def test_dense_fuzz_vs_hash_path(eng, orc):
                 denom = np.abs(gh["agg_d"][a]) + np.maximum(gh["agg_i"][0], 1)
                 err = np.abs(gd["agg_d"][a] - gh["agg_d"][a])
                 assert np.all(err <= DTOL_REL * denom), (it, a, err.max())
+
+
+# ---- hot-bucket absorption (ABS eager scatter, bkdpart.inc) -------------
+# The host gates absorption on the measured per-bucket survivor histogram;
+# BK_DABS_MIN=0 forces it on any data, BK_DABS=0 disables it. Results must
+# match the oracle AND the non-absorbing run (int aggs bit-exact; f64 within
+# DTOL — absorption only reorders the f64 atomic adds).
+
+def _with_env(k, v):
+    import contextlib
+
+    @contextlib.contextmanager
+    def cm():
+        old = os.environ.get(k)
+        os.environ[k] = v
+        try:
+            yield
+        finally:
+            if old is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = old
+    return cm()
+
+
+def test_dense_absorb_skewed_vs_oracle(eng, orc):
+    """Zipf group key: the hottest bucket absorbs in-LDS during the eager
+    scatter; parity vs the oracle on all agg kinds the eager path carries."""
+    aggs = [("count_star", -1), ("sum", 2), ("sum", 3), ("avg", 3),
+            ("min", 2), ("max", 3)]
+    with _with_env("BK_DABS_MIN", "0.0"):
+        got, exp = run_both(eng, orc, BASE5, 600_000,
+                            [(0, "<", int((1 << 31) * 0.75))], [1], aggs)
+    assert_parity(got, exp, aggs, CT)
+
+
+def test_dense_absorb_two_keys_dict(eng, orc):
+    aggs = [("count_star", -1), ("sum", 2), ("avg", 3)]
+    with _with_env("BK_DABS_MIN", "0.0"):
+        got, exp = run_both(eng, orc, BASE5, 400_000,
+                            [(0, "<", int((1 << 31) * 0.8)), (2, "!=", 17)],
+                            [1, 4], aggs)
+    assert_parity(got, exp, aggs, CT)
+
+
+def test_dense_absorb_on_off_identical(eng):
+    """BK_DABS on vs off: int64 aggregates must be bit-identical (absorption
+    changes only where the adds happen, not what is added)."""
+    from baikaldb_amd import QueryPlan
+
+    def run():
+        t = eng.create_table(BASE5, 500_000)
+        try:
+            eng.generate(t, SEED)
+            plan = QueryPlan(t.col_types,
+                             conjuncts=[(0, "<", int((1 << 31) * 0.7))],
+                             group=[1],
+                             aggs=[("count_star", -1), ("sum", 2),
+                                   ("min", 2), ("max", 2)])
+            res = eng.filter_agg(t, plan, expected_groups=1 << 14)
+            try:
+                return res.fetch(sorted=True)
+            finally:
+                res.free()
+        finally:
+            t.free()
+
+    with _with_env("BK_DABS_MIN", "0.0"):
+        a = run()
+    with _with_env("BK_DABS", "0"):
+        b = run()
+    assert len(a) == len(b)
+    for ra, rb in zip(a, b):
+        assert ra == rb, (ra, rb)
