@@ -293,3 +293,30 @@ def test_dense_absorb_on_off_identical(eng):
     assert len(a) == len(b)
     for ra, rb in zip(a, b):
         assert ra == rb, (ra, rb)
+
+
+def test_dense_avg_over_wide_int64(eng, orc):
+    """AVG over a physically-8-byte INT64 column (span >= 2^32): the eager
+    record must carry (double)value bits (agg_fn_call.cpp casts AVG inputs
+    to double), not raw int bits — wmode 3 in dense_eager_plan."""
+    specs = [(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+             (TYPE_INT64, D_SKEW, 50_000, 0, 0),
+             (TYPE_INT64, D_UNI, 0, 1 << 40, 0),     # wide: no narrow store
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 0)]
+    aggs = [("count_star", -1), ("avg", 2), ("sum", 3)]
+    got, exp = run_both(eng, orc, specs, 400_000,
+                        [(0, "<", int((1 << 31) * 0.7))], [1], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])
+
+
+def test_dense_absorb_avg_wide_int64(eng, orc):
+    """Same shape with absorption forced on."""
+    specs = [(TYPE_INT64, D_UNI, 0, 1 << 31, 0),
+             (TYPE_INT64, D_SKEW, 50_000, 0, 0),
+             (TYPE_INT64, D_UNI, 0, 1 << 40, 0),
+             (TYPE_DOUBLE, D_SUM16, 0, 0, 0)]
+    aggs = [("count_star", -1), ("avg", 2), ("min", 2), ("sum", 3)]
+    with _with_env("BK_DABS_MIN", "0.0"):
+        got, exp = run_both(eng, orc, specs, 400_000,
+                            [(0, "<", int((1 << 31) * 0.7))], [1], aggs)
+    assert_parity(got, exp, aggs, [s[0] for s in specs])

@@ -32,7 +32,7 @@ def run(tag):
         ng = r.ngroups
         if rep == 2:
             rows = r.fetch(sorted=True)
-            cs = sum(hash(x[:3]) for x in rows[:5000]) & 0xFFFFFFFF
+            cs = sum(hash(tuple(x[:3])) for x in rows[:5000]) & 0xFFFFFFFF
             scount = sum(x[2] for x in rows)
             sums = (ng, len(rows), cs, scount)
         r.free()
