@@ -227,20 +227,23 @@ def test_dense_fuzz_vs_hash_path(eng, orc):
 # match the oracle AND the non-absorbing run (int aggs bit-exact; f64 within
 # DTOL — absorption only reorders the f64 atomic adds).
 
-def _with_env(k, v):
+def _with_env(*kv):
+    """_with_env(k1, v1, k2, v2, ...)"""
     import contextlib
 
     @contextlib.contextmanager
     def cm():
-        old = os.environ.get(k)
-        os.environ[k] = v
+        olds = [(k, os.environ.get(k)) for k in kv[::2]]
+        for k, v in zip(kv[::2], kv[1::2]):
+            os.environ[k] = v
         try:
             yield
         finally:
-            if old is None:
-                os.environ.pop(k, None)
-            else:
-                os.environ[k] = old
+            for k, old in olds:
+                if old is None:
+                    os.environ.pop(k, None)
+                else:
+                    os.environ[k] = old
     return cm()
 
 
@@ -249,7 +252,7 @@ def test_dense_absorb_skewed_vs_oracle(eng, orc):
     scatter; parity vs the oracle on all agg kinds the eager path carries."""
     aggs = [("count_star", -1), ("sum", 2), ("sum", 3), ("avg", 3),
             ("min", 2), ("max", 3)]
-    with _with_env("BK_DABS_MIN", "0.0"):
+    with _with_env("BK_DABS", "1", "BK_DABS_MIN", "0.0"):
         got, exp = run_both(eng, orc, BASE5, 600_000,
                             [(0, "<", int((1 << 31) * 0.75))], [1], aggs)
     assert_parity(got, exp, aggs, CT)
@@ -257,7 +260,7 @@ def test_dense_absorb_skewed_vs_oracle(eng, orc):
 
 def test_dense_absorb_two_keys_dict(eng, orc):
     aggs = [("count_star", -1), ("sum", 2), ("avg", 3)]
-    with _with_env("BK_DABS_MIN", "0.0"):
+    with _with_env("BK_DABS", "1", "BK_DABS_MIN", "0.0"):
         got, exp = run_both(eng, orc, BASE5, 400_000,
                             [(0, "<", int((1 << 31) * 0.8)), (2, "!=", 17)],
                             [1, 4], aggs)
@@ -265,8 +268,9 @@ def test_dense_absorb_two_keys_dict(eng, orc):
 
 
 def test_dense_absorb_on_off_identical(eng):
-    """BK_DABS on vs off: int64 aggregates must be bit-identical (absorption
-    changes only where the adds happen, not what is added)."""
+    """BK_DABS on vs off (off is the default): int64 aggregates must be
+    bit-identical (absorption changes only where the adds happen, not what
+    is added)."""
     from baikaldb_amd import QueryPlan
 
     def run():
@@ -286,10 +290,9 @@ def test_dense_absorb_on_off_identical(eng):
         finally:
             t.free()
 
-    with _with_env("BK_DABS_MIN", "0.0"):
+    with _with_env("BK_DABS", "1", "BK_DABS_MIN", "0.0"):
         a = run()
-    with _with_env("BK_DABS", "0"):
-        b = run()
+    b = run()
     assert len(a) == len(b)
     for ra, rb in zip(a, b):
         assert ra == rb, (ra, rb)
@@ -316,7 +319,7 @@ def test_dense_absorb_avg_wide_int64(eng, orc):
              (TYPE_INT64, D_UNI, 0, 1 << 40, 0),
              (TYPE_DOUBLE, D_SUM16, 0, 0, 0)]
     aggs = [("count_star", -1), ("avg", 2), ("min", 2), ("sum", 3)]
-    with _with_env("BK_DABS_MIN", "0.0"):
+    with _with_env("BK_DABS", "1", "BK_DABS_MIN", "0.0"):
         got, exp = run_both(eng, orc, specs, 400_000,
                             [(0, "<", int((1 << 31) * 0.7))], [1], aggs)
     assert_parity(got, exp, aggs, [s[0] for s in specs])
