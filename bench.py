@@ -403,3 +403,11 @@ def main():
 
 if __name__ == "__main__":
     main()
+    # clean rendezvous shutdown (all ranks reach here: rank!=0 returns
+    # early from main after the timed loop's final barrier)
+    try:
+        import torch.distributed as _td
+        if _td.is_initialized():
+            _td.destroy_process_group()
+    except Exception:
+        pass
