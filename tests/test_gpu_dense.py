@@ -323,3 +323,62 @@ def test_dense_absorb_avg_wide_int64(eng, orc):
         got, exp = run_both(eng, orc, specs, 400_000,
                             [(0, "<", int((1 << 31) * 0.7))], [1], aggs)
     assert_parity(got, exp, aggs, [s[0] for s in specs])
+
+
+# ---- PACK32 split records (BK_DREC_PACK=1, measured-dead default off) ----
+# The eager scatter can split the record into a u32 hdr stream (did + narrow
+# fields bit-packed) + an aligned wide payload when the bit widths fit; it
+# measured slower than the 24-B AoS record and defaults off, but the layout
+# stays covered: results must be identical to the oracle and to the AoS run.
+
+
+def test_dense_pack32_all_aggs_vs_oracle(eng, orc):
+    """Single skewed key (17 bits) + narrow sum field (10 bits): PACK32
+    eligible; parity vs the oracle on all agg kinds the eager path carries."""
+    aggs = [("count_star", -1), ("sum", 2), ("sum", 3), ("avg", 3),
+            ("min", 2), ("max", 3)]
+    with _with_env("BK_DREC_PACK", "1"):
+        got, exp = run_both(eng, orc, BASE5, 600_000,
+                            [(0, "<", int((1 << 31) * 0.75))], [1], aggs)
+    assert_parity(got, exp, aggs, CT)
+
+
+def test_dense_pack32_two_keys_bit_boundary(eng, orc):
+    """Two keys (10+12 did bits) + one 10-bit narrow field = exactly 32
+    hdr bits — the eligibility boundary."""
+    aggs = [("count_star", -1), ("sum", 2), ("avg", 3), ("max", 2)]
+    with _with_env("BK_DREC_PACK", "1"):
+        got, exp = run_both(eng, orc, BASE5, 400_000,
+                            [(0, "<", int((1 << 31) * 0.8))],
+                            [2, 4], aggs)
+    assert_parity(got, exp, aggs, CT)
+
+
+def test_dense_pack32_on_off_identical(eng):
+    """PACK32 on vs off: int64 aggregates must be bit-identical (the split
+    layout changes where record bytes live, not what is aggregated)."""
+    from baikaldb_amd import QueryPlan
+
+    def run():
+        t = eng.create_table(BASE5, 500_000)
+        try:
+            eng.generate(t, SEED)
+            plan = QueryPlan(t.col_types,
+                             conjuncts=[(0, "<", int((1 << 31) * 0.7))],
+                             group=[1],
+                             aggs=[("count_star", -1), ("sum", 2),
+                                   ("min", 2), ("max", 2)])
+            res = eng.filter_agg(t, plan, expected_groups=1 << 14)
+            try:
+                return res.fetch(sorted=True)
+            finally:
+                res.free()
+        finally:
+            t.free()
+
+    with _with_env("BK_DREC_PACK", "1"):
+        a = run()
+    b = run()
+    assert len(a) == len(b)
+    for ra, rb in zip(a, b):
+        assert ra == rb, (ra, rb)
