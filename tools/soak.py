@@ -137,12 +137,16 @@ dnfails = 0
 for cs in range(DN_CASES):
     rng = random.Random(130_000 + SEED_OFF + cs)
     span1 = rng.choice([50, 3000, 200_000])
-    span2 = rng.choice([4, 100, 4000])
+    group = [1] if rng.random() < 0.5 else [1, 2]
+    # col2 doubles as group key (narrow spans) or agg input; when it is
+    # agg-only, sometimes make it WIDE (span >= 2^32) so the eager record's
+    # wide-int64 field modes (wmode 1 MIN/MAX, wmode 3 AVG-cast — the class
+    # that carried a real parity bug this round) stay fuzzed
+    span2 = rng.choice([4, 100, 4000] + ([1 << 41] if group == [1] else []))
     specs = [(fz.TYPE_INT64, 0, 0, 1 << 31, 0),
              (fz.TYPE_INT64, 1, span1, 0, 0),
              (fz.TYPE_INT64, 0, 0, span2, 0),
              (fz.TYPE_DOUBLE, 3, 0, 0, 0)]
-    group = [1] if rng.random() < 0.5 else [1, 2]
     aggs = [("count_star", -1)]
     for _ in range(rng.randint(1, 4)):
         aggs.append((rng.choice(["sum", "avg", "min", "max", "count"]),
