@@ -332,21 +332,35 @@ def test_dense_absorb_avg_wide_int64(eng, orc):
 # stays covered: results must be identical to the oracle and to the AoS run.
 
 
-def test_dense_pack32_all_aggs_vs_oracle(eng, orc):
-    """Single skewed key (17 bits) + narrow sum field (10 bits): PACK32
-    eligible; parity vs the oracle on all agg kinds the eager path carries."""
+def test_dense_pack32_sum_narrow_vs_oracle(eng, orc):
+    """Single skewed key (17 did bits) + one 10-bit narrow SUM field + wide
+    f64 fields: 27 hdr bits -> PACK32 engages; parity vs the oracle."""
     aggs = [("count_star", -1), ("sum", 2), ("sum", 3), ("avg", 3),
-            ("min", 2), ("max", 3)]
+            ("max", 3)]
     with _with_env("BK_DREC_PACK", "1"):
         got, exp = run_both(eng, orc, BASE5, 600_000,
                             [(0, "<", int((1 << 31) * 0.75))], [1], aggs)
     assert_parity(got, exp, aggs, CT)
 
 
+def test_dense_pack32_minmax_narrow(eng, orc):
+    """Narrow MIN/MAX fields carry encoding deltas in the hdr: 10-bit key +
+    17-bit min field (27 bits), then 10-bit key + 17-bit max (each <= 32)."""
+    with _with_env("BK_DREC_PACK", "1"):
+        aggs = [("count_star", -1), ("min", 1), ("sum", 3)]
+        got, exp = run_both(eng, orc, BASE5, 400_000,
+                            [(0, "<", int((1 << 31) * 0.8))], [2], aggs)
+        assert_parity(got, exp, aggs, CT)
+        aggs = [("count_star", -1), ("max", 1), ("avg", 3)]
+        got, exp = run_both(eng, orc, BASE5, 400_000,
+                            [(0, ">", int((1 << 31) * 0.3))], [2], aggs)
+        assert_parity(got, exp, aggs, CT)
+
+
 def test_dense_pack32_two_keys_bit_boundary(eng, orc):
     """Two keys (10+12 did bits) + one 10-bit narrow field = exactly 32
     hdr bits — the eligibility boundary."""
-    aggs = [("count_star", -1), ("sum", 2), ("avg", 3), ("max", 2)]
+    aggs = [("count_star", -1), ("sum", 2), ("avg", 3)]
     with _with_env("BK_DREC_PACK", "1"):
         got, exp = run_both(eng, orc, BASE5, 400_000,
                             [(0, "<", int((1 << 31) * 0.8))],
