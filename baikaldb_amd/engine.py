@@ -15,6 +15,21 @@ _HERE = os.path.dirname(os.path.abspath(__file__))
 _LIB = os.path.join(_HERE, "libbkgpu.so")
 
 
+def substr_ref(w: str, start: int, ln=None) -> str:
+    """SQL SUBSTR exactly as /root/reference/src/expr/internal_functions.cpp
+    substr(): 1-based; start<0 counts from the string end (pos = size+start,
+    no extra -1); start 0 / past-end -> empty; len<=0 -> empty; the 2-arg
+    form (ln=None) runs to the end."""
+    pos = len(w) + start if start < 0 else start - 1
+    if pos < 0 or pos >= len(w):
+        return ""
+    if ln is None:
+        return w[pos:]
+    if ln <= 0:
+        return ""
+    return w[pos:pos + ln]
+
+
 class NativeEngineMissing(RuntimeError):
     pass
 
@@ -363,8 +378,9 @@ class GpuEngine:
         (new_col_index, new_words). fn: a STRING_FNS name, a callable, or
         ("substr", start, len) mirroring internal_functions.cpp substr."""
         if isinstance(fn, tuple) and fn[0] == "substr":
-            start, ln = fn[1], fn[2]
-            f = lambda w: w[start - 1:start - 1 + ln]  # 1-based, substr()
+            start = fn[1]
+            ln = fn[2] if len(fn) > 2 else None
+            f = lambda w: substr_ref(w, start, ln)
         elif callable(fn):
             f = fn
         else:

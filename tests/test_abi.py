@@ -46,3 +46,29 @@ def test_engine_refuses_without_gpu():
     from baikaldb_amd import GpuEngine, NativeEngineMissing
     with pytest.raises((NativeEngineMissing, RuntimeError)):
         GpuEngine()
+
+
+def test_substr_reference_semantics():
+    """substr_ref restates internal_functions.cpp substr() exactly: 1-based
+    positions, negative start counts from the end (pos = size+start), start
+    0 or past-end -> empty, len<=0 -> empty, 2-arg form runs to the end.
+    (The old remap lambda mishandled start<=0 — real parity fix.)"""
+    from baikaldb_amd.engine import substr_ref
+
+    s = "abcdef"
+    # (start, len, expected) — MySQL SUBSTRING semantics as the reference
+    # implements them
+    cases = [
+        (1, 3, "abc"), (2, 2, "bc"), (6, 1, "f"), (6, 10, "f"),
+        (7, 1, ""), (100, 5, ""),
+        (0, 3, ""),                      # pos 0: --pos -> -1 -> empty
+        (-1, 1, "f"), (-3, 2, "de"), (-6, 2, "ab"),
+        (-7, 2, ""),                     # size+start < 0 -> empty
+        (1, 0, ""), (1, -2, ""),         # len <= 0 -> empty
+        (3, None, "cdef"), (-2, None, "ef"), (0, None, ""),
+    ]
+    for start, ln, exp in cases:
+        got = substr_ref(s, start, ln)
+        assert got == exp, (start, ln, got, exp)
+    assert substr_ref("", 1, 1) == ""
+    assert substr_ref("", -1, None) == ""

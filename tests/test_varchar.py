@@ -160,3 +160,4 @@ def test_string_nulls_and_like(eng):
         assert got["agg_i"][0][0] == expect
     finally:
         t.free()
+
