@@ -363,11 +363,21 @@ class GpuEngine:
         return AggResult(self, h, plan)
 
     # unary string scalar fns the engine compiles to dict remaps
-    # (internal_functions.cpp upper/lower/substr via fn_manager.cpp:97-137)
+    # (internal_functions.cpp upper/lower/reverse/substr via
+    # fn_manager.cpp:97-137). The reference transforms BYTES (::toupper /
+    # ::tolower per byte, std::reverse on bytes) — so upper/lower touch only
+    # ASCII letters (UTF-8 continuation bytes pass through, matching C
+    # tolower), and reverse is a byte reversal (surrogateescape keeps
+    # non-UTF-8 results round-trippable, as the reference's raw bytes are).
+    _UP = str.maketrans("abcdefghijklmnopqrstuvwxyz",
+                        "ABCDEFGHIJKLMNOPQRSTUVWXYZ")
+    _LO = str.maketrans("ABCDEFGHIJKLMNOPQRSTUVWXYZ",
+                        "abcdefghijklmnopqrstuvwxyz")
     STRING_FNS = {
-        "upper": lambda w: w.upper(),
-        "lower": lambda w: w.lower(),
-        "reverse": lambda w: w[::-1],
+        "upper": lambda w: w.translate(GpuEngine._UP),
+        "lower": lambda w: w.translate(GpuEngine._LO),
+        "reverse": lambda w: w.encode("utf-8", "surrogateescape")[::-1]
+                              .decode("utf-8", "surrogateescape"),
     }
 
     def derive_string_fn(self, table, col, fn, words):

@@ -72,3 +72,24 @@ def test_substr_reference_semantics():
         assert got == exp, (start, ln, got, exp)
     assert substr_ref("", 1, 1) == ""
     assert substr_ref("", -1, None) == ""
+
+
+def test_string_fn_byte_semantics():
+    """upper/lower/reverse match internal_functions.cpp byte-wise behavior:
+    ::toupper/::tolower touch only ASCII letters (non-ASCII UTF-8 bytes pass
+    through), reverse reverses BYTES (not codepoints)."""
+    from baikaldb_amd.engine import GpuEngine
+
+    up = GpuEngine.STRING_FNS["upper"]
+    lo = GpuEngine.STRING_FNS["lower"]
+    rv = GpuEngine.STRING_FNS["reverse"]
+    assert up("aBc9_z") == "ABC9_Z"
+    assert lo("AbC9_Z") == "abc9_z"
+    # non-ASCII letters are NOT case-mapped (C locale tolower on bytes)
+    assert up("café") == "CAFé"
+    assert lo("CAFÉ") == "cafÉ"
+    assert rv("abc") == "cba"
+    # byte reversal of multibyte input mirrors std::reverse on the raw
+    # bytes: the result's bytes are exactly the reversed input bytes
+    got = rv("aé").encode("utf-8", "surrogateescape")
+    assert got == "aé".encode("utf-8")[::-1]
