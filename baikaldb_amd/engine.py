@@ -17,17 +17,50 @@ _LIB = os.path.join(_HERE, "libbkgpu.so")
 
 def substr_ref(w: str, start: int, ln=None) -> str:
     """SQL SUBSTR exactly as /root/reference/src/expr/internal_functions.cpp
-    substr(): 1-based; start<0 counts from the string end (pos = size+start,
-    no extra -1); start 0 / past-end -> empty; len<=0 -> empty; the 2-arg
-    form (ln=None) runs to the end."""
-    pos = len(w) + start if start < 0 else start - 1
-    if pos < 0 or pos >= len(w):
+    substr(): 1-based BYTE positions (std::string indexing); start<0 counts
+    from the string end (pos = size+start, no extra -1); start 0 / past-end
+    -> empty; len<=0 -> empty; the 2-arg form (ln=None) runs to the end.
+    Byte slices that split a multibyte char survive via surrogateescape
+    (the reference's raw bytes do the same)."""
+    b = w.encode("utf-8", "surrogateescape")
+    pos = len(b) + start if start < 0 else start - 1
+    if pos < 0 or pos >= len(b):
         return ""
     if ln is None:
-        return w[pos:]
-    if ln <= 0:
+        out = b[pos:]
+    elif ln <= 0:
         return ""
-    return w[pos:pos + ln]
+    else:
+        out = b[pos:pos + ln]
+    return out.decode("utf-8", "surrogateescape")
+
+
+def resolve_string_fn(fn, named=None):
+    """Resolve a derive_string_fn spec to a word->word callable, restating
+    /root/reference/src/expr/internal_functions.cpp semantics (all positions
+    are BYTE positions, as std::string indexes):
+    - ("substr", start[, len]): substr_ref above
+    - ("left", len): first len bytes; len<=0 -> ""
+    - ("right", len): last len bytes (whole string when len > size);
+      len<=0 -> ""
+    - a callable passes through; a str looks up the named map."""
+    if isinstance(fn, tuple) and fn[0] == "substr":
+        start = fn[1]
+        ln = fn[2] if len(fn) > 2 else None
+        return lambda w: substr_ref(w, start, ln)
+    if isinstance(fn, tuple) and fn[0] == "left":
+        ln = fn[1]
+        return (lambda w: "") if ln <= 0 else (
+            lambda w: w.encode("utf-8", "surrogateescape")[:ln]
+                       .decode("utf-8", "surrogateescape"))
+    if isinstance(fn, tuple) and fn[0] == "right":
+        ln = fn[1]
+        return (lambda w: "") if ln <= 0 else (
+            lambda w: w.encode("utf-8", "surrogateescape")[-ln:]
+                       .decode("utf-8", "surrogateescape"))
+    if callable(fn):
+        return fn
+    return named[fn]
 
 
 class NativeEngineMissing(RuntimeError):
@@ -386,15 +419,9 @@ class GpuEngine:
         (order-preserving new codes), append a derived dict column with
         newcode = remap[oldcode] (bkgpu_table_derive_remap). Returns
         (new_col_index, new_words). fn: a STRING_FNS name, a callable, or
-        ("substr", start, len) mirroring internal_functions.cpp substr."""
-        if isinstance(fn, tuple) and fn[0] == "substr":
-            start = fn[1]
-            ln = fn[2] if len(fn) > 2 else None
-            f = lambda w: substr_ref(w, start, ln)
-        elif callable(fn):
-            f = fn
-        else:
-            f = self.STRING_FNS[fn]
+        ("substr", start[, len]) / ("left", len) / ("right", len) mirroring
+        internal_functions.cpp (byte positions, like std::string)."""
+        f = resolve_string_fn(fn, self.STRING_FNS)
         tw = [f(w) for w in words]
         new_words = sorted(set(tw))
         idx = {w: i for i, w in enumerate(new_words)}

@@ -93,3 +93,27 @@ def test_string_fn_byte_semantics():
     # bytes: the result's bytes are exactly the reversed input bytes
     got = rv("aé").encode("utf-8", "surrogateescape")
     assert got == "aé".encode("utf-8")[::-1]
+
+
+def test_substr_left_right_byte_positions():
+    """substr/left/right index BYTES like std::string (multibyte chars can
+    split; surrogateescape keeps the raw bytes) — via the same
+    resolve_string_fn the dict remap uses."""
+    from baikaldb_amd.engine import substr_ref, resolve_string_fn
+
+    # "a\u00e9" = bytes 61 C3 A9: substr(2,2) -> C3 A9 = "\u00e9"
+    assert substr_ref("a\u00e9", 2, 2) == "\u00e9"
+    # split the multibyte char: substr(2,1) -> the lone C3 byte
+    assert substr_ref("a\u00e9", 2, 1).encode("utf-8", "surrogateescape") \
+        == b"\xc3"
+    assert substr_ref("a\u00e9", -2, None) == "\u00e9"
+
+    for fn, w, exp_bytes in [
+        (("left", 2), "a\u00e9", b"a\xc3"),
+        (("right", 2), "\u00e9a", b"\xa9a"),
+        (("left", 0), "abc", b""),
+        (("right", 99), "abc", b"abc"),
+        (("substr", 2, 2), "abcd", b"bc"),
+    ]:
+        f = resolve_string_fn(fn)
+        assert f(w).encode("utf-8", "surrogateescape") == exp_bytes, (fn, w)
