@@ -183,3 +183,12 @@ def test_topk_dense_collect_stage(eng, orc):
                         [(0, 1, 1), (1, 0, 0)], n)
     import numpy as np
     assert np.array_equal(got, exp)
+
+
+def test_sort_filter_double_conjunct_generic(eng, orc):
+    """A DOUBLE-typed conjunct is non-SIMPLE, so the scan routes through the
+    generic CJ=2 k_topk_scan variant (the other filtered tests take the
+    padded-SIMPLE CJ=1 path)."""
+    got, exp = run_both(eng, orc, BASE, 120_000, [(0, 1, 1)], 2000,
+                        conjuncts=[(2, "<", 0.3)])
+    assert np.array_equal(got, exp)
