@@ -3397,16 +3397,20 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
     int dense_pref = 1;
     if (const char* e = getenv("BK_DENSE")) dense_pref = atoi(e);
 
-    /* small/mid ranges: the sort-dedup path beats the partitioned pipeline
-     * (histo+scatter+agg fixed passes dominate at low survivor volume; a
-     * 1e8-row config2-shaped query measured ~3x faster sorted). An EQUALITY
-     * conjunct raises the range bound: it marks a selective scan (the
-     * reference's index selector uses exactly this signal,
-     * src/physical_plan/index_selector.cpp) and the sorted path's cost
-     * scales with SURVIVORS, not rows (1e9-row 8-int64 c1<K AND c2=K2:
-     * sorted 7.6 ms vs partitioned 12.4 ms). The sorted path auto-packs
-     * keys from column stats and falls through here when the shape does
-     * not qualify. BK_SORTED_RANGE overrides (0 = off). */
+    /* Sort-dedup routing, re-measured at round-2 close: the DENSE pipeline
+     * now beats the sorted path at every plain-GROUP-BY size once a shape
+     * is dense-eligible (c2a shape, same box: 3e7 rows 0.50 vs 1.25 ms,
+     * 1e8 0.68 vs 1.63, 2e8 0.94 vs 2.21; at the 1e9 c2b EQ-selective
+     * shape they tie, 3.018 vs 3.02) — the round-2 staged-load dense work
+     * obsoleted the earlier "sorted ~3x faster at 1e8" measurement this
+     * block was built on. Dense-eligible shapes therefore go dense
+     * outright; the sorted path remains the route for shapes dense cannot
+     * take (wide key spans, nullable keys) within its range caps, and the
+     * designed path for high-cardinality DISTINCT level 1. An EQUALITY
+     * conjunct still raises the sorted range bound for those shapes (the
+     * reference's index selector uses the same signal,
+     * src/physical_plan/index_selector.cpp), gated on a survivor estimate
+     * from column stats. BK_SORTED_RANGE overrides (0 = off). */
     if (partitioned) {
         /* the raised cap applies only when the EQ plausibly SELECTS: a
          * low-selectivity EQ (2-valued column) over 1e9 rows would
@@ -3431,9 +3435,7 @@ extern "C" BkgAggOut* bkgpu_filter_agg(BkgTable* t, const BkQuerySpec* q,
         int64_t smax = eq_selective ? 1200 * 1000 * 1000ll
                                     : 200 * 1000 * 1000;
         if (const char* e = getenv("BK_SORTED_RANGE")) smax = atoll(e);
-        bool dense_first = use_dense &&
-                           (dense_pref >= 2 ||
-                            row_end - row_begin > 200 * 1000 * 1000);
+        bool dense_first = use_dense && dense_pref >= 1;
         if (!dense_first && row_end - row_begin <= smax) {
             BkgAggOut* so = bkgpu_filter_agg_sorted(t, q, row_begin, row_end);
             if (so) { delete o; return so; }
