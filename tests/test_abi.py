@@ -143,3 +143,20 @@ def test_exports_cover_all_capi_headers():
             if not hasattr(lib, s):
                 missing.append(f"{hdr}:{s}")
     assert not missing, missing
+
+
+def test_headers_compile_standalone_as_c():
+    """Every include/*.h is a self-contained C header (the FFI boundary a
+    cgo/JNI/ctypes consumer includes with no C++ toolchain)."""
+    import glob
+    import tempfile
+    for h in sorted(glob.glob(os.path.join(REPO, "include", "*.h"))):
+        with tempfile.NamedTemporaryFile("w", suffix=".c", delete=False) as f:
+            f.write(f'#include "{h}"\n')
+            path = f.name
+        try:
+            r = subprocess.run(["gcc", "-fsyntax-only", "-I", REPO, path],
+                               capture_output=True, text=True)
+            assert r.returncode == 0, (h, r.stderr[:400])
+        finally:
+            os.unlink(path)
